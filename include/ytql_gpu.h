@@ -1,0 +1,283 @@
+/* ytql_gpu.h — C-ABI drop-in boundary for the YTsaurus dynamic-table query
+ * engine hot path (scan → filter → hash-aggregate) on AMD MI355X (gfx950).
+ *
+ * This header declares the seam that replaces, for this path:
+ *   - IEvaluator::Run                (reference: yt/yt/library/query/engine_api/evaluator.h;
+ *                                     impl yt/yt/library/query/engine/evaluator.cpp:51-105)
+ *   - TCGQuerySignature              (engine_api/evaluation_helpers.h:330)
+ *   - TExecutionContext              (engine_api/evaluation_helpers.h:251-280)
+ *   - TQueryStatistics               (yt/yt/client/query_client/query_statistics.h:49-79)
+ *   - the columnar-batch ingress     (yt/yt/client/table_client/row_batch.h:39-202)
+ *   - the unversioned column writer  (yt/yt/ytlib/table_chunk_format/integer_column_writer.cpp)
+ *     (synthetic-chunk generator; encoded bytes match the reference formats)
+ *
+ * Plain pointers and sizes only; no C++ or torch types cross this boundary.
+ * Errors: non-zero status + message copied into the caller's buffer
+ * (the reference uses C++ exceptions; TInterruptedIncompleteException maps to
+ * statistics.incomplete_output = 1 with status YT_OK, mirroring
+ * engine/cg_routines/registry.cpp:1902-1907).
+ */
+#ifndef YTQL_GPU_H
+#define YTQL_GPU_H
+
+#include <stddef.h>
+#include <stdint.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+/* ---- status codes ---- */
+enum {
+    YT_OK = 0,
+    YT_ERR_INVALID_PLAN = 1,
+    YT_ERR_INVALID_CHUNK = 2,
+    YT_ERR_CAPACITY = 3,      /* caller-provided output buffer too small */
+    YT_ERR_NO_GPU = 4,        /* HIP device unavailable — NO CPU fallback exists */
+    YT_ERR_HIP = 5,           /* HIP runtime error (message has details) */
+    YT_ERR_LIMIT = 6,         /* hard limit exceeded (non-interrupt kind) */
+    YT_ERR_UNSUPPORTED = 7,
+    YT_ERR_DIV_ZERO = 8,      /* integer division/modulo by zero in an expression */
+};
+
+/* ---- value model ----
+ * 16-byte mirror of TUnversionedValue (client/table_client/unversioned_value.h)
+ * / TPIValue (engine_api/position_independent_value.h:37-66). */
+enum {
+    YT_VT_NULL = 0x02,    /* EValueType::Null */
+    YT_VT_INT64 = 0x03,
+    YT_VT_UINT64 = 0x04,
+    YT_VT_DOUBLE = 0x05,
+    YT_VT_BOOLEAN = 0x06,
+    YT_VT_STRING = 0x10,
+};
+
+typedef struct YtValue {
+    uint16_t id;
+    uint8_t type;     /* YT_VT_* */
+    uint8_t flags;
+    uint32_t length;  /* string length */
+    union {
+        int64_t i64;
+        uint64_t u64;
+        double dbl;
+        const char* str;
+        uint64_t bits;
+    } data;
+} YtValue;
+
+/* ---- encoded column segments ----
+ * Byte layouts identical to the reference's unversioned segment formats
+ * (table_chunk_format/integer_column_writer.cpp, private.h:25-30,
+ *  floating_point_column_writer.cpp). */
+enum {
+    YT_SEG_DICTIONARY_RLE = 0,
+    YT_SEG_DICTIONARY_DENSE = 1,
+    YT_SEG_DIRECT_RLE = 2,
+    YT_SEG_DIRECT_DENSE = 3,
+    YT_SEG_DOUBLE = 16,       /* unversioned double segment (single layout) */
+};
+
+typedef struct YtSegment {
+    int32_t type;         /* YT_SEG_* */
+    int32_t row_count;    /* rows covered by this segment (<= 128*1024) */
+    uint64_t min_value;   /* TIntegerSegmentMeta.min_value (zigzag space) */
+    const void* data;     /* segment blob (device ptr for GPU entries, host ptr for oracle/encode) */
+    int64_t data_size;    /* bytes */
+} YtSegment;
+
+typedef struct YtColumn {
+    int32_t value_type;   /* YT_VT_INT64 / YT_VT_UINT64 / YT_VT_DOUBLE / YT_VT_BOOLEAN */
+    int32_t segment_count;
+    const YtSegment* segments;
+} YtColumn;
+
+typedef struct YtChunk {
+    int64_t row_count;
+    int32_t column_count;
+    const YtColumn* columns;
+} YtChunk;
+
+/* ---- plan ----
+ * Restates TQuery{WhereClause,GroupClause} (base/query.h:483-578) for the
+ * scan→filter→group-by shape; expression grammar covers the hot-path tests.
+ */
+enum {
+    YT_EX_COLUMN = 0,     /* leaf: input column[col] */
+    YT_EX_LIT_I64 = 1,
+    YT_EX_LIT_NULL = 2,
+    YT_EX_LIT_DOUBLE = 3,
+    YT_EX_ADD = 10, YT_EX_SUB = 11, YT_EX_MUL = 12, YT_EX_DIV = 13, YT_EX_MOD = 14,
+    YT_EX_EQ = 20, YT_EX_NE = 21, YT_EX_LT = 22, YT_EX_LE = 23, YT_EX_GT = 24, YT_EX_GE = 25,
+    YT_EX_AND = 30, YT_EX_OR = 31, YT_EX_NOT = 32,
+};
+
+typedef struct YtExpr {
+    int32_t op;           /* YT_EX_* */
+    int32_t col;          /* for YT_EX_COLUMN */
+    int64_t lit_i64;      /* for YT_EX_LIT_I64 */
+    double lit_dbl;       /* for YT_EX_LIT_DOUBLE */
+    const struct YtExpr* a;
+    const struct YtExpr* b;
+} YtExpr;
+
+enum {
+    YT_AGG_SUM = 0,       /* udf/sum.c — null-propagating add; int64 wraps mod 2^64 */
+    YT_AGG_SUM1 = 1,      /* sum(1) == count; QL has no count aggregate (ql_query_ut.cpp:3200) */
+    YT_AGG_MIN = 2,       /* udf/min.c */
+    YT_AGG_MAX = 3,       /* udf/max.c */
+};
+
+typedef struct YtAgg {
+    int32_t func;         /* YT_AGG_* */
+    const YtExpr* arg;    /* NULL for YT_AGG_SUM1 */
+} YtAgg;
+
+typedef struct YtPlan {
+    const YtExpr* filter;         /* NULL = no WHERE clause */
+    int32_t key_count;            /* 0 + agg_count>0 = global aggregate (one group) */
+    const YtExpr* const* keys;
+    int32_t agg_count;
+    const YtAgg* const* aggs;
+    int32_t project_count;        /* used when agg_count == 0 (plain scan): projected exprs */
+    const YtExpr* const* projects;
+    int32_t is_merge;             /* front-query mode: input = [keys..., states...] rows
+                                     (cg_fragment_compiler.cpp:4016-4124) */
+} YtPlan;
+
+/* ---- execution context / statistics ----
+ * Mirrors TExecutionContext limits and TQueryStatistics counters. */
+typedef struct YtExecOptions {
+    int64_t input_row_limit;      /* 0 = unlimited */
+    int64_t output_row_limit;
+    int64_t group_row_limit;
+    int32_t device;               /* HIP device ordinal */
+    uint64_t stream;              /* hipStream_t as integer; 0 = default stream */
+    int64_t max_groups_hint;      /* sizes the device hash table; 0 = default */
+} YtExecOptions;
+
+typedef struct YtStatistics {
+    int64_t rows_read;
+    int64_t data_weight_read;     /* encoded bytes consumed */
+    int64_t rows_written;
+    int64_t grouped_row_count;
+    int32_t incomplete_input;
+    int32_t incomplete_output;
+    double decode_time_ms;        /* host wall around device pipeline */
+    double execute_time_ms;
+    /* device-event timings of the dominant kernels (for roofline evidence) */
+    double kernel_scan_ms;        /* total GPU time in scan_*_agg kernels */
+    int64_t kernel_scan_launches;
+    double kernel_other_ms;
+} YtStatistics;
+
+/* ---- result rowset ----
+ * Caller provides capacity; library writes row-major 16-byte YtValues.
+ * (Matches the writer seam IUnversionedRowsetWriter::Write,
+ *  client/table_client/unversioned_writer.h:21-32.) */
+typedef struct YtRowset {
+    YtValue* values;              /* capacity_rows * column_count values, row-major */
+    int64_t capacity_rows;
+    int64_t row_count;            /* out */
+    int32_t column_count;         /* out */
+    char* string_pool;            /* optional pool for string payloads */
+    int64_t string_pool_capacity;
+    int64_t string_pool_used;     /* out */
+} YtRowset;
+
+/* =========================== entry points =========================== */
+
+/* Library/device probe. Returns YT_OK when a gfx950 HIP device is usable. */
+int yt_gpu_available(char* errbuf, size_t errlen);
+
+/* The evaluator seam (replaces IEvaluator::Run for this plan shape).
+ * Chunk segment data pointers must be DEVICE pointers (HBM-resident);
+ * output rowset buffers are HOST memory. Synchronous on `options->stream`. */
+int yt_gpu_query_execute(
+    const YtPlan* plan,
+    const YtChunk* chunk,
+    const YtExecOptions* options,
+    YtRowset* output,
+    YtStatistics* stats,
+    char* errbuf, size_t errlen);
+
+/* Two-phase (coordinated) path, mirroring the bottom/front query split
+ * (engine_api/coordinator.h:26,78,89; shuffle engine_api/shuffling_reader.cpp:21-88).
+ *
+ * yt_gpu_query_partial: bottom query — scan+filter+group into partial states,
+ * then hash-partition state rows by hash(group key) % partition_count into
+ * caller's DEVICE buffer `states` (capacity_rows state rows). A state row is
+ * (key_count + 3*agg_count... ) — packed as YtStateRow records below.
+ * part_offsets/part_counts (length partition_count, host) describe the layout.
+ */
+/* Round-1 state-row record: the {one int64/boolean key, sum(expr), sum(1)}
+ * plan family (BASELINE configs 3/4). Other shapes return YT_ERR_UNSUPPORTED
+ * from the two-phase entries for now. 32 bytes, device-native layout. */
+typedef struct YtStateRow {
+    uint64_t key_bits;               /* raw key bits (i64), undefined when key null */
+    uint64_t meta;                   /* bit0 = key is null; bits 8..63 = nonnull count of sum arg */
+    uint64_t sum_bits;               /* wrapping int64 sum of the sum() argument */
+    uint64_t row_count;              /* rows in group == sum(1) state */
+} YtStateRow;
+
+int yt_gpu_query_partial(
+    const YtPlan* plan,
+    const YtChunk* chunk,
+    const YtExecOptions* options,
+    int32_t partition_count,
+    void* states_device,             /* device buffer for partitioned YtStateRow[] */
+    int64_t capacity_rows,
+    int64_t* part_counts,            /* out, host, length partition_count */
+    YtStatistics* stats,
+    char* errbuf, size_t errlen);
+
+/* front query: merge state rows (device buffer, e.g. post all-to-all) and
+ * finalize into `output` (host). Implements Merge+finalize semantics of
+ * cg_fragment_compiler.cpp:4116-4134 + udf/sum.c:47-65. */
+int yt_gpu_merge_states(
+    const YtPlan* plan,
+    const void* states_device,
+    int64_t state_row_count,
+    const YtExecOptions* options,
+    YtRowset* output,
+    YtStatistics* stats,
+    char* errbuf, size_t errlen);
+
+/* ---- chunk encoder (host-side product component; the synthetic-data
+ * generator — reference writer semantics, integer_column_writer.cpp). ---- */
+
+/* Encodes one int64/uint64 column into reference-format segments.
+ * nulls: optional bytemask (1 = null), length n. Segments split every
+ * max_segment_values rows (DefaultMaxSegmentValueCount = 128*1024,
+ * table_chunk_format/public.h:11). The returned blob is a single allocation;
+ * call yt_encoded_column_free. On return, *segments points into the blob. */
+typedef struct YtEncodedColumn {
+    int32_t segment_count;
+    YtSegment* segments;          /* data pointers are HOST pointers into blob */
+    void* blob;
+    int64_t blob_size;
+} YtEncodedColumn;
+
+int yt_encode_int64_column(
+    const int64_t* values, const uint8_t* nulls, int64_t n,
+    int32_t max_segment_values, int32_t is_unsigned,
+    YtEncodedColumn* out, char* errbuf, size_t errlen);
+
+int yt_encode_double_column(
+    const double* values, const uint8_t* nulls, int64_t n,
+    int32_t max_segment_values,
+    YtEncodedColumn* out, char* errbuf, size_t errlen);
+
+void yt_encoded_column_free(YtEncodedColumn* col);
+
+/* Bit-pack primitive, exposed for format tests
+ * (core/misc/bit_packed_unsigned_vector-inl.h:34-82). dst must be
+ * zero-initialized and hold yt_bitpack_size_words(max_value, n) words.
+ * Returns words written. */
+int64_t yt_bitpack_size_words(uint64_t max_value, int64_t n);
+int64_t yt_bitpack(const uint64_t* values, int64_t n, uint64_t max_value, uint64_t* dst);
+
+#ifdef __cplusplus
+} /* extern "C" */
+#endif
+#endif /* YTQL_GPU_H */

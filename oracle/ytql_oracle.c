@@ -1,0 +1,986 @@
+/* ytql_oracle.c — CPU restatement of the YTsaurus dynamic-table query hot
+ * path (scan → filter → group-by → merge), used to pin parity of the GPU
+ * executor.
+ *
+ * TEST INFRASTRUCTURE ONLY. This library is the parity CHECKER and the
+ * reported CPU baseline; it is imported only by tests/, __graft_entry__.smoke()
+ * and bench.py's cpu_baseline leg. The product path (ytsaurus_amd/libytql_gpu.so)
+ * never calls it and has no CPU fallback.
+ *
+ * Every function cites the reference code (/root/reference, ytsaurus/ytsaurus
+ * @2026-08-21) it restates:
+ *   - bit-unpack:      yt/yt/core/misc/bit_packed_unsigned_vector-inl.h:108-117,157-186
+ *   - int64 decode:    yt/yt/ytlib/table_chunk_format/integer_column_reader.cpp:19-127,391-449
+ *   - double decode:   yt/yt/ytlib/table_chunk_format/floating_point_column_reader.cpp:36-60
+ *   - null bitmap:     yt/yt/core/misc/bitmap.h (ui8, LSB-first, 1 = null;
+ *                      serialization aligned to 8 bytes)
+ *   - expressions:     yt/yt/library/query/engine/cg_fragment_compiler.cpp
+ *                      (arithmetic :1440-1540 null-propagating, div-by-zero at :2019;
+ *                       relational :1601-1720: non-canonical null relations compare
+ *                       (rhsIsNull, lhsIsNull) i.e. null < any, result Boolean;
+ *                       logical :1547-1599 Kleene)
+ *   - group-by:        yt/yt/library/query/engine/cg_routines/registry.cpp:1231-1916
+ *                      (InsertGroupRow :1783, InsertIntermediate :1517; flush in
+ *                       insert order :1571-1650)
+ *   - aggregates:      yt/yt/library/query/engine/udf/sum.c (null-propagating add,
+ *                      int64 wraps mod 2^64), min.c, max.c
+ *   - merge mode:      cg_fragment_compiler.cpp:4116-4134 (front query re-groups by
+ *                      key and Merge(state,state))
+ *
+ * Parity pinning: the reference binary cannot be built in this container
+ * (conanfile.py:20-23 needs bison/m4/ragel/yasm + network) and this path has
+ * no Python implementation, so the oracle is pinned against golden vectors
+ * transcribed from the reference's own unit tests
+ * (yt/yt/library/query/unittests/ql_query_ut.cpp GroupBy family) in
+ * tests/golden/ — see tests/test_oracle_golden.py.
+ */
+
+#include <stdint.h>
+#include <stddef.h>
+#include <stdlib.h>
+#include <string.h>
+#include <stdio.h>
+#include <pthread.h>
+
+#include "../include/ytql_gpu.h"   /* shared struct/type declarations only */
+
+#define ORACLE_EXPORT __attribute__((visibility("default")))
+
+/* ------------------------------------------------------------------ */
+/* small helpers                                                       */
+
+static void set_err(char* errbuf, size_t errlen, const char* msg)
+{
+    if (errbuf && errlen) {
+        snprintf(errbuf, errlen, "%s", msg);
+    }
+}
+
+/* ZigZag codec — library/cpp/yt/coding/zig_zag-inl.h:24-35 */
+static inline uint64_t zigzag_encode64(int64_t n)
+{
+    return ((uint64_t)n << 1) ^ (uint64_t)(n >> 63);
+}
+static inline int64_t zigzag_decode64(uint64_t n)
+{
+    return (int64_t)((n >> 1) ^ (~(n & 1) + 1));
+}
+
+/* ------------------------------------------------------------------ */
+/* bit-packed unsigned vector reader                                   */
+/* header word = count(low 56b) | width(top 8b)                        */
+/* bit_packed_unsigned_vector-inl.h:108-133,157-186                    */
+
+typedef struct {
+    const uint64_t* data;   /* first data word (header + 1) */
+    uint64_t size;          /* element count */
+    unsigned width;
+} BitReader;
+
+static BitReader bitreader_init(const void* ptr)
+{
+    BitReader r;
+    uint64_t header = *(const uint64_t*)ptr;
+    r.data = (const uint64_t*)ptr + 1;
+    r.size = header & ((1ULL << 56) - 1);
+    r.width = (unsigned)(header >> 56);
+    return r;
+}
+
+/* bytes including header — TBitPackedUnsignedVectorReader::GetByteSize */
+static int64_t bitreader_byte_size(const BitReader* r)
+{
+    return (int64_t)(1 + (((uint64_t)r->width * r->size + 63ULL) >> 6)) * 8;
+}
+
+/* GetValue — -inl.h:157-178 */
+static inline uint64_t bitreader_get(const BitReader* r, uint64_t index)
+{
+    if (r->width == 0) {
+        return 0;
+    }
+    if (r->width == 64) {
+        return r->data[index];
+    }
+    uint64_t bit_index = index * r->width;
+    const uint64_t* word = r->data + (bit_index >> 6);
+    unsigned offset = bit_index & 63;
+    uint64_t w1 = *word >> offset;
+    if (offset + r->width > 64) {
+        uint64_t w2 = (word[1] & ((1ULL << ((offset + r->width) & 63)) - 1)) << (64 - offset);
+        return w1 | w2;
+    }
+    return w1 & ((r->width < 64) ? ((1ULL << r->width) - 1) : ~0ULL);
+}
+
+/* null bitmap: bit set = null; ui8 LSB-first — core/misc/bitmap.h */
+static inline int bitmap_get(const uint8_t* bm, uint64_t index)
+{
+    return (bm[index >> 3] >> (index & 7)) & 1;
+}
+
+static inline int64_t align_up8(int64_t x) { return (x + 7) & ~(int64_t)7; }
+
+/* ------------------------------------------------------------------ */
+/* segment decode → (int64 bits, null bytemask)                        */
+
+/* Decodes one integer segment into vals (i64 for Int64, raw u64 for Uint64)
+ * and nulls bytemask. Returns 0 on success.
+ * integer_column_reader.cpp: direct :54-88, dictionary :92-130, RLE via
+ * TRleValueExtractorBase (column_reader_detail.h:246-262) + writer run
+ * construction integer_column_writer.cpp:394-489. */
+static int decode_int_segment(const YtSegment* seg, int is_signed,
+                              int64_t* vals, uint8_t* nulls)
+{
+    const char* ptr = (const char*)seg->data;
+    const char* end = ptr + seg->data_size;
+    int64_t n = seg->row_count;
+    uint64_t minv = seg->min_value;
+
+    switch (seg->type) {
+    case YT_SEG_DIRECT_DENSE: {
+        BitReader values = bitreader_init(ptr);
+        ptr += bitreader_byte_size(&values);
+        const uint8_t* nb = (const uint8_t*)ptr;
+        ptr += align_up8((n + 7) / 8);
+        if ((int64_t)values.size != n || ptr != end) return -1;
+        for (int64_t i = 0; i < n; i++) {
+            int isnull = bitmap_get(nb, i);
+            nulls[i] = (uint8_t)isnull;
+            uint64_t data = minv + bitreader_get(&values, i);
+            vals[i] = is_signed ? zigzag_decode64(data) : (int64_t)data;
+        }
+        return 0;
+    }
+    case YT_SEG_DICTIONARY_DENSE: {
+        BitReader dict = bitreader_init(ptr);
+        ptr += bitreader_byte_size(&dict);
+        BitReader ids = bitreader_init(ptr);
+        ptr += bitreader_byte_size(&ids);
+        if ((int64_t)ids.size != n || ptr != end) return -1;
+        for (int64_t i = 0; i < n; i++) {
+            uint64_t id = bitreader_get(&ids, i);
+            if (id == 0) {
+                nulls[i] = 1;
+                vals[i] = 0;
+            } else {
+                nulls[i] = 0;
+                uint64_t data = minv + bitreader_get(&dict, id - 1);
+                vals[i] = is_signed ? zigzag_decode64(data) : (int64_t)data;
+            }
+        }
+        return 0;
+    }
+    case YT_SEG_DIRECT_RLE: {
+        BitReader values = bitreader_init(ptr);
+        ptr += bitreader_byte_size(&values);
+        int64_t run_count = (int64_t)values.size;
+        const uint8_t* nb = (const uint8_t*)ptr;
+        ptr += align_up8((run_count + 7) / 8);
+        BitReader starts = bitreader_init(ptr);
+        ptr += bitreader_byte_size(&starts);
+        if ((int64_t)starts.size != run_count || ptr != end) return -1;
+        int64_t run = 0;
+        for (int64_t i = 0; i < n; i++) {
+            while (run + 1 < run_count && (int64_t)bitreader_get(&starts, run + 1) <= i) {
+                run++;
+            }
+            int isnull = bitmap_get(nb, run);
+            nulls[i] = (uint8_t)isnull;
+            uint64_t data = minv + bitreader_get(&values, run);
+            vals[i] = is_signed ? zigzag_decode64(data) : (int64_t)data;
+        }
+        return 0;
+    }
+    case YT_SEG_DICTIONARY_RLE: {
+        BitReader dict = bitreader_init(ptr);
+        ptr += bitreader_byte_size(&dict);
+        BitReader ids = bitreader_init(ptr);
+        ptr += bitreader_byte_size(&ids);
+        int64_t run_count = (int64_t)ids.size;
+        BitReader starts = bitreader_init(ptr);
+        ptr += bitreader_byte_size(&starts);
+        if ((int64_t)starts.size != run_count || ptr != end) return -1;
+        int64_t run = 0;
+        for (int64_t i = 0; i < n; i++) {
+            while (run + 1 < run_count && (int64_t)bitreader_get(&starts, run + 1) <= i) {
+                run++;
+            }
+            uint64_t id = bitreader_get(&ids, run);
+            if (id == 0) {
+                nulls[i] = 1;
+                vals[i] = 0;
+            } else {
+                nulls[i] = 0;
+                uint64_t data = minv + bitreader_get(&dict, id - 1);
+                vals[i] = is_signed ? zigzag_decode64(data) : (int64_t)data;
+            }
+        }
+        return 0;
+    }
+    default:
+        return -1;
+    }
+}
+
+/* floating_point_column_reader.cpp:36-60: [u64 count][doubles][null bitmap] */
+static int decode_double_segment(const YtSegment* seg, int64_t* vals, uint8_t* nulls)
+{
+    const char* ptr = (const char*)seg->data;
+    const char* end = ptr + seg->data_size;
+    uint64_t count = *(const uint64_t*)ptr;
+    ptr += 8;
+    const double* d = (const double*)ptr;
+    ptr += 8 * count;
+    const uint8_t* nb = (const uint8_t*)ptr;
+    ptr += align_up8(((int64_t)count + 7) / 8);
+    if ((int64_t)count != seg->row_count || ptr != end) return -1;
+    for (uint64_t i = 0; i < count; i++) {
+        nulls[i] = (uint8_t)bitmap_get(nb, i);
+        memcpy(&vals[i], &d[i], 8);
+    }
+    return 0;
+}
+
+/* Decode a whole column into arrays. vals carries i64 / u64 / double bits. */
+ORACLE_EXPORT
+int yto_decode_column(const YtColumn* col, int64_t row_count,
+                      int64_t* vals, uint8_t* nulls)
+{
+    int64_t row = 0;
+    for (int s = 0; s < col->segment_count; s++) {
+        const YtSegment* seg = &col->segments[s];
+        int rc;
+        if (col->value_type == YT_VT_DOUBLE) {
+            rc = decode_double_segment(seg, vals + row, nulls + row);
+        } else {
+            rc = decode_int_segment(seg, col->value_type == YT_VT_INT64,
+                                    vals + row, nulls + row);
+        }
+        if (rc != 0) return YT_ERR_INVALID_CHUNK;
+        row += seg->row_count;
+    }
+    return (row == row_count) ? YT_OK : YT_ERR_INVALID_CHUNK;
+}
+
+/* ------------------------------------------------------------------ */
+/* expression evaluation                                               */
+
+typedef struct {
+    uint8_t type;     /* YT_VT_* */
+    uint64_t bits;    /* i64/u64/double bits/bool(0/1) */
+} Val;
+
+typedef struct {
+    const int64_t* const* col_vals;
+    const uint8_t* const* col_nulls;
+    const uint8_t* col_types;
+    int ncols;
+    int64_t row;
+    int error;        /* YT_ERR_DIV_ZERO etc. */
+} EvalCtx;
+
+static Val VNULL(void) { Val v; v.type = YT_VT_NULL; v.bits = 0; return v; }
+
+static Val eval_expr(const YtExpr* e, EvalCtx* ctx)
+{
+    Val v = VNULL();
+    switch (e->op) {
+    case YT_EX_COLUMN: {
+        if (ctx->col_nulls[e->col][ctx->row]) return VNULL();
+        v.type = ctx->col_types[e->col];
+        v.bits = (uint64_t)ctx->col_vals[e->col][ctx->row];
+        return v;
+    }
+    case YT_EX_LIT_I64:
+        v.type = YT_VT_INT64;
+        v.bits = (uint64_t)e->lit_i64;
+        return v;
+    case YT_EX_LIT_DOUBLE:
+        v.type = YT_VT_DOUBLE;
+        memcpy(&v.bits, &e->lit_dbl, 8);
+        return v;
+    case YT_EX_LIT_NULL:
+        return VNULL();
+    case YT_EX_NOT: {
+        Val a = eval_expr(e->a, ctx);
+        if (a.type == YT_VT_NULL) return VNULL();
+        v.type = YT_VT_BOOLEAN;
+        v.bits = !a.bits;
+        return v;
+    }
+    default:
+        break;
+    }
+
+    Val a = eval_expr(e->a, ctx);
+    Val b = eval_expr(e->b, ctx);
+
+    if (e->op >= YT_EX_ADD && e->op <= YT_EX_MOD) {
+        /* arithmetic: null-propagating (cg_fragment_compiler.cpp arithmetic op) */
+        if (a.type == YT_VT_NULL || b.type == YT_VT_NULL) return VNULL();
+        if (a.type == YT_VT_DOUBLE) {
+            double x, y, r = 0;
+            memcpy(&x, &a.bits, 8);
+            memcpy(&y, &b.bits, 8);
+            switch (e->op) {
+            case YT_EX_ADD: r = x + y; break;
+            case YT_EX_SUB: r = x - y; break;
+            case YT_EX_MUL: r = x * y; break;
+            case YT_EX_DIV: r = x / y; break;
+            default: ctx->error = YT_ERR_UNSUPPORTED; return VNULL();
+            }
+            v.type = YT_VT_DOUBLE;
+            memcpy(&v.bits, &r, 8);
+            return v;
+        }
+        /* int64/uint64: wrapping two's-complement arithmetic */
+        uint64_t x = a.bits, y = b.bits, r = 0;
+        int sgn = (a.type == YT_VT_INT64);
+        switch (e->op) {
+        case YT_EX_ADD: r = x + y; break;
+        case YT_EX_SUB: r = x - y; break;
+        case YT_EX_MUL: r = x * y; break;
+        case YT_EX_DIV:
+        case YT_EX_MOD:
+            if (y == 0) { ctx->error = YT_ERR_DIV_ZERO; return VNULL(); }
+            if (sgn) {
+                int64_t sx = (int64_t)x, sy = (int64_t)y;
+                if (sx == INT64_MIN && sy == -1) {
+                    r = (e->op == YT_EX_DIV) ? (uint64_t)INT64_MIN : 0;
+                } else {
+                    r = (uint64_t)((e->op == YT_EX_DIV) ? sx / sy : sx % sy);
+                }
+            } else {
+                r = (e->op == YT_EX_DIV) ? x / y : x % y;
+            }
+            break;
+        default: break;
+        }
+        v.type = a.type;
+        v.bits = r;
+        return v;
+    }
+
+    if (e->op >= YT_EX_EQ && e->op <= YT_EX_GE) {
+        /* relational: non-canonical null relations — when either side is
+         * null, compare (rhsIsNull, lhsIsNull) unsigned: null < any value,
+         * null == null; result is a non-null Boolean
+         * (cg_fragment_compiler.cpp:1621-1649). */
+        int lt, eq;
+        if (a.type == YT_VT_NULL || b.type == YT_VT_NULL) {
+            unsigned ln = (a.type == YT_VT_NULL), rn = (b.type == YT_VT_NULL);
+            lt = rn < ln;     /* lhs < rhs  <=>  rhsIsNull < lhsIsNull */
+            eq = ln == rn;
+        } else if (a.type == YT_VT_DOUBLE) {
+            double x, y;
+            memcpy(&x, &a.bits, 8);
+            memcpy(&y, &b.bits, 8);
+            /* FCmpU*: unordered-or — NaN makes every comparison true in the
+             * reference codegen; restate exactly */
+            int unordered = (x != x) || (y != y);
+            lt = unordered || (x < y);
+            eq = unordered || (x == y);
+            if (unordered) {
+                v.type = YT_VT_BOOLEAN;
+                switch (e->op) { /* all unordered comparisons are true */
+                case YT_EX_EQ: case YT_EX_NE: case YT_EX_LT:
+                case YT_EX_LE: case YT_EX_GT: case YT_EX_GE:
+                    v.bits = 1; return v;
+                }
+            }
+        } else if (a.type == YT_VT_INT64) {
+            int64_t x = (int64_t)a.bits, y = (int64_t)b.bits;
+            lt = x < y;
+            eq = x == y;
+        } else { /* uint64 / boolean */
+            lt = a.bits < b.bits;
+            eq = a.bits == b.bits;
+        }
+        int r = 0;
+        switch (e->op) {
+        case YT_EX_EQ: r = eq; break;
+        case YT_EX_NE: r = !eq; break;
+        case YT_EX_LT: r = lt; break;
+        case YT_EX_LE: r = lt || eq; break;
+        case YT_EX_GT: r = !(lt || eq); break;
+        case YT_EX_GE: r = !lt; break;
+        }
+        v.type = YT_VT_BOOLEAN;
+        v.bits = (uint64_t)r;
+        return v;
+    }
+
+    if (e->op == YT_EX_AND || e->op == YT_EX_OR) {
+        /* Kleene logic — cg_fragment_compiler.cpp:1547-1599 */
+        int an = (a.type == YT_VT_NULL), bn = (b.type == YT_VT_NULL);
+        int av = an ? 0 : (int)(a.bits != 0), bv = bn ? 0 : (int)(b.bits != 0);
+        if (e->op == YT_EX_AND) {
+            if ((!an && !av) || (!bn && !bv)) { v.type = YT_VT_BOOLEAN; v.bits = 0; return v; }
+            if (an || bn) return VNULL();
+            v.type = YT_VT_BOOLEAN; v.bits = 1; return v;
+        } else {
+            if ((!an && av) || (!bn && bv)) { v.type = YT_VT_BOOLEAN; v.bits = 1; return v; }
+            if (an || bn) return VNULL();
+            v.type = YT_VT_BOOLEAN; v.bits = 0; return v;
+        }
+    }
+
+    ctx->error = YT_ERR_UNSUPPORTED;
+    return VNULL();
+}
+
+/* ------------------------------------------------------------------ */
+/* group-by hash table (insert-order preserving — flush order mirrors
+ * registry.cpp:1571-1650: rows flush in hash-insert order)             */
+
+typedef struct {
+    Val* keys;            /* ngroups * key_count */
+    Val* states;          /* ngroups * agg_count (sum state: typed or null) */
+    uint64_t* rowcounts;  /* rows per group (sum(1) state) */
+    int64_t ngroups;
+    int64_t cap;
+    int64_t* slots;       /* open addressing: index into keys/-1 */
+    int64_t nslots;       /* power of 2 */
+    int key_count;
+    int agg_count;
+} GroupTable;
+
+static uint64_t splitmix64(uint64_t x)
+{
+    x += 0x9E3779B97F4A7C15ULL;
+    x = (x ^ (x >> 30)) * 0xBF58476D1CE4E5B9ULL;
+    x = (x ^ (x >> 27)) * 0x94D049BB133111EBULL;
+    return x ^ (x >> 31);
+}
+
+static uint64_t hash_keys(const Val* keys, int key_count)
+{
+    uint64_t h = 0x12345678ULL;
+    for (int i = 0; i < key_count; i++) {
+        h = splitmix64(h ^ splitmix64(((uint64_t)keys[i].type << 56) ^ keys[i].bits));
+    }
+    return h;
+}
+
+static int keys_eq(const Val* a, const Val* b, int key_count)
+{
+    for (int i = 0; i < key_count; i++) {
+        if (a[i].type != b[i].type) return 0;
+        if (a[i].type != YT_VT_NULL && a[i].bits != b[i].bits) return 0;
+    }
+    return 1;
+}
+
+static int gt_init(GroupTable* t, int key_count, int agg_count, int64_t cap_hint)
+{
+    memset(t, 0, sizeof(*t));
+    t->key_count = key_count;
+    t->agg_count = agg_count;
+    t->cap = cap_hint > 16 ? cap_hint : 16;
+    t->nslots = 32;
+    while (t->nslots < t->cap * 2) t->nslots <<= 1;
+    t->keys = malloc(sizeof(Val) * t->cap * (key_count ? key_count : 1));
+    t->states = malloc(sizeof(Val) * t->cap * (agg_count ? agg_count : 1));
+    t->rowcounts = malloc(sizeof(uint64_t) * t->cap);
+    t->slots = malloc(sizeof(int64_t) * t->nslots);
+    if (!t->keys || !t->states || !t->rowcounts || !t->slots) return -1;
+    for (int64_t i = 0; i < t->nslots; i++) t->slots[i] = -1;
+    return 0;
+}
+
+static void gt_free(GroupTable* t)
+{
+    free(t->keys); free(t->states); free(t->rowcounts); free(t->slots);
+}
+
+static int gt_grow(GroupTable* t)
+{
+    int64_t newcap = t->cap * 2;
+    int kc = t->key_count ? t->key_count : 1;
+    int ac = t->agg_count ? t->agg_count : 1;
+    t->keys = realloc(t->keys, sizeof(Val) * newcap * kc);
+    t->states = realloc(t->states, sizeof(Val) * newcap * ac);
+    t->rowcounts = realloc(t->rowcounts, sizeof(uint64_t) * newcap);
+    if (!t->keys || !t->states || !t->rowcounts) return -1;
+    t->cap = newcap;
+    if (t->nslots < newcap * 2) {
+        int64_t newslots = t->nslots;
+        while (newslots < newcap * 2) newslots <<= 1;
+        t->slots = realloc(t->slots, sizeof(int64_t) * newslots);
+        if (!t->slots) return -1;
+        t->nslots = newslots;
+        for (int64_t i = 0; i < t->nslots; i++) t->slots[i] = -1;
+        for (int64_t g = 0; g < t->ngroups; g++) {
+            uint64_t h = hash_keys(&t->keys[g * t->key_count], t->key_count);
+            int64_t s = (int64_t)(h & (uint64_t)(t->nslots - 1));
+            while (t->slots[s] != -1) s = (s + 1) & (t->nslots - 1);
+            t->slots[s] = g;
+        }
+    }
+    return 0;
+}
+
+/* find-or-insert; returns group index, or -1 on OOM */
+static int64_t gt_upsert(GroupTable* t, const Val* keys)
+{
+    uint64_t h = hash_keys(keys, t->key_count);
+    int64_t s = (int64_t)(h & (uint64_t)(t->nslots - 1));
+    for (;;) {
+        int64_t g = t->slots[s];
+        if (g == -1) break;
+        if (keys_eq(&t->keys[g * t->key_count], keys, t->key_count)) return g;
+        s = (s + 1) & (t->nslots - 1);
+    }
+    if (t->ngroups == t->cap) {
+        if (gt_grow(t) != 0) return -1;
+        /* slots were rebuilt; re-probe */
+        s = (int64_t)(h & (uint64_t)(t->nslots - 1));
+        while (t->slots[s] != -1) s = (s + 1) & (t->nslots - 1);
+    }
+    int64_t g = t->ngroups++;
+    t->slots[s] = g;
+    memcpy(&t->keys[g * t->key_count], keys, sizeof(Val) * t->key_count);
+    for (int a = 0; a < t->agg_count; a++) {
+        t->states[g * t->agg_count + a] = VNULL();   /* sum_init — udf/sum.c:3-10 */
+    }
+    t->rowcounts[g] = 0;
+    return g;
+}
+
+/* sum_update / sum_merge — udf/sum.c:12-65 (null-propagating; int64 wraps) */
+static void sum_update_val(Val* state, Val nv)
+{
+    if (nv.type == YT_VT_NULL) return;
+    if (state->type == YT_VT_NULL) { *state = nv; return; }
+    if (nv.type == YT_VT_INT64 || nv.type == YT_VT_UINT64) {
+        state->bits = state->bits + nv.bits;
+    } else if (nv.type == YT_VT_DOUBLE) {
+        double x, y;
+        memcpy(&x, &state->bits, 8);
+        memcpy(&y, &nv.bits, 8);
+        x += y;
+        memcpy(&state->bits, &x, 8);
+    }
+}
+
+static void minmax_update_val(Val* state, Val nv, int is_max)
+{
+    if (nv.type == YT_VT_NULL) return;
+    int take = 0;
+    if (state->type == YT_VT_NULL) {
+        take = 1;
+    } else if (nv.type == YT_VT_INT64) {
+        int64_t s = (int64_t)state->bits, x = (int64_t)nv.bits;
+        take = is_max ? (s < x) : (s >= x);
+    } else if (nv.type == YT_VT_UINT64 || nv.type == YT_VT_BOOLEAN) {
+        take = is_max ? (state->bits < nv.bits) : (state->bits >= nv.bits);
+    } else if (nv.type == YT_VT_DOUBLE) {
+        double s, x;
+        memcpy(&s, &state->bits, 8);
+        memcpy(&x, &nv.bits, 8);
+        take = is_max ? (s < x) : (s >= x);
+    }
+    if (take) *state = nv;
+}
+
+/* ------------------------------------------------------------------ */
+/* execution                                                           */
+
+typedef struct {
+    const YtPlan* plan;
+    const YtChunk* chunk;
+    int64_t row_begin, row_end;
+    int64_t** vals;        /* decoded columns (shared) */
+    uint8_t** nulls;
+    uint8_t* types;
+    GroupTable table;
+    int error;
+    int64_t rows_read;
+} ScanTask;
+
+static void* scan_worker(void* arg)
+{
+    ScanTask* t = (ScanTask*)arg;
+    const YtPlan* plan = t->plan;
+    EvalCtx ctx;
+    ctx.col_vals = (const int64_t* const*)t->vals;
+    ctx.col_nulls = (const uint8_t* const*)t->nulls;
+    ctx.col_types = t->types;
+    ctx.ncols = t->chunk->column_count;
+    ctx.error = 0;
+
+    Val keybuf[16];
+    int kc = plan->key_count;
+
+    for (int64_t r = t->row_begin; r < t->row_end; r++) {
+        ctx.row = r;
+        t->rows_read++;
+        if (plan->filter) {
+            Val f = eval_expr(plan->filter, &ctx);
+            if (ctx.error) { t->error = ctx.error; return NULL; }
+            if (f.type == YT_VT_NULL || f.bits == 0) continue;
+        }
+        for (int k = 0; k < kc; k++) {
+            keybuf[k] = eval_expr(plan->keys[k], &ctx);
+            if (ctx.error) { t->error = ctx.error; return NULL; }
+        }
+        int64_t g = gt_upsert(&t->table, keybuf);
+        if (g < 0) { t->error = YT_ERR_CAPACITY; return NULL; }
+        t->table.rowcounts[g]++;
+        for (int a = 0; a < plan->agg_count; a++) {
+            const YtAgg* agg = plan->aggs[a];
+            if (agg->func == YT_AGG_SUM1) continue;  /* rowcounts covers it */
+            Val nv = eval_expr(agg->arg, &ctx);
+            if (ctx.error) { t->error = ctx.error; return NULL; }
+            Val* st = &t->table.states[g * plan->agg_count + a];
+            if (agg->func == YT_AGG_SUM) sum_update_val(st, nv);
+            else minmax_update_val(st, nv, agg->func == YT_AGG_MAX);
+        }
+    }
+    return NULL;
+}
+
+/* write one output row from group g of table */
+static int emit_group_row(const YtPlan* plan, GroupTable* t, int64_t g,
+                          YtRowset* out)
+{
+    int kc = plan->key_count, ac = plan->agg_count;
+    int ncols = plan->project_count ? plan->project_count : kc + ac;
+    if (out->row_count >= out->capacity_rows) return YT_ERR_CAPACITY;
+
+    /* materialize [keys..., finalized aggs...] as a row */
+    Val rowvals[32];
+    for (int k = 0; k < kc; k++) rowvals[k] = t->keys[g * kc + k];
+    for (int a = 0; a < ac; a++) {
+        if (plan->aggs[a]->func == YT_AGG_SUM1) {
+            Val v; v.type = YT_VT_INT64; v.bits = t->rowcounts[g];
+            rowvals[kc + a] = v;
+        } else {
+            rowvals[kc + a] = t->states[g * ac + a];  /* sum_finalize = copy */
+        }
+    }
+
+    YtValue* dst = out->values + out->row_count * ncols;
+    if (plan->project_count) {
+        /* projection over the group row: column i refers to rowvals[i] */
+        int64_t pv[32];
+        uint8_t pn[32];
+        uint8_t pt[32];
+        const int64_t* pvp[32];
+        const uint8_t* pnp[32];
+        for (int i = 0; i < kc + ac; i++) {
+            pv[i] = (int64_t)rowvals[i].bits;
+            pn[i] = (rowvals[i].type == YT_VT_NULL);
+            pt[i] = pn[i] ? YT_VT_INT64 : rowvals[i].type;
+            pvp[i] = &pv[i];
+            pnp[i] = &pn[i];
+        }
+        EvalCtx ctx;
+        ctx.col_vals = pvp;
+        ctx.col_nulls = pnp;
+        ctx.col_types = pt;
+        ctx.ncols = kc + ac;
+        ctx.row = 0;
+        ctx.error = 0;
+        for (int p = 0; p < plan->project_count; p++) {
+            Val v = eval_expr(plan->projects[p], &ctx);
+            if (ctx.error) return ctx.error;
+            dst[p].id = (uint16_t)p;
+            dst[p].type = v.type;
+            dst[p].flags = 0;
+            dst[p].length = 0;
+            dst[p].data.bits = v.bits;
+        }
+    } else {
+        for (int i = 0; i < ncols; i++) {
+            dst[i].id = (uint16_t)i;
+            dst[i].type = rowvals[i].type;
+            dst[i].flags = 0;
+            dst[i].length = 0;
+            dst[i].data.bits = rowvals[i].bits;
+        }
+    }
+    out->row_count++;
+    out->column_count = ncols;
+    return YT_OK;
+}
+
+ORACLE_EXPORT
+int yto_execute(const YtPlan* plan, const YtChunk* chunk,
+                YtRowset* output, YtStatistics* stats,
+                int nthreads, char* errbuf, size_t errlen)
+{
+    if (nthreads < 1) nthreads = 1;
+    int ncols = chunk->column_count;
+    int64_t n = chunk->row_count;
+    int rc = YT_OK;
+
+    int64_t** vals = calloc(ncols, sizeof(int64_t*));
+    uint8_t** nulls = calloc(ncols, sizeof(uint8_t*));
+    uint8_t* types = calloc(ncols, 1);
+    for (int c = 0; c < ncols; c++) {
+        vals[c] = malloc(sizeof(int64_t) * (n ? n : 1));
+        nulls[c] = malloc(n ? n : 1);
+        types[c] = (uint8_t)chunk->columns[c].value_type;
+        if (!vals[c] || !nulls[c]) { rc = YT_ERR_CAPACITY; goto done; }
+        rc = yto_decode_column(&chunk->columns[c], n, vals[c], nulls[c]);
+        if (rc != YT_OK) { set_err(errbuf, errlen, "oracle: bad segment"); goto done; }
+    }
+
+    output->row_count = 0;
+    output->string_pool_used = 0;
+
+    if (plan->agg_count == 0 && plan->key_count == 0) {
+        /* plain scan+filter+project */
+        EvalCtx ctx;
+        ctx.col_vals = (const int64_t* const*)vals;
+        ctx.col_nulls = (const uint8_t* const*)nulls;
+        ctx.col_types = types;
+        ctx.ncols = ncols;
+        ctx.error = 0;
+        int np = plan->project_count;
+        for (int64_t r = 0; r < n; r++) {
+            ctx.row = r;
+            if (plan->filter) {
+                Val f = eval_expr(plan->filter, &ctx);
+                if (ctx.error) { rc = ctx.error; set_err(errbuf, errlen, "expr error"); goto done; }
+                if (f.type == YT_VT_NULL || f.bits == 0) continue;
+            }
+            if (output->row_count >= output->capacity_rows) { rc = YT_ERR_CAPACITY; goto done; }
+            YtValue* dst = output->values + output->row_count * np;
+            for (int p = 0; p < np; p++) {
+                Val v = eval_expr(plan->projects[p], &ctx);
+                if (ctx.error) { rc = ctx.error; set_err(errbuf, errlen, "expr error"); goto done; }
+                dst[p].id = (uint16_t)p;
+                dst[p].type = v.type;
+                dst[p].flags = 0;
+                dst[p].length = 0;
+                dst[p].data.bits = v.bits;
+            }
+            output->row_count++;
+        }
+        output->column_count = np;
+        if (stats) {
+            stats->rows_read = n;
+            stats->rows_written = output->row_count;
+            stats->grouped_row_count = 0;
+        }
+        goto done;
+    }
+
+    /* group-by: partition rows across threads on segment-ish boundaries,
+     * per-thread tables, then merge in thread order (mirrors the reference's
+     * per-tablet bottom queries + front merge, executor.cpp:761) */
+    {
+        ScanTask* tasks = calloc(nthreads, sizeof(ScanTask));
+        pthread_t* tids = malloc(sizeof(pthread_t) * nthreads);
+        int64_t per = (n + nthreads - 1) / nthreads;
+        int actual = 0;
+        for (int i = 0; i < nthreads; i++) {
+            int64_t b = (int64_t)i * per;
+            int64_t e = b + per > n ? n : b + per;
+            if (b >= e) break;
+            tasks[actual].plan = plan;
+            tasks[actual].chunk = chunk;
+            tasks[actual].row_begin = b;
+            tasks[actual].row_end = e;
+            tasks[actual].vals = vals;
+            tasks[actual].nulls = nulls;
+            tasks[actual].types = types;
+            gt_init(&tasks[actual].table, plan->key_count, plan->agg_count, 1024);
+            actual++;
+        }
+        if (actual == 1) {
+            scan_worker(&tasks[0]);
+        } else {
+            for (int i = 0; i < actual; i++) pthread_create(&tids[i], NULL, scan_worker, &tasks[i]);
+            for (int i = 0; i < actual; i++) pthread_join(tids[i], NULL);
+        }
+        for (int i = 0; i < actual; i++) {
+            if (tasks[i].error) { rc = tasks[i].error; set_err(errbuf, errlen, "scan error"); }
+        }
+        if (rc == YT_OK) {
+            GroupTable* final_t;
+            GroupTable merged;
+            if (actual == 1) {
+                final_t = &tasks[0].table;
+            } else {
+                gt_init(&merged, plan->key_count, plan->agg_count, 1024);
+                for (int i = 0; i < actual; i++) {
+                    GroupTable* pt = &tasks[i].table;
+                    for (int64_t g = 0; g < pt->ngroups; g++) {
+                        int64_t mg = gt_upsert(&merged, &pt->keys[g * pt->key_count]);
+                        merged.rowcounts[mg] += pt->rowcounts[g];
+                        for (int a = 0; a < plan->agg_count; a++) {
+                            Val st = pt->states[g * plan->agg_count + a];
+                            Val* dst = &merged.states[mg * plan->agg_count + a];
+                            if (plan->aggs[a]->func == YT_AGG_SUM) sum_update_val(dst, st);
+                            else if (plan->aggs[a]->func != YT_AGG_SUM1)
+                                minmax_update_val(dst, st, plan->aggs[a]->func == YT_AGG_MAX);
+                        }
+                    }
+                }
+                final_t = &merged;
+            }
+            for (int64_t g = 0; g < final_t->ngroups && rc == YT_OK; g++) {
+                rc = emit_group_row(plan, final_t, g, output);
+            }
+            if (stats) {
+                stats->rows_read = n;
+                stats->rows_written = output->row_count;
+                stats->grouped_row_count = final_t->ngroups;
+            }
+            if (actual != 1) gt_free(&merged);
+        }
+        for (int i = 0; i < actual; i++) gt_free(&tasks[i].table);
+        free(tasks);
+        free(tids);
+    }
+
+done:
+    for (int c = 0; c < ncols; c++) {
+        if (vals) free(vals[c]);
+        if (nulls) free(nulls[c]);
+    }
+    free(vals); free(nulls); free(types);
+    return rc;
+}
+
+/* ------------------------------------------------------------------ */
+/* two-phase path: partial states + partition, and merge               */
+/* (restates the bottom/front split + in-process shuffle:               */
+/*  engine/coordinator.cpp:420-505, engine_api/shuffling_reader.cpp:21-88) */
+
+/* Partition hash over the group key — internal choice (need not match the
+ * reference's farm fingerprint, results are order-free); MUST match the GPU
+ * library's yt_partition_hash. */
+ORACLE_EXPORT
+uint64_t yto_partition_hash(uint64_t key_bits, int key_is_null)
+{
+    return splitmix64(key_bits ^ (key_is_null ? 0xDEADBEEF12345678ULL : 0));
+}
+
+/* Bottom query for the round-1 state family: 1 int64/bool key, aggs =
+ * {sum(expr), sum(1)} in any order. Emits YtStateRow records grouped into
+ * partition_count buckets (bucket-major). */
+ORACLE_EXPORT
+int yto_partial(const YtPlan* plan, const YtChunk* chunk,
+                int32_t partition_count,
+                YtStateRow* out, int64_t capacity_rows,
+                int64_t* part_counts,
+                int nthreads, char* errbuf, size_t errlen)
+{
+    if (plan->key_count != 1) { set_err(errbuf, errlen, "partial: need 1 key"); return YT_ERR_UNSUPPORTED; }
+    int sum_idx = -1;
+    for (int a = 0; a < plan->agg_count; a++) {
+        if (plan->aggs[a]->func == YT_AGG_SUM) sum_idx = a;
+        else if (plan->aggs[a]->func != YT_AGG_SUM1) { set_err(errbuf, errlen, "partial: sum/sum1 only"); return YT_ERR_UNSUPPORTED; }
+    }
+
+    /* run the local group-by via yto_execute on a plan without projection */
+    YtPlan local = *plan;
+    local.project_count = 0;
+    local.projects = NULL;
+
+    int64_t cap = capacity_rows;
+    YtValue* tmp = malloc(sizeof(YtValue) * cap * (1 + plan->agg_count));
+    if (!tmp) return YT_ERR_CAPACITY;
+    YtRowset rs;
+    memset(&rs, 0, sizeof(rs));
+    rs.values = tmp;
+    rs.capacity_rows = cap;
+    YtStatistics st;
+    memset(&st, 0, sizeof(st));
+    int rc = yto_execute(&local, chunk, &rs, &st, nthreads, errbuf, errlen);
+    if (rc != YT_OK) { free(tmp); return rc; }
+
+    /* To know nonnull counts we recompute: a state with non-null sum means
+     * nonnull >= 1; exact nonnull count is not needed for sum-merge parity —
+     * only null-ness matters (sum.c:12-22). Encode nonnull = 1 for non-null
+     * state, 0 for null state. */
+    int ncols = 1 + plan->agg_count;
+    int64_t* counts = calloc(partition_count, sizeof(int64_t));
+    for (int64_t r = 0; r < rs.row_count; r++) {
+        const YtValue* row = rs.values + r * ncols;
+        int knull = (row[0].type == YT_VT_NULL);
+        counts[yto_partition_hash(row[0].data.bits, knull) % (uint64_t)partition_count]++;
+    }
+    int64_t total = 0;
+    int64_t* offs = calloc(partition_count, sizeof(int64_t));
+    for (int p = 0; p < partition_count; p++) { offs[p] = total; total += counts[p]; }
+    if (total > capacity_rows) { free(tmp); free(counts); free(offs); return YT_ERR_CAPACITY; }
+
+    for (int64_t r = 0; r < rs.row_count; r++) {
+        const YtValue* row = rs.values + r * ncols;
+        int knull = (row[0].type == YT_VT_NULL);
+        int64_t p = (int64_t)(yto_partition_hash(row[0].data.bits, knull) % (uint64_t)partition_count);
+        YtStateRow* sr = &out[offs[p]++];
+        sr->key_bits = row[0].data.bits;
+        uint64_t rowcount = 0, sum_bits = 0, nonnull = 0;
+        for (int a = 0; a < plan->agg_count; a++) {
+            if (plan->aggs[a]->func == YT_AGG_SUM1) rowcount = row[1 + a].data.bits;
+            else if (a == sum_idx) {
+                sum_bits = row[1 + a].data.bits;
+                nonnull = (row[1 + a].type != YT_VT_NULL);
+            }
+        }
+        sr->meta = (uint64_t)knull | (nonnull << 8);
+        sr->sum_bits = sum_bits;
+        sr->row_count = rowcount;
+    }
+    for (int p = 0; p < partition_count; p++) part_counts[p] = counts[p];
+    free(tmp); free(counts); free(offs);
+    return YT_OK;
+}
+
+/* Front-query merge + finalize over YtStateRow records.
+ * Output columns follow the plan: [key, aggs...] or projection. */
+ORACLE_EXPORT
+int yto_merge(const YtPlan* plan, const YtStateRow* states, int64_t nstates,
+              YtRowset* output, char* errbuf, size_t errlen)
+{
+    if (plan->key_count != 1) { set_err(errbuf, errlen, "merge: need 1 key"); return YT_ERR_UNSUPPORTED; }
+    GroupTable t;
+    gt_init(&t, 1, plan->agg_count, 1024);
+    int key_type = YT_VT_INT64;
+    /* key static type: a boolean-valued key expr yields boolean outputs */
+    for (int64_t i = 0; i < nstates; i++) {
+        Val key;
+        if (states[i].meta & 1) {
+            key = VNULL();
+        } else {
+            key.type = (uint8_t)key_type;
+            key.bits = states[i].key_bits;
+        }
+        int64_t g = gt_upsert(&t, &key);
+        t.rowcounts[g] += states[i].row_count;
+        uint64_t nonnull = states[i].meta >> 8;
+        for (int a = 0; a < plan->agg_count; a++) {
+            if (plan->aggs[a]->func != YT_AGG_SUM) continue;
+            if (nonnull) {
+                Val nv;
+                nv.type = YT_VT_INT64;
+                nv.bits = states[i].sum_bits;
+                sum_update_val(&t.states[g * plan->agg_count + a], nv);
+            }
+        }
+    }
+    output->row_count = 0;
+    int rc = YT_OK;
+    for (int64_t g = 0; g < t.ngroups && rc == YT_OK; g++) {
+        rc = emit_group_row(plan, &t, g, output);
+    }
+    gt_free(&t);
+    return rc;
+}
+
+/* bit-unpack KAT helper: unpack a packed vector (with header) into out */
+ORACLE_EXPORT
+int64_t yto_bitunpack(const void* packed, uint64_t* out, int64_t max_out)
+{
+    BitReader r = bitreader_init(packed);
+    int64_t n = (int64_t)r.size < max_out ? (int64_t)r.size : max_out;
+    for (int64_t i = 0; i < n; i++) out[i] = bitreader_get(&r, i);
+    return (int64_t)r.size;
+}

@@ -1,0 +1,398 @@
+"""High-level Python helpers around the C-ABI (plan building, chunk encoding,
+execution). Python here is plumbing only: encoding and all query compute run
+in native code (libytql_gpu.so / the oracle).
+"""
+import ctypes as C
+
+import numpy as np
+
+from . import _abi
+from ._abi import (
+    YtValue, YtSegment, YtColumn, YtChunk, YtExpr, YtAgg, YtPlan,
+    YtExecOptions, YtStatistics, YtRowset, YtStateRow, YtEncodedColumn,
+    VT_NULL, VT_INT64, VT_UINT64, VT_DOUBLE, VT_BOOLEAN,
+    EX_COLUMN, EX_LIT_I64, EX_LIT_NULL, EX_LIT_DOUBLE,
+    EX_ADD, EX_SUB, EX_MUL, EX_DIV, EX_MOD,
+    EX_EQ, EX_NE, EX_LT, EX_LE, EX_GT, EX_GE, EX_AND, EX_OR, EX_NOT,
+    AGG_SUM, AGG_SUM1, YT_OK,
+)
+
+__all__ = [
+    "col", "lit", "null", "litf", "Plan", "agg_sum", "agg_sum1",
+    "encode_int64", "encode_double", "Chunk", "oracle_execute",
+    "gpu_available", "gpu_execute", "gpu_partial", "gpu_merge",
+    "rows_from_rowset", "sort_rows",
+]
+
+
+# ---------------- expression builder ----------------
+
+class Expr:
+    """Builds YtExpr trees; keeps ctypes objects alive via _keep."""
+
+    def __init__(self, op, col=-1, lit_i64=0, lit_dbl=0.0, a=None, b=None):
+        self.c = YtExpr(op=op, col=col, lit_i64=lit_i64, lit_dbl=lit_dbl)
+        self._keep = []
+        if a is not None:
+            self.c.a = C.pointer(a.c)
+            self._keep.append(a)
+        if b is not None:
+            self.c.b = C.pointer(b.c)
+            self._keep.append(b)
+
+    def _bin(self, op, other):
+        return Expr(op, a=self, b=_coerce(other))
+
+    def __add__(self, o): return self._bin(EX_ADD, o)
+    def __sub__(self, o): return self._bin(EX_SUB, o)
+    def __mul__(self, o): return self._bin(EX_MUL, o)
+    def __floordiv__(self, o): return self._bin(EX_DIV, o)
+    def __mod__(self, o): return self._bin(EX_MOD, o)
+    def __eq__(self, o): return self._bin(EX_EQ, o)      # noqa: E704
+    def __ne__(self, o): return self._bin(EX_NE, o)      # noqa: E704
+    def __lt__(self, o): return self._bin(EX_LT, o)
+    def __le__(self, o): return self._bin(EX_LE, o)
+    def __gt__(self, o): return self._bin(EX_GT, o)
+    def __ge__(self, o): return self._bin(EX_GE, o)
+    def and_(self, o): return self._bin(EX_AND, o)
+    def or_(self, o): return self._bin(EX_OR, o)
+    def not_(self): return Expr(EX_NOT, a=self)
+    __hash__ = None
+
+
+def _coerce(x):
+    if isinstance(x, Expr):
+        return x
+    if isinstance(x, bool):
+        return lit(int(x))
+    if isinstance(x, int):
+        return lit(x)
+    if isinstance(x, float):
+        return litf(x)
+    if x is None:
+        return null()
+    raise TypeError(x)
+
+
+def col(i):
+    return Expr(EX_COLUMN, col=i)
+
+
+def lit(v):
+    return Expr(EX_LIT_I64, lit_i64=v)
+
+
+def litf(v):
+    return Expr(EX_LIT_DOUBLE, lit_dbl=v)
+
+
+def null():
+    return Expr(EX_LIT_NULL)
+
+
+def agg_sum(e):
+    return (AGG_SUM, e)
+
+
+def agg_sum1():
+    return (AGG_SUM1, None)
+
+
+class Plan:
+    """Mirror of the restated TQuery{WhereClause,GroupClause} shape."""
+
+    def __init__(self, filter=None, keys=(), aggs=(), projects=(), is_merge=False):
+        self.filter = filter
+        self.keys = list(keys)
+        self.aggs = list(aggs)
+        self.projects = list(projects)
+        self.is_merge = is_merge
+        self._build()
+
+    def _build(self):
+        self._keep = []
+        p = YtPlan()
+        if self.filter is not None:
+            p.filter = C.pointer(self.filter.c)
+            self._keep.append(self.filter)
+        p.key_count = len(self.keys)
+        if self.keys:
+            arr = (C.POINTER(YtExpr) * len(self.keys))()
+            for i, k in enumerate(self.keys):
+                arr[i] = C.pointer(k.c)
+                self._keep.append(k)
+            p.keys = arr
+            self._keep.append(arr)
+        p.agg_count = len(self.aggs)
+        if self.aggs:
+            aggobjs = []
+            arr = (C.POINTER(YtAgg) * len(self.aggs))()
+            for i, (func, arg) in enumerate(self.aggs):
+                a = YtAgg(func=func)
+                if arg is not None:
+                    a.arg = C.pointer(arg.c)
+                    self._keep.append(arg)
+                aggobjs.append(a)
+                arr[i] = C.pointer(a)
+            p.aggs = arr
+            self._keep += [aggobjs, arr]
+        p.project_count = len(self.projects)
+        if self.projects:
+            arr = (C.POINTER(YtExpr) * len(self.projects))()
+            for i, e in enumerate(self.projects):
+                arr[i] = C.pointer(e.c)
+                self._keep.append(e)
+            p.projects = arr
+            self._keep.append(arr)
+        p.is_merge = 1 if self.is_merge else 0
+        self.c = p
+
+
+# ---------------- chunk encoding ----------------
+
+class EncodedColumn:
+    """Owns a YtEncodedColumn (host blob) produced by the product encoder."""
+
+    def __init__(self, value_type, cenc):
+        self.value_type = value_type
+        self._cenc = cenc
+        self.segments = [cenc.segments[i] for i in range(cenc.segment_count)]
+
+    def __del__(self):
+        try:
+            _abi.gpu_lib().yt_encoded_column_free(C.byref(self._cenc))
+        except Exception:
+            pass
+
+    def segment_blobs(self):
+        """list of (type, row_count, min_value, bytes)"""
+        out = []
+        for s in self.segments:
+            buf = C.string_at(s.data, s.data_size)
+            out.append((s.type, s.row_count, s.min_value, buf))
+        return out
+
+
+def _check(rc, err):
+    if rc != YT_OK:
+        raise RuntimeError("ytql error %d: %s" % (rc, err.value.decode()))
+
+
+def encode_int64(values, nulls=None, max_segment_values=0, unsigned=False):
+    values = np.ascontiguousarray(values, dtype=np.int64)
+    n = len(values)
+    nullp = None
+    if nulls is not None:
+        nulls = np.ascontiguousarray(nulls, dtype=np.uint8)
+        assert len(nulls) == n
+        nullp = nulls.ctypes.data_as(C.POINTER(C.c_uint8))
+    enc = YtEncodedColumn()
+    err = C.create_string_buffer(256)
+    rc = _abi.gpu_lib().yt_encode_int64_column(
+        values.ctypes.data_as(C.POINTER(C.c_int64)), nullp, n,
+        max_segment_values, 1 if unsigned else 0, C.byref(enc), err, 256)
+    _check(rc, err)
+    vt = VT_UINT64 if unsigned else VT_INT64
+    return EncodedColumn(vt, enc)
+
+
+def encode_double(values, nulls=None, max_segment_values=0):
+    values = np.ascontiguousarray(values, dtype=np.float64)
+    n = len(values)
+    nullp = None
+    if nulls is not None:
+        nulls = np.ascontiguousarray(nulls, dtype=np.uint8)
+        nullp = nulls.ctypes.data_as(C.POINTER(C.c_uint8))
+    enc = YtEncodedColumn()
+    err = C.create_string_buffer(256)
+    rc = _abi.gpu_lib().yt_encode_double_column(
+        values.ctypes.data_as(C.POINTER(C.c_double)), nullp, n,
+        max_segment_values, C.byref(enc), err, 256)
+    _check(rc, err)
+    return EncodedColumn(VT_DOUBLE, enc)
+
+
+class Chunk:
+    """A set of encoded columns; can present host- or device-pointer views."""
+
+    def __init__(self, columns, row_count):
+        self.columns = columns       # list of EncodedColumn
+        self.row_count = row_count
+        self._keep = []
+
+    def c_host(self):
+        """YtChunk with host segment pointers (for the oracle)."""
+        cols = (YtColumn * len(self.columns))()
+        keep = [cols]
+        for i, ec in enumerate(self.columns):
+            nseg = ec._cenc.segment_count
+            cols[i] = YtColumn(value_type=ec.value_type, segment_count=nseg,
+                               segments=ec._cenc.segments)
+        ch = YtChunk(row_count=self.row_count, column_count=len(self.columns),
+                     columns=cols)
+        self._keep.append(keep)
+        return ch
+
+    def c_device(self, torch):
+        """YtChunk whose segment data pointers live in HBM. Each column's
+        whole encoder blob is uploaded once as a torch uint8 tensor (torch =
+        device allocator only); segment pointers are offsets into it."""
+        cols = (YtColumn * len(self.columns))()
+        keep = [cols]
+        for i, ec in enumerate(self.columns):
+            nseg = ec._cenc.segment_count
+            segs = (YtSegment * nseg)()
+            keep.append(segs)
+            blob_base = ec._cenc.blob
+            blob_size = ec._cenc.blob_size
+            host = np.frombuffer(C.string_at(blob_base, blob_size), dtype=np.uint8)
+            t = torch.from_numpy(host.copy()).cuda()
+            keep.append(t)
+            for j in range(nseg):
+                s = ec._cenc.segments[j]
+                off = s.data - blob_base
+                segs[j] = YtSegment(type=s.type, row_count=s.row_count,
+                                    min_value=s.min_value,
+                                    data=t.data_ptr() + off, data_size=s.data_size)
+            cols[i] = YtColumn(value_type=ec.value_type, segment_count=nseg,
+                               segments=segs)
+        ch = YtChunk(row_count=self.row_count, column_count=len(self.columns),
+                     columns=cols)
+        self._keep.append(keep)
+        return ch
+
+
+# ---------------- row extraction ----------------
+
+def rows_from_rowset(rs):
+    out = []
+    ncols = rs.column_count
+    for r in range(rs.row_count):
+        row = []
+        for c in range(ncols):
+            v = rs.values[r * ncols + c]
+            if v.type == VT_NULL:
+                row.append(None)
+            elif v.type == VT_INT64:
+                row.append(int(C.c_int64(v.data.i64).value))
+            elif v.type == VT_UINT64:
+                row.append(int(v.data.u64))
+            elif v.type == VT_DOUBLE:
+                row.append(float(v.data.dbl))
+            elif v.type == VT_BOOLEAN:
+                row.append(bool(v.data.bits))
+            else:
+                row.append(("?", v.type, v.data.bits))
+        out.append(tuple(row))
+    return out
+
+
+def sort_rows(rows):
+    """Order-insensitive compare helper — mirrors the reference's
+    OrderedResultMatcher (unittests/evaluate/test_evaluate.cpp:174-199):
+    group-by output order is hash-insert order, so both sides are sorted."""
+    def key(row):
+        return tuple((x is None, isinstance(x, bool), x if x is not None else 0)
+                     for x in row)
+    return sorted(rows, key=key)
+
+
+def _mk_rowset(capacity, ncols_max=8):
+    buf = (YtValue * (capacity * ncols_max))()
+    rs = YtRowset(values=buf, capacity_rows=capacity)
+    rs._buf = buf  # keep alive
+    return rs
+
+
+# ---------------- execution ----------------
+
+def oracle_execute(plan, chunk, nthreads=1, expect_error=False):
+    """TEST/BASELINE ONLY — runs the CPU oracle restatement."""
+    ch = chunk.c_host()
+    rs = _mk_rowset(max(chunk.row_count + 16, 1 << 16))
+    st = YtStatistics()
+    err = C.create_string_buffer(256)
+    rc = _abi.oracle_lib().yto_execute(C.byref(plan.c), C.byref(ch), C.byref(rs),
+                                       C.byref(st), nthreads, err, 256)
+    if expect_error:
+        return rc, rows_from_rowset(rs), st
+    _check(rc, err)
+    return rows_from_rowset(rs), st
+
+
+def oracle_partial(plan, chunk, nparts, nthreads=1):
+    ch = chunk.c_host()
+    cap = chunk.row_count + 16
+    states = (YtStateRow * cap)()
+    counts = (C.c_int64 * nparts)()
+    err = C.create_string_buffer(256)
+    rc = _abi.oracle_lib().yto_partial(C.byref(plan.c), C.byref(ch), nparts,
+                                       states, cap, counts, nthreads, err, 256)
+    _check(rc, err)
+    return states, [counts[i] for i in range(nparts)]
+
+
+def oracle_merge(plan, states_list):
+    """states_list: list of (YtStateRow array-like, count) or a flat numpy
+    structured buffer."""
+    total = sum(n for _, n in states_list)
+    flat = (YtStateRow * max(total, 1))()
+    at = 0
+    for arr, n in states_list:
+        for i in range(n):
+            flat[at] = arr[i]
+            at += 1
+    rs = _mk_rowset(max(total + 16, 1024))
+    err = C.create_string_buffer(256)
+    rc = _abi.oracle_lib().yto_merge(C.byref(plan.c), flat, total, C.byref(rs),
+                                     err, 256)
+    _check(rc, err)
+    return rows_from_rowset(rs)
+
+
+def gpu_available():
+    err = C.create_string_buffer(256)
+    return _abi.gpu_lib().yt_gpu_available(err, 256) == YT_OK
+
+
+def gpu_execute(plan, device_chunk, max_groups_hint=0, group_row_limit=0,
+                stream=0, out_capacity=None):
+    """device_chunk: YtChunk with device pointers (Chunk.c_device)."""
+    opts = YtExecOptions(max_groups_hint=max_groups_hint,
+                         group_row_limit=group_row_limit, stream=stream)
+    cap = out_capacity or max(int(max_groups_hint) * 2 + 1024, 1 << 16)
+    rs = _mk_rowset(cap)
+    st = YtStatistics()
+    err = C.create_string_buffer(512)
+    rc = _abi.gpu_lib().yt_gpu_query_execute(
+        C.byref(plan.c), C.byref(device_chunk), C.byref(opts), C.byref(rs),
+        C.byref(st), err, 512)
+    _check(rc, err)
+    return rows_from_rowset(rs), st
+
+
+def gpu_partial(plan, device_chunk, nparts, states_dev_ptr, capacity_rows,
+                max_groups_hint=0, stream=0):
+    opts = YtExecOptions(max_groups_hint=max_groups_hint, stream=stream)
+    counts = (C.c_int64 * nparts)()
+    st = YtStatistics()
+    err = C.create_string_buffer(512)
+    rc = _abi.gpu_lib().yt_gpu_query_partial(
+        C.byref(plan.c), C.byref(device_chunk), C.byref(opts), nparts,
+        C.c_void_p(states_dev_ptr), capacity_rows, counts, C.byref(st), err, 512)
+    _check(rc, err)
+    return [counts[i] for i in range(nparts)], st
+
+
+def gpu_merge(plan, states_dev_ptr, n_states, max_groups_hint=0, stream=0,
+              out_capacity=None):
+    opts = YtExecOptions(max_groups_hint=max_groups_hint, stream=stream)
+    cap = out_capacity or max(n_states + 1024, 1 << 16)
+    rs = _mk_rowset(cap)
+    st = YtStatistics()
+    err = C.create_string_buffer(512)
+    rc = _abi.gpu_lib().yt_gpu_merge_states(
+        C.byref(plan.c), C.c_void_p(states_dev_ptr), n_states, C.byref(opts),
+        C.byref(rs), C.byref(st), err, 512)
+    _check(rc, err)
+    return rows_from_rowset(rs), st

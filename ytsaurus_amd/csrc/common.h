@@ -1,0 +1,141 @@
+/* Internal structures shared between the host evaluator and the HIP kernels.
+ * Product code (libytql_gpu.so). Not part of the public C-ABI
+ * (see /root/repo/include/ytql_gpu.h for the boundary).
+ */
+#pragma once
+#include <stdint.h>
+
+namespace ytql {
+
+/* Resolved device-side segment descriptor. The kernel parses the sub-blob
+ * offsets itself from the 8-byte bit-pack headers
+ * (reference layout: bit_packed_unsigned_vector-inl.h:34-45 header word =
+ * count(56b)|width(8b); segment assembly integer_column_writer.cpp:66-116,
+ * 394-489). */
+struct DevSeg {
+    int32_t type;        /* YT_SEG_* */
+    int32_t is_signed;   /* 1 = int64 (zigzag), 0 = uint64 */
+    int64_t start_row;   /* first chunk row covered */
+    int32_t row_count;
+    int32_t pad_;
+    uint64_t min_value;
+    const uint64_t* blob;
+    int64_t blob_bytes;
+};
+
+/* Parsed per-segment sub-blob map (built on device by k_parse_segments). */
+struct SegEx {
+    uint32_t w_values;
+    uint32_t w_ids;
+    uint32_t w_starts;
+    uint32_t run_count;       /* RLE run count / dictionary id count */
+    uint32_t dict_size;
+    uint32_t flags;
+    int64_t off_values_words; /* first data word of the values vector (past header) */
+    int64_t off_bitmap_bytes; /* null bitmap (direct formats) */
+    int64_t off_ids_words;    /* dictionary ids / RLE ids data words */
+    int64_t off_starts_words; /* RLE run-start data words */
+    int64_t off_doubles_bytes;
+};
+
+/* Postfix expression program; mirrors the oracle's eval_expr semantics
+ * (cg_fragment_compiler.cpp arithmetic/relational/logical codegen). */
+enum POp : int32_t {
+    P_COL = 0, P_LIT_I64 = 1, P_LIT_NULL = 2, P_LIT_DOUBLE = 3,
+    P_ADD = 10, P_SUB = 11, P_MUL = 12, P_DIV = 13, P_MOD = 14,
+    P_EQ = 20, P_NE = 21, P_LT = 22, P_LE = 23, P_GT = 24, P_GE = 25,
+    P_AND = 30, P_OR = 31, P_NOT = 32,
+};
+
+struct PInst {
+    int32_t op;
+    int32_t col;
+    uint64_t bits;     /* literal payload */
+};
+
+constexpr int kMaxProg = 48;
+constexpr int kMaxCols = 8;
+constexpr int kMaxAggs = 4;
+
+/* One compiled plan for the device: programs are concatenated postfix
+ * streams with (offset,len) per role. */
+struct DevPlan {
+    int32_t ncols;
+    uint8_t col_types[kMaxCols];       /* YT_VT_* */
+    int32_t filter_off, filter_len;    /* -1 len 0 = none */
+    int32_t key_off, key_len;          /* key_count==1 only; len 0 = global agg */
+    int32_t agg_count;
+    int32_t agg_func[kMaxAggs];        /* YT_AGG_* */
+    int32_t agg_off[kMaxAggs], agg_len[kMaxAggs];
+    PInst prog[kMaxProg];
+    int32_t prog_len;
+};
+
+/* Group hash-table slot layout (generic path):
+ *   u64 key_bits | u64 cnt | per-agg { u64 bits, u64 nonnull }
+ * stride_u64 = 2 + 2*agg_count. key_bits == 0 means EMPTY; real key 0 and
+ * null key live in the side accumulators of TableHdr. */
+struct TableHdr {
+    uint64_t nslots;          /* power of two */
+    uint64_t mask;
+    /* side groups: [0] = key bits == 0, [1] = null key */
+    uint64_t side_used[2];    /* 0/1 */
+    uint64_t side_cnt[2];
+    uint64_t side_agg[2][2 * kMaxAggs];  /* bits, nonnull per agg */
+    unsigned long long ngroups;          /* inserted (excl. side) */
+    uint64_t overflow;        /* set when probe failed / group limit hit */
+    int64_t group_limit;      /* 0 = unlimited */
+};
+
+/* Compacted output record (device→host), generic path */
+struct OutGroup {
+    uint64_t key_bits;
+    uint64_t key_meta;        /* bit0 null-key */
+    uint64_t cnt;
+    uint64_t agg_bits[kMaxAggs];
+    uint64_t agg_nonnull[kMaxAggs];
+};
+
+/* Fast-shape descriptor (analysed by the host):
+ * filter: optional int64 range on one column; key: direct column or none;
+ * aggs: sum(direct col) and/or sum(1). */
+struct FastShape {
+    int32_t valid;
+    int32_t filter_col;       /* -1 = none */
+    int64_t filter_lo, filter_hi;   /* inclusive */
+    int32_t key_col;          /* -1 = global */
+    int32_t nsum;             /* number of sum(col) aggs */
+    int32_t sum_col[kMaxAggs];
+    int32_t sum_slot[kMaxAggs];     /* agg index of each sum */
+    int32_t have_sum1;
+};
+
+/* fast-kernel launch descriptors */
+struct FastCol {
+    int32_t seg_off;      /* into segs/segex: first segment of this column */
+    int32_t seg_cnt;
+};
+
+struct FastParams {
+    int32_t nused;               /* staged columns */
+    int32_t tile_rows;
+    int32_t tiles_per_seg;       /* uniform by the 128Ki interior-segment cap */
+    int32_t filter_idx;          /* index into used cols; -1 none */
+    int32_t key_idx;             /* -1 = global aggregate */
+    int32_t nsum;
+    int32_t sum_idx[kMaxAggs];   /* used-col index of each sum arg */
+    int32_t sum_slot[kMaxAggs];  /* agg slot in table layout */
+    int32_t agg_count;           /* table stride basis */
+    int64_t filter_lo, filter_hi;
+    int64_t row_count;
+    int32_t nsegs_per_col;
+    int32_t ntiles;
+};
+
+struct KernelTimes {
+    float scan_ms;
+    int64_t scan_launches;
+    float other_ms;
+};
+
+} /* namespace ytql */

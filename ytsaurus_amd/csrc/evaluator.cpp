@@ -1,0 +1,1050 @@
+/* evaluator.cpp — host side of the MI355X query executor: the C-ABI
+ * implementation of the IEvaluator::Run seam (reference
+ * engine/evaluator.cpp:51-105), plan compilation (AOT replacement of the
+ * LLVM codegen in folding_profiler.cpp / cg_fragment_compiler.cpp), device
+ * orchestration, limits and statistics (TExecutionContext /
+ * TQueryStatistics, engine_api/evaluation_helpers.h:251-280).
+ *
+ * Product code. No CPU fallback: without a HIP device every execute entry
+ * returns YT_ERR_NO_GPU / YT_ERR_HIP.
+ */
+#include <hip/hip_runtime.h>
+#include <stdint.h>
+#include <string.h>
+#include <stdio.h>
+#include <stdlib.h>
+#include <vector>
+#include <algorithm>
+
+#include "common.h"
+#include "../../include/ytql_gpu.h"
+
+using namespace ytql;
+
+/* launch wrappers from kernels.hip */
+extern "C" {
+hipError_t ytql_launch_parse_segments(const DevSeg*, int, SegEx*, unsigned*, hipStream_t);
+hipError_t ytql_launch_scan_generic(const DevPlan*, const DevSeg*, const SegEx*,
+                                    const int32_t*, const int32_t*, int64_t,
+                                    TableHdr*, unsigned long long*, unsigned*, hipStream_t);
+hipError_t ytql_launch_scan_fast(const FastParams*, const DevSeg*, const SegEx*,
+                                 const FastCol*, TableHdr*, unsigned long long*,
+                                 unsigned long long*, size_t, int, hipStream_t);
+hipError_t ytql_launch_compact(TableHdr*, TableHdr*, const unsigned long long*, int,
+                               OutGroup*, unsigned long long*, uint64_t, hipStream_t);
+hipError_t ytql_launch_part_count(const OutGroup*, int64_t, int, int,
+                                  unsigned long long*, hipStream_t);
+hipError_t ytql_launch_part_scatter(const OutGroup*, int64_t, int, int,
+                                    unsigned long long*, YtStateRow*, hipStream_t);
+hipError_t ytql_launch_merge_states(const YtStateRow*, int64_t, int, int,
+                                    TableHdr*, unsigned long long*, hipStream_t);
+}
+
+/* ------------------------------------------------------------------ */
+
+static void set_err(char* errbuf, size_t errlen, const char* msg)
+{
+    if (errbuf && errlen) snprintf(errbuf, errlen, "%s", msg);
+}
+
+#define HIP_CHECK(call)                                                     \
+    do {                                                                    \
+        hipError_t e_ = (call);                                             \
+        if (e_ != hipSuccess) {                                             \
+            if (errbuf) snprintf(errbuf, errlen, "HIP error %s at %s:%d",   \
+                                 hipGetErrorString(e_), __FILE__, __LINE__);\
+            rc = YT_ERR_HIP;                                                \
+            goto fail;                                                      \
+        }                                                                   \
+    } while (0)
+
+extern "C" int yt_gpu_available(char* errbuf, size_t errlen)
+{
+    int n = 0;
+    hipError_t e = hipGetDeviceCount(&n);
+    if (e != hipSuccess || n == 0) {
+        set_err(errbuf, errlen, "no HIP device");
+        return YT_ERR_NO_GPU;
+    }
+    return YT_OK;
+}
+
+/* ------------------------------------------------------------------ */
+/* plan compilation: YtExpr tree → postfix DevPlan program              */
+
+static int compile_expr(const YtExpr* e, DevPlan* p, int* len, char* errbuf, size_t errlen)
+{
+    if (!e) { set_err(errbuf, errlen, "null expr"); return YT_ERR_INVALID_PLAN; }
+    switch (e->op) {
+    case YT_EX_COLUMN:
+    case YT_EX_LIT_I64:
+    case YT_EX_LIT_NULL:
+    case YT_EX_LIT_DOUBLE:
+        break;
+    case YT_EX_NOT: {
+        int rc = compile_expr(e->a, p, len, errbuf, errlen);
+        if (rc) return rc;
+        break;
+    }
+    default: {
+        int rc = compile_expr(e->a, p, len, errbuf, errlen);
+        if (rc) return rc;
+        rc = compile_expr(e->b, p, len, errbuf, errlen);
+        if (rc) return rc;
+        break;
+    }
+    }
+    if (p->prog_len >= kMaxProg) { set_err(errbuf, errlen, "expression too long"); return YT_ERR_INVALID_PLAN; }
+    PInst& in = p->prog[p->prog_len++];
+    in.op = e->op;     /* YT_EX_* values coincide with P_* by construction */
+    in.col = e->col;
+    in.bits = 0;
+    if (e->op == YT_EX_LIT_I64) in.bits = (uint64_t)e->lit_i64;
+    if (e->op == YT_EX_LIT_DOUBLE) memcpy(&in.bits, &e->lit_dbl, 8);
+    if (e->op == YT_EX_COLUMN && (e->col < 0 || e->col >= p->ncols)) {
+        set_err(errbuf, errlen, "column index out of range");
+        return YT_ERR_INVALID_PLAN;
+    }
+    (*len)++;
+    return YT_OK;
+}
+
+/* static result type of an expression (for output formatting) */
+static uint8_t expr_static_type(const YtExpr* e, const uint8_t* col_types)
+{
+    switch (e->op) {
+    case YT_EX_COLUMN: return col_types[e->col];
+    case YT_EX_LIT_I64: return YT_VT_INT64;
+    case YT_EX_LIT_DOUBLE: return YT_VT_DOUBLE;
+    case YT_EX_LIT_NULL: return YT_VT_NULL;
+    case YT_EX_NOT: return YT_VT_BOOLEAN;
+    default:
+        if (e->op >= YT_EX_EQ) return YT_VT_BOOLEAN;   /* cmp / and / or */
+        return expr_static_type(e->a, col_types);      /* arithmetic */
+    }
+}
+
+static int build_devplan(const YtPlan* plan, const YtChunk* chunk, DevPlan* p,
+                         char* errbuf, size_t errlen)
+{
+    memset(p, 0, sizeof(*p));
+    p->ncols = chunk->column_count;
+    if (p->ncols > kMaxCols) { set_err(errbuf, errlen, "too many columns"); return YT_ERR_UNSUPPORTED; }
+    for (int c = 0; c < p->ncols; c++) p->col_types[c] = (uint8_t)chunk->columns[c].value_type;
+
+    if (plan->key_count > 1) { set_err(errbuf, errlen, "GPU path: at most 1 group key this round"); return YT_ERR_UNSUPPORTED; }
+    if (plan->agg_count < 1 || plan->agg_count > kMaxAggs) { set_err(errbuf, errlen, "need 1..4 aggregates"); return YT_ERR_UNSUPPORTED; }
+    for (int a = 0; a < plan->agg_count; a++) {
+        if (plan->aggs[a]->func != YT_AGG_SUM && plan->aggs[a]->func != YT_AGG_SUM1) {
+            set_err(errbuf, errlen, "GPU path: sum/sum(1) aggregates this round");
+            return YT_ERR_UNSUPPORTED;
+        }
+    }
+
+    int rc;
+    if (plan->filter) {
+        p->filter_off = p->prog_len;
+        rc = compile_expr(plan->filter, p, &p->filter_len, errbuf, errlen);
+        if (rc) return rc;
+        p->filter_len = p->prog_len - p->filter_off;
+    }
+    if (plan->key_count == 1) {
+        p->key_off = p->prog_len;
+        rc = compile_expr(plan->keys[0], p, &p->key_len, errbuf, errlen);
+        if (rc) return rc;
+        p->key_len = p->prog_len - p->key_off;
+    }
+    p->agg_count = plan->agg_count;
+    for (int a = 0; a < plan->agg_count; a++) {
+        p->agg_func[a] = plan->aggs[a]->func;
+        if (plan->aggs[a]->func == YT_AGG_SUM) {
+            p->agg_off[a] = p->prog_len;
+            rc = compile_expr(plan->aggs[a]->arg, p, &p->agg_len[a], errbuf, errlen);
+            if (rc) return rc;
+            p->agg_len[a] = p->prog_len - p->agg_off[a];
+        }
+    }
+    return YT_OK;
+}
+
+/* fast-shape analysis: filter = none | cmp(col, lit) | AND of two bounds;
+ * key = direct int64 column | none; sum args = direct int64 columns */
+static bool expr_is_col(const YtExpr* e, int* col)
+{
+    if (e && e->op == YT_EX_COLUMN) { *col = e->col; return true; }
+    return false;
+}
+
+static bool match_bound(const YtExpr* e, const uint8_t* col_types,
+                        int* col, int64_t* lo, int64_t* hi)
+{
+    /* col cmp lit / lit cmp col, int64 only */
+    int c;
+    int64_t k;
+    int op = e->op;
+    if (op < YT_EX_EQ || op > YT_EX_GE || op == YT_EX_NE) return false;
+    if (expr_is_col(e->a, &c) && e->b && e->b->op == YT_EX_LIT_I64) {
+        k = e->b->lit_i64;
+    } else if (expr_is_col(e->b, &c) && e->a && e->a->op == YT_EX_LIT_I64) {
+        k = e->a->lit_i64;
+        /* mirror: lit cmp col → col cmp' lit */
+        switch (op) {
+        case YT_EX_LT: op = YT_EX_GT; break;
+        case YT_EX_LE: op = YT_EX_GE; break;
+        case YT_EX_GT: op = YT_EX_LT; break;
+        case YT_EX_GE: op = YT_EX_LE; break;
+        default: break;
+        }
+    } else {
+        return false;
+    }
+    if (col_types[c] != YT_VT_INT64) return false;
+    *col = c;
+    *lo = INT64_MIN; *hi = INT64_MAX;
+    switch (op) {
+    case YT_EX_EQ: *lo = *hi = k; break;
+    case YT_EX_LT: if (k == INT64_MIN) return false; *hi = k - 1; break;
+    case YT_EX_LE: *hi = k; break;
+    case YT_EX_GT: if (k == INT64_MAX) return false; *lo = k + 1; break;
+    case YT_EX_GE: *lo = k; break;
+    default: return false;
+    }
+    return true;
+}
+
+static void analyze_fast(const YtPlan* plan, const YtChunk* chunk, FastShape* fs)
+{
+    memset(fs, 0, sizeof(*fs));
+    fs->filter_col = -1;
+    fs->key_col = -1;
+
+    const uint8_t* ct = nullptr;
+    static uint8_t types[kMaxCols];
+    for (int c = 0; c < chunk->column_count && c < kMaxCols; c++)
+        types[c] = (uint8_t)chunk->columns[c].value_type;
+    ct = types;
+
+    /* all segments of all columns must be DirectDense int64, uniform interior
+     * row counts */
+    int32_t seg_cnt0 = chunk->columns[0].segment_count;
+    int32_t rows0 = seg_cnt0 ? chunk->columns[0].segments[0].row_count : 0;
+    for (int c = 0; c < chunk->column_count; c++) {
+        const YtColumn& col = chunk->columns[c];
+        if (col.value_type != YT_VT_INT64) return;
+        if (col.segment_count != seg_cnt0) return;
+        for (int s = 0; s < col.segment_count; s++) {
+            if (col.segments[s].type != YT_SEG_DIRECT_DENSE) return;
+            if (s < col.segment_count - 1 && col.segments[s].row_count != rows0) return;
+            if (col.segments[s].row_count > rows0) return;
+        }
+    }
+    if (seg_cnt0 == 0) return;
+
+    if (plan->filter) {
+        int64_t lo, hi;
+        int col;
+        if (match_bound(plan->filter, ct, &col, &lo, &hi)) {
+            fs->filter_col = col; fs->filter_lo = lo; fs->filter_hi = hi;
+        } else if (plan->filter->op == YT_EX_AND) {
+            int c1, c2;
+            int64_t lo1, hi1, lo2, hi2;
+            if (!match_bound(plan->filter->a, ct, &c1, &lo1, &hi1)) return;
+            if (!match_bound(plan->filter->b, ct, &c2, &lo2, &hi2)) return;
+            if (c1 != c2) return;
+            fs->filter_col = c1;
+            fs->filter_lo = std::max(lo1, lo2);
+            fs->filter_hi = std::min(hi1, hi2);
+        } else {
+            return;
+        }
+    }
+    if (plan->key_count == 1) {
+        int col;
+        if (!expr_is_col(plan->keys[0], &col)) return;
+        if (ct[col] != YT_VT_INT64) return;
+        fs->key_col = col;
+    } else if (plan->key_count > 1) {
+        return;
+    }
+    fs->nsum = 0;
+    for (int a = 0; a < plan->agg_count; a++) {
+        if (plan->aggs[a]->func == YT_AGG_SUM1) { fs->have_sum1 = 1; continue; }
+        if (plan->aggs[a]->func != YT_AGG_SUM) return;
+        int col;
+        if (!expr_is_col(plan->aggs[a]->arg, &col)) return;
+        if (ct[col] != YT_VT_INT64) return;
+        if (fs->nsum >= kMaxAggs) return;
+        fs->sum_col[fs->nsum] = col;
+        fs->sum_slot[fs->nsum] = a;
+        fs->nsum++;
+    }
+    fs->valid = 1;
+}
+
+/* ------------------------------------------------------------------ */
+/* host-side projection evaluation over a finalized group row           */
+/* (mirrors MakeCodegenProjectOp over [keys..., aggregates...])        */
+
+struct HVal { uint8_t type; uint64_t bits; };
+
+static int heval(const YtExpr* e, const HVal* row, int nrow, HVal* out)
+{
+    switch (e->op) {
+    case YT_EX_COLUMN:
+        if (e->col < 0 || e->col >= nrow) return YT_ERR_INVALID_PLAN;
+        *out = row[e->col];
+        return YT_OK;
+    case YT_EX_LIT_I64: out->type = YT_VT_INT64; out->bits = (uint64_t)e->lit_i64; return YT_OK;
+    case YT_EX_LIT_DOUBLE: out->type = YT_VT_DOUBLE; memcpy(&out->bits, &e->lit_dbl, 8); return YT_OK;
+    case YT_EX_LIT_NULL: out->type = YT_VT_NULL; out->bits = 0; return YT_OK;
+    case YT_EX_NOT: {
+        HVal a;
+        int rc = heval(e->a, row, nrow, &a);
+        if (rc) return rc;
+        if (a.type == YT_VT_NULL) { *out = a; return YT_OK; }
+        out->type = YT_VT_BOOLEAN; out->bits = !a.bits;
+        return YT_OK;
+    }
+    default: break;
+    }
+    HVal a, b;
+    int rc = heval(e->a, row, nrow, &a);
+    if (rc) return rc;
+    rc = heval(e->b, row, nrow, &b);
+    if (rc) return rc;
+    out->type = YT_VT_NULL; out->bits = 0;
+    if (e->op >= YT_EX_ADD && e->op <= YT_EX_MOD) {
+        if (a.type == YT_VT_NULL || b.type == YT_VT_NULL) return YT_OK;
+        if (a.type == YT_VT_DOUBLE) {
+            double x, y, z = 0;
+            memcpy(&x, &a.bits, 8);
+            memcpy(&y, &b.bits, 8);
+            switch (e->op) {
+            case YT_EX_ADD: z = x + y; break;
+            case YT_EX_SUB: z = x - y; break;
+            case YT_EX_MUL: z = x * y; break;
+            case YT_EX_DIV: z = x / y; break;
+            default: return YT_ERR_UNSUPPORTED;
+            }
+            out->type = YT_VT_DOUBLE;
+            memcpy(&out->bits, &z, 8);
+            return YT_OK;
+        }
+        uint64_t x = a.bits, y = b.bits, z = 0;
+        int sgn = (a.type == YT_VT_INT64);
+        switch (e->op) {
+        case YT_EX_ADD: z = x + y; break;
+        case YT_EX_SUB: z = x - y; break;
+        case YT_EX_MUL: z = x * y; break;
+        case YT_EX_DIV:
+        case YT_EX_MOD:
+            if (y == 0) return YT_ERR_DIV_ZERO;
+            if (sgn) {
+                int64_t sx = (int64_t)x, sy = (int64_t)y;
+                if (sx == INT64_MIN && sy == -1) z = (e->op == YT_EX_DIV) ? (uint64_t)INT64_MIN : 0;
+                else z = (uint64_t)((e->op == YT_EX_DIV) ? sx / sy : sx % sy);
+            } else {
+                z = (e->op == YT_EX_DIV) ? x / y : x % y;
+            }
+            break;
+        }
+        out->type = a.type;
+        out->bits = z;
+        return YT_OK;
+    }
+    if (e->op >= YT_EX_EQ && e->op <= YT_EX_GE) {
+        int lt, eq, force_true = 0;
+        if (a.type == YT_VT_NULL || b.type == YT_VT_NULL) {
+            unsigned ln = (a.type == YT_VT_NULL), rn = (b.type == YT_VT_NULL);
+            lt = rn < ln; eq = ln == rn;
+        } else if (a.type == YT_VT_DOUBLE) {
+            double x, y;
+            memcpy(&x, &a.bits, 8);
+            memcpy(&y, &b.bits, 8);
+            int unordered = (x != x) || (y != y);
+            lt = unordered || (x < y);
+            eq = unordered || (x == y);
+            force_true = unordered;
+        } else if (a.type == YT_VT_INT64) {
+            int64_t x = (int64_t)a.bits, y = (int64_t)b.bits;
+            lt = x < y; eq = x == y;
+        } else {
+            lt = a.bits < b.bits; eq = a.bits == b.bits;
+        }
+        int r = 1;
+        if (!force_true) {
+            switch (e->op) {
+            case YT_EX_EQ: r = eq; break;
+            case YT_EX_NE: r = !eq; break;
+            case YT_EX_LT: r = lt; break;
+            case YT_EX_LE: r = lt || eq; break;
+            case YT_EX_GT: r = !(lt || eq); break;
+            case YT_EX_GE: r = !lt; break;
+            }
+        }
+        out->type = YT_VT_BOOLEAN;
+        out->bits = (uint64_t)r;
+        return YT_OK;
+    }
+    if (e->op == YT_EX_AND || e->op == YT_EX_OR) {
+        int an = (a.type == YT_VT_NULL), bn = (b.type == YT_VT_NULL);
+        int av = an ? 0 : (a.bits != 0), bv = bn ? 0 : (b.bits != 0);
+        if (e->op == YT_EX_AND) {
+            if ((!an && !av) || (!bn && !bv)) { out->type = YT_VT_BOOLEAN; out->bits = 0; }
+            else if (!an && !bn) { out->type = YT_VT_BOOLEAN; out->bits = 1; }
+        } else {
+            if ((!an && av) || (!bn && bv)) { out->type = YT_VT_BOOLEAN; out->bits = 1; }
+            else if (!an && !bn) { out->type = YT_VT_BOOLEAN; out->bits = 0; }
+        }
+        return YT_OK;
+    }
+    return YT_ERR_UNSUPPORTED;
+}
+
+/* ------------------------------------------------------------------ */
+/* device chunk setup                                                  */
+
+struct DeviceRun {
+    std::vector<DevSeg> h_segs;
+    std::vector<int32_t> h_off, h_cnt;
+    DevSeg* d_segs = nullptr;
+    SegEx* d_segex = nullptr;
+    int32_t* d_off = nullptr;
+    int32_t* d_cnt = nullptr;
+    unsigned* d_maxw = nullptr;
+    unsigned* d_err = nullptr;
+    TableHdr* d_th = nullptr;
+    unsigned long long* d_slots = nullptr;
+    OutGroup* d_groups = nullptr;
+    unsigned long long* d_counter = nullptr;
+    unsigned long long* d_gaccum = nullptr;
+    FastCol* d_fastcols = nullptr;
+    uint64_t nslots = 0;
+    int nsegs = 0;
+    hipStream_t stream = 0;
+
+    ~DeviceRun()
+    {
+        hipError_t e;
+        (void)e;
+        if (d_segs) e = hipFree(d_segs);
+        if (d_segex) e = hipFree(d_segex);
+        if (d_off) e = hipFree(d_off);
+        if (d_cnt) e = hipFree(d_cnt);
+        if (d_maxw) e = hipFree(d_maxw);
+        if (d_err) e = hipFree(d_err);
+        if (d_th) e = hipFree(d_th);
+        if (d_slots) e = hipFree(d_slots);
+        if (d_groups) e = hipFree(d_groups);
+        if (d_counter) e = hipFree(d_counter);
+        if (d_gaccum) e = hipFree(d_gaccum);
+        if (d_fastcols) e = hipFree(d_fastcols);
+    }
+};
+
+static uint64_t next_pow2(uint64_t x)
+{
+    uint64_t p = 1;
+    while (p < x) p <<= 1;
+    return p;
+}
+
+/* builds device segment descriptors; returns YT status */
+static int setup_chunk(const YtChunk* chunk, DeviceRun* R, unsigned* maxw_out,
+                       char* errbuf, size_t errlen)
+{
+    int rc = YT_OK;
+    int ncols = chunk->column_count;
+    R->h_off.resize(ncols);
+    R->h_cnt.resize(ncols);
+    for (int c = 0; c < ncols; c++) {
+        const YtColumn& col = chunk->columns[c];
+        R->h_off[c] = (int32_t)R->h_segs.size();
+        R->h_cnt[c] = col.segment_count;
+        int64_t row = 0;
+        for (int s = 0; s < col.segment_count; s++) {
+            const YtSegment& seg = col.segments[s];
+            DevSeg d;
+            d.type = seg.type;
+            d.is_signed = (col.value_type == YT_VT_INT64);
+            d.start_row = row;
+            d.row_count = seg.row_count;
+            d.pad_ = 0;
+            d.min_value = seg.min_value;
+            d.blob = (const uint64_t*)seg.data;
+            d.blob_bytes = seg.data_size;
+            R->h_segs.push_back(d);
+            row += seg.row_count;
+        }
+        if (row != chunk->row_count) {
+            set_err(errbuf, errlen, "segment row counts do not sum to chunk rows");
+            return YT_ERR_INVALID_CHUNK;
+        }
+    }
+    R->nsegs = (int)R->h_segs.size();
+    if (R->nsegs == 0) return YT_OK;
+
+    HIP_CHECK(hipMalloc(&R->d_segs, sizeof(DevSeg) * R->nsegs));
+    HIP_CHECK(hipMalloc(&R->d_segex, sizeof(SegEx) * R->nsegs));
+    HIP_CHECK(hipMalloc(&R->d_off, sizeof(int32_t) * ncols));
+    HIP_CHECK(hipMalloc(&R->d_cnt, sizeof(int32_t) * ncols));
+    HIP_CHECK(hipMalloc(&R->d_maxw, sizeof(unsigned)));
+    HIP_CHECK(hipMalloc(&R->d_err, sizeof(unsigned)));
+    HIP_CHECK(hipMemcpyAsync(R->d_segs, R->h_segs.data(), sizeof(DevSeg) * R->nsegs,
+                             hipMemcpyHostToDevice, R->stream));
+    HIP_CHECK(hipMemcpyAsync(R->d_off, R->h_off.data(), sizeof(int32_t) * ncols,
+                             hipMemcpyHostToDevice, R->stream));
+    HIP_CHECK(hipMemcpyAsync(R->d_cnt, R->h_cnt.data(), sizeof(int32_t) * ncols,
+                             hipMemcpyHostToDevice, R->stream));
+    HIP_CHECK(hipMemsetAsync(R->d_maxw, 0, sizeof(unsigned), R->stream));
+    HIP_CHECK(hipMemsetAsync(R->d_err, 0, sizeof(unsigned), R->stream));
+    HIP_CHECK(ytql_launch_parse_segments(R->d_segs, R->nsegs, R->d_segex, R->d_maxw, R->stream));
+    HIP_CHECK(hipMemcpyAsync(maxw_out, R->d_maxw, sizeof(unsigned),
+                             hipMemcpyDeviceToHost, R->stream));
+    HIP_CHECK(hipStreamSynchronize(R->stream));
+    return YT_OK;
+fail:
+    return rc;
+}
+
+static int setup_table(DeviceRun* R, int agg_count, int64_t max_groups,
+                       int64_t group_limit, char* errbuf, size_t errlen)
+{
+    int rc = YT_OK;
+    R->nslots = next_pow2((uint64_t)(max_groups > 0 ? max_groups : (1 << 20)) * 2);
+    if (R->nslots < 2048) R->nslots = 2048;
+    int stride = 2 + 2 * agg_count;
+    HIP_CHECK(hipMalloc(&R->d_th, sizeof(TableHdr)));
+    HIP_CHECK(hipMalloc(&R->d_slots, sizeof(uint64_t) * R->nslots * stride));
+    TableHdr hh;
+    memset(&hh, 0, sizeof(hh));
+    hh.nslots = R->nslots;
+    hh.mask = R->nslots - 1;
+    hh.group_limit = group_limit;
+    HIP_CHECK(hipMemcpyAsync(R->d_th, &hh, sizeof(hh), hipMemcpyHostToDevice, R->stream));
+    HIP_CHECK(hipMemsetAsync(R->d_slots, 0, sizeof(uint64_t) * R->nslots * stride, R->stream));
+    return YT_OK;
+fail:
+    return rc;
+}
+
+/* run the scan (fast when possible), leaving results in the table/gaccum.
+ * Fills stats->kernel_scan_* from device events. */
+static int run_scan(const YtPlan* plan, const YtChunk* chunk,
+                    const YtExecOptions* options, DeviceRun* R,
+                    const DevPlan* dp, const FastShape* fs, unsigned maxw,
+                    YtStatistics* stats, char* errbuf, size_t errlen)
+{
+    int rc = YT_OK;
+    hipEvent_t ev0 = nullptr, ev1 = nullptr;
+    HIP_CHECK(hipEventCreate(&ev0));
+    HIP_CHECK(hipEventCreate(&ev1));
+
+    HIP_CHECK(hipMalloc(&R->d_gaccum, sizeof(uint64_t) * (1 + 2 * kMaxAggs)));
+    HIP_CHECK(hipMemsetAsync(R->d_gaccum, 0, sizeof(uint64_t) * (1 + 2 * kMaxAggs), R->stream));
+
+    if (fs->valid) {
+        /* fast fused kernel */
+        FastParams fp;
+        memset(&fp, 0, sizeof(fp));
+        /* used columns: filter first (if distinct), then key, then sums */
+        int used[kMaxCols];
+        int nused = 0;
+        auto add_used = [&](int col) -> int {
+            for (int i = 0; i < nused; i++) if (used[i] == col) return i;
+            used[nused] = col;
+            return nused++;
+        };
+        fp.filter_idx = fs->filter_col >= 0 ? add_used(fs->filter_col) : -1;
+        fp.key_idx = fs->key_col >= 0 ? add_used(fs->key_col) : -1;
+        fp.nsum = fs->nsum;
+        for (int a = 0; a < fs->nsum; a++) {
+            fp.sum_idx[a] = add_used(fs->sum_col[a]);
+            fp.sum_slot[a] = fs->sum_slot[a];
+        }
+        fp.nused = nused;
+        fp.agg_count = plan->agg_count;
+        fp.filter_lo = fs->filter_lo;
+        fp.filter_hi = fs->filter_hi;
+        fp.row_count = chunk->row_count;
+        fp.nsegs_per_col = chunk->columns[0].segment_count;
+
+        int32_t seg0_rows = chunk->columns[0].segments[0].row_count;
+
+        /* LDS sizing: per col align16((tile_rows*w/64+2)*8) + align16(tile_rows/8) */
+        unsigned w = maxw ? maxw : 1;
+        int tile_rows = 8192;
+        size_t lds = 0;
+        for (;;) {
+            size_t per = ((((size_t)tile_rows * w / 64 + 2) * 8 + 15) & ~(size_t)15)
+                       + (((size_t)tile_rows / 8 + 15) & ~(size_t)15);
+            lds = (size_t)nused * per + 1024;
+            if (lds <= 144 * 1024 || tile_rows == 256) break;
+            tile_rows >>= 1;
+        }
+        if (lds > 160 * 1024) { set_err(errbuf, errlen, "LDS overflow"); rc = YT_ERR_UNSUPPORTED; goto fail; }
+        if (tile_rows > seg0_rows) {
+            /* keep at least one tile per segment */
+            while (tile_rows > 256 && tile_rows > seg0_rows) tile_rows >>= 1;
+        }
+        fp.tile_rows = tile_rows;
+        fp.tiles_per_seg = (seg0_rows + tile_rows - 1) / tile_rows;
+        {
+            int nseg = chunk->columns[0].segment_count;
+            int32_t last_rows = chunk->columns[0].segments[nseg - 1].row_count;
+            fp.ntiles = (nseg - 1) * fp.tiles_per_seg + (last_rows + tile_rows - 1) / tile_rows;
+        }
+
+        std::vector<FastCol> fc(nused);
+        for (int u = 0; u < nused; u++) {
+            fc[u].seg_off = R->h_off[used[u]];
+            fc[u].seg_cnt = R->h_cnt[used[u]];
+        }
+        HIP_CHECK(hipMalloc(&R->d_fastcols, sizeof(FastCol) * nused));
+        HIP_CHECK(hipMemcpyAsync(R->d_fastcols, fc.data(), sizeof(FastCol) * nused,
+                                 hipMemcpyHostToDevice, R->stream));
+
+        int grid = fp.ntiles < 2048 ? (fp.ntiles ? fp.ntiles : 1) : 2048;
+        HIP_CHECK(hipEventRecord(ev0, R->stream));
+        HIP_CHECK(ytql_launch_scan_fast(&fp, R->d_segs, R->d_segex, R->d_fastcols,
+                                        R->d_th, R->d_slots, R->d_gaccum,
+                                        lds, grid, R->stream));
+        HIP_CHECK(hipEventRecord(ev1, R->stream));
+    } else {
+        HIP_CHECK(hipEventRecord(ev0, R->stream));
+        HIP_CHECK(ytql_launch_scan_generic(dp, R->d_segs, R->d_segex, R->d_off, R->d_cnt,
+                                           chunk->row_count, R->d_th, R->d_slots,
+                                           R->d_err, R->stream));
+        HIP_CHECK(hipEventRecord(ev1, R->stream));
+    }
+    HIP_CHECK(hipStreamSynchronize(R->stream));
+    {
+        float ms = 0;
+        HIP_CHECK(hipEventElapsedTime(&ms, ev0, ev1));
+        if (stats) {
+            stats->kernel_scan_ms += ms;
+            stats->kernel_scan_launches += 1;
+        }
+    }
+    if (!fs->valid) {
+        unsigned kerr = 0;
+        HIP_CHECK(hipMemcpy(&kerr, R->d_err, sizeof(unsigned), hipMemcpyDeviceToHost));
+        if (kerr) {
+            set_err(errbuf, errlen, kerr == YT_ERR_DIV_ZERO ? "Division by zero" : "expression error");
+            rc = (int)kerr;
+            goto fail;
+        }
+    }
+    hipEventDestroy(ev0);
+    hipEventDestroy(ev1);
+    return YT_OK;
+fail:
+    if (ev0) hipEventDestroy(ev0);
+    if (ev1) hipEventDestroy(ev1);
+    return rc;
+}
+
+/* finalize a group (OutGroup record or side accumulator) into HVal row
+ * [keys..., aggs...] */
+static void finalize_row(const YtPlan* plan, uint8_t key_type,
+                         uint8_t sum_type[kMaxAggs],
+                         uint64_t key_bits, int key_null, uint64_t cnt,
+                         const uint64_t* agg_bits, const uint64_t* agg_nonnull,
+                         HVal* row, int* nrow, int has_key)
+{
+    int n = 0;
+    if (has_key) {
+        row[n].type = key_null ? YT_VT_NULL : key_type;
+        row[n].bits = key_null ? 0 : key_bits;
+        n++;
+    }
+    for (int a = 0; a < plan->agg_count; a++) {
+        if (plan->aggs[a]->func == YT_AGG_SUM1) {
+            row[n].type = YT_VT_INT64;
+            row[n].bits = cnt;
+        } else if (agg_nonnull[a] == 0) {
+            row[n].type = YT_VT_NULL;
+            row[n].bits = 0;
+        } else {
+            row[n].type = sum_type[a];
+            row[n].bits = agg_bits[a];
+        }
+        n++;
+    }
+    *nrow = n;
+}
+
+static int emit_rows(const YtPlan* plan, const YtChunk* chunk,
+                     const std::vector<OutGroup>& groups,
+                     const TableHdr& th, int has_any_row_global,
+                     const uint64_t* gaccum, int used_fast_global,
+                     YtRowset* output, char* errbuf, size_t errlen)
+{
+    uint8_t col_types[kMaxCols];
+    for (int c = 0; c < chunk->column_count && c < kMaxCols; c++)
+        col_types[c] = (uint8_t)chunk->columns[c].value_type;
+
+    uint8_t key_type = YT_VT_INT64;
+    int has_key = plan->key_count == 1;
+    if (has_key) key_type = expr_static_type(plan->keys[0], col_types);
+    uint8_t sum_type[kMaxAggs];
+    for (int a = 0; a < plan->agg_count; a++) {
+        sum_type[a] = (plan->aggs[a]->func == YT_AGG_SUM)
+            ? expr_static_type(plan->aggs[a]->arg, col_types) : YT_VT_INT64;
+        if (sum_type[a] == YT_VT_NULL) sum_type[a] = YT_VT_INT64;
+        if (sum_type[a] == YT_VT_BOOLEAN) sum_type[a] = YT_VT_BOOLEAN;
+    }
+
+    int base_cols = (has_key ? 1 : 0) + plan->agg_count;
+    int out_cols = plan->project_count ? plan->project_count : base_cols;
+
+    auto emit = [&](uint64_t key_bits, int key_null, uint64_t cnt,
+                    const uint64_t* ab, const uint64_t* an) -> int {
+        if (output->row_count >= output->capacity_rows) return YT_ERR_CAPACITY;
+        HVal row[2 + kMaxAggs];
+        int nrow = 0;
+        finalize_row(plan, key_type, sum_type, key_bits, key_null, cnt, ab, an,
+                     row, &nrow, has_key);
+        YtValue* dst = output->values + output->row_count * out_cols;
+        if (plan->project_count) {
+            for (int p = 0; p < plan->project_count; p++) {
+                HVal v;
+                int rc = heval(plan->projects[p], row, nrow, &v);
+                if (rc) return rc;
+                dst[p].id = (uint16_t)p;
+                dst[p].type = v.type;
+                dst[p].flags = 0;
+                dst[p].length = 0;
+                dst[p].data.bits = v.bits;
+            }
+        } else {
+            for (int i = 0; i < nrow; i++) {
+                dst[i].id = (uint16_t)i;
+                dst[i].type = row[i].type;
+                dst[i].flags = 0;
+                dst[i].length = 0;
+                dst[i].data.bits = row[i].bits;
+            }
+        }
+        output->row_count++;
+        return YT_OK;
+    };
+
+    output->row_count = 0;
+    output->column_count = out_cols;
+    int rc = YT_OK;
+
+    if (!has_key) {
+        /* global aggregate: one row iff any row passed the filter */
+        uint64_t cnt;
+        uint64_t ab[kMaxAggs] = {0}, an[kMaxAggs] = {0};
+        if (used_fast_global) {
+            cnt = gaccum[0];
+            for (int a = 0; a < plan->agg_count; a++) {
+                ab[a] = gaccum[1 + 2 * a];
+                an[a] = gaccum[2 + 2 * a];
+            }
+            if (cnt == 0) return YT_OK;
+        } else {
+            /* generic path routed rows to synthetic key bits=1 */
+            if (groups.empty()) return YT_OK;
+            cnt = groups[0].cnt;
+            for (int a = 0; a < plan->agg_count; a++) {
+                ab[a] = groups[0].agg_bits[a];
+                an[a] = groups[0].agg_nonnull[a];
+            }
+        }
+        (void)has_any_row_global;
+        return emit(0, 0, cnt, ab, an);
+    }
+
+    for (const OutGroup& g : groups) {
+        rc = emit(g.key_bits, (int)(g.key_meta & 1), g.cnt, g.agg_bits, g.agg_nonnull);
+        if (rc) return rc;
+    }
+    /* side groups: key bits == 0, then null key */
+    for (int side = 0; side < 2; side++) {
+        if (!th.side_used[side]) continue;
+        uint64_t ab[kMaxAggs], an[kMaxAggs];
+        for (int a = 0; a < plan->agg_count; a++) {
+            ab[a] = th.side_agg[side][2 * a];
+            an[a] = th.side_agg[side][2 * a + 1];
+        }
+        rc = emit(0, side == 1, th.side_cnt[side], ab, an);
+        if (rc) return rc;
+    }
+    (void)errbuf; (void)errlen;
+    return rc;
+}
+
+/* ------------------------------------------------------------------ */
+/* public entries                                                      */
+
+extern "C" int yt_gpu_query_execute(
+    const YtPlan* plan, const YtChunk* chunk, const YtExecOptions* options,
+    YtRowset* output, YtStatistics* stats, char* errbuf, size_t errlen)
+{
+    int rc = yt_gpu_available(errbuf, errlen);
+    if (rc != YT_OK) return rc;
+    if (!plan || !chunk || !output) { set_err(errbuf, errlen, "null argument"); return YT_ERR_INVALID_PLAN; }
+
+    YtExecOptions defopt;
+    memset(&defopt, 0, sizeof(defopt));
+    if (!options) options = &defopt;
+    if (stats) memset(stats, 0, sizeof(*stats));
+
+    if (plan->agg_count == 0) {
+        set_err(errbuf, errlen, "GPU path: scan-without-aggregation not built this round (oracle covers it)");
+        return YT_ERR_UNSUPPORTED;
+    }
+
+    DevPlan dp;
+    rc = build_devplan(plan, chunk, &dp, errbuf, errlen);
+    if (rc) return rc;
+
+    FastShape fs;
+    analyze_fast(plan, chunk, &fs);
+
+    DeviceRun R;
+    R.stream = (hipStream_t)(uintptr_t)options->stream;
+    if (options->device) {
+        hipError_t e = hipSetDevice(options->device);
+        if (e != hipSuccess) { set_err(errbuf, errlen, "bad device"); return YT_ERR_HIP; }
+    }
+
+    unsigned maxw = 0;
+    rc = setup_chunk(chunk, &R, &maxw, errbuf, errlen);
+    if (rc) return rc;
+
+    rc = setup_table(&R, plan->agg_count,
+                     options->max_groups_hint,
+                     options->group_row_limit, errbuf, errlen);
+    if (rc) return rc;
+
+    rc = run_scan(plan, chunk, options, &R, &dp, &fs, maxw, stats, errbuf, errlen);
+    if (rc) return rc;
+
+    /* compact + readback */
+    TableHdr th;
+    std::vector<OutGroup> groups;
+    {
+        HIP_CHECK(hipMemcpy(&th, R.d_th, sizeof(th), hipMemcpyDeviceToHost));
+        if (th.overflow == 1) { set_err(errbuf, errlen, "group table overflow — raise max_groups_hint"); rc = YT_ERR_CAPACITY; goto fail; }
+        int64_t ngroups = (int64_t)th.ngroups;
+        if (ngroups > 0) {
+            HIP_CHECK(hipMalloc(&R.d_groups, sizeof(OutGroup) * ngroups));
+            HIP_CHECK(hipMalloc(&R.d_counter, sizeof(unsigned long long)));
+            HIP_CHECK(hipMemsetAsync(R.d_counter, 0, sizeof(unsigned long long), R.stream));
+            HIP_CHECK(ytql_launch_compact(nullptr, R.d_th, R.d_slots, plan->agg_count,
+                                          R.d_groups, R.d_counter, R.nslots, R.stream));
+            groups.resize(ngroups);
+            HIP_CHECK(hipMemcpyAsync(groups.data(), R.d_groups, sizeof(OutGroup) * ngroups,
+                                     hipMemcpyDeviceToHost, R.stream));
+            HIP_CHECK(hipStreamSynchronize(R.stream));
+        }
+    }
+    {
+        std::vector<uint64_t> gaccum(1 + 2 * kMaxAggs, 0);
+        if (fs.valid && fs.key_col < 0) {
+            HIP_CHECK(hipMemcpy(gaccum.data(), R.d_gaccum,
+                                sizeof(uint64_t) * (1 + 2 * kMaxAggs), hipMemcpyDeviceToHost));
+        }
+        rc = emit_rows(plan, chunk, groups, th, 0, gaccum.data(),
+                       fs.valid && fs.key_col < 0, output, errbuf, errlen);
+        if (rc) return rc;
+    }
+    if (stats) {
+        stats->rows_read = chunk->row_count;
+        int64_t bytes = 0;
+        for (int c = 0; c < chunk->column_count; c++)
+            for (int s = 0; s < chunk->columns[c].segment_count; s++)
+                bytes += chunk->columns[c].segments[s].data_size;
+        stats->data_weight_read = bytes;
+        stats->rows_written = output->row_count;
+        stats->grouped_row_count = (int64_t)th.ngroups
+            + (plan->key_count ? (int64_t)(th.side_used[0] + th.side_used[1]) : 0);
+        stats->incomplete_output = (th.overflow == 2);
+    }
+    return YT_OK;
+fail:
+    return rc;
+}
+
+extern "C" int yt_gpu_query_partial(
+    const YtPlan* plan, const YtChunk* chunk, const YtExecOptions* options,
+    int32_t partition_count, void* states_device, int64_t capacity_rows,
+    int64_t* part_counts, YtStatistics* stats, char* errbuf, size_t errlen)
+{
+    int rc = yt_gpu_available(errbuf, errlen);
+    if (rc != YT_OK) return rc;
+    if (plan->key_count != 1) { set_err(errbuf, errlen, "partial: need exactly 1 key"); return YT_ERR_UNSUPPORTED; }
+    int sum_slot = -1;
+    for (int a = 0; a < plan->agg_count; a++) {
+        if (plan->aggs[a]->func == YT_AGG_SUM) {
+            if (sum_slot >= 0) { set_err(errbuf, errlen, "partial: one sum agg max this round"); return YT_ERR_UNSUPPORTED; }
+            sum_slot = a;
+        } else if (plan->aggs[a]->func != YT_AGG_SUM1) {
+            set_err(errbuf, errlen, "partial: sum/sum(1) only");
+            return YT_ERR_UNSUPPORTED;
+        }
+    }
+
+    YtExecOptions defopt;
+    memset(&defopt, 0, sizeof(defopt));
+    if (!options) options = &defopt;
+    if (stats) memset(stats, 0, sizeof(*stats));
+
+    DevPlan dp;
+    rc = build_devplan(plan, chunk, &dp, errbuf, errlen);
+    if (rc) return rc;
+    FastShape fs;
+    analyze_fast(plan, chunk, &fs);
+
+    DeviceRun R;
+    R.stream = (hipStream_t)(uintptr_t)options->stream;
+
+    unsigned maxw = 0;
+    rc = setup_chunk(chunk, &R, &maxw, errbuf, errlen);
+    if (rc) return rc;
+    rc = setup_table(&R, plan->agg_count, options->max_groups_hint,
+                     options->group_row_limit, errbuf, errlen);
+    if (rc) return rc;
+    rc = run_scan(plan, chunk, options, &R, &dp, &fs, maxw, stats, errbuf, errlen);
+    if (rc) return rc;
+
+    TableHdr th;
+    {
+        HIP_CHECK(hipMemcpy(&th, R.d_th, sizeof(th), hipMemcpyDeviceToHost));
+        if (th.overflow == 1) { set_err(errbuf, errlen, "group table overflow"); rc = YT_ERR_CAPACITY; goto fail; }
+        int64_t ngroups = (int64_t)th.ngroups;
+        int64_t total = ngroups + th.side_used[0] + th.side_used[1];
+        if (total > capacity_rows) { set_err(errbuf, errlen, "state buffer too small"); rc = YT_ERR_CAPACITY; goto fail; }
+
+        /* compact table into OutGroups (device) */
+        HIP_CHECK(hipMalloc(&R.d_groups, sizeof(OutGroup) * (total ? total : 1)));
+        HIP_CHECK(hipMalloc(&R.d_counter, sizeof(unsigned long long)));
+        HIP_CHECK(hipMemsetAsync(R.d_counter, 0, sizeof(unsigned long long), R.stream));
+        if (ngroups > 0) {
+            HIP_CHECK(ytql_launch_compact(nullptr, R.d_th, R.d_slots, plan->agg_count,
+                                          R.d_groups, R.d_counter, R.nslots, R.stream));
+        }
+        /* append side groups from host */
+        int64_t pos = ngroups;
+        for (int side = 0; side < 2; side++) {
+            if (!th.side_used[side]) continue;
+            OutGroup g;
+            memset(&g, 0, sizeof(g));
+            g.key_bits = 0;
+            g.key_meta = (side == 1);
+            g.cnt = th.side_cnt[side];
+            for (int a = 0; a < plan->agg_count; a++) {
+                g.agg_bits[a] = th.side_agg[side][2 * a];
+                g.agg_nonnull[a] = th.side_agg[side][2 * a + 1];
+            }
+            HIP_CHECK(hipMemcpyAsync(R.d_groups + pos, &g, sizeof(g),
+                                     hipMemcpyHostToDevice, R.stream));
+            pos++;
+        }
+
+        /* partition counts → host prefix → scatter */
+        unsigned long long* d_counts = nullptr;
+        HIP_CHECK(hipMalloc(&d_counts, sizeof(unsigned long long) * partition_count));
+        HIP_CHECK(hipMemsetAsync(d_counts, 0, sizeof(unsigned long long) * partition_count, R.stream));
+        HIP_CHECK(ytql_launch_part_count(R.d_groups, total, partition_count, sum_slot,
+                                         d_counts, R.stream));
+        std::vector<unsigned long long> counts(partition_count);
+        HIP_CHECK(hipMemcpyAsync(counts.data(), d_counts,
+                                 sizeof(unsigned long long) * partition_count,
+                                 hipMemcpyDeviceToHost, R.stream));
+        HIP_CHECK(hipStreamSynchronize(R.stream));
+        std::vector<unsigned long long> cursors(partition_count);
+        unsigned long long run = 0;
+        for (int p = 0; p < partition_count; p++) { cursors[p] = run; run += counts[p]; }
+        HIP_CHECK(hipMemcpyAsync(d_counts, cursors.data(),
+                                 sizeof(unsigned long long) * partition_count,
+                                 hipMemcpyHostToDevice, R.stream));
+        HIP_CHECK(ytql_launch_part_scatter(R.d_groups, total, partition_count, sum_slot,
+                                           d_counts, (YtStateRow*)states_device, R.stream));
+        HIP_CHECK(hipStreamSynchronize(R.stream));
+        hipFree(d_counts);
+        for (int p = 0; p < partition_count; p++) part_counts[p] = (int64_t)counts[p];
+        if (stats) {
+            stats->rows_read = chunk->row_count;
+            stats->grouped_row_count = total;
+            stats->incomplete_output = (th.overflow == 2);
+        }
+    }
+    return YT_OK;
+fail:
+    return rc;
+}
+
+extern "C" int yt_gpu_merge_states(
+    const YtPlan* plan, const void* states_device, int64_t state_row_count,
+    const YtExecOptions* options, YtRowset* output, YtStatistics* stats,
+    char* errbuf, size_t errlen)
+{
+    int rc = yt_gpu_available(errbuf, errlen);
+    if (rc != YT_OK) return rc;
+    if (plan->key_count != 1) { set_err(errbuf, errlen, "merge: need exactly 1 key"); return YT_ERR_UNSUPPORTED; }
+    int sum_slot = -1;
+    for (int a = 0; a < plan->agg_count; a++) {
+        if (plan->aggs[a]->func == YT_AGG_SUM) sum_slot = a;
+    }
+
+    YtExecOptions defopt;
+    memset(&defopt, 0, sizeof(defopt));
+    if (!options) options = &defopt;
+    if (stats) memset(stats, 0, sizeof(*stats));
+
+    DeviceRun R;
+    R.stream = (hipStream_t)(uintptr_t)options->stream;
+    rc = setup_table(&R, plan->agg_count, options->max_groups_hint, 0, errbuf, errlen);
+    if (rc) return rc;
+
+    HIP_CHECK(ytql_launch_merge_states((const YtStateRow*)states_device, state_row_count,
+                                       plan->agg_count, sum_slot, R.d_th, R.d_slots,
+                                       R.stream));
+    HIP_CHECK(hipStreamSynchronize(R.stream));
+    {
+        TableHdr th;
+        HIP_CHECK(hipMemcpy(&th, R.d_th, sizeof(th), hipMemcpyDeviceToHost));
+        if (th.overflow == 1) { set_err(errbuf, errlen, "merge table overflow"); rc = YT_ERR_CAPACITY; goto fail; }
+        int64_t ngroups = (int64_t)th.ngroups;
+        std::vector<OutGroup> groups;
+        if (ngroups > 0) {
+            HIP_CHECK(hipMalloc(&R.d_groups, sizeof(OutGroup) * ngroups));
+            HIP_CHECK(hipMalloc(&R.d_counter, sizeof(unsigned long long)));
+            HIP_CHECK(hipMemsetAsync(R.d_counter, 0, sizeof(unsigned long long), R.stream));
+            HIP_CHECK(ytql_launch_compact(nullptr, R.d_th, R.d_slots, plan->agg_count,
+                                          R.d_groups, R.d_counter, R.nslots, R.stream));
+            groups.resize(ngroups);
+            HIP_CHECK(hipMemcpyAsync(groups.data(), R.d_groups, sizeof(OutGroup) * ngroups,
+                                     hipMemcpyDeviceToHost, R.stream));
+            HIP_CHECK(hipStreamSynchronize(R.stream));
+        }
+        /* fabricate a single-int64-column chunk view for type inference:
+         * the key static type is carried by plan->keys[0] over col types.
+         * Merge callers use the same chunk schema as partial; we only need
+         * col_types for expr_static_type — rebuild from the key expr columns
+         * is not possible here, so default key/sum types to int64. */
+        YtColumn col;
+        memset(&col, 0, sizeof(col));
+        col.value_type = YT_VT_INT64;
+        col.segment_count = 0;
+        std::vector<YtColumn> cols(kMaxCols, col);
+        YtChunk fake;
+        fake.row_count = 0;
+        fake.column_count = kMaxCols;
+        fake.columns = cols.data();
+        rc = emit_rows(plan, &fake, groups, th, 0, nullptr, 0, output, errbuf, errlen);
+        if (rc) return rc;
+        if (stats) {
+            stats->rows_written = output->row_count;
+            stats->grouped_row_count = ngroups + th.side_used[0] + th.side_used[1];
+        }
+    }
+    return YT_OK;
+fail:
+    return rc;
+}

@@ -1,0 +1,877 @@
+/* kernels.hip — hand-written CDNA4 (gfx950) kernels for the YTsaurus
+ * dynamic-table query hot path: fused columnar decode → filter →
+ * hash-aggregate, plus state partition/merge for the 8-GPU exchange.
+ *
+ * The compute semantics restate (MI355X-native, not ported):
+ *   - bit-unpack:   reference bit_packed_unsigned_vector-inl.h:108-186
+ *   - int64 decode: integer_column_reader.cpp:19-127,391-449 (zigzag-space
+ *                   frame of reference: v = ZigZagDecode64(min + packed))
+ *   - filter:       cg_fragment_compiler.cpp:3460-3510 (here fused into the
+ *                   decode loop instead of a row-at-a-time branch)
+ *   - group insert: cg_routines/registry.cpp:1783-1834 (open addressing; our
+ *                   table is a device CAS-claimed linear-probe table; the
+ *                   reference's hash function is internal and not mirrored —
+ *                   results are order-free)
+ *   - aggregates:   udf/sum.c (null-propagating; int64 wraps mod 2^64, so
+ *                   unordered atomic u64 adds are bit-exact)
+ *   - merge/shuffle: engine/coordinator.cpp:420-505 + shuffling_reader.cpp:21-88
+ *
+ * Everything is HBM-bandwidth-bound integer work — no MFMA on this path.
+ * All tables are accessed with device-scope atomics (per-XCD L2s are not
+ * coherent; see MI355X microarch notes).
+ */
+#include <hip/hip_runtime.h>
+#include <stdint.h>
+#include "common.h"
+#include "../../include/ytql_gpu.h"
+
+namespace ytql {
+
+/* ------------------------------------------------------------------ */
+/* device helpers                                                      */
+
+__device__ __forceinline__ int64_t zz_dec(uint64_t n)
+{
+    return (int64_t)((n >> 1) ^ (~(n & 1) + 1));
+}
+
+__device__ __forceinline__ uint64_t mix64(uint64_t x)
+{
+    x += 0x9E3779B97F4A7C15ULL;
+    x = (x ^ (x >> 30)) * 0xBF58476D1CE4E5B9ULL;
+    x = (x ^ (x >> 27)) * 0x94D049BB133111EBULL;
+    return x ^ (x >> 31);
+}
+
+/* must match oracle yto_partition_hash */
+__device__ __forceinline__ uint64_t partition_hash(uint64_t key_bits, int key_null)
+{
+    return mix64(key_bits ^ (key_null ? 0xDEADBEEF12345678ULL : 0));
+}
+
+struct DVal {
+    uint64_t bits;
+    uint8_t type;    /* YT_VT_* */
+    uint8_t null_;
+};
+
+__device__ __forceinline__ uint64_t bp_get(const uint64_t* data, uint32_t width, uint64_t index)
+{
+    if (width == 0) return 0;
+    if (width == 64) return data[index];
+    uint64_t bit = index * width;
+    const uint64_t* word = data + (bit >> 6);
+    unsigned off = (unsigned)(bit & 63);
+    uint64_t w1 = *word >> off;
+    if (off + width > 64) {
+        uint64_t w2 = (word[1] & ((1ULL << ((off + width) & 63)) - 1)) << (64 - off);
+        w1 |= w2;
+    } else {
+        w1 &= (1ULL << width) - 1;
+    }
+    return w1;
+}
+
+__device__ __forceinline__ int bm_get(const uint8_t* bm, uint64_t index)
+{
+    return (bm[index >> 3] >> (index & 7)) & 1;
+}
+
+/* extract value `index` (absolute in segment) from an LDS window that holds
+ * words [w0, ...] of the packed vector */
+__device__ __forceinline__ uint64_t bp_get_win(const uint64_t* win, uint32_t width,
+                                               uint64_t index, uint64_t w0)
+{
+    if (width == 0) return 0;
+    uint64_t bit = index * width;
+    uint64_t wi = (bit >> 6) - w0;
+    unsigned off = (unsigned)(bit & 63);
+    if (width == 64) return win[wi];
+    uint64_t w1 = win[wi] >> off;
+    if (off + width > 64) {
+        w1 |= (win[wi + 1] & ((1ULL << ((off + width) & 63)) - 1)) << (64 - off);
+    } else {
+        w1 &= (1ULL << width) - 1;
+    }
+    return w1;
+}
+
+/* ------------------------------------------------------------------ */
+/* segment header parsing (one thread per segment)                     */
+
+__global__ void k_parse_segments(const DevSeg* segs, int nsegs, SegEx* out,
+                                 unsigned* max_width_out /* per column? single */)
+{
+    int i = blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= nsegs) return;
+    const DevSeg& s = segs[i];
+    SegEx e = {};
+    const uint64_t* b = s.blob;
+    if (s.type == YT_SEG_DOUBLE) {
+        /* [u64 count][doubles][null bitmap] */
+        e.off_doubles_bytes = 8;
+        e.off_bitmap_bytes = 8 + 8 * (int64_t)s.row_count;
+        e.w_values = 64;
+    } else {
+        uint64_t h0 = b[0];
+        uint64_t n0 = h0 & ((1ULL << 56) - 1);
+        uint32_t w0 = (uint32_t)(h0 >> 56);
+        int64_t words0 = 1 + (int64_t)((w0 * n0 + 63) >> 6);
+        switch (s.type) {
+        case YT_SEG_DIRECT_DENSE: {
+            e.w_values = w0;
+            e.off_values_words = 1;
+            e.off_bitmap_bytes = words0 * 8;
+            break;
+        }
+        case YT_SEG_DICTIONARY_DENSE: {
+            e.w_values = w0;                 /* dictionary vector */
+            e.dict_size = (uint32_t)n0;
+            e.off_values_words = 1;
+            uint64_t h1 = b[words0];
+            e.w_ids = (uint32_t)(h1 >> 56);
+            e.run_count = (uint32_t)(h1 & ((1ULL << 56) - 1));
+            e.off_ids_words = words0 + 1;
+            break;
+        }
+        case YT_SEG_DIRECT_RLE: {
+            e.w_values = w0;
+            e.run_count = (uint32_t)n0;
+            e.off_values_words = 1;
+            e.off_bitmap_bytes = words0 * 8;
+            int64_t bm_bytes = ((int64_t)n0 + 7) / 8;
+            bm_bytes = (bm_bytes + 7) & ~(int64_t)7;
+            int64_t starts_word = words0 + bm_bytes / 8;
+            uint64_t h2 = b[starts_word];
+            e.w_starts = (uint32_t)(h2 >> 56);
+            e.off_starts_words = starts_word + 1;
+            break;
+        }
+        case YT_SEG_DICTIONARY_RLE: {
+            e.w_values = w0;
+            e.dict_size = (uint32_t)n0;
+            e.off_values_words = 1;
+            uint64_t h1 = b[words0];
+            e.w_ids = (uint32_t)(h1 >> 56);
+            e.run_count = (uint32_t)(h1 & ((1ULL << 56) - 1));
+            int64_t words1 = 1 + (int64_t)(((uint64_t)e.w_ids * e.run_count + 63) >> 6);
+            e.off_ids_words = words0 + 1;
+            int64_t starts_word = words0 + words1;
+            uint64_t h2 = b[starts_word];
+            e.w_starts = (uint32_t)(h2 >> 56);
+            e.off_starts_words = starts_word + 1;
+            break;
+        }
+        }
+    }
+    out[i] = e;
+    atomicMax(max_width_out, e.w_values);
+}
+
+/* ------------------------------------------------------------------ */
+/* generic row-at-a-time value fetch (correctness path; any segment)   */
+
+__device__ DVal seg_value_at(const DevSeg& s, const SegEx& e, int64_t row_in_seg,
+                             uint8_t vt)
+{
+    DVal v;
+    v.type = vt;
+    v.null_ = 0;
+    const uint64_t* b = s.blob;
+    switch (s.type) {
+    case YT_SEG_DOUBLE: {
+        const uint8_t* bm = (const uint8_t*)b + e.off_bitmap_bytes;
+        if (bm_get(bm, row_in_seg)) { v.null_ = 1; v.bits = 0; return v; }
+        v.bits = ((const uint64_t*)((const uint8_t*)b + e.off_doubles_bytes))[row_in_seg];
+        return v;
+    }
+    case YT_SEG_DIRECT_DENSE: {
+        const uint8_t* bm = (const uint8_t*)b + e.off_bitmap_bytes;
+        if (bm_get(bm, row_in_seg)) { v.null_ = 1; v.bits = 0; return v; }
+        uint64_t data = s.min_value + bp_get(b + e.off_values_words, e.w_values, row_in_seg);
+        v.bits = s.is_signed ? (uint64_t)zz_dec(data) : data;
+        return v;
+    }
+    case YT_SEG_DICTIONARY_DENSE: {
+        uint64_t id = bp_get(b + e.off_ids_words, e.w_ids, row_in_seg);
+        if (id == 0) { v.null_ = 1; v.bits = 0; return v; }
+        uint64_t data = s.min_value + bp_get(b + e.off_values_words, e.w_values, id - 1);
+        v.bits = s.is_signed ? (uint64_t)zz_dec(data) : data;
+        return v;
+    }
+    case YT_SEG_DIRECT_RLE:
+    case YT_SEG_DICTIONARY_RLE: {
+        /* binary search the run whose start <= row; starts are sorted
+         * (run starts, integer_column_writer.cpp:403-419) */
+        const uint64_t* starts = b + e.off_starts_words;
+        uint32_t lo = 0, hi = e.run_count;   /* first run with start > row, minus 1 */
+        while (lo + 1 < hi) {
+            uint32_t mid = (lo + hi) / 2;
+            if (bp_get(starts, e.w_starts, mid) <= (uint64_t)row_in_seg) lo = mid;
+            else hi = mid;
+        }
+        uint32_t run = lo;
+        if (s.type == YT_SEG_DIRECT_RLE) {
+            const uint8_t* bm = (const uint8_t*)b + e.off_bitmap_bytes;
+            if (bm_get(bm, run)) { v.null_ = 1; v.bits = 0; return v; }
+            uint64_t data = s.min_value + bp_get(b + e.off_values_words, e.w_values, run);
+            v.bits = s.is_signed ? (uint64_t)zz_dec(data) : data;
+        } else {
+            uint64_t id = bp_get(b + e.off_ids_words, e.w_ids, run);
+            if (id == 0) { v.null_ = 1; v.bits = 0; return v; }
+            uint64_t data = s.min_value + bp_get(b + e.off_values_words, e.w_values, id - 1);
+            v.bits = s.is_signed ? (uint64_t)zz_dec(data) : data;
+        }
+        return v;
+    }
+    }
+    v.null_ = 1;
+    v.bits = 0;
+    return v;
+}
+
+/* ------------------------------------------------------------------ */
+/* postfix expression interpreter — same semantics as the oracle        */
+/* (cg_fragment_compiler.cpp: arithmetic null-propagating :1440-1540,  */
+/*  relational non-canonical :1601-1720, logical Kleene :1547-1599)    */
+
+struct ColCtx {
+    const DevSeg* segs;        /* all columns' segments, concatenated */
+    const SegEx* segex;
+    const int32_t* col_seg_off;   /* per column: first segment index */
+    const int32_t* col_seg_cnt;
+    int64_t row;                  /* chunk row */
+    uint32_t error;               /* YT_ERR_* */
+};
+
+__device__ DVal col_value(const DevPlan& p, ColCtx& c, int col)
+{
+    int off = c.col_seg_off[col];
+    int cnt = c.col_seg_cnt[col];
+    /* binary search segment by start_row */
+    int lo = 0, hi = cnt;
+    while (lo + 1 < hi) {
+        int mid = (lo + hi) / 2;
+        if (c.segs[off + mid].start_row <= c.row) lo = mid;
+        else hi = mid;
+    }
+    const DevSeg& s = c.segs[off + lo];
+    return seg_value_at(s, c.segex[off + lo], c.row - s.start_row, p.col_types[col]);
+}
+
+__device__ DVal eval_prog(const DevPlan& p, ColCtx& c, int off, int len)
+{
+    DVal stk[12];
+    int sp = 0;
+    for (int i = 0; i < len; i++) {
+        const PInst& in = p.prog[off + i];
+        switch (in.op) {
+        case P_COL:
+            stk[sp++] = col_value(p, c, in.col);
+            break;
+        case P_LIT_I64: {
+            DVal v; v.bits = in.bits; v.type = YT_VT_INT64; v.null_ = 0;
+            stk[sp++] = v;
+            break;
+        }
+        case P_LIT_DOUBLE: {
+            DVal v; v.bits = in.bits; v.type = YT_VT_DOUBLE; v.null_ = 0;
+            stk[sp++] = v;
+            break;
+        }
+        case P_LIT_NULL: {
+            DVal v; v.bits = 0; v.type = YT_VT_NULL; v.null_ = 1;
+            stk[sp++] = v;
+            break;
+        }
+        case P_NOT: {
+            DVal a = stk[--sp];
+            if (!a.null_) { a.bits = !a.bits; a.type = YT_VT_BOOLEAN; }
+            stk[sp++] = a;
+            break;
+        }
+        default: {
+            DVal b = stk[--sp];
+            DVal a = stk[--sp];
+            DVal r; r.bits = 0; r.type = YT_VT_NULL; r.null_ = 1;
+            if (in.op >= P_ADD && in.op <= P_MOD) {
+                if (!a.null_ && !b.null_) {
+                    if (a.type == YT_VT_DOUBLE) {
+                        double x = __longlong_as_double(a.bits);
+                        double y = __longlong_as_double(b.bits);
+                        double z = 0;
+                        switch (in.op) {
+                        case P_ADD: z = x + y; break;
+                        case P_SUB: z = x - y; break;
+                        case P_MUL: z = x * y; break;
+                        case P_DIV: z = x / y; break;
+                        default: c.error = YT_ERR_UNSUPPORTED; break;
+                        }
+                        r.bits = __double_as_longlong(z);
+                        r.type = YT_VT_DOUBLE; r.null_ = 0;
+                    } else {
+                        uint64_t x = a.bits, y = b.bits, z = 0;
+                        int sgn = (a.type == YT_VT_INT64);
+                        switch (in.op) {
+                        case P_ADD: z = x + y; break;
+                        case P_SUB: z = x - y; break;
+                        case P_MUL: z = x * y; break;
+                        case P_DIV:
+                        case P_MOD:
+                            if (y == 0) { c.error = YT_ERR_DIV_ZERO; }
+                            else if (sgn) {
+                                int64_t sx = (int64_t)x, sy = (int64_t)y;
+                                if (sx == INT64_MIN && sy == -1) {
+                                    z = (in.op == P_DIV) ? (uint64_t)INT64_MIN : 0;
+                                } else {
+                                    z = (uint64_t)((in.op == P_DIV) ? sx / sy : sx % sy);
+                                }
+                            } else {
+                                z = (in.op == P_DIV) ? x / y : x % y;
+                            }
+                            break;
+                        }
+                        r.bits = z; r.type = a.type; r.null_ = 0;
+                    }
+                }
+            } else if (in.op >= P_EQ && in.op <= P_GE) {
+                int lt, eq;
+                int force_true = 0;
+                if (a.null_ || b.null_) {
+                    unsigned ln = a.null_, rn = b.null_;
+                    lt = rn < ln;
+                    eq = ln == rn;
+                } else if (a.type == YT_VT_DOUBLE) {
+                    double x = __longlong_as_double(a.bits);
+                    double y = __longlong_as_double(b.bits);
+                    int unordered = (x != x) || (y != y);
+                    lt = unordered || (x < y);
+                    eq = unordered || (x == y);
+                    force_true = unordered;   /* FCmpU*: NaN → true */
+                } else if (a.type == YT_VT_INT64) {
+                    int64_t x = (int64_t)a.bits, y = (int64_t)b.bits;
+                    lt = x < y; eq = x == y;
+                } else {
+                    lt = a.bits < b.bits; eq = a.bits == b.bits;
+                }
+                int res = 1;
+                if (!force_true) {
+                    switch (in.op) {
+                    case P_EQ: res = eq; break;
+                    case P_NE: res = !eq; break;
+                    case P_LT: res = lt; break;
+                    case P_LE: res = lt || eq; break;
+                    case P_GT: res = !(lt || eq); break;
+                    case P_GE: res = !lt; break;
+                    }
+                }
+                r.bits = (uint64_t)res; r.type = YT_VT_BOOLEAN; r.null_ = 0;
+            } else if (in.op == P_AND || in.op == P_OR) {
+                int an = a.null_, bn = b.null_;
+                int av = an ? 0 : (a.bits != 0), bv = bn ? 0 : (b.bits != 0);
+                if (in.op == P_AND) {
+                    if ((!an && !av) || (!bn && !bv)) { r.bits = 0; r.type = YT_VT_BOOLEAN; r.null_ = 0; }
+                    else if (!an && !bn) { r.bits = 1; r.type = YT_VT_BOOLEAN; r.null_ = 0; }
+                } else {
+                    if ((!an && av) || (!bn && bv)) { r.bits = 1; r.type = YT_VT_BOOLEAN; r.null_ = 0; }
+                    else if (!an && !bn) { r.bits = 0; r.type = YT_VT_BOOLEAN; r.null_ = 0; }
+                }
+            } else {
+                c.error = YT_ERR_UNSUPPORTED;
+            }
+            stk[sp++] = r;
+            break;
+        }
+        }
+    }
+    return stk[0];
+}
+
+/* ------------------------------------------------------------------ */
+/* group table update                                                  */
+
+__device__ __forceinline__ void agg_update_slot(unsigned long long* aggp,
+                                                const DevPlan& p, int a,
+                                                DVal v)
+{
+    /* aggp -> { bits, nonnull } for agg a; sum semantics udf/sum.c:12-45 */
+    if (v.null_) return;
+    if (v.type == YT_VT_DOUBLE) {
+        atomicAdd((double*)aggp, __longlong_as_double(v.bits));
+    } else {
+        atomicAdd(aggp, (unsigned long long)v.bits);
+    }
+    atomicAdd(aggp + 1, 1ULL);
+}
+
+/* returns pointer to slot (stride u64s) for key, claiming if new; nullptr on
+ * overflow. side groups handled by caller. */
+__device__ unsigned long long* table_probe(TableHdr* th, unsigned long long* slots,
+                                           int stride, uint64_t key_bits)
+{
+    uint64_t h = mix64(key_bits);
+    uint64_t mask = th->mask;
+    uint64_t s = h & mask;
+    for (uint64_t iter = 0; iter <= mask; iter++) {
+        unsigned long long* slot = slots + s * stride;
+        unsigned long long old = atomicCAS(slot, 0ULL, (unsigned long long)key_bits);
+        if (old == 0ULL) {
+            unsigned long long ticket = atomicAdd(&th->ngroups, 1ULL);
+            if (th->group_limit > 0 && (int64_t)ticket >= th->group_limit) {
+                th->overflow = 2;   /* group row limit — incomplete output */
+            }
+            return slot;
+        }
+        if (old == (unsigned long long)key_bits) return slot;
+        s = (s + 1) & mask;
+    }
+    th->overflow = 1;
+    return nullptr;
+}
+
+__device__ void table_update_generic(TableHdr* th, unsigned long long* slots,
+                                     const DevPlan& p, DVal key, DVal* aggv)
+{
+    int stride = 2 + 2 * p.agg_count;
+    unsigned long long* aggp;
+    unsigned long long* cntp;
+    if (key.null_ || key.bits == 0) {
+        int side = key.null_ ? 1 : 0;
+        th->side_used[side] = 1;
+        cntp = (unsigned long long*)&th->side_cnt[side];
+        aggp = (unsigned long long*)&th->side_agg[side][0];
+    } else {
+        unsigned long long* slot = table_probe(th, slots, stride, key.bits);
+        if (!slot) return;
+        cntp = slot + 1;
+        aggp = slot + 2;
+    }
+    atomicAdd(cntp, 1ULL);
+    for (int a = 0; a < p.agg_count; a++) {
+        if (p.agg_func[a] == YT_AGG_SUM) {
+            agg_update_slot(aggp + 2 * a, p, a, aggv[a]);
+        }
+        /* YT_AGG_SUM1: cnt covers it */
+    }
+}
+
+/* ------------------------------------------------------------------ */
+/* generic fused scan (any segment types / expressions)                */
+
+__global__ void __launch_bounds__(256)
+k_scan_generic(DevPlan p, const DevSeg* segs, const SegEx* segex,
+               const int32_t* col_seg_off, const int32_t* col_seg_cnt,
+               int64_t row_count,
+               TableHdr* th, unsigned long long* slots,
+               unsigned* error_out)
+{
+    ColCtx c;
+    c.segs = segs;
+    c.segex = segex;
+    c.col_seg_off = col_seg_off;
+    c.col_seg_cnt = col_seg_cnt;
+    c.error = 0;
+
+    DVal aggv[kMaxAggs];
+
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         r < row_count; r += stride) {
+        c.row = r;
+        if (p.filter_len) {
+            DVal f = eval_prog(p, c, p.filter_off, p.filter_len);
+            if (f.null_ || f.bits == 0) continue;
+        }
+        DVal key;
+        if (p.key_len) {
+            key = eval_prog(p, c, p.key_off, p.key_len);
+        } else {
+            key.bits = 1; key.type = YT_VT_INT64; key.null_ = 0;  /* single group, side-stepped below */
+        }
+        for (int a = 0; a < p.agg_count; a++) {
+            if (p.agg_func[a] == YT_AGG_SUM) {
+                aggv[a] = eval_prog(p, c, p.agg_off[a], p.agg_len[a]);
+            }
+        }
+        if (c.error) { atomicMax(error_out, c.error); return; }
+        table_update_generic(th, slots, p, key, aggv);
+    }
+    if (c.error) atomicMax(error_out, c.error);
+}
+
+/* ------------------------------------------------------------------ */
+/* fast fused kernel — DirectDense segments, direct-column shapes      */
+/*                                                                     */
+/* One workgroup (256 threads) per tile of `tile_rows` rows inside one */
+/* segment. Packed words of every used column + null bitmaps staged in */
+/* LDS via coalesced loads; per-thread strided extraction; aggregation */
+/* either into registers (global aggregate: wave shfl-reduce → LDS →   */
+/* one atomic per WG) or the shared table (group by).                  */
+
+/* global accumulators for the no-key path:
+ * [0] = row count (pass filter), [1 + 2a], [2 + 2a] = sum bits / nonnull */
+__global__ void __launch_bounds__(256)
+k_scan_fast(FastParams fp, const DevSeg* segs, const SegEx* segex,
+            const FastCol* cols,
+            TableHdr* th, unsigned long long* slots,
+            unsigned long long* gaccum)
+{
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+
+    const int tid = threadIdx.x;
+
+    for (int tile = blockIdx.x; tile < fp.ntiles; tile += gridDim.x) {
+        const int seg_idx = tile / fp.tiles_per_seg;
+        const int tile_in_seg = tile % fp.tiles_per_seg;
+        const int64_t t0 = (int64_t)tile_in_seg * fp.tile_rows;
+
+        /* ---- stage all used columns for this tile into LDS ---- */
+        int64_t lds_off = 0;
+        uint64_t* col_words[6];
+        const uint8_t* col_bitmap[6];
+        uint32_t col_w[6];
+        int64_t col_w0[6];    /* first staged word index (absolute in segment values) */
+        int32_t seg_rows = 0;
+
+        for (int u = 0; u < fp.nused; u++) {
+            const DevSeg& s = segs[cols[u].seg_off + seg_idx];
+            const SegEx& e = segex[cols[u].seg_off + seg_idx];
+            seg_rows = s.row_count;
+            int64_t t1 = t0 + fp.tile_rows;
+            if (t1 > seg_rows) t1 = seg_rows;
+            uint32_t w = e.w_values;
+            col_w[u] = w;
+            int64_t W0 = ((uint64_t)t0 * w) >> 6;
+            int64_t W1 = w ? ((((uint64_t)t1 * w) + 63) >> 6) : 0;
+            if (W1 > W0 + 1 || w == 0) { /* at least cover straddle */ }
+            int64_t nwords = (w == 0) ? 0 : (W1 - W0 + 1);
+            col_w0[u] = W0;
+            uint64_t* dst = (uint64_t*)(smem + lds_off);
+            col_words[u] = dst;
+            const uint64_t* src = s.blob + e.off_values_words + W0;
+            /* clamp to the vector's actual words to avoid OOB on last word */
+            int64_t vec_words = (w == 0) ? 0 : (((uint64_t)seg_rows * w + 63) >> 6);
+            int64_t avail = vec_words - W0;
+            if (nwords > avail) nwords = avail;
+            for (int64_t i = tid; i < nwords; i += 256) {
+                dst[i] = src[i];
+            }
+            lds_off += ((nwords * 8) + 15) & ~(int64_t)15;
+
+            /* stage null bitmap slice: bytes for rows [t0, t1) */
+            uint8_t* bdst = (uint8_t*)(smem + lds_off);
+            col_bitmap[u] = bdst;
+            const uint8_t* bsrc = (const uint8_t*)s.blob + e.off_bitmap_bytes + (t0 >> 3);
+            int64_t bbytes = ((t1 - t0) + 7) / 8;
+            for (int64_t i = tid; i < bbytes; i += 256) {
+                bdst[i] = bsrc[i];
+            }
+            lds_off += (bbytes + 15) & ~(int64_t)15;
+        }
+        __syncthreads();
+
+        const DevSeg& s0 = segs[cols[0].seg_off + seg_idx];
+        int64_t t1 = t0 + fp.tile_rows;
+        if (t1 > seg_rows) t1 = seg_rows;
+        const int64_t chunk_row0 = s0.start_row;
+
+        /* per-thread accumulators for the global-agg path */
+        uint64_t acc_sum[kMaxAggs] = {0, 0, 0, 0};
+        uint64_t acc_nn[kMaxAggs] = {0, 0, 0, 0};
+        uint64_t acc_cnt = 0;
+
+        const int R = (fp.tile_rows + 255) / 256;
+        for (int i = 0; i < R; i++) {
+            int64_t j = t0 + (int64_t)i * 256 + tid;   /* row in segment */
+            if (j >= t1) break;
+
+            /* filter */
+            if (fp.filter_idx >= 0) {
+                int u = fp.filter_idx;
+                if (bm_get(col_bitmap[u], j - t0)) continue;   /* null → range false */
+                uint64_t packed = bp_get_win(col_words[u], col_w[u], j, col_w0[u]);
+                const DevSeg& sf = segs[cols[u].seg_off + seg_idx];
+                int64_t v = zz_dec(sf.min_value + packed);
+                if (v < fp.filter_lo || v > fp.filter_hi) continue;
+            }
+
+            /* key */
+            DVal key;
+            if (fp.key_idx >= 0) {
+                int u = fp.key_idx;
+                if (bm_get(col_bitmap[u], j - t0)) {
+                    key.null_ = 1; key.bits = 0; key.type = YT_VT_INT64;
+                } else {
+                    uint64_t packed = bp_get_win(col_words[u], col_w[u], j, col_w0[u]);
+                    const DevSeg& sk = segs[cols[u].seg_off + seg_idx];
+                    key.bits = (uint64_t)zz_dec(sk.min_value + packed);
+                    key.null_ = 0; key.type = YT_VT_INT64;
+                }
+            }
+
+            if (fp.key_idx < 0) {
+                acc_cnt++;
+                for (int a = 0; a < fp.nsum; a++) {
+                    int u = fp.sum_idx[a];
+                    if (bm_get(col_bitmap[u], j - t0)) continue;
+                    uint64_t packed = bp_get_win(col_words[u], col_w[u], j, col_w0[u]);
+                    const DevSeg& sv = segs[cols[u].seg_off + seg_idx];
+                    acc_sum[a] += (uint64_t)zz_dec(sv.min_value + packed);
+                    acc_nn[a]++;
+                }
+            } else {
+                /* group path: direct table update */
+                unsigned long long* cntp;
+                unsigned long long* aggp;
+                int stride = 2 + 2 * fp.agg_count;
+                if (key.null_ || key.bits == 0) {
+                    int side = key.null_ ? 1 : 0;
+                    th->side_used[side] = 1;
+                    cntp = (unsigned long long*)&th->side_cnt[side];
+                    aggp = (unsigned long long*)&th->side_agg[side][0];
+                } else {
+                    unsigned long long* slot = table_probe(th, slots, stride, key.bits);
+                    if (!slot) continue;
+                    cntp = slot + 1;
+                    aggp = slot + 2;
+                }
+                atomicAdd(cntp, 1ULL);
+                for (int a = 0; a < fp.nsum; a++) {
+                    int u = fp.sum_idx[a];
+                    if (bm_get(col_bitmap[u], j - t0)) continue;
+                    uint64_t packed = bp_get_win(col_words[u], col_w[u], j, col_w0[u]);
+                    const DevSeg& sv = segs[cols[u].seg_off + seg_idx];
+                    uint64_t v = (uint64_t)zz_dec(sv.min_value + packed);
+                    unsigned long long* ap = aggp + 2 * fp.sum_slot[a];
+                    atomicAdd(ap, (unsigned long long)v);
+                    atomicAdd(ap + 1, 1ULL);
+                }
+            }
+        }
+
+        /* global-agg reduction: wave shfl → LDS → one atomic per WG */
+        if (fp.key_idx < 0) {
+            __syncthreads();   /* LDS reuse */
+            uint64_t* red = (uint64_t*)smem;
+            const int lane = tid & 63;
+            const int wave = tid >> 6;
+            for (int a = 0; a < 2 * fp.nsum + 1; a++) {
+                uint64_t v = (a == 0) ? acc_cnt
+                           : (a & 1) ? acc_sum[a >> 1]
+                                     : acc_nn[(a >> 1) - 1];
+                for (int sh = 32; sh >= 1; sh >>= 1) {
+                    v += (uint64_t)__shfl_down((long long)v, sh, 64);
+                }
+                if (lane == 0) red[wave * 16 + a] = v;
+            }
+            __syncthreads();
+            if (tid == 0) {
+                for (int a = 0; a < 2 * fp.nsum + 1; a++) {
+                    uint64_t v = red[a] + red[16 + a] + red[32 + a] + red[48 + a];
+                    if (a == 0) {
+                        atomicAdd(&gaccum[0], (unsigned long long)v);
+                    } else if (a & 1) {
+                        int slot = fp.sum_slot[a >> 1];
+                        atomicAdd(&gaccum[1 + 2 * slot], (unsigned long long)v);
+                    } else {
+                        int slot = fp.sum_slot[(a >> 1) - 1];
+                        atomicAdd(&gaccum[2 + 2 * slot], (unsigned long long)v);
+                    }
+                }
+            }
+            __syncthreads();
+        } else {
+            __syncthreads();   /* before next tile overwrites LDS */
+        }
+        (void)chunk_row0;
+    }
+}
+
+/* ------------------------------------------------------------------ */
+/* table compaction / partition / merge                                */
+
+__global__ void k_compact(TableHdr* th, const unsigned long long* slots,
+                          int agg_count, OutGroup* out,
+                          unsigned long long* counter)
+{
+    int stride = 2 + 2 * agg_count;
+    uint64_t nslots = th->nslots;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         i < nslots; i += (uint64_t)gridDim.x * blockDim.x) {
+        const unsigned long long* slot = slots + i * stride;
+        uint64_t key = slot[0];
+        if (key == 0) continue;
+        unsigned long long idx = atomicAdd(counter, 1ULL);
+        OutGroup& g = out[idx];
+        g.key_bits = key;
+        g.key_meta = 0;
+        g.cnt = slot[1];
+        for (int a = 0; a < agg_count; a++) {
+            g.agg_bits[a] = slot[2 + 2 * a];
+            g.agg_nonnull[a] = slot[3 + 2 * a];
+        }
+    }
+}
+
+/* partition counting + scatter of compacted groups into YtStateRow buckets.
+ * Mirrors the reference's in-process shuffle hash partition
+ * (shuffling_reader.cpp:40-42: destination = hash(key) % destinationCount). */
+__global__ void k_part_count(const OutGroup* groups, int64_t n, int nparts,
+                             int sum_slot, unsigned long long* counts)
+{
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         i < n; i += (int64_t)gridDim.x * blockDim.x) {
+        uint64_t h = partition_hash(groups[i].key_bits, (int)(groups[i].key_meta & 1));
+        atomicAdd(&counts[h % (uint64_t)nparts], 1ULL);
+    }
+}
+
+__global__ void k_part_scatter(const OutGroup* groups, int64_t n, int nparts,
+                               int sum_slot, unsigned long long* cursors,
+                               YtStateRow* out)
+{
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         i < n; i += (int64_t)gridDim.x * blockDim.x) {
+        const OutGroup& g = groups[i];
+        int knull = (int)(g.key_meta & 1);
+        uint64_t h = partition_hash(g.key_bits, knull);
+        unsigned long long pos = atomicAdd(&cursors[h % (uint64_t)nparts], 1ULL);
+        YtStateRow& sr = out[pos];
+        sr.key_bits = g.key_bits;
+        uint64_t nonnull = (sum_slot >= 0) ? (g.agg_nonnull[sum_slot] ? 1 : 0) : 0;
+        sr.meta = (uint64_t)knull | (nonnull << 8);
+        sr.sum_bits = (sum_slot >= 0) ? g.agg_bits[sum_slot] : 0;
+        sr.row_count = g.cnt;
+    }
+}
+
+/* merge state rows into a (fresh) table — front-query Merge semantics
+ * (cg_fragment_compiler.cpp:4116-4134 + udf/sum.c sum_merge). */
+__global__ void k_merge_states(const YtStateRow* states, int64_t n,
+                               int agg_count, int sum_slot,
+                               TableHdr* th, unsigned long long* slots)
+{
+    int stride = 2 + 2 * agg_count;
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         i < n; i += (int64_t)gridDim.x * blockDim.x) {
+        const YtStateRow& sr = states[i];
+        int knull = (int)(sr.meta & 1);
+        unsigned long long* cntp;
+        unsigned long long* aggp;
+        if (knull || sr.key_bits == 0) {
+            int side = knull ? 1 : 0;
+            th->side_used[side] = 1;
+            cntp = (unsigned long long*)&th->side_cnt[side];
+            aggp = (unsigned long long*)&th->side_agg[side][0];
+        } else {
+            unsigned long long* slot = table_probe(th, slots, stride, sr.key_bits);
+            if (!slot) continue;
+            cntp = slot + 1;
+            aggp = slot + 2;
+        }
+        atomicAdd(cntp, (unsigned long long)sr.row_count);
+        if (sum_slot >= 0 && (sr.meta >> 8)) {
+            atomicAdd(aggp + 2 * sum_slot, (unsigned long long)sr.sum_bits);
+            atomicAdd(aggp + 2 * sum_slot + 1, 1ULL);
+        }
+    }
+}
+
+} /* namespace ytql */
+
+/* ------------------------------------------------------------------ */
+/* extern-C launch wrappers (called from evaluator.cpp)                */
+
+using namespace ytql;
+
+extern "C" {
+
+hipError_t ytql_launch_parse_segments(const DevSeg* segs, int nsegs, SegEx* out,
+                                      unsigned* max_width, hipStream_t st)
+{
+    int block = 256;
+    int grid = (nsegs + block - 1) / block;
+    hipLaunchKernelGGL(k_parse_segments, dim3(grid), dim3(block), 0, st,
+                       segs, nsegs, out, max_width);
+    return hipGetLastError();
+}
+
+hipError_t ytql_launch_scan_generic(const DevPlan* p, const DevSeg* segs,
+                                    const SegEx* segex,
+                                    const int32_t* col_seg_off,
+                                    const int32_t* col_seg_cnt,
+                                    int64_t row_count,
+                                    TableHdr* th, unsigned long long* slots,
+                                    unsigned* error_out, hipStream_t st)
+{
+    int block = 256;
+    int64_t want = (row_count + block - 1) / block;
+    int grid = (int)(want > 4096 ? 4096 : (want > 0 ? want : 1));
+    hipLaunchKernelGGL(k_scan_generic, dim3(grid), dim3(block), 0, st,
+                       *p, segs, segex, col_seg_off, col_seg_cnt, row_count,
+                       th, slots, error_out);
+    return hipGetLastError();
+}
+
+hipError_t ytql_launch_scan_fast(const FastParams* fp, const DevSeg* segs,
+                                 const SegEx* segex, const FastCol* cols,
+                                 TableHdr* th, unsigned long long* slots,
+                                 unsigned long long* gaccum,
+                                 size_t lds_bytes, int grid, hipStream_t st)
+{
+    hipLaunchKernelGGL(k_scan_fast, dim3(grid), dim3(256), lds_bytes, st,
+                       *fp, segs, segex, cols, th, slots, gaccum);
+    return hipGetLastError();
+}
+
+hipError_t ytql_launch_compact(TableHdr* th_host_nslots, TableHdr* th,
+                               const unsigned long long* slots, int agg_count,
+                               OutGroup* out, unsigned long long* counter,
+                               uint64_t nslots, hipStream_t st)
+{
+    int block = 256;
+    uint64_t want = (nslots + block - 1) / block;
+    int grid = (int)(want > 2048 ? 2048 : (want ? want : 1));
+    hipLaunchKernelGGL(k_compact, dim3(grid), dim3(block), 0, st,
+                       th, slots, agg_count, out, counter);
+    return hipGetLastError();
+}
+
+hipError_t ytql_launch_part_count(const OutGroup* groups, int64_t n, int nparts,
+                                  int sum_slot, unsigned long long* counts,
+                                  hipStream_t st)
+{
+    int block = 256;
+    int64_t want = (n + block - 1) / block;
+    int grid = (int)(want > 2048 ? 2048 : (want ? want : 1));
+    hipLaunchKernelGGL(k_part_count, dim3(grid), dim3(block), 0, st,
+                       groups, n, nparts, sum_slot, counts);
+    return hipGetLastError();
+}
+
+hipError_t ytql_launch_part_scatter(const OutGroup* groups, int64_t n, int nparts,
+                                    int sum_slot, unsigned long long* cursors,
+                                    YtStateRow* out, hipStream_t st)
+{
+    int block = 256;
+    int64_t want = (n + block - 1) / block;
+    int grid = (int)(want > 2048 ? 2048 : (want ? want : 1));
+    hipLaunchKernelGGL(k_part_scatter, dim3(grid), dim3(block), 0, st,
+                       groups, n, nparts, sum_slot, cursors, out);
+    return hipGetLastError();
+}
+
+hipError_t ytql_launch_merge_states(const YtStateRow* states, int64_t n,
+                                    int agg_count, int sum_slot,
+                                    TableHdr* th, unsigned long long* slots,
+                                    hipStream_t st)
+{
+    int block = 256;
+    int64_t want = (n + block - 1) / block;
+    int grid = (int)(want > 2048 ? 2048 : (want ? want : 1));
+    hipLaunchKernelGGL(k_merge_states, dim3(grid), dim3(block), 0, st,
+                       states, n, agg_count, sum_slot, th, slots);
+    return hipGetLastError();
+}
+
+} /* extern C */
