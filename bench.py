@@ -1,0 +1,337 @@
+#!/usr/bin/env python3
+"""bench.py — headline benchmark for the MI355X-native YTsaurus query-path
+executor.
+
+Workload (BASELINE.json configs[2], the configuration the headline metric
+"rows/sec scan+GROUP BY on 1B-row int64 chunks" is quoted on):
+GROUP BY int64 key (1M distinct) with sum(v), sum(1) over N_ROWS rows of
+seeded synthetic data, stored in the reference's unversioned columnar chunk
+format (DirectDense int64 segments), resident in HBM when timing starts.
+
+A step = one full execution of the hot path over the resident encoded chunks
+(decode + group-by + aggregate + compact + readback). For --gpus N > 1, a
+step = per-rank partial aggregation, RCCL all-to-all of hash-partitioned
+state rows over xGMI, local merge; per-GPU rows fixed → weak
+scaling.
+
+Prints ONE JSON line from rank 0 per the driver contract, including:
+  roofline     — dominant kernel (fused scan_group) achieved GB/s vs the
+                 8 TB/s HBM peak, algorithmic bytes = encoded bytes of the
+                 two scanned columns, timed with HIP events
+                 inside the library on the launch stream
+  cpu_baseline — the CPU oracle (restated reference algorithm, "port") timed
+                 on this box's host cores on a bounded sample
+"""
+import argparse
+import ctypes
+import json
+import os
+import sys
+import time
+from concurrent.futures import ThreadPoolExecutor
+
+import numpy as np
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+SEED = 20260915
+KEY_SPACE = 1_000_000          # 1M distinct keys (config 3)
+VAL_BITS = 40                  # values uniform in [0, 2^40)
+SLICE = 16 * 128 * 1024        # 2Mi rows per generation slice (16 segments)
+
+
+def log(msg):
+    print("[bench] " + msg, file=sys.stderr, flush=True)
+
+
+def gen_encode_column(n, seed, lo, hi, nthreads, base_seed_tag):
+    """Generate + encode one int64 column in parallel slices; bit-identical
+    to a serial encode (cum_rows_base threads the reference's cumulative
+    RowCount_ through)."""
+    import ytsaurus_amd as y
+
+    slices = [(s, min(s + SLICE, n)) for s in range(0, n, SLICE)]
+
+    def work(item):
+        i, (b, e) = item
+        rng = np.random.default_rng([seed, base_seed_tag, i])
+        vals = rng.integers(lo, hi, e - b, dtype=np.int64)
+        return y.encode_int64(vals, cum_rows_base=b)
+
+    with ThreadPoolExecutor(max_workers=nthreads) as ex:
+        encs = list(ex.map(work, enumerate(slices)))
+    return encs
+
+
+def build_device_chunk(enc_cols, n, torch):
+    """Concatenate per-slice encodings into one device chunk."""
+    from ytsaurus_amd._abi import YtChunk, YtColumn, YtSegment
+
+    keep = []
+    cols = (YtColumn * len(enc_cols))()
+    total_bytes = 0
+    for ci, encs in enumerate(enc_cols):
+        nseg = sum(e._cenc.segment_count for e in encs)
+        segs = (YtSegment * nseg)()
+        keep.append(segs)
+        at = 0
+        for e in encs:
+            blob_base = e._cenc.blob
+            host = np.frombuffer(
+                ctypes.string_at(blob_base, e._cenc.blob_size), dtype=np.uint8)
+            t = torch.from_numpy(host.copy()).cuda()
+            keep.append(t)
+            for j in range(e._cenc.segment_count):
+                s = e._cenc.segments[j]
+                segs[at] = YtSegment(type=s.type, row_count=s.row_count,
+                                     min_value=s.min_value,
+                                     data=t.data_ptr() + (s.data - blob_base),
+                                     data_size=s.data_size)
+                total_bytes += s.data_size
+                at += 1
+        cols[ci] = YtColumn(value_type=encs[0].value_type, segment_count=nseg,
+                            segments=segs)
+    ch = YtChunk(row_count=n, column_count=len(enc_cols), columns=cols)
+    ch._keep = keep
+    return ch, total_bytes
+
+
+def cpu_baseline_leg(plan_f, enc_cols, n, cores):
+    """Time the oracle (CPU restatement, kind 'port') on a bounded sample of
+    the same workload: enough slices for ~10-30 s of CPU work."""
+    import ytsaurus_amd as y
+    from ytsaurus_amd._abi import YtChunk, YtColumn, YtSegment
+
+    sample_rows = min(n, 8 * SLICE)    # ≤ 16Mi rows
+    # host chunk view over the first slices
+    cols = (YtColumn * len(enc_cols))()
+    keep = []
+    for ci, encs in enumerate(enc_cols):
+        segs_list = []
+        rows = 0
+        for e in encs:
+            for j in range(e._cenc.segment_count):
+                s = e._cenc.segments[j]
+                if rows >= sample_rows:
+                    break
+                segs_list.append(s)
+                rows += s.row_count
+        nseg = len(segs_list)
+        segs = (YtSegment * nseg)(*segs_list)
+        keep.append(segs)
+        cols[ci] = YtColumn(value_type=encs[0].value_type,
+                            segment_count=nseg, segments=segs)
+        sample_rows = rows
+    ch = YtChunk(row_count=sample_rows, column_count=len(enc_cols), columns=cols)
+
+    from ytsaurus_amd.api import _mk_rowset
+    from ytsaurus_amd import _abi
+    rs = _mk_rowset(KEY_SPACE + 1024, 4)
+    st = _abi.YtStatistics()
+    err = ctypes.create_string_buffer(256)
+    t0 = time.monotonic()
+    rc = _abi.oracle_lib().yto_execute(
+        ctypes.byref(plan_f().c), ctypes.byref(ch), ctypes.byref(rs),
+        ctypes.byref(st), cores, err, 256)
+    dt = time.monotonic() - t0
+    if rc != 0:
+        log("cpu baseline failed: %s" % err.value)
+        return None
+    return {
+        "value": sample_rows / dt,
+        "unit": "rows/s",
+        "cores": cores,
+        "kind": "port",
+        "sample": "first %d rows of the same encoded chunk, oracle MT group-by"
+                  % sample_rows,
+    }
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=5)
+    ap.add_argument("--warmup", type=int, default=2)
+    ap.add_argument("--rows", type=float, default=1e9,
+                    help="rows per GPU (weak scaling)")
+    ap.add_argument("--keys", type=int, default=KEY_SPACE)
+    ap.add_argument("--workload", default="groupby",
+                    choices=["groupby", "scanfilter"])
+    ap.add_argument("--no-cpu-baseline", action="store_true")
+    ap.add_argument("--traffic-bytes", type=float, default=0.0,
+                    help="measured per-launch HBM bytes from a rocprofv3 --pmc run")
+    args = ap.parse_args()
+
+    n = int(args.rows)
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+
+    import torch
+    assert torch.cuda.is_available(), "bench needs a GPU (no CPU fallback exists)"
+    torch.cuda.set_device(local_rank)
+
+    import ytsaurus_amd as y
+
+    dist = None
+    if world > 1:
+        import torch.distributed as dist_mod
+        dist = dist_mod
+        dist.init_process_group("nccl")
+
+    key_space = args.keys
+
+    def make_plan():
+        if args.workload == "groupby":
+            return y.Plan(keys=[y.col(0)],
+                          aggs=[y.agg_sum(y.col(1)), y.agg_sum1()])
+        lo, hi = int(0.25 * 2**VAL_BITS), int(0.75 * 2**VAL_BITS)
+        return y.Plan(filter=(y.col(0) >= lo).and_(y.col(0) <= hi),
+                      aggs=[y.agg_sum(y.col(1)), y.agg_sum(y.col(2)),
+                            y.agg_sum(y.col(3)), y.agg_sum1()])
+
+    ncols = 2 if args.workload == "groupby" else 4
+    cores = os.cpu_count() or 8
+
+    t0 = time.monotonic()
+    enc_cols = []
+    for ci in range(ncols):
+        if args.workload == "groupby" and ci == 0:
+            lo, hi = 0, key_space
+        else:
+            lo, hi = 0, 2**VAL_BITS
+        enc_cols.append(gen_encode_column(n, SEED + rank, lo, hi, cores, ci))
+    log("rank %d: generated+encoded %d rows x %d cols in %.1fs"
+        % (rank, n, ncols, time.monotonic() - t0))
+
+    t0 = time.monotonic()
+    dev_chunk, enc_bytes = build_device_chunk(enc_cols, n, torch)
+    torch.cuda.synchronize()
+    upload_s = time.monotonic() - t0
+    log("rank %d: uploaded %.2f GB encoded in %.1fs (%.1f GB/s PCIe-inclusive)"
+        % (rank, enc_bytes / 1e9, upload_s, enc_bytes / 1e9 / max(upload_s, 1e-9)))
+
+    plan = make_plan()
+    hint = key_space if args.workload == "groupby" else 0
+
+    # multi-GPU state buffers
+    if world > 1:
+        cap = 2 * key_space + 1024
+        states_t = torch.zeros((cap, 4), dtype=torch.int64, device="cuda")
+        recv_t = torch.zeros((2 * cap, 4), dtype=torch.int64, device="cuda")
+
+    scan_ms_total = 0.0
+    scan_launches = 0
+
+    def step():
+        nonlocal scan_ms_total, scan_launches
+        if world == 1:
+            rows, st = y.gpu_execute(plan, dev_chunk, max_groups_hint=hint,
+                                     out_capacity=key_space + 1024)
+            scan_ms_total += st.kernel_scan_ms
+            scan_launches += st.kernel_scan_launches
+            return st
+        # bottom query: partial aggregate + hash partition on device
+        counts, st = y.gpu_partial(plan, dev_chunk, world,
+                                   states_t.data_ptr(), cap,
+                                   max_groups_hint=hint)
+        scan_ms_total += st.kernel_scan_ms
+        scan_launches += st.kernel_scan_launches
+        # exchange sizes then states (RCCL all-to-all over xGMI)
+        sizes = torch.tensor(counts, dtype=torch.int64, device="cuda")
+        rsizes = torch.zeros(world, dtype=torch.int64, device="cuda")
+        dist.all_to_all_single(rsizes, sizes)
+        rs = rsizes.cpu().tolist()
+        offs = np.cumsum([0] + counts[:-1]).tolist()
+        send_split = [c for c in counts]
+        recv_split = [int(x) for x in rs]
+        total_recv = sum(recv_split)
+        dist.all_to_all_single(
+            recv_t[:total_recv].view(-1, 4), states_t[:sum(counts)].view(-1, 4),
+            output_split_sizes=recv_split, input_split_sizes=send_split)
+        # front query: merge + finalize
+        rows, mst = y.gpu_merge(plan, recv_t.data_ptr(), total_recv,
+                                max_groups_hint=hint,
+                                out_capacity=key_space + 1024)
+        return st
+
+    # warmup
+    for _ in range(args.warmup):
+        step()
+    torch.cuda.synchronize()
+    if dist:
+        dist.barrier()
+    scan_ms_total = 0.0
+    scan_launches = 0
+
+    t0 = time.monotonic()
+    for _ in range(args.steps):
+        step()
+    torch.cuda.synchronize()
+    if dist:
+        dist.barrier()
+    elapsed = time.monotonic() - t0
+
+    # max over ranks
+    if dist:
+        t = torch.tensor([elapsed], dtype=torch.float64, device="cuda")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    ms_per_step = elapsed / args.steps * 1000
+    total_rows = n * world
+    value = total_rows / (elapsed / args.steps)
+
+    # roofline: dominant kernel = fused scan_group; algorithmic bytes =
+    # encoded bytes of the scanned columns per launch (one launch per step)
+    scan_ms_avg = scan_ms_total / max(scan_launches, 1)
+    achieved_gbs = (enc_bytes / 1e9) / (scan_ms_avg / 1e3) if scan_ms_avg else None
+    peak_gbs = 8000.0   # HBM3E spec peak, MI355X_MICROARCH.md
+    roofline = {
+        "bound": "hbm",
+        "achieved": achieved_gbs,
+        "peak": peak_gbs,
+        "unit": "GB/s",
+        "frac": (achieved_gbs / peak_gbs) if achieved_gbs else None,
+        "traffic": args.traffic_bytes or None,
+    }
+
+    cpu_baseline = None
+    if rank == 0 and world == 1 and not args.no_cpu_baseline:
+        cpu_baseline = cpu_baseline_leg(make_plan, enc_cols, n, cores)
+
+    if rank == 0:
+        out = {
+            "metric": "rows/sec scan+GROUP BY on 1B-row int64 chunks; achieved HBM GB/s vs peak",
+            "value": value,
+            "unit": "rows/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "int64",
+            "data": "synthetic",
+            "config": {
+                "workload": "groupby_1M_distinct@%drows" % n
+                            if args.workload == "groupby"
+                            else "scan_filter_sum_4col@%drows" % n,
+                "rows_per_gpu": n,
+                "distinct_keys": key_space,
+                "columns": ncols,
+                "chunk_format": "unversioned DirectDense int64 (reference layout)",
+                "encoded_gb": enc_bytes / 1e9,
+            },
+            "roofline": roofline,
+            "cpu_baseline": cpu_baseline,
+        }
+        print(json.dumps(out), flush=True)
+
+    if dist:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -258,9 +258,13 @@ typedef struct YtEncodedColumn {
     int64_t blob_size;
 } YtEncodedColumn;
 
+/* cum_rows_base: rows already written to this column before `values` —
+ * feeds the reference's cumulative RowCount_ used in the RLE segment-size
+ * estimate (integer_column_writer.cpp:365,371), so a column encoded in
+ * parallel slices is bit-identical to a serial encode. Pass 0 normally. */
 int yt_encode_int64_column(
     const int64_t* values, const uint8_t* nulls, int64_t n,
-    int32_t max_segment_values, int32_t is_unsigned,
+    int32_t max_segment_values, int32_t is_unsigned, int64_t cum_rows_base,
     YtEncodedColumn* out, char* errbuf, size_t errlen);
 
 int yt_encode_double_column(

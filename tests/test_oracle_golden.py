@@ -1,0 +1,147 @@
+"""Oracle vs golden vectors transcribed from the reference's own tests
+(tests/golden/cases.json; citations inside). This is what pins parity, since
+the reference binary cannot be built here.
+"""
+import numpy as np
+import pytest
+
+import ytsaurus_amd as y
+from helpers import build_chunk, build_plan, load_cases, norm_rows
+
+CASES = load_cases()
+
+
+@pytest.mark.parametrize("case", CASES, ids=[c["name"] for c in CASES])
+def test_golden(case):
+    chunk = build_chunk(case["columns"], case["rows"])
+    plan = build_plan(case["plan"])
+    rows, _ = y.oracle_execute(plan, chunk)
+    got = norm_rows(rows)
+    want = norm_rows([tuple(r) for r in case["expected"]])
+    if case.get("ordered"):
+        assert got == want
+    else:
+        assert y.sort_rows(got) == y.sort_rows(want)
+
+
+@pytest.mark.parametrize("case", [c for c in CASES if c["plan"].get("keys")],
+                         ids=[c["name"] for c in CASES if c["plan"].get("keys")])
+def test_golden_segmented_and_mt(case):
+    """Same goldens with tiny segments (exercises multi-segment decode) and
+    the multithreaded oracle (exercises partial-merge; order-free compare,
+    like the reference's OrderedResultMatcher)."""
+    chunk = build_chunk(case["columns"], case["rows"], max_seg=4)
+    plan = build_plan(case["plan"])
+    rows, _ = y.oracle_execute(plan, chunk, nthreads=3)
+    got = y.sort_rows(norm_rows(rows))
+    want = y.sort_rows(norm_rows([tuple(r) for r in case["expected"]]))
+    assert got == want
+
+
+def test_group_by_no_limit_coordinated():
+    """GroupByNoLimitCoordinated (ql_query_ut.cpp:3685-3733): the expected
+    result is computed in the reference test itself; restated here.
+    100 rows a=i/10, b=i%10, v=i; query: k, x, sum(b) group by a as k, v%2 as x.
+    Exercised through the single pass AND the two-phase partial/merge path
+    (= the reference's bottom/front coordinated split)."""
+    n = 100
+    a = np.array([i // 10 for i in range(n)], dtype=np.int64)
+    b = np.array([i % 10 for i in range(n)], dtype=np.int64)
+    v = np.arange(n, dtype=np.int64)
+
+    grouped = {}
+    for i in range(n):
+        grouped.setdefault((i // 10, i % 2), 0)
+        grouped[(i // 10, i % 2)] += i % 10
+    want = sorted((k, x, s) for (k, x), s in grouped.items())
+
+    chunk = y.Chunk([y.encode_int64(a), y.encode_int64(b), y.encode_int64(v)], n)
+    # our GPU round-1 plan family is single-key; emulate the two-key group
+    # with key = a*2 + v%2 (injective for this data), then project back
+    plan = y.Plan(keys=[y.col(0) * 2 + (y.col(2) % 2)],
+                  aggs=[y.agg_sum(y.col(1))],
+                  projects=[y.col(0) // 2, y.col(0) % 2, y.col(1)])
+    rows, _ = y.oracle_execute(plan, chunk)
+    assert sorted(rows) == want
+
+    # two-phase: shard rows 0..29 / 30..59 / 60..99 like the reference test
+    plan2 = y.Plan(keys=[y.col(0) * 2 + (y.col(2) % 2)],
+                   aggs=[y.agg_sum(y.col(1)), y.agg_sum1()])
+    shards = [(0, 30), (30, 60), (60, 100)]
+    parts = 4
+    all_states = [[] for _ in range(parts)]
+    for lo, hi in shards:
+        ch = y.Chunk([y.encode_int64(a[lo:hi]), y.encode_int64(b[lo:hi]),
+                      y.encode_int64(v[lo:hi])], hi - lo)
+        states, counts = y.oracle_partial(plan2, ch, parts)
+        at = 0
+        for p in range(parts):
+            for i in range(counts[p]):
+                all_states[p].append(states[at + i])
+            at += counts[p]
+    # merge each partition separately (as each destination GPU would), union
+    merged = []
+    for p in range(parts):
+        import ctypes as C
+        from ytsaurus_amd._abi import YtStateRow
+        arr = (YtStateRow * max(len(all_states[p]), 1))()
+        for i, s in enumerate(all_states[p]):
+            arr[i] = s
+        merged += y.oracle_merge(plan2, [(arr, len(all_states[p]))])
+    got = sorted((r[0] // 2, r[0] % 2, r[1]) for r in merged)
+    assert got == want
+    # sum(1) must equal group sizes: 5 rows per (k,x) group
+    assert all(r[2] == 5 for r in merged)
+
+
+def test_division_by_zero():
+    chunk = build_chunk([["a", "int64"]], [[1], [2]])
+    plan = y.Plan(keys=[y.col(0) % 0], aggs=[y.agg_sum1()])
+    rc, _, _ = y.oracle_execute(plan, chunk, expect_error=True)
+    from ytsaurus_amd._abi import YT_ERR_DIV_ZERO
+    assert rc == YT_ERR_DIV_ZERO
+
+
+def test_null_comparison_semantics():
+    """Non-canonical null relations (cg_fragment_compiler.cpp:1621-1649):
+    null < any value, null == null; result is non-null Boolean."""
+    chunk = build_chunk([["a", "int64"], ["b", "int64"]],
+                        [[1, None], [2, 5]])
+    # where b < 0 keeps only the null row (null is smallest)
+    plan = y.Plan(filter=y.col(1) < 0, projects=[y.col(0)])
+    rows, _ = y.oracle_execute(plan, chunk)
+    assert rows == [(1,)]
+    # where b == null is true only for the null row under non-canonical rules
+    plan = y.Plan(filter=y.col(1) == y.null(), projects=[y.col(0)])
+    rows, _ = y.oracle_execute(plan, chunk)
+    assert rows == [(1,)]
+
+
+def test_global_aggregate():
+    """BASELINE config 1 shape: SELECT sum(v) (implicit empty group key)."""
+    rng = np.random.default_rng(9)
+    vals = rng.integers(-10**9, 10**9, 10000, dtype=np.int64)
+    chunk = y.Chunk([y.encode_int64(vals)], len(vals))
+    plan = y.Plan(aggs=[y.agg_sum(y.col(0)), y.agg_sum1()])
+    rows, st = y.oracle_execute(plan, chunk)
+    assert rows == [(int(vals.sum()), 10000)]
+    # empty input → no output rows (single final flush of zero groups)
+    empty = y.Chunk([y.encode_int64(np.array([], dtype=np.int64))], 0)
+    rows, _ = y.oracle_execute(plan, empty)
+    assert rows == []
+    # all-null input → sum null, count present
+    nulls = np.ones(100, dtype=np.uint8)
+    chn = y.Chunk([y.encode_int64(np.zeros(100, dtype=np.int64), nulls)], 100)
+    rows, _ = y.oracle_execute(plan, chn)
+    assert rows == [(None, 100)]
+
+
+def test_int64_wrap():
+    """int64 sum wraps mod 2^64 (udf/sum.c add on two's complement)."""
+    vals = np.array([2**63 - 1, 2**63 - 1, 5], dtype=np.int64)
+    chunk = y.Chunk([y.encode_int64(vals)], 3)
+    plan = y.Plan(aggs=[y.agg_sum(y.col(0))])
+    rows, _ = y.oracle_execute(plan, chunk)
+    want = (2**63 - 1) * 2 + 5
+    want = ((want + 2**63) % 2**64) - 2**63
+    assert rows == [(want,)]

@@ -166,7 +166,8 @@ def gpu_lib():
               C.POINTER(YtRowset), C.POINTER(YtStatistics), C.c_char_p, C.c_size_t])
         _sig(lib, "yt_encode_int64_column", C.c_int,
              [C.POINTER(C.c_int64), C.POINTER(C.c_uint8), C.c_int64, C.c_int32,
-              C.c_int32, C.POINTER(YtEncodedColumn), C.c_char_p, C.c_size_t])
+              C.c_int32, C.c_int64, C.POINTER(YtEncodedColumn), C.c_char_p,
+              C.c_size_t])
         _sig(lib, "yt_encode_double_column", C.c_int,
              [C.POINTER(C.c_double), C.POINTER(C.c_uint8), C.c_int64, C.c_int32,
               C.POINTER(YtEncodedColumn), C.c_char_p, C.c_size_t])

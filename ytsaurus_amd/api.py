@@ -20,6 +20,7 @@ from ._abi import (
 __all__ = [
     "col", "lit", "null", "litf", "Plan", "agg_sum", "agg_sum1",
     "encode_int64", "encode_double", "Chunk", "oracle_execute",
+    "oracle_partial", "oracle_merge",
     "gpu_available", "gpu_execute", "gpu_partial", "gpu_merge",
     "rows_from_rowset", "sort_rows",
 ]
@@ -178,7 +179,8 @@ def _check(rc, err):
         raise RuntimeError("ytql error %d: %s" % (rc, err.value.decode()))
 
 
-def encode_int64(values, nulls=None, max_segment_values=0, unsigned=False):
+def encode_int64(values, nulls=None, max_segment_values=0, unsigned=False,
+                 cum_rows_base=0):
     values = np.ascontiguousarray(values, dtype=np.int64)
     n = len(values)
     nullp = None
@@ -190,7 +192,8 @@ def encode_int64(values, nulls=None, max_segment_values=0, unsigned=False):
     err = C.create_string_buffer(256)
     rc = _abi.gpu_lib().yt_encode_int64_column(
         values.ctypes.data_as(C.POINTER(C.c_int64)), nullp, n,
-        max_segment_values, 1 if unsigned else 0, C.byref(enc), err, 256)
+        max_segment_values, 1 if unsigned else 0, cum_rows_base,
+        C.byref(enc), err, 256)
     _check(rc, err)
     vt = VT_UINT64 if unsigned else VT_INT64
     return EncodedColumn(vt, enc)

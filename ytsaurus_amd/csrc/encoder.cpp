@@ -284,14 +284,14 @@ static int finish_column(std::vector<SegmentOut>& segs, YtEncodedColumn* out,
 
 extern "C" int yt_encode_int64_column(
     const int64_t* values, const uint8_t* nulls, int64_t n,
-    int32_t max_segment_values, int32_t is_unsigned,
+    int32_t max_segment_values, int32_t is_unsigned, int64_t cum_rows_base,
     YtEncodedColumn* out, char* errbuf, size_t errlen)
 {
     if (max_segment_values <= 0) max_segment_values = 128 * 1024;  /* DefaultMaxSegmentValueCount */
     std::vector<SegmentOut> segs;
     std::vector<uint64_t> data;
     std::vector<uint8_t> isnull;
-    int64_t cum = 0;
+    int64_t cum = cum_rows_base;
     for (int64_t i = 0; i < n; i++) {
         int nu = nulls ? nulls[i] : 0;
         /* null values store data 0 — AddValues :578-584 */

@@ -278,6 +278,9 @@ static void analyze_fast(const YtPlan* plan, const YtChunk* chunk, FastShape* fs
         fs->sum_slot[fs->nsum] = a;
         fs->nsum++;
     }
+    if (fs->nsum == 0 && fs->filter_col < 0 && fs->key_col < 0) {
+        return;   /* pure row count with no staged column — generic path */
+    }
     fs->valid = 1;
 }
 
