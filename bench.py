@@ -213,6 +213,7 @@ def main():
 
     plan = make_plan()
     hint = key_space if args.workload == "groupby" else 0
+    out_rs = y.make_rowset(key_space + 1024, 1 + len(plan.aggs))
 
     # multi-GPU state buffers
     if world > 1:
@@ -226,8 +227,8 @@ def main():
     def step():
         nonlocal scan_ms_total, scan_launches
         if world == 1:
-            rows, st = y.gpu_execute(plan, dev_chunk, max_groups_hint=hint,
-                                     out_capacity=key_space + 1024)
+            _, st = y.gpu_execute(plan, dev_chunk, max_groups_hint=hint,
+                                  rowset=out_rs, raw_rowset=True)
             scan_ms_total += st.kernel_scan_ms
             scan_launches += st.kernel_scan_launches
             return st

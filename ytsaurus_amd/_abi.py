@@ -148,6 +148,13 @@ def gpu_lib():
     """The product library. Raises if not built — no silent fallback."""
     global _gpu_lib
     if _gpu_lib is None:
+        # bind the HIP runtime torch already loaded (torch ships its own
+        # libamdhip64; loading ours first would map a second runtime that
+        # sees no device once torch initializes)
+        try:
+            import torch  # noqa: F401
+        except Exception:
+            pass
         path = os.path.join(_HERE, "libytql_gpu.so")
         if not os.path.exists(path):
             raise RuntimeError(

@@ -22,7 +22,7 @@ __all__ = [
     "encode_int64", "encode_double", "Chunk", "oracle_execute",
     "oracle_partial", "oracle_merge",
     "gpu_available", "gpu_execute", "gpu_partial", "gpu_merge",
-    "rows_from_rowset", "sort_rows",
+    "rows_from_rowset", "sort_rows", "make_rowset",
 ]
 
 
@@ -359,19 +359,30 @@ def gpu_available():
 
 
 def gpu_execute(plan, device_chunk, max_groups_hint=0, group_row_limit=0,
-                stream=0, out_capacity=None):
-    """device_chunk: YtChunk with device pointers (Chunk.c_device)."""
+                stream=0, out_capacity=None, rowset=None, raw_rowset=False):
+    """device_chunk: YtChunk with device pointers (Chunk.c_device).
+    Pass a preallocated `rowset` (from make_rowset) to avoid per-call
+    allocation; raw_rowset=True skips the Python row conversion."""
     opts = YtExecOptions(max_groups_hint=max_groups_hint,
                          group_row_limit=group_row_limit, stream=stream)
-    cap = out_capacity or max(int(max_groups_hint) * 2 + 1024, 1 << 16)
-    rs = _mk_rowset(cap)
+    if rowset is not None:
+        rs = rowset
+    else:
+        cap = out_capacity or max(int(max_groups_hint) * 2 + 1024, 1 << 16)
+        rs = _mk_rowset(cap)
     st = YtStatistics()
     err = C.create_string_buffer(512)
     rc = _abi.gpu_lib().yt_gpu_query_execute(
         C.byref(plan.c), C.byref(device_chunk), C.byref(opts), C.byref(rs),
         C.byref(st), err, 512)
     _check(rc, err)
+    if raw_rowset:
+        return rs, st
     return rows_from_rowset(rs), st
+
+
+def make_rowset(capacity, ncols):
+    return _mk_rowset(capacity, ncols)
 
 
 def gpu_partial(plan, device_chunk, nparts, states_dev_ptr, capacity_rows,

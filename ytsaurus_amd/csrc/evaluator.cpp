@@ -582,7 +582,7 @@ static int run_scan(const YtPlan* plan, const YtChunk* chunk,
             size_t per = ((((size_t)tile_rows * w / 64 + 2) * 8 + 15) & ~(size_t)15)
                        + (((size_t)tile_rows / 8 + 15) & ~(size_t)15);
             lds = (size_t)nused * per + 1024;
-            if (lds <= 144 * 1024 || tile_rows == 256) break;
+            if (lds <= 50 * 1024 || tile_rows == 256) break;
             tile_rows >>= 1;
         }
         if (lds > 160 * 1024) { set_err(errbuf, errlen, "LDS overflow"); rc = YT_ERR_UNSUPPORTED; goto fail; }
@@ -819,6 +819,14 @@ extern "C" int yt_gpu_query_execute(
     rc = setup_chunk(chunk, &R, &maxw, errbuf, errlen);
     if (rc) return rc;
 
+    if (chunk->row_count == 0 || R.nsegs == 0) {
+        /* zero groups -> single final flush emits nothing (registry.cpp:1481) */
+        output->row_count = 0;
+        output->column_count = plan->project_count ? plan->project_count
+            : (plan->key_count + plan->agg_count);
+        return YT_OK;
+    }
+
     rc = setup_table(&R, plan->agg_count,
                      options->max_groups_hint,
                      options->group_row_limit, errbuf, errlen);
@@ -909,6 +917,10 @@ extern "C" int yt_gpu_query_partial(
     unsigned maxw = 0;
     rc = setup_chunk(chunk, &R, &maxw, errbuf, errlen);
     if (rc) return rc;
+    if (chunk->row_count == 0 || R.nsegs == 0) {
+        for (int p = 0; p < partition_count; p++) part_counts[p] = 0;
+        return YT_OK;
+    }
     rc = setup_table(&R, plan->agg_count, options->max_groups_hint,
                      options->group_row_limit, errbuf, errlen);
     if (rc) return rc;
