@@ -194,13 +194,13 @@ def test_two_phase_gpu_matches_single(cuda):
     assert y.sort_rows(merged) == y.sort_rows(single)
 
     # property checks at size: sums of sums and counts match a global run
+    # (single rows are [key, sum(v), sum(1)])
     glob = y.Plan(aggs=[y.agg_sum(y.col(1)), y.agg_sum1()])
     grow, _ = y.gpu_execute(glob, chunk.c_device(cuda))
-    tot_cnt = sum(r[1] for r in single)
+    tot_cnt = sum(r[2] for r in single)
     assert tot_cnt == grow[0][1] == n
     M = 1 << 64
-    tot_sum = sum((r[0] if r[0] is not None else 0) % M for r in single
-                  if True) % M
+    tot_sum = sum((r[1] if r[1] is not None else 0) % M for r in single) % M
     want_sum = (grow[0][0] or 0) % M
     assert tot_sum == want_sum
     assert total == len(single)
