@@ -17,7 +17,7 @@ struct DevSeg {
     int32_t is_signed;   /* 1 = int64 (zigzag), 0 = uint64 */
     int64_t start_row;   /* first chunk row covered */
     int32_t row_count;
-    int32_t pad_;
+    int32_t col;         /* owning column index */
     uint64_t min_value;
     const uint64_t* blob;
     int64_t blob_bytes;
@@ -78,8 +78,10 @@ struct DevPlan {
 struct TableHdr {
     uint64_t nslots;          /* power of two */
     uint64_t mask;
-    /* side groups: [0] = key bits == 0, [1] = null key */
+    /* side groups: [0] = the in-table empty-sentinel key (bits in
+     * side_key_bits[0]), [1] = null key */
     uint64_t side_used[2];    /* 0/1 */
+    uint64_t side_key_bits[2];
     uint64_t side_cnt[2];
     uint64_t side_agg[2][2 * kMaxAggs];  /* bits, nonnull per agg */
     unsigned long long ngroups;          /* inserted (excl. side) */
@@ -130,6 +132,35 @@ struct FastParams {
     int64_t row_count;
     int32_t nsegs_per_col;
     int32_t ntiles;
+};
+
+/* two-phase partitioned group-by (BASELINE configs 3/4 family):
+ * Phase A (k_scan_partition): fused decode+filter, hash-partition rows into
+ * kNB buckets as 16-byte {key,val} records (mirrors the reference's
+ * in-process shuffle partition, shuffling_reader.cpp:40-42, pushed down to
+ * the row level so aggregation can run in LDS).
+ * Phase B (k_bucket_agg): one workgroup per bucket aggregates its records
+ * in an LDS open-addressing table and emits compacted OutGroups. */
+constexpr int kNB = 1024;          /* partition buckets */
+constexpr int kHSlots = 4096;      /* LDS table slots per bucket */
+constexpr uint64_t kEmptyKey = 0x8000000000000000ULL;  /* INT64_MIN bits */
+
+struct PartParams {
+    int32_t tile_rows;
+    int32_t tiles_per_seg;
+    int32_t ntiles;
+    int32_t filter_idx;           /* used-col index; -1 none */
+    int32_t key_idx;              /* used-col index (required) */
+    int32_t val_idx;              /* used-col index of the sum arg; -1 none */
+    int32_t nused;
+    int32_t has_val_nulls;
+    int32_t has_key_nulls;
+    int32_t sum_slot;             /* agg slot of the sum; -1 none */
+    int32_t agg_count;
+    int32_t pad_;
+    int64_t filter_lo, filter_hi;
+    int64_t bucket_stride;        /* record capacity per bucket */
+    int64_t nbucket_stride;       /* null-stream capacity per bucket */
 };
 
 struct KernelTimes {

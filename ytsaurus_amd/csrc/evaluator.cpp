@@ -23,7 +23,15 @@ using namespace ytql;
 
 /* launch wrappers from kernels.hip */
 extern "C" {
-hipError_t ytql_launch_parse_segments(const DevSeg*, int, SegEx*, unsigned*, hipStream_t);
+hipError_t ytql_launch_parse_segments(const DevSeg*, int, SegEx*, unsigned*, unsigned*, hipStream_t);
+hipError_t ytql_launch_scan_partition(const PartParams*, const DevSeg*, const SegEx*,
+                                      const FastCol*, TableHdr*, unsigned long long*,
+                                      void*, unsigned long long*, uint64_t*,
+                                      size_t, int, hipStream_t);
+hipError_t ytql_launch_bucket_agg(const void*, const unsigned long long*, int64_t,
+                                  const uint64_t*, const unsigned long long*, int64_t,
+                                  OutGroup*, unsigned long long*, int64_t,
+                                  TableHdr*, int, int, hipStream_t);
 hipError_t ytql_launch_scan_generic(const DevPlan*, const DevSeg*, const SegEx*,
                                     const int32_t*, const int32_t*, int64_t,
                                     TableHdr*, unsigned long long*, unsigned*, hipStream_t);
@@ -422,8 +430,16 @@ struct DeviceRun {
     unsigned long long* d_counter = nullptr;
     unsigned long long* d_gaccum = nullptr;
     FastCol* d_fastcols = nullptr;
+    unsigned* d_colnull = nullptr;
+    unsigned long long* d_cursors = nullptr;
+    void* d_recs = nullptr;
+    unsigned long long* d_ncursors = nullptr;
+    uint64_t* d_nrecs = nullptr;
     uint64_t nslots = 0;
     int nsegs = 0;
+    int64_t groups_capacity = 0;
+    bool groups_compacted = false;     /* d_groups already holds final groups */
+    unsigned col_null_flags[kMaxCols] = {0};
     hipStream_t stream = 0;
 
     ~DeviceRun()
@@ -442,6 +458,11 @@ struct DeviceRun {
         if (d_counter) e = hipFree(d_counter);
         if (d_gaccum) e = hipFree(d_gaccum);
         if (d_fastcols) e = hipFree(d_fastcols);
+        if (d_colnull) e = hipFree(d_colnull);
+        if (d_cursors) e = hipFree(d_cursors);
+        if (d_recs) e = hipFree(d_recs);
+        if (d_ncursors) e = hipFree(d_ncursors);
+        if (d_nrecs) e = hipFree(d_nrecs);
     }
 };
 
@@ -472,7 +493,7 @@ static int setup_chunk(const YtChunk* chunk, DeviceRun* R, unsigned* maxw_out,
             d.is_signed = (col.value_type == YT_VT_INT64);
             d.start_row = row;
             d.row_count = seg.row_count;
-            d.pad_ = 0;
+            d.col = c;
             d.min_value = seg.min_value;
             d.blob = (const uint64_t*)seg.data;
             d.blob_bytes = seg.data_size;
@@ -493,6 +514,8 @@ static int setup_chunk(const YtChunk* chunk, DeviceRun* R, unsigned* maxw_out,
     HIP_CHECK(hipMalloc(&R->d_cnt, sizeof(int32_t) * ncols));
     HIP_CHECK(hipMalloc(&R->d_maxw, sizeof(unsigned)));
     HIP_CHECK(hipMalloc(&R->d_err, sizeof(unsigned)));
+    HIP_CHECK(hipMalloc(&R->d_colnull, sizeof(unsigned) * kMaxCols));
+    HIP_CHECK(hipMemsetAsync(R->d_colnull, 0, sizeof(unsigned) * kMaxCols, R->stream));
     HIP_CHECK(hipMemcpyAsync(R->d_segs, R->h_segs.data(), sizeof(DevSeg) * R->nsegs,
                              hipMemcpyHostToDevice, R->stream));
     HIP_CHECK(hipMemcpyAsync(R->d_off, R->h_off.data(), sizeof(int32_t) * ncols,
@@ -501,8 +524,12 @@ static int setup_chunk(const YtChunk* chunk, DeviceRun* R, unsigned* maxw_out,
                              hipMemcpyHostToDevice, R->stream));
     HIP_CHECK(hipMemsetAsync(R->d_maxw, 0, sizeof(unsigned), R->stream));
     HIP_CHECK(hipMemsetAsync(R->d_err, 0, sizeof(unsigned), R->stream));
-    HIP_CHECK(ytql_launch_parse_segments(R->d_segs, R->nsegs, R->d_segex, R->d_maxw, R->stream));
+    HIP_CHECK(ytql_launch_parse_segments(R->d_segs, R->nsegs, R->d_segex, R->d_maxw,
+                                         R->d_colnull, R->stream));
     HIP_CHECK(hipMemcpyAsync(maxw_out, R->d_maxw, sizeof(unsigned),
+                             hipMemcpyDeviceToHost, R->stream));
+    HIP_CHECK(hipMemcpyAsync(R->col_null_flags, R->d_colnull,
+                             sizeof(unsigned) * kMaxCols,
                              hipMemcpyDeviceToHost, R->stream));
     HIP_CHECK(hipStreamSynchronize(R->stream));
     return YT_OK;
@@ -531,6 +558,164 @@ fail:
     return rc;
 }
 
+/* Two-phase partitioned group-by for the {1 direct int64 key, <=1 direct
+ * sum, sum(1), optional range filter} family. On success sets *done and
+ * leaves COMPACTED groups in R->d_groups (+ counter in R->d_counter);
+ * sets *done=false (no error) when a capacity guard tripped so the caller
+ * falls back to the direct-table kernel. */
+static int run_partitioned(const YtPlan* plan, const YtChunk* chunk,
+                           const YtExecOptions* options, DeviceRun* R,
+                           const FastShape* fs, unsigned maxw,
+                           YtStatistics* stats, bool* done,
+                           char* errbuf, size_t errlen)
+{
+    int rc = YT_OK;
+    *done = false;
+    hipEvent_t ev0 = nullptr, ev1 = nullptr, ev2 = nullptr;
+
+    PartParams pp;
+    memset(&pp, 0, sizeof(pp));
+    int used[4];
+    int nused = 0;
+    auto add_used = [&](int col) -> int {
+        for (int i = 0; i < nused; i++) if (used[i] == col) return i;
+        used[nused] = col;
+        return nused++;
+    };
+    pp.filter_idx = fs->filter_col >= 0 ? add_used(fs->filter_col) : -1;
+    pp.key_idx = add_used(fs->key_col);
+    pp.val_idx = fs->nsum == 1 ? add_used(fs->sum_col[0]) : -1;
+    pp.nused = nused;
+    pp.sum_slot = fs->nsum == 1 ? fs->sum_slot[0] : -1;
+    pp.agg_count = plan->agg_count;
+    pp.filter_lo = fs->filter_lo;
+    pp.filter_hi = fs->filter_hi;
+    pp.has_key_nulls = (R->col_null_flags[fs->key_col] != 0);
+    pp.has_val_nulls = pp.val_idx >= 0 && (R->col_null_flags[fs->sum_col[0]] != 0);
+
+    const int64_t rows = chunk->row_count;
+    int32_t seg0_rows = chunk->columns[0].segments[0].row_count;
+
+    /* LDS: 4*kNB u32 histogram structures + column staging */
+    unsigned w = maxw ? maxw : 1;
+    int tile_rows = 8192;
+    size_t lds = 0;
+    for (;;) {
+        size_t per = ((((size_t)tile_rows * w / 64 + 2) * 8 + 15) & ~(size_t)15)
+                   + (((size_t)tile_rows / 8 + 15) & ~(size_t)15);
+        lds = 4 * kNB * 4 + (size_t)nused * per + 256;
+        if (lds <= 64 * 1024 || tile_rows == 256) break;
+        tile_rows >>= 1;
+    }
+    if (lds > 160 * 1024) return YT_OK;   /* fall back */
+    while (tile_rows > 256 && tile_rows > seg0_rows) tile_rows >>= 1;
+    pp.tile_rows = tile_rows;
+    pp.tiles_per_seg = (seg0_rows + tile_rows - 1) / tile_rows;
+    {
+        int nseg = chunk->columns[0].segment_count;
+        int32_t last_rows = chunk->columns[0].segments[nseg - 1].row_count;
+        pp.ntiles = (nseg - 1) * pp.tiles_per_seg + (last_rows + tile_rows - 1) / tile_rows;
+    }
+    pp.bucket_stride = rows / kNB + (rows / kNB) / 2 + 8192;
+    pp.nbucket_stride = pp.has_val_nulls ? pp.bucket_stride : 0;
+
+    std::vector<FastCol> fc(nused);
+    for (int u = 0; u < nused; u++) {
+        fc[u].seg_off = R->h_off[used[u]];
+        fc[u].seg_cnt = R->h_cnt[used[u]];
+    }
+
+    {
+        HIP_CHECK(hipMalloc(&R->d_fastcols, sizeof(FastCol) * nused));
+        HIP_CHECK(hipMemcpyAsync(R->d_fastcols, fc.data(), sizeof(FastCol) * nused,
+                                 hipMemcpyHostToDevice, R->stream));
+        HIP_CHECK(hipMalloc(&R->d_cursors, sizeof(uint64_t) * kNB));
+        HIP_CHECK(hipMemsetAsync(R->d_cursors, 0, sizeof(uint64_t) * kNB, R->stream));
+        HIP_CHECK(hipMalloc(&R->d_recs, (size_t)kNB * pp.bucket_stride * 16));
+        if (pp.has_val_nulls) {
+            HIP_CHECK(hipMalloc(&R->d_ncursors, sizeof(uint64_t) * kNB));
+            HIP_CHECK(hipMemsetAsync(R->d_ncursors, 0, sizeof(uint64_t) * kNB, R->stream));
+            HIP_CHECK(hipMalloc(&R->d_nrecs, (size_t)kNB * pp.nbucket_stride * 8));
+        }
+        /* the partitioned path's in-table sentinel is INT64_MIN bits */
+        uint64_t sk = kEmptyKey;
+        HIP_CHECK(hipMemcpyAsync((char*)R->d_th + offsetof(TableHdr, side_key_bits),
+                                 &sk, 8, hipMemcpyHostToDevice, R->stream));
+
+        int64_t cap_groups = (int64_t)kNB * kHSlots;
+        if (options->max_groups_hint > 0) {
+            int64_t want = options->max_groups_hint * 2 + 4096;
+            if (want < cap_groups) cap_groups = want;
+        }
+        if (cap_groups > rows + 16) cap_groups = rows + 16;
+        R->groups_capacity = cap_groups;
+        HIP_CHECK(hipMalloc(&R->d_groups, sizeof(OutGroup) * cap_groups));
+        HIP_CHECK(hipMalloc(&R->d_counter, sizeof(unsigned long long)));
+        HIP_CHECK(hipMemsetAsync(R->d_counter, 0, sizeof(unsigned long long), R->stream));
+
+        HIP_CHECK(hipEventCreate(&ev0));
+        HIP_CHECK(hipEventCreate(&ev1));
+        HIP_CHECK(hipEventCreate(&ev2));
+        int grid = pp.ntiles < 2048 ? (pp.ntiles ? pp.ntiles : 1) : 2048;
+        HIP_CHECK(hipEventRecord(ev0, R->stream));
+        HIP_CHECK(ytql_launch_scan_partition(&pp, R->d_segs, R->d_segex, R->d_fastcols,
+                                             R->d_th, R->d_cursors, R->d_recs,
+                                             R->d_ncursors, R->d_nrecs,
+                                             lds, grid, R->stream));
+        HIP_CHECK(hipEventRecord(ev1, R->stream));
+        HIP_CHECK(ytql_launch_bucket_agg(R->d_recs, R->d_cursors, pp.bucket_stride,
+                                         R->d_nrecs, R->d_ncursors, pp.nbucket_stride,
+                                         R->d_groups, R->d_counter, cap_groups,
+                                         R->d_th, pp.sum_slot, pp.agg_count,
+                                         R->stream));
+        HIP_CHECK(hipEventRecord(ev2, R->stream));
+        HIP_CHECK(hipStreamSynchronize(R->stream));
+        float msA = 0, msB = 0;
+        HIP_CHECK(hipEventElapsedTime(&msA, ev0, ev1));
+        HIP_CHECK(hipEventElapsedTime(&msB, ev1, ev2));
+        if (stats) {
+            stats->kernel_scan_ms += msA;
+            stats->kernel_scan_launches += 1;
+            stats->kernel_other_ms += msB;
+        }
+        /* overflow==1 → capacity guard; retry on the direct path */
+        TableHdr th;
+        HIP_CHECK(hipMemcpy(&th, R->d_th, sizeof(th), hipMemcpyDeviceToHost));
+        if (th.overflow == 1) {
+            /* reset and fall back */
+            hipFree(R->d_groups); R->d_groups = nullptr;
+            hipFree(R->d_counter); R->d_counter = nullptr;
+            hipFree(R->d_recs); R->d_recs = nullptr;
+            hipFree(R->d_cursors); R->d_cursors = nullptr;
+            if (R->d_nrecs) { hipFree(R->d_nrecs); R->d_nrecs = nullptr; }
+            if (R->d_ncursors) { hipFree(R->d_ncursors); R->d_ncursors = nullptr; }
+            hipFree(R->d_fastcols); R->d_fastcols = nullptr;
+            TableHdr hh;
+            memset(&hh, 0, sizeof(hh));
+            hh.nslots = R->nslots;
+            hh.mask = R->nslots - 1;
+            hh.group_limit = options->group_row_limit;
+            HIP_CHECK(hipMemcpyAsync(R->d_th, &hh, sizeof(hh), hipMemcpyHostToDevice, R->stream));
+            int stride = 2 + 2 * plan->agg_count;
+            HIP_CHECK(hipMemsetAsync(R->d_slots, 0,
+                                     sizeof(uint64_t) * R->nslots * stride, R->stream));
+            *done = false;
+        } else {
+            R->groups_compacted = true;
+            *done = true;
+        }
+    }
+    hipEventDestroy(ev0);
+    hipEventDestroy(ev1);
+    hipEventDestroy(ev2);
+    return YT_OK;
+fail:
+    if (ev0) hipEventDestroy(ev0);
+    if (ev1) hipEventDestroy(ev1);
+    if (ev2) hipEventDestroy(ev2);
+    return rc;
+}
+
 /* run the scan (fast when possible), leaving results in the table/gaccum.
  * Fills stats->kernel_scan_* from device events. */
 static int run_scan(const YtPlan* plan, const YtChunk* chunk,
@@ -540,6 +725,16 @@ static int run_scan(const YtPlan* plan, const YtChunk* chunk,
 {
     int rc = YT_OK;
     hipEvent_t ev0 = nullptr, ev1 = nullptr;
+
+    if (fs->valid && fs->key_col >= 0 && fs->nsum <= 1) {
+        bool done = false;
+        rc = run_partitioned(plan, chunk, options, R, fs, maxw, stats, &done,
+                             errbuf, errlen);
+        if (rc != YT_OK) return rc;
+        if (done) return YT_OK;
+        /* else: capacity guard tripped — fall through to the direct path */
+    }
+
     HIP_CHECK(hipEventCreate(&ev0));
     HIP_CHECK(hipEventCreate(&ev1));
 
@@ -765,7 +960,7 @@ static int emit_rows(const YtPlan* plan, const YtChunk* chunk,
         rc = emit(g.key_bits, (int)(g.key_meta & 1), g.cnt, g.agg_bits, g.agg_nonnull);
         if (rc) return rc;
     }
-    /* side groups: key bits == 0, then null key */
+    /* side groups: the in-table sentinel key, then the null key */
     for (int side = 0; side < 2; side++) {
         if (!th.side_used[side]) continue;
         uint64_t ab[kMaxAggs], an[kMaxAggs];
@@ -773,7 +968,7 @@ static int emit_rows(const YtPlan* plan, const YtChunk* chunk,
             ab[a] = th.side_agg[side][2 * a];
             an[a] = th.side_agg[side][2 * a + 1];
         }
-        rc = emit(0, side == 1, th.side_cnt[side], ab, an);
+        rc = emit(th.side_key_bits[side], side == 1, th.side_cnt[side], ab, an);
         if (rc) return rc;
     }
     (void)errbuf; (void)errlen;
@@ -842,7 +1037,12 @@ extern "C" int yt_gpu_query_execute(
         HIP_CHECK(hipMemcpy(&th, R.d_th, sizeof(th), hipMemcpyDeviceToHost));
         if (th.overflow == 1) { set_err(errbuf, errlen, "group table overflow — raise max_groups_hint"); rc = YT_ERR_CAPACITY; goto fail; }
         int64_t ngroups = (int64_t)th.ngroups;
-        if (ngroups > 0) {
+        if (ngroups > 0 && R.groups_compacted) {
+            groups.resize(ngroups);
+            HIP_CHECK(hipMemcpyAsync(groups.data(), R.d_groups, sizeof(OutGroup) * ngroups,
+                                     hipMemcpyDeviceToHost, R.stream));
+            HIP_CHECK(hipStreamSynchronize(R.stream));
+        } else if (ngroups > 0) {
             HIP_CHECK(hipMalloc(&R.d_groups, sizeof(OutGroup) * ngroups));
             HIP_CHECK(hipMalloc(&R.d_counter, sizeof(unsigned long long)));
             HIP_CHECK(hipMemsetAsync(R.d_counter, 0, sizeof(unsigned long long), R.stream));
@@ -935,13 +1135,19 @@ extern "C" int yt_gpu_query_partial(
         int64_t total = ngroups + th.side_used[0] + th.side_used[1];
         if (total > capacity_rows) { set_err(errbuf, errlen, "state buffer too small"); rc = YT_ERR_CAPACITY; goto fail; }
 
-        /* compact table into OutGroups (device) */
-        HIP_CHECK(hipMalloc(&R.d_groups, sizeof(OutGroup) * (total ? total : 1)));
-        HIP_CHECK(hipMalloc(&R.d_counter, sizeof(unsigned long long)));
-        HIP_CHECK(hipMemsetAsync(R.d_counter, 0, sizeof(unsigned long long), R.stream));
-        if (ngroups > 0) {
-            HIP_CHECK(ytql_launch_compact(nullptr, R.d_th, R.d_slots, plan->agg_count,
-                                          R.d_groups, R.d_counter, R.nslots, R.stream));
+        if (!R.groups_compacted) {
+            /* compact table into OutGroups (device) */
+            HIP_CHECK(hipMalloc(&R.d_groups, sizeof(OutGroup) * (total ? total : 1)));
+            HIP_CHECK(hipMalloc(&R.d_counter, sizeof(unsigned long long)));
+            HIP_CHECK(hipMemsetAsync(R.d_counter, 0, sizeof(unsigned long long), R.stream));
+            if (ngroups > 0) {
+                HIP_CHECK(ytql_launch_compact(nullptr, R.d_th, R.d_slots, plan->agg_count,
+                                              R.d_groups, R.d_counter, R.nslots, R.stream));
+            }
+        } else if (R.groups_capacity < total) {
+            set_err(errbuf, errlen, "state buffer too small");
+            rc = YT_ERR_CAPACITY;
+            goto fail;
         }
         /* append side groups from host */
         int64_t pos = ngroups;
@@ -949,7 +1155,7 @@ extern "C" int yt_gpu_query_partial(
             if (!th.side_used[side]) continue;
             OutGroup g;
             memset(&g, 0, sizeof(g));
-            g.key_bits = 0;
+            g.key_bits = th.side_key_bits[side];
             g.key_meta = (side == 1);
             g.cnt = th.side_cnt[side];
             for (int a = 0; a < plan->agg_count; a++) {

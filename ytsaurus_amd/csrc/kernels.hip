@@ -100,7 +100,8 @@ __device__ __forceinline__ uint64_t bp_get_win(const uint64_t* win, uint32_t wid
 /* segment header parsing (one thread per segment)                     */
 
 __global__ void k_parse_segments(const DevSeg* segs, int nsegs, SegEx* out,
-                                 unsigned* max_width_out /* per column? single */)
+                                 unsigned* max_width_out,
+                                 unsigned* col_null_flags /* kMaxCols words */)
 {
     int i = blockIdx.x * blockDim.x + threadIdx.x;
     if (i >= nsegs) return;
@@ -163,6 +164,22 @@ __global__ void k_parse_segments(const DevSeg* segs, int nsegs, SegEx* out,
             break;
         }
         }
+    }
+    /* value-null presence (DirectDense): OR over the null bitmap */
+    if (s.type == YT_SEG_DIRECT_DENSE) {
+        const uint64_t* bm = (const uint64_t*)((const uint8_t*)b + e.off_bitmap_bytes);
+        int64_t words = (s.row_count + 63) / 64;
+        uint64_t any = 0;
+        /* last partial byte-word is zero-padded by the writer (bitmap.h) */
+        for (int64_t wI = 0; wI < words; wI++) any |= bm[wI];
+        /* mask tail bits beyond row_count (padding is zero, but be safe) */
+        if (any) {
+            e.flags |= 1;
+            atomicOr(&col_null_flags[s.col], 1u);
+        }
+    } else if (s.type != YT_SEG_DOUBLE) {
+        /* other formats: unknown without a scan — be conservative */
+        atomicOr(&col_null_flags[s.col], 2u);
     }
     out[i] = e;
     atomicMax(max_width_out, e.w_values);
@@ -414,6 +431,12 @@ __device__ unsigned long long* table_probe(TableHdr* th, unsigned long long* slo
     uint64_t s = h & mask;
     for (uint64_t iter = 0; iter <= mask; iter++) {
         unsigned long long* slot = slots + s * stride;
+        /* relaxed agent-scope load first: on the hot path the group exists,
+         * and a read is far cheaper than an RMW (probe measurement V5) */
+        unsigned long long cur = __hip_atomic_load(slot, __ATOMIC_RELAXED,
+                                                   __HIP_MEMORY_SCOPE_AGENT);
+        if (cur == (unsigned long long)key_bits) return slot;
+        if (cur != 0ULL) { s = (s + 1) & mask; continue; }
         unsigned long long old = atomicCAS(slot, 0ULL, (unsigned long long)key_bits);
         if (old == 0ULL) {
             unsigned long long ticket = atomicAdd(&th->ngroups, 1ULL);
@@ -688,6 +711,267 @@ k_scan_fast(FastParams fp, const DevSeg* segs, const SegEx* segex,
 }
 
 /* ------------------------------------------------------------------ */
+/* two-phase partitioned group-by (see common.h PartParams)            */
+
+__global__ void __launch_bounds__(256)
+k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
+                 const FastCol* cols, TableHdr* th,
+                 unsigned long long* cursors,        /* kNB main */
+                 ulonglong2* recs,
+                 unsigned long long* ncursors,       /* kNB null-value stream */
+                 uint64_t* nrecs)
+{
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    const int tid = threadIdx.x;
+
+    /* LDS layout: hist[kNB] u32 | gbase[kNB] u32 | nhist[kNB] u32 |
+     *             ngbase[kNB] u32 | column staging */
+    unsigned* hist = (unsigned*)smem;
+    unsigned* gbase = hist + kNB;
+    unsigned* nhist = gbase + kNB;
+    unsigned* ngbase = nhist + kNB;
+    char* stage_base = (char*)(ngbase + kNB);
+
+    for (int tile = blockIdx.x; tile < pp.ntiles; tile += gridDim.x) {
+        const int seg_idx = tile / pp.tiles_per_seg;
+        const int tile_in_seg = tile % pp.tiles_per_seg;
+        const int64_t t0 = (int64_t)tile_in_seg * pp.tile_rows;
+
+        int64_t lds_off = 0;
+        uint64_t* col_words[4];
+        const uint8_t* col_bitmap[4];
+        uint32_t col_w[4];
+        int64_t col_w0[4];
+        int32_t seg_rows = 0;
+        for (int u = 0; u < pp.nused; u++) {
+            const DevSeg& sg = segs[cols[u].seg_off + seg_idx];
+            const SegEx& e = segex[cols[u].seg_off + seg_idx];
+            seg_rows = sg.row_count;
+            int64_t t1 = t0 + pp.tile_rows;
+            if (t1 > seg_rows) t1 = seg_rows;
+            uint32_t w = e.w_values;
+            col_w[u] = w;
+            int64_t W0 = ((uint64_t)t0 * w) >> 6;
+            int64_t W1 = w ? ((((uint64_t)t1 * w) + 63) >> 6) : 0;
+            int64_t nwords = (w == 0) ? 0 : (W1 - W0 + 1);
+            col_w0[u] = W0;
+            uint64_t* dst = (uint64_t*)(stage_base + lds_off);
+            col_words[u] = dst;
+            const uint64_t* src = sg.blob + e.off_values_words + W0;
+            int64_t vec_words = (w == 0) ? 0 : (((uint64_t)seg_rows * w + 63) >> 6);
+            int64_t avail = vec_words - W0;
+            if (nwords > avail) nwords = avail;
+            for (int64_t i = tid; i < nwords; i += 256) dst[i] = src[i];
+            lds_off += ((nwords * 8) + 15) & ~(int64_t)15;
+
+            uint8_t* bdst = (uint8_t*)(stage_base + lds_off);
+            col_bitmap[u] = bdst;
+            const uint8_t* bsrc = (const uint8_t*)sg.blob + e.off_bitmap_bytes + (t0 >> 3);
+            int64_t bbytes = ((t1 - t0) + 7) / 8;
+            for (int64_t i = tid; i < bbytes; i += 256) bdst[i] = bsrc[i];
+            lds_off += (bbytes + 15) & ~(int64_t)15;
+        }
+        for (int i = tid; i < kNB; i += 256) { hist[i] = 0; nhist[i] = 0; }
+        __syncthreads();
+
+        const DevSeg& sk = segs[cols[pp.key_idx].seg_off + seg_idx];
+        const DevSeg* sv = pp.val_idx >= 0 ? &segs[cols[pp.val_idx].seg_off + seg_idx] : nullptr;
+        int64_t t1 = t0 + pp.tile_rows;
+        if (t1 > seg_rows) t1 = seg_rows;
+
+        const int R = (pp.tile_rows + 255) / 256;
+        uint32_t row_b[32];
+        uint32_t row_off[32];
+        for (int i = 0; i < R; i++) {
+            row_b[i] = 0xFFFFFFFFu;
+            int64_t j = t0 + (int64_t)i * 256 + tid;
+            if (j >= t1) continue;
+
+            if (pp.filter_idx >= 0) {
+                int u = pp.filter_idx;
+                if (bm_get(col_bitmap[u], j - t0)) continue;
+                uint64_t packed = bp_get_win(col_words[u], col_w[u], j, col_w0[u]);
+                const DevSeg& sf = segs[cols[u].seg_off + seg_idx];
+                int64_t v = zz_dec(sf.min_value + packed);
+                if (v < pp.filter_lo || v > pp.filter_hi) continue;
+            }
+
+            uint64_t key;
+            int key_null = pp.has_key_nulls && bm_get(col_bitmap[pp.key_idx], j - t0);
+            if (!key_null) {
+                uint64_t packed = bp_get_win(col_words[pp.key_idx], col_w[pp.key_idx], j,
+                                             col_w0[pp.key_idx]);
+                key = (uint64_t)zz_dec(sk.min_value + packed);
+            }
+            int val_null = 1;
+            uint64_t val = 0;
+            if (pp.val_idx >= 0) {
+                val_null = pp.has_val_nulls && bm_get(col_bitmap[pp.val_idx], j - t0);
+                if (!val_null) {
+                    uint64_t packed = bp_get_win(col_words[pp.val_idx], col_w[pp.val_idx], j,
+                                                 col_w0[pp.val_idx]);
+                    val = (uint64_t)zz_dec(sv->min_value + packed);
+                }
+            }
+
+            if (key_null || (!key_null && key == kEmptyKey)) {
+                /* side groups: null key, or the LDS-table sentinel key
+                 * (INT64_MIN) — rare; direct device-scope atomics */
+                int side = key_null ? 1 : 0;
+                th->side_used[side] = 1;
+                atomicAdd((unsigned long long*)&th->side_cnt[side], 1ULL);
+                if (pp.sum_slot >= 0 && !val_null) {
+                    atomicAdd((unsigned long long*)&th->side_agg[side][2 * pp.sum_slot], val);
+                    atomicAdd((unsigned long long*)&th->side_agg[side][2 * pp.sum_slot + 1], 1ULL);
+                }
+                continue;
+            }
+
+            unsigned b = (unsigned)(mix64(key) >> 40) & (kNB - 1);
+            if (pp.val_idx >= 0 && val_null) {
+                row_b[i] = b | 0x80000000u;
+                row_off[i] = atomicAdd(&nhist[b], 1u);
+            } else {
+                row_b[i] = b;
+                row_off[i] = atomicAdd(&hist[b], 1u);
+            }
+        }
+        __syncthreads();
+        for (int i = tid; i < kNB; i += 256) {
+            unsigned c = hist[i];
+            if (c) {
+                unsigned long long base = atomicAdd(&cursors[i], (unsigned long long)c);
+                if ((int64_t)(base + c) > pp.bucket_stride) { th->overflow = 1; base = 0; }
+                gbase[i] = (unsigned)base;
+            }
+            unsigned nc = nhist[i];
+            if (nc) {
+                unsigned long long base = atomicAdd(&ncursors[i], (unsigned long long)nc);
+                if ((int64_t)(base + nc) > pp.nbucket_stride) { th->overflow = 1; base = 0; }
+                ngbase[i] = (unsigned)base;
+            }
+        }
+        __syncthreads();
+        if (th->overflow != 1) {
+            for (int i = 0; i < R; i++) {
+                if (row_b[i] == 0xFFFFFFFFu) continue;
+                int64_t j = t0 + (int64_t)i * 256 + tid;
+                uint64_t packed = bp_get_win(col_words[pp.key_idx], col_w[pp.key_idx], j,
+                                             col_w0[pp.key_idx]);
+                uint64_t key = (uint64_t)zz_dec(sk.min_value + packed);
+                unsigned b = row_b[i] & 0x7FFFFFFFu;
+                if (row_b[i] & 0x80000000u) {
+                    nrecs[(int64_t)b * pp.nbucket_stride + ngbase[b] + row_off[i]] = key;
+                } else {
+                    uint64_t val = 0;
+                    if (pp.val_idx >= 0) {
+                        uint64_t pv = bp_get_win(col_words[pp.val_idx], col_w[pp.val_idx], j,
+                                                 col_w0[pp.val_idx]);
+                        val = (uint64_t)zz_dec(sv->min_value + pv);
+                    }
+                    recs[(int64_t)b * pp.bucket_stride + gbase[b] + row_off[i]] =
+                        make_ulonglong2(key, val);
+                }
+            }
+        }
+        __syncthreads();
+    }
+}
+
+/* Phase B: one workgroup per bucket. LDS table slot = {key, cnt|nonnull<<32
+ * packed, sum}; per-workgroup counts stay < 2^32 because a bucket's rows do.
+ * Emits compacted OutGroups directly (buckets are key-disjoint). */
+__global__ void __launch_bounds__(256)
+k_bucket_agg(const ulonglong2* recs, const unsigned long long* cursors,
+             int64_t bucket_stride,
+             const uint64_t* nrecs, const unsigned long long* ncursors,
+             int64_t nbucket_stride,
+             OutGroup* out, unsigned long long* out_counter, int64_t out_cap,
+             TableHdr* th, int sum_slot, int agg_count)
+{
+    __shared__ unsigned long long tab[kHSlots * 3];
+    const int tid = threadIdx.x;
+    const int bucket = blockIdx.x;
+
+    for (int i = tid; i < kHSlots; i += 256) {
+        tab[i * 3] = kEmptyKey;
+        tab[i * 3 + 1] = 0;
+        tab[i * 3 + 2] = 0;
+    }
+    __syncthreads();
+
+    int64_t n = (int64_t)cursors[bucket];
+    const ulonglong2* rows = recs + (int64_t)bucket * bucket_stride;
+    bool full = false;
+    for (int64_t i = tid; i < n; i += 256) {
+        ulonglong2 kv = rows[i];
+        uint64_t s = mix64(kv.x) & (kHSlots - 1);
+        int found = 0;
+        for (int it = 0; it < kHSlots; it++) {
+            unsigned long long k = tab[s * 3];
+            if (k == (unsigned long long)kv.x) { found = 1; break; }
+            if (k == (unsigned long long)kEmptyKey) {
+                k = atomicCAS(&tab[s * 3], (unsigned long long)kEmptyKey,
+                              (unsigned long long)kv.x);
+                if (k == (unsigned long long)kEmptyKey ||
+                    k == (unsigned long long)kv.x) { found = 1; break; }
+            }
+            s = (s + 1) & (kHSlots - 1);
+        }
+        if (!found) { full = true; break; }
+        atomicAdd(&tab[s * 3 + 1], 1ULL | (1ULL << 32));
+        if (sum_slot >= 0) atomicAdd(&tab[s * 3 + 2], (unsigned long long)kv.y);
+    }
+    if (nrecs) {
+        int64_t nn = (int64_t)ncursors[bucket];
+        const uint64_t* nrows = nrecs + (int64_t)bucket * nbucket_stride;
+        for (int64_t i = tid; i < nn && !full; i += 256) {
+            uint64_t key = nrows[i];
+            uint64_t s = mix64(key) & (kHSlots - 1);
+            int found = 0;
+            for (int it = 0; it < kHSlots; it++) {
+                unsigned long long k = tab[s * 3];
+                if (k == (unsigned long long)key) { found = 1; break; }
+                if (k == (unsigned long long)kEmptyKey) {
+                    k = atomicCAS(&tab[s * 3], (unsigned long long)kEmptyKey,
+                                  (unsigned long long)key);
+                    if (k == (unsigned long long)kEmptyKey ||
+                        k == (unsigned long long)key) { found = 1; break; }
+                }
+                s = (s + 1) & (kHSlots - 1);
+            }
+            if (!found) { full = true; break; }
+            atomicAdd(&tab[s * 3 + 1], 1ULL);   /* cnt only; sum stays null-contributing */
+        }
+    }
+    if (full) th->overflow = 1;
+    __syncthreads();
+
+    /* flush compacted groups */
+    for (int i = tid; i < kHSlots; i += 256) {
+        unsigned long long key = tab[i * 3];
+        if (key == (unsigned long long)kEmptyKey) continue;
+        unsigned long long idx = atomicAdd(out_counter, 1ULL);
+        unsigned long long t = atomicAdd(&th->ngroups, 1ULL);
+        if (th->group_limit > 0 && (int64_t)t >= th->group_limit) th->overflow = 2;
+        if ((int64_t)idx >= out_cap) { th->overflow = 1; continue; }
+        OutGroup& g = out[idx];
+        g.key_bits = key;
+        g.key_meta = 0;
+        uint64_t cntnn = tab[i * 3 + 1];
+        g.cnt = cntnn & 0xFFFFFFFFULL;
+        for (int a = 0; a < agg_count; a++) {
+            g.agg_bits[a] = 0;
+            g.agg_nonnull[a] = 0;
+        }
+        if (sum_slot >= 0) {
+            g.agg_bits[sum_slot] = tab[i * 3 + 2];
+            g.agg_nonnull[sum_slot] = cntnn >> 32;
+        }
+    }
+}
+
+/* ------------------------------------------------------------------ */
 /* table compaction / partition / merge                                */
 
 __global__ void k_compact(TableHdr* th, const unsigned long long* slots,
@@ -787,12 +1071,42 @@ using namespace ytql;
 extern "C" {
 
 hipError_t ytql_launch_parse_segments(const DevSeg* segs, int nsegs, SegEx* out,
-                                      unsigned* max_width, hipStream_t st)
+                                      unsigned* max_width, unsigned* col_null_flags,
+                                      hipStream_t st)
 {
     int block = 256;
     int grid = (nsegs + block - 1) / block;
     hipLaunchKernelGGL(k_parse_segments, dim3(grid), dim3(block), 0, st,
-                       segs, nsegs, out, max_width);
+                       segs, nsegs, out, max_width, col_null_flags);
+    return hipGetLastError();
+}
+
+hipError_t ytql_launch_scan_partition(const PartParams* pp, const DevSeg* segs,
+                                      const SegEx* segex, const FastCol* cols,
+                                      TableHdr* th,
+                                      unsigned long long* cursors, void* recs,
+                                      unsigned long long* ncursors, uint64_t* nrecs,
+                                      size_t lds_bytes, int grid, hipStream_t st)
+{
+    hipLaunchKernelGGL(k_scan_partition, dim3(grid), dim3(256), lds_bytes, st,
+                       *pp, segs, segex, cols, th, cursors, (ulonglong2*)recs,
+                       ncursors, nrecs);
+    return hipGetLastError();
+}
+
+hipError_t ytql_launch_bucket_agg(const void* recs, const unsigned long long* cursors,
+                                  int64_t bucket_stride,
+                                  const uint64_t* nrecs, const unsigned long long* ncursors,
+                                  int64_t nbucket_stride,
+                                  OutGroup* out, unsigned long long* out_counter,
+                                  int64_t out_cap,
+                                  TableHdr* th, int sum_slot, int agg_count,
+                                  hipStream_t st)
+{
+    hipLaunchKernelGGL(k_bucket_agg, dim3(kNB), dim3(256), 0, st,
+                       (const ulonglong2*)recs, cursors, bucket_stride,
+                       nrecs, ncursors, nbucket_stride,
+                       out, out_counter, out_cap, th, sum_slot, agg_count);
     return hipGetLastError();
 }
 
