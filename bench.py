@@ -231,6 +231,7 @@ def main():
                                   rowset=out_rs, raw_rowset=True)
             scan_ms_total += st.kernel_scan_ms
             scan_launches += st.kernel_scan_launches
+            step.last = st
             return st
         # bottom query: partial aggregate + hash partition on device
         counts, st = y.gpu_partial(plan, dev_chunk, world,
@@ -278,6 +279,12 @@ def main():
         t = torch.tensor([elapsed], dtype=torch.float64, device="cuda")
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t.item())
+
+    st = getattr(step, "last", None)
+    if st is not None:
+        log("last step: exec %.1fms scanK %.2fms otherK %.2fms rows_written %d"
+            % (st.execute_time_ms, st.kernel_scan_ms / max(st.kernel_scan_launches, 1),
+               st.kernel_other_ms, st.rows_written))
 
     ms_per_step = elapsed / args.steps * 1000
     total_rows = n * world

@@ -15,6 +15,8 @@
 #include <stdlib.h>
 #include <vector>
 #include <algorithm>
+#include <mutex>
+#include <chrono>
 
 #include "common.h"
 #include "../../include/ytql_gpu.h"
@@ -49,6 +51,105 @@ hipError_t ytql_launch_merge_states(const YtStateRow*, int64_t, int, int,
 }
 
 /* ------------------------------------------------------------------ */
+/* device/pinned buffer pool: query executions reuse large allocations
+ * across calls (the reference's evaluator similarly reuses codegen and
+ * memory-chunk pools per query class, evaluator.cpp:256, TExpressionContext
+ * chunk providers). Freed via yt_gpu_pool_trim(). */
+
+namespace {
+
+struct Pool {
+    struct Blk { void* p; size_t sz; bool used; bool host; };
+    std::mutex m;
+    std::vector<Blk> blks;
+
+    hipError_t get(size_t sz, bool host, void** out)
+    {
+        std::lock_guard<std::mutex> g(m);
+        Blk* best = nullptr;
+        for (auto& b : blks) {
+            if (!b.used && b.host == host && b.sz >= sz &&
+                (!best || b.sz < best->sz)) {
+                best = &b;
+            }
+        }
+        if (best && best->sz <= sz * 2 + (64 << 20)) {
+            best->used = true;
+            *out = best->p;
+            return hipSuccess;
+        }
+        void* p = nullptr;
+        hipError_t e = host ? hipHostMalloc(&p, sz) : hipMalloc(&p, sz);
+        if (e != hipSuccess) {
+            /* retry after trimming the pool */
+            for (auto& b : blks) {
+                if (!b.used) {
+                    if (b.host) { (void)hipHostFree(b.p); } else { (void)hipFree(b.p); }
+                    b.p = nullptr;
+                    b.sz = 0;
+                }
+            }
+            blks.erase(std::remove_if(blks.begin(), blks.end(),
+                                      [](const Blk& b) { return !b.p; }),
+                       blks.end());
+            e = host ? hipHostMalloc(&p, sz) : hipMalloc(&p, sz);
+            if (e != hipSuccess) return e;
+        }
+        blks.push_back({p, sz, true, host});
+        *out = p;
+        return hipSuccess;
+    }
+
+    void put(void* p)
+    {
+        if (!p) return;
+        std::lock_guard<std::mutex> g(m);
+        for (auto& b : blks) {
+            if (b.p == p) { b.used = false; return; }
+        }
+    }
+
+    void trim()
+    {
+        std::lock_guard<std::mutex> g(m);
+        for (auto& b : blks) {
+            if (!b.used) {
+                if (b.host) { (void)hipHostFree(b.p); } else { (void)hipFree(b.p); }
+                b.p = nullptr;
+            }
+        }
+        blks.erase(std::remove_if(blks.begin(), blks.end(),
+                                  [](const Blk& b) { return !b.p; }),
+                   blks.end());
+    }
+};
+
+Pool g_pool;
+
+template <class T>
+hipError_t pool_alloc(T** p, size_t sz)
+{
+    return g_pool.get(sz, false, (void**)p);
+}
+
+template <class T>
+hipError_t pool_alloc_host(T** p, size_t sz)
+{
+    return g_pool.get(sz, true, (void**)p);
+}
+
+double now_ms()
+{
+    return std::chrono::duration<double, std::milli>(
+        std::chrono::steady_clock::now().time_since_epoch()).count();
+}
+
+} /* namespace */
+
+extern "C" void yt_gpu_pool_trim(void)
+{
+    g_pool.trim();
+}
 
 static void set_err(char* errbuf, size_t errlen, const char* msg)
 {
@@ -444,25 +545,23 @@ struct DeviceRun {
 
     ~DeviceRun()
     {
-        hipError_t e;
-        (void)e;
-        if (d_segs) e = hipFree(d_segs);
-        if (d_segex) e = hipFree(d_segex);
-        if (d_off) e = hipFree(d_off);
-        if (d_cnt) e = hipFree(d_cnt);
-        if (d_maxw) e = hipFree(d_maxw);
-        if (d_err) e = hipFree(d_err);
-        if (d_th) e = hipFree(d_th);
-        if (d_slots) e = hipFree(d_slots);
-        if (d_groups) e = hipFree(d_groups);
-        if (d_counter) e = hipFree(d_counter);
-        if (d_gaccum) e = hipFree(d_gaccum);
-        if (d_fastcols) e = hipFree(d_fastcols);
-        if (d_colnull) e = hipFree(d_colnull);
-        if (d_cursors) e = hipFree(d_cursors);
-        if (d_recs) e = hipFree(d_recs);
-        if (d_ncursors) e = hipFree(d_ncursors);
-        if (d_nrecs) e = hipFree(d_nrecs);
+        g_pool.put(d_segs);
+        g_pool.put(d_segex);
+        g_pool.put(d_off);
+        g_pool.put(d_cnt);
+        g_pool.put(d_maxw);
+        g_pool.put(d_err);
+        g_pool.put(d_th);
+        g_pool.put(d_slots);
+        g_pool.put(d_groups);
+        g_pool.put(d_counter);
+        g_pool.put(d_gaccum);
+        g_pool.put(d_fastcols);
+        g_pool.put(d_colnull);
+        g_pool.put(d_cursors);
+        g_pool.put(d_recs);
+        g_pool.put(d_ncursors);
+        g_pool.put(d_nrecs);
     }
 };
 
@@ -508,13 +607,13 @@ static int setup_chunk(const YtChunk* chunk, DeviceRun* R, unsigned* maxw_out,
     R->nsegs = (int)R->h_segs.size();
     if (R->nsegs == 0) return YT_OK;
 
-    HIP_CHECK(hipMalloc(&R->d_segs, sizeof(DevSeg) * R->nsegs));
-    HIP_CHECK(hipMalloc(&R->d_segex, sizeof(SegEx) * R->nsegs));
-    HIP_CHECK(hipMalloc(&R->d_off, sizeof(int32_t) * ncols));
-    HIP_CHECK(hipMalloc(&R->d_cnt, sizeof(int32_t) * ncols));
-    HIP_CHECK(hipMalloc(&R->d_maxw, sizeof(unsigned)));
-    HIP_CHECK(hipMalloc(&R->d_err, sizeof(unsigned)));
-    HIP_CHECK(hipMalloc(&R->d_colnull, sizeof(unsigned) * kMaxCols));
+    HIP_CHECK(pool_alloc(&R->d_segs, sizeof(DevSeg) * R->nsegs));
+    HIP_CHECK(pool_alloc(&R->d_segex, sizeof(SegEx) * R->nsegs));
+    HIP_CHECK(pool_alloc(&R->d_off, sizeof(int32_t) * ncols));
+    HIP_CHECK(pool_alloc(&R->d_cnt, sizeof(int32_t) * ncols));
+    HIP_CHECK(pool_alloc(&R->d_maxw, sizeof(unsigned)));
+    HIP_CHECK(pool_alloc(&R->d_err, sizeof(unsigned)));
+    HIP_CHECK(pool_alloc(&R->d_colnull, sizeof(unsigned) * kMaxCols));
     HIP_CHECK(hipMemsetAsync(R->d_colnull, 0, sizeof(unsigned) * kMaxCols, R->stream));
     HIP_CHECK(hipMemcpyAsync(R->d_segs, R->h_segs.data(), sizeof(DevSeg) * R->nsegs,
                              hipMemcpyHostToDevice, R->stream));
@@ -544,8 +643,8 @@ static int setup_table(DeviceRun* R, int agg_count, int64_t max_groups,
     R->nslots = next_pow2((uint64_t)(max_groups > 0 ? max_groups : (1 << 20)) * 2);
     if (R->nslots < 2048) R->nslots = 2048;
     int stride = 2 + 2 * agg_count;
-    HIP_CHECK(hipMalloc(&R->d_th, sizeof(TableHdr)));
-    HIP_CHECK(hipMalloc(&R->d_slots, sizeof(uint64_t) * R->nslots * stride));
+    HIP_CHECK(pool_alloc(&R->d_th, sizeof(TableHdr)));
+    HIP_CHECK(pool_alloc(&R->d_slots, sizeof(uint64_t) * R->nslots * stride));
     TableHdr hh;
     memset(&hh, 0, sizeof(hh));
     hh.nslots = R->nslots;
@@ -626,16 +725,16 @@ static int run_partitioned(const YtPlan* plan, const YtChunk* chunk,
     }
 
     {
-        HIP_CHECK(hipMalloc(&R->d_fastcols, sizeof(FastCol) * nused));
+        HIP_CHECK(pool_alloc(&R->d_fastcols, sizeof(FastCol) * nused));
         HIP_CHECK(hipMemcpyAsync(R->d_fastcols, fc.data(), sizeof(FastCol) * nused,
                                  hipMemcpyHostToDevice, R->stream));
-        HIP_CHECK(hipMalloc(&R->d_cursors, sizeof(uint64_t) * kNB));
+        HIP_CHECK(pool_alloc(&R->d_cursors, sizeof(uint64_t) * kNB));
         HIP_CHECK(hipMemsetAsync(R->d_cursors, 0, sizeof(uint64_t) * kNB, R->stream));
-        HIP_CHECK(hipMalloc(&R->d_recs, (size_t)kNB * pp.bucket_stride * 16));
+        HIP_CHECK(pool_alloc(&R->d_recs, (size_t)kNB * pp.bucket_stride * 16));
         if (pp.has_val_nulls) {
-            HIP_CHECK(hipMalloc(&R->d_ncursors, sizeof(uint64_t) * kNB));
+            HIP_CHECK(pool_alloc(&R->d_ncursors, sizeof(uint64_t) * kNB));
             HIP_CHECK(hipMemsetAsync(R->d_ncursors, 0, sizeof(uint64_t) * kNB, R->stream));
-            HIP_CHECK(hipMalloc(&R->d_nrecs, (size_t)kNB * pp.nbucket_stride * 8));
+            HIP_CHECK(pool_alloc(&R->d_nrecs, (size_t)kNB * pp.nbucket_stride * 8));
         }
         /* the partitioned path's in-table sentinel is INT64_MIN bits */
         uint64_t sk = kEmptyKey;
@@ -649,8 +748,8 @@ static int run_partitioned(const YtPlan* plan, const YtChunk* chunk,
         }
         if (cap_groups > rows + 16) cap_groups = rows + 16;
         R->groups_capacity = cap_groups;
-        HIP_CHECK(hipMalloc(&R->d_groups, sizeof(OutGroup) * cap_groups));
-        HIP_CHECK(hipMalloc(&R->d_counter, sizeof(unsigned long long)));
+        HIP_CHECK(pool_alloc(&R->d_groups, sizeof(OutGroup) * cap_groups));
+        HIP_CHECK(pool_alloc(&R->d_counter, sizeof(unsigned long long)));
         HIP_CHECK(hipMemsetAsync(R->d_counter, 0, sizeof(unsigned long long), R->stream));
 
         HIP_CHECK(hipEventCreate(&ev0));
@@ -683,13 +782,13 @@ static int run_partitioned(const YtPlan* plan, const YtChunk* chunk,
         HIP_CHECK(hipMemcpy(&th, R->d_th, sizeof(th), hipMemcpyDeviceToHost));
         if (th.overflow == 1) {
             /* reset and fall back */
-            hipFree(R->d_groups); R->d_groups = nullptr;
-            hipFree(R->d_counter); R->d_counter = nullptr;
-            hipFree(R->d_recs); R->d_recs = nullptr;
-            hipFree(R->d_cursors); R->d_cursors = nullptr;
-            if (R->d_nrecs) { hipFree(R->d_nrecs); R->d_nrecs = nullptr; }
-            if (R->d_ncursors) { hipFree(R->d_ncursors); R->d_ncursors = nullptr; }
-            hipFree(R->d_fastcols); R->d_fastcols = nullptr;
+            g_pool.put(R->d_groups); R->d_groups = nullptr;
+            g_pool.put(R->d_counter); R->d_counter = nullptr;
+            g_pool.put(R->d_recs); R->d_recs = nullptr;
+            g_pool.put(R->d_cursors); R->d_cursors = nullptr;
+            g_pool.put(R->d_nrecs); R->d_nrecs = nullptr;
+            g_pool.put(R->d_ncursors); R->d_ncursors = nullptr;
+            g_pool.put(R->d_fastcols); R->d_fastcols = nullptr;
             TableHdr hh;
             memset(&hh, 0, sizeof(hh));
             hh.nslots = R->nslots;
@@ -738,7 +837,7 @@ static int run_scan(const YtPlan* plan, const YtChunk* chunk,
     HIP_CHECK(hipEventCreate(&ev0));
     HIP_CHECK(hipEventCreate(&ev1));
 
-    HIP_CHECK(hipMalloc(&R->d_gaccum, sizeof(uint64_t) * (1 + 2 * kMaxAggs)));
+    HIP_CHECK(pool_alloc(&R->d_gaccum, sizeof(uint64_t) * (1 + 2 * kMaxAggs)));
     HIP_CHECK(hipMemsetAsync(R->d_gaccum, 0, sizeof(uint64_t) * (1 + 2 * kMaxAggs), R->stream));
 
     if (fs->valid) {
@@ -798,7 +897,7 @@ static int run_scan(const YtPlan* plan, const YtChunk* chunk,
             fc[u].seg_off = R->h_off[used[u]];
             fc[u].seg_cnt = R->h_cnt[used[u]];
         }
-        HIP_CHECK(hipMalloc(&R->d_fastcols, sizeof(FastCol) * nused));
+        HIP_CHECK(pool_alloc(&R->d_fastcols, sizeof(FastCol) * nused));
         HIP_CHECK(hipMemcpyAsync(R->d_fastcols, fc.data(), sizeof(FastCol) * nused,
                                  hipMemcpyHostToDevice, R->stream));
 
@@ -873,7 +972,7 @@ static void finalize_row(const YtPlan* plan, uint8_t key_type,
 }
 
 static int emit_rows(const YtPlan* plan, const YtChunk* chunk,
-                     const std::vector<OutGroup>& groups,
+                     const OutGroup* groups, int64_t ngroups,
                      const TableHdr& th, int has_any_row_global,
                      const uint64_t* gaccum, int used_fast_global,
                      YtRowset* output, char* errbuf, size_t errlen)
@@ -945,7 +1044,7 @@ static int emit_rows(const YtPlan* plan, const YtChunk* chunk,
             if (cnt == 0) return YT_OK;
         } else {
             /* generic path routed rows to synthetic key bits=1 */
-            if (groups.empty()) return YT_OK;
+            if (ngroups == 0) return YT_OK;
             cnt = groups[0].cnt;
             for (int a = 0; a < plan->agg_count; a++) {
                 ab[a] = groups[0].agg_bits[a];
@@ -956,7 +1055,8 @@ static int emit_rows(const YtPlan* plan, const YtChunk* chunk,
         return emit(0, 0, cnt, ab, an);
     }
 
-    for (const OutGroup& g : groups) {
+    for (int64_t i = 0; i < ngroups; i++) {
+        const OutGroup& g = groups[i];
         rc = emit(g.key_bits, (int)(g.key_meta & 1), g.cnt, g.agg_bits, g.agg_nonnull);
         if (rc) return rc;
     }
@@ -986,6 +1086,7 @@ extern "C" int yt_gpu_query_execute(
     if (rc != YT_OK) return rc;
     if (!plan || !chunk || !output) { set_err(errbuf, errlen, "null argument"); return YT_ERR_INVALID_PLAN; }
 
+    double tw0 = now_ms();
     YtExecOptions defopt;
     memset(&defopt, 0, sizeof(defopt));
     if (!options) options = &defopt;
@@ -1030,26 +1131,24 @@ extern "C" int yt_gpu_query_execute(
     rc = run_scan(plan, chunk, options, &R, &dp, &fs, maxw, stats, errbuf, errlen);
     if (rc) return rc;
 
-    /* compact + readback */
+    /* compact + readback (pinned staging from the pool) */
     TableHdr th;
-    std::vector<OutGroup> groups;
+    OutGroup* hgroups = nullptr;
+    int64_t ngroups = 0;
     {
         HIP_CHECK(hipMemcpy(&th, R.d_th, sizeof(th), hipMemcpyDeviceToHost));
         if (th.overflow == 1) { set_err(errbuf, errlen, "group table overflow — raise max_groups_hint"); rc = YT_ERR_CAPACITY; goto fail; }
-        int64_t ngroups = (int64_t)th.ngroups;
-        if (ngroups > 0 && R.groups_compacted) {
-            groups.resize(ngroups);
-            HIP_CHECK(hipMemcpyAsync(groups.data(), R.d_groups, sizeof(OutGroup) * ngroups,
-                                     hipMemcpyDeviceToHost, R.stream));
-            HIP_CHECK(hipStreamSynchronize(R.stream));
-        } else if (ngroups > 0) {
-            HIP_CHECK(hipMalloc(&R.d_groups, sizeof(OutGroup) * ngroups));
-            HIP_CHECK(hipMalloc(&R.d_counter, sizeof(unsigned long long)));
+        ngroups = (int64_t)th.ngroups;
+        if (ngroups > 0 && !R.groups_compacted) {
+            HIP_CHECK(pool_alloc(&R.d_groups, sizeof(OutGroup) * ngroups));
+            HIP_CHECK(pool_alloc(&R.d_counter, sizeof(unsigned long long)));
             HIP_CHECK(hipMemsetAsync(R.d_counter, 0, sizeof(unsigned long long), R.stream));
             HIP_CHECK(ytql_launch_compact(nullptr, R.d_th, R.d_slots, plan->agg_count,
                                           R.d_groups, R.d_counter, R.nslots, R.stream));
-            groups.resize(ngroups);
-            HIP_CHECK(hipMemcpyAsync(groups.data(), R.d_groups, sizeof(OutGroup) * ngroups,
+        }
+        if (ngroups > 0) {
+            HIP_CHECK(pool_alloc_host(&hgroups, sizeof(OutGroup) * ngroups));
+            HIP_CHECK(hipMemcpyAsync(hgroups, R.d_groups, sizeof(OutGroup) * ngroups,
                                      hipMemcpyDeviceToHost, R.stream));
             HIP_CHECK(hipStreamSynchronize(R.stream));
         }
@@ -1060,8 +1159,9 @@ extern "C" int yt_gpu_query_execute(
             HIP_CHECK(hipMemcpy(gaccum.data(), R.d_gaccum,
                                 sizeof(uint64_t) * (1 + 2 * kMaxAggs), hipMemcpyDeviceToHost));
         }
-        rc = emit_rows(plan, chunk, groups, th, 0, gaccum.data(),
+        rc = emit_rows(plan, chunk, hgroups, ngroups, th, 0, gaccum.data(),
                        fs.valid && fs.key_col < 0, output, errbuf, errlen);
+        g_pool.put(hgroups);
         if (rc) return rc;
     }
     if (stats) {
@@ -1075,6 +1175,7 @@ extern "C" int yt_gpu_query_execute(
         stats->grouped_row_count = (int64_t)th.ngroups
             + (plan->key_count ? (int64_t)(th.side_used[0] + th.side_used[1]) : 0);
         stats->incomplete_output = (th.overflow == 2);
+        stats->execute_time_ms = now_ms() - tw0;
     }
     return YT_OK;
 fail:
@@ -1137,8 +1238,8 @@ extern "C" int yt_gpu_query_partial(
 
         if (!R.groups_compacted) {
             /* compact table into OutGroups (device) */
-            HIP_CHECK(hipMalloc(&R.d_groups, sizeof(OutGroup) * (total ? total : 1)));
-            HIP_CHECK(hipMalloc(&R.d_counter, sizeof(unsigned long long)));
+            HIP_CHECK(pool_alloc(&R.d_groups, sizeof(OutGroup) * (total ? total : 1)));
+            HIP_CHECK(pool_alloc(&R.d_counter, sizeof(unsigned long long)));
             HIP_CHECK(hipMemsetAsync(R.d_counter, 0, sizeof(unsigned long long), R.stream));
             if (ngroups > 0) {
                 HIP_CHECK(ytql_launch_compact(nullptr, R.d_th, R.d_slots, plan->agg_count,
@@ -1169,7 +1270,7 @@ extern "C" int yt_gpu_query_partial(
 
         /* partition counts → host prefix → scatter */
         unsigned long long* d_counts = nullptr;
-        HIP_CHECK(hipMalloc(&d_counts, sizeof(unsigned long long) * partition_count));
+        HIP_CHECK(pool_alloc(&d_counts, sizeof(unsigned long long) * partition_count));
         HIP_CHECK(hipMemsetAsync(d_counts, 0, sizeof(unsigned long long) * partition_count, R.stream));
         HIP_CHECK(ytql_launch_part_count(R.d_groups, total, partition_count, sum_slot,
                                          d_counts, R.stream));
@@ -1187,7 +1288,7 @@ extern "C" int yt_gpu_query_partial(
         HIP_CHECK(ytql_launch_part_scatter(R.d_groups, total, partition_count, sum_slot,
                                            d_counts, (YtStateRow*)states_device, R.stream));
         HIP_CHECK(hipStreamSynchronize(R.stream));
-        hipFree(d_counts);
+        g_pool.put(d_counts);
         for (int p = 0; p < partition_count; p++) part_counts[p] = (int64_t)counts[p];
         if (stats) {
             stats->rows_read = chunk->row_count;
@@ -1234,8 +1335,8 @@ extern "C" int yt_gpu_merge_states(
         int64_t ngroups = (int64_t)th.ngroups;
         std::vector<OutGroup> groups;
         if (ngroups > 0) {
-            HIP_CHECK(hipMalloc(&R.d_groups, sizeof(OutGroup) * ngroups));
-            HIP_CHECK(hipMalloc(&R.d_counter, sizeof(unsigned long long)));
+            HIP_CHECK(pool_alloc(&R.d_groups, sizeof(OutGroup) * ngroups));
+            HIP_CHECK(pool_alloc(&R.d_counter, sizeof(unsigned long long)));
             HIP_CHECK(hipMemsetAsync(R.d_counter, 0, sizeof(unsigned long long), R.stream));
             HIP_CHECK(ytql_launch_compact(nullptr, R.d_th, R.d_slots, plan->agg_count,
                                           R.d_groups, R.d_counter, R.nslots, R.stream));
@@ -1258,7 +1359,8 @@ extern "C" int yt_gpu_merge_states(
         fake.row_count = 0;
         fake.column_count = kMaxCols;
         fake.columns = cols.data();
-        rc = emit_rows(plan, &fake, groups, th, 0, nullptr, 0, output, errbuf, errlen);
+        rc = emit_rows(plan, &fake, groups.data(), (int64_t)groups.size(), th, 0,
+                       nullptr, 0, output, errbuf, errlen);
         if (rc) return rc;
         if (stats) {
             stats->rows_written = output->row_count;
