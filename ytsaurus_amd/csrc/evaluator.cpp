@@ -26,6 +26,7 @@ using namespace ytql;
 /* launch wrappers from kernels.hip */
 extern "C" {
 hipError_t ytql_launch_parse_segments(const DevSeg*, int, SegEx*, unsigned*, unsigned*, hipStream_t);
+hipError_t ytql_launch_scan_nullflags(const DevSeg*, const SegEx*, int, unsigned*, hipStream_t);
 hipError_t ytql_launch_scan_partition(const PartParams*, const DevSeg*, const SegEx*,
                                       const FastCol*, TableHdr*, unsigned long long*,
                                       void*, unsigned long long*, uint64_t*,
@@ -624,6 +625,8 @@ static int setup_chunk(const YtChunk* chunk, DeviceRun* R, unsigned* maxw_out,
     HIP_CHECK(hipMemsetAsync(R->d_maxw, 0, sizeof(unsigned), R->stream));
     HIP_CHECK(hipMemsetAsync(R->d_err, 0, sizeof(unsigned), R->stream));
     HIP_CHECK(ytql_launch_parse_segments(R->d_segs, R->nsegs, R->d_segex, R->d_maxw,
+                                         R->d_colnull, R->stream));
+    HIP_CHECK(ytql_launch_scan_nullflags(R->d_segs, R->d_segex, R->nsegs,
                                          R->d_colnull, R->stream));
     HIP_CHECK(hipMemcpyAsync(maxw_out, R->d_maxw, sizeof(unsigned),
                              hipMemcpyDeviceToHost, R->stream));

@@ -165,24 +165,38 @@ __global__ void k_parse_segments(const DevSeg* segs, int nsegs, SegEx* out,
         }
         }
     }
-    /* value-null presence (DirectDense): OR over the null bitmap */
-    if (s.type == YT_SEG_DIRECT_DENSE) {
-        const uint64_t* bm = (const uint64_t*)((const uint8_t*)b + e.off_bitmap_bytes);
-        int64_t words = (s.row_count + 63) / 64;
-        uint64_t any = 0;
-        /* last partial byte-word is zero-padded by the writer (bitmap.h) */
-        for (int64_t wI = 0; wI < words; wI++) any |= bm[wI];
-        /* mask tail bits beyond row_count (padding is zero, but be safe) */
-        if (any) {
-            e.flags |= 1;
-            atomicOr(&col_null_flags[s.col], 1u);
-        }
-    } else if (s.type != YT_SEG_DOUBLE) {
-        /* other formats: unknown without a scan — be conservative */
+    if (s.type != YT_SEG_DIRECT_DENSE && s.type != YT_SEG_DOUBLE) {
+        /* dictionary/RLE null presence is unknown without an id scan —
+         * be conservative (fast paths only run on DirectDense anyway) */
         atomicOr(&col_null_flags[s.col], 2u);
     }
     out[i] = e;
     atomicMax(max_width_out, e.w_values);
+}
+
+/* null-presence scan: one workgroup per DirectDense segment, threads stride
+ * the null bitmap words, wave-OR reduce (writer zero-pads to 8 bytes,
+ * bitmap.h) */
+__global__ void __launch_bounds__(256)
+k_scan_nullflags(const DevSeg* segs, const SegEx* segex, int nsegs,
+                 unsigned* col_null_flags)
+{
+    int seg = blockIdx.x;
+    if (seg >= nsegs) return;
+    const DevSeg& s = segs[seg];
+    if (s.type != YT_SEG_DIRECT_DENSE && s.type != YT_SEG_DOUBLE) return;
+    const SegEx& e = segex[seg];
+    const uint64_t* bm = (const uint64_t*)((const uint8_t*)s.blob + e.off_bitmap_bytes);
+    int64_t words = ((int64_t)s.row_count + 63) / 64;
+    uint64_t any = 0;
+    for (int64_t i = threadIdx.x; i < words; i += 256) any |= bm[i];
+    any = (uint64_t)__any((long long)any != 0);
+    __shared__ int found;
+    if (threadIdx.x == 0) found = 0;
+    __syncthreads();
+    if (any && (threadIdx.x & 63) == 0) found = 1;
+    __syncthreads();
+    if (threadIdx.x == 0 && found) atomicOr(&col_null_flags[s.col], 1u);
 }
 
 /* ------------------------------------------------------------------ */
@@ -1078,6 +1092,15 @@ hipError_t ytql_launch_parse_segments(const DevSeg* segs, int nsegs, SegEx* out,
     int grid = (nsegs + block - 1) / block;
     hipLaunchKernelGGL(k_parse_segments, dim3(grid), dim3(block), 0, st,
                        segs, nsegs, out, max_width, col_null_flags);
+    return hipGetLastError();
+}
+
+hipError_t ytql_launch_scan_nullflags(const DevSeg* segs, const SegEx* segex,
+                                      int nsegs, unsigned* col_null_flags,
+                                      hipStream_t st)
+{
+    hipLaunchKernelGGL(k_scan_nullflags, dim3(nsegs), dim3(256), 0, st,
+                       segs, segex, nsegs, col_null_flags);
     return hipGetLastError();
 }
 
