@@ -698,14 +698,16 @@ static int run_partitioned(const YtPlan* plan, const YtChunk* chunk,
     const int64_t rows = chunk->row_count;
     int32_t seg0_rows = chunk->columns[0].segments[0].row_count;
 
-    /* LDS: 4*kNB u32 histogram structures + column staging */
+    /* LDS: 4*kNB u32 histogram structures + staged columns (value column's
+     * packed words are NOT staged — read from global in the write pass) */
     unsigned w = maxw ? maxw : 1;
+    int staged_cols = nused - (pp.val_idx >= 0 ? 1 : 0);
     int tile_rows = 8192;
     size_t lds = 0;
     for (;;) {
-        size_t per = ((((size_t)tile_rows * w / 64 + 2) * 8 + 15) & ~(size_t)15)
-                   + (((size_t)tile_rows / 8 + 15) & ~(size_t)15);
-        lds = 4 * kNB * 4 + (size_t)nused * per + 256;
+        size_t words = ((((size_t)tile_rows * w / 64 + 2) * 8 + 15) & ~(size_t)15);
+        size_t bm = (((size_t)tile_rows / 8 + 15) & ~(size_t)15);
+        lds = 4 * kNB * 4 + (size_t)staged_cols * words + (size_t)nused * bm + 256;
         if (lds <= 64 * 1024 || tile_rows == 256) break;
         tile_rows >>= 1;
     }

@@ -739,7 +739,12 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
     const int tid = threadIdx.x;
 
     /* LDS layout: hist[kNB] u32 | gbase[kNB] u32 | nhist[kNB] u32 |
-     *             ngbase[kNB] u32 | column staging */
+     *             ngbase[kNB] u32 | staged filter/key words + bitmaps |
+     *             value-column null bitmap.
+     * The value column's packed words are NOT staged: they are read once,
+     * directly from L2/HBM in the write pass (per-lane strided reads of a
+     * hot tile region are coalesced enough, and skipping the staging both
+     * halves LDS use and removes a full LDS round trip). */
     unsigned* hist = (unsigned*)smem;
     unsigned* gbase = hist + kNB;
     unsigned* nhist = gbase + kNB;
@@ -756,28 +761,32 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
         const uint8_t* col_bitmap[4];
         uint32_t col_w[4];
         int64_t col_w0[4];
-        int32_t seg_rows = 0;
+        int32_t seg_rows = segs[cols[0].seg_off + seg_idx].row_count;
+        int64_t t1 = t0 + pp.tile_rows;
+        if (t1 > seg_rows) t1 = seg_rows;
+
         for (int u = 0; u < pp.nused; u++) {
             const DevSeg& sg = segs[cols[u].seg_off + seg_idx];
             const SegEx& e = segex[cols[u].seg_off + seg_idx];
-            seg_rows = sg.row_count;
-            int64_t t1 = t0 + pp.tile_rows;
-            if (t1 > seg_rows) t1 = seg_rows;
             uint32_t w = e.w_values;
             col_w[u] = w;
             int64_t W0 = ((uint64_t)t0 * w) >> 6;
-            int64_t W1 = w ? ((((uint64_t)t1 * w) + 63) >> 6) : 0;
-            int64_t nwords = (w == 0) ? 0 : (W1 - W0 + 1);
             col_w0[u] = W0;
-            uint64_t* dst = (uint64_t*)(stage_base + lds_off);
-            col_words[u] = dst;
-            const uint64_t* src = sg.blob + e.off_values_words + W0;
-            int64_t vec_words = (w == 0) ? 0 : (((uint64_t)seg_rows * w + 63) >> 6);
-            int64_t avail = vec_words - W0;
-            if (nwords > avail) nwords = avail;
-            for (int64_t i = tid; i < nwords; i += 256) dst[i] = src[i];
-            lds_off += ((nwords * 8) + 15) & ~(int64_t)15;
-
+            if (u != pp.val_idx) {
+                int64_t W1 = w ? ((((uint64_t)t1 * w) + 63) >> 6) : 0;
+                int64_t nwords = (w == 0) ? 0 : (W1 - W0 + 1);
+                uint64_t* dst = (uint64_t*)(stage_base + lds_off);
+                col_words[u] = dst;
+                const uint64_t* src = sg.blob + e.off_values_words + W0;
+                int64_t vec_words = (w == 0) ? 0 : (((uint64_t)seg_rows * w + 63) >> 6);
+                int64_t avail = vec_words - W0;
+                if (nwords > avail) nwords = avail;
+                for (int64_t i = tid; i < nwords; i += 256) dst[i] = src[i];
+                lds_off += ((nwords * 8) + 15) & ~(int64_t)15;
+            } else {
+                col_words[u] = (uint64_t*)(sg.blob + e.off_values_words);  /* global */
+            }
+            /* bitmap staged for every used column (validity checks) */
             uint8_t* bdst = (uint8_t*)(stage_base + lds_off);
             col_bitmap[u] = bdst;
             const uint8_t* bsrc = (const uint8_t*)sg.blob + e.off_bitmap_bytes + (t0 >> 3);
@@ -790,8 +799,6 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
 
         const DevSeg& sk = segs[cols[pp.key_idx].seg_off + seg_idx];
         const DevSeg* sv = pp.val_idx >= 0 ? &segs[cols[pp.val_idx].seg_off + seg_idx] : nullptr;
-        int64_t t1 = t0 + pp.tile_rows;
-        if (t1 > seg_rows) t1 = seg_rows;
 
         const int R = (pp.tile_rows + 255) / 256;
         uint32_t row_b[32];
@@ -818,23 +825,18 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
                 key = (uint64_t)zz_dec(sk.min_value + packed);
             }
             int val_null = 1;
-            uint64_t val = 0;
             if (pp.val_idx >= 0) {
                 val_null = pp.has_val_nulls && bm_get(col_bitmap[pp.val_idx], j - t0);
-                if (!val_null) {
-                    uint64_t packed = bp_get_win(col_words[pp.val_idx], col_w[pp.val_idx], j,
-                                                 col_w0[pp.val_idx]);
-                    val = (uint64_t)zz_dec(sv->min_value + packed);
-                }
             }
 
             if (key_null || (!key_null && key == kEmptyKey)) {
-                /* side groups: null key, or the LDS-table sentinel key
-                 * (INT64_MIN) — rare; direct device-scope atomics */
                 int side = key_null ? 1 : 0;
                 th->side_used[side] = 1;
                 atomicAdd((unsigned long long*)&th->side_cnt[side], 1ULL);
                 if (pp.sum_slot >= 0 && !val_null) {
+                    uint64_t pv = bp_get_win(col_words[pp.val_idx],
+                                             col_w[pp.val_idx], j, 0);
+                    uint64_t val = (uint64_t)zz_dec(sv->min_value + pv);
                     atomicAdd((unsigned long long*)&th->side_agg[side][2 * pp.sum_slot], val);
                     atomicAdd((unsigned long long*)&th->side_agg[side][2 * pp.sum_slot + 1], 1ULL);
                 }
@@ -879,8 +881,9 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
                 } else {
                     uint64_t val = 0;
                     if (pp.val_idx >= 0) {
-                        uint64_t pv = bp_get_win(col_words[pp.val_idx], col_w[pp.val_idx], j,
-                                                 col_w0[pp.val_idx]);
+                        /* value read straight from global (not staged) */
+                        uint64_t pv = bp_get_win(col_words[pp.val_idx],
+                                                 col_w[pp.val_idx], j, 0);
                         val = (uint64_t)zz_dec(sv->min_value + pv);
                     }
                     recs[(int64_t)b * pp.bucket_stride + gbase[b] + row_off[i]] =
@@ -917,25 +920,51 @@ k_bucket_agg(const ulonglong2* recs, const unsigned long long* cursors,
     int64_t n = (int64_t)cursors[bucket];
     const ulonglong2* rows = recs + (int64_t)bucket * bucket_stride;
     bool full = false;
-    for (int64_t i = tid; i < n; i += 256) {
-        ulonglong2 kv = rows[i];
-        uint64_t s = mix64(kv.x) & (kHSlots - 1);
-        int found = 0;
-        for (int it = 0; it < kHSlots; it++) {
-            unsigned long long k = tab[s * 3];
-            if (k == (unsigned long long)kv.x) { found = 1; break; }
-            if (k == (unsigned long long)kEmptyKey) {
-                k = atomicCAS(&tab[s * 3], (unsigned long long)kEmptyKey,
-                              (unsigned long long)kv.x);
-                if (k == (unsigned long long)kEmptyKey ||
-                    k == (unsigned long long)kv.x) { found = 1; break; }
-            }
-            s = (s + 1) & (kHSlots - 1);
+    /* 4 records per thread per pass: independent probes overlap LDS latency */
+    int64_t i = tid;
+    for (; i + 768 < n; i += 1024) {
+        ulonglong2 kv0 = rows[i];
+        ulonglong2 kv1 = rows[i + 256];
+        ulonglong2 kv2 = rows[i + 512];
+        ulonglong2 kv3 = rows[i + 768];
+        uint64_t s0 = mix64(kv0.x) & (kHSlots - 1);
+        uint64_t s1 = mix64(kv1.x) & (kHSlots - 1);
+        uint64_t s2 = mix64(kv2.x) & (kHSlots - 1);
+        uint64_t s3 = mix64(kv3.x) & (kHSlots - 1);
+        #define PROBE(kv, sv_)                                               \
+        {                                                                    \
+            uint64_t sp = sv_;                                               \
+            int found = 0;                                                   \
+            for (int it = 0; it < kHSlots; it++) {                           \
+                unsigned long long k = tab[sp * 3];                          \
+                if (k == (unsigned long long)kv.x) { found = 1; break; }     \
+                if (k == (unsigned long long)kEmptyKey) {                    \
+                    k = atomicCAS(&tab[sp * 3], (unsigned long long)kEmptyKey,\
+                                  (unsigned long long)kv.x);                 \
+                    if (k == (unsigned long long)kEmptyKey ||                \
+                        k == (unsigned long long)kv.x) { found = 1; break; } \
+                }                                                            \
+                sp = (sp + 1) & (kHSlots - 1);                               \
+            }                                                                \
+            if (!found) { full = true; }                                     \
+            else {                                                           \
+                atomicAdd(&tab[sp * 3 + 1], 1ULL | (1ULL << 32));            \
+                if (sum_slot >= 0)                                           \
+                    atomicAdd(&tab[sp * 3 + 2], (unsigned long long)kv.y);   \
+            }                                                                \
         }
-        if (!found) { full = true; break; }
-        atomicAdd(&tab[s * 3 + 1], 1ULL | (1ULL << 32));
-        if (sum_slot >= 0) atomicAdd(&tab[s * 3 + 2], (unsigned long long)kv.y);
+        PROBE(kv0, s0)
+        PROBE(kv1, s1)
+        PROBE(kv2, s2)
+        PROBE(kv3, s3)
+        if (full) break;
     }
+    for (; i < n && !full; i += 256) {
+        ulonglong2 kv = rows[i];
+        uint64_t s0 = mix64(kv.x) & (kHSlots - 1);
+        PROBE(kv, s0)
+    }
+    #undef PROBE
     if (nrecs) {
         int64_t nn = (int64_t)ncursors[bucket];
         const uint64_t* nrows = nrecs + (int64_t)bucket * nbucket_stride;
