@@ -161,6 +161,12 @@ struct PartParams {
     int64_t filter_lo, filter_hi;
     int64_t bucket_stride;        /* record capacity per bucket */
     int64_t nbucket_stride;       /* null-stream capacity per bucket */
+    /* 8-byte packed records: when the key and value zigzag spans fit 64
+     * bits together, a record is (kzz - gmin_k) | (vzz - gmin_v) << bits_k
+     * — half the partition traffic of {key,val} pairs. */
+    int32_t packed_mode;
+    int32_t bits_k;
+    uint64_t gmin_k, gmin_v;
 };
 
 struct KernelTimes {

@@ -25,7 +25,7 @@ using namespace ytql;
 
 /* launch wrappers from kernels.hip */
 extern "C" {
-hipError_t ytql_launch_parse_segments(const DevSeg*, int, SegEx*, unsigned*, unsigned*, hipStream_t);
+hipError_t ytql_launch_parse_segments(const DevSeg*, int, SegEx*, unsigned*, unsigned*, unsigned long long*, unsigned long long*, hipStream_t);
 hipError_t ytql_launch_scan_nullflags(const DevSeg*, const SegEx*, int, unsigned*, hipStream_t);
 hipError_t ytql_launch_scan_partition(const PartParams*, const DevSeg*, const SegEx*,
                                       const FastCol*, TableHdr*, unsigned long long*,
@@ -34,7 +34,8 @@ hipError_t ytql_launch_scan_partition(const PartParams*, const DevSeg*, const Se
 hipError_t ytql_launch_bucket_agg(const void*, const unsigned long long*, int64_t,
                                   const uint64_t*, const unsigned long long*, int64_t,
                                   OutGroup*, unsigned long long*, int64_t,
-                                  TableHdr*, int, int, hipStream_t);
+                                  TableHdr*, int, int, int, int,
+                                  uint64_t, uint64_t, hipStream_t);
 hipError_t ytql_launch_scan_generic(const DevPlan*, const DevSeg*, const SegEx*,
                                     const int32_t*, const int32_t*, int64_t,
                                     TableHdr*, unsigned long long*, unsigned*, hipStream_t);
@@ -533,6 +534,7 @@ struct DeviceRun {
     unsigned long long* d_gaccum = nullptr;
     FastCol* d_fastcols = nullptr;
     unsigned* d_colnull = nullptr;
+    unsigned long long* d_zzrange = nullptr;   /* [0..kMaxCols) min, [kMaxCols..) max */
     unsigned long long* d_cursors = nullptr;
     void* d_recs = nullptr;
     unsigned long long* d_ncursors = nullptr;
@@ -542,6 +544,8 @@ struct DeviceRun {
     int64_t groups_capacity = 0;
     bool groups_compacted = false;     /* d_groups already holds final groups */
     unsigned col_null_flags[kMaxCols] = {0};
+    uint64_t col_zzmin[kMaxCols] = {0};
+    uint64_t col_zzmax[kMaxCols] = {0};
     hipStream_t stream = 0;
 
     ~DeviceRun()
@@ -559,6 +563,7 @@ struct DeviceRun {
         g_pool.put(d_gaccum);
         g_pool.put(d_fastcols);
         g_pool.put(d_colnull);
+        g_pool.put(d_zzrange);
         g_pool.put(d_cursors);
         g_pool.put(d_recs);
         g_pool.put(d_ncursors);
@@ -616,6 +621,9 @@ static int setup_chunk(const YtChunk* chunk, DeviceRun* R, unsigned* maxw_out,
     HIP_CHECK(pool_alloc(&R->d_err, sizeof(unsigned)));
     HIP_CHECK(pool_alloc(&R->d_colnull, sizeof(unsigned) * kMaxCols));
     HIP_CHECK(hipMemsetAsync(R->d_colnull, 0, sizeof(unsigned) * kMaxCols, R->stream));
+    HIP_CHECK(pool_alloc(&R->d_zzrange, sizeof(uint64_t) * 2 * kMaxCols));
+    HIP_CHECK(hipMemsetAsync(R->d_zzrange, 0xFF, sizeof(uint64_t) * kMaxCols, R->stream));
+    HIP_CHECK(hipMemsetAsync(R->d_zzrange + kMaxCols, 0, sizeof(uint64_t) * kMaxCols, R->stream));
     HIP_CHECK(hipMemcpyAsync(R->d_segs, R->h_segs.data(), sizeof(DevSeg) * R->nsegs,
                              hipMemcpyHostToDevice, R->stream));
     HIP_CHECK(hipMemcpyAsync(R->d_off, R->h_off.data(), sizeof(int32_t) * ncols,
@@ -625,13 +633,20 @@ static int setup_chunk(const YtChunk* chunk, DeviceRun* R, unsigned* maxw_out,
     HIP_CHECK(hipMemsetAsync(R->d_maxw, 0, sizeof(unsigned), R->stream));
     HIP_CHECK(hipMemsetAsync(R->d_err, 0, sizeof(unsigned), R->stream));
     HIP_CHECK(ytql_launch_parse_segments(R->d_segs, R->nsegs, R->d_segex, R->d_maxw,
-                                         R->d_colnull, R->stream));
+                                         R->d_colnull, R->d_zzrange,
+                                         R->d_zzrange + kMaxCols, R->stream));
     HIP_CHECK(ytql_launch_scan_nullflags(R->d_segs, R->d_segex, R->nsegs,
                                          R->d_colnull, R->stream));
     HIP_CHECK(hipMemcpyAsync(maxw_out, R->d_maxw, sizeof(unsigned),
                              hipMemcpyDeviceToHost, R->stream));
     HIP_CHECK(hipMemcpyAsync(R->col_null_flags, R->d_colnull,
                              sizeof(unsigned) * kMaxCols,
+                             hipMemcpyDeviceToHost, R->stream));
+    HIP_CHECK(hipMemcpyAsync(R->col_zzmin, R->d_zzrange,
+                             sizeof(uint64_t) * kMaxCols,
+                             hipMemcpyDeviceToHost, R->stream));
+    HIP_CHECK(hipMemcpyAsync(R->col_zzmax, R->d_zzrange + kMaxCols,
+                             sizeof(uint64_t) * kMaxCols,
                              hipMemcpyDeviceToHost, R->stream));
     HIP_CHECK(hipStreamSynchronize(R->stream));
     return YT_OK;
@@ -695,6 +710,31 @@ static int run_partitioned(const YtPlan* plan, const YtChunk* chunk,
     pp.has_key_nulls = (R->col_null_flags[fs->key_col] != 0);
     pp.has_val_nulls = pp.val_idx >= 0 && (R->col_null_flags[fs->sum_col[0]] != 0);
 
+    /* 8-byte packed records when the combined zigzag spans fit 64 bits */
+    {
+        auto bits_of = [](uint64_t span) -> int {
+            int b = 0;
+            while (span) { b++; span >>= 1; }
+            return b;
+        };
+        uint64_t span_k = R->col_zzmax[fs->key_col] - R->col_zzmin[fs->key_col];
+        int bk = bits_of(span_k);
+        int bv = 0;
+        uint64_t gmin_v = 0;
+        if (pp.val_idx >= 0) {
+            uint64_t span_v = R->col_zzmax[fs->sum_col[0]] - R->col_zzmin[fs->sum_col[0]];
+            bv = bits_of(span_v);
+            gmin_v = R->col_zzmin[fs->sum_col[0]];
+        }
+        if (bk + bv <= 64) {
+            pp.packed_mode = 1;
+            pp.bits_k = bk ? bk : 1;
+            if (pp.bits_k + bv > 64) pp.bits_k = bk;  /* bk>=1 here */
+            pp.gmin_k = R->col_zzmin[fs->key_col];
+            pp.gmin_v = gmin_v;
+        }
+    }
+
     const int64_t rows = chunk->row_count;
     int32_t seg0_rows = chunk->columns[0].segments[0].row_count;
 
@@ -735,7 +775,8 @@ static int run_partitioned(const YtPlan* plan, const YtChunk* chunk,
                                  hipMemcpyHostToDevice, R->stream));
         HIP_CHECK(pool_alloc(&R->d_cursors, sizeof(uint64_t) * kNB));
         HIP_CHECK(hipMemsetAsync(R->d_cursors, 0, sizeof(uint64_t) * kNB, R->stream));
-        HIP_CHECK(pool_alloc(&R->d_recs, (size_t)kNB * pp.bucket_stride * 16));
+        HIP_CHECK(pool_alloc(&R->d_recs,
+                             (size_t)kNB * pp.bucket_stride * (pp.packed_mode ? 8 : 16)));
         if (pp.has_val_nulls) {
             HIP_CHECK(pool_alloc(&R->d_ncursors, sizeof(uint64_t) * kNB));
             HIP_CHECK(hipMemsetAsync(R->d_ncursors, 0, sizeof(uint64_t) * kNB, R->stream));
@@ -771,7 +812,8 @@ static int run_partitioned(const YtPlan* plan, const YtChunk* chunk,
                                          R->d_nrecs, R->d_ncursors, pp.nbucket_stride,
                                          R->d_groups, R->d_counter, cap_groups,
                                          R->d_th, pp.sum_slot, pp.agg_count,
-                                         R->stream));
+                                         pp.packed_mode, pp.bits_k,
+                                         pp.gmin_k, pp.gmin_v, R->stream));
         HIP_CHECK(hipEventRecord(ev2, R->stream));
         HIP_CHECK(hipStreamSynchronize(R->stream));
         float msA = 0, msB = 0;

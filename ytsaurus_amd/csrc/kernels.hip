@@ -101,7 +101,9 @@ __device__ __forceinline__ uint64_t bp_get_win(const uint64_t* win, uint32_t wid
 
 __global__ void k_parse_segments(const DevSeg* segs, int nsegs, SegEx* out,
                                  unsigned* max_width_out,
-                                 unsigned* col_null_flags /* kMaxCols words */)
+                                 unsigned* col_null_flags /* kMaxCols words */,
+                                 unsigned long long* col_zzmin /* kMaxCols, init ~0 */,
+                                 unsigned long long* col_zzmax /* kMaxCols, init 0 */)
 {
     int i = blockIdx.x * blockDim.x + threadIdx.x;
     if (i >= nsegs) return;
@@ -172,6 +174,17 @@ __global__ void k_parse_segments(const DevSeg* segs, int nsegs, SegEx* out,
     }
     out[i] = e;
     atomicMax(max_width_out, e.w_values);
+    /* per-column zigzag-space range, from segment meta alone
+     * (value = min_value + packed, packed < 2^w) */
+    {
+        unsigned long long lo = (unsigned long long)s.min_value;
+        unsigned long long span = (e.w_values >= 64) ? ~0ULL
+                                : ((1ULL << e.w_values) - 1);
+        unsigned long long hi = lo + span;   /* may wrap for degenerate metas */
+        if (hi < lo) hi = ~0ULL;
+        atomicMin(&col_zzmin[s.col], lo);
+        atomicMax(&col_zzmax[s.col], hi);
+    }
 }
 
 /* null-presence scan: one workgroup per DirectDense segment, threads stride
@@ -852,6 +865,7 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
                 row_off[i] = atomicAdd(&hist[b], 1u);
             }
         }
+        uint64_t* recs8 = (uint64_t*)recs;
         __syncthreads();
         for (int i = tid; i < kNB; i += 256) {
             unsigned c = hist[i];
@@ -878,6 +892,18 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
                 unsigned b = row_b[i] & 0x7FFFFFFFu;
                 if (row_b[i] & 0x80000000u) {
                     nrecs[(int64_t)b * pp.nbucket_stride + ngbase[b] + row_off[i]] = key;
+                } else if (pp.packed_mode) {
+                    uint64_t kzz = (sk.min_value +
+                        bp_get_win(col_words[pp.key_idx], col_w[pp.key_idx], j,
+                                   col_w0[pp.key_idx])) - pp.gmin_k;
+                    uint64_t rec = kzz;
+                    if (pp.val_idx >= 0) {
+                        uint64_t pv = bp_get_win(col_words[pp.val_idx],
+                                                 col_w[pp.val_idx], j, 0);
+                        uint64_t vzz = (sv->min_value + pv) - pp.gmin_v;
+                        rec |= vzz << pp.bits_k;
+                    }
+                    recs8[(int64_t)b * pp.bucket_stride + gbase[b] + row_off[i]] = rec;
                 } else {
                     uint64_t val = 0;
                     if (pp.val_idx >= 0) {
@@ -904,7 +930,8 @@ k_bucket_agg(const ulonglong2* recs, const unsigned long long* cursors,
              const uint64_t* nrecs, const unsigned long long* ncursors,
              int64_t nbucket_stride,
              OutGroup* out, unsigned long long* out_counter, int64_t out_cap,
-             TableHdr* th, int sum_slot, int agg_count)
+             TableHdr* th, int sum_slot, int agg_count,
+             int packed_mode, int bits_k, uint64_t gmin_k, uint64_t gmin_v)
 {
     __shared__ unsigned long long tab[kHSlots * 3];
     const int tid = threadIdx.x;
@@ -919,14 +946,25 @@ k_bucket_agg(const ulonglong2* recs, const unsigned long long* cursors,
 
     int64_t n = (int64_t)cursors[bucket];
     const ulonglong2* rows = recs + (int64_t)bucket * bucket_stride;
+    const uint64_t* rows8 = (const uint64_t*)recs + (int64_t)bucket * bucket_stride;
+    const uint64_t kmask = (bits_k >= 64) ? ~0ULL : ((1ULL << bits_k) - 1);
     bool full = false;
     /* 4 records per thread per pass: independent probes overlap LDS latency */
     int64_t i = tid;
+    #define LOADKV(kv, idx)                                                  \
+        ulonglong2 kv;                                                       \
+        if (packed_mode) {                                                   \
+            uint64_t r_ = rows8[idx];                                        \
+            kv.x = (uint64_t)zz_dec(gmin_k + (r_ & kmask));                  \
+            kv.y = (uint64_t)zz_dec(gmin_v + (r_ >> bits_k));                \
+        } else {                                                             \
+            kv = rows[idx];                                                  \
+        }
     for (; i + 768 < n; i += 1024) {
-        ulonglong2 kv0 = rows[i];
-        ulonglong2 kv1 = rows[i + 256];
-        ulonglong2 kv2 = rows[i + 512];
-        ulonglong2 kv3 = rows[i + 768];
+        LOADKV(kv0, i)
+        LOADKV(kv1, i + 256)
+        LOADKV(kv2, i + 512)
+        LOADKV(kv3, i + 768)
         uint64_t s0 = mix64(kv0.x) & (kHSlots - 1);
         uint64_t s1 = mix64(kv1.x) & (kHSlots - 1);
         uint64_t s2 = mix64(kv2.x) & (kHSlots - 1);
@@ -960,11 +998,12 @@ k_bucket_agg(const ulonglong2* recs, const unsigned long long* cursors,
         if (full) break;
     }
     for (; i < n && !full; i += 256) {
-        ulonglong2 kv = rows[i];
+        LOADKV(kv, i)
         uint64_t s0 = mix64(kv.x) & (kHSlots - 1);
         PROBE(kv, s0)
     }
     #undef PROBE
+    #undef LOADKV
     if (nrecs) {
         int64_t nn = (int64_t)ncursors[bucket];
         const uint64_t* nrows = nrecs + (int64_t)bucket * nbucket_stride;
@@ -1115,12 +1154,15 @@ extern "C" {
 
 hipError_t ytql_launch_parse_segments(const DevSeg* segs, int nsegs, SegEx* out,
                                       unsigned* max_width, unsigned* col_null_flags,
+                                      unsigned long long* col_zzmin,
+                                      unsigned long long* col_zzmax,
                                       hipStream_t st)
 {
     int block = 256;
     int grid = (nsegs + block - 1) / block;
     hipLaunchKernelGGL(k_parse_segments, dim3(grid), dim3(block), 0, st,
-                       segs, nsegs, out, max_width, col_null_flags);
+                       segs, nsegs, out, max_width, col_null_flags,
+                       col_zzmin, col_zzmax);
     return hipGetLastError();
 }
 
@@ -1153,12 +1195,15 @@ hipError_t ytql_launch_bucket_agg(const void* recs, const unsigned long long* cu
                                   OutGroup* out, unsigned long long* out_counter,
                                   int64_t out_cap,
                                   TableHdr* th, int sum_slot, int agg_count,
+                                  int packed_mode, int bits_k,
+                                  uint64_t gmin_k, uint64_t gmin_v,
                                   hipStream_t st)
 {
     hipLaunchKernelGGL(k_bucket_agg, dim3(kNB), dim3(256), 0, st,
                        (const ulonglong2*)recs, cursors, bucket_stride,
                        nrecs, ncursors, nbucket_stride,
-                       out, out_counter, out_cap, th, sum_slot, agg_count);
+                       out, out_counter, out_cap, th, sum_slot, agg_count,
+                       packed_mode, bits_k, gmin_k, gmin_v);
     return hipGetLastError();
 }
 
