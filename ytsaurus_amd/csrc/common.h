@@ -167,6 +167,8 @@ struct PartParams {
     int32_t packed_mode;
     int32_t bits_k;
     uint64_t gmin_k, gmin_v;
+    int32_t stage_bm_mask;        /* bit u: stage used-col u's null bitmap */
+    int32_t has_filter_nulls;
 };
 
 struct KernelTimes {

@@ -709,6 +709,11 @@ static int run_partitioned(const YtPlan* plan, const YtChunk* chunk,
     pp.filter_hi = fs->filter_hi;
     pp.has_key_nulls = (R->col_null_flags[fs->key_col] != 0);
     pp.has_val_nulls = pp.val_idx >= 0 && (R->col_null_flags[fs->sum_col[0]] != 0);
+    pp.has_filter_nulls = pp.filter_idx >= 0 && (R->col_null_flags[fs->filter_col] != 0);
+    pp.stage_bm_mask = 0;
+    for (int u = 0; u < nused; u++) {
+        if (R->col_null_flags[used[u]] != 0) pp.stage_bm_mask |= 1 << u;
+    }
 
     /* 8-byte packed records when the combined zigzag spans fit 64 bits */
     {
@@ -743,6 +748,10 @@ static int run_partitioned(const YtPlan* plan, const YtChunk* chunk,
     unsigned w = maxw ? maxw : 1;
     int staged_cols = nused - (pp.val_idx >= 0 ? 1 : 0);
     int tile_rows = 8192;
+    {
+        const char* ev = getenv("YTQL_TILE");   /* perf experiments only */
+        if (ev && atoi(ev) >= 256) tile_rows = atoi(ev);
+    }
     size_t lds = 0;
     for (;;) {
         size_t words = ((((size_t)tile_rows * w / 64 + 2) * 8 + 15) & ~(size_t)15);

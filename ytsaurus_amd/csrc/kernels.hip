@@ -799,13 +799,18 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
             } else {
                 col_words[u] = (uint64_t*)(sg.blob + e.off_values_words);  /* global */
             }
-            /* bitmap staged for every used column (validity checks) */
-            uint8_t* bdst = (uint8_t*)(stage_base + lds_off);
-            col_bitmap[u] = bdst;
-            const uint8_t* bsrc = (const uint8_t*)sg.blob + e.off_bitmap_bytes + (t0 >> 3);
-            int64_t bbytes = ((t1 - t0) + 7) / 8;
-            for (int64_t i = tid; i < bbytes; i += 256) bdst[i] = bsrc[i];
-            lds_off += (bbytes + 15) & ~(int64_t)15;
+            /* null bitmap staged only for columns that HAVE nulls
+             * (parse-time fact); null-free columns skip the checks */
+            if ((pp.stage_bm_mask >> u) & 1) {
+                uint8_t* bdst = (uint8_t*)(stage_base + lds_off);
+                col_bitmap[u] = bdst;
+                const uint8_t* bsrc = (const uint8_t*)sg.blob + e.off_bitmap_bytes + (t0 >> 3);
+                int64_t bbytes = ((t1 - t0) + 7) / 8;
+                for (int64_t i = tid; i < bbytes; i += 256) bdst[i] = bsrc[i];
+                lds_off += (bbytes + 15) & ~(int64_t)15;
+            } else {
+                col_bitmap[u] = nullptr;
+            }
         }
         for (int i = tid; i < kNB; i += 256) { hist[i] = 0; nhist[i] = 0; }
         __syncthreads();
@@ -823,7 +828,7 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
 
             if (pp.filter_idx >= 0) {
                 int u = pp.filter_idx;
-                if (bm_get(col_bitmap[u], j - t0)) continue;
+                if (pp.has_filter_nulls && bm_get(col_bitmap[u], j - t0)) continue;
                 uint64_t packed = bp_get_win(col_words[u], col_w[u], j, col_w0[u]);
                 const DevSeg& sf = segs[cols[u].seg_off + seg_idx];
                 int64_t v = zz_dec(sf.min_value + packed);
