@@ -169,6 +169,7 @@ struct PartParams {
     uint64_t gmin_k, gmin_v;
     int32_t stage_bm_mask;        /* bit u: stage used-col u's null bitmap */
     int32_t has_filter_nulls;
+    int32_t stage_val;            /* stage the value column's packed words too */
 };
 
 struct KernelTimes {

@@ -746,18 +746,24 @@ static int run_partitioned(const YtPlan* plan, const YtChunk* chunk,
     /* LDS: 4*kNB u32 histogram structures + staged columns (value column's
      * packed words are NOT staged — read from global in the write pass) */
     unsigned w = maxw ? maxw : 1;
-    int staged_cols = nused - (pp.val_idx >= 0 ? 1 : 0);
-    int tile_rows = 8192;
+    int tile_rows = 4096;
     {
         const char* ev = getenv("YTQL_TILE");   /* perf experiments only */
         if (ev && atoi(ev) >= 256) tile_rows = atoi(ev);
     }
+    /* stage the value column too when the budget allows (avoids a dependent
+     * global load per row in the write pass); otherwise read it from L2 */
     size_t lds = 0;
     for (;;) {
         size_t words = ((((size_t)tile_rows * w / 64 + 2) * 8 + 15) & ~(size_t)15);
         size_t bm = (((size_t)tile_rows / 8 + 15) & ~(size_t)15);
-        lds = 4 * kNB * 4 + (size_t)staged_cols * words + (size_t)nused * bm + 256;
-        if (lds <= 64 * 1024 || tile_rows == 256) break;
+        pp.stage_val = 1;
+        lds = 4 * kNB * 4 + (size_t)nused * (words + bm) + 256;
+        if (lds <= 52 * 1024 || tile_rows == 256) break;
+        pp.stage_val = 0;
+        lds = 4 * kNB * 4 + (size_t)(nused - (pp.val_idx >= 0 ? 1 : 0)) * words
+            + (size_t)nused * bm + 256;
+        if (lds <= 52 * 1024 || tile_rows == 256) break;
         tile_rows >>= 1;
     }
     if (lds > 160 * 1024) return YT_OK;   /* fall back */

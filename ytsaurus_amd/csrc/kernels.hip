@@ -785,7 +785,7 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
             col_w[u] = w;
             int64_t W0 = ((uint64_t)t0 * w) >> 6;
             col_w0[u] = W0;
-            if (u != pp.val_idx) {
+            if (u != pp.val_idx || pp.stage_val) {
                 int64_t W1 = w ? ((((uint64_t)t1 * w) + 63) >> 6) : 0;
                 int64_t nwords = (w == 0) ? 0 : (W1 - W0 + 1);
                 uint64_t* dst = (uint64_t*)(stage_base + lds_off);
@@ -798,6 +798,7 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
                 lds_off += ((nwords * 8) + 15) & ~(int64_t)15;
             } else {
                 col_words[u] = (uint64_t*)(sg.blob + e.off_values_words);  /* global */
+                col_w0[u] = 0;
             }
             /* null bitmap staged only for columns that HAVE nulls
              * (parse-time fact); null-free columns skip the checks */
@@ -853,7 +854,7 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
                 atomicAdd((unsigned long long*)&th->side_cnt[side], 1ULL);
                 if (pp.sum_slot >= 0 && !val_null) {
                     uint64_t pv = bp_get_win(col_words[pp.val_idx],
-                                             col_w[pp.val_idx], j, 0);
+                                             col_w[pp.val_idx], j, col_w0[pp.val_idx]);
                     uint64_t val = (uint64_t)zz_dec(sv->min_value + pv);
                     atomicAdd((unsigned long long*)&th->side_agg[side][2 * pp.sum_slot], val);
                     atomicAdd((unsigned long long*)&th->side_agg[side][2 * pp.sum_slot + 1], 1ULL);
@@ -904,7 +905,7 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
                     uint64_t rec = kzz;
                     if (pp.val_idx >= 0) {
                         uint64_t pv = bp_get_win(col_words[pp.val_idx],
-                                                 col_w[pp.val_idx], j, 0);
+                                                 col_w[pp.val_idx], j, col_w0[pp.val_idx]);
                         uint64_t vzz = (sv->min_value + pv) - pp.gmin_v;
                         rec |= vzz << pp.bits_k;
                     }
@@ -912,9 +913,8 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
                 } else {
                     uint64_t val = 0;
                     if (pp.val_idx >= 0) {
-                        /* value read straight from global (not staged) */
                         uint64_t pv = bp_get_win(col_words[pp.val_idx],
-                                                 col_w[pp.val_idx], j, 0);
+                                                 col_w[pp.val_idx], j, col_w0[pp.val_idx]);
                         val = (uint64_t)zz_dec(sv->min_value + pv);
                     }
                     recs[(int64_t)b * pp.bucket_stride + gbase[b] + row_off[i]] =
