@@ -132,6 +132,8 @@ struct FastParams {
     int64_t row_count;
     int32_t nsegs_per_col;
     int32_t ntiles;
+    int32_t stage_bm_mask;       /* bit u: used col u has nulls — stage+check */
+    int32_t pad2_;
 };
 
 /* two-phase partitioned group-by (BASELINE configs 3/4 family):

@@ -927,6 +927,10 @@ static int run_scan(const YtPlan* plan, const YtChunk* chunk,
         fp.filter_hi = fs->filter_hi;
         fp.row_count = chunk->row_count;
         fp.nsegs_per_col = chunk->columns[0].segment_count;
+        fp.stage_bm_mask = 0;
+        for (int u = 0; u < nused; u++) {
+            if (R->col_null_flags[used[u]] != 0) fp.stage_bm_mask |= 1 << u;
+        }
 
         int32_t seg0_rows = chunk->columns[0].segments[0].row_count;
 

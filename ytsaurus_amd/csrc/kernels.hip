@@ -608,15 +608,19 @@ k_scan_fast(FastParams fp, const DevSeg* segs, const SegEx* segex,
             }
             lds_off += ((nwords * 8) + 15) & ~(int64_t)15;
 
-            /* stage null bitmap slice: bytes for rows [t0, t1) */
-            uint8_t* bdst = (uint8_t*)(smem + lds_off);
-            col_bitmap[u] = bdst;
-            const uint8_t* bsrc = (const uint8_t*)s.blob + e.off_bitmap_bytes + (t0 >> 3);
-            int64_t bbytes = ((t1 - t0) + 7) / 8;
-            for (int64_t i = tid; i < bbytes; i += 256) {
-                bdst[i] = bsrc[i];
+            /* stage null bitmap slice only for nullable columns */
+            if ((fp.stage_bm_mask >> u) & 1) {
+                uint8_t* bdst = (uint8_t*)(smem + lds_off);
+                col_bitmap[u] = bdst;
+                const uint8_t* bsrc = (const uint8_t*)s.blob + e.off_bitmap_bytes + (t0 >> 3);
+                int64_t bbytes = ((t1 - t0) + 7) / 8;
+                for (int64_t i = tid; i < bbytes; i += 256) {
+                    bdst[i] = bsrc[i];
+                }
+                lds_off += (bbytes + 15) & ~(int64_t)15;
+            } else {
+                col_bitmap[u] = nullptr;
             }
-            lds_off += (bbytes + 15) & ~(int64_t)15;
         }
         __syncthreads();
 
@@ -638,7 +642,7 @@ k_scan_fast(FastParams fp, const DevSeg* segs, const SegEx* segex,
             /* filter */
             if (fp.filter_idx >= 0) {
                 int u = fp.filter_idx;
-                if (bm_get(col_bitmap[u], j - t0)) continue;   /* null → range false */
+                if (col_bitmap[u] && bm_get(col_bitmap[u], j - t0)) continue;
                 uint64_t packed = bp_get_win(col_words[u], col_w[u], j, col_w0[u]);
                 const DevSeg& sf = segs[cols[u].seg_off + seg_idx];
                 int64_t v = zz_dec(sf.min_value + packed);
@@ -649,7 +653,7 @@ k_scan_fast(FastParams fp, const DevSeg* segs, const SegEx* segex,
             DVal key;
             if (fp.key_idx >= 0) {
                 int u = fp.key_idx;
-                if (bm_get(col_bitmap[u], j - t0)) {
+                if (col_bitmap[u] && bm_get(col_bitmap[u], j - t0)) {
                     key.null_ = 1; key.bits = 0; key.type = YT_VT_INT64;
                 } else {
                     uint64_t packed = bp_get_win(col_words[u], col_w[u], j, col_w0[u]);
@@ -661,9 +665,11 @@ k_scan_fast(FastParams fp, const DevSeg* segs, const SegEx* segex,
 
             if (fp.key_idx < 0) {
                 acc_cnt++;
-                for (int a = 0; a < fp.nsum; a++) {
+                #pragma unroll
+                for (int a = 0; a < kMaxAggs; a++) {
+                    if (a >= fp.nsum) break;
                     int u = fp.sum_idx[a];
-                    if (bm_get(col_bitmap[u], j - t0)) continue;
+                    if (col_bitmap[u] && bm_get(col_bitmap[u], j - t0)) continue;
                     uint64_t packed = bp_get_win(col_words[u], col_w[u], j, col_w0[u]);
                     const DevSeg& sv = segs[cols[u].seg_off + seg_idx];
                     acc_sum[a] += (uint64_t)zz_dec(sv.min_value + packed);
@@ -688,7 +694,7 @@ k_scan_fast(FastParams fp, const DevSeg* segs, const SegEx* segex,
                 atomicAdd(cntp, 1ULL);
                 for (int a = 0; a < fp.nsum; a++) {
                     int u = fp.sum_idx[a];
-                    if (bm_get(col_bitmap[u], j - t0)) continue;
+                    if (col_bitmap[u] && bm_get(col_bitmap[u], j - t0)) continue;
                     uint64_t packed = bp_get_win(col_words[u], col_w[u], j, col_w0[u]);
                     const DevSeg& sv = segs[cols[u].seg_off + seg_idx];
                     uint64_t v = (uint64_t)zz_dec(sv.min_value + packed);
