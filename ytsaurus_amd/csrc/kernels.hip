@@ -581,12 +581,14 @@ k_scan_fast(FastParams fp, const DevSeg* segs, const SegEx* segex,
         const uint8_t* col_bitmap[6];
         uint32_t col_w[6];
         int64_t col_w0[6];    /* first staged word index (absolute in segment values) */
+        uint64_t col_min[6];  /* per-tile min_value, hoisted out of the row loop */
         int32_t seg_rows = 0;
 
         for (int u = 0; u < fp.nused; u++) {
             const DevSeg& s = segs[cols[u].seg_off + seg_idx];
             const SegEx& e = segex[cols[u].seg_off + seg_idx];
             seg_rows = s.row_count;
+            col_min[u] = s.min_value;
             int64_t t1 = t0 + fp.tile_rows;
             if (t1 > seg_rows) t1 = seg_rows;
             uint32_t w = e.w_values;
@@ -644,8 +646,7 @@ k_scan_fast(FastParams fp, const DevSeg* segs, const SegEx* segex,
                 int u = fp.filter_idx;
                 if (col_bitmap[u] && bm_get(col_bitmap[u], j - t0)) continue;
                 uint64_t packed = bp_get_win(col_words[u], col_w[u], j, col_w0[u]);
-                const DevSeg& sf = segs[cols[u].seg_off + seg_idx];
-                int64_t v = zz_dec(sf.min_value + packed);
+                int64_t v = zz_dec(col_min[u] + packed);
                 if (v < fp.filter_lo || v > fp.filter_hi) continue;
             }
 
@@ -657,8 +658,7 @@ k_scan_fast(FastParams fp, const DevSeg* segs, const SegEx* segex,
                     key.null_ = 1; key.bits = 0; key.type = YT_VT_INT64;
                 } else {
                     uint64_t packed = bp_get_win(col_words[u], col_w[u], j, col_w0[u]);
-                    const DevSeg& sk = segs[cols[u].seg_off + seg_idx];
-                    key.bits = (uint64_t)zz_dec(sk.min_value + packed);
+                    key.bits = (uint64_t)zz_dec(col_min[u] + packed);
                     key.null_ = 0; key.type = YT_VT_INT64;
                 }
             }
@@ -671,8 +671,7 @@ k_scan_fast(FastParams fp, const DevSeg* segs, const SegEx* segex,
                     int u = fp.sum_idx[a];
                     if (col_bitmap[u] && bm_get(col_bitmap[u], j - t0)) continue;
                     uint64_t packed = bp_get_win(col_words[u], col_w[u], j, col_w0[u]);
-                    const DevSeg& sv = segs[cols[u].seg_off + seg_idx];
-                    acc_sum[a] += (uint64_t)zz_dec(sv.min_value + packed);
+                    acc_sum[a] += (uint64_t)zz_dec(col_min[u] + packed);
                     acc_nn[a]++;
                 }
             } else {
@@ -696,8 +695,7 @@ k_scan_fast(FastParams fp, const DevSeg* segs, const SegEx* segex,
                     int u = fp.sum_idx[a];
                     if (col_bitmap[u] && bm_get(col_bitmap[u], j - t0)) continue;
                     uint64_t packed = bp_get_win(col_words[u], col_w[u], j, col_w0[u]);
-                    const DevSeg& sv = segs[cols[u].seg_off + seg_idx];
-                    uint64_t v = (uint64_t)zz_dec(sv.min_value + packed);
+                    uint64_t v = (uint64_t)zz_dec(col_min[u] + packed);
                     unsigned long long* ap = aggp + 2 * fp.sum_slot[a];
                     atomicAdd(ap, (unsigned long long)v);
                     atomicAdd(ap + 1, 1ULL);
@@ -780,6 +778,7 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
         const uint8_t* col_bitmap[4];
         uint32_t col_w[4];
         int64_t col_w0[4];
+        uint64_t col_min[4];
         int32_t seg_rows = segs[cols[0].seg_off + seg_idx].row_count;
         int64_t t1 = t0 + pp.tile_rows;
         if (t1 > seg_rows) t1 = seg_rows;
@@ -787,6 +786,7 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
         for (int u = 0; u < pp.nused; u++) {
             const DevSeg& sg = segs[cols[u].seg_off + seg_idx];
             const SegEx& e = segex[cols[u].seg_off + seg_idx];
+            col_min[u] = sg.min_value;
             uint32_t w = e.w_values;
             col_w[u] = w;
             int64_t W0 = ((uint64_t)t0 * w) >> 6;
@@ -822,8 +822,8 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
         for (int i = tid; i < kNB; i += 256) { hist[i] = 0; nhist[i] = 0; }
         __syncthreads();
 
-        const DevSeg& sk = segs[cols[pp.key_idx].seg_off + seg_idx];
-        const DevSeg* sv = pp.val_idx >= 0 ? &segs[cols[pp.val_idx].seg_off + seg_idx] : nullptr;
+        const uint64_t kmin = col_min[pp.key_idx];
+        const uint64_t vmin = pp.val_idx >= 0 ? col_min[pp.val_idx] : 0;
 
         const int R = (pp.tile_rows + 255) / 256;
         uint32_t row_b[32];
@@ -837,8 +837,7 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
                 int u = pp.filter_idx;
                 if (pp.has_filter_nulls && bm_get(col_bitmap[u], j - t0)) continue;
                 uint64_t packed = bp_get_win(col_words[u], col_w[u], j, col_w0[u]);
-                const DevSeg& sf = segs[cols[u].seg_off + seg_idx];
-                int64_t v = zz_dec(sf.min_value + packed);
+                int64_t v = zz_dec(col_min[u] + packed);
                 if (v < pp.filter_lo || v > pp.filter_hi) continue;
             }
 
@@ -847,7 +846,7 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
             if (!key_null) {
                 uint64_t packed = bp_get_win(col_words[pp.key_idx], col_w[pp.key_idx], j,
                                              col_w0[pp.key_idx]);
-                key = (uint64_t)zz_dec(sk.min_value + packed);
+                key = (uint64_t)zz_dec(kmin + packed);
             }
             int val_null = 1;
             if (pp.val_idx >= 0) {
@@ -861,7 +860,7 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
                 if (pp.sum_slot >= 0 && !val_null) {
                     uint64_t pv = bp_get_win(col_words[pp.val_idx],
                                              col_w[pp.val_idx], j, col_w0[pp.val_idx]);
-                    uint64_t val = (uint64_t)zz_dec(sv->min_value + pv);
+                    uint64_t val = (uint64_t)zz_dec(vmin + pv);
                     atomicAdd((unsigned long long*)&th->side_agg[side][2 * pp.sum_slot], val);
                     atomicAdd((unsigned long long*)&th->side_agg[side][2 * pp.sum_slot + 1], 1ULL);
                 }
@@ -900,19 +899,19 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
                 int64_t j = t0 + (int64_t)i * 256 + tid;
                 uint64_t packed = bp_get_win(col_words[pp.key_idx], col_w[pp.key_idx], j,
                                              col_w0[pp.key_idx]);
-                uint64_t key = (uint64_t)zz_dec(sk.min_value + packed);
+                uint64_t key = (uint64_t)zz_dec(kmin + packed);
                 unsigned b = row_b[i] & 0x7FFFFFFFu;
                 if (row_b[i] & 0x80000000u) {
                     nrecs[(int64_t)b * pp.nbucket_stride + ngbase[b] + row_off[i]] = key;
                 } else if (pp.packed_mode) {
-                    uint64_t kzz = (sk.min_value +
+                    uint64_t kzz = (kmin +
                         bp_get_win(col_words[pp.key_idx], col_w[pp.key_idx], j,
                                    col_w0[pp.key_idx])) - pp.gmin_k;
                     uint64_t rec = kzz;
                     if (pp.val_idx >= 0) {
                         uint64_t pv = bp_get_win(col_words[pp.val_idx],
                                                  col_w[pp.val_idx], j, col_w0[pp.val_idx]);
-                        uint64_t vzz = (sv->min_value + pv) - pp.gmin_v;
+                        uint64_t vzz = (vmin + pv) - pp.gmin_v;
                         rec |= vzz << pp.bits_k;
                     }
                     recs8[(int64_t)b * pp.bucket_stride + gbase[b] + row_off[i]] = rec;
@@ -921,7 +920,7 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
                     if (pp.val_idx >= 0) {
                         uint64_t pv = bp_get_win(col_words[pp.val_idx],
                                                  col_w[pp.val_idx], j, col_w0[pp.val_idx]);
-                        val = (uint64_t)zz_dec(sv->min_value + pv);
+                        val = (uint64_t)zz_dec(vmin + pv);
                     }
                     recs[(int64_t)b * pp.bucket_stride + gbase[b] + row_off[i]] =
                         make_ulonglong2(key, val);
