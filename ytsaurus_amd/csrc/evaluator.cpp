@@ -692,16 +692,15 @@ static int run_partitioned(const YtPlan* plan, const YtChunk* chunk,
 
     PartParams pp;
     memset(&pp, 0, sizeof(pp));
-    int used[4];
-    int nused = 0;
-    auto add_used = [&](int col) -> int {
-        for (int i = 0; i < nused; i++) if (used[i] == col) return i;
-        used[nused] = col;
-        return nused++;
-    };
-    pp.filter_idx = fs->filter_col >= 0 ? add_used(fs->filter_col) : -1;
-    pp.key_idx = add_used(fs->key_col);
-    pp.val_idx = fs->nsum == 1 ? add_used(fs->sum_col[0]) : -1;
+    /* canonical slots: 0 = filter, 1 = key, 2 = value (see kernel comment) */
+    int used[3];
+    used[0] = fs->filter_col >= 0 ? fs->filter_col : fs->key_col;
+    used[1] = fs->key_col;
+    used[2] = fs->nsum == 1 ? fs->sum_col[0] : fs->key_col;
+    int nused = 3;
+    pp.filter_idx = fs->filter_col >= 0 ? 0 : -1;
+    pp.key_idx = 1;
+    pp.val_idx = fs->nsum == 1 ? 2 : -1;
     pp.nused = nused;
     pp.sum_slot = fs->nsum == 1 ? fs->sum_slot[0] : -1;
     pp.agg_count = plan->agg_count;
@@ -753,16 +752,19 @@ static int run_partitioned(const YtPlan* plan, const YtChunk* chunk,
     }
     /* stage the value column too when the budget allows (avoids a dependent
      * global load per row in the write pass); otherwise read it from L2 */
+    int base_cols = (pp.filter_idx >= 0 ? 1 : 0) + 1;
+    int nbms = (pp.has_filter_nulls ? 1 : 0) + (pp.has_key_nulls ? 1 : 0)
+             + (pp.has_val_nulls ? 1 : 0);
     size_t lds = 0;
     for (;;) {
         size_t words = ((((size_t)tile_rows * w / 64 + 2) * 8 + 15) & ~(size_t)15);
         size_t bm = (((size_t)tile_rows / 8 + 15) & ~(size_t)15);
         pp.stage_val = 1;
-        lds = 4 * kNB * 4 + (size_t)nused * (words + bm) + 256;
+        lds = 4 * kNB * 4 + (size_t)(base_cols + (pp.val_idx >= 0 ? 1 : 0)) * words
+            + (size_t)nbms * bm + 256;
         if (lds <= 52 * 1024 || tile_rows == 256) break;
         pp.stage_val = 0;
-        lds = 4 * kNB * 4 + (size_t)(nused - (pp.val_idx >= 0 ? 1 : 0)) * words
-            + (size_t)nused * bm + 256;
+        lds = 4 * kNB * 4 + (size_t)base_cols * words + (size_t)nbms * bm + 256;
         if (lds <= 52 * 1024 || tile_rows == 256) break;
         tile_rows >>= 1;
     }

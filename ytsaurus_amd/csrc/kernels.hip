@@ -752,78 +752,93 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
                  unsigned long long* ncursors,       /* kNB null-value stream */
                  uint64_t* nrecs)
 {
+    /* canonical column slots (host builds `cols` in this order):
+     *   0 = filter column (staged iff filter_idx >= 0)
+     *   1 = key column    (always staged)
+     *   2 = value column  (staged iff stage_val)
+     * Constant slot indices keep every per-row access in plain registers —
+     * runtime-indexed local arrays compile to s_set_gpr_idx register
+     * indexing, which serialized the row loop (measured ~14x). */
     extern __shared__ __attribute__((aligned(16))) char smem[];
     const int tid = threadIdx.x;
 
-    /* LDS layout: hist[kNB] u32 | gbase[kNB] u32 | nhist[kNB] u32 |
-     *             ngbase[kNB] u32 | staged filter/key words + bitmaps |
-     *             value-column null bitmap.
-     * The value column's packed words are NOT staged: they are read once,
-     * directly from L2/HBM in the write pass (per-lane strided reads of a
-     * hot tile region are coalesced enough, and skipping the staging both
-     * halves LDS use and removes a full LDS round trip). */
     unsigned* hist = (unsigned*)smem;
     unsigned* gbase = hist + kNB;
     unsigned* nhist = gbase + kNB;
     unsigned* ngbase = nhist + kNB;
     char* stage_base = (char*)(ngbase + kNB);
 
+    const bool has_filter = pp.filter_idx >= 0;
+    const bool has_val = pp.val_idx >= 0;
+
     for (int tile = blockIdx.x; tile < pp.ntiles; tile += gridDim.x) {
         const int seg_idx = tile / pp.tiles_per_seg;
         const int tile_in_seg = tile % pp.tiles_per_seg;
         const int64_t t0 = (int64_t)tile_in_seg * pp.tile_rows;
 
-        int64_t lds_off = 0;
-        uint64_t* col_words[4];
-        const uint8_t* col_bitmap[4];
-        uint32_t col_w[4];
-        int64_t col_w0[4];
-        uint64_t col_min[4];
-        int32_t seg_rows = segs[cols[0].seg_off + seg_idx].row_count;
+        const int32_t seg_rows = segs[cols[1].seg_off + seg_idx].row_count;
         int64_t t1 = t0 + pp.tile_rows;
         if (t1 > seg_rows) t1 = seg_rows;
 
-        for (int u = 0; u < pp.nused; u++) {
-            const DevSeg& sg = segs[cols[u].seg_off + seg_idx];
-            const SegEx& e = segex[cols[u].seg_off + seg_idx];
-            col_min[u] = sg.min_value;
+        int64_t lds_off = 0;
+        /* stage one column's packed words; returns LDS base or global base */
+        auto stage_words = [&](int slot, bool in_lds, uint32_t* w_out,
+                               int64_t* w0_out) -> const uint64_t* {
+            const DevSeg& sg = segs[cols[slot].seg_off + seg_idx];
+            const SegEx& e = segex[cols[slot].seg_off + seg_idx];
             uint32_t w = e.w_values;
-            col_w[u] = w;
+            *w_out = w;
+            if (!in_lds) {
+                *w0_out = 0;
+                return (const uint64_t*)(sg.blob + e.off_values_words);
+            }
             int64_t W0 = ((uint64_t)t0 * w) >> 6;
-            col_w0[u] = W0;
-            if (u != pp.val_idx || pp.stage_val) {
-                int64_t W1 = w ? ((((uint64_t)t1 * w) + 63) >> 6) : 0;
-                int64_t nwords = (w == 0) ? 0 : (W1 - W0 + 1);
-                uint64_t* dst = (uint64_t*)(stage_base + lds_off);
-                col_words[u] = dst;
-                const uint64_t* src = sg.blob + e.off_values_words + W0;
-                int64_t vec_words = (w == 0) ? 0 : (((uint64_t)seg_rows * w + 63) >> 6);
-                int64_t avail = vec_words - W0;
-                if (nwords > avail) nwords = avail;
-                for (int64_t i = tid; i < nwords; i += 256) dst[i] = src[i];
-                lds_off += ((nwords * 8) + 15) & ~(int64_t)15;
-            } else {
-                col_words[u] = (uint64_t*)(sg.blob + e.off_values_words);  /* global */
-                col_w0[u] = 0;
-            }
-            /* null bitmap staged only for columns that HAVE nulls
-             * (parse-time fact); null-free columns skip the checks */
-            if ((pp.stage_bm_mask >> u) & 1) {
-                uint8_t* bdst = (uint8_t*)(stage_base + lds_off);
-                col_bitmap[u] = bdst;
-                const uint8_t* bsrc = (const uint8_t*)sg.blob + e.off_bitmap_bytes + (t0 >> 3);
-                int64_t bbytes = ((t1 - t0) + 7) / 8;
-                for (int64_t i = tid; i < bbytes; i += 256) bdst[i] = bsrc[i];
-                lds_off += (bbytes + 15) & ~(int64_t)15;
-            } else {
-                col_bitmap[u] = nullptr;
-            }
+            *w0_out = W0;
+            int64_t W1 = w ? ((((uint64_t)t1 * w) + 63) >> 6) : 0;
+            int64_t nwords = (w == 0) ? 0 : (W1 - W0 + 1);
+            uint64_t* dst = (uint64_t*)(stage_base + lds_off);
+            const uint64_t* src = sg.blob + e.off_values_words + W0;
+            int64_t vec_words = (w == 0) ? 0 : (((uint64_t)seg_rows * w + 63) >> 6);
+            int64_t avail = vec_words - W0;
+            if (nwords > avail) nwords = avail;
+            for (int64_t i = tid; i < nwords; i += 256) dst[i] = src[i];
+            lds_off += ((nwords * 8) + 15) & ~(int64_t)15;
+            return dst;
+        };
+        auto stage_bitmap = [&](int slot, bool want) -> const uint8_t* {
+            if (!want) return nullptr;
+            const DevSeg& sg = segs[cols[slot].seg_off + seg_idx];
+            const SegEx& e = segex[cols[slot].seg_off + seg_idx];
+            uint8_t* bdst = (uint8_t*)(stage_base + lds_off);
+            const uint8_t* bsrc = (const uint8_t*)sg.blob + e.off_bitmap_bytes + (t0 >> 3);
+            int64_t bbytes = ((t1 - t0) + 7) / 8;
+            for (int64_t i = tid; i < bbytes; i += 256) bdst[i] = bsrc[i];
+            lds_off += (bbytes + 15) & ~(int64_t)15;
+            return bdst;
+        };
+
+        uint32_t fw = 0, kw = 0, vw = 0;
+        int64_t fw0 = 0, kw0 = 0, vw0 = 0;
+        const uint64_t* fwords = nullptr;
+        const uint8_t* fbm = nullptr;
+        if (has_filter) {
+            fwords = stage_words(0, true, &fw, &fw0);
+            fbm = stage_bitmap(0, pp.has_filter_nulls);
         }
+        const uint64_t* kwords = stage_words(1, true, &kw, &kw0);
+        const uint8_t* kbm = stage_bitmap(1, pp.has_key_nulls);
+        const uint64_t* vwords = nullptr;
+        const uint8_t* vbm = nullptr;
+        if (has_val) {
+            vwords = stage_words(2, pp.stage_val != 0, &vw, &vw0);
+            vbm = stage_bitmap(2, pp.has_val_nulls);
+        }
+        const uint64_t kmin = segs[cols[1].seg_off + seg_idx].min_value;
+        const uint64_t vmin = has_val ? segs[cols[2].seg_off + seg_idx].min_value : 0;
+        const uint64_t fmin = has_filter ? segs[cols[0].seg_off + seg_idx].min_value : 0;
+
         for (int i = tid; i < kNB; i += 256) { hist[i] = 0; nhist[i] = 0; }
         __syncthreads();
-
-        const uint64_t kmin = col_min[pp.key_idx];
-        const uint64_t vmin = pp.val_idx >= 0 ? col_min[pp.val_idx] : 0;
 
         const int R = (pp.tile_rows + 255) / 256;
         uint32_t row_b[32];
@@ -833,34 +848,25 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
             int64_t j = t0 + (int64_t)i * 256 + tid;
             if (j >= t1) continue;
 
-            if (pp.filter_idx >= 0) {
-                int u = pp.filter_idx;
-                if (pp.has_filter_nulls && bm_get(col_bitmap[u], j - t0)) continue;
-                uint64_t packed = bp_get_win(col_words[u], col_w[u], j, col_w0[u]);
-                int64_t v = zz_dec(col_min[u] + packed);
+            if (has_filter) {
+                if (fbm && bm_get(fbm, j - t0)) continue;
+                int64_t v = zz_dec(fmin + bp_get_win(fwords, fw, j, fw0));
                 if (v < pp.filter_lo || v > pp.filter_hi) continue;
             }
 
-            uint64_t key;
-            int key_null = pp.has_key_nulls && bm_get(col_bitmap[pp.key_idx], j - t0);
+            uint64_t key = 0;
+            int key_null = kbm && bm_get(kbm, j - t0);
             if (!key_null) {
-                uint64_t packed = bp_get_win(col_words[pp.key_idx], col_w[pp.key_idx], j,
-                                             col_w0[pp.key_idx]);
-                key = (uint64_t)zz_dec(kmin + packed);
+                key = (uint64_t)zz_dec(kmin + bp_get_win(kwords, kw, j, kw0));
             }
-            int val_null = 1;
-            if (pp.val_idx >= 0) {
-                val_null = pp.has_val_nulls && bm_get(col_bitmap[pp.val_idx], j - t0);
-            }
+            int val_null = has_val ? (vbm && bm_get(vbm, j - t0)) : 1;
 
-            if (key_null || (!key_null && key == kEmptyKey)) {
+            if (key_null || key == kEmptyKey) {
                 int side = key_null ? 1 : 0;
                 th->side_used[side] = 1;
                 atomicAdd((unsigned long long*)&th->side_cnt[side], 1ULL);
                 if (pp.sum_slot >= 0 && !val_null) {
-                    uint64_t pv = bp_get_win(col_words[pp.val_idx],
-                                             col_w[pp.val_idx], j, col_w0[pp.val_idx]);
-                    uint64_t val = (uint64_t)zz_dec(vmin + pv);
+                    uint64_t val = (uint64_t)zz_dec(vmin + bp_get_win(vwords, vw, j, vw0));
                     atomicAdd((unsigned long long*)&th->side_agg[side][2 * pp.sum_slot], val);
                     atomicAdd((unsigned long long*)&th->side_agg[side][2 * pp.sum_slot + 1], 1ULL);
                 }
@@ -868,7 +874,7 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
             }
 
             unsigned b = (unsigned)(mix64(key) >> 40) & (kNB - 1);
-            if (pp.val_idx >= 0 && val_null) {
+            if (has_val && val_null) {
                 row_b[i] = b | 0x80000000u;
                 row_off[i] = atomicAdd(&nhist[b], 1u);
             } else {
@@ -876,7 +882,6 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
                 row_off[i] = atomicAdd(&hist[b], 1u);
             }
         }
-        uint64_t* recs8 = (uint64_t*)recs;
         __syncthreads();
         for (int i = tid; i < kNB; i += 256) {
             unsigned c = hist[i];
@@ -893,37 +898,30 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
             }
         }
         __syncthreads();
+        uint64_t* recs8 = (uint64_t*)recs;
         if (th->overflow != 1) {
             for (int i = 0; i < R; i++) {
                 if (row_b[i] == 0xFFFFFFFFu) continue;
                 int64_t j = t0 + (int64_t)i * 256 + tid;
-                uint64_t packed = bp_get_win(col_words[pp.key_idx], col_w[pp.key_idx], j,
-                                             col_w0[pp.key_idx]);
-                uint64_t key = (uint64_t)zz_dec(kmin + packed);
+                uint64_t kzzfull = kmin + bp_get_win(kwords, kw, j, kw0);
                 unsigned b = row_b[i] & 0x7FFFFFFFu;
                 if (row_b[i] & 0x80000000u) {
-                    nrecs[(int64_t)b * pp.nbucket_stride + ngbase[b] + row_off[i]] = key;
+                    nrecs[(int64_t)b * pp.nbucket_stride + ngbase[b] + row_off[i]] =
+                        (uint64_t)zz_dec(kzzfull);
                 } else if (pp.packed_mode) {
-                    uint64_t kzz = (kmin +
-                        bp_get_win(col_words[pp.key_idx], col_w[pp.key_idx], j,
-                                   col_w0[pp.key_idx])) - pp.gmin_k;
-                    uint64_t rec = kzz;
-                    if (pp.val_idx >= 0) {
-                        uint64_t pv = bp_get_win(col_words[pp.val_idx],
-                                                 col_w[pp.val_idx], j, col_w0[pp.val_idx]);
-                        uint64_t vzz = (vmin + pv) - pp.gmin_v;
+                    uint64_t rec = kzzfull - pp.gmin_k;
+                    if (has_val) {
+                        uint64_t vzz = (vmin + bp_get_win(vwords, vw, j, vw0)) - pp.gmin_v;
                         rec |= vzz << pp.bits_k;
                     }
                     recs8[(int64_t)b * pp.bucket_stride + gbase[b] + row_off[i]] = rec;
                 } else {
                     uint64_t val = 0;
-                    if (pp.val_idx >= 0) {
-                        uint64_t pv = bp_get_win(col_words[pp.val_idx],
-                                                 col_w[pp.val_idx], j, col_w0[pp.val_idx]);
-                        val = (uint64_t)zz_dec(vmin + pv);
+                    if (has_val) {
+                        val = (uint64_t)zz_dec(vmin + bp_get_win(vwords, vw, j, vw0));
                     }
                     recs[(int64_t)b * pp.bucket_stride + gbase[b] + row_off[i]] =
-                        make_ulonglong2(key, val);
+                        make_ulonglong2((uint64_t)zz_dec(kzzfull), val);
                 }
             }
         }
