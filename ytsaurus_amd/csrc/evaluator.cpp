@@ -905,22 +905,22 @@ static int run_scan(const YtPlan* plan, const YtChunk* chunk,
     HIP_CHECK(hipMemsetAsync(R->d_gaccum, 0, sizeof(uint64_t) * (1 + 2 * kMaxAggs), R->stream));
 
     if (fs->valid) {
-        /* fast fused kernel */
+        /* fast fused kernel — canonical slots: 0 = filter, 1..4 = sum args,
+         * 5 = key (see k_scan_fast) */
         FastParams fp;
         memset(&fp, 0, sizeof(fp));
-        /* used columns: filter first (if distinct), then key, then sums */
-        int used[kMaxCols];
-        int nused = 0;
-        auto add_used = [&](int col) -> int {
-            for (int i = 0; i < nused; i++) if (used[i] == col) return i;
-            used[nused] = col;
-            return nused++;
-        };
-        fp.filter_idx = fs->filter_col >= 0 ? add_used(fs->filter_col) : -1;
-        fp.key_idx = fs->key_col >= 0 ? add_used(fs->key_col) : -1;
+        int used[6];
+        used[0] = fs->filter_col >= 0 ? fs->filter_col : 0;
+        for (int a = 0; a < kMaxAggs; a++) {
+            used[1 + a] = (a < fs->nsum) ? fs->sum_col[a] : 0;
+        }
+        used[5] = fs->key_col >= 0 ? fs->key_col : 0;
+        int nused = 6;
+        fp.filter_idx = fs->filter_col >= 0 ? 0 : -1;
+        fp.key_idx = fs->key_col >= 0 ? 5 : -1;
         fp.nsum = fs->nsum;
         for (int a = 0; a < fs->nsum; a++) {
-            fp.sum_idx[a] = add_used(fs->sum_col[a]);
+            fp.sum_idx[a] = 1 + a;
             fp.sum_slot[a] = fs->sum_slot[a];
         }
         fp.nused = nused;
@@ -930,26 +930,29 @@ static int run_scan(const YtPlan* plan, const YtChunk* chunk,
         fp.row_count = chunk->row_count;
         fp.nsegs_per_col = chunk->columns[0].segment_count;
         fp.stage_bm_mask = 0;
-        for (int u = 0; u < nused; u++) {
-            if (R->col_null_flags[used[u]] != 0) fp.stage_bm_mask |= 1 << u;
+        if (fs->filter_col >= 0 && R->col_null_flags[fs->filter_col]) fp.stage_bm_mask |= 1;
+        for (int a = 0; a < fs->nsum; a++) {
+            if (R->col_null_flags[fs->sum_col[a]]) fp.stage_bm_mask |= 1 << (1 + a);
         }
+        if (fs->key_col >= 0 && R->col_null_flags[fs->key_col]) fp.stage_bm_mask |= 1 << 5;
 
         int32_t seg0_rows = chunk->columns[0].segments[0].row_count;
 
-        /* LDS sizing: per col align16((tile_rows*w/64+2)*8) + align16(tile_rows/8) */
+        /* LDS sizing over the ACTUALLY staged arrays */
         unsigned w = maxw ? maxw : 1;
+        int staged = (fs->filter_col >= 0 ? 1 : 0) + fs->nsum + (fs->key_col >= 0 ? 1 : 0);
+        int nbms = __builtin_popcount((unsigned)fp.stage_bm_mask);
         int tile_rows = 8192;
         size_t lds = 0;
         for (;;) {
-            size_t per = ((((size_t)tile_rows * w / 64 + 2) * 8 + 15) & ~(size_t)15)
-                       + (((size_t)tile_rows / 8 + 15) & ~(size_t)15);
-            lds = (size_t)nused * per + 1024;
-            if (lds <= 50 * 1024 || tile_rows == 256) break;
+            size_t words = ((((size_t)tile_rows * w / 64 + 2) * 8 + 15) & ~(size_t)15);
+            size_t bm = (((size_t)tile_rows / 8 + 15) & ~(size_t)15);
+            lds = (size_t)staged * words + (size_t)nbms * bm + 1024;
+            if (lds <= 52 * 1024 || tile_rows == 256) break;
             tile_rows >>= 1;
         }
         if (lds > 160 * 1024) { set_err(errbuf, errlen, "LDS overflow"); rc = YT_ERR_UNSUPPORTED; goto fail; }
         if (tile_rows > seg0_rows) {
-            /* keep at least one tile per segment */
             while (tile_rows > 256 && tile_rows > seg0_rows) tile_rows >>= 1;
         }
         fp.tile_rows = tile_rows;

@@ -566,116 +566,119 @@ k_scan_fast(FastParams fp, const DevSeg* segs, const SegEx* segex,
             TableHdr* th, unsigned long long* slots,
             unsigned long long* gaccum)
 {
+    /* canonical column slots (host builds `cols` in this order):
+     *   0       = filter column (used iff filter_idx >= 0)
+     *   1..4    = sum arguments (first nsum valid; rest duplicate slot 1)
+     *   5       = key column (direct-table fallback path; key_idx >= 0)
+     * Constant indices only — runtime-indexed register arrays serialize
+     * (s_set_gpr_idx), measured ~14x on this loop. */
     extern __shared__ __attribute__((aligned(16))) char smem[];
-
     const int tid = threadIdx.x;
+    const bool has_filter = fp.filter_idx >= 0;
+    const bool has_key = fp.key_idx >= 0;
 
     for (int tile = blockIdx.x; tile < fp.ntiles; tile += gridDim.x) {
         const int seg_idx = tile / fp.tiles_per_seg;
         const int tile_in_seg = tile % fp.tiles_per_seg;
         const int64_t t0 = (int64_t)tile_in_seg * fp.tile_rows;
+        const int32_t seg_rows = segs[cols[1].seg_off + seg_idx].row_count;
+        int64_t t1 = t0 + fp.tile_rows;
+        if (t1 > seg_rows) t1 = seg_rows;
 
-        /* ---- stage all used columns for this tile into LDS ---- */
         int64_t lds_off = 0;
-        uint64_t* col_words[6];
-        const uint8_t* col_bitmap[6];
-        uint32_t col_w[6];
-        int64_t col_w0[6];    /* first staged word index (absolute in segment values) */
-        uint64_t col_min[6];  /* per-tile min_value, hoisted out of the row loop */
-        int32_t seg_rows = 0;
-
-        for (int u = 0; u < fp.nused; u++) {
-            const DevSeg& s = segs[cols[u].seg_off + seg_idx];
-            const SegEx& e = segex[cols[u].seg_off + seg_idx];
-            seg_rows = s.row_count;
-            col_min[u] = s.min_value;
-            int64_t t1 = t0 + fp.tile_rows;
-            if (t1 > seg_rows) t1 = seg_rows;
+        auto stage_words = [&](int slot, uint32_t* w_out, int64_t* w0_out)
+            -> const uint64_t* {
+            const DevSeg& sg = segs[cols[slot].seg_off + seg_idx];
+            const SegEx& e = segex[cols[slot].seg_off + seg_idx];
             uint32_t w = e.w_values;
-            col_w[u] = w;
+            *w_out = w;
             int64_t W0 = ((uint64_t)t0 * w) >> 6;
+            *w0_out = W0;
             int64_t W1 = w ? ((((uint64_t)t1 * w) + 63) >> 6) : 0;
-            if (W1 > W0 + 1 || w == 0) { /* at least cover straddle */ }
             int64_t nwords = (w == 0) ? 0 : (W1 - W0 + 1);
-            col_w0[u] = W0;
             uint64_t* dst = (uint64_t*)(smem + lds_off);
-            col_words[u] = dst;
-            const uint64_t* src = s.blob + e.off_values_words + W0;
-            /* clamp to the vector's actual words to avoid OOB on last word */
+            const uint64_t* src = sg.blob + e.off_values_words + W0;
             int64_t vec_words = (w == 0) ? 0 : (((uint64_t)seg_rows * w + 63) >> 6);
             int64_t avail = vec_words - W0;
             if (nwords > avail) nwords = avail;
-            for (int64_t i = tid; i < nwords; i += 256) {
-                dst[i] = src[i];
-            }
+            for (int64_t i = tid; i < nwords; i += 256) dst[i] = src[i];
             lds_off += ((nwords * 8) + 15) & ~(int64_t)15;
+            return dst;
+        };
+        auto stage_bitmap = [&](int slot, bool want) -> const uint8_t* {
+            if (!want) return nullptr;
+            const DevSeg& sg = segs[cols[slot].seg_off + seg_idx];
+            const SegEx& e = segex[cols[slot].seg_off + seg_idx];
+            uint8_t* bdst = (uint8_t*)(smem + lds_off);
+            const uint8_t* bsrc = (const uint8_t*)sg.blob + e.off_bitmap_bytes + (t0 >> 3);
+            int64_t bbytes = ((t1 - t0) + 7) / 8;
+            for (int64_t i = tid; i < bbytes; i += 256) bdst[i] = bsrc[i];
+            lds_off += (bbytes + 15) & ~(int64_t)15;
+            return bdst;
+        };
 
-            /* stage null bitmap slice only for nullable columns */
-            if ((fp.stage_bm_mask >> u) & 1) {
-                uint8_t* bdst = (uint8_t*)(smem + lds_off);
-                col_bitmap[u] = bdst;
-                const uint8_t* bsrc = (const uint8_t*)s.blob + e.off_bitmap_bytes + (t0 >> 3);
-                int64_t bbytes = ((t1 - t0) + 7) / 8;
-                for (int64_t i = tid; i < bbytes; i += 256) {
-                    bdst[i] = bsrc[i];
-                }
-                lds_off += (bbytes + 15) & ~(int64_t)15;
-            } else {
-                col_bitmap[u] = nullptr;
-            }
+        uint32_t fw = 0, kw = 0, sw_[4] = {0, 0, 0, 0};
+        int64_t fw0 = 0, kw0 = 0, sw0_[4] = {0, 0, 0, 0};
+        const uint64_t* fwords = nullptr;
+        const uint64_t* kwords = nullptr;
+        const uint64_t* swords_[4] = {nullptr, nullptr, nullptr, nullptr};
+        const uint8_t* fbm = nullptr;
+        const uint8_t* kbm = nullptr;
+        const uint8_t* sbm_[4] = {nullptr, nullptr, nullptr, nullptr};
+        uint64_t fmin = 0, kmin = 0, smin_[4] = {0, 0, 0, 0};
+
+        if (has_filter) {
+            fwords = stage_words(0, &fw, &fw0);
+            fbm = stage_bitmap(0, (fp.stage_bm_mask >> 0) & 1);
+            fmin = segs[cols[0].seg_off + seg_idx].min_value;
+        }
+        #pragma unroll
+        for (int a = 0; a < kMaxAggs; a++) {
+            if (a >= fp.nsum) break;
+            swords_[a] = stage_words(1 + a, &sw_[a], &sw0_[a]);
+            sbm_[a] = stage_bitmap(1 + a, (fp.stage_bm_mask >> (1 + a)) & 1);
+            smin_[a] = segs[cols[1 + a].seg_off + seg_idx].min_value;
+        }
+        if (has_key) {
+            kwords = stage_words(5, &kw, &kw0);
+            kbm = stage_bitmap(5, (fp.stage_bm_mask >> 5) & 1);
+            kmin = segs[cols[5].seg_off + seg_idx].min_value;
         }
         __syncthreads();
 
-        const DevSeg& s0 = segs[cols[0].seg_off + seg_idx];
-        int64_t t1 = t0 + fp.tile_rows;
-        if (t1 > seg_rows) t1 = seg_rows;
-        const int64_t chunk_row0 = s0.start_row;
-
-        /* per-thread accumulators for the global-agg path */
         uint64_t acc_sum[kMaxAggs] = {0, 0, 0, 0};
         uint64_t acc_nn[kMaxAggs] = {0, 0, 0, 0};
         uint64_t acc_cnt = 0;
 
         const int R = (fp.tile_rows + 255) / 256;
         for (int i = 0; i < R; i++) {
-            int64_t j = t0 + (int64_t)i * 256 + tid;   /* row in segment */
+            int64_t j = t0 + (int64_t)i * 256 + tid;
             if (j >= t1) break;
 
-            /* filter */
-            if (fp.filter_idx >= 0) {
-                int u = fp.filter_idx;
-                if (col_bitmap[u] && bm_get(col_bitmap[u], j - t0)) continue;
-                uint64_t packed = bp_get_win(col_words[u], col_w[u], j, col_w0[u]);
-                int64_t v = zz_dec(col_min[u] + packed);
+            if (has_filter) {
+                if (fbm && bm_get(fbm, j - t0)) continue;
+                int64_t v = zz_dec(fmin + bp_get_win(fwords, fw, j, fw0));
                 if (v < fp.filter_lo || v > fp.filter_hi) continue;
             }
 
-            /* key */
-            DVal key;
-            if (fp.key_idx >= 0) {
-                int u = fp.key_idx;
-                if (col_bitmap[u] && bm_get(col_bitmap[u], j - t0)) {
-                    key.null_ = 1; key.bits = 0; key.type = YT_VT_INT64;
-                } else {
-                    uint64_t packed = bp_get_win(col_words[u], col_w[u], j, col_w0[u]);
-                    key.bits = (uint64_t)zz_dec(col_min[u] + packed);
-                    key.null_ = 0; key.type = YT_VT_INT64;
-                }
-            }
-
-            if (fp.key_idx < 0) {
+            if (!has_key) {
                 acc_cnt++;
                 #pragma unroll
                 for (int a = 0; a < kMaxAggs; a++) {
                     if (a >= fp.nsum) break;
-                    int u = fp.sum_idx[a];
-                    if (col_bitmap[u] && bm_get(col_bitmap[u], j - t0)) continue;
-                    uint64_t packed = bp_get_win(col_words[u], col_w[u], j, col_w0[u]);
-                    acc_sum[a] += (uint64_t)zz_dec(col_min[u] + packed);
+                    if (sbm_[a] && bm_get(sbm_[a], j - t0)) continue;
+                    acc_sum[a] += (uint64_t)zz_dec(smin_[a] + bp_get_win(swords_[a], sw_[a], j, sw0_[a]));
                     acc_nn[a]++;
                 }
             } else {
-                /* group path: direct table update */
+                /* direct-table fallback path */
+                DVal key;
+                if (kbm && bm_get(kbm, j - t0)) {
+                    key.null_ = 1; key.bits = 0; key.type = YT_VT_INT64;
+                } else {
+                    key.bits = (uint64_t)zz_dec(kmin + bp_get_win(kwords, kw, j, kw0));
+                    key.null_ = 0; key.type = YT_VT_INT64;
+                }
                 unsigned long long* cntp;
                 unsigned long long* aggp;
                 int stride = 2 + 2 * fp.agg_count;
@@ -691,11 +694,11 @@ k_scan_fast(FastParams fp, const DevSeg* segs, const SegEx* segex,
                     aggp = slot + 2;
                 }
                 atomicAdd(cntp, 1ULL);
-                for (int a = 0; a < fp.nsum; a++) {
-                    int u = fp.sum_idx[a];
-                    if (col_bitmap[u] && bm_get(col_bitmap[u], j - t0)) continue;
-                    uint64_t packed = bp_get_win(col_words[u], col_w[u], j, col_w0[u]);
-                    uint64_t v = (uint64_t)zz_dec(col_min[u] + packed);
+                #pragma unroll
+                for (int a = 0; a < kMaxAggs; a++) {
+                    if (a >= fp.nsum) break;
+                    if (sbm_[a] && bm_get(sbm_[a], j - t0)) continue;
+                    uint64_t v = (uint64_t)zz_dec(smin_[a] + bp_get_win(swords_[a], sw_[a], j, sw0_[a]));
                     unsigned long long* ap = aggp + 2 * fp.sum_slot[a];
                     atomicAdd(ap, (unsigned long long)v);
                     atomicAdd(ap + 1, 1ULL);
@@ -704,7 +707,7 @@ k_scan_fast(FastParams fp, const DevSeg* segs, const SegEx* segex,
         }
 
         /* global-agg reduction: wave shfl → LDS → one atomic per WG */
-        if (fp.key_idx < 0) {
+        if (!has_key) {
             __syncthreads();   /* LDS reuse */
             uint64_t* red = (uint64_t*)smem;
             const int lane = tid & 63;
@@ -737,7 +740,6 @@ k_scan_fast(FastParams fp, const DevSeg* segs, const SegEx* segex,
         } else {
             __syncthreads();   /* before next tile overwrites LDS */
         }
-        (void)chunk_row0;
     }
 }
 
