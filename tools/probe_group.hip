@@ -465,6 +465,65 @@ k_bucket_agg(const ulonglong2* inbuf, const unsigned long long* cursors,
     }
 }
 
+/* V12: V0-style decode-only but with RUNTIME widths + branchless funnel
+ * extraction (mask hoisted per tile) — isolates the cost of runtime vs
+ * compile-time widths. */
+__device__ __forceinline__ uint64_t bp_get_bl(const uint64_t* win, uint64_t mask,
+                                              uint32_t width, uint64_t index, uint64_t w0)
+{
+    uint64_t bit = index * width;
+    uint64_t wi = (bit >> 6) - w0;
+    unsigned off = (unsigned)(bit & 63);
+    uint64_t lo = win[wi] >> off;
+    uint64_t hi = off ? (win[wi + 1] << (64 - off)) : 0;
+    return (lo | hi) & mask;
+}
+
+__global__ void __launch_bounds__(256)
+k_scan_rtw(const uint64_t* kwords, const uint64_t* vwords, int64_t rows,
+           uint32_t wk, uint32_t wv, int branchless,
+           unsigned long long* out)
+{
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    const int tid = threadIdx.x;
+    const int64_t ntiles = (rows + TILE - 1) / TILE;
+    uint64_t acc = 0;
+    const uint64_t kmask = (wk >= 64) ? ~0ULL : ((1ULL << wk) - 1);
+    const uint64_t vmask = (wv >= 64) ? ~0ULL : ((1ULL << wv) - 1);
+
+    for (int64_t tile = blockIdx.x; tile < ntiles; tile += gridDim.x) {
+        const int64_t t0 = tile * TILE;
+        int64_t t1 = t0 + TILE < rows ? t0 + TILE : rows;
+        int64_t kW0 = (uint64_t)t0 * wk >> 6;
+        int64_t kW1 = (((uint64_t)t1 * wk) + 63) >> 6;
+        uint64_t* klds = (uint64_t*)smem;
+        int64_t knw = kW1 - kW0 + 1;
+        for (int64_t i = tid; i < knw; i += 256) klds[i] = kwords[kW0 + i];
+        int64_t vW0 = (uint64_t)t0 * wv >> 6;
+        int64_t vW1 = (((uint64_t)t1 * wv) + 63) >> 6;
+        uint64_t* vlds = klds + knw + 2;
+        int64_t vnw = vW1 - vW0 + 1;
+        for (int64_t i = tid; i < vnw; i += 256) vlds[i] = vwords[vW0 + i];
+        __syncthreads();
+        const int R = TILE / 256;
+        for (int i = 0; i < R; i++) {
+            int64_t j = t0 + (int64_t)i * 256 + tid;
+            if (j >= t1) break;
+            uint64_t key, val;
+            if (branchless) {
+                key = bp_get_bl(klds, kmask, wk, j, kW0);
+                val = bp_get_bl(vlds, vmask, wv, j, vW0);
+            } else {
+                key = bp_get_win(klds, wk, j, kW0);
+                val = bp_get_win(vlds, wv, j, vW0);
+            }
+            acc += key + val;
+        }
+        __syncthreads();
+    }
+    if (acc) atomicAdd(out, acc);
+}
+
 int main(int argc, char** argv)
 {
     int64_t rows = (argc > 1 ? atoll(argv[1]) : 200) * 1000000LL;
@@ -629,5 +688,30 @@ int main(int argc, char** argv)
         CHECK(hipFree(outbuf));
         CHECK(hipFree(cursors));
     }
+    /* V12: runtime-width decode */
+    {
+        int64_t ntiles = (rows + TILE - 1) / TILE;
+        int grid = ntiles < 2048 ? (int)ntiles : 2048;
+        size_t lds = ((size_t)TILE * WK / 64 + 2 + (size_t)TILE * WV / 64 + 4) * 8 + 64;
+        hipEvent_t e0, e1;
+        CHECK(hipEventCreate(&e0));
+        CHECK(hipEventCreate(&e1));
+        for (int bl = 0; bl <= 1; bl++) {
+            hipLaunchKernelGGL(k_scan_rtw, dim3(grid), dim3(256), lds, 0,
+                               kw, vw, rows, WK, WV, bl, out);
+            CHECK(hipDeviceSynchronize());
+            CHECK(hipEventRecord(e0));
+            hipLaunchKernelGGL(k_scan_rtw, dim3(grid), dim3(256), lds, 0,
+                               kw, vw, rows, WK, WV, bl, out);
+            CHECK(hipEventRecord(e1));
+            CHECK(hipEventSynchronize(e1));
+            float ms;
+            CHECK(hipEventElapsedTime(&ms, e0, e1));
+            printf("V12 rtw decode bl=%d    %8.3f ms  %8.1f GB/s  %8.2f Grows/s\n",
+                   bl, ms, gb / ms * 1000, rows / ms / 1e6);
+        }
+    }
     return 0;
 }
+
+
