@@ -123,6 +123,24 @@ __device__ __forceinline__ void stage_copy(uint64_t* dst, const uint64_t* src,
         dst[base + 1536] = r6;
         dst[base + 1792] = r7;
     }
+    if (base + 256 * 3 < nwords) {
+        uint64_t r0 = src[base];
+        uint64_t r1 = src[base + 256];
+        uint64_t r2 = src[base + 512];
+        uint64_t r3 = src[base + 768];
+        dst[base] = r0;
+        dst[base + 256] = r1;
+        dst[base + 512] = r2;
+        dst[base + 768] = r3;
+        base += 256 * 4;
+    }
+    if (base + 256 < nwords) {
+        uint64_t r0 = src[base];
+        uint64_t r1 = src[base + 256];
+        dst[base] = r0;
+        dst[base + 256] = r1;
+        base += 256 * 2;
+    }
     for (; base < nwords; base += 256) dst[base] = src[base];
 }
 
