@@ -524,6 +524,36 @@ k_scan_rtw(const uint64_t* kwords, const uint64_t* vwords, int64_t rows,
     if (acc) atomicAdd(out, acc);
 }
 
+
+/* V13: decode straight from GLOBAL (no LDS staging): per-value 1-2 loads,
+ * L1 absorbs the window overlap, unrolled rows give MLP. */
+__global__ void __launch_bounds__(256)
+k_scan_noLDS(const uint64_t* kwords, const uint64_t* vwords, int64_t rows,
+             uint32_t wk, uint32_t wv, unsigned long long* out)
+{
+    uint64_t acc = 0;
+    const uint64_t kmask = (wk >= 64) ? ~0ULL : ((1ULL << wk) - 1);
+    const uint64_t vmask = (wv >= 64) ? ~0ULL : ((1ULL << wv) - 1);
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    int64_t i0 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    for (int64_t j = i0; j + 3 * stride < rows; j += 4 * stride) {
+        #pragma unroll 4
+        for (int u = 0; u < 4; u++) {
+            int64_t jj = j + u * stride;
+            uint64_t kb = jj * wk;
+            uint64_t lo = kwords[kb >> 6] >> (kb & 63);
+            uint64_t hi = (kb & 63) ? (kwords[(kb >> 6) + 1] << (64 - (kb & 63))) : 0;
+            uint64_t key = (lo | hi) & kmask;
+            uint64_t vb = jj * wv;
+            uint64_t lo2 = vwords[vb >> 6] >> (vb & 63);
+            uint64_t hi2 = (vb & 63) ? (vwords[(vb >> 6) + 1] << (64 - (vb & 63))) : 0;
+            uint64_t val = (lo2 | hi2) & vmask;
+            acc += key + val;
+        }
+    }
+    if (acc) atomicAdd(out, acc);
+}
+
 int main(int argc, char** argv)
 {
     int64_t rows = (argc > 1 ? atoll(argv[1]) : 200) * 1000000LL;
@@ -710,6 +740,22 @@ int main(int argc, char** argv)
             printf("V12 rtw decode bl=%d    %8.3f ms  %8.1f GB/s  %8.2f Grows/s\n",
                    bl, ms, gb / ms * 1000, rows / ms / 1e6);
         }
+    }
+    /* V13 */
+    {
+        hipEvent_t e0, e1;
+        CHECK(hipEventCreate(&e0));
+        CHECK(hipEventCreate(&e1));
+        hipLaunchKernelGGL(k_scan_noLDS, dim3(2048), dim3(256), 0, 0, kw, vw, rows, WK, WV, out);
+        CHECK(hipDeviceSynchronize());
+        CHECK(hipEventRecord(e0));
+        hipLaunchKernelGGL(k_scan_noLDS, dim3(2048), dim3(256), 0, 0, kw, vw, rows, WK, WV, out);
+        CHECK(hipEventRecord(e1));
+        CHECK(hipEventSynchronize(e1));
+        float ms;
+        CHECK(hipEventElapsedTime(&ms, e0, e1));
+        printf("V13 noLDS rtw decode    %8.3f ms  %8.1f GB/s  %8.2f Grows/s\n",
+               ms, gb / ms * 1000, rows / ms / 1e6);
     }
     return 0;
 }

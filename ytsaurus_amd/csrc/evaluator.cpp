@@ -938,20 +938,9 @@ static int run_scan(const YtPlan* plan, const YtChunk* chunk,
 
         int32_t seg0_rows = chunk->columns[0].segments[0].row_count;
 
-        /* LDS sizing over the ACTUALLY staged arrays */
-        unsigned w = maxw ? maxw : 1;
-        int staged = (fs->filter_col >= 0 ? 1 : 0) + fs->nsum + (fs->key_col >= 0 ? 1 : 0);
-        int nbms = __builtin_popcount((unsigned)fp.stage_bm_mask);
-        int tile_rows = 8192;
+        /* no LDS staging: tile size only shapes the grid */
         size_t lds = 0;
-        for (;;) {
-            size_t words = ((((size_t)tile_rows * w / 64 + 2) * 8 + 15) & ~(size_t)15);
-            size_t bm = (((size_t)tile_rows / 8 + 15) & ~(size_t)15);
-            lds = (size_t)staged * words + (size_t)nbms * bm + 1024;
-            if (lds <= 52 * 1024 || tile_rows == 256) break;
-            tile_rows >>= 1;
-        }
-        if (lds > 160 * 1024) { set_err(errbuf, errlen, "LDS overflow"); rc = YT_ERR_UNSUPPORTED; goto fail; }
+        int tile_rows = 8192;
         if (tile_rows > seg0_rows) {
             while (tile_rows > 256 && tile_rows > seg0_rows) tile_rows >>= 1;
         }

@@ -144,6 +144,18 @@ __device__ __forceinline__ void stage_copy(uint64_t* dst, const uint64_t* src,
     for (; base < nwords; base += 256) dst[base] = src[base];
 }
 
+/* branchless global-memory funnel extraction; safe to read one word past
+ * the packed vector (the segment's null bitmap follows it) */
+__device__ __forceinline__ uint64_t bp_gl(const uint64_t* base, uint64_t mask,
+                                          uint32_t width, uint64_t index)
+{
+    uint64_t bit = index * width;
+    unsigned off = (unsigned)(bit & 63);
+    uint64_t lo = base[bit >> 6] >> off;
+    uint64_t hi = off ? (base[(bit >> 6) + 1] << (64 - off)) : 0;
+    return (lo | hi) & mask;
+}
+
 /* ------------------------------------------------------------------ */
 /* segment header parsing (one thread per segment)                     */
 
@@ -614,16 +626,20 @@ k_scan_fast(FastParams fp, const DevSeg* segs, const SegEx* segex,
             TableHdr* th, unsigned long long* slots,
             unsigned long long* gaccum)
 {
-    /* canonical column slots (host builds `cols` in this order):
-     *   0       = filter column (used iff filter_idx >= 0)
-     *   1..4    = sum arguments (first nsum valid; rest duplicate slot 1)
-     *   5       = key column (direct-table fallback path; key_idx >= 0)
-     * Constant indices only — runtime-indexed register arrays serialize
-     * (s_set_gpr_idx), measured ~14x on this loop. */
-    extern __shared__ __attribute__((aligned(16))) char smem[];
+    /* No LDS staging: every value is decoded straight from global memory
+     * with a branchless two-word funnel; the L1/L2 absorb the window
+     * overlap between lanes and the unrolled row loop supplies enough
+     * loads in flight to reach streaming bandwidth (a staged-LDS version
+     * of this loop measured 83% WAIT_ANY — phases serialized).
+     * Canonical column slots: 0 = filter, 1..4 = sum args, 5 = key. */
+    __shared__ uint64_t red[64];
     const int tid = threadIdx.x;
     const bool has_filter = fp.filter_idx >= 0;
     const bool has_key = fp.key_idx >= 0;
+
+    uint64_t acc_sum[kMaxAggs] = {0, 0, 0, 0};
+    uint64_t acc_nn[kMaxAggs] = {0, 0, 0, 0};
+    uint64_t acc_cnt = 0;
 
     for (int tile = blockIdx.x; tile < fp.ntiles; tile += gridDim.x) {
         const int seg_idx = tile / fp.tiles_per_seg;
@@ -633,101 +649,62 @@ k_scan_fast(FastParams fp, const DevSeg* segs, const SegEx* segex,
         int64_t t1 = t0 + fp.tile_rows;
         if (t1 > seg_rows) t1 = seg_rows;
 
-        int64_t lds_off = 0;
-        auto stage_words = [&](int slot, uint32_t* w_out, int64_t* w0_out)
-            -> const uint64_t* {
-            const DevSeg& sg = segs[cols[slot].seg_off + seg_idx];
-            const SegEx& e = segex[cols[slot].seg_off + seg_idx];
-            uint32_t w = e.w_values;
-            *w_out = w;
-            int64_t W0 = ((uint64_t)t0 * w) >> 6;
-            *w0_out = W0;
-            int64_t W1 = w ? ((((uint64_t)t1 * w) + 63) >> 6) : 0;
-            int64_t nwords = (w == 0) ? 0 : (W1 - W0 + 1);
-            uint64_t* dst = (uint64_t*)(smem + lds_off);
-            const uint64_t* src = sg.blob + e.off_values_words + W0;
-            int64_t vec_words = (w == 0) ? 0 : (((uint64_t)seg_rows * w + 63) >> 6);
-            int64_t avail = vec_words - W0;
-            if (nwords > avail) nwords = avail;
-            stage_copy(dst, src, nwords, tid);
-            lds_off += ((nwords * 8) + 15) & ~(int64_t)15;
-            return dst;
-        };
-        auto stage_bitmap = [&](int slot, bool want) -> const uint8_t* {
-            if (!want) return nullptr;
-            const DevSeg& sg = segs[cols[slot].seg_off + seg_idx];
-            const SegEx& e = segex[cols[slot].seg_off + seg_idx];
-            uint8_t* bdst = (uint8_t*)(smem + lds_off);
-            const uint8_t* bsrc = (const uint8_t*)sg.blob + e.off_bitmap_bytes + (t0 >> 3);
-            int64_t bbytes = ((t1 - t0) + 7) / 8;
-            for (int64_t i = tid; i < bbytes; i += 256) bdst[i] = bsrc[i];
-            lds_off += (bbytes + 15) & ~(int64_t)15;
-            return bdst;
-        };
-
-        uint32_t fw = 0, kw = 0, sw_[4] = {0, 0, 0, 0};
-        int64_t fw0 = 0, kw0 = 0, sw0_[4] = {0, 0, 0, 0};
+        /* per-tile uniform column state, all constant-indexed */
         const uint64_t* fwords = nullptr;
-        const uint64_t* kwords = nullptr;
-        const uint64_t* swords_[4] = {nullptr, nullptr, nullptr, nullptr};
         const uint8_t* fbm = nullptr;
-        const uint8_t* kbm = nullptr;
-        const uint8_t* sbm_[4] = {nullptr, nullptr, nullptr, nullptr};
-        uint64_t fmin = 0, kmin = 0, smin_[4] = {0, 0, 0, 0};
-
+        uint32_t fwd = 0;
+        uint64_t fmask = 0, fmin = 0;
         if (has_filter) {
-            fwords = stage_words(0, &fw, &fw0);
-            fbm = stage_bitmap(0, (fp.stage_bm_mask >> 0) & 1);
-            fmin = segs[cols[0].seg_off + seg_idx].min_value;
+            const DevSeg& sg = segs[cols[0].seg_off + seg_idx];
+            const SegEx& e = segex[cols[0].seg_off + seg_idx];
+            fwords = sg.blob + e.off_values_words;
+            fwd = e.w_values;
+            fmask = (fwd >= 64) ? ~0ULL : ((1ULL << fwd) - 1);
+            fmin = sg.min_value;
+            if ((fp.stage_bm_mask >> 0) & 1)
+                fbm = (const uint8_t*)sg.blob + e.off_bitmap_bytes;
         }
+        const uint64_t* swords_[4] = {nullptr, nullptr, nullptr, nullptr};
+        const uint8_t* sbm_[4] = {nullptr, nullptr, nullptr, nullptr};
+        uint32_t swd_[4] = {0, 0, 0, 0};
+        uint64_t smask_[4] = {0, 0, 0, 0};
+        uint64_t smin_[4] = {0, 0, 0, 0};
         #pragma unroll
         for (int a = 0; a < kMaxAggs; a++) {
             if (a >= fp.nsum) break;
-            swords_[a] = stage_words(1 + a, &sw_[a], &sw0_[a]);
-            sbm_[a] = stage_bitmap(1 + a, (fp.stage_bm_mask >> (1 + a)) & 1);
-            smin_[a] = segs[cols[1 + a].seg_off + seg_idx].min_value;
+            const DevSeg& sg = segs[cols[1 + a].seg_off + seg_idx];
+            const SegEx& e = segex[cols[1 + a].seg_off + seg_idx];
+            swords_[a] = sg.blob + e.off_values_words;
+            swd_[a] = e.w_values;
+            smask_[a] = (swd_[a] >= 64) ? ~0ULL : ((1ULL << swd_[a]) - 1);
+            smin_[a] = sg.min_value;
+            if ((fp.stage_bm_mask >> (1 + a)) & 1)
+                sbm_[a] = (const uint8_t*)sg.blob + e.off_bitmap_bytes;
         }
+        const uint64_t* kwords = nullptr;
+        const uint8_t* kbm = nullptr;
+        uint32_t kwd = 0;
+        uint64_t kmask = 0, kmin = 0;
         if (has_key) {
-            kwords = stage_words(5, &kw, &kw0);
-            kbm = stage_bitmap(5, (fp.stage_bm_mask >> 5) & 1);
-            kmin = segs[cols[5].seg_off + seg_idx].min_value;
+            const DevSeg& sg = segs[cols[5].seg_off + seg_idx];
+            const SegEx& e = segex[cols[5].seg_off + seg_idx];
+            kwords = sg.blob + e.off_values_words;
+            kwd = e.w_values;
+            kmask = (kwd >= 64) ? ~0ULL : ((1ULL << kwd) - 1);
+            kmin = sg.min_value;
+            if ((fp.stage_bm_mask >> 5) & 1)
+                kbm = (const uint8_t*)sg.blob + e.off_bitmap_bytes;
         }
-        __syncthreads();
-
-        uint64_t acc_sum[kMaxAggs] = {0, 0, 0, 0};
-        uint64_t acc_nn[kMaxAggs] = {0, 0, 0, 0};
-        uint64_t acc_cnt = 0;
 
         const int R = (fp.tile_rows + 255) / 256;
-        const bool full_tile = (t1 - t0) == fp.tile_rows;
-        if (full_tile && !has_key) {
-            /* bounds-check-free unrolled body: lets the compiler overlap the
-             * per-row LDS-read chains across iterations */
-            #pragma unroll 4
-            for (int i = 0; i < R; i++) {
-                int64_t j = t0 + (int64_t)i * 256 + tid;
-                if (has_filter) {
-                    if (fbm && bm_get(fbm, j - t0)) continue;
-                    int64_t v = zz_dec(fmin + bp_get_win(fwords, fw, j, fw0));
-                    if (v < fp.filter_lo || v > fp.filter_hi) continue;
-                }
-                acc_cnt++;
-                #pragma unroll
-                for (int a = 0; a < kMaxAggs; a++) {
-                    if (a >= fp.nsum) break;
-                    if (sbm_[a] && bm_get(sbm_[a], j - t0)) continue;
-                    acc_sum[a] += (uint64_t)zz_dec(smin_[a] + bp_get_win(swords_[a], sw_[a], j, sw0_[a]));
-                    acc_nn[a]++;
-                }
-            }
-        } else
+        #pragma unroll 4
         for (int i = 0; i < R; i++) {
             int64_t j = t0 + (int64_t)i * 256 + tid;
-            if (j >= t1) break;
+            if (j >= t1) continue;
 
             if (has_filter) {
-                if (fbm && bm_get(fbm, j - t0)) continue;
-                int64_t v = zz_dec(fmin + bp_get_win(fwords, fw, j, fw0));
+                if (fbm && bm_get(fbm, j)) continue;
+                int64_t v = zz_dec(fmin + bp_gl(fwords, fmask, fwd, j));
                 if (v < fp.filter_lo || v > fp.filter_hi) continue;
             }
 
@@ -736,17 +713,16 @@ k_scan_fast(FastParams fp, const DevSeg* segs, const SegEx* segex,
                 #pragma unroll
                 for (int a = 0; a < kMaxAggs; a++) {
                     if (a >= fp.nsum) break;
-                    if (sbm_[a] && bm_get(sbm_[a], j - t0)) continue;
-                    acc_sum[a] += (uint64_t)zz_dec(smin_[a] + bp_get_win(swords_[a], sw_[a], j, sw0_[a]));
+                    if (sbm_[a] && bm_get(sbm_[a], j)) continue;
+                    acc_sum[a] += (uint64_t)zz_dec(smin_[a] + bp_gl(swords_[a], smask_[a], swd_[a], j));
                     acc_nn[a]++;
                 }
             } else {
-                /* direct-table fallback path */
                 DVal key;
-                if (kbm && bm_get(kbm, j - t0)) {
+                if (kbm && bm_get(kbm, j)) {
                     key.null_ = 1; key.bits = 0; key.type = YT_VT_INT64;
                 } else {
-                    key.bits = (uint64_t)zz_dec(kmin + bp_get_win(kwords, kw, j, kw0));
+                    key.bits = (uint64_t)zz_dec(kmin + bp_gl(kwords, kmask, kwd, j));
                     key.null_ = 0; key.type = YT_VT_INT64;
                 }
                 unsigned long long* cntp;
@@ -767,51 +743,48 @@ k_scan_fast(FastParams fp, const DevSeg* segs, const SegEx* segex,
                 #pragma unroll
                 for (int a = 0; a < kMaxAggs; a++) {
                     if (a >= fp.nsum) break;
-                    if (sbm_[a] && bm_get(sbm_[a], j - t0)) continue;
-                    uint64_t v = (uint64_t)zz_dec(smin_[a] + bp_get_win(swords_[a], sw_[a], j, sw0_[a]));
+                    if (sbm_[a] && bm_get(sbm_[a], j)) continue;
+                    uint64_t v = (uint64_t)zz_dec(smin_[a] + bp_gl(swords_[a], smask_[a], swd_[a], j));
                     unsigned long long* ap = aggp + 2 * fp.sum_slot[a];
                     atomicAdd(ap, (unsigned long long)v);
                     atomicAdd(ap + 1, 1ULL);
                 }
             }
         }
+    }
 
-        /* global-agg reduction: wave shfl → LDS → one atomic per WG */
-        if (!has_key) {
-            __syncthreads();   /* LDS reuse */
-            uint64_t* red = (uint64_t*)smem;
-            const int lane = tid & 63;
-            const int wave = tid >> 6;
+    /* one block-level reduction + one set of atomics per WORKGROUP (not per
+     * tile): accumulators persist across the block's tiles */
+    if (!has_key) {
+        const int lane = tid & 63;
+        const int wave = tid >> 6;
+        for (int a = 0; a < 2 * fp.nsum + 1; a++) {
+            uint64_t v = (a == 0) ? acc_cnt
+                       : (a & 1) ? acc_sum[a >> 1]
+                                 : acc_nn[(a >> 1) - 1];
+            for (int sh = 32; sh >= 1; sh >>= 1) {
+                v += (uint64_t)__shfl_down((long long)v, sh, 64);
+            }
+            if (lane == 0) red[wave * 16 + a] = v;
+        }
+        __syncthreads();
+        if (tid == 0) {
             for (int a = 0; a < 2 * fp.nsum + 1; a++) {
-                uint64_t v = (a == 0) ? acc_cnt
-                           : (a & 1) ? acc_sum[a >> 1]
-                                     : acc_nn[(a >> 1) - 1];
-                for (int sh = 32; sh >= 1; sh >>= 1) {
-                    v += (uint64_t)__shfl_down((long long)v, sh, 64);
-                }
-                if (lane == 0) red[wave * 16 + a] = v;
-            }
-            __syncthreads();
-            if (tid == 0) {
-                for (int a = 0; a < 2 * fp.nsum + 1; a++) {
-                    uint64_t v = red[a] + red[16 + a] + red[32 + a] + red[48 + a];
-                    if (a == 0) {
-                        atomicAdd(&gaccum[0], (unsigned long long)v);
-                    } else if (a & 1) {
-                        int slot = fp.sum_slot[a >> 1];
-                        atomicAdd(&gaccum[1 + 2 * slot], (unsigned long long)v);
-                    } else {
-                        int slot = fp.sum_slot[(a >> 1) - 1];
-                        atomicAdd(&gaccum[2 + 2 * slot], (unsigned long long)v);
-                    }
+                uint64_t v = red[a] + red[16 + a] + red[32 + a] + red[48 + a];
+                if (a == 0) {
+                    atomicAdd(&gaccum[0], (unsigned long long)v);
+                } else if (a & 1) {
+                    int slot = fp.sum_slot[a >> 1];
+                    atomicAdd(&gaccum[1 + 2 * slot], (unsigned long long)v);
+                } else {
+                    int slot = fp.sum_slot[(a >> 1) - 1];
+                    atomicAdd(&gaccum[2 + 2 * slot], (unsigned long long)v);
                 }
             }
-            __syncthreads();
-        } else {
-            __syncthreads();   /* before next tile overwrites LDS */
         }
     }
 }
+
 
 /* ------------------------------------------------------------------ */
 /* two-phase partitioned group-by (see common.h PartParams)            */
