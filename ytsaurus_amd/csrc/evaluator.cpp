@@ -750,25 +750,9 @@ static int run_partitioned(const YtPlan* plan, const YtChunk* chunk,
         const char* ev = getenv("YTQL_TILE");   /* perf experiments only */
         if (ev && atoi(ev) >= 256) tile_rows = atoi(ev);
     }
-    /* stage the value column too when the budget allows (avoids a dependent
-     * global load per row in the write pass); otherwise read it from L2 */
-    int base_cols = (pp.filter_idx >= 0 ? 1 : 0) + 1;
-    int nbms = (pp.has_filter_nulls ? 1 : 0) + (pp.has_key_nulls ? 1 : 0)
-             + (pp.has_val_nulls ? 1 : 0);
-    size_t lds = 0;
-    for (;;) {
-        size_t words = ((((size_t)tile_rows * w / 64 + 2) * 8 + 15) & ~(size_t)15);
-        size_t bm = (((size_t)tile_rows / 8 + 15) & ~(size_t)15);
-        pp.stage_val = 1;
-        lds = 4 * kNB * 4 + (size_t)(base_cols + (pp.val_idx >= 0 ? 1 : 0)) * words
-            + (size_t)nbms * bm + 256;
-        if (lds <= 52 * 1024 || tile_rows == 256) break;
-        pp.stage_val = 0;
-        lds = 4 * kNB * 4 + (size_t)base_cols * words + (size_t)nbms * bm + 256;
-        if (lds <= 52 * 1024 || tile_rows == 256) break;
-        tile_rows >>= 1;
-    }
-    if (lds > 160 * 1024) return YT_OK;   /* fall back */
+    /* no value staging: LDS holds only the per-tile bucket histograms */
+    (void)w;
+    size_t lds = 4 * kNB * 4 + 256;
     while (tile_rows > 256 && tile_rows > seg0_rows) tile_rows >>= 1;
     pp.tile_rows = tile_rows;
     pp.tiles_per_seg = (seg0_rows + tile_rows - 1) / tile_rows;

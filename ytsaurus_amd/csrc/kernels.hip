@@ -797,13 +797,10 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
                  unsigned long long* ncursors,       /* kNB null-value stream */
                  uint64_t* nrecs)
 {
-    /* canonical column slots (host builds `cols` in this order):
-     *   0 = filter column (staged iff filter_idx >= 0)
-     *   1 = key column    (always staged)
-     *   2 = value column  (staged iff stage_val)
-     * Constant slot indices keep every per-row access in plain registers —
-     * runtime-indexed local arrays compile to s_set_gpr_idx register
-     * indexing, which serialized the row loop (measured ~14x). */
+    /* Canonical column slots: 0 = filter, 1 = key, 2 = value.
+     * Values are decoded straight from global memory (branchless funnel,
+     * L1/L2 absorb window overlap); LDS holds only the per-tile bucket
+     * histograms. See k_scan_fast for the measured rationale. */
     extern __shared__ __attribute__((aligned(16))) char smem[];
     const int tid = threadIdx.x;
 
@@ -811,7 +808,6 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
     unsigned* gbase = hist + kNB;
     unsigned* nhist = gbase + kNB;
     unsigned* ngbase = nhist + kNB;
-    char* stage_base = (char*)(ngbase + kNB);
 
     const bool has_filter = pp.filter_idx >= 0;
     const bool has_val = pp.val_idx >= 0;
@@ -820,67 +816,46 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
         const int seg_idx = tile / pp.tiles_per_seg;
         const int tile_in_seg = tile % pp.tiles_per_seg;
         const int64_t t0 = (int64_t)tile_in_seg * pp.tile_rows;
-
         const int32_t seg_rows = segs[cols[1].seg_off + seg_idx].row_count;
         int64_t t1 = t0 + pp.tile_rows;
         if (t1 > seg_rows) t1 = seg_rows;
 
-        int64_t lds_off = 0;
-        /* stage one column's packed words; returns LDS base or global base */
-        auto stage_words = [&](int slot, bool in_lds, uint32_t* w_out,
-                               int64_t* w0_out) -> const uint64_t* {
-            const DevSeg& sg = segs[cols[slot].seg_off + seg_idx];
-            const SegEx& e = segex[cols[slot].seg_off + seg_idx];
-            uint32_t w = e.w_values;
-            *w_out = w;
-            if (!in_lds) {
-                *w0_out = 0;
-                return (const uint64_t*)(sg.blob + e.off_values_words);
-            }
-            int64_t W0 = ((uint64_t)t0 * w) >> 6;
-            *w0_out = W0;
-            int64_t W1 = w ? ((((uint64_t)t1 * w) + 63) >> 6) : 0;
-            int64_t nwords = (w == 0) ? 0 : (W1 - W0 + 1);
-            uint64_t* dst = (uint64_t*)(stage_base + lds_off);
-            const uint64_t* src = sg.blob + e.off_values_words + W0;
-            int64_t vec_words = (w == 0) ? 0 : (((uint64_t)seg_rows * w + 63) >> 6);
-            int64_t avail = vec_words - W0;
-            if (nwords > avail) nwords = avail;
-            stage_copy(dst, src, nwords, tid);
-            lds_off += ((nwords * 8) + 15) & ~(int64_t)15;
-            return dst;
-        };
-        auto stage_bitmap = [&](int slot, bool want) -> const uint8_t* {
-            if (!want) return nullptr;
-            const DevSeg& sg = segs[cols[slot].seg_off + seg_idx];
-            const SegEx& e = segex[cols[slot].seg_off + seg_idx];
-            uint8_t* bdst = (uint8_t*)(stage_base + lds_off);
-            const uint8_t* bsrc = (const uint8_t*)sg.blob + e.off_bitmap_bytes + (t0 >> 3);
-            int64_t bbytes = ((t1 - t0) + 7) / 8;
-            for (int64_t i = tid; i < bbytes; i += 256) bdst[i] = bsrc[i];
-            lds_off += (bbytes + 15) & ~(int64_t)15;
-            return bdst;
-        };
-
-        uint32_t fw = 0, kw = 0, vw = 0;
-        int64_t fw0 = 0, kw0 = 0, vw0 = 0;
         const uint64_t* fwords = nullptr;
         const uint8_t* fbm = nullptr;
+        uint32_t fwd = 0;
+        uint64_t fmask = 0, fmin = 0;
         if (has_filter) {
-            fwords = stage_words(0, true, &fw, &fw0);
-            fbm = stage_bitmap(0, pp.has_filter_nulls);
+            const DevSeg& sg = segs[cols[0].seg_off + seg_idx];
+            const SegEx& e = segex[cols[0].seg_off + seg_idx];
+            fwords = sg.blob + e.off_values_words;
+            fwd = e.w_values;
+            fmask = (fwd >= 64) ? ~0ULL : ((1ULL << fwd) - 1);
+            fmin = sg.min_value;
+            if (pp.has_filter_nulls)
+                fbm = (const uint8_t*)sg.blob + e.off_bitmap_bytes;
         }
-        const uint64_t* kwords = stage_words(1, true, &kw, &kw0);
-        const uint8_t* kbm = stage_bitmap(1, pp.has_key_nulls);
+        const DevSeg& sgk = segs[cols[1].seg_off + seg_idx];
+        const SegEx& ek = segex[cols[1].seg_off + seg_idx];
+        const uint64_t* kwords = sgk.blob + ek.off_values_words;
+        const uint32_t kwd = ek.w_values;
+        const uint64_t kmask = (kwd >= 64) ? ~0ULL : ((1ULL << kwd) - 1);
+        const uint64_t kmin = sgk.min_value;
+        const uint8_t* kbm = pp.has_key_nulls
+            ? (const uint8_t*)sgk.blob + ek.off_bitmap_bytes : nullptr;
         const uint64_t* vwords = nullptr;
         const uint8_t* vbm = nullptr;
+        uint32_t vwd = 0;
+        uint64_t vmask = 0, vmin = 0;
         if (has_val) {
-            vwords = stage_words(2, pp.stage_val != 0, &vw, &vw0);
-            vbm = stage_bitmap(2, pp.has_val_nulls);
+            const DevSeg& sg = segs[cols[2].seg_off + seg_idx];
+            const SegEx& e = segex[cols[2].seg_off + seg_idx];
+            vwords = sg.blob + e.off_values_words;
+            vwd = e.w_values;
+            vmask = (vwd >= 64) ? ~0ULL : ((1ULL << vwd) - 1);
+            vmin = sg.min_value;
+            if (pp.has_val_nulls)
+                vbm = (const uint8_t*)sg.blob + e.off_bitmap_bytes;
         }
-        const uint64_t kmin = segs[cols[1].seg_off + seg_idx].min_value;
-        const uint64_t vmin = has_val ? segs[cols[2].seg_off + seg_idx].min_value : 0;
-        const uint64_t fmin = has_filter ? segs[cols[0].seg_off + seg_idx].min_value : 0;
 
         for (int i = tid; i < kNB; i += 256) { hist[i] = 0; nhist[i] = 0; }
         __syncthreads();
@@ -888,30 +863,31 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
         const int R = (pp.tile_rows + 255) / 256;
         uint32_t row_b[32];
         uint32_t row_off[32];
+        #pragma unroll 4
         for (int i = 0; i < R; i++) {
             row_b[i] = 0xFFFFFFFFu;
             int64_t j = t0 + (int64_t)i * 256 + tid;
             if (j >= t1) continue;
 
             if (has_filter) {
-                if (fbm && bm_get(fbm, j - t0)) continue;
-                int64_t v = zz_dec(fmin + bp_get_win(fwords, fw, j, fw0));
+                if (fbm && bm_get(fbm, j)) continue;
+                int64_t v = zz_dec(fmin + bp_gl(fwords, fmask, fwd, j));
                 if (v < pp.filter_lo || v > pp.filter_hi) continue;
             }
 
             uint64_t key = 0;
-            int key_null = kbm && bm_get(kbm, j - t0);
+            int key_null = kbm && bm_get(kbm, j);
             if (!key_null) {
-                key = (uint64_t)zz_dec(kmin + bp_get_win(kwords, kw, j, kw0));
+                key = (uint64_t)zz_dec(kmin + bp_gl(kwords, kmask, kwd, j));
             }
-            int val_null = has_val ? (vbm && bm_get(vbm, j - t0)) : 1;
+            int val_null = has_val ? (vbm && bm_get(vbm, j)) : 1;
 
             if (key_null || key == kEmptyKey) {
                 int side = key_null ? 1 : 0;
                 th->side_used[side] = 1;
                 atomicAdd((unsigned long long*)&th->side_cnt[side], 1ULL);
                 if (pp.sum_slot >= 0 && !val_null) {
-                    uint64_t val = (uint64_t)zz_dec(vmin + bp_get_win(vwords, vw, j, vw0));
+                    uint64_t val = (uint64_t)zz_dec(vmin + bp_gl(vwords, vmask, vwd, j));
                     atomicAdd((unsigned long long*)&th->side_agg[side][2 * pp.sum_slot], val);
                     atomicAdd((unsigned long long*)&th->side_agg[side][2 * pp.sum_slot + 1], 1ULL);
                 }
@@ -945,10 +921,11 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
         __syncthreads();
         uint64_t* recs8 = (uint64_t*)recs;
         if (th->overflow != 1) {
+            #pragma unroll 4
             for (int i = 0; i < R; i++) {
                 if (row_b[i] == 0xFFFFFFFFu) continue;
                 int64_t j = t0 + (int64_t)i * 256 + tid;
-                uint64_t kzzfull = kmin + bp_get_win(kwords, kw, j, kw0);
+                uint64_t kzzfull = kmin + bp_gl(kwords, kmask, kwd, j);
                 unsigned b = row_b[i] & 0x7FFFFFFFu;
                 if (row_b[i] & 0x80000000u) {
                     nrecs[(int64_t)b * pp.nbucket_stride + ngbase[b] + row_off[i]] =
@@ -956,14 +933,14 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
                 } else if (pp.packed_mode) {
                     uint64_t rec = kzzfull - pp.gmin_k;
                     if (has_val) {
-                        uint64_t vzz = (vmin + bp_get_win(vwords, vw, j, vw0)) - pp.gmin_v;
+                        uint64_t vzz = (vmin + bp_gl(vwords, vmask, vwd, j)) - pp.gmin_v;
                         rec |= vzz << pp.bits_k;
                     }
                     recs8[(int64_t)b * pp.bucket_stride + gbase[b] + row_off[i]] = rec;
                 } else {
                     uint64_t val = 0;
                     if (has_val) {
-                        val = (uint64_t)zz_dec(vmin + bp_get_win(vwords, vw, j, vw0));
+                        val = (uint64_t)zz_dec(vmin + bp_gl(vwords, vmask, vwd, j));
                     }
                     recs[(int64_t)b * pp.bucket_stride + gbase[b] + row_off[i]] =
                         make_ulonglong2((uint64_t)zz_dec(kzzfull), val);
@@ -973,6 +950,7 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
         __syncthreads();
     }
 }
+
 
 /* Phase B: one workgroup per bucket. LDS table slot = {key, cnt|nonnull<<32
  * packed, sum}; per-workgroup counts stay < 2^32 because a bucket's rows do.
