@@ -745,14 +745,15 @@ static int run_partitioned(const YtPlan* plan, const YtChunk* chunk,
     /* LDS: 4*kNB u32 histogram structures + staged columns (value column's
      * packed words are NOT staged — read from global in the write pass) */
     unsigned w = maxw ? maxw : 1;
-    int tile_rows = 4096;
+    int tile_rows = 8192;
     {
         const char* ev = getenv("YTQL_TILE");   /* perf experiments only */
         if (ev && atoi(ev) >= 256) tile_rows = atoi(ev);
     }
-    /* no value staging: LDS holds only the per-tile bucket histograms */
-    (void)w;
-    size_t lds = 4 * kNB * 4 + 256;
+    if (tile_rows > 8192) tile_rows = 8192;   /* per-thread row arrays bound */
+    /* LDS: per-tile bucket histograms + the staged key column */
+    size_t lds = 4 * kNB * 4 + ((size_t)tile_rows * w / 64 + 2) * 8 + 256;
+    if (lds > 64 * 1024) { tile_rows = 4096; lds = 4 * kNB * 4 + ((size_t)tile_rows * w / 64 + 2) * 8 + 256; }
     while (tile_rows > 256 && tile_rows > seg0_rows) tile_rows >>= 1;
     pp.tile_rows = tile_rows;
     pp.tiles_per_seg = (seg0_rows + tile_rows - 1) / tile_rows;

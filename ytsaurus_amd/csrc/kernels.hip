@@ -808,6 +808,7 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
     unsigned* gbase = hist + kNB;
     unsigned* nhist = gbase + kNB;
     unsigned* ngbase = nhist + kNB;
+    uint64_t* klds = (uint64_t*)(ngbase + kNB);   /* staged key words */
 
     const bool has_filter = pp.filter_idx >= 0;
     const bool has_val = pp.val_idx >= 0;
@@ -834,14 +835,23 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
             if (pp.has_filter_nulls)
                 fbm = (const uint8_t*)sg.blob + e.off_bitmap_bytes;
         }
+        /* the key column is decoded in BOTH passes — stage it in LDS */
         const DevSeg& sgk = segs[cols[1].seg_off + seg_idx];
         const SegEx& ek = segex[cols[1].seg_off + seg_idx];
-        const uint64_t* kwords = sgk.blob + ek.off_values_words;
         const uint32_t kwd = ek.w_values;
         const uint64_t kmask = (kwd >= 64) ? ~0ULL : ((1ULL << kwd) - 1);
         const uint64_t kmin = sgk.min_value;
         const uint8_t* kbm = pp.has_key_nulls
             ? (const uint8_t*)sgk.blob + ek.off_bitmap_bytes : nullptr;
+        const int64_t kW0 = ((uint64_t)t0 * kwd) >> 6;
+        {
+            int64_t kW1 = kwd ? ((((uint64_t)t1 * kwd) + 63) >> 6) : 0;
+            int64_t nwords = (kwd == 0) ? 0 : (kW1 - kW0 + 1);
+            int64_t vec_words = (kwd == 0) ? 0 : (((uint64_t)seg_rows * kwd + 63) >> 6);
+            int64_t avail = vec_words - kW0;
+            if (nwords > avail) nwords = avail;
+            stage_copy(klds, sgk.blob + ek.off_values_words + kW0, nwords, tid);
+        }
         const uint64_t* vwords = nullptr;
         const uint8_t* vbm = nullptr;
         uint32_t vwd = 0;
@@ -878,7 +888,7 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
             uint64_t key = 0;
             int key_null = kbm && bm_get(kbm, j);
             if (!key_null) {
-                key = (uint64_t)zz_dec(kmin + bp_gl(kwords, kmask, kwd, j));
+                key = (uint64_t)zz_dec(kmin + (bp_get_win(klds, kwd, j, kW0) & kmask));
             }
             int val_null = has_val ? (vbm && bm_get(vbm, j)) : 1;
 
@@ -925,7 +935,7 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
             for (int i = 0; i < R; i++) {
                 if (row_b[i] == 0xFFFFFFFFu) continue;
                 int64_t j = t0 + (int64_t)i * 256 + tid;
-                uint64_t kzzfull = kmin + bp_gl(kwords, kmask, kwd, j);
+                uint64_t kzzfull = kmin + (bp_get_win(klds, kwd, j, kW0) & kmask);
                 unsigned b = row_b[i] & 0x7FFFFFFFu;
                 if (row_b[i] & 0x80000000u) {
                     nrecs[(int64_t)b * pp.nbucket_stride + ngbase[b] + row_off[i]] =
