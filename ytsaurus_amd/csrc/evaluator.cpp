@@ -660,15 +660,25 @@ static int setup_table(DeviceRun* R, int agg_count, int64_t max_groups,
     int rc = YT_OK;
     R->nslots = next_pow2((uint64_t)(max_groups > 0 ? max_groups : (1 << 20)) * 2);
     if (R->nslots < 2048) R->nslots = 2048;
-    int stride = 2 + 2 * agg_count;
     HIP_CHECK(pool_alloc(&R->d_th, sizeof(TableHdr)));
-    HIP_CHECK(pool_alloc(&R->d_slots, sizeof(uint64_t) * R->nslots * stride));
     TableHdr hh;
     memset(&hh, 0, sizeof(hh));
     hh.nslots = R->nslots;
     hh.mask = R->nslots - 1;
     hh.group_limit = group_limit;
     HIP_CHECK(hipMemcpyAsync(R->d_th, &hh, sizeof(hh), hipMemcpyHostToDevice, R->stream));
+    return YT_OK;
+fail:
+    return rc;
+}
+
+/* the direct-table slot array is only needed on the non-partitioned paths */
+static int ensure_slots(DeviceRun* R, int agg_count, char* errbuf, size_t errlen)
+{
+    int rc = YT_OK;
+    if (R->d_slots) return YT_OK;
+    int stride = 2 + 2 * agg_count;
+    HIP_CHECK(pool_alloc(&R->d_slots, sizeof(uint64_t) * R->nslots * stride));
     HIP_CHECK(hipMemsetAsync(R->d_slots, 0, sizeof(uint64_t) * R->nslots * stride, R->stream));
     return YT_OK;
 fail:
@@ -844,9 +854,6 @@ static int run_partitioned(const YtPlan* plan, const YtChunk* chunk,
             hh.mask = R->nslots - 1;
             hh.group_limit = options->group_row_limit;
             HIP_CHECK(hipMemcpyAsync(R->d_th, &hh, sizeof(hh), hipMemcpyHostToDevice, R->stream));
-            int stride = 2 + 2 * plan->agg_count;
-            HIP_CHECK(hipMemsetAsync(R->d_slots, 0,
-                                     sizeof(uint64_t) * R->nslots * stride, R->stream));
             *done = false;
         } else {
             R->groups_compacted = true;
@@ -882,6 +889,9 @@ static int run_scan(const YtPlan* plan, const YtChunk* chunk,
         if (done) return YT_OK;
         /* else: capacity guard tripped — fall through to the direct path */
     }
+
+    rc = ensure_slots(R, plan->agg_count, errbuf, errlen);
+    if (rc != YT_OK) return rc;
 
     HIP_CHECK(hipEventCreate(&ev0));
     HIP_CHECK(hipEventCreate(&ev1));
@@ -1367,6 +1377,8 @@ extern "C" int yt_gpu_merge_states(
     DeviceRun R;
     R.stream = (hipStream_t)(uintptr_t)options->stream;
     rc = setup_table(&R, plan->agg_count, options->max_groups_hint, 0, errbuf, errlen);
+    if (rc) return rc;
+    rc = ensure_slots(&R, plan->agg_count, errbuf, errlen);
     if (rc) return rc;
 
     HIP_CHECK(ytql_launch_merge_states((const YtStateRow*)states_device, state_row_count,
