@@ -174,7 +174,8 @@ def main():
     import ytsaurus_amd as y
 
     dist = None
-    if world > 1:
+    force_two = os.environ.get("YTQL_FORCE_TWOPHASE") == "1"
+    if world > 1 or (force_two and "RANK" in os.environ):
         import torch.distributed as dist_mod
         dist = dist_mod
         dist.init_process_group("nccl")
@@ -216,7 +217,7 @@ def main():
     out_rs = y.make_rowset(key_space + 1024, 1 + len(plan.aggs))
 
     # multi-GPU state buffers
-    if world > 1:
+    if dist is not None:
         cap = 2 * key_space + 1024
         states_t = torch.zeros((cap, 4), dtype=torch.int64, device="cuda")
         recv_t = torch.zeros((2 * cap, 4), dtype=torch.int64, device="cuda")
@@ -226,7 +227,7 @@ def main():
 
     def step():
         nonlocal scan_ms_total, scan_launches
-        if world == 1:
+        if dist is None:
             _, st = y.gpu_execute(plan, dev_chunk, max_groups_hint=hint,
                                   rowset=out_rs, raw_rowset=True)
             scan_ms_total += st.kernel_scan_ms
