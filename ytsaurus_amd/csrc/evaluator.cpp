@@ -772,7 +772,8 @@ static int run_partitioned(const YtPlan* plan, const YtChunk* chunk,
         int32_t last_rows = chunk->columns[0].segments[nseg - 1].row_count;
         pp.ntiles = (nseg - 1) * pp.tiles_per_seg + (last_rows + tile_rows - 1) / tile_rows;
     }
-    pp.bucket_stride = rows / kNB + (rows / kNB) / 2 + 8192;
+    /* per (bucket, XCD) sub-streams: 8x more cursors, 1/8 the rows each */
+    pp.bucket_stride = rows / (kNB * 8) + (rows / (kNB * 8)) / 2 + 4096;
     pp.nbucket_stride = pp.has_val_nulls ? pp.bucket_stride : 0;
 
     std::vector<FastCol> fc(nused);
@@ -785,14 +786,14 @@ static int run_partitioned(const YtPlan* plan, const YtChunk* chunk,
         HIP_CHECK(pool_alloc(&R->d_fastcols, sizeof(FastCol) * nused));
         HIP_CHECK(hipMemcpyAsync(R->d_fastcols, fc.data(), sizeof(FastCol) * nused,
                                  hipMemcpyHostToDevice, R->stream));
-        HIP_CHECK(pool_alloc(&R->d_cursors, sizeof(uint64_t) * kNB));
-        HIP_CHECK(hipMemsetAsync(R->d_cursors, 0, sizeof(uint64_t) * kNB, R->stream));
+        HIP_CHECK(pool_alloc(&R->d_cursors, sizeof(uint64_t) * kNB * 8));
+        HIP_CHECK(hipMemsetAsync(R->d_cursors, 0, sizeof(uint64_t) * kNB * 8, R->stream));
         HIP_CHECK(pool_alloc(&R->d_recs,
-                             (size_t)kNB * pp.bucket_stride * (pp.packed_mode ? 8 : 16)));
+                             (size_t)kNB * 8 * pp.bucket_stride * (pp.packed_mode ? 8 : 16)));
         if (pp.has_val_nulls) {
-            HIP_CHECK(pool_alloc(&R->d_ncursors, sizeof(uint64_t) * kNB));
-            HIP_CHECK(hipMemsetAsync(R->d_ncursors, 0, sizeof(uint64_t) * kNB, R->stream));
-            HIP_CHECK(pool_alloc(&R->d_nrecs, (size_t)kNB * pp.nbucket_stride * 8));
+            HIP_CHECK(pool_alloc(&R->d_ncursors, sizeof(uint64_t) * kNB * 8));
+            HIP_CHECK(hipMemsetAsync(R->d_ncursors, 0, sizeof(uint64_t) * kNB * 8, R->stream));
+            HIP_CHECK(pool_alloc(&R->d_nrecs, (size_t)kNB * 8 * pp.nbucket_stride * 8));
         }
         /* the partitioned path's in-table sentinel is INT64_MIN bits */
         uint64_t sk = kEmptyKey;
