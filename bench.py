@@ -253,9 +253,9 @@ def main():
             recv_t[:total_recv].view(-1, 4), states_t[:sum(counts)].view(-1, 4),
             output_split_sizes=recv_split, input_split_sizes=send_split)
         # front query: merge + finalize
-        rows, mst = y.gpu_merge(plan, recv_t.data_ptr(), total_recv,
-                                max_groups_hint=hint,
-                                out_capacity=key_space + 1024)
+        _, mst = y.gpu_merge(plan, recv_t.data_ptr(), total_recv,
+                             max_groups_hint=hint, rowset=out_rs,
+                             raw_rowset=True)
         return st
 
     # warmup

@@ -190,6 +190,10 @@ typedef struct YtRowset {
 /* Library/device probe. Returns YT_OK when a gfx950 HIP device is usable. */
 int yt_gpu_available(char* errbuf, size_t errlen);
 
+/* Releases the library's cached device/pinned buffers (queries reuse large
+ * allocations across calls, mirroring the reference evaluator's pooling). */
+void yt_gpu_pool_trim(void);
+
 /* The evaluator seam (replaces IEvaluator::Run for this plan shape).
  * Chunk segment data pointers must be DEVICE pointers (HBM-resident);
  * output rowset buffers are HOST memory. Synchronous on `options->stream`. */

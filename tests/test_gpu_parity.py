@@ -217,3 +217,41 @@ def test_group_row_limit_incomplete(cuda):
     got, stats = y.gpu_execute(plan, dev, max_groups_hint=1 << 16,
                                group_row_limit=10)
     assert stats.incomplete_output == 1
+
+
+def test_double_sum_parity(cuda):
+    """sum over a DOUBLE column (generic kernel, fp64 atomics): within 1e-6
+    relative of the oracle — the bound the reference's own nondeterministic
+    cross-tablet merge order implies (north_star; test_evaluate.cpp:220-260
+    checks doubles at 1e-5 absolute)."""
+    rng = np.random.default_rng(11)
+    n = 200_000
+    keys = rng.integers(0, 500, n, dtype=np.int64)
+    vals = rng.random(n)
+    vn = (rng.random(n) < 0.05).astype(np.uint8)
+    chunk = y.Chunk([y.encode_int64(keys), y.encode_double(vals, vn)], n)
+    plan = y.Plan(keys=[y.col(0)], aggs=[y.agg_sum(y.col(1)), y.agg_sum1()])
+    got, stats = y.gpu_execute(plan, chunk.c_device(cuda), max_groups_hint=2048)
+    want, _ = y.oracle_execute(plan, chunk)
+    gm = {r[0]: r for r in got}
+    assert len(got) == len(want)
+    for k, ws, wc in want:
+        gk, gs, gc = gm[k]
+        assert gc == wc
+        if ws is None:
+            assert gs is None
+        else:
+            assert abs(gs - ws) <= 1e-6 * max(abs(ws), 1e-30)
+
+
+def test_statistics_counters(cuda):
+    """TQueryStatistics parity: RowsRead / RowsWritten / GroupedRowCount
+    (query_statistics.h:49-79 semantics)."""
+    rng = np.random.default_rng(12)
+    n = 100_000
+    chunk = _mk(rng, n, 100, 0.0)
+    got, stats = y.gpu_execute(group_plan(), chunk.c_device(cuda), max_groups_hint=1024)
+    assert stats.rows_read == n
+    assert stats.rows_written == len(got)
+    assert stats.grouped_row_count == len(got)
+    assert stats.data_weight_read > 0

@@ -161,6 +161,7 @@ def gpu_lib():
                 "libytql_gpu.so not built; run python -c 'import __graft_entry__; __graft_entry__.build()'")
         lib = C.CDLL(path)
         _sig(lib, "yt_gpu_available", C.c_int, [C.c_char_p, C.c_size_t])
+        _sig(lib, "yt_gpu_pool_trim", None, [])
         _sig(lib, "yt_gpu_query_execute", C.c_int,
              [C.POINTER(YtPlan), C.POINTER(YtChunk), C.POINTER(YtExecOptions),
               C.POINTER(YtRowset), C.POINTER(YtStatistics), C.c_char_p, C.c_size_t])

@@ -399,14 +399,19 @@ def gpu_partial(plan, device_chunk, nparts, states_dev_ptr, capacity_rows,
 
 
 def gpu_merge(plan, states_dev_ptr, n_states, max_groups_hint=0, stream=0,
-              out_capacity=None):
+              out_capacity=None, rowset=None, raw_rowset=False):
     opts = YtExecOptions(max_groups_hint=max_groups_hint, stream=stream)
-    cap = out_capacity or max(n_states + 1024, 1 << 16)
-    rs = _mk_rowset(cap)
+    if rowset is not None:
+        rs = rowset
+    else:
+        cap = out_capacity or max(n_states + 1024, 1 << 16)
+        rs = _mk_rowset(cap)
     st = YtStatistics()
     err = C.create_string_buffer(512)
     rc = _abi.gpu_lib().yt_gpu_merge_states(
         C.byref(plan.c), C.c_void_p(states_dev_ptr), n_states, C.byref(opts),
         C.byref(rs), C.byref(st), err, 512)
     _check(rc, err)
+    if raw_rowset:
+        return rs, st
     return rows_from_rowset(rs), st
