@@ -798,24 +798,30 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
                  uint64_t* nrecs)
 {
     /* Canonical column slots: 0 = filter, 1 = key, 2 = value.
-     * Values are decoded straight from global memory (branchless funnel,
-     * L1/L2 absorb window overlap); LDS holds only the per-tile bucket
-     * histograms. See k_scan_fast for the measured rationale. */
+     * Records are appended to WORKGROUP-PRIVATE chunks of each bucket's
+     * region: one workgroup (one XCD) fills each 1-KiB chunk with full
+     * cache lines, so no line is ever part-written by two L2s and no
+     * partially-filled line lingers (the naive per-tile global reservation
+     * measured 5x HBM write amplification). Abandoned chunk tails are
+     * PAD-filled; Phase B skips PADs. */
     extern __shared__ __attribute__((aligned(16))) char smem[];
     const int tid = threadIdx.x;
 
-    unsigned* hist = (unsigned*)smem;
-    unsigned* gbase = hist + kNB;
+    unsigned* hist = (unsigned*)smem;                 /* per-tile counts */
+    unsigned* gbase = hist + kNB;                     /* per-tile write base (chunk-relative) */
     unsigned* nhist = gbase + kNB;
     unsigned* ngbase = nhist + kNB;
-    uint64_t* klds = (uint64_t*)(ngbase + kNB);   /* staged key words */
+    unsigned* cpos = ngbase + kNB;                    /* WG chunk cursor (records index) */
+    unsigned* crem = cpos + kNB;                      /* WG chunk remaining */
+    uint64_t* klds = (uint64_t*)(crem + kNB);         /* staged key words */
 
     const bool has_filter = pp.filter_idx >= 0;
     const bool has_val = pp.val_idx >= 0;
-    /* per-XCD sub-buckets: workgroups on different XCDs write disjoint
-     * record regions, so no cache line is filled by two L2s (cross-XCD
-     * partial-line sharing measured 3.7x write amplification) */
-    const int sub = blockIdx.x & 7;
+    const bool packed = pp.packed_mode != 0;
+    constexpr unsigned CHUNK = 128;                   /* records per reservation */
+
+    for (int i = tid; i < kNB; i += 256) { cpos[i] = 0; crem[i] = 0; }
+    __syncthreads();
 
     for (int tile = blockIdx.x; tile < pp.ntiles; tile += gridDim.x) {
         const int seg_idx = tile / pp.tiles_per_seg;
@@ -918,22 +924,42 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
             }
         }
         __syncthreads();
+        /* reserve space: serve this tile's counts from the WG's private
+         * chunk of each bucket; refill (and PAD the abandoned tail) when
+         * short. Null-stream rows are rare: direct global reservation. */
+        uint64_t* recs8 = (uint64_t*)recs;
         for (int i = tid; i < kNB; i += 256) {
             unsigned c = hist[i];
             if (c) {
-                unsigned long long base = atomicAdd(&cursors[i * 8 + sub], (unsigned long long)c);
-                if ((int64_t)(base + c) > pp.bucket_stride) { th->overflow = 1; base = 0; }
-                gbase[i] = (unsigned)base;
+                if (crem[i] < c) {
+                    /* pad abandoned tail of the old chunk */
+                    uint64_t base = (uint64_t)cpos[i];
+                    for (unsigned p = 0; p < crem[i]; p++) {
+                        if (packed) {
+                            recs8[(int64_t)i * pp.bucket_stride + base + p] = ~0ULL;
+                        } else {
+                            recs[(int64_t)i * pp.bucket_stride + base + p] =
+                                make_ulonglong2(kEmptyKey, 0);
+                        }
+                    }
+                    unsigned want = c > CHUNK ? c : CHUNK;
+                    unsigned long long nb = atomicAdd(&cursors[i], (unsigned long long)want);
+                    if ((int64_t)(nb + want) > pp.bucket_stride) { th->overflow = 1; nb = 0; }
+                    cpos[i] = (unsigned)nb;
+                    crem[i] = want;
+                }
+                gbase[i] = cpos[i];
+                cpos[i] += c;
+                crem[i] -= c;
             }
             unsigned nc = nhist[i];
             if (nc) {
-                unsigned long long base = atomicAdd(&ncursors[i * 8 + sub], (unsigned long long)nc);
+                unsigned long long base = atomicAdd(&ncursors[i], (unsigned long long)nc);
                 if ((int64_t)(base + nc) > pp.nbucket_stride) { th->overflow = 1; base = 0; }
                 ngbase[i] = (unsigned)base;
             }
         }
         __syncthreads();
-        uint64_t* recs8 = (uint64_t*)recs;
         if (th->overflow != 1) {
             #pragma unroll 4
             for (int i = 0; i < R; i++) {
@@ -941,23 +967,22 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
                 int64_t j = t0 + (int64_t)i * 256 + tid;
                 uint64_t kzzfull = kmin + (bp_get_win(klds, kwd, j, kW0) & kmask);
                 unsigned b = row_b[i] & 0x7FFFFFFFu;
-                int64_t sb = (int64_t)b * 8 + sub;
                 if (row_b[i] & 0x80000000u) {
-                    nrecs[sb * pp.nbucket_stride + ngbase[b] + row_off[i]] =
+                    nrecs[(int64_t)b * pp.nbucket_stride + ngbase[b] + row_off[i]] =
                         (uint64_t)zz_dec(kzzfull);
-                } else if (pp.packed_mode) {
+                } else if (packed) {
                     uint64_t rec = kzzfull - pp.gmin_k;
                     if (has_val) {
                         uint64_t vzz = (vmin + bp_gl(vwords, vmask, vwd, j)) - pp.gmin_v;
                         rec |= vzz << pp.bits_k;
                     }
-                    recs8[sb * pp.bucket_stride + gbase[b] + row_off[i]] = rec;
+                    recs8[(int64_t)b * pp.bucket_stride + gbase[b] + row_off[i]] = rec;
                 } else {
                     uint64_t val = 0;
                     if (has_val) {
                         val = (uint64_t)zz_dec(vmin + bp_gl(vwords, vmask, vwd, j));
                     }
-                    recs[sb * pp.bucket_stride + gbase[b] + row_off[i]] =
+                    recs[(int64_t)b * pp.bucket_stride + gbase[b] + row_off[i]] =
                         make_ulonglong2((uint64_t)zz_dec(kzzfull), val);
                 }
             }
@@ -992,18 +1017,21 @@ k_bucket_agg(const ulonglong2* recs, const unsigned long long* cursors,
 
     const uint64_t kmask = (bits_k >= 64) ? ~0ULL : ((1ULL << bits_k) - 1);
     bool full = false;
-    for (int sub = 0; sub < 8 && !full; sub++) {
-    int64_t n = (int64_t)cursors[bucket * 8 + sub];
-    const ulonglong2* rows = recs + ((int64_t)bucket * 8 + sub) * bucket_stride;
-    const uint64_t* rows8 = (const uint64_t*)recs + ((int64_t)bucket * 8 + sub) * bucket_stride;
+    {
+    int64_t n = (int64_t)cursors[bucket];
+    const ulonglong2* rows = recs + (int64_t)bucket * bucket_stride;
+    const uint64_t* rows8 = (const uint64_t*)recs + (int64_t)bucket * bucket_stride;
     /* 4 records per thread per pass: independent probes overlap LDS latency */
     int64_t i = tid;
     #define LOADKV(kv, idx)                                                  \
         ulonglong2 kv;                                                       \
         if (packed_mode) {                                                   \
             uint64_t r_ = rows8[idx];                                        \
-            kv.x = (uint64_t)zz_dec(gmin_k + (r_ & kmask));                  \
-            kv.y = (uint64_t)zz_dec(gmin_v + (r_ >> bits_k));                \
+            if (r_ == ~0ULL) { kv.x = kEmptyKey; kv.y = 0; }   /* PAD */     \
+            else {                                                           \
+                kv.x = (uint64_t)zz_dec(gmin_k + (r_ & kmask));              \
+                kv.y = (uint64_t)zz_dec(gmin_v + (r_ >> bits_k));            \
+            }                                                                \
         } else {                                                             \
             kv = rows[idx];                                                  \
         }
@@ -1017,7 +1045,7 @@ k_bucket_agg(const ulonglong2* recs, const unsigned long long* cursors,
         uint64_t s2 = mix64(kv2.x) & (kHSlots - 1);
         uint64_t s3 = mix64(kv3.x) & (kHSlots - 1);
         #define PROBE(kv, sv_)                                               \
-        {                                                                    \
+        if (kv.x != (uint64_t)kEmptyKey) {   /* PAD-skip */                  \
             uint64_t sp = sv_;                                               \
             int found = 0;                                                   \
             for (int it = 0; it < kHSlots; it++) {                           \
@@ -1049,13 +1077,13 @@ k_bucket_agg(const ulonglong2* recs, const unsigned long long* cursors,
         uint64_t s0 = mix64(kv.x) & (kHSlots - 1);
         PROBE(kv, s0)
     }
-    }   /* sub-stream loop */
+    }
     #undef PROBE
     #undef LOADKV
     if (nrecs) {
-        for (int sub = 0; sub < 8 && !full; sub++) {
-        int64_t nn = (int64_t)ncursors[bucket * 8 + sub];
-        const uint64_t* nrows = nrecs + ((int64_t)bucket * 8 + sub) * nbucket_stride;
+        {
+        int64_t nn = (int64_t)ncursors[bucket];
+        const uint64_t* nrows = nrecs + (int64_t)bucket * nbucket_stride;
         for (int64_t i = tid; i < nn && !full; i += 256) {
             uint64_t key = nrows[i];
             uint64_t s = mix64(key) & (kHSlots - 1);
@@ -1074,7 +1102,7 @@ k_bucket_agg(const ulonglong2* recs, const unsigned long long* cursors,
             if (!found) { full = true; break; }
             atomicAdd(&tab[s * 3 + 1], 1ULL);   /* cnt only; sum stays null-contributing */
         }
-        }   /* sub-stream loop */
+        }
     }
     if (full) th->overflow = 1;
     __syncthreads();
