@@ -989,6 +989,21 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
         }
         __syncthreads();
     }
+
+    /* PAD the unused tails of this workgroup's final chunks (reserved but
+     * unwritten positions would otherwise surface as garbage records) */
+    uint64_t* recs8f = (uint64_t*)recs;
+    for (int i = tid; i < kNB; i += 256) {
+        uint64_t base = (uint64_t)cpos[i];
+        for (unsigned p = 0; p < crem[i]; p++) {
+            if (pp.packed_mode) {
+                recs8f[(int64_t)i * pp.bucket_stride + base + p] = ~0ULL;
+            } else {
+                recs[(int64_t)i * pp.bucket_stride + base + p] =
+                    make_ulonglong2(kEmptyKey, 0);
+            }
+        }
+    }
 }
 
 
