@@ -772,9 +772,12 @@ static int run_partitioned(const YtPlan* plan, const YtChunk* chunk,
         int32_t last_rows = chunk->columns[0].segments[nseg - 1].row_count;
         pp.ntiles = (nseg - 1) * pp.tiles_per_seg + (last_rows + tile_rows - 1) / tile_rows;
     }
-    /* WG-private chunked regions: + grid*CHUNK slack per bucket for
-     * abandoned chunk tails (grid <= 2048, CHUNK = 128) */
-    pp.bucket_stride = rows / kNB + (rows / kNB) / 4 + 2048 * 128 / 8 + 8192;
+    /* WG-private chunked regions: every workgroup may strand one partial
+     * chunk per bucket, so the worst-case slack is grid * CHUNK records */
+    const int part_grid = 1024;
+    const int part_chunk = 64;
+    pp.bucket_stride = rows / kNB + (rows / kNB) / 4
+                     + (int64_t)part_grid * part_chunk + 8192;
     pp.nbucket_stride = pp.has_val_nulls ? pp.bucket_stride : 0;
 
     std::vector<FastCol> fc(nused);
@@ -815,7 +818,7 @@ static int run_partitioned(const YtPlan* plan, const YtChunk* chunk,
         HIP_CHECK(hipEventCreate(&ev0));
         HIP_CHECK(hipEventCreate(&ev1));
         HIP_CHECK(hipEventCreate(&ev2));
-        int grid = pp.ntiles < 2048 ? (pp.ntiles ? pp.ntiles : 1) : 2048;
+        int grid = pp.ntiles < part_grid ? (pp.ntiles ? pp.ntiles : 1) : part_grid;
         HIP_CHECK(hipEventRecord(ev0, R->stream));
         HIP_CHECK(ytql_launch_scan_partition(&pp, R->d_segs, R->d_segex, R->d_fastcols,
                                              R->d_th, R->d_cursors, R->d_recs,

@@ -818,7 +818,7 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
     const bool has_filter = pp.filter_idx >= 0;
     const bool has_val = pp.val_idx >= 0;
     const bool packed = pp.packed_mode != 0;
-    constexpr unsigned CHUNK = 128;                   /* records per reservation */
+    constexpr unsigned CHUNK = 64;                    /* records per reservation (512 B) */
 
     for (int i = tid; i < kNB; i += 256) { cpos[i] = 0; crem[i] = 0; }
     __syncthreads();
