@@ -761,9 +761,21 @@ static int run_partitioned(const YtPlan* plan, const YtChunk* chunk,
         if (ev && atoi(ev) >= 256) tile_rows = atoi(ev);
     }
     if (tile_rows > 8192) tile_rows = 8192;   /* per-thread row arrays bound */
-    /* LDS: per-tile bucket histograms + chunk cursors + the staged key */
-    size_t lds = 6 * kNB * 4 + ((size_t)tile_rows * w / 64 + 2) * 8 + 256;
-    if (lds > 64 * 1024) { tile_rows = 4096; lds = 6 * kNB * 4 + ((size_t)tile_rows * w / 64 + 2) * 8 + 256; }
+    /* LDS: histograms/cursors/prefix (7*kNB u32 + scan scratch) + the
+     * bucket-major record buffer + the staged key column */
+    tile_rows = 4096;
+    {
+        const char* ev2 = getenv("YTQL_TILE");
+        if (ev2 && atoi(ev2) >= 256) tile_rows = atoi(ev2) > 8192 ? 8192 : atoi(ev2);
+    }
+    size_t lds = 0;
+    for (;;) {
+        size_t buf = (size_t)tile_rows * 8 * (pp.packed_mode ? 1 : 2);
+        lds = 7 * kNB * 4 + (256 + 4) * 4 + buf
+            + ((size_t)tile_rows * w / 64 + 2) * 8 + 256;
+        if (lds <= 100 * 1024 || tile_rows == 256) break;
+        tile_rows >>= 1;
+    }
     while (tile_rows > 256 && tile_rows > seg0_rows) tile_rows >>= 1;
     pp.tile_rows = tile_rows;
     pp.tiles_per_seg = (seg0_rows + tile_rows - 1) / tile_rows;

@@ -813,7 +813,10 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
     unsigned* ngbase = nhist + kNB;
     unsigned* cpos = ngbase + kNB;                    /* WG chunk cursor (records index) */
     unsigned* crem = cpos + kNB;                      /* WG chunk remaining */
-    uint64_t* klds = (uint64_t*)(crem + kNB);         /* staged key words */
+    unsigned* lstart = crem + kNB;                    /* per-tile exclusive prefix */
+    unsigned* partials = lstart + kNB;                /* 256 scan partials */
+    uint64_t* lbuf = (uint64_t*)(partials + 256 + 4); /* bucket-major record buffer */
+    uint64_t* klds = lbuf + (pp.packed_mode ? pp.tile_rows : 2 * pp.tile_rows);
 
     const bool has_filter = pp.filter_idx >= 0;
     const bool has_val = pp.val_idx >= 0;
@@ -959,8 +962,34 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
                 ngbase[i] = (unsigned)base;
             }
         }
+        /* exclusive prefix of hist → lstart (thread t owns 4 contiguous
+         * buckets; wave scan + tiny cross-wave combine) */
+        {
+            const int own = tid * (kNB / 256);
+            unsigned mysum = 0;
+            #pragma unroll
+            for (int q = 0; q < kNB / 256; q++) mysum += hist[own + q];
+            const int lane = tid & 63, wave = tid >> 6;
+            unsigned inc = mysum;
+            #pragma unroll
+            for (int off = 1; off < 64; off <<= 1) {
+                unsigned up = (unsigned)__shfl_up((int)inc, off, 64);
+                if (lane >= off) inc += up;
+            }
+            if (lane == 63) partials[256 + wave] = inc;
+            __syncthreads();
+            unsigned wbase = 0;
+            for (int q = 0; q < wave; q++) wbase += partials[256 + q];
+            unsigned run = wbase + inc - mysum;
+            #pragma unroll
+            for (int q = 0; q < kNB / 256; q++) {
+                lstart[own + q] = run;
+                run += hist[own + q];
+            }
+        }
         __syncthreads();
         if (th->overflow != 1) {
+            /* pass 1: scatter records into the bucket-major LDS buffer */
             #pragma unroll 4
             for (int i = 0; i < R; i++) {
                 if (row_b[i] == 0xFFFFFFFFu) continue;
@@ -968,6 +997,7 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
                 uint64_t kzzfull = kmin + (bp_get_win(klds, kwd, j, kW0) & kmask);
                 unsigned b = row_b[i] & 0x7FFFFFFFu;
                 if (row_b[i] & 0x80000000u) {
+                    /* null-value stream: rare, direct scatter */
                     nrecs[(int64_t)b * pp.nbucket_stride + ngbase[b] + row_off[i]] =
                         (uint64_t)zz_dec(kzzfull);
                 } else if (packed) {
@@ -976,16 +1006,38 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
                         uint64_t vzz = (vmin + bp_gl(vwords, vmask, vwd, j)) - pp.gmin_v;
                         rec |= vzz << pp.bits_k;
                     }
-                    recs8[(int64_t)b * pp.bucket_stride + gbase[b] + row_off[i]] = rec;
+                    lbuf[lstart[b] + row_off[i]] = rec;
                 } else {
                     uint64_t val = 0;
                     if (has_val) {
                         val = (uint64_t)zz_dec(vmin + bp_gl(vwords, vmask, vwd, j));
                     }
-                    recs[(int64_t)b * pp.bucket_stride + gbase[b] + row_off[i]] =
-                        make_ulonglong2((uint64_t)zz_dec(kzzfull), val);
+                    lbuf[2 * (lstart[b] + row_off[i])] = (uint64_t)zz_dec(kzzfull);
+                    lbuf[2 * (lstart[b] + row_off[i]) + 1] = val;
                 }
             }
+            __syncthreads();
+            /* pass 2: bucket-major burst — each thread streams its buckets'
+             * runs as contiguous stores, so every destination line fills
+             * back-to-back (short open-line lifetime, no partial-line
+             * write-back churn) */
+            #pragma unroll
+            for (int q = 0; q < kNB / 256; q++) {
+                int b = tid * (kNB / 256) + q;
+                unsigned c = hist[b];
+                if (!c) continue;
+                unsigned ls = lstart[b];
+                if (packed) {
+                    uint64_t* dst = recs8 + (int64_t)b * pp.bucket_stride + gbase[b];
+                    for (unsigned r = 0; r < c; r++) dst[r] = lbuf[ls + r];
+                } else {
+                    ulonglong2* dst = recs + (int64_t)b * pp.bucket_stride + gbase[b];
+                    for (unsigned r = 0; r < c; r++)
+                        dst[r] = make_ulonglong2(lbuf[2 * (ls + r)], lbuf[2 * (ls + r) + 1]);
+                }
+            }
+        } else {
+            __syncthreads();
         }
         __syncthreads();
     }
