@@ -740,7 +740,7 @@ static int run_partitioned(const YtPlan* plan, const YtChunk* chunk,
             bv = bits_of(span_v);
             gmin_v = R->col_zzmin[fs->sum_col[0]];
         }
-        if (bk + bv <= 63) {   /* <= 63: all-ones stays free as the PAD */
+        if (bk + bv <= 64) {
             pp.packed_mode = 1;
             pp.bits_k = bk ? bk : 1;
             if (pp.bits_k + bv > 64) pp.bits_k = bk;  /* bk>=1 here */
@@ -761,21 +761,9 @@ static int run_partitioned(const YtPlan* plan, const YtChunk* chunk,
         if (ev && atoi(ev) >= 256) tile_rows = atoi(ev);
     }
     if (tile_rows > 8192) tile_rows = 8192;   /* per-thread row arrays bound */
-    /* LDS: histograms/cursors/prefix (7*kNB u32 + scan scratch) + the
-     * bucket-major record buffer + the staged key column */
-    tile_rows = 4096;
-    {
-        const char* ev2 = getenv("YTQL_TILE");
-        if (ev2 && atoi(ev2) >= 256) tile_rows = atoi(ev2) > 8192 ? 8192 : atoi(ev2);
-    }
-    size_t lds = 0;
-    for (;;) {
-        size_t buf = (size_t)tile_rows * 8 * (pp.packed_mode ? 1 : 2);
-        lds = 7 * kNB * 4 + (256 + 4) * 4 + buf
-            + ((size_t)tile_rows * w / 64 + 2) * 8 + 256;
-        if (lds <= 100 * 1024 || tile_rows == 256) break;
-        tile_rows >>= 1;
-    }
+    /* LDS: per-tile bucket histograms + the staged key column */
+    size_t lds = 4 * kNB * 4 + ((size_t)tile_rows * w / 64 + 2) * 8 + 256;
+    if (lds > 64 * 1024) { tile_rows = 4096; lds = 4 * kNB * 4 + ((size_t)tile_rows * w / 64 + 2) * 8 + 256; }
     while (tile_rows > 256 && tile_rows > seg0_rows) tile_rows >>= 1;
     pp.tile_rows = tile_rows;
     pp.tiles_per_seg = (seg0_rows + tile_rows - 1) / tile_rows;
@@ -784,12 +772,8 @@ static int run_partitioned(const YtPlan* plan, const YtChunk* chunk,
         int32_t last_rows = chunk->columns[0].segments[nseg - 1].row_count;
         pp.ntiles = (nseg - 1) * pp.tiles_per_seg + (last_rows + tile_rows - 1) / tile_rows;
     }
-    /* WG-private chunked regions: every workgroup may strand one partial
-     * chunk per bucket, so the worst-case slack is grid * CHUNK records */
-    const int part_grid = 1024;
-    const int part_chunk = 64;
-    pp.bucket_stride = rows / kNB + (rows / kNB) / 4
-                     + (int64_t)part_grid * part_chunk + 8192;
+    /* per (bucket, XCD) sub-streams: 8x more cursors, 1/8 the rows each */
+    pp.bucket_stride = rows / (kNB * 8) + (rows / (kNB * 8)) / 2 + 4096;
     pp.nbucket_stride = pp.has_val_nulls ? pp.bucket_stride : 0;
 
     std::vector<FastCol> fc(nused);
@@ -802,14 +786,14 @@ static int run_partitioned(const YtPlan* plan, const YtChunk* chunk,
         HIP_CHECK(pool_alloc(&R->d_fastcols, sizeof(FastCol) * nused));
         HIP_CHECK(hipMemcpyAsync(R->d_fastcols, fc.data(), sizeof(FastCol) * nused,
                                  hipMemcpyHostToDevice, R->stream));
-        HIP_CHECK(pool_alloc(&R->d_cursors, sizeof(uint64_t) * kNB));
-        HIP_CHECK(hipMemsetAsync(R->d_cursors, 0, sizeof(uint64_t) * kNB, R->stream));
+        HIP_CHECK(pool_alloc(&R->d_cursors, sizeof(uint64_t) * kNB * 8));
+        HIP_CHECK(hipMemsetAsync(R->d_cursors, 0, sizeof(uint64_t) * kNB * 8, R->stream));
         HIP_CHECK(pool_alloc(&R->d_recs,
-                             (size_t)kNB * pp.bucket_stride * (pp.packed_mode ? 8 : 16)));
+                             (size_t)kNB * 8 * pp.bucket_stride * (pp.packed_mode ? 8 : 16)));
         if (pp.has_val_nulls) {
-            HIP_CHECK(pool_alloc(&R->d_ncursors, sizeof(uint64_t) * kNB));
-            HIP_CHECK(hipMemsetAsync(R->d_ncursors, 0, sizeof(uint64_t) * kNB, R->stream));
-            HIP_CHECK(pool_alloc(&R->d_nrecs, (size_t)kNB * pp.nbucket_stride * 8));
+            HIP_CHECK(pool_alloc(&R->d_ncursors, sizeof(uint64_t) * kNB * 8));
+            HIP_CHECK(hipMemsetAsync(R->d_ncursors, 0, sizeof(uint64_t) * kNB * 8, R->stream));
+            HIP_CHECK(pool_alloc(&R->d_nrecs, (size_t)kNB * 8 * pp.nbucket_stride * 8));
         }
         /* the partitioned path's in-table sentinel is INT64_MIN bits */
         uint64_t sk = kEmptyKey;
@@ -830,7 +814,7 @@ static int run_partitioned(const YtPlan* plan, const YtChunk* chunk,
         HIP_CHECK(hipEventCreate(&ev0));
         HIP_CHECK(hipEventCreate(&ev1));
         HIP_CHECK(hipEventCreate(&ev2));
-        int grid = pp.ntiles < part_grid ? (pp.ntiles ? pp.ntiles : 1) : part_grid;
+        int grid = pp.ntiles < 2048 ? (pp.ntiles ? pp.ntiles : 1) : 2048;
         HIP_CHECK(hipEventRecord(ev0, R->stream));
         HIP_CHECK(ytql_launch_scan_partition(&pp, R->d_segs, R->d_segex, R->d_fastcols,
                                              R->d_th, R->d_cursors, R->d_recs,

@@ -798,33 +798,24 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
                  uint64_t* nrecs)
 {
     /* Canonical column slots: 0 = filter, 1 = key, 2 = value.
-     * Records are appended to WORKGROUP-PRIVATE chunks of each bucket's
-     * region: one workgroup (one XCD) fills each 1-KiB chunk with full
-     * cache lines, so no line is ever part-written by two L2s and no
-     * partially-filled line lingers (the naive per-tile global reservation
-     * measured 5x HBM write amplification). Abandoned chunk tails are
-     * PAD-filled; Phase B skips PADs. */
+     * Values are decoded straight from global memory (branchless funnel,
+     * L1/L2 absorb window overlap); LDS holds only the per-tile bucket
+     * histograms. See k_scan_fast for the measured rationale. */
     extern __shared__ __attribute__((aligned(16))) char smem[];
     const int tid = threadIdx.x;
 
-    unsigned* hist = (unsigned*)smem;                 /* per-tile counts */
-    unsigned* gbase = hist + kNB;                     /* per-tile write base (chunk-relative) */
+    unsigned* hist = (unsigned*)smem;
+    unsigned* gbase = hist + kNB;
     unsigned* nhist = gbase + kNB;
     unsigned* ngbase = nhist + kNB;
-    unsigned* cpos = ngbase + kNB;                    /* WG chunk cursor (records index) */
-    unsigned* crem = cpos + kNB;                      /* WG chunk remaining */
-    unsigned* lstart = crem + kNB;                    /* per-tile exclusive prefix */
-    unsigned* partials = lstart + kNB;                /* 256 scan partials */
-    uint64_t* lbuf = (uint64_t*)(partials + 256 + 4); /* bucket-major record buffer */
-    uint64_t* klds = lbuf + (pp.packed_mode ? pp.tile_rows : 2 * pp.tile_rows);
+    uint64_t* klds = (uint64_t*)(ngbase + kNB);   /* staged key words */
 
     const bool has_filter = pp.filter_idx >= 0;
     const bool has_val = pp.val_idx >= 0;
-    const bool packed = pp.packed_mode != 0;
-    constexpr unsigned CHUNK = 64;                    /* records per reservation (512 B) */
-
-    for (int i = tid; i < kNB; i += 256) { cpos[i] = 0; crem[i] = 0; }
-    __syncthreads();
+    /* per-XCD sub-buckets: workgroups on different XCDs write disjoint
+     * record regions, so no cache line is filled by two L2s (cross-XCD
+     * partial-line sharing measured 3.7x write amplification) */
+    const int sub = blockIdx.x & 7;
 
     for (int tile = blockIdx.x; tile < pp.ntiles; tile += gridDim.x) {
         const int seg_idx = tile / pp.tiles_per_seg;
@@ -927,134 +918,51 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
             }
         }
         __syncthreads();
-        /* reserve space: serve this tile's counts from the WG's private
-         * chunk of each bucket; refill (and PAD the abandoned tail) when
-         * short. Null-stream rows are rare: direct global reservation. */
-        uint64_t* recs8 = (uint64_t*)recs;
         for (int i = tid; i < kNB; i += 256) {
             unsigned c = hist[i];
             if (c) {
-                if (crem[i] < c) {
-                    /* pad abandoned tail of the old chunk */
-                    uint64_t base = (uint64_t)cpos[i];
-                    for (unsigned p = 0; p < crem[i]; p++) {
-                        if (packed) {
-                            recs8[(int64_t)i * pp.bucket_stride + base + p] = ~0ULL;
-                        } else {
-                            recs[(int64_t)i * pp.bucket_stride + base + p] =
-                                make_ulonglong2(kEmptyKey, 0);
-                        }
-                    }
-                    unsigned want = c > CHUNK ? c : CHUNK;
-                    unsigned long long nb = atomicAdd(&cursors[i], (unsigned long long)want);
-                    if ((int64_t)(nb + want) > pp.bucket_stride) { th->overflow = 1; nb = 0; }
-                    cpos[i] = (unsigned)nb;
-                    crem[i] = want;
-                }
-                gbase[i] = cpos[i];
-                cpos[i] += c;
-                crem[i] -= c;
+                unsigned long long base = atomicAdd(&cursors[i * 8 + sub], (unsigned long long)c);
+                if ((int64_t)(base + c) > pp.bucket_stride) { th->overflow = 1; base = 0; }
+                gbase[i] = (unsigned)base;
             }
             unsigned nc = nhist[i];
             if (nc) {
-                unsigned long long base = atomicAdd(&ncursors[i], (unsigned long long)nc);
+                unsigned long long base = atomicAdd(&ncursors[i * 8 + sub], (unsigned long long)nc);
                 if ((int64_t)(base + nc) > pp.nbucket_stride) { th->overflow = 1; base = 0; }
                 ngbase[i] = (unsigned)base;
             }
         }
-        /* exclusive prefix of hist → lstart (thread t owns 4 contiguous
-         * buckets; wave scan + tiny cross-wave combine) */
-        {
-            const int own = tid * (kNB / 256);
-            unsigned mysum = 0;
-            #pragma unroll
-            for (int q = 0; q < kNB / 256; q++) mysum += hist[own + q];
-            const int lane = tid & 63, wave = tid >> 6;
-            unsigned inc = mysum;
-            #pragma unroll
-            for (int off = 1; off < 64; off <<= 1) {
-                unsigned up = (unsigned)__shfl_up((int)inc, off, 64);
-                if (lane >= off) inc += up;
-            }
-            if (lane == 63) partials[256 + wave] = inc;
-            __syncthreads();
-            unsigned wbase = 0;
-            for (int q = 0; q < wave; q++) wbase += partials[256 + q];
-            unsigned run = wbase + inc - mysum;
-            #pragma unroll
-            for (int q = 0; q < kNB / 256; q++) {
-                lstart[own + q] = run;
-                run += hist[own + q];
-            }
-        }
         __syncthreads();
+        uint64_t* recs8 = (uint64_t*)recs;
         if (th->overflow != 1) {
-            /* pass 1: scatter records into the bucket-major LDS buffer */
             #pragma unroll 4
             for (int i = 0; i < R; i++) {
                 if (row_b[i] == 0xFFFFFFFFu) continue;
                 int64_t j = t0 + (int64_t)i * 256 + tid;
                 uint64_t kzzfull = kmin + (bp_get_win(klds, kwd, j, kW0) & kmask);
                 unsigned b = row_b[i] & 0x7FFFFFFFu;
+                int64_t sb = (int64_t)b * 8 + sub;
                 if (row_b[i] & 0x80000000u) {
-                    /* null-value stream: rare, direct scatter */
-                    nrecs[(int64_t)b * pp.nbucket_stride + ngbase[b] + row_off[i]] =
+                    nrecs[sb * pp.nbucket_stride + ngbase[b] + row_off[i]] =
                         (uint64_t)zz_dec(kzzfull);
-                } else if (packed) {
+                } else if (pp.packed_mode) {
                     uint64_t rec = kzzfull - pp.gmin_k;
                     if (has_val) {
                         uint64_t vzz = (vmin + bp_gl(vwords, vmask, vwd, j)) - pp.gmin_v;
                         rec |= vzz << pp.bits_k;
                     }
-                    lbuf[lstart[b] + row_off[i]] = rec;
+                    recs8[sb * pp.bucket_stride + gbase[b] + row_off[i]] = rec;
                 } else {
                     uint64_t val = 0;
                     if (has_val) {
                         val = (uint64_t)zz_dec(vmin + bp_gl(vwords, vmask, vwd, j));
                     }
-                    lbuf[2 * (lstart[b] + row_off[i])] = (uint64_t)zz_dec(kzzfull);
-                    lbuf[2 * (lstart[b] + row_off[i]) + 1] = val;
+                    recs[sb * pp.bucket_stride + gbase[b] + row_off[i]] =
+                        make_ulonglong2((uint64_t)zz_dec(kzzfull), val);
                 }
             }
-            __syncthreads();
-            /* pass 2: bucket-major burst — each thread streams its buckets'
-             * runs as contiguous stores, so every destination line fills
-             * back-to-back (short open-line lifetime, no partial-line
-             * write-back churn) */
-            #pragma unroll
-            for (int q = 0; q < kNB / 256; q++) {
-                int b = tid * (kNB / 256) + q;
-                unsigned c = hist[b];
-                if (!c) continue;
-                unsigned ls = lstart[b];
-                if (packed) {
-                    uint64_t* dst = recs8 + (int64_t)b * pp.bucket_stride + gbase[b];
-                    for (unsigned r = 0; r < c; r++) dst[r] = lbuf[ls + r];
-                } else {
-                    ulonglong2* dst = recs + (int64_t)b * pp.bucket_stride + gbase[b];
-                    for (unsigned r = 0; r < c; r++)
-                        dst[r] = make_ulonglong2(lbuf[2 * (ls + r)], lbuf[2 * (ls + r) + 1]);
-                }
-            }
-        } else {
-            __syncthreads();
         }
         __syncthreads();
-    }
-
-    /* PAD the unused tails of this workgroup's final chunks (reserved but
-     * unwritten positions would otherwise surface as garbage records) */
-    uint64_t* recs8f = (uint64_t*)recs;
-    for (int i = tid; i < kNB; i += 256) {
-        uint64_t base = (uint64_t)cpos[i];
-        for (unsigned p = 0; p < crem[i]; p++) {
-            if (pp.packed_mode) {
-                recs8f[(int64_t)i * pp.bucket_stride + base + p] = ~0ULL;
-            } else {
-                recs[(int64_t)i * pp.bucket_stride + base + p] =
-                    make_ulonglong2(kEmptyKey, 0);
-            }
-        }
     }
 }
 
@@ -1084,21 +992,18 @@ k_bucket_agg(const ulonglong2* recs, const unsigned long long* cursors,
 
     const uint64_t kmask = (bits_k >= 64) ? ~0ULL : ((1ULL << bits_k) - 1);
     bool full = false;
-    {
-    int64_t n = (int64_t)cursors[bucket];
-    const ulonglong2* rows = recs + (int64_t)bucket * bucket_stride;
-    const uint64_t* rows8 = (const uint64_t*)recs + (int64_t)bucket * bucket_stride;
+    for (int sub = 0; sub < 8 && !full; sub++) {
+    int64_t n = (int64_t)cursors[bucket * 8 + sub];
+    const ulonglong2* rows = recs + ((int64_t)bucket * 8 + sub) * bucket_stride;
+    const uint64_t* rows8 = (const uint64_t*)recs + ((int64_t)bucket * 8 + sub) * bucket_stride;
     /* 4 records per thread per pass: independent probes overlap LDS latency */
     int64_t i = tid;
     #define LOADKV(kv, idx)                                                  \
         ulonglong2 kv;                                                       \
         if (packed_mode) {                                                   \
             uint64_t r_ = rows8[idx];                                        \
-            if (r_ == ~0ULL) { kv.x = kEmptyKey; kv.y = 0; }   /* PAD */     \
-            else {                                                           \
-                kv.x = (uint64_t)zz_dec(gmin_k + (r_ & kmask));              \
-                kv.y = (uint64_t)zz_dec(gmin_v + (r_ >> bits_k));            \
-            }                                                                \
+            kv.x = (uint64_t)zz_dec(gmin_k + (r_ & kmask));                  \
+            kv.y = (uint64_t)zz_dec(gmin_v + (r_ >> bits_k));                \
         } else {                                                             \
             kv = rows[idx];                                                  \
         }
@@ -1112,7 +1017,7 @@ k_bucket_agg(const ulonglong2* recs, const unsigned long long* cursors,
         uint64_t s2 = mix64(kv2.x) & (kHSlots - 1);
         uint64_t s3 = mix64(kv3.x) & (kHSlots - 1);
         #define PROBE(kv, sv_)                                               \
-        if (kv.x != (uint64_t)kEmptyKey) {   /* PAD-skip */                  \
+        {                                                                    \
             uint64_t sp = sv_;                                               \
             int found = 0;                                                   \
             for (int it = 0; it < kHSlots; it++) {                           \
@@ -1144,13 +1049,13 @@ k_bucket_agg(const ulonglong2* recs, const unsigned long long* cursors,
         uint64_t s0 = mix64(kv.x) & (kHSlots - 1);
         PROBE(kv, s0)
     }
-    }
+    }   /* sub-stream loop */
     #undef PROBE
     #undef LOADKV
     if (nrecs) {
-        {
-        int64_t nn = (int64_t)ncursors[bucket];
-        const uint64_t* nrows = nrecs + (int64_t)bucket * nbucket_stride;
+        for (int sub = 0; sub < 8 && !full; sub++) {
+        int64_t nn = (int64_t)ncursors[bucket * 8 + sub];
+        const uint64_t* nrows = nrecs + ((int64_t)bucket * 8 + sub) * nbucket_stride;
         for (int64_t i = tid; i < nn && !full; i += 256) {
             uint64_t key = nrows[i];
             uint64_t s = mix64(key) & (kHSlots - 1);
@@ -1169,7 +1074,7 @@ k_bucket_agg(const ulonglong2* recs, const unsigned long long* cursors,
             if (!found) { full = true; break; }
             atomicAdd(&tab[s * 3 + 1], 1ULL);   /* cnt only; sum stays null-contributing */
         }
-        }
+        }   /* sub-stream loop */
     }
     if (full) th->overflow = 1;
     __syncthreads();
