@@ -1007,11 +1007,15 @@ k_bucket_agg(const ulonglong2* recs, const unsigned long long* cursors,
         } else {                                                             \
             kv = rows[idx];                                                  \
         }
-    for (; i + 768 < n; i += 1024) {
+    for (; i + 1792 < n; i += 2048) {
         LOADKV(kv0, i)
         LOADKV(kv1, i + 256)
         LOADKV(kv2, i + 512)
         LOADKV(kv3, i + 768)
+        LOADKV(kv4, i + 1024)
+        LOADKV(kv5, i + 1280)
+        LOADKV(kv6, i + 1536)
+        LOADKV(kv7, i + 1792)
         uint64_t s0 = mix64(kv0.x) & (kHSlots - 1);
         uint64_t s1 = mix64(kv1.x) & (kHSlots - 1);
         uint64_t s2 = mix64(kv2.x) & (kHSlots - 1);
@@ -1038,10 +1042,18 @@ k_bucket_agg(const ulonglong2* recs, const unsigned long long* cursors,
                     atomicAdd(&tab[sp * 3 + 2], (unsigned long long)kv.y);   \
             }                                                                \
         }
+        uint64_t s4 = mix64(kv4.x) & (kHSlots - 1);
+        uint64_t s5 = mix64(kv5.x) & (kHSlots - 1);
+        uint64_t s6 = mix64(kv6.x) & (kHSlots - 1);
+        uint64_t s7 = mix64(kv7.x) & (kHSlots - 1);
         PROBE(kv0, s0)
         PROBE(kv1, s1)
         PROBE(kv2, s2)
         PROBE(kv3, s3)
+        PROBE(kv4, s4)
+        PROBE(kv5, s5)
+        PROBE(kv6, s6)
+        PROBE(kv7, s7)
         if (full) break;
     }
     for (; i < n && !full; i += 256) {
