@@ -255,3 +255,22 @@ def test_statistics_counters(cuda):
     assert stats.rows_written == len(got)
     assert stats.grouped_row_count == len(got)
     assert stats.data_weight_read > 0
+
+
+def test_min_max_parity(cuda):
+    """min/max aggregates (udf/min.c, max.c semantics) on the generic path,
+    int64 and double, with nulls."""
+    from ytsaurus_amd._abi import AGG_MIN, AGG_MAX
+    rng = np.random.default_rng(13)
+    n = 100_000
+    keys = rng.integers(0, 300, n, dtype=np.int64)
+    iv = rng.integers(-2**62, 2**62, n, dtype=np.int64)
+    dv = rng.standard_normal(n) * 1e6
+    ivn = (rng.random(n) < 0.1).astype(np.uint8)
+    chunk = y.Chunk([y.encode_int64(keys), y.encode_int64(iv, ivn),
+                     y.encode_double(dv)], n)
+    plan = y.Plan(keys=[y.col(0)],
+                  aggs=[(AGG_MIN, y.col(1)), (AGG_MAX, y.col(1)),
+                        (AGG_MIN, y.col(2)), (AGG_MAX, y.col(2)), y.agg_sum1()])
+    got, want, _ = run_both(plan, chunk, cuda, hint=1024)
+    assert y.sort_rows(got) == y.sort_rows(want)

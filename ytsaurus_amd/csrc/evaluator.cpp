@@ -246,8 +246,9 @@ static int build_devplan(const YtPlan* plan, const YtChunk* chunk, DevPlan* p,
     if (plan->key_count > 1) { set_err(errbuf, errlen, "GPU path: at most 1 group key this round"); return YT_ERR_UNSUPPORTED; }
     if (plan->agg_count < 1 || plan->agg_count > kMaxAggs) { set_err(errbuf, errlen, "need 1..4 aggregates"); return YT_ERR_UNSUPPORTED; }
     for (int a = 0; a < plan->agg_count; a++) {
-        if (plan->aggs[a]->func != YT_AGG_SUM && plan->aggs[a]->func != YT_AGG_SUM1) {
-            set_err(errbuf, errlen, "GPU path: sum/sum(1) aggregates this round");
+        int f = plan->aggs[a]->func;
+        if (f != YT_AGG_SUM && f != YT_AGG_SUM1 && f != YT_AGG_MIN && f != YT_AGG_MAX) {
+            set_err(errbuf, errlen, "unsupported aggregate");
             return YT_ERR_UNSUPPORTED;
         }
     }
@@ -268,7 +269,7 @@ static int build_devplan(const YtPlan* plan, const YtChunk* chunk, DevPlan* p,
     p->agg_count = plan->agg_count;
     for (int a = 0; a < plan->agg_count; a++) {
         p->agg_func[a] = plan->aggs[a]->func;
-        if (plan->aggs[a]->func == YT_AGG_SUM) {
+        if (plan->aggs[a]->func != YT_AGG_SUM1) {
             p->agg_off[a] = p->prog_len;
             rc = compile_expr(plan->aggs[a]->arg, p, &p->agg_len[a], errbuf, errlen);
             if (rc) return rc;
@@ -1020,7 +1021,22 @@ static void finalize_row(const YtPlan* plan, uint8_t key_type,
             row[n].bits = 0;
         } else {
             row[n].type = sum_type[a];
-            row[n].bits = agg_bits[a];
+            uint64_t bits = agg_bits[a];
+            int f = plan->aggs[a]->func;
+            if (f == YT_AGG_MIN || f == YT_AGG_MAX) {
+                /* undo the order-preserving map (kernels.hip ord_map):
+                 * MIN stored ~m(x); then m inverse per type */
+                uint64_t m = (f == YT_AGG_MIN) ? ~bits : bits;
+                const uint64_t SIGN = 0x8000000000000000ULL;
+                if (sum_type[a] == YT_VT_DOUBLE) {
+                    bits = (m & SIGN) ? (m & ~SIGN) : ~m;
+                } else if (sum_type[a] == YT_VT_INT64) {
+                    bits = m ^ SIGN;
+                } else {
+                    bits = m;
+                }
+            }
+            row[n].bits = bits;
         }
         n++;
     }

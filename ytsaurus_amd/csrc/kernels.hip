@@ -494,16 +494,37 @@ __device__ DVal eval_prog(const DevPlan& p, ColCtx& c, int off, int len)
 /* ------------------------------------------------------------------ */
 /* group table update                                                  */
 
+/* order-preserving map to u64 with 0 as the min/max identity (the table is
+ * zero-initialized; emptiness is distinguished by nonnull == 0):
+ *   MAX: m(x) such that x < y  <=>  m(x) < m(y); store max m(x)
+ *   MIN: store max of ~m(x) (order reversed)                            */
+__device__ __forceinline__ uint64_t ord_map(uint64_t bits, uint8_t type)
+{
+    if (type == YT_VT_DOUBLE) {
+        /* IEEE754 total-order map */
+        return (bits & 0x8000000000000000ULL) ? ~bits : (bits | 0x8000000000000000ULL);
+    }
+    if (type == YT_VT_INT64) return bits ^ 0x8000000000000000ULL;
+    return bits;   /* uint64 / boolean */
+}
+
 __device__ __forceinline__ void agg_update_slot(unsigned long long* aggp,
                                                 const DevPlan& p, int a,
                                                 DVal v)
 {
-    /* aggp -> { bits, nonnull } for agg a; sum semantics udf/sum.c:12-45 */
+    /* aggp -> { bits, nonnull } for agg a; semantics udf/sum.c, min.c, max.c */
     if (v.null_) return;
-    if (v.type == YT_VT_DOUBLE) {
-        atomicAdd((double*)aggp, __longlong_as_double(v.bits));
-    } else {
-        atomicAdd(aggp, (unsigned long long)v.bits);
+    int f = p.agg_func[a];
+    if (f == YT_AGG_SUM) {
+        if (v.type == YT_VT_DOUBLE) {
+            atomicAdd((double*)aggp, __longlong_as_double(v.bits));
+        } else {
+            atomicAdd(aggp, (unsigned long long)v.bits);
+        }
+    } else if (f == YT_AGG_MAX) {
+        atomicMax(aggp, (unsigned long long)ord_map(v.bits, v.type));
+    } else if (f == YT_AGG_MIN) {
+        atomicMax(aggp, (unsigned long long)~ord_map(v.bits, v.type));
     }
     atomicAdd(aggp + 1, 1ULL);
 }
@@ -558,7 +579,7 @@ __device__ void table_update_generic(TableHdr* th, unsigned long long* slots,
     }
     atomicAdd(cntp, 1ULL);
     for (int a = 0; a < p.agg_count; a++) {
-        if (p.agg_func[a] == YT_AGG_SUM) {
+        if (p.agg_func[a] != YT_AGG_SUM1) {
             agg_update_slot(aggp + 2 * a, p, a, aggv[a]);
         }
         /* YT_AGG_SUM1: cnt covers it */
@@ -599,7 +620,7 @@ k_scan_generic(DevPlan p, const DevSeg* segs, const SegEx* segex,
             key.bits = 1; key.type = YT_VT_INT64; key.null_ = 0;  /* single group, side-stepped below */
         }
         for (int a = 0; a < p.agg_count; a++) {
-            if (p.agg_func[a] == YT_AGG_SUM) {
+            if (p.agg_func[a] != YT_AGG_SUM1) {
                 aggv[a] = eval_prog(p, c, p.agg_off[a], p.agg_len[a]);
             }
         }
