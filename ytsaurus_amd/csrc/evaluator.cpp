@@ -1058,10 +1058,9 @@ static int emit_rows(const YtPlan* plan, const YtChunk* chunk,
     if (has_key) key_type = expr_static_type(plan->keys[0], col_types);
     uint8_t sum_type[kMaxAggs];
     for (int a = 0; a < plan->agg_count; a++) {
-        sum_type[a] = (plan->aggs[a]->func == YT_AGG_SUM)
+        sum_type[a] = (plan->aggs[a]->func != YT_AGG_SUM1)
             ? expr_static_type(plan->aggs[a]->arg, col_types) : YT_VT_INT64;
         if (sum_type[a] == YT_VT_NULL) sum_type[a] = YT_VT_INT64;
-        if (sum_type[a] == YT_VT_BOOLEAN) sum_type[a] = YT_VT_BOOLEAN;
     }
 
     int base_cols = (has_key ? 1 : 0) + plan->agg_count;
