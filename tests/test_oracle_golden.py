@@ -145,3 +145,24 @@ def test_int64_wrap():
     want = (2**63 - 1) * 2 + 5
     want = ((want + 2**63) % 2**64) - 2**63
     assert rows == [(want,)]
+
+
+def test_min_max_oracle_vs_numpy():
+    """min/max aggregate semantics (udf/min.c:40-47, max.c:40-47) against
+    numpy on the same seeded data."""
+    from ytsaurus_amd._abi import AGG_MIN, AGG_MAX
+    rng = np.random.default_rng(21)
+    n = 20_000
+    keys = rng.integers(0, 50, n, dtype=np.int64)
+    vals = rng.integers(-2**60, 2**60, n, dtype=np.int64)
+    vn = (rng.random(n) < 0.2).astype(np.uint8)
+    chunk = y.Chunk([y.encode_int64(keys), y.encode_int64(vals, vn)], n)
+    plan = y.Plan(keys=[y.col(0)], aggs=[(AGG_MIN, y.col(1)), (AGG_MAX, y.col(1))])
+    rows, _ = y.oracle_execute(plan, chunk)
+    got = {r[0]: (r[1], r[2]) for r in rows}
+    for k in np.unique(keys):
+        mask = (keys == k) & (vn == 0)
+        if mask.any():
+            assert got[k] == (int(vals[mask].min()), int(vals[mask].max()))
+        else:
+            assert got[k] == (None, None)
