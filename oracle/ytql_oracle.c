@@ -984,3 +984,140 @@ int64_t yto_bitunpack(const void* packed, uint64_t* out, int64_t max_out)
     for (int64_t i = 0; i < n; i++) out[i] = bitreader_get(&r, i);
     return (int64_t)r.size;
 }
+
+/* ------------------------------------------------------------------ */
+/* string segment decode — string_column_reader.cpp:
+ *   GetOffset :39-42 (offset(i) = expected_length*(i+1) + ZigZagDecode32(packed[i])),
+ *   SetStringValue :44-66, dictionary extract :77-124, direct extract :130-168,
+ *   RLE via TRleStringValueExtractorBase; blob layouts mirror
+ *   string_column_writer.cpp dumps (ids/offsets/data orders per dump fn).
+ * YtSegment.min_value carries expected_length.                         */
+
+static inline int32_t zigzag_decode32(uint32_t n)
+{
+    return (int32_t)((n >> 1) ^ (~(n & 1) + 1));
+}
+
+static uint64_t str_offset(const BitReader* offsets, uint64_t expected, int64_t i)
+{
+    if (i < 0) return 0;
+    return expected * (uint64_t)(i + 1)
+         + (uint64_t)(int64_t)zigzag_decode32((uint32_t)bitreader_get(offsets, i));
+}
+
+/* decode one string segment; appends string bytes to out_blob (cursor
+ * *blob_used), writes per-row cumulative ends via out_end[row+1] and nulls. */
+static int decode_string_segment(const YtSegment* seg,
+                                 char* out_blob, int64_t blob_cap, int64_t* blob_used,
+                                 int64_t* out_end, uint8_t* nulls)
+{
+    const char* ptr = (const char*)seg->data;
+    int64_t n = seg->row_count;
+    uint64_t expected = seg->min_value;
+
+    BitReader starts = {0}, ids = {0}, offs = {0};
+    const uint8_t* nb = NULL;
+    const char* data = NULL;
+    int64_t run_count = 0;
+    int is_dict = 0, is_rle = 0;
+
+    switch (seg->type) {
+    case YT_SEG_DIRECT_DENSE: {
+        offs = bitreader_init(ptr);
+        ptr += bitreader_byte_size(&offs);
+        nb = (const uint8_t*)ptr;
+        ptr += align_up8((n + 7) / 8);
+        data = ptr;
+        if ((int64_t)offs.size != n) return -1;
+        break;
+    }
+    case YT_SEG_DICTIONARY_DENSE: {
+        ids = bitreader_init(ptr);
+        ptr += bitreader_byte_size(&ids);
+        offs = bitreader_init(ptr);
+        ptr += bitreader_byte_size(&offs);
+        data = ptr;
+        is_dict = 1;
+        if ((int64_t)ids.size != n) return -1;
+        break;
+    }
+    case YT_SEG_DIRECT_RLE: {
+        starts = bitreader_init(ptr);
+        ptr += bitreader_byte_size(&starts);
+        run_count = (int64_t)starts.size;
+        offs = bitreader_init(ptr);
+        ptr += bitreader_byte_size(&offs);
+        nb = (const uint8_t*)ptr;
+        ptr += align_up8((run_count + 7) / 8);
+        data = ptr;
+        is_rle = 1;
+        break;
+    }
+    case YT_SEG_DICTIONARY_RLE: {
+        starts = bitreader_init(ptr);
+        ptr += bitreader_byte_size(&starts);
+        run_count = (int64_t)starts.size;
+        ids = bitreader_init(ptr);
+        ptr += bitreader_byte_size(&ids);
+        offs = bitreader_init(ptr);
+        ptr += bitreader_byte_size(&offs);
+        data = ptr;
+        is_rle = 1;
+        is_dict = 1;
+        break;
+    }
+    default:
+        return -1;
+    }
+
+    int64_t run = 0;
+    for (int64_t i = 0; i < n; i++) {
+        int64_t idx = i;
+        if (is_rle) {
+            while (run + 1 < run_count && (int64_t)bitreader_get(&starts, run + 1) <= i) run++;
+            idx = run;
+        }
+        int isnull;
+        int64_t sbeg = 0, slen = 0;
+        if (is_dict) {
+            uint64_t id = bitreader_get(&ids, idx);
+            isnull = (id == 0);
+            if (!isnull) {
+                sbeg = (int64_t)str_offset(&offs, expected, (int64_t)id - 2);
+                slen = (int64_t)str_offset(&offs, expected, (int64_t)id - 1) - sbeg;
+            }
+        } else {
+            isnull = bitmap_get(nb, idx);
+            if (!isnull) {
+                sbeg = (int64_t)str_offset(&offs, expected, idx - 1);
+                slen = (int64_t)str_offset(&offs, expected, idx) - sbeg;
+            }
+        }
+        nulls[i] = (uint8_t)isnull;
+        if (!isnull) {
+            if (*blob_used + slen > blob_cap) return -2;
+            memcpy(out_blob + *blob_used, data + sbeg, slen);
+            *blob_used += slen;
+        }
+        out_end[i + 1] = *blob_used;
+    }
+    return 0;
+}
+
+ORACLE_EXPORT
+int yto_decode_string_column(const YtColumn* col, int64_t row_count,
+                             char* out_blob, int64_t blob_cap,
+                             int64_t* out_end /* row_count+1 */, uint8_t* nulls)
+{
+    int64_t row = 0;
+    int64_t used = 0;
+    out_end[0] = 0;
+    for (int s = 0; s < col->segment_count; s++) {
+        const YtSegment* seg = &col->segments[s];
+        int rc = decode_string_segment(seg, out_blob, blob_cap, &used,
+                                       out_end + row, nulls + row);
+        if (rc != 0) return rc == -2 ? YT_ERR_CAPACITY : YT_ERR_INVALID_CHUNK;
+        row += seg->row_count;
+    }
+    return (row == row_count) ? YT_OK : YT_ERR_INVALID_CHUNK;
+}

@@ -179,6 +179,10 @@ def gpu_lib():
         _sig(lib, "yt_encode_double_column", C.c_int,
              [C.POINTER(C.c_double), C.POINTER(C.c_uint8), C.c_int64, C.c_int32,
               C.POINTER(YtEncodedColumn), C.c_char_p, C.c_size_t])
+        _sig(lib, "yt_encode_string_column", C.c_int,
+             [C.c_char_p, C.POINTER(C.c_uint64), C.POINTER(C.c_uint32),
+              C.POINTER(C.c_uint8), C.c_int64, C.c_int32,
+              C.POINTER(YtEncodedColumn), C.c_char_p, C.c_size_t])
         _sig(lib, "yt_encoded_column_free", None, [C.POINTER(YtEncodedColumn)])
         _sig(lib, "yt_bitpack_size_words", C.c_int64, [C.c_uint64, C.c_int64])
         _sig(lib, "yt_bitpack", C.c_int64,
@@ -211,5 +215,8 @@ def oracle_lib():
         _sig(lib, "yto_partition_hash", C.c_uint64, [C.c_uint64, C.c_int])
         _sig(lib, "yto_bitunpack", C.c_int64,
              [C.c_void_p, C.POINTER(C.c_uint64), C.c_int64])
+        _sig(lib, "yto_decode_string_column", C.c_int,
+             [C.POINTER(YtColumn), C.c_int64, C.c_char_p, C.c_int64,
+              C.POINTER(C.c_int64), C.POINTER(C.c_uint8)])
         _oracle_lib = lib
     return _oracle_lib

@@ -10,7 +10,7 @@ from . import _abi
 from ._abi import (
     YtValue, YtSegment, YtColumn, YtChunk, YtExpr, YtAgg, YtPlan,
     YtExecOptions, YtStatistics, YtRowset, YtStateRow, YtEncodedColumn,
-    VT_NULL, VT_INT64, VT_UINT64, VT_DOUBLE, VT_BOOLEAN,
+    VT_NULL, VT_INT64, VT_UINT64, VT_DOUBLE, VT_BOOLEAN, VT_STRING,
     EX_COLUMN, EX_LIT_I64, EX_LIT_NULL, EX_LIT_DOUBLE,
     EX_ADD, EX_SUB, EX_MUL, EX_DIV, EX_MOD,
     EX_EQ, EX_NE, EX_LT, EX_LE, EX_GT, EX_GE, EX_AND, EX_OR, EX_NOT,
@@ -19,7 +19,8 @@ from ._abi import (
 
 __all__ = [
     "col", "lit", "null", "litf", "Plan", "agg_sum", "agg_sum1",
-    "encode_int64", "encode_double", "Chunk", "oracle_execute",
+    "encode_int64", "encode_double", "encode_string", "oracle_decode_strings",
+    "Chunk", "oracle_execute",
     "oracle_partial", "oracle_merge",
     "gpu_available", "gpu_execute", "gpu_partial", "gpu_merge",
     "rows_from_rowset", "sort_rows", "make_rowset",
@@ -197,6 +198,62 @@ def encode_int64(values, nulls=None, max_segment_values=0, unsigned=False,
     _check(rc, err)
     vt = VT_UINT64 if unsigned else VT_INT64
     return EncodedColumn(vt, enc)
+
+
+def encode_string(strings, max_segment_values=0):
+    """strings: list of bytes/str or None (null). Encodes into the
+    reference's unversioned string segment formats."""
+    blobs = []
+    begins = np.zeros(len(strings), dtype=np.uint64)
+    lens = np.zeros(len(strings), dtype=np.uint32)
+    nulls = np.zeros(len(strings), dtype=np.uint8)
+    at = 0
+    for i, v in enumerate(strings):
+        if v is None:
+            nulls[i] = 1
+            continue
+        if isinstance(v, str):
+            v = v.encode()
+        blobs.append(v)
+        begins[i] = at
+        lens[i] = len(v)
+        at += len(v)
+    blob = b"".join(blobs)
+    enc = YtEncodedColumn()
+    err = C.create_string_buffer(256)
+    rc = _abi.gpu_lib().yt_encode_string_column(
+        blob, begins.ctypes.data_as(C.POINTER(C.c_uint64)),
+        lens.ctypes.data_as(C.POINTER(C.c_uint32)),
+        nulls.ctypes.data_as(C.POINTER(C.c_uint8)),
+        len(strings), max_segment_values, C.byref(enc), err, 256)
+    _check(rc, err)
+    return EncodedColumn(VT_STRING, enc)
+
+
+def oracle_decode_strings(enc, n):
+    """TEST ONLY: decode a string column via the oracle; returns list of
+    bytes-or-None."""
+    colc = YtColumn(value_type=enc.value_type,
+                    segment_count=enc._cenc.segment_count,
+                    segments=enc._cenc.segments)
+    cap = enc._cenc.blob_size + 16
+    # dictionary/RLE segments decode to more bytes than they store
+    cap = max(cap, 64 * 1024 * 1024)
+    blob = C.create_string_buffer(cap)
+    ends = (C.c_int64 * (n + 1))()
+    nulls = (C.c_uint8 * max(n, 1))()
+    rc = _abi.oracle_lib().yto_decode_string_column(
+        C.byref(colc), n, blob, cap, ends, nulls)
+    if rc != YT_OK:
+        raise RuntimeError("string decode failed rc=%d" % rc)
+    raw = blob.raw[:ends[n] if n else 0]
+    out = []
+    for i in range(n):
+        if nulls[i]:
+            out.append(None)
+        else:
+            out.append(raw[ends[i]:ends[i + 1]])
+    return out
 
 
 def encode_double(values, nulls=None, max_segment_values=0):

@@ -27,6 +27,8 @@
 #include <stdlib.h>
 #include <vector>
 #include <unordered_map>
+#include <string_view>
+#include <string>
 
 #include "../../include/ytql_gpu.h"
 
@@ -351,4 +353,243 @@ extern "C" void yt_encoded_column_free(YtEncodedColumn* col)
         col->segments = nullptr;
         col->segment_count = 0;
     }
+}
+
+/* ------------------------------------------------------------------ */
+/* string columns — string_column_writer.cpp (unversioned):
+ *   CaptureValue :100-151 (dictionary ids in first-appearance order, 1-based,
+ *   empty-vs-null via pointer sentinel), DumpDictionaryValues :153-202,
+ *   DumpDirectValues :205-231, RLE dumps :497-594, segment choice :592-648,
+ *   size estimates :651-680; offsets stored as ZigZagEncode32 of the diff
+ *   from an expected running length (PrepareDiffFromExpected,
+ *   core/misc/bit_packed_unsigned_vector.cpp:11-31, expected =
+ *   DivRound(last, count), numeric_helpers-inl.h:29-33).
+ * YtSegment.min_value carries TStringSegmentMeta.expected_length. */
+
+static inline uint32_t zigzag_encode32(int32_t n)
+{
+    return ((uint32_t)n << 1) ^ (uint32_t)(n >> 31);
+}
+
+static uint32_t prepare_diff_from_expected(std::vector<uint64_t>* values,
+                                           uint64_t* max_diff)
+{
+    *max_diff = 0;
+    if (values->empty()) return 0;
+    int64_t last = (int64_t)values->back();
+    int64_t cnt = (int64_t)values->size();
+    lldiv_t d = lldiv(last, cnt);
+    uint32_t expected = (uint32_t)(d.quot + (d.rem >= (cnt + 1) / 2 ? 1 : 0));
+    int64_t expected_value = 0;
+    for (size_t i = 0; i < values->size(); i++) {
+        expected_value += expected;
+        int32_t diff = (int32_t)((int64_t)(*values)[i] - expected_value);
+        (*values)[i] = zigzag_encode32(diff);
+        if ((*values)[i] > *max_diff) *max_diff = (*values)[i];
+    }
+    return expected;
+}
+
+namespace {
+
+struct StrSeg {
+    /* per-segment string state (Values_ as (off,len,null) views) */
+    std::vector<uint64_t> begins;
+    std::vector<uint32_t> lens;
+    std::vector<uint8_t> nulls;
+    std::unordered_map<std::string_view, uint32_t> dict;  /* first-appearance ids */
+    uint64_t dict_bytes = 0;
+    uint32_t max_len = 0;
+    int64_t direct_bytes = 0;      /* Σ lens (DirectBuffer_ size) */
+    int64_t rle_bytes = 0;         /* Σ run-head lens */
+    std::vector<uint64_t> rle_starts;
+};
+
+static void dump_string_segment(const char* blob, StrSeg& st, int64_t cum_rows,
+                                SegmentOut* out)
+{
+    const int64_t n = (int64_t)st.begins.size();
+    const int64_t run_count = (int64_t)st.rle_starts.size();
+    const int64_t ndict = (int64_t)st.dict.size();
+    (void)cum_rows;
+
+    auto sv = [&](int64_t i) -> std::string_view {
+        return std::string_view(blob + st.begins[i], st.lens[i]);
+    };
+
+    /* GetSegmentSize :651-680; enum order DictionaryRle=0, DictionaryDense=1,
+     * DirectRle=2, DirectDense=3 (private.h:33-38), first minimum wins */
+    int64_t sz[4];
+    sz[YT_SEG_DICTIONARY_RLE] = st.dict_bytes
+        + cs_bytes(st.max_len, ndict)
+        + cs_bytes((uint64_t)ndict + 1, run_count)
+        + cs_bytes((uint64_t)n, run_count);
+    sz[YT_SEG_DICTIONARY_DENSE] = st.dict_bytes
+        + cs_bytes(st.max_len, ndict)
+        + cs_bytes((uint64_t)ndict + 1, n);
+    sz[YT_SEG_DIRECT_RLE] = st.rle_bytes
+        + cs_bytes(st.max_len, run_count)
+        + cs_bytes((uint64_t)n, run_count)
+        + n / 8;
+    sz[YT_SEG_DIRECT_DENSE] = st.direct_bytes
+        + cs_bytes(st.max_len, n)
+        + n / 8;
+    int best = 0;
+    for (int t = 1; t < 4; t++) {
+        if (sz[t] < sz[best]) best = t;
+    }
+
+    out->type = best;
+    out->row_count = (int32_t)n;
+
+    uint64_t expected = 0;
+    switch (best) {
+    case YT_SEG_DIRECT_DENSE: {
+        /* [packed zigzag-diff end-offsets][null bitmap][string data] */
+        std::vector<uint64_t> offsets(n);
+        uint64_t run = 0;
+        for (int64_t i = 0; i < n; i++) {
+            run += st.lens[i];
+            offsets[i] = run;
+        }
+        uint64_t maxd;
+        expected = prepare_diff_from_expected(&offsets, &maxd);
+        out->blob.pack(offsets, maxd);
+        out->blob.bitmap(st.nulls);
+        for (int64_t i = 0; i < n; i++) {
+            if (!st.nulls[i]) out->blob.raw(blob + st.begins[i], st.lens[i]);
+        }
+        break;
+    }
+    case YT_SEG_DICTIONARY_DENSE: {
+        /* [packed ids][packed zigzag-diff dict end-offsets][dict data] */
+        std::vector<uint64_t> ids(n);
+        std::vector<uint64_t> dict_offsets;
+        std::vector<int64_t> dict_rows;   /* row index of each dict entry's first use */
+        uint64_t doff = 0;
+        uint32_t dsize = 0;
+        for (int64_t i = 0; i < n; i++) {
+            if (st.nulls[i]) { ids[i] = 0; continue; }
+            uint32_t id = st.dict[sv(i)];
+            ids[i] = id;
+            if (id > dsize) {
+                doff += st.lens[i];
+                dict_offsets.push_back(doff);
+                dict_rows.push_back(i);
+                dsize++;
+            }
+        }
+        out->blob.pack(ids, (uint64_t)dsize + 1);
+        uint64_t maxd;
+        expected = prepare_diff_from_expected(&dict_offsets, &maxd);
+        out->blob.pack(dict_offsets, maxd);
+        for (int64_t r : dict_rows) out->blob.raw(blob + st.begins[r], st.lens[r]);
+        break;
+    }
+    case YT_SEG_DIRECT_RLE: {
+        /* [packed run starts][packed zigzag-diff run end-offsets]
+         * [run null bitmap][run string data] */
+        std::vector<uint64_t> offsets;
+        std::vector<uint8_t> run_null;
+        uint64_t run = 0;
+        for (uint64_t ri : st.rle_starts) {
+            run += st.lens[ri];
+            offsets.push_back(run);
+            run_null.push_back(st.nulls[ri]);
+        }
+        out->blob.pack(st.rle_starts, st.rle_starts.back());
+        uint64_t maxd;
+        expected = prepare_diff_from_expected(&offsets, &maxd);
+        out->blob.pack(offsets, maxd);
+        out->blob.bitmap(run_null);
+        for (uint64_t ri : st.rle_starts) {
+            if (!st.nulls[ri]) out->blob.raw(blob + st.begins[ri], st.lens[ri]);
+        }
+        break;
+    }
+    case YT_SEG_DICTIONARY_RLE: {
+        /* [packed run starts][packed run ids][packed zigzag-diff dict
+         * end-offsets][dict data] — NB ids maxValue = dict.size() here
+         * (DumpDictionaryRleData :578), unlike the dense variant's size+1 */
+        std::vector<uint64_t> ids;
+        std::vector<uint64_t> dict_offsets;
+        std::vector<int64_t> dict_rows;
+        uint64_t doff = 0;
+        uint32_t dsize = 0;
+        for (uint64_t ri : st.rle_starts) {
+            if (st.nulls[ri]) { ids.push_back(0); continue; }
+            uint32_t id = st.dict[sv(ri)];
+            ids.push_back(id);
+            if (id > dsize) {
+                doff += st.lens[ri];
+                dict_offsets.push_back(doff);
+                dict_rows.push_back((int64_t)ri);
+                dsize++;
+            }
+        }
+        out->blob.pack(st.rle_starts, st.rle_starts.back());
+        out->blob.pack(ids, (uint64_t)st.dict.size());
+        uint64_t maxd;
+        expected = prepare_diff_from_expected(&dict_offsets, &maxd);
+        out->blob.pack(dict_offsets, maxd);
+        for (int64_t r : dict_rows) out->blob.raw(blob + st.begins[r], st.lens[r]);
+        break;
+    }
+    }
+    out->min_value = expected;    /* carries expected_length for strings */
+}
+
+} /* namespace */
+
+extern "C" int yt_encode_string_column(
+    const char* blob, const uint64_t* begins, const uint32_t* lens,
+    const uint8_t* nulls, int64_t n, int32_t max_segment_values,
+    YtEncodedColumn* out, char* errbuf, size_t errlen)
+{
+    if (max_segment_values <= 0) max_segment_values = 128 * 1024;
+    std::vector<SegmentOut> segs;
+    StrSeg st;
+    int64_t cum = 0;
+    auto flush = [&]() {
+        if (st.begins.empty()) return;
+        segs.emplace_back();
+        dump_string_segment(blob, st, cum, &segs.back());
+        st = StrSeg();
+    };
+    for (int64_t i = 0; i < n; i++) {
+        int nu = nulls ? nulls[i] : 0;
+        uint64_t b = nu ? 0 : begins[i];
+        uint32_t l = nu ? 0 : lens[i];
+        /* RLE run break — AreValuesEqual :233-247 (null==null; bytes equal) */
+        bool same = false;
+        if (!st.begins.empty()) {
+            size_t k = st.begins.size() - 1;
+            if (nu && st.nulls[k]) same = true;
+            else if (!nu && !st.nulls[k] && st.lens[k] == l &&
+                     memcmp(blob + st.begins[k], blob + b, l) == 0) same = true;
+        }
+        if (!same) {
+            st.rle_starts.push_back(st.begins.size());
+            st.rle_bytes += l;
+        }
+        if (!nu) {
+            auto ins = st.dict.emplace(std::string_view(blob + b, l),
+                                       (uint32_t)st.dict.size() + 1);
+            if (ins.second) {
+                st.dict_bytes += l;
+                if (l > st.max_len) st.max_len = l;
+            }
+            st.direct_bytes += l;
+        }
+        st.begins.push_back(b);
+        st.lens.push_back(l);
+        st.nulls.push_back((uint8_t)nu);
+        cum++;
+        if ((int64_t)st.begins.size() >= max_segment_values ||
+            st.direct_bytes > (int64_t)32 * 1024 * 1024) {
+            flush();
+        }
+    }
+    flush();
+    return finish_column(segs, out, errbuf, errlen);
 }
