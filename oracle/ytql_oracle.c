@@ -121,6 +121,10 @@ static inline int bitmap_get(const uint8_t* bm, uint64_t index)
 
 static inline int64_t align_up8(int64_t x) { return (x + 7) & ~(int64_t)7; }
 
+int yto_decode_string_column(const YtColumn* col, int64_t row_count,
+                             char* out_blob, int64_t blob_cap,
+                             int64_t* out_end, uint8_t* nulls);
+
 /* ------------------------------------------------------------------ */
 /* segment decode → (int64 bits, null bytemask)                        */
 
@@ -269,18 +273,23 @@ int yto_decode_column(const YtColumn* col, int64_t row_count,
 typedef struct {
     uint8_t type;     /* YT_VT_* */
     uint64_t bits;    /* i64/u64/double bits/bool(0/1) */
+    const char* str;  /* YT_VT_STRING payload */
+    uint32_t len;
 } Val;
 
 typedef struct {
     const int64_t* const* col_vals;
     const uint8_t* const* col_nulls;
     const uint8_t* col_types;
+    /* string columns: decoded blob + cumulative end offsets (len n+1) */
+    const char* const* col_str;
+    const int64_t* const* col_str_end;
     int ncols;
     int64_t row;
     int error;        /* YT_ERR_DIV_ZERO etc. */
 } EvalCtx;
 
-static Val VNULL(void) { Val v; v.type = YT_VT_NULL; v.bits = 0; return v; }
+static Val VNULL(void) { Val v; v.type = YT_VT_NULL; v.bits = 0; v.str = 0; v.len = 0; return v; }
 
 static Val eval_expr(const YtExpr* e, EvalCtx* ctx)
 {
@@ -289,6 +298,13 @@ static Val eval_expr(const YtExpr* e, EvalCtx* ctx)
     case YT_EX_COLUMN: {
         if (ctx->col_nulls[e->col][ctx->row]) return VNULL();
         v.type = ctx->col_types[e->col];
+        if (v.type == YT_VT_STRING) {
+            const int64_t* ends = ctx->col_str_end[e->col];
+            v.str = ctx->col_str[e->col] + ends[ctx->row];
+            v.len = (uint32_t)(ends[ctx->row + 1] - ends[ctx->row]);
+            v.bits = 0;
+            return v;
+        }
         v.bits = (uint64_t)ctx->col_vals[e->col][ctx->row];
         return v;
     }
@@ -372,6 +388,11 @@ static Val eval_expr(const YtExpr* e, EvalCtx* ctx)
             unsigned ln = (a.type == YT_VT_NULL), rn = (b.type == YT_VT_NULL);
             lt = rn < ln;     /* lhs < rhs  <=>  rhsIsNull < lhsIsNull */
             eq = ln == rn;
+        } else if (a.type == YT_VT_STRING) {
+            uint32_t minlen = a.len < b.len ? a.len : b.len;
+            int c = memcmp(a.str, b.str, minlen);
+            lt = (c < 0) || (c == 0 && a.len < b.len);
+            eq = (c == 0) && (a.len == b.len);
         } else if (a.type == YT_VT_DOUBLE) {
             double x, y;
             memcpy(&x, &a.bits, 8);
@@ -458,7 +479,15 @@ static uint64_t hash_keys(const Val* keys, int key_count)
 {
     uint64_t h = 0x12345678ULL;
     for (int i = 0; i < key_count; i++) {
-        h = splitmix64(h ^ splitmix64(((uint64_t)keys[i].type << 56) ^ keys[i].bits));
+        if (keys[i].type == YT_VT_STRING) {
+            uint64_t sh = 0xCBF29CE484222325ULL;
+            for (uint32_t k = 0; k < keys[i].len; k++) {
+                sh = (sh ^ (uint8_t)keys[i].str[k]) * 0x100000001B3ULL;
+            }
+            h = splitmix64(h ^ splitmix64(((uint64_t)keys[i].type << 56) ^ sh));
+        } else {
+            h = splitmix64(h ^ splitmix64(((uint64_t)keys[i].type << 56) ^ keys[i].bits));
+        }
     }
     return h;
 }
@@ -467,7 +496,12 @@ static int keys_eq(const Val* a, const Val* b, int key_count)
 {
     for (int i = 0; i < key_count; i++) {
         if (a[i].type != b[i].type) return 0;
-        if (a[i].type != YT_VT_NULL && a[i].bits != b[i].bits) return 0;
+        if (a[i].type == YT_VT_STRING) {
+            if (a[i].len != b[i].len) return 0;
+            if (memcmp(a[i].str, b[i].str, a[i].len) != 0) return 0;
+        } else if (a[i].type != YT_VT_NULL && a[i].bits != b[i].bits) {
+            return 0;
+        }
     }
     return 1;
 }
@@ -594,6 +628,8 @@ typedef struct {
     int64_t** vals;        /* decoded columns (shared) */
     uint8_t** nulls;
     uint8_t* types;
+    char** strblob;
+    int64_t** strend;
     GroupTable table;
     int error;
     int64_t rows_read;
@@ -607,6 +643,8 @@ static void* scan_worker(void* arg)
     ctx.col_vals = (const int64_t* const*)t->vals;
     ctx.col_nulls = (const uint8_t* const*)t->nulls;
     ctx.col_types = t->types;
+    ctx.col_str = (const char* const*)t->strblob;
+    ctx.col_str_end = (const int64_t* const*)t->strend;
     ctx.ncols = t->chunk->column_count;
     ctx.error = 0;
 
@@ -662,6 +700,20 @@ static int emit_group_row(const YtPlan* plan, GroupTable* t, int64_t g,
     }
 
     YtValue* dst = out->values + out->row_count * ncols;
+    if (!plan->project_count) {
+        /* copy string payloads into the caller's pool */
+        for (int i = 0; i < kc + ac; i++) {
+            if (rowvals[i].type == YT_VT_STRING) {
+                if (out->string_pool_used + rowvals[i].len > out->string_pool_capacity) {
+                    return YT_ERR_CAPACITY;
+                }
+                memcpy(out->string_pool + out->string_pool_used,
+                       rowvals[i].str, rowvals[i].len);
+                rowvals[i].str = out->string_pool + out->string_pool_used;
+                out->string_pool_used += rowvals[i].len;
+            }
+        }
+    }
     if (plan->project_count) {
         /* projection over the group row: column i refers to rowvals[i] */
         int64_t pv[32];
@@ -697,8 +749,13 @@ static int emit_group_row(const YtPlan* plan, GroupTable* t, int64_t g,
             dst[i].id = (uint16_t)i;
             dst[i].type = rowvals[i].type;
             dst[i].flags = 0;
-            dst[i].length = 0;
-            dst[i].data.bits = rowvals[i].bits;
+            if (rowvals[i].type == YT_VT_STRING) {
+                dst[i].length = rowvals[i].len;
+                dst[i].data.str = rowvals[i].str;
+            } else {
+                dst[i].length = 0;
+                dst[i].data.bits = rowvals[i].bits;
+            }
         }
     }
     out->row_count++;
@@ -719,11 +776,35 @@ int yto_execute(const YtPlan* plan, const YtChunk* chunk,
     int64_t** vals = calloc(ncols, sizeof(int64_t*));
     uint8_t** nulls = calloc(ncols, sizeof(uint8_t*));
     uint8_t* types = calloc(ncols, 1);
+    char** strblob = calloc(ncols, sizeof(char*));
+    int64_t** strend = calloc(ncols, sizeof(int64_t*));
     for (int c = 0; c < ncols; c++) {
-        vals[c] = malloc(sizeof(int64_t) * (n ? n : 1));
         nulls[c] = malloc(n ? n : 1);
         types[c] = (uint8_t)chunk->columns[c].value_type;
-        if (!vals[c] || !nulls[c]) { rc = YT_ERR_CAPACITY; goto done; }
+        if (types[c] == YT_VT_STRING) {
+            /* size: sum of encoded segment bytes is a lower bound only —
+             * decode with growth */
+            int64_t cap = 1024;
+            for (int s2 = 0; s2 < chunk->columns[c].segment_count; s2++) {
+                cap += chunk->columns[c].segments[s2].data_size * 4;
+            }
+            for (;;) {
+                strblob[c] = malloc(cap ? cap : 1);
+                strend[c] = malloc(sizeof(int64_t) * (n + 1));
+                if (!strblob[c] || !strend[c]) { rc = YT_ERR_CAPACITY; goto done; }
+                rc = yto_decode_string_column(&chunk->columns[c], n,
+                                              strblob[c], cap, strend[c], nulls[c]);
+                if (rc != YT_ERR_CAPACITY) break;
+                free(strblob[c]); free(strend[c]);
+                strblob[c] = NULL; strend[c] = NULL;
+                cap *= 4;
+            }
+            if (rc != YT_OK) { set_err(errbuf, errlen, "oracle: bad string segment"); goto done; }
+            vals[c] = calloc(n ? n : 1, sizeof(int64_t));
+            continue;
+        }
+        vals[c] = malloc(sizeof(int64_t) * (n ? n : 1));
+        if (!vals[c]) { rc = YT_ERR_CAPACITY; goto done; }
         rc = yto_decode_column(&chunk->columns[c], n, vals[c], nulls[c]);
         if (rc != YT_OK) { set_err(errbuf, errlen, "oracle: bad segment"); goto done; }
     }
@@ -737,6 +818,8 @@ int yto_execute(const YtPlan* plan, const YtChunk* chunk,
         ctx.col_vals = (const int64_t* const*)vals;
         ctx.col_nulls = (const uint8_t* const*)nulls;
         ctx.col_types = types;
+        ctx.col_str = (const char* const*)strblob;
+        ctx.col_str_end = (const int64_t* const*)strend;
         ctx.ncols = ncols;
         ctx.error = 0;
         int np = plan->project_count;
@@ -788,6 +871,8 @@ int yto_execute(const YtPlan* plan, const YtChunk* chunk,
             tasks[actual].vals = vals;
             tasks[actual].nulls = nulls;
             tasks[actual].types = types;
+            tasks[actual].strblob = strblob;
+            tasks[actual].strend = strend;
             gt_init(&tasks[actual].table, plan->key_count, plan->agg_count, 1024);
             actual++;
         }
@@ -842,8 +927,10 @@ done:
     for (int c = 0; c < ncols; c++) {
         if (vals) free(vals[c]);
         if (nulls) free(nulls[c]);
+        if (strblob) free(strblob[c]);
+        if (strend) free(strend[c]);
     }
-    free(vals); free(nulls); free(types);
+    free(vals); free(nulls); free(types); free(strblob); free(strend);
     return rc;
 }
 

@@ -74,3 +74,43 @@ def test_rle_with_nulls():
         block = [None] * 50 if i % 3 == 0 else ["blk%d" % i] * 50
         strs += block
     roundtrip(strs)
+
+
+def test_string_group_by_oracle():
+    """config-5 semantics on the oracle: GROUP BY string key + sum(double)
+    (reference: GROUP BY over string columns, ql_query_ut.cpp GroupByKeyTypes
+    key part; aggregates per udf/sum.c)."""
+    rng = np.random.default_rng(41)
+    n = 30_000
+    keyset = ["k%04d" % i for i in range(200)]
+    keys = [keyset[int(i)] for i in rng.integers(0, 200, n)]
+    kn = rng.random(n) < 0.02
+    keys = [None if kn[i] else keys[i] for i in range(n)]
+    vals = rng.random(n)
+    chunk = y.Chunk([y.encode_string(keys), y.encode_double(vals)], n)
+    plan = y.Plan(keys=[y.col(0)], aggs=[y.agg_sum(y.col(1)), y.agg_sum1()])
+    rows, st = y.oracle_execute(plan, chunk)
+
+    import collections
+    want_sum = collections.defaultdict(float)
+    want_cnt = collections.defaultdict(int)
+    for i in range(n):
+        k = keys[i].encode() if keys[i] is not None else None
+        want_sum[k] += vals[i]
+        want_cnt[k] += 1
+    assert len(rows) == len(want_cnt)
+    for k, sv, cv in rows:
+        assert cv == want_cnt[k]
+        assert abs(sv - want_sum[k]) < 1e-6 * max(abs(want_sum[k]), 1e-30)
+
+
+def test_string_filter_oracle():
+    """string comparisons (string_less_than semantics, udf/min.c:6-19)."""
+    strs = ["apple", "banana", None, "apricot", "b", "banana"]
+    # no string literals in the round-1 plan grammar: compare against a
+    # constant string COLUMN instead
+    chunk2 = y.Chunk([y.encode_string(strs), y.encode_string(["banana"] * 6)], 6)
+    plan = y.Plan(filter=y.col(0) == y.col(1),
+                  keys=[y.col(0)], aggs=[y.agg_sum1()])
+    rows, _ = y.oracle_execute(plan, chunk2)
+    assert rows == [(b"banana", 2)]

@@ -341,6 +341,8 @@ def rows_from_rowset(rs):
                 row.append(float(v.data.dbl))
             elif v.type == VT_BOOLEAN:
                 row.append(bool(v.data.bits))
+            elif v.type == VT_STRING:
+                row.append(C.string_at(v.data.str, v.length))
             else:
                 row.append(("?", v.type, v.data.bits))
         out.append(tuple(row))
@@ -357,10 +359,15 @@ def sort_rows(rows):
     return sorted(rows, key=key)
 
 
-def _mk_rowset(capacity, ncols_max=8):
+def _mk_rowset(capacity, ncols_max=8, pool_bytes=0):
     buf = (YtValue * (capacity * ncols_max))()
     rs = YtRowset(values=buf, capacity_rows=capacity)
     rs._buf = buf  # keep alive
+    if pool_bytes:
+        pool = C.create_string_buffer(pool_bytes)
+        rs.string_pool = C.cast(pool, C.c_char_p)
+        rs.string_pool_capacity = pool_bytes
+        rs._pool = pool
     return rs
 
 
@@ -369,7 +376,7 @@ def _mk_rowset(capacity, ncols_max=8):
 def oracle_execute(plan, chunk, nthreads=1, expect_error=False):
     """TEST/BASELINE ONLY — runs the CPU oracle restatement."""
     ch = chunk.c_host()
-    rs = _mk_rowset(max(chunk.row_count + 16, 1 << 16))
+    rs = _mk_rowset(max(chunk.row_count + 16, 1 << 16), pool_bytes=32 << 20)
     st = YtStatistics()
     err = C.create_string_buffer(256)
     rc = _abi.oracle_lib().yto_execute(C.byref(plan.c), C.byref(ch), C.byref(rs),
