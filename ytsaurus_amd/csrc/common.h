@@ -59,6 +59,8 @@ constexpr int kMaxAggs = 4;
 
 /* One compiled plan for the device: programs are concatenated postfix
  * streams with (offset,len) per role. */
+constexpr int kMaxProj = 8;
+
 struct DevPlan {
     int32_t ncols;
     uint8_t col_types[kMaxCols];       /* YT_VT_* */
@@ -67,8 +69,17 @@ struct DevPlan {
     int32_t agg_count;
     int32_t agg_func[kMaxAggs];        /* YT_AGG_* */
     int32_t agg_off[kMaxAggs], agg_len[kMaxAggs];
+    int32_t proj_count;                /* scan-without-aggregation mode */
+    int32_t proj_off[kMaxProj], proj_len[kMaxProj];
     PInst prog[kMaxProg];
     int32_t prog_len;
+};
+
+/* device output value for the scan+project path (16 B) */
+struct DevOutVal {
+    uint64_t bits;
+    uint32_t type;     /* YT_VT_*; YT_VT_NULL for null */
+    uint32_t pad_;
 };
 
 /* Group hash-table slot layout (generic path):

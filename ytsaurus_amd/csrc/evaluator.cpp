@@ -36,6 +36,9 @@ hipError_t ytql_launch_bucket_agg(const void*, const unsigned long long*, int64_
                                   OutGroup*, unsigned long long*, int64_t,
                                   TableHdr*, int, int, int, int,
                                   uint64_t, uint64_t, hipStream_t);
+hipError_t ytql_launch_scan_project(const DevPlan*, const DevSeg*, const SegEx*,
+                                    const int32_t*, const int32_t*, int64_t,
+                                    DevOutVal*, uint8_t*, unsigned*, hipStream_t);
 hipError_t ytql_launch_scan_generic(const DevPlan*, const DevSeg*, const SegEx*,
                                     const int32_t*, const int32_t*, int64_t,
                                     TableHdr*, unsigned long long*, unsigned*, hipStream_t);
@@ -244,6 +247,12 @@ static int build_devplan(const YtPlan* plan, const YtChunk* chunk, DevPlan* p,
     for (int c = 0; c < p->ncols; c++) p->col_types[c] = (uint8_t)chunk->columns[c].value_type;
 
     if (plan->key_count > 1) { set_err(errbuf, errlen, "GPU path: at most 1 group key this round"); return YT_ERR_UNSUPPORTED; }
+    if (plan->agg_count == 0) {
+        if (plan->project_count < 1 || plan->project_count > kMaxProj) {
+            set_err(errbuf, errlen, "scan mode needs 1..8 projections");
+            return YT_ERR_UNSUPPORTED;
+        }
+    } else
     if (plan->agg_count < 1 || plan->agg_count > kMaxAggs) { set_err(errbuf, errlen, "need 1..4 aggregates"); return YT_ERR_UNSUPPORTED; }
     for (int a = 0; a < plan->agg_count; a++) {
         int f = plan->aggs[a]->func;
@@ -274,6 +283,15 @@ static int build_devplan(const YtPlan* plan, const YtChunk* chunk, DevPlan* p,
             rc = compile_expr(plan->aggs[a]->arg, p, &p->agg_len[a], errbuf, errlen);
             if (rc) return rc;
             p->agg_len[a] = p->prog_len - p->agg_off[a];
+        }
+    }
+    if (plan->agg_count == 0) {
+        p->proj_count = plan->project_count;
+        for (int pj = 0; pj < plan->project_count; pj++) {
+            p->proj_off[pj] = p->prog_len;
+            rc = compile_expr(plan->projects[pj], p, &p->proj_len[pj], errbuf, errlen);
+            if (rc) return rc;
+            p->proj_len[pj] = p->prog_len - p->proj_off[pj];
         }
     }
     return YT_OK;
@@ -1146,6 +1164,94 @@ static int emit_rows(const YtPlan* plan, const YtChunk* chunk,
     return rc;
 }
 
+
+/* scan + filter + project: order-preserving (MakeCodegenProjectOp +
+ * WriteOpHelper, registry.cpp:1999-2047). The device writes per-row values
+ * and a pass mask; the host compacts in row order. */
+static int run_scan_project(const YtPlan* plan, const YtChunk* chunk,
+                            const YtExecOptions* options, const DevPlan* dp,
+                            YtRowset* output, YtStatistics* stats, double tw0,
+                            char* errbuf, size_t errlen)
+{
+    int rc = YT_OK;
+    DeviceRun R2;
+    R2.stream = (hipStream_t)(uintptr_t)options->stream;
+    unsigned mw = 0;
+    rc = setup_chunk(chunk, &R2, &mw, errbuf, errlen);
+    if (rc) return rc;
+    int64_t n = chunk->row_count;
+    output->row_count = 0;
+    output->column_count = plan->project_count;
+    if (n == 0 || R2.nsegs == 0) return YT_OK;
+    {
+        DevOutVal* d_out = nullptr;
+        uint8_t* d_pass = nullptr;
+        DevOutVal* h_out = nullptr;
+        uint8_t* h_pass = nullptr;
+        HIP_CHECK(pool_alloc(&d_out, sizeof(DevOutVal) * n * plan->project_count));
+        HIP_CHECK(pool_alloc(&d_pass, (size_t)n));
+        HIP_CHECK(hipMemsetAsync(R2.d_err, 0, sizeof(unsigned), R2.stream));
+        hipEvent_t e0, e1;
+        HIP_CHECK(hipEventCreate(&e0));
+        HIP_CHECK(hipEventCreate(&e1));
+        HIP_CHECK(hipEventRecord(e0, R2.stream));
+        HIP_CHECK(ytql_launch_scan_project(dp, R2.d_segs, R2.d_segex, R2.d_off,
+                                           R2.d_cnt, n, d_out, d_pass, R2.d_err,
+                                           R2.stream));
+        HIP_CHECK(hipEventRecord(e1, R2.stream));
+        HIP_CHECK(pool_alloc_host(&h_out, sizeof(DevOutVal) * n * plan->project_count));
+        HIP_CHECK(pool_alloc_host(&h_pass, (size_t)n));
+        HIP_CHECK(hipMemcpyAsync(h_out, d_out, sizeof(DevOutVal) * n * plan->project_count,
+                                 hipMemcpyDeviceToHost, R2.stream));
+        HIP_CHECK(hipMemcpyAsync(h_pass, d_pass, (size_t)n,
+                                 hipMemcpyDeviceToHost, R2.stream));
+        HIP_CHECK(hipStreamSynchronize(R2.stream));
+        float ms = 0;
+        HIP_CHECK(hipEventElapsedTime(&ms, e0, e1));
+        hipEventDestroy(e0);
+        hipEventDestroy(e1);
+        unsigned kerr = 0;
+        HIP_CHECK(hipMemcpy(&kerr, R2.d_err, sizeof(unsigned), hipMemcpyDeviceToHost));
+        if (kerr) {
+            g_pool.put(d_out); g_pool.put(d_pass);
+            g_pool.put(h_out); g_pool.put(h_pass);
+            set_err(errbuf, errlen, kerr == YT_ERR_DIV_ZERO ? "Division by zero" : "expression error");
+            return (int)kerr;
+        }
+        int np = plan->project_count;
+        for (int64_t r2 = 0; r2 < n; r2++) {
+            if (!h_pass[r2]) continue;
+            if (output->row_count >= output->capacity_rows) {
+                g_pool.put(d_out); g_pool.put(d_pass);
+                g_pool.put(h_out); g_pool.put(h_pass);
+                return YT_ERR_CAPACITY;
+            }
+            YtValue* dst = output->values + output->row_count * np;
+            for (int pj = 0; pj < np; pj++) {
+                const DevOutVal& o = h_out[r2 * np + pj];
+                dst[pj].id = (uint16_t)pj;
+                dst[pj].type = (uint8_t)o.type;
+                dst[pj].flags = 0;
+                dst[pj].length = 0;
+                dst[pj].data.bits = o.bits;
+            }
+            output->row_count++;
+        }
+        if (stats) {
+            stats->rows_read = n;
+            stats->rows_written = output->row_count;
+            stats->kernel_scan_ms += ms;
+            stats->kernel_scan_launches += 1;
+            stats->execute_time_ms = now_ms() - tw0;
+        }
+        g_pool.put(d_out); g_pool.put(d_pass);
+        g_pool.put(h_out); g_pool.put(h_pass);
+    }
+    return YT_OK;
+fail:
+    return rc;
+}
+
 /* ------------------------------------------------------------------ */
 /* public entries                                                      */
 
@@ -1163,14 +1269,24 @@ extern "C" int yt_gpu_query_execute(
     if (!options) options = &defopt;
     if (stats) memset(stats, 0, sizeof(*stats));
 
-    if (plan->agg_count == 0) {
-        set_err(errbuf, errlen, "GPU path: scan-without-aggregation not built this round (oracle covers it)");
+    if (plan->agg_count == 0 && plan->project_count == 0) {
+        set_err(errbuf, errlen, "empty plan");
+        return YT_ERR_INVALID_PLAN;
+    }
+    if (plan->agg_count == 0 && chunk->row_count > (int64_t)1 << 24) {
+        set_err(errbuf, errlen,
+                "GPU scan-project materializes per-row output; capped at 16M rows this round");
         return YT_ERR_UNSUPPORTED;
     }
 
     DevPlan dp;
     rc = build_devplan(plan, chunk, &dp, errbuf, errlen);
     if (rc) return rc;
+
+    if (plan->agg_count == 0) {
+        return run_scan_project(plan, chunk, options, &dp, output, stats, tw0,
+                                errbuf, errlen);
+    }
 
     FastShape fs;
     analyze_fast(plan, chunk, &fs);

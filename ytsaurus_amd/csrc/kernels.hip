@@ -630,6 +630,46 @@ k_scan_generic(DevPlan p, const DevSeg* segs, const SegEx* segex,
     if (c.error) atomicMax(error_out, c.error);
 }
 
+/* scan + filter + project (no aggregation): preserves input row order —
+ * MakeCodegenProjectOp + WriteOpHelper semantics (registry.cpp:1999-2047)
+ * with row-at-a-time evaluation replaced by one thread per row. Rows that
+ * fail the filter leave pass[j] = 0; the host compacts in row order. */
+__global__ void __launch_bounds__(256)
+k_scan_project(DevPlan p, const DevSeg* segs, const SegEx* segex,
+               const int32_t* col_seg_off, const int32_t* col_seg_cnt,
+               int64_t row_count, DevOutVal* out, uint8_t* pass,
+               unsigned* error_out)
+{
+    ColCtx c;
+    c.segs = segs;
+    c.segex = segex;
+    c.col_seg_off = col_seg_off;
+    c.col_seg_cnt = col_seg_cnt;
+    c.error = 0;
+
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         r < row_count; r += stride) {
+        c.row = r;
+        pass[r] = 0;
+        if (p.filter_len) {
+            DVal f = eval_prog(p, c, p.filter_off, p.filter_len);
+            if (f.null_ || f.bits == 0) continue;
+        }
+        pass[r] = 1;
+        for (int pj = 0; pj < p.proj_count; pj++) {
+            DVal v = eval_prog(p, c, p.proj_off[pj], p.proj_len[pj]);
+            DevOutVal o;
+            o.bits = v.bits;
+            o.type = v.null_ ? YT_VT_NULL : v.type;
+            o.pad_ = 0;
+            out[r * p.proj_count + pj] = o;
+        }
+        if (c.error) { atomicMax(error_out, c.error); return; }
+    }
+    if (c.error) atomicMax(error_out, c.error);
+}
+
 /* ------------------------------------------------------------------ */
 /* fast fused kernel — DirectDense segments, direct-column shapes      */
 /*                                                                     */
@@ -1287,6 +1327,23 @@ hipError_t ytql_launch_bucket_agg(const void* recs, const unsigned long long* cu
                        nrecs, ncursors, nbucket_stride,
                        out, out_counter, out_cap, th, sum_slot, agg_count,
                        packed_mode, bits_k, gmin_k, gmin_v);
+    return hipGetLastError();
+}
+
+hipError_t ytql_launch_scan_project(const DevPlan* p, const DevSeg* segs,
+                                    const SegEx* segex,
+                                    const int32_t* col_seg_off,
+                                    const int32_t* col_seg_cnt,
+                                    int64_t row_count, DevOutVal* out,
+                                    uint8_t* pass, unsigned* error_out,
+                                    hipStream_t st)
+{
+    int block = 256;
+    int64_t want = (row_count + block - 1) / block;
+    int grid = (int)(want > 4096 ? 4096 : (want > 0 ? want : 1));
+    hipLaunchKernelGGL(k_scan_project, dim3(grid), dim3(block), 0, st,
+                       *p, segs, segex, col_seg_off, col_seg_cnt, row_count,
+                       out, pass, error_out);
     return hipGetLastError();
 }
 
