@@ -859,9 +859,12 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
                  uint64_t* nrecs)
 {
     /* Canonical column slots: 0 = filter, 1 = key, 2 = value.
-     * Values are decoded straight from global memory (branchless funnel,
-     * L1/L2 absorb window overlap); LDS holds only the per-tile bucket
-     * histograms. See k_scan_fast for the measured rationale. */
+     * Values are decoded straight from global memory (branchless funnel;
+     * the twice-read key column is staged in LDS); LDS otherwise holds only
+     * the per-tile bucket histograms.
+     * Pass 1 counts per bucket with fire-and-forget LDS adds; pass 2 claims
+     * each row's offset at write time — no per-row offset state lives in
+     * registers, so tiles can be large and the loops unroll freely. */
     extern __shared__ __attribute__((aligned(16))) char smem[];
     const int tid = threadIdx.x;
 
@@ -874,8 +877,8 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
     const bool has_filter = pp.filter_idx >= 0;
     const bool has_val = pp.val_idx >= 0;
     /* per-XCD sub-buckets: workgroups on different XCDs write disjoint
-     * record regions, so no cache line is filled by two L2s (cross-XCD
-     * partial-line sharing measured 3.7x write amplification) */
+     * record regions (cross-XCD partial-line sharing measured 3.7x write
+     * amplification) */
     const int sub = blockIdx.x & 7;
 
     for (int tile = blockIdx.x; tile < pp.ntiles; tile += gridDim.x) {
@@ -936,90 +939,98 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
         __syncthreads();
 
         const int R = (pp.tile_rows + 255) / 256;
-        uint32_t row_b[32];
-        uint32_t row_off[32];
+        /* pass 1: COUNT per bucket (no returns, no per-row state) */
         #pragma unroll 4
         for (int i = 0; i < R; i++) {
-            row_b[i] = 0xFFFFFFFFu;
             int64_t j = t0 + (int64_t)i * 256 + tid;
             if (j >= t1) continue;
-
             if (has_filter) {
                 if (fbm && bm_get(fbm, j)) continue;
                 int64_t v = zz_dec(fmin + bp_gl(fwords, fmask, fwd, j));
                 if (v < pp.filter_lo || v > pp.filter_hi) continue;
             }
-
-            uint64_t key = 0;
             int key_null = kbm && bm_get(kbm, j);
-            if (!key_null) {
-                key = (uint64_t)zz_dec(kmin + (bp_get_win(klds, kwd, j, kW0) & kmask));
-            }
-            int val_null = has_val ? (vbm && bm_get(vbm, j)) : 1;
-
-            if (key_null || key == kEmptyKey) {
-                int side = key_null ? 1 : 0;
-                th->side_used[side] = 1;
-                atomicAdd((unsigned long long*)&th->side_cnt[side], 1ULL);
-                if (pp.sum_slot >= 0 && !val_null) {
-                    uint64_t val = (uint64_t)zz_dec(vmin + bp_gl(vwords, vmask, vwd, j));
-                    atomicAdd((unsigned long long*)&th->side_agg[side][2 * pp.sum_slot], val);
-                    atomicAdd((unsigned long long*)&th->side_agg[side][2 * pp.sum_slot + 1], 1ULL);
-                }
-                continue;
-            }
-
+            if (key_null) continue;              /* side rows counted in pass 2 */
+            uint64_t key = (uint64_t)zz_dec(kmin + (bp_get_win(klds, kwd, j, kW0) & kmask));
+            if (key == kEmptyKey) continue;
             unsigned b = (unsigned)(mix64(key) >> 40) & (kNB - 1);
-            if (has_val && val_null) {
-                row_b[i] = b | 0x80000000u;
-                row_off[i] = atomicAdd(&nhist[b], 1u);
-            } else {
-                row_b[i] = b;
-                row_off[i] = atomicAdd(&hist[b], 1u);
-            }
+            if (has_val && vbm && bm_get(vbm, j)) atomicAdd(&nhist[b], 1u);
+            else atomicAdd(&hist[b], 1u);
         }
         __syncthreads();
+        /* reserve global space per bucket-sub (1 atomic per bucket/tile),
+         * then reuse hist/nhist as the pass-2 claim counters */
         for (int i = tid; i < kNB; i += 256) {
             unsigned c = hist[i];
             if (c) {
                 unsigned long long base = atomicAdd(&cursors[i * 8 + sub], (unsigned long long)c);
                 if ((int64_t)(base + c) > pp.bucket_stride) { th->overflow = 1; base = 0; }
                 gbase[i] = (unsigned)base;
+                hist[i] = 0;
             }
             unsigned nc = nhist[i];
             if (nc) {
                 unsigned long long base = atomicAdd(&ncursors[i * 8 + sub], (unsigned long long)nc);
                 if ((int64_t)(base + nc) > pp.nbucket_stride) { th->overflow = 1; base = 0; }
                 ngbase[i] = (unsigned)base;
+                nhist[i] = 0;
             }
         }
         __syncthreads();
         uint64_t* recs8 = (uint64_t*)recs;
         if (th->overflow != 1) {
+            /* pass 2: claim offset, decode, write */
             #pragma unroll 4
             for (int i = 0; i < R; i++) {
-                if (row_b[i] == 0xFFFFFFFFu) continue;
                 int64_t j = t0 + (int64_t)i * 256 + tid;
-                uint64_t kzzfull = kmin + (bp_get_win(klds, kwd, j, kW0) & kmask);
-                unsigned b = row_b[i] & 0x7FFFFFFFu;
+                if (j >= t1) continue;
+                if (has_filter) {
+                    if (fbm && bm_get(fbm, j)) continue;
+                    int64_t v = zz_dec(fmin + bp_gl(fwords, fmask, fwd, j));
+                    if (v < pp.filter_lo || v > pp.filter_hi) continue;
+                }
+                int key_null = kbm && bm_get(kbm, j);
+                uint64_t kzzfull = 0;
+                uint64_t key = 0;
+                if (!key_null) {
+                    kzzfull = kmin + (bp_get_win(klds, kwd, j, kW0) & kmask);
+                    key = (uint64_t)zz_dec(kzzfull);
+                }
+                int val_null = has_val ? (vbm && bm_get(vbm, j)) : 1;
+
+                if (key_null || key == kEmptyKey) {
+                    int side = key_null ? 1 : 0;
+                    th->side_used[side] = 1;
+                    atomicAdd((unsigned long long*)&th->side_cnt[side], 1ULL);
+                    if (pp.sum_slot >= 0 && !val_null) {
+                        uint64_t val = (uint64_t)zz_dec(vmin + bp_gl(vwords, vmask, vwd, j));
+                        atomicAdd((unsigned long long*)&th->side_agg[side][2 * pp.sum_slot], val);
+                        atomicAdd((unsigned long long*)&th->side_agg[side][2 * pp.sum_slot + 1], 1ULL);
+                    }
+                    continue;
+                }
+
+                unsigned b = (unsigned)(mix64(key) >> 40) & (kNB - 1);
                 int64_t sb = (int64_t)b * 8 + sub;
-                if (row_b[i] & 0x80000000u) {
-                    nrecs[sb * pp.nbucket_stride + ngbase[b] + row_off[i]] =
-                        (uint64_t)zz_dec(kzzfull);
+                if (has_val && val_null) {
+                    unsigned off = atomicAdd(&nhist[b], 1u);
+                    nrecs[sb * pp.nbucket_stride + ngbase[b] + off] = key;
                 } else if (pp.packed_mode) {
+                    unsigned off = atomicAdd(&hist[b], 1u);
                     uint64_t rec = kzzfull - pp.gmin_k;
                     if (has_val) {
                         uint64_t vzz = (vmin + bp_gl(vwords, vmask, vwd, j)) - pp.gmin_v;
                         rec |= vzz << pp.bits_k;
                     }
-                    recs8[sb * pp.bucket_stride + gbase[b] + row_off[i]] = rec;
+                    recs8[sb * pp.bucket_stride + gbase[b] + off] = rec;
                 } else {
+                    unsigned off = atomicAdd(&hist[b], 1u);
                     uint64_t val = 0;
                     if (has_val) {
                         val = (uint64_t)zz_dec(vmin + bp_gl(vwords, vmask, vwd, j));
                     }
-                    recs[sb * pp.bucket_stride + gbase[b] + row_off[i]] =
-                        make_ulonglong2((uint64_t)zz_dec(kzzfull), val);
+                    recs[sb * pp.bucket_stride + gbase[b] + off] =
+                        make_ulonglong2(key, val);
                 }
             }
         }
