@@ -155,7 +155,7 @@ struct FastParams {
  * Phase B (k_bucket_agg): one workgroup per bucket aggregates its records
  * in an LDS open-addressing table and emits compacted OutGroups. */
 constexpr int kNB = 1024;          /* partition buckets */
-constexpr int kHSlots = 2048;      /* LDS table slots per bucket (48 KB table -> 3 WGs/CU) */
+constexpr int kHSlots = 2048;      /* LDS table slots per bucket (48 KB -> 3 WGs/CU) */
 constexpr uint64_t kEmptyKey = 0x8000000000000000ULL;  /* INT64_MIN bits */
 
 struct PartParams {
