@@ -304,3 +304,42 @@ def test_scan_project_large(cuda):
     want, _ = y.oracle_execute(plan, chunk)
     assert got == want         # exact, order-preserving
     assert st.rows_written == len(want)
+
+
+def test_input_output_row_limits(cuda):
+    """TExecutionContext InputRowLimit / OutputRowLimit semantics
+    (registry.cpp:259-265, :297-305): first-N-rows scan with
+    incomplete_input, output soft-stop with incomplete_output."""
+    import ctypes as C
+    from ytsaurus_amd import _abi
+    from ytsaurus_amd.api import _mk_rowset
+    rng = np.random.default_rng(15)
+    n = 300_000
+    keys = rng.integers(0, 1000, n, dtype=np.int64)
+    vals = rng.integers(0, 100, n, dtype=np.int64)
+    chunk = y.Chunk([y.encode_int64(keys), y.encode_int64(vals)], n)
+    plan = group_plan()
+    dev = chunk.c_device(cuda)
+
+    # input limit: equals a full run over the first 200k rows
+    opts = _abi.YtExecOptions(input_row_limit=200_000, max_groups_hint=4096)
+    rs = _mk_rowset(1 << 16)
+    st = _abi.YtStatistics()
+    err = C.create_string_buffer(512)
+    rc = _abi.gpu_lib().yt_gpu_query_execute(
+        C.byref(plan.c), C.byref(dev), C.byref(opts), C.byref(rs), C.byref(st), err, 512)
+    assert rc == 0, err.value
+    assert st.incomplete_input == 1 and st.rows_read == 200_000
+    got = y.rows_from_rowset(rs)
+    sub = y.Chunk([y.encode_int64(keys[:200_000]), y.encode_int64(vals[:200_000])], 200_000)
+    want, _ = y.oracle_execute(plan, sub)
+    assert y.sort_rows(got) == y.sort_rows(want)
+
+    # output limit: at most 10 rows, flagged incomplete
+    opts = _abi.YtExecOptions(output_row_limit=10, max_groups_hint=4096)
+    rs = _mk_rowset(1 << 16)
+    rc = _abi.gpu_lib().yt_gpu_query_execute(
+        C.byref(plan.c), C.byref(dev), C.byref(opts), C.byref(rs), C.byref(st), err, 512)
+    assert rc == 0, err.value
+    assert rs.row_count == 10
+    assert st.incomplete_output == 1

@@ -597,33 +597,52 @@ static uint64_t next_pow2(uint64_t x)
     return p;
 }
 
-/* builds device segment descriptors; returns YT status */
+/* builds device segment descriptors; returns YT status.
+ * input_row_limit > 0 truncates the scan to the first N rows — exactly the
+ * reference's InputRowLimit interrupt semantics (rows are consumed in row
+ * order; registry.cpp:259-265, TInterruptedIncompleteException). */
 static int setup_chunk(const YtChunk* chunk, DeviceRun* R, unsigned* maxw_out,
+                       int64_t input_row_limit, int* clamped,
                        char* errbuf, size_t errlen)
 {
     int rc = YT_OK;
     int ncols = chunk->column_count;
+    if (clamped) *clamped = 0;
     R->h_off.resize(ncols);
     R->h_cnt.resize(ncols);
     for (int c = 0; c < ncols; c++) {
         const YtColumn& col = chunk->columns[c];
         R->h_off[c] = (int32_t)R->h_segs.size();
-        R->h_cnt[c] = col.segment_count;
+        int32_t kept = 0;
         int64_t row = 0;
         for (int s = 0; s < col.segment_count; s++) {
             const YtSegment& seg = col.segments[s];
+            int32_t rows_here = seg.row_count;
+            if (input_row_limit > 0) {
+                if (row >= input_row_limit) {
+                    if (clamped) *clamped = 1;
+                    row += seg.row_count;
+                    continue;
+                }
+                if (row + rows_here > input_row_limit) {
+                    rows_here = (int32_t)(input_row_limit - row);
+                    if (clamped) *clamped = 1;
+                }
+            }
             DevSeg d;
             d.type = seg.type;
             d.is_signed = (col.value_type == YT_VT_INT64);
             d.start_row = row;
-            d.row_count = seg.row_count;
+            d.row_count = rows_here;
             d.col = c;
             d.min_value = seg.min_value;
             d.blob = (const uint64_t*)seg.data;
             d.blob_bytes = seg.data_size;
             R->h_segs.push_back(d);
+            kept++;
             row += seg.row_count;
         }
+        R->h_cnt[c] = kept;
         if (row != chunk->row_count) {
             set_err(errbuf, errlen, "segment row counts do not sum to chunk rows");
             return YT_ERR_INVALID_CHUNK;
@@ -768,8 +787,10 @@ static int run_partitioned(const YtPlan* plan, const YtChunk* chunk,
         }
     }
 
-    const int64_t rows = chunk->row_count;
-    int32_t seg0_rows = chunk->columns[0].segments[0].row_count;
+    int64_t rows = chunk->row_count;
+    if (options->input_row_limit > 0 && options->input_row_limit < rows)
+        rows = options->input_row_limit;
+    int32_t seg0_rows = R->h_segs[R->h_off[fs->key_col]].row_count;
 
     /* LDS: 4*kNB u32 histogram structures + staged columns (value column's
      * packed words are NOT staged — read from global in the write pass) */
@@ -787,8 +808,8 @@ static int run_partitioned(const YtPlan* plan, const YtChunk* chunk,
     pp.tile_rows = tile_rows;
     pp.tiles_per_seg = (seg0_rows + tile_rows - 1) / tile_rows;
     {
-        int nseg = chunk->columns[0].segment_count;
-        int32_t last_rows = chunk->columns[0].segments[nseg - 1].row_count;
+        int nseg = R->h_cnt[fs->key_col];
+        int32_t last_rows = R->h_segs[R->h_off[fs->key_col] + nseg - 1].row_count;
         pp.ntiles = (nseg - 1) * pp.tiles_per_seg + (last_rows + tile_rows - 1) / tile_rows;
     }
     /* per (bucket, XCD) sub-streams: 8x more cursors, 1/8 the rows each */
@@ -951,7 +972,7 @@ static int run_scan(const YtPlan* plan, const YtChunk* chunk,
         }
         if (fs->key_col >= 0 && R->col_null_flags[fs->key_col]) fp.stage_bm_mask |= 1 << 5;
 
-        int32_t seg0_rows = chunk->columns[0].segments[0].row_count;
+        int32_t seg0_rows = R->h_segs[R->h_off[used[1]]].row_count;
 
         /* no LDS staging: tile size only shapes the grid */
         size_t lds = 0;
@@ -962,8 +983,8 @@ static int run_scan(const YtPlan* plan, const YtChunk* chunk,
         fp.tile_rows = tile_rows;
         fp.tiles_per_seg = (seg0_rows + tile_rows - 1) / tile_rows;
         {
-            int nseg = chunk->columns[0].segment_count;
-            int32_t last_rows = chunk->columns[0].segments[nseg - 1].row_count;
+            int nseg = R->h_cnt[used[1]];
+            int32_t last_rows = R->h_segs[R->h_off[used[1]] + nseg - 1].row_count;
             fp.ntiles = (nseg - 1) * fp.tiles_per_seg + (last_rows + tile_rows - 1) / tile_rows;
         }
 
@@ -984,8 +1005,11 @@ static int run_scan(const YtPlan* plan, const YtChunk* chunk,
         HIP_CHECK(hipEventRecord(ev1, R->stream));
     } else {
         HIP_CHECK(hipEventRecord(ev0, R->stream));
+        int64_t gen_rows = chunk->row_count;
+        if (options->input_row_limit > 0 && options->input_row_limit < gen_rows)
+            gen_rows = options->input_row_limit;
         HIP_CHECK(ytql_launch_scan_generic(dp, R->d_segs, R->d_segex, R->d_off, R->d_cnt,
-                                           chunk->row_count, R->d_th, R->d_slots,
+                                           gen_rows, R->d_th, R->d_slots,
                                            R->d_err, R->stream));
         HIP_CHECK(hipEventRecord(ev1, R->stream));
     }
@@ -1065,6 +1089,7 @@ static int emit_rows(const YtPlan* plan, const YtChunk* chunk,
                      const OutGroup* groups, int64_t ngroups,
                      const TableHdr& th, int has_any_row_global,
                      const uint64_t* gaccum, int used_fast_global,
+                     int64_t output_row_limit, int* out_limited,
                      YtRowset* output, char* errbuf, size_t errlen)
 {
     uint8_t col_types[kMaxCols];
@@ -1086,6 +1111,11 @@ static int emit_rows(const YtPlan* plan, const YtChunk* chunk,
 
     auto emit = [&](uint64_t key_bits, int key_null, uint64_t cnt,
                     const uint64_t* ab, const uint64_t* an) -> int {
+        if (output_row_limit > 0 && output->row_count >= output_row_limit) {
+            /* OutputRowLimit: soft stop — registry.cpp:297-305 WriteRow */
+            if (out_limited) *out_limited = 1;
+            return -1;
+        }
         if (output->row_count >= output->capacity_rows) return YT_ERR_CAPACITY;
         HVal row[2 + kMaxAggs];
         int nrow = 0;
@@ -1141,12 +1171,14 @@ static int emit_rows(const YtPlan* plan, const YtChunk* chunk,
             }
         }
         (void)has_any_row_global;
-        return emit(0, 0, cnt, ab, an);
+        rc = emit(0, 0, cnt, ab, an);
+        return rc == -1 ? YT_OK : rc;
     }
 
     for (int64_t i = 0; i < ngroups; i++) {
         const OutGroup& g = groups[i];
         rc = emit(g.key_bits, (int)(g.key_meta & 1), g.cnt, g.agg_bits, g.agg_nonnull);
+        if (rc == -1) return YT_OK;
         if (rc) return rc;
     }
     /* side groups: the in-table sentinel key, then the null key */
@@ -1158,6 +1190,7 @@ static int emit_rows(const YtPlan* plan, const YtChunk* chunk,
             an[a] = th.side_agg[side][2 * a + 1];
         }
         rc = emit(th.side_key_bits[side], side == 1, th.side_cnt[side], ab, an);
+        if (rc == -1) return YT_OK;
         if (rc) return rc;
     }
     (void)errbuf; (void)errlen;
@@ -1177,9 +1210,14 @@ static int run_scan_project(const YtPlan* plan, const YtChunk* chunk,
     DeviceRun R2;
     R2.stream = (hipStream_t)(uintptr_t)options->stream;
     unsigned mw = 0;
-    rc = setup_chunk(chunk, &R2, &mw, errbuf, errlen);
+    int in_clamped = 0;
+    rc = setup_chunk(chunk, &R2, &mw, options->input_row_limit, &in_clamped,
+                     errbuf, errlen);
     if (rc) return rc;
+    (void)in_clamped;
     int64_t n = chunk->row_count;
+    if (options->input_row_limit > 0 && options->input_row_limit < n)
+        n = options->input_row_limit;
     output->row_count = 0;
     output->column_count = plan->project_count;
     if (n == 0 || R2.nsegs == 0) return YT_OK;
@@ -1221,6 +1259,11 @@ static int run_scan_project(const YtPlan* plan, const YtChunk* chunk,
         int np = plan->project_count;
         for (int64_t r2 = 0; r2 < n; r2++) {
             if (!h_pass[r2]) continue;
+            if (options->output_row_limit > 0 &&
+                output->row_count >= options->output_row_limit) {
+                if (stats) stats->incomplete_output = 1;
+                break;
+            }
             if (output->row_count >= output->capacity_rows) {
                 g_pool.put(d_out); g_pool.put(d_pass);
                 g_pool.put(h_out); g_pool.put(h_pass);
@@ -1299,7 +1342,9 @@ extern "C" int yt_gpu_query_execute(
     }
 
     unsigned maxw = 0;
-    rc = setup_chunk(chunk, &R, &maxw, errbuf, errlen);
+    int in_clamped = 0;
+    rc = setup_chunk(chunk, &R, &maxw, options->input_row_limit, &in_clamped,
+                     errbuf, errlen);
     if (rc) return rc;
 
     if (chunk->row_count == 0 || R.nsegs == 0) {
@@ -1346,13 +1391,20 @@ extern "C" int yt_gpu_query_execute(
             HIP_CHECK(hipMemcpy(gaccum.data(), R.d_gaccum,
                                 sizeof(uint64_t) * (1 + 2 * kMaxAggs), hipMemcpyDeviceToHost));
         }
+        int out_limited = 0;
         rc = emit_rows(plan, chunk, hgroups, ngroups, th, 0, gaccum.data(),
-                       fs.valid && fs.key_col < 0, output, errbuf, errlen);
+                       fs.valid && fs.key_col < 0,
+                       options->output_row_limit, &out_limited,
+                       output, errbuf, errlen);
         g_pool.put(hgroups);
         if (rc) return rc;
+        if (out_limited && stats) stats->incomplete_output = 1;
     }
     if (stats) {
-        stats->rows_read = chunk->row_count;
+        stats->rows_read = (options->input_row_limit > 0 &&
+                            options->input_row_limit < chunk->row_count)
+            ? options->input_row_limit : chunk->row_count;
+        stats->incomplete_input = in_clamped;
         int64_t bytes = 0;
         for (int c = 0; c < chunk->column_count; c++)
             for (int s = 0; s < chunk->columns[c].segment_count; s++)
@@ -1361,7 +1413,7 @@ extern "C" int yt_gpu_query_execute(
         stats->rows_written = output->row_count;
         stats->grouped_row_count = (int64_t)th.ngroups
             + (plan->key_count ? (int64_t)(th.side_used[0] + th.side_used[1]) : 0);
-        stats->incomplete_output = (th.overflow == 2);
+        stats->incomplete_output |= (th.overflow == 2);
         stats->execute_time_ms = now_ms() - tw0;
     }
     return YT_OK;
@@ -1403,7 +1455,9 @@ extern "C" int yt_gpu_query_partial(
     R.stream = (hipStream_t)(uintptr_t)options->stream;
 
     unsigned maxw = 0;
-    rc = setup_chunk(chunk, &R, &maxw, errbuf, errlen);
+    int in_clamped = 0;
+    rc = setup_chunk(chunk, &R, &maxw, options->input_row_limit, &in_clamped,
+                     errbuf, errlen);
     if (rc) return rc;
     if (chunk->row_count == 0 || R.nsegs == 0) {
         for (int p = 0; p < partition_count; p++) part_counts[p] = 0;
@@ -1548,9 +1602,12 @@ extern "C" int yt_gpu_merge_states(
         fake.row_count = 0;
         fake.column_count = kMaxCols;
         fake.columns = cols.data();
+        int out_limited = 0;
         rc = emit_rows(plan, &fake, groups.data(), (int64_t)groups.size(), th, 0,
-                       nullptr, 0, output, errbuf, errlen);
+                       nullptr, 0, options->output_row_limit, &out_limited,
+                       output, errbuf, errlen);
         if (rc) return rc;
+        if (out_limited && stats) stats->incomplete_output = 1;
         if (stats) {
             stats->rows_written = output->row_count;
             stats->grouped_row_count = ngroups + th.side_used[0] + th.side_used[1];
