@@ -76,6 +76,7 @@ enum {
     YT_SEG_DIRECT_RLE = 2,
     YT_SEG_DIRECT_DENSE = 3,
     YT_SEG_DOUBLE = 16,       /* unversioned double segment (single layout) */
+    YT_SEG_BOOLEAN = 17,      /* unversioned boolean segment (single layout) */
 };
 
 typedef struct YtSegment {
@@ -273,6 +274,13 @@ int yt_encode_int64_column(
 
 int yt_encode_double_column(
     const double* values, const uint8_t* nulls, int64_t n,
+    int32_t max_segment_values,
+    YtEncodedColumn* out, char* errbuf, size_t errlen);
+
+/* boolean columns — boolean_column_writer.cpp DumpBooleanValues:
+ * [ui64 count][value bitmap, 8-aligned][null bitmap, 8-aligned] */
+int yt_encode_bool_column(
+    const uint8_t* values, const uint8_t* nulls, int64_t n,
     int32_t max_segment_values,
     YtEncodedColumn* out, char* errbuf, size_t errlen);
 

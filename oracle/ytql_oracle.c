@@ -246,6 +246,25 @@ static int decode_double_segment(const YtSegment* seg, int64_t* vals, uint8_t* n
     return 0;
 }
 
+/* boolean_column_reader.cpp:35-50: [u64 count][value bitmap][null bitmap] */
+static int decode_bool_segment(const YtSegment* seg, int64_t* vals, uint8_t* nulls)
+{
+    const char* ptr = (const char*)seg->data;
+    const char* end = ptr + seg->data_size;
+    uint64_t count = *(const uint64_t*)ptr;
+    ptr += 8;
+    const uint8_t* vb = (const uint8_t*)ptr;
+    ptr += align_up8(((int64_t)count + 7) / 8);
+    const uint8_t* nb = (const uint8_t*)ptr;
+    ptr += align_up8(((int64_t)count + 7) / 8);
+    if ((int64_t)count != seg->row_count || ptr != end) return -1;
+    for (uint64_t i = 0; i < count; i++) {
+        nulls[i] = (uint8_t)bitmap_get(nb, i);
+        vals[i] = bitmap_get(vb, i);
+    }
+    return 0;
+}
+
 /* Decode a whole column into arrays. vals carries i64 / u64 / double bits. */
 ORACLE_EXPORT
 int yto_decode_column(const YtColumn* col, int64_t row_count,
@@ -257,6 +276,8 @@ int yto_decode_column(const YtColumn* col, int64_t row_count,
         int rc;
         if (col->value_type == YT_VT_DOUBLE) {
             rc = decode_double_segment(seg, vals + row, nulls + row);
+        } else if (col->value_type == YT_VT_BOOLEAN) {
+            rc = decode_bool_segment(seg, vals + row, nulls + row);
         } else {
             rc = decode_int_segment(seg, col->value_type == YT_VT_INT64,
                                     vals + row, nulls + row);

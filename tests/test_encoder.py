@@ -141,3 +141,26 @@ def test_double_roundtrip():
     assert rc == 0
     np.testing.assert_array_equal(gn, nulls)
     np.testing.assert_array_equal(got.view(np.float64)[nulls == 0], vals[nulls == 0])
+
+
+def test_boolean_roundtrip():
+    """boolean segments — boolean_column_writer.cpp DumpBooleanValues:
+    [u64 count][value bitmap][null bitmap], both 8-aligned."""
+    rng = np.random.default_rng(7)
+    n = 3000
+    vals = (rng.random(n) < 0.5).astype(np.uint8)
+    nulls = (rng.random(n) < 0.1).astype(np.uint8)
+    enc = y.encode_bool(vals, nulls, max_segment_values=1000)
+    assert enc._cenc.segment_count == 3
+    colc = YtColumn(value_type=enc.value_type,
+                    segment_count=enc._cenc.segment_count,
+                    segments=enc._cenc.segments)
+    got = np.zeros(n, dtype=np.int64)
+    gn = np.zeros(n, dtype=np.uint8)
+    rc = _abi.oracle_lib().yto_decode_column(
+        C.byref(colc), n,
+        got.ctypes.data_as(C.POINTER(C.c_int64)),
+        gn.ctypes.data_as(C.POINTER(C.c_uint8)))
+    assert rc == 0
+    np.testing.assert_array_equal(gn, nulls)
+    np.testing.assert_array_equal(got[nulls == 0], vals[nulls == 0])

@@ -343,3 +343,17 @@ def test_input_output_row_limits(cuda):
     assert rc == 0, err.value
     assert rs.row_count == 10
     assert st.incomplete_output == 1
+
+
+def test_bool_and_uint64_columns(cuda):
+    """boolean group keys (boolean column format) and uint64 sums on the
+    generic path."""
+    rng = np.random.default_rng(16)
+    n = 80_000
+    bk = (rng.random(n) < 0.4).astype(np.uint8)
+    bn = (rng.random(n) < 0.1).astype(np.uint8)
+    uv = rng.integers(0, 2**63, n, dtype=np.int64)  # bits used as u64
+    chunk = y.Chunk([y.encode_bool(bk, bn), y.encode_int64(uv, unsigned=True)], n)
+    plan = y.Plan(keys=[y.col(0)], aggs=[y.agg_sum(y.col(1)), y.agg_sum1()])
+    got, want, _ = run_both(plan, chunk, cuda, hint=64)
+    assert y.sort_rows(got) == y.sort_rows(want)

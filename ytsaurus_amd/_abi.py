@@ -179,6 +179,9 @@ def gpu_lib():
         _sig(lib, "yt_encode_double_column", C.c_int,
              [C.POINTER(C.c_double), C.POINTER(C.c_uint8), C.c_int64, C.c_int32,
               C.POINTER(YtEncodedColumn), C.c_char_p, C.c_size_t])
+        _sig(lib, "yt_encode_bool_column", C.c_int,
+             [C.POINTER(C.c_uint8), C.POINTER(C.c_uint8), C.c_int64, C.c_int32,
+              C.POINTER(YtEncodedColumn), C.c_char_p, C.c_size_t])
         _sig(lib, "yt_encode_string_column", C.c_int,
              [C.c_char_p, C.POINTER(C.c_uint64), C.POINTER(C.c_uint32),
               C.POINTER(C.c_uint8), C.c_int64, C.c_int32,

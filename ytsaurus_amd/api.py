@@ -19,7 +19,8 @@ from ._abi import (
 
 __all__ = [
     "col", "lit", "null", "litf", "Plan", "agg_sum", "agg_sum1",
-    "encode_int64", "encode_double", "encode_string", "oracle_decode_strings",
+    "encode_int64", "encode_double", "encode_bool", "encode_string",
+    "oracle_decode_strings",
     "Chunk", "oracle_execute",
     "oracle_partial", "oracle_merge",
     "gpu_available", "gpu_execute", "gpu_partial", "gpu_merge",
@@ -254,6 +255,22 @@ def oracle_decode_strings(enc, n):
         else:
             out.append(raw[ends[i]:ends[i + 1]])
     return out
+
+
+def encode_bool(values, nulls=None, max_segment_values=0):
+    values = np.ascontiguousarray(values, dtype=np.uint8)
+    n = len(values)
+    nullp = None
+    if nulls is not None:
+        nulls = np.ascontiguousarray(nulls, dtype=np.uint8)
+        nullp = nulls.ctypes.data_as(C.POINTER(C.c_uint8))
+    enc = YtEncodedColumn()
+    err = C.create_string_buffer(256)
+    rc = _abi.gpu_lib().yt_encode_bool_column(
+        values.ctypes.data_as(C.POINTER(C.c_uint8)), nullp, n,
+        max_segment_values, C.byref(enc), err, 256)
+    _check(rc, err)
+    return EncodedColumn(VT_BOOLEAN, enc)
 
 
 def encode_double(values, nulls=None, max_segment_values=0):

@@ -345,6 +345,36 @@ extern "C" int yt_encode_double_column(
     return finish_column(segs, out, errbuf, errlen);
 }
 
+/* boolean_column_writer.cpp:17-28 DumpBooleanValues */
+extern "C" int yt_encode_bool_column(
+    const uint8_t* values, const uint8_t* nulls, int64_t n,
+    int32_t max_segment_values,
+    YtEncodedColumn* out, char* errbuf, size_t errlen)
+{
+    if (max_segment_values <= 0) max_segment_values = 128 * 1024;
+    std::vector<SegmentOut> segs;
+    int64_t at = 0;
+    while (at < n) {
+        int64_t cnt = n - at < max_segment_values ? n - at : max_segment_values;
+        segs.emplace_back();
+        SegmentOut& s = segs.back();
+        s.type = YT_SEG_BOOLEAN;
+        s.row_count = (int32_t)cnt;
+        s.min_value = 0;
+        uint64_t c64 = (uint64_t)cnt;
+        s.blob.raw(&c64, 8);
+        std::vector<uint8_t> vb(cnt), nb(cnt, 0);
+        for (int64_t i = 0; i < cnt; i++) {
+            vb[i] = values[at + i] ? 1 : 0;
+            if (nulls) nb[i] = nulls[at + i];
+        }
+        s.blob.bitmap(vb);     /* value bitmap (bit set = true) */
+        s.blob.bitmap(nb);
+        at += cnt;
+    }
+    return finish_column(segs, out, errbuf, errlen);
+}
+
 extern "C" void yt_encoded_column_free(YtEncodedColumn* col)
 {
     if (col && col->blob) {

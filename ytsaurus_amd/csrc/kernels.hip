@@ -175,6 +175,11 @@ __global__ void k_parse_segments(const DevSeg* segs, int nsegs, SegEx* out,
         e.off_doubles_bytes = 8;
         e.off_bitmap_bytes = 8 + 8 * (int64_t)s.row_count;
         e.w_values = 64;
+    } else if (s.type == YT_SEG_BOOLEAN) {
+        /* [u64 count][value bitmap][null bitmap], both 8-aligned */
+        e.off_doubles_bytes = 8;     /* value bitmap */
+        e.off_bitmap_bytes = 8 + (((int64_t)s.row_count + 7) / 8 + 7) / 8 * 8;
+        e.w_values = 1;
     } else {
         uint64_t h0 = b[0];
         uint64_t n0 = h0 & ((1ULL << 56) - 1);
@@ -287,6 +292,12 @@ __device__ DVal seg_value_at(const DevSeg& s, const SegEx& e, int64_t row_in_seg
         const uint8_t* bm = (const uint8_t*)b + e.off_bitmap_bytes;
         if (bm_get(bm, row_in_seg)) { v.null_ = 1; v.bits = 0; return v; }
         v.bits = ((const uint64_t*)((const uint8_t*)b + e.off_doubles_bytes))[row_in_seg];
+        return v;
+    }
+    case YT_SEG_BOOLEAN: {
+        const uint8_t* bm = (const uint8_t*)b + e.off_bitmap_bytes;
+        if (bm_get(bm, row_in_seg)) { v.null_ = 1; v.bits = 0; return v; }
+        v.bits = (uint64_t)bm_get((const uint8_t*)b + e.off_doubles_bytes, row_in_seg);
         return v;
     }
     case YT_SEG_DIRECT_DENSE: {
