@@ -450,7 +450,7 @@ def gpu_execute(plan, device_chunk, max_groups_hint=0, group_row_limit=0,
         rs = rowset
     else:
         cap = out_capacity or max(int(max_groups_hint) * 2 + 1024, 1 << 16)
-        rs = _mk_rowset(cap)
+        rs = _mk_rowset(cap, pool_bytes=32 << 20)
     st = YtStatistics()
     err = C.create_string_buffer(512)
     rc = _abi.gpu_lib().yt_gpu_query_execute(

@@ -14,7 +14,7 @@ namespace ytql {
  * 394-489). */
 struct DevSeg {
     int32_t type;        /* YT_SEG_* */
-    int32_t is_signed;   /* 1 = int64 (zigzag), 0 = uint64 */
+    int32_t is_signed;   /* 1 = int64 (zigzag), 0 = uint64; 2 = string column */
     int64_t start_row;   /* first chunk row covered */
     int32_t row_count;
     int32_t col;         /* owning column index */
@@ -183,6 +183,47 @@ struct PartParams {
     int32_t stage_bm_mask;        /* bit u: stage used-col u's null bitmap */
     int32_t has_filter_nulls;
     int32_t stage_val;            /* stage the value column's packed words too */
+};
+
+/* string-keyed GROUP BY (BASELINE config 5 family): dictionary-encoded
+ * string key segments aggregate by per-segment dictionary id into dense
+ * per-segment accumulators, then dictionary entries merge across segments
+ * in a global table keyed by (hash, byte-exact string compare). */
+struct StrGroupParams {
+    int32_t key_seg_off;          /* key column segments */
+    int32_t key_seg_cnt;
+    int32_t val_seg_off;          /* sum-arg column segments; -1 = none */
+    int32_t val_seg_cnt;
+    int32_t val_is_double;
+    int32_t sum_slot;             /* agg slot of the sum; -1 = none */
+    int32_t agg_count;
+    int32_t tile_rows;
+    int32_t tiles_per_seg;
+    int32_t ntiles;
+    int32_t has_val_nulls;
+    int32_t pad_;
+    int64_t row_count;
+};
+
+/* merge-table slot: rep = (seg+1)<<32 | id (1-based), 0 = empty.
+ * The representative's hash is NOT stored: readers recompute it from the
+ * precomputed per-entry hash array via (seg, id), which avoids any
+ * publication ordering between the claim and a hash field. */
+struct StrSlot {
+    unsigned long long rep;
+    uint64_t sum_bits;
+    uint64_t cnt;
+    uint64_t nonnull;
+};
+
+/* compacted string group (device→host) */
+struct OutStrGroup {
+    uint64_t pool_off;
+    uint32_t len;
+    uint32_t pad_;
+    uint64_t sum_bits;
+    uint64_t cnt;
+    uint64_t nonnull;
 };
 
 struct KernelTimes {

@@ -53,6 +53,19 @@ hipError_t ytql_launch_part_scatter(const OutGroup*, int64_t, int, int,
                                     unsigned long long*, YtStateRow*, hipStream_t);
 hipError_t ytql_launch_merge_states(const YtStateRow*, int64_t, int, int,
                                     TableHdr*, unsigned long long*, hipStream_t);
+hipError_t ytql_launch_strgrp_accum(const StrGroupParams*, const DevSeg*, const SegEx*,
+                                    const int64_t*, unsigned long long*, TableHdr*,
+                                    hipStream_t);
+hipError_t ytql_launch_strgrp_hash(const DevSeg*, const SegEx*, int, int,
+                                   const int64_t*, uint64_t*, int64_t, hipStream_t);
+hipError_t ytql_launch_strgrp_merge(const DevSeg*, const SegEx*, int, int,
+                                    const int64_t*, const unsigned long long*,
+                                    const uint64_t*, StrSlot*, uint64_t, int,
+                                    TableHdr*, int64_t, hipStream_t);
+hipError_t ytql_launch_strgrp_compact(const DevSeg*, const SegEx*, int,
+                                      const StrSlot*, uint64_t, OutStrGroup*,
+                                      unsigned long long*, char*, unsigned long long*,
+                                      uint64_t, TableHdr*, hipStream_t);
 }
 
 /* ------------------------------------------------------------------ */
@@ -631,7 +644,8 @@ static int setup_chunk(const YtChunk* chunk, DeviceRun* R, unsigned* maxw_out,
             }
             DevSeg d;
             d.type = seg.type;
-            d.is_signed = (col.value_type == YT_VT_INT64);
+            d.is_signed = (col.value_type == YT_VT_STRING) ? 2
+                        : (col.value_type == YT_VT_INT64);
             d.start_row = row;
             d.row_count = rows_here;
             d.col = c;
@@ -1295,6 +1309,276 @@ fail:
     return rc;
 }
 
+
+/* string-keyed GROUP BY (config-5 family): dictionary-encoded string key,
+ * aggregates {sum(col), sum(1)}. Dense per-segment id accumulators, then an
+ * exact cross-segment merge (hash gate + byte compare), then device-side
+ * key materialization into the caller's string pool. */
+static int run_string_group(const YtPlan* plan, const YtChunk* chunk,
+                            const YtExecOptions* options, int key_col,
+                            YtRowset* output, YtStatistics* stats, double tw0,
+                            char* errbuf, size_t errlen)
+{
+    int rc = YT_OK;
+    int sum_slot = -1, sum_col = -1, val_is_double = 0;
+    for (int a = 0; a < plan->agg_count; a++) {
+        int f = plan->aggs[a]->func;
+        if (f == YT_AGG_SUM) {
+            int c;
+            if (sum_slot >= 0 || !expr_is_col(plan->aggs[a]->arg, &c)) {
+                set_err(errbuf, errlen, "string keys: one direct sum this round");
+                return YT_ERR_UNSUPPORTED;
+            }
+            sum_slot = a;
+            sum_col = c;
+            val_is_double = chunk->columns[c].value_type == YT_VT_DOUBLE;
+            if (!val_is_double && chunk->columns[c].value_type != YT_VT_INT64) {
+                set_err(errbuf, errlen, "string keys: int64/double sums this round");
+                return YT_ERR_UNSUPPORTED;
+            }
+        } else if (f != YT_AGG_SUM1) {
+            set_err(errbuf, errlen, "string keys: sum/sum(1) this round");
+            return YT_ERR_UNSUPPORTED;
+        }
+    }
+    if (plan->project_count) {
+        set_err(errbuf, errlen, "string keys: no post-projection this round");
+        return YT_ERR_UNSUPPORTED;
+    }
+
+    DeviceRun R;
+    R.stream = (hipStream_t)(uintptr_t)options->stream;
+    unsigned mw = 0;
+    int in_clamped = 0;
+    rc = setup_chunk(chunk, &R, &mw, options->input_row_limit, &in_clamped,
+                     errbuf, errlen);
+    if (rc) return rc;
+    output->row_count = 0;
+    output->column_count = 1 + plan->agg_count;
+    if (chunk->row_count == 0 || R.nsegs == 0) return YT_OK;
+    rc = setup_table(&R, plan->agg_count, options->max_groups_hint,
+                     options->group_row_limit, errbuf, errlen);
+    if (rc) return rc;
+
+    /* read back the parsed key-column segments: dictionary sizes + layout */
+    int knseg = R.h_cnt[key_col];
+    int koff = R.h_off[key_col];
+    std::vector<SegEx> ksegex(knseg);
+    TableHdr th;
+    int64_t total_dict = 0;
+    {
+        HIP_CHECK(hipMemcpy(ksegex.data(), R.d_segex + koff,
+                            sizeof(SegEx) * knseg, hipMemcpyDeviceToHost));
+        for (int i = 0; i < knseg; i++) {
+            if (!(ksegex[i].flags & 8)) {
+                set_err(errbuf, errlen,
+                        "string keys: only dictionary-encoded segments this round");
+                return YT_ERR_UNSUPPORTED;
+            }
+        }
+        int32_t rows0 = R.h_segs[koff].row_count;
+        for (int i = 0; i + 1 < knseg; i++) {
+            if (R.h_segs[koff + i].row_count != rows0) {
+                set_err(errbuf, errlen, "string keys: non-uniform segments");
+                return YT_ERR_UNSUPPORTED;
+            }
+        }
+        if (sum_col >= 0 && R.h_cnt[sum_col] != knseg) {
+            set_err(errbuf, errlen, "string keys: column segmentation mismatch");
+            return YT_ERR_UNSUPPORTED;
+        }
+
+        std::vector<int64_t> acc_base(knseg + 1);
+        for (int i = 0; i < knseg; i++) {
+            acc_base[i] = total_dict;
+            total_dict += ksegex[i].dict_size;
+        }
+        acc_base[knseg] = total_dict;
+
+        StrGroupParams sp;
+        memset(&sp, 0, sizeof(sp));
+        sp.key_seg_off = koff;
+        sp.key_seg_cnt = knseg;
+        sp.val_seg_off = sum_col >= 0 ? R.h_off[sum_col] : -1;
+        sp.val_seg_cnt = sum_col >= 0 ? R.h_cnt[sum_col] : 0;
+        sp.val_is_double = val_is_double;
+        sp.sum_slot = sum_slot;
+        sp.agg_count = plan->agg_count;
+        sp.has_val_nulls = sum_col >= 0 && (R.col_null_flags[sum_col] != 0);
+        sp.tile_rows = 8192;
+        if (sp.tile_rows > rows0) sp.tile_rows = rows0 > 256 ? rows0 : 256;
+        sp.tiles_per_seg = (rows0 + sp.tile_rows - 1) / sp.tile_rows;
+        int32_t last_rows = R.h_segs[koff + knseg - 1].row_count;
+        sp.ntiles = (knseg - 1) * sp.tiles_per_seg
+                  + (last_rows + sp.tile_rows - 1) / sp.tile_rows;
+        sp.row_count = chunk->row_count;
+
+        int64_t* d_accbase = nullptr;
+        unsigned long long* d_acc = nullptr;
+        uint64_t* d_hashes = nullptr;
+        StrSlot* d_slots = nullptr;
+        HIP_CHECK(pool_alloc(&d_accbase, sizeof(int64_t) * (knseg + 1)));
+        HIP_CHECK(hipMemcpyAsync(d_accbase, acc_base.data(),
+                                 sizeof(int64_t) * (knseg + 1),
+                                 hipMemcpyHostToDevice, R.stream));
+        HIP_CHECK(pool_alloc(&d_acc, sizeof(uint64_t) * 2 * (total_dict ? total_dict : 1)));
+        HIP_CHECK(hipMemsetAsync(d_acc, 0, sizeof(uint64_t) * 2 * (total_dict ? total_dict : 1), R.stream));
+        HIP_CHECK(pool_alloc(&d_hashes, sizeof(uint64_t) * (total_dict ? total_dict : 1)));
+        uint64_t nslots = next_pow2((uint64_t)(total_dict ? total_dict : 1) * 2);
+        if (nslots < 2048) nslots = 2048;
+        HIP_CHECK(pool_alloc(&d_slots, sizeof(StrSlot) * nslots));
+        HIP_CHECK(hipMemsetAsync(d_slots, 0, sizeof(StrSlot) * nslots, R.stream));
+
+        hipEvent_t e0, e1;
+        HIP_CHECK(hipEventCreate(&e0));
+        HIP_CHECK(hipEventCreate(&e1));
+        HIP_CHECK(hipEventRecord(e0, R.stream));
+        HIP_CHECK(ytql_launch_strgrp_accum(&sp, R.d_segs, R.d_segex, d_accbase,
+                                           d_acc, R.d_th, R.stream));
+        HIP_CHECK(ytql_launch_strgrp_hash(R.d_segs, R.d_segex, koff, knseg,
+                                          d_accbase, d_hashes, total_dict, R.stream));
+        HIP_CHECK(ytql_launch_strgrp_merge(R.d_segs, R.d_segex, koff, knseg,
+                                           d_accbase, d_acc, d_hashes,
+                                           d_slots, nslots, val_is_double,
+                                           R.d_th, total_dict, R.stream));
+        HIP_CHECK(hipEventRecord(e1, R.stream));
+
+        /* compact + materialize keys into a device pool */
+        HIP_CHECK(hipMemcpy(&th, R.d_th, sizeof(th), hipMemcpyDeviceToHost));
+        int64_t ngroups = (int64_t)th.ngroups;
+        uint64_t pool_cap = 0;
+        for (int i = 0; i < knseg; i++) {
+            /* dictionary blob bytes upper-bound the key bytes */
+            pool_cap += (uint64_t)R.h_segs[koff + i].blob_bytes;
+        }
+        if (pool_cap < 1024) pool_cap = 1024;
+        OutStrGroup* d_out = nullptr;
+        char* d_pool = nullptr;
+        unsigned long long* d_ctr = nullptr;
+        HIP_CHECK(pool_alloc(&d_out, sizeof(OutStrGroup) * (ngroups ? ngroups : 1)));
+        HIP_CHECK(pool_alloc(&d_pool, pool_cap));
+        HIP_CHECK(pool_alloc(&d_ctr, 2 * sizeof(unsigned long long)));
+        HIP_CHECK(hipMemsetAsync(d_ctr, 0, 2 * sizeof(unsigned long long), R.stream));
+        if (ngroups > 0) {
+            HIP_CHECK(ytql_launch_strgrp_compact(R.d_segs, R.d_segex, koff,
+                                                 d_slots, nslots, d_out, d_ctr,
+                                                 d_pool, d_ctr + 1, pool_cap,
+                                                 R.d_th, R.stream));
+        }
+        std::vector<OutStrGroup> groups(ngroups ? ngroups : 0);
+        unsigned long long hctr[2] = {0, 0};
+        if (ngroups > 0) {
+            HIP_CHECK(hipMemcpyAsync(groups.data(), d_out,
+                                     sizeof(OutStrGroup) * ngroups,
+                                     hipMemcpyDeviceToHost, R.stream));
+        }
+        HIP_CHECK(hipMemcpyAsync(hctr, d_ctr, 2 * sizeof(unsigned long long),
+                                 hipMemcpyDeviceToHost, R.stream));
+        HIP_CHECK(hipStreamSynchronize(R.stream));
+        std::vector<char> pool(hctr[1] ? hctr[1] : 1);
+        if (hctr[1]) {
+            HIP_CHECK(hipMemcpy(pool.data(), d_pool, hctr[1], hipMemcpyDeviceToHost));
+        }
+        float ms = 0;
+        HIP_CHECK(hipEventElapsedTime(&ms, e0, e1));
+        hipEventDestroy(e0);
+        hipEventDestroy(e1);
+        HIP_CHECK(hipMemcpy(&th, R.d_th, sizeof(th), hipMemcpyDeviceToHost));
+        if (th.overflow == 1) {
+            g_pool.put(d_accbase); g_pool.put(d_acc); g_pool.put(d_hashes);
+            g_pool.put(d_slots); g_pool.put(d_out); g_pool.put(d_pool); g_pool.put(d_ctr);
+            set_err(errbuf, errlen, "string merge table/pool overflow");
+            return YT_ERR_CAPACITY;
+        }
+
+        /* emit [key(string), aggs...] rows */
+        int ncols = 1 + plan->agg_count;
+        int out_limited = 0;
+        for (int64_t gI = 0; gI < (int64_t)groups.size() + 2; gI++) {
+            const char* kstr = nullptr;
+            uint32_t klen = 0;
+            int knull = 0;
+            uint64_t cnt, sum_bits, nonnull;
+            if (gI < (int64_t)groups.size()) {
+                const OutStrGroup& g = groups[gI];
+                kstr = pool.data() + g.pool_off;
+                klen = g.len;
+                cnt = g.cnt;
+                sum_bits = g.sum_bits;
+                nonnull = g.nonnull;
+            } else {
+                int side = (int)(gI - (int64_t)groups.size());
+                if (side == 0) continue;                /* no sentinel for strings */
+                if (!th.side_used[1]) continue;
+                knull = 1;
+                cnt = th.side_cnt[1];
+                sum_bits = sum_slot >= 0 ? th.side_agg[1][2 * sum_slot] : 0;
+                nonnull = sum_slot >= 0 ? th.side_agg[1][2 * sum_slot + 1] : 0;
+            }
+            if (options->output_row_limit > 0 &&
+                output->row_count >= options->output_row_limit) {
+                out_limited = 1;
+                break;
+            }
+            if (output->row_count >= output->capacity_rows) {
+                rc = YT_ERR_CAPACITY;
+                break;
+            }
+            YtValue* dst = output->values + output->row_count * ncols;
+            if (knull) {
+                dst[0].type = YT_VT_NULL;
+                dst[0].length = 0;
+                dst[0].data.bits = 0;
+            } else {
+                if (output->string_pool_used + klen > output->string_pool_capacity) {
+                    rc = YT_ERR_CAPACITY;
+                    set_err(errbuf, errlen, "string pool too small");
+                    break;
+                }
+                memcpy(output->string_pool + output->string_pool_used, kstr, klen);
+                dst[0].type = YT_VT_STRING;
+                dst[0].length = klen;
+                dst[0].data.str = output->string_pool + output->string_pool_used;
+                output->string_pool_used += klen;
+            }
+            dst[0].id = 0;
+            dst[0].flags = 0;
+            for (int a = 0; a < plan->agg_count; a++) {
+                YtValue& v = dst[1 + a];
+                v.id = (uint16_t)(1 + a);
+                v.flags = 0;
+                v.length = 0;
+                if (plan->aggs[a]->func == YT_AGG_SUM1) {
+                    v.type = YT_VT_INT64;
+                    v.data.bits = cnt;
+                } else if (nonnull == 0) {
+                    v.type = YT_VT_NULL;
+                    v.data.bits = 0;
+                } else {
+                    v.type = val_is_double ? YT_VT_DOUBLE : YT_VT_INT64;
+                    v.data.bits = sum_bits;
+                }
+            }
+            output->row_count++;
+        }
+        if (stats) {
+            stats->rows_read = chunk->row_count;
+            stats->rows_written = output->row_count;
+            stats->grouped_row_count = ngroups + (th.side_used[1] ? 1 : 0);
+            stats->incomplete_input = in_clamped;
+            stats->incomplete_output = out_limited || (th.overflow == 2);
+            stats->kernel_scan_ms += ms;
+            stats->kernel_scan_launches += 1;
+            stats->execute_time_ms = now_ms() - tw0;
+        }
+        g_pool.put(d_accbase); g_pool.put(d_acc); g_pool.put(d_hashes);
+        g_pool.put(d_slots); g_pool.put(d_out); g_pool.put(d_pool); g_pool.put(d_ctr);
+    }
+    return rc;
+fail:
+    return rc;
+}
+
 /* ------------------------------------------------------------------ */
 /* public entries                                                      */
 
@@ -1320,6 +1604,16 @@ extern "C" int yt_gpu_query_execute(
         set_err(errbuf, errlen,
                 "GPU scan-project materializes per-row output; capped at 16M rows this round");
         return YT_ERR_UNSUPPORTED;
+    }
+
+    {
+        int kc;
+        if (plan->key_count == 1 && plan->agg_count > 0 &&
+            expr_is_col(plan->keys[0], &kc) && kc < chunk->column_count &&
+            chunk->columns[kc].value_type == YT_VT_STRING) {
+            return run_string_group(plan, chunk, options, kc, output, stats,
+                                    tw0, errbuf, errlen);
+        }
     }
 
     DevPlan dp;

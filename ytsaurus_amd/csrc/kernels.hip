@@ -175,6 +175,55 @@ __global__ void k_parse_segments(const DevSeg* segs, int nsegs, SegEx* out,
         e.off_doubles_bytes = 8;
         e.off_bitmap_bytes = 8 + 8 * (int64_t)s.row_count;
         e.w_values = 64;
+    } else if (s.is_signed == 2) {
+        /* STRING column — string_column_writer.cpp dump layouts.
+         * off_values_words = offsets vector data; off_doubles_bytes = blob
+         * byte offset of the string data; dict_size = dictionary entries. */
+        uint64_t h0 = b[0];
+        uint64_t n0 = h0 & ((1ULL << 56) - 1);
+        uint32_t w0 = (uint32_t)(h0 >> 56);
+        int64_t words0 = 1 + (int64_t)((w0 * n0 + 63) >> 6);
+        switch (s.type) {
+        case YT_SEG_DICTIONARY_DENSE: {
+            /* [ids][offsets][data] */
+            e.w_ids = w0;
+            e.run_count = (uint32_t)n0;
+            e.off_ids_words = 1;
+            uint64_t h1 = b[words0];
+            e.w_values = (uint32_t)(h1 >> 56);
+            e.dict_size = (uint32_t)(h1 & ((1ULL << 56) - 1));
+            e.off_values_words = words0 + 1;
+            int64_t words1 = 1 + (int64_t)(((uint64_t)e.w_values * e.dict_size + 63) >> 6);
+            e.off_doubles_bytes = (words0 + words1) * 8;
+            e.flags |= 8;   /* dictionary string segment */
+            break;
+        }
+        case YT_SEG_DICTIONARY_RLE: {
+            /* [row starts][ids][offsets][data] */
+            e.w_starts = w0;
+            e.run_count = (uint32_t)n0;
+            e.off_starts_words = 1;
+            uint64_t h1 = b[words0];
+            e.w_ids = (uint32_t)(h1 >> 56);
+            uint32_t nids = (uint32_t)(h1 & ((1ULL << 56) - 1));
+            e.run_count = nids;
+            e.off_ids_words = words0 + 1;
+            int64_t words1 = 1 + (int64_t)(((uint64_t)e.w_ids * nids + 63) >> 6);
+            uint64_t h2 = b[words0 + words1];
+            e.w_values = (uint32_t)(h2 >> 56);
+            e.dict_size = (uint32_t)(h2 & ((1ULL << 56) - 1));
+            e.off_values_words = words0 + words1 + 1;
+            int64_t words2 = 1 + (int64_t)(((uint64_t)e.w_values * e.dict_size + 63) >> 6);
+            e.off_doubles_bytes = (words0 + words1 + words2) * 8;
+            e.flags |= 8;
+            break;
+        }
+        default:
+            e.flags |= 16;  /* direct string segment: no GPU key path yet */
+            break;
+        }
+        out[i] = e;
+        return;
     } else if (s.type == YT_SEG_BOOLEAN) {
         /* [u64 count][value bitmap][null bitmap], both 8-aligned */
         e.off_doubles_bytes = 8;     /* value bitmap */
@@ -1198,6 +1247,265 @@ k_bucket_agg(const ulonglong2* recs, const unsigned long long* cursors,
     }
 }
 
+
+/* ------------------------------------------------------------------ */
+/* string-keyed GROUP BY (config-5 family; see common.h StrGroupParams) */
+
+/* string dictionary entry bounds — string_column_reader.cpp:39-42:
+ * offset(i) = expected_length*(i+1) + ZigZagDecode32(packed[i]) */
+__device__ __forceinline__ int32_t zz_dec32(uint32_t n)
+{
+    return (int32_t)((n >> 1) ^ (~(n & 1) + 1));
+}
+
+__device__ __forceinline__ uint64_t str_off(const uint64_t* offs, uint32_t w,
+                                            uint64_t expected, int64_t i)
+{
+    if (i < 0) return 0;
+    return expected * (uint64_t)(i + 1)
+         + (uint64_t)(int64_t)zz_dec32((uint32_t)bp_get(offs, w, i));
+}
+
+/* dictionary entry i of a parsed string segment → (ptr, len) */
+__device__ __forceinline__ const char* dict_entry(const DevSeg& s, const SegEx& e,
+                                                  int64_t i, uint32_t* len)
+{
+    const uint64_t* offs = s.blob + e.off_values_words;
+    uint64_t b = str_off(offs, e.w_values, s.min_value, i - 1);
+    uint64_t en = str_off(offs, e.w_values, s.min_value, i);
+    *len = (uint32_t)(en - b);
+    return (const char*)s.blob + e.off_doubles_bytes + b;
+}
+
+/* S1: accumulate rows into per-segment per-dictionary-id accumulators
+ * acc[base + id-1] = { cnt|nonnull<<32 (u32 pair, bounded by the 128Ki
+ * segment row cap), sum bits }. Null ids go to the null-key side group. */
+__global__ void __launch_bounds__(256)
+k_strgrp_accum(StrGroupParams sp, const DevSeg* segs, const SegEx* segex,
+               const int64_t* acc_base, unsigned long long* acc,
+               TableHdr* th)
+{
+    for (int tile = blockIdx.x; tile < sp.ntiles; tile += gridDim.x) {
+        const int seg_idx = tile / sp.tiles_per_seg;
+        const int tile_in_seg = tile % sp.tiles_per_seg;
+        const DevSeg& sk = segs[sp.key_seg_off + seg_idx];
+        const SegEx& ek = segex[sp.key_seg_off + seg_idx];
+        const int64_t t0 = (int64_t)tile_in_seg * sp.tile_rows;
+        int64_t t1 = t0 + sp.tile_rows;
+        if (t1 > sk.row_count) t1 = sk.row_count;
+        unsigned long long* seg_acc = acc + 2 * acc_base[seg_idx];
+
+        const DevSeg* sv = sp.val_seg_off >= 0 ? &segs[sp.val_seg_off + seg_idx] : nullptr;
+        const SegEx* ev = sp.val_seg_off >= 0 ? &segex[sp.val_seg_off + seg_idx] : nullptr;
+        const uint8_t* vbm = nullptr;
+        uint64_t vmask = 0, vmin = 0;
+        uint32_t vwd = 0;
+        const uint64_t* vwords = nullptr;
+        if (sv) {
+            vwords = sv->blob + ev->off_values_words;
+            vwd = ev->w_values;
+            vmask = (vwd >= 64) ? ~0ULL : ((1ULL << vwd) - 1);
+            vmin = sv->min_value;
+            if (sp.has_val_nulls)
+                vbm = (const uint8_t*)sv->blob + ev->off_bitmap_bytes;
+        }
+
+        const bool is_rle = (sk.type == YT_SEG_DICTIONARY_RLE);
+        const uint64_t* ids = sk.blob + ek.off_ids_words;
+        const uint64_t* starts = sk.blob + ek.off_starts_words;
+        const int R = (sp.tile_rows + 255) / 256;
+        for (int i = 0; i < R; i++) {
+            int64_t j = t0 + (int64_t)i * 256 + threadIdx.x;
+            if (j >= t1) continue;
+            uint64_t id;
+            if (is_rle) {
+                uint32_t lo = 0, hi = ek.run_count;
+                while (lo + 1 < hi) {
+                    uint32_t mid = (lo + hi) / 2;
+                    if (bp_get(starts, ek.w_starts, mid) <= (uint64_t)j) lo = mid;
+                    else hi = mid;
+                }
+                id = bp_get(ids, ek.w_ids, lo);
+            } else {
+                id = bp_get(ids, ek.w_ids, j);
+            }
+            int val_null = 1;
+            uint64_t vbits = 0;
+            double vdbl = 0;
+            if (sv) {
+                val_null = vbm && bm_get(vbm, j);
+                if (!val_null) {
+                    if (sv->type == YT_SEG_DOUBLE) {
+                        vbits = ((const uint64_t*)((const uint8_t*)sv->blob
+                                 + ev->off_doubles_bytes))[j];
+                        vdbl = __longlong_as_double(vbits);
+                    } else {
+                        uint64_t pv = bp_gl(vwords, vmask, vwd, j);
+                        vbits = (uint64_t)zz_dec(vmin + pv);
+                    }
+                }
+            }
+            if (id == 0) {
+                /* null string key → side group (registry.cpp group-key null) */
+                th->side_used[1] = 1;
+                atomicAdd((unsigned long long*)&th->side_cnt[1], 1ULL);
+                if (sp.sum_slot >= 0 && !val_null) {
+                    if (sp.val_is_double)
+                        atomicAdd((double*)&th->side_agg[1][2 * sp.sum_slot], vdbl);
+                    else
+                        atomicAdd((unsigned long long*)&th->side_agg[1][2 * sp.sum_slot], vbits);
+                    atomicAdd((unsigned long long*)&th->side_agg[1][2 * sp.sum_slot + 1], 1ULL);
+                }
+                continue;
+            }
+            unsigned long long* a = seg_acc + 2 * (id - 1);
+            atomicAdd(a, 1ULL | ((unsigned long long)(!val_null) << 32));
+            if (sp.sum_slot >= 0 && !val_null) {
+                if (sp.val_is_double) atomicAdd((double*)(a + 1), vdbl);
+                else atomicAdd(a + 1, vbits);
+            }
+        }
+    }
+}
+
+/* S2: FNV-1a hash of every dictionary entry */
+__global__ void k_strgrp_hash(const DevSeg* segs, const SegEx* segex,
+                              int key_seg_off, int nsegs,
+                              const int64_t* acc_base, uint64_t* hashes)
+{
+    int64_t total = acc_base[nsegs];
+    for (int64_t g = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         g < total; g += (int64_t)gridDim.x * blockDim.x) {
+        int lo = 0, hi = nsegs;
+        while (lo + 1 < hi) {
+            int mid = (lo + hi) / 2;
+            if (acc_base[mid] <= g) lo = mid;
+            else hi = mid;
+        }
+        const DevSeg& sk = segs[key_seg_off + lo];
+        const SegEx& ek = segex[key_seg_off + lo];
+        uint32_t len;
+        const char* p = dict_entry(sk, ek, g - acc_base[lo] + 1, &len);
+        uint64_t h = 0xCBF29CE484222325ULL;
+        for (uint32_t k = 0; k < len; k++) {
+            h = (h ^ (uint8_t)p[k]) * 0x100000001B3ULL;
+        }
+        hashes[g] = h;
+    }
+}
+
+/* S3: merge dictionary entries across segments by exact string identity.
+ * One thread per dictionary entry with a nonzero count. Identity test:
+ * hash gate (precomputed array, no publication race) then byte compare. */
+__global__ void k_strgrp_merge(const DevSeg* segs, const SegEx* segex,
+                               int key_seg_off, int nsegs,
+                               const int64_t* acc_base,
+                               const unsigned long long* acc,
+                               const uint64_t* hashes,
+                               StrSlot* slots, uint64_t nslots,
+                               int val_is_double, TableHdr* th)
+{
+    int64_t total = acc_base[nsegs];
+    uint64_t mask = nslots - 1;
+    for (int64_t g = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         g < total; g += (int64_t)gridDim.x * blockDim.x) {
+        unsigned long long cn = acc[2 * g];
+        if (cn == 0) continue;
+        int lo = 0, hi = nsegs;
+        while (lo + 1 < hi) {
+            int mid = (lo + hi) / 2;
+            if (acc_base[mid] <= g) lo = mid;
+            else hi = mid;
+        }
+        const DevSeg& my_s = segs[key_seg_off + lo];
+        const SegEx& my_e = segex[key_seg_off + lo];
+        int64_t my_id = g - acc_base[lo] + 1;
+        uint32_t my_len;
+        const char* my_p = dict_entry(my_s, my_e, my_id, &my_len);
+        uint64_t h = hashes[g];
+        unsigned long long rep = ((unsigned long long)(lo + 1) << 32)
+                               | (unsigned long long)my_id;
+
+        uint64_t sidx = mix64(h) & mask;
+        StrSlot* slot = nullptr;
+        for (uint64_t it = 0; it <= mask; it++) {
+            StrSlot* cand = &slots[sidx];
+            unsigned long long cur = __hip_atomic_load(&cand->rep, __ATOMIC_RELAXED,
+                                                       __HIP_MEMORY_SCOPE_AGENT);
+            if (cur == 0ULL) {
+                cur = atomicCAS(&cand->rep, 0ULL, rep);
+                if (cur == 0ULL) {
+                    unsigned long long t = atomicAdd(&th->ngroups, 1ULL);
+                    if (th->group_limit > 0 && (int64_t)t >= th->group_limit)
+                        th->overflow = 2;
+                    slot = cand;
+                    break;
+                }
+            }
+            if (cur == rep) { slot = cand; break; }
+            {
+                int oseg = (int)(cur >> 32) - 1;
+                int64_t oid = (int64_t)(cur & 0xFFFFFFFFULL);
+                uint64_t oh = hashes[acc_base[oseg] + oid - 1];
+                if (oh == h) {
+                    uint32_t olen;
+                    const char* op = dict_entry(segs[key_seg_off + oseg],
+                                                segex[key_seg_off + oseg], oid, &olen);
+                    if (olen == my_len) {
+                        uint32_t k = 0;
+                        while (k < my_len && op[k] == my_p[k]) k++;
+                        if (k == my_len) { slot = cand; break; }
+                    }
+                }
+            }
+            sidx = (sidx + 1) & mask;
+        }
+        if (!slot) { th->overflow = 1; continue; }
+        atomicAdd((unsigned long long*)&slot->cnt, cn & 0xFFFFFFFFULL);
+        unsigned long long nn = cn >> 32;
+        if (nn) {
+            atomicAdd((unsigned long long*)&slot->nonnull, nn);
+            if (val_is_double)
+                atomicAdd((double*)&slot->sum_bits,
+                          __longlong_as_double(acc[2 * g + 1]));
+            else
+                atomicAdd((unsigned long long*)&slot->sum_bits, acc[2 * g + 1]);
+        }
+    }
+}
+
+/* S4: compact occupied slots; copy each group's key string into the device
+ * pool */
+__global__ void k_strgrp_compact(const DevSeg* segs, const SegEx* segex,
+                                 int key_seg_off,
+                                 const StrSlot* slots, uint64_t nslots,
+                                 OutStrGroup* out, unsigned long long* counter,
+                                 char* pool, unsigned long long* pool_cursor,
+                                 uint64_t pool_cap, TableHdr* th)
+{
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         i < nslots; i += (uint64_t)gridDim.x * blockDim.x) {
+        const StrSlot& sl = slots[i];
+        if (sl.rep == 0) continue;
+        int seg = (int)(sl.rep >> 32) - 1;
+        int64_t id = (int64_t)(sl.rep & 0xFFFFFFFFULL);
+        uint32_t len;
+        const char* p = dict_entry(segs[key_seg_off + seg], segex[key_seg_off + seg],
+                                   id, &len);
+        unsigned long long off = atomicAdd(pool_cursor, (unsigned long long)len);
+        if (off + len > pool_cap) { th->overflow = 1; continue; }
+        for (uint32_t k = 0; k < len; k++) pool[off + k] = p[k];
+        unsigned long long idx = atomicAdd(counter, 1ULL);
+        OutStrGroup& g = out[idx];
+        g.pool_off = off;
+        g.len = len;
+        g.pad_ = 0;
+        g.sum_bits = sl.sum_bits;
+        g.cnt = sl.cnt;
+        g.nonnull = sl.nonnull;
+    }
+}
+
 /* ------------------------------------------------------------------ */
 /* table compaction / partition / merge                                */
 
@@ -1431,6 +1739,62 @@ hipError_t ytql_launch_part_scatter(const OutGroup* groups, int64_t n, int npart
     int grid = (int)(want > 2048 ? 2048 : (want ? want : 1));
     hipLaunchKernelGGL(k_part_scatter, dim3(grid), dim3(block), 0, st,
                        groups, n, nparts, sum_slot, cursors, out);
+    return hipGetLastError();
+}
+
+hipError_t ytql_launch_strgrp_accum(const StrGroupParams* sp, const DevSeg* segs,
+                                    const SegEx* segex, const int64_t* acc_base,
+                                    unsigned long long* acc, TableHdr* th,
+                                    hipStream_t st)
+{
+    int grid = sp->ntiles < 2048 ? (sp->ntiles ? sp->ntiles : 1) : 2048;
+    hipLaunchKernelGGL(k_strgrp_accum, dim3(grid), dim3(256), 0, st,
+                       *sp, segs, segex, acc_base, acc, th);
+    return hipGetLastError();
+}
+
+hipError_t ytql_launch_strgrp_hash(const DevSeg* segs, const SegEx* segex,
+                                   int key_seg_off, int nsegs,
+                                   const int64_t* acc_base, uint64_t* hashes,
+                                   int64_t total, hipStream_t st)
+{
+    int64_t want = (total + 255) / 256;
+    int grid = (int)(want > 4096 ? 4096 : (want ? want : 1));
+    hipLaunchKernelGGL(k_strgrp_hash, dim3(grid), dim3(256), 0, st,
+                       segs, segex, key_seg_off, nsegs, acc_base, hashes);
+    return hipGetLastError();
+}
+
+hipError_t ytql_launch_strgrp_merge(const DevSeg* segs, const SegEx* segex,
+                                    int key_seg_off, int nsegs,
+                                    const int64_t* acc_base,
+                                    const unsigned long long* acc,
+                                    const uint64_t* hashes,
+                                    StrSlot* slots, uint64_t nslots,
+                                    int val_is_double, TableHdr* th,
+                                    int64_t total, hipStream_t st)
+{
+    int64_t want = (total + 255) / 256;
+    int grid = (int)(want > 4096 ? 4096 : (want ? want : 1));
+    hipLaunchKernelGGL(k_strgrp_merge, dim3(grid), dim3(256), 0, st,
+                       segs, segex, key_seg_off, nsegs, acc_base, acc, hashes,
+                       slots, nslots, val_is_double, th);
+    return hipGetLastError();
+}
+
+hipError_t ytql_launch_strgrp_compact(const DevSeg* segs, const SegEx* segex,
+                                      int key_seg_off,
+                                      const StrSlot* slots, uint64_t nslots,
+                                      OutStrGroup* out, unsigned long long* counter,
+                                      char* pool, unsigned long long* pool_cursor,
+                                      uint64_t pool_cap, TableHdr* th,
+                                      hipStream_t st)
+{
+    uint64_t want = (nslots + 255) / 256;
+    int grid = (int)(want > 2048 ? 2048 : (want ? want : 1));
+    hipLaunchKernelGGL(k_strgrp_compact, dim3(grid), dim3(256), 0, st,
+                       segs, segex, key_seg_off, slots, nslots, out, counter,
+                       pool, pool_cursor, pool_cap, th);
     return hipGetLastError();
 }
 
