@@ -1266,13 +1266,15 @@ __device__ __forceinline__ uint64_t str_off(const uint64_t* offs, uint32_t w,
          + (uint64_t)(int64_t)zz_dec32((uint32_t)bp_get(offs, w, i));
 }
 
-/* dictionary entry i of a parsed string segment → (ptr, len) */
+/* dictionary entry j (0-BASED index = id-1) of a parsed string segment →
+ * (ptr, len): spans [offset(j-1), offset(j)), offset(-1) = 0
+ * (string_column_reader.cpp:44-66) */
 __device__ __forceinline__ const char* dict_entry(const DevSeg& s, const SegEx& e,
-                                                  int64_t i, uint32_t* len)
+                                                  int64_t j, uint32_t* len)
 {
     const uint64_t* offs = s.blob + e.off_values_words;
-    uint64_t b = str_off(offs, e.w_values, s.min_value, i - 1);
-    uint64_t en = str_off(offs, e.w_values, s.min_value, i);
+    uint64_t b = str_off(offs, e.w_values, s.min_value, j - 1);
+    uint64_t en = str_off(offs, e.w_values, s.min_value, j);
     *len = (uint32_t)(en - b);
     return (const char*)s.blob + e.off_doubles_bytes + b;
 }
@@ -1385,7 +1387,7 @@ __global__ void k_strgrp_hash(const DevSeg* segs, const SegEx* segex,
         const DevSeg& sk = segs[key_seg_off + lo];
         const SegEx& ek = segex[key_seg_off + lo];
         uint32_t len;
-        const char* p = dict_entry(sk, ek, g - acc_base[lo] + 1, &len);
+        const char* p = dict_entry(sk, ek, g - acc_base[lo], &len);
         uint64_t h = 0xCBF29CE484222325ULL;
         for (uint32_t k = 0; k < len; k++) {
             h = (h ^ (uint8_t)p[k]) * 0x100000001B3ULL;
@@ -1421,7 +1423,7 @@ __global__ void k_strgrp_merge(const DevSeg* segs, const SegEx* segex,
         const SegEx& my_e = segex[key_seg_off + lo];
         int64_t my_id = g - acc_base[lo] + 1;
         uint32_t my_len;
-        const char* my_p = dict_entry(my_s, my_e, my_id, &my_len);
+        const char* my_p = dict_entry(my_s, my_e, my_id - 1, &my_len);
         uint64_t h = hashes[g];
         unsigned long long rep = ((unsigned long long)(lo + 1) << 32)
                                | (unsigned long long)my_id;
@@ -1450,7 +1452,7 @@ __global__ void k_strgrp_merge(const DevSeg* segs, const SegEx* segex,
                 if (oh == h) {
                     uint32_t olen;
                     const char* op = dict_entry(segs[key_seg_off + oseg],
-                                                segex[key_seg_off + oseg], oid, &olen);
+                                                segex[key_seg_off + oseg], oid - 1, &olen);
                     if (olen == my_len) {
                         uint32_t k = 0;
                         while (k < my_len && op[k] == my_p[k]) k++;
@@ -1491,7 +1493,7 @@ __global__ void k_strgrp_compact(const DevSeg* segs, const SegEx* segex,
         int64_t id = (int64_t)(sl.rep & 0xFFFFFFFFULL);
         uint32_t len;
         const char* p = dict_entry(segs[key_seg_off + seg], segex[key_seg_off + seg],
-                                   id, &len);
+                                   id - 1, &len);
         unsigned long long off = atomicAdd(pool_cursor, (unsigned long long)len);
         if (off + len > pool_cap) { th->overflow = 1; continue; }
         for (uint32_t k = 0; k < len; k++) pool[off + k] = p[k];
