@@ -63,6 +63,51 @@ def gen_encode_column(n, seed, lo, hi, nthreads, base_seed_tag):
     return encs
 
 
+STR_WINDOW = 65536             # distinct keys per generation slice (strgroup)
+
+
+def gen_encode_strkey_column(n, seed, nthreads):
+    """String key column for the config-5 family: each 2Mi-row slice draws
+    from its own disjoint window of STR_WINDOW keys, so every 128Ki-row
+    segment sees ~56K distinct values and the reference's min-size rule
+    picks DictionaryDense — global distinct ≈ (n/SLICE)·STR_WINDOW."""
+    import ytsaurus_amd as y
+
+    slices = [(s, min(s + SLICE, n)) for s in range(0, n, SLICE)]
+
+    def work(item):
+        i, (b, e) = item
+        m = e - b
+        rng = np.random.default_rng([seed, 777, i])
+        ids = i * STR_WINDOW + rng.integers(0, STR_WINDOW, m)
+        # fixed-width 10-byte keys "k%09d", built with numpy (no per-row Python)
+        dig = (ids[:, None] // 10 ** np.arange(8, -1, -1)) % 10
+        arr = np.empty((m, 10), dtype=np.uint8)
+        arr[:, 0] = ord("k")
+        arr[:, 1:] = dig + ord("0")
+        begins = np.arange(m, dtype=np.uint64) * 10
+        lens = np.full(m, 10, dtype=np.uint32)
+        nulls = np.zeros(m, dtype=np.uint8)
+        return y.encode_string_raw(arr.tobytes(), begins, lens, nulls)
+
+    with ThreadPoolExecutor(max_workers=nthreads) as ex:
+        return list(ex.map(work, enumerate(slices)))
+
+
+def gen_encode_double_column(n, seed, nthreads):
+    import ytsaurus_amd as y
+
+    slices = [(s, min(s + SLICE, n)) for s in range(0, n, SLICE)]
+
+    def work(item):
+        i, (b, e) = item
+        rng = np.random.default_rng([seed, 778, i])
+        return y.encode_double(rng.random(e - b))
+
+    with ThreadPoolExecutor(max_workers=nthreads) as ex:
+        return list(ex.map(work, enumerate(slices)))
+
+
 def build_device_chunk(enc_cols, n, torch):
     """Concatenate per-slice encodings into one device chunk."""
     from ytsaurus_amd._abi import YtChunk, YtColumn, YtSegment
@@ -96,7 +141,7 @@ def build_device_chunk(enc_cols, n, torch):
     return ch, total_bytes
 
 
-def cpu_baseline_leg(plan_f, enc_cols, n, cores):
+def cpu_baseline_leg(plan_f, enc_cols, n, cores, out_cap, pool_bytes=0):
     """Time the oracle (CPU restatement, kind 'port') on a bounded sample of
     the same workload: enough slices for ~10-30 s of CPU work."""
     import ytsaurus_amd as y
@@ -126,7 +171,7 @@ def cpu_baseline_leg(plan_f, enc_cols, n, cores):
 
     from ytsaurus_amd.api import _mk_rowset
     from ytsaurus_amd import _abi
-    rs = _mk_rowset(KEY_SPACE + 1024, 4)
+    rs = _mk_rowset(out_cap, 4, pool_bytes=pool_bytes)
     st = _abi.YtStatistics()
     err = ctypes.create_string_buffer(256)
     t0 = time.monotonic()
@@ -156,7 +201,7 @@ def main():
                     help="rows per GPU (weak scaling)")
     ap.add_argument("--keys", type=int, default=KEY_SPACE)
     ap.add_argument("--workload", default="groupby",
-                    choices=["groupby", "scanfilter"])
+                    choices=["groupby", "scanfilter", "strgroup"])
     ap.add_argument("--no-cpu-baseline", action="store_true")
     ap.add_argument("--traffic-bytes", type=float, default=0.0,
                     help="measured per-launch HBM bytes from a rocprofv3 --pmc run")
@@ -183,7 +228,7 @@ def main():
     key_space = args.keys
 
     def make_plan():
-        if args.workload == "groupby":
+        if args.workload in ("groupby", "strgroup"):
             return y.Plan(keys=[y.col(0)],
                           aggs=[y.agg_sum(y.col(1)), y.agg_sum1()])
         lo, hi = int(0.25 * 2**VAL_BITS), int(0.75 * 2**VAL_BITS)
@@ -191,17 +236,29 @@ def main():
                       aggs=[y.agg_sum(y.col(1)), y.agg_sum(y.col(2)),
                             y.agg_sum(y.col(3)), y.agg_sum1()])
 
-    ncols = 2 if args.workload == "groupby" else 4
+    ncols = 4 if args.workload == "scanfilter" else 2
     cores = os.cpu_count() or 8
 
+    if args.workload == "strgroup":
+        assert world == 1, "strgroup: single-GPU this round (string partial " \
+                           "states not in the exchange format yet)"
+        # BASELINE configs[4] family: dictionary-encoded string key GROUP BY
+        # + double sum, via the global-atomic per-dictionary-id accumulate
+        # path; distinct keys ≈ slices × STR_WINDOW
+        key_space = ((n + SLICE - 1) // SLICE) * STR_WINDOW
+
     t0 = time.monotonic()
-    enc_cols = []
-    for ci in range(ncols):
-        if args.workload == "groupby" and ci == 0:
-            lo, hi = 0, key_space
-        else:
-            lo, hi = 0, 2**VAL_BITS
-        enc_cols.append(gen_encode_column(n, SEED + rank, lo, hi, cores, ci))
+    if args.workload == "strgroup":
+        enc_cols = [gen_encode_strkey_column(n, SEED + rank, cores),
+                    gen_encode_double_column(n, SEED + rank, cores)]
+    else:
+        enc_cols = []
+        for ci in range(ncols):
+            if args.workload == "groupby" and ci == 0:
+                lo, hi = 0, key_space
+            else:
+                lo, hi = 0, 2**VAL_BITS
+            enc_cols.append(gen_encode_column(n, SEED + rank, lo, hi, cores, ci))
     log("rank %d: generated+encoded %d rows x %d cols in %.1fs"
         % (rank, n, ncols, time.monotonic() - t0))
 
@@ -214,7 +271,9 @@ def main():
 
     plan = make_plan()
     hint = key_space if args.workload == "groupby" else 0
-    out_rs = y.make_rowset(key_space + 1024, 1 + len(plan.aggs))
+    pool_b = (key_space * 10 + (1 << 20)) if args.workload == "strgroup" else 0
+    out_rs = y.make_rowset(key_space + 4096, 1 + len(plan.aggs),
+                           pool_bytes=pool_b)
 
     # multi-GPU state buffers
     if dist is not None:
@@ -307,7 +366,13 @@ def main():
 
     cpu_baseline = None
     if rank == 0 and world == 1 and not args.no_cpu_baseline:
-        cpu_baseline = cpu_baseline_leg(make_plan, enc_cols, n, cores)
+        bl_pool = 0
+        bl_cap = KEY_SPACE + 1024
+        if args.workload == "strgroup":
+            bl_cap = key_space + 4096
+            bl_pool = key_space * 10 + (1 << 20)
+        cpu_baseline = cpu_baseline_leg(make_plan, enc_cols, n, cores,
+                                        bl_cap, bl_pool)
 
     if rank == 0:
         out = {
@@ -324,13 +389,18 @@ def main():
             "dtype": "int64",
             "data": "synthetic",
             "config": {
-                "workload": "groupby_1M_distinct@%drows" % n
-                            if args.workload == "groupby"
-                            else "scan_filter_sum_4col@%drows" % n,
+                "workload": {
+                    "groupby": "groupby_1M_distinct@%drows" % n,
+                    "scanfilter": "scan_filter_sum_4col@%drows" % n,
+                    "strgroup": "strgroup_dict_%ddistinct@%drows"
+                                % (key_space, n),
+                }[args.workload],
                 "rows_per_gpu": n,
                 "distinct_keys": key_space,
                 "columns": ncols,
-                "chunk_format": "unversioned DirectDense int64 (reference layout)",
+                "chunk_format": "unversioned DictionaryDense string + double"
+                                if args.workload == "strgroup"
+                                else "unversioned DirectDense int64 (reference layout)",
                 "encoded_gb": enc_bytes / 1e9,
             },
             "roofline": roofline,

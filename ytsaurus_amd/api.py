@@ -19,7 +19,7 @@ from ._abi import (
 
 __all__ = [
     "col", "lit", "null", "litf", "Plan", "agg_sum", "agg_sum1",
-    "encode_int64", "encode_double", "encode_bool", "encode_string",
+    "encode_int64", "encode_double", "encode_bool", "encode_string", "encode_string_raw",
     "oracle_decode_strings",
     "Chunk", "oracle_execute",
     "oracle_partial", "oracle_merge",
@@ -201,6 +201,24 @@ def encode_int64(values, nulls=None, max_segment_values=0, unsigned=False,
     return EncodedColumn(vt, enc)
 
 
+def encode_string_raw(blob, begins, lens, nulls, max_segment_values=0):
+    """Encode a string column from a flat byte blob + per-row (begin, len,
+    null) arrays — the fast path for synthetic-data generation (numpy builds
+    the blob, no per-row Python)."""
+    begins = np.ascontiguousarray(begins, dtype=np.uint64)
+    lens = np.ascontiguousarray(lens, dtype=np.uint32)
+    nulls = np.ascontiguousarray(nulls, dtype=np.uint8)
+    enc = YtEncodedColumn()
+    err = C.create_string_buffer(256)
+    rc = _abi.gpu_lib().yt_encode_string_column(
+        blob, begins.ctypes.data_as(C.POINTER(C.c_uint64)),
+        lens.ctypes.data_as(C.POINTER(C.c_uint32)),
+        nulls.ctypes.data_as(C.POINTER(C.c_uint8)),
+        len(begins), max_segment_values, C.byref(enc), err, 256)
+    _check(rc, err)
+    return EncodedColumn(VT_STRING, enc)
+
+
 def encode_string(strings, max_segment_values=0):
     """strings: list of bytes/str or None (null). Encodes into the
     reference's unversioned string segment formats."""
@@ -220,15 +238,7 @@ def encode_string(strings, max_segment_values=0):
         lens[i] = len(v)
         at += len(v)
     blob = b"".join(blobs)
-    enc = YtEncodedColumn()
-    err = C.create_string_buffer(256)
-    rc = _abi.gpu_lib().yt_encode_string_column(
-        blob, begins.ctypes.data_as(C.POINTER(C.c_uint64)),
-        lens.ctypes.data_as(C.POINTER(C.c_uint32)),
-        nulls.ctypes.data_as(C.POINTER(C.c_uint8)),
-        len(strings), max_segment_values, C.byref(enc), err, 256)
-    _check(rc, err)
-    return EncodedColumn(VT_STRING, enc)
+    return encode_string_raw(blob, begins, lens, nulls, max_segment_values)
 
 
 def oracle_decode_strings(enc, n):
@@ -462,8 +472,8 @@ def gpu_execute(plan, device_chunk, max_groups_hint=0, group_row_limit=0,
     return rows_from_rowset(rs), st
 
 
-def make_rowset(capacity, ncols):
-    return _mk_rowset(capacity, ncols)
+def make_rowset(capacity, ncols, pool_bytes=0):
+    return _mk_rowset(capacity, ncols, pool_bytes=pool_bytes)
 
 
 def gpu_partial(plan, device_chunk, nparts, states_dev_ptr, capacity_rows,
