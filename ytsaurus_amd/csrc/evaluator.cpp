@@ -1354,6 +1354,7 @@ static int run_string_group(const YtPlan* plan, const YtChunk* chunk,
                      errbuf, errlen);
     if (rc) return rc;
     output->row_count = 0;
+    output->string_pool_used = 0;   /* rowsets are reusable across queries */
     output->column_count = 1 + plan->agg_count;
     if (chunk->row_count == 0 || R.nsegs == 0) return YT_OK;
     rc = setup_table(&R, plan->agg_count, options->max_groups_hint,
