@@ -17,6 +17,7 @@
 #include <algorithm>
 #include <mutex>
 #include <chrono>
+#include <thread>
 
 #include "common.h"
 #include "../../include/ytql_gpu.h"
@@ -1476,10 +1477,6 @@ static int run_string_group(const YtPlan* plan, const YtChunk* chunk,
         HIP_CHECK(hipMemcpyAsync(hctr, d_ctr, 2 * sizeof(unsigned long long),
                                  hipMemcpyDeviceToHost, R.stream));
         HIP_CHECK(hipStreamSynchronize(R.stream));
-        std::vector<char> pool(hctr[1] ? hctr[1] : 1);
-        if (hctr[1]) {
-            HIP_CHECK(hipMemcpy(pool.data(), d_pool, hctr[1], hipMemcpyDeviceToHost));
-        }
         float ms = 0;
         HIP_CHECK(hipEventElapsedTime(&ms, e0, e1));
         hipEventDestroy(e0);
@@ -1492,9 +1489,106 @@ static int run_string_group(const YtPlan* plan, const YtChunk* chunk,
             return YT_ERR_CAPACITY;
         }
 
-        /* emit [key(string), aggs...] rows */
+        /* emit [key(string), aggs...] rows. Fast path (no output limit,
+         * everything fits): copy the device key pool wholesale — compact's
+         * pool_off values are global offsets into it — and fill the YtValue
+         * rows with host threads. */
         int ncols = 1 + plan->agg_count;
         int out_limited = 0;
+        int64_t ng = (int64_t)groups.size();
+        int has_null_row = th.side_used[1] ? 1 : 0;
+        if (options->output_row_limit == 0 &&
+            ng + has_null_row <= output->capacity_rows &&
+            (unsigned long long)hctr[1] <= (unsigned long long)output->string_pool_capacity &&
+            (output->string_pool != nullptr || hctr[1] == 0)) {
+            if (hctr[1]) {
+                HIP_CHECK(hipMemcpy(output->string_pool, d_pool, hctr[1],
+                                    hipMemcpyDeviceToHost));
+            }
+            output->string_pool_used = hctr[1];
+            int agg_is_sum1[kMaxAggs];
+            for (int a = 0; a < plan->agg_count; a++)
+                agg_is_sum1[a] = plan->aggs[a]->func == YT_AGG_SUM1;
+            auto emit_range = [&](int64_t b, int64_t e) {
+                for (int64_t gI = b; gI < e; gI++) {
+                    const OutStrGroup& g = groups[gI];
+                    YtValue* dst = output->values + gI * ncols;
+                    dst[0].id = 0;
+                    dst[0].flags = 0;
+                    dst[0].type = YT_VT_STRING;
+                    dst[0].length = g.len;
+                    dst[0].data.str = output->string_pool + g.pool_off;
+                    for (int a = 0; a < plan->agg_count; a++) {
+                        YtValue& v = dst[1 + a];
+                        v.id = (uint16_t)(1 + a);
+                        v.flags = 0;
+                        v.length = 0;
+                        if (agg_is_sum1[a]) {
+                            v.type = YT_VT_INT64;
+                            v.data.bits = g.cnt;
+                        } else if (g.nonnull == 0) {
+                            v.type = YT_VT_NULL;
+                            v.data.bits = 0;
+                        } else {
+                            v.type = val_is_double ? YT_VT_DOUBLE : YT_VT_INT64;
+                            v.data.bits = g.sum_bits;
+                        }
+                    }
+                }
+            };
+            int nt = (int)std::thread::hardware_concurrency();
+            if (nt < 1) nt = 1;
+            if (nt > 64) nt = 64;
+            if (ng < (1 << 16)) nt = 1;
+            if (nt == 1) {
+                emit_range(0, ng);
+            } else {
+                std::vector<std::thread> ths;
+                int64_t per = (ng + nt - 1) / nt;
+                for (int t = 0; t < nt; t++) {
+                    int64_t b = (int64_t)t * per, e = b + per;
+                    if (b >= ng) break;
+                    if (e > ng) e = ng;
+                    ths.emplace_back(emit_range, b, e);
+                }
+                for (auto& t : ths) t.join();
+            }
+            output->row_count = ng;
+            if (has_null_row) {
+                YtValue* dst = output->values + output->row_count * ncols;
+                dst[0].id = 0;
+                dst[0].flags = 0;
+                dst[0].type = YT_VT_NULL;
+                dst[0].length = 0;
+                dst[0].data.bits = 0;
+                uint64_t cnt = th.side_cnt[1];
+                uint64_t sum_bits = sum_slot >= 0 ? th.side_agg[1][2 * sum_slot] : 0;
+                uint64_t nonnull = sum_slot >= 0 ? th.side_agg[1][2 * sum_slot + 1] : 0;
+                for (int a = 0; a < plan->agg_count; a++) {
+                    YtValue& v = dst[1 + a];
+                    v.id = (uint16_t)(1 + a);
+                    v.flags = 0;
+                    v.length = 0;
+                    if (plan->aggs[a]->func == YT_AGG_SUM1) {
+                        v.type = YT_VT_INT64;
+                        v.data.bits = cnt;
+                    } else if (nonnull == 0) {
+                        v.type = YT_VT_NULL;
+                        v.data.bits = 0;
+                    } else {
+                        v.type = val_is_double ? YT_VT_DOUBLE : YT_VT_INT64;
+                        v.data.bits = sum_bits;
+                    }
+                }
+                output->row_count++;
+            }
+            goto emitted;
+        }
+        {
+        std::vector<char> pool(hctr[1] ? hctr[1] : 1);
+        if (hctr[1]) {
+            HIP_CHECK(hipMemcpy(pool.data(), d_pool, hctr[1], hipMemcpyDeviceToHost));
+        }
         for (int64_t gI = 0; gI < (int64_t)groups.size() + 2; gI++) {
             const char* kstr = nullptr;
             uint32_t klen = 0;
@@ -1562,6 +1656,8 @@ static int run_string_group(const YtPlan* plan, const YtChunk* chunk,
             }
             output->row_count++;
         }
+        }
+emitted:
         if (stats) {
             stats->rows_read = chunk->row_count;
             stats->rows_written = output->row_count;
