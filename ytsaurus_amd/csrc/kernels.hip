@@ -1485,19 +1485,47 @@ __global__ void k_strgrp_compact(const DevSeg* segs, const SegEx* segex,
                                  char* pool, unsigned long long* pool_cursor,
                                  uint64_t pool_cap, TableHdr* th)
 {
+    /* counter/pool_cursor are single hot addresses: aggregate per wavefront
+     * (ballot + shfl scan), one atomic pair per 64 slots instead of per
+     * group — the naive form measured 78 ms at 6.3 M groups. */
+    const int lane = threadIdx.x & 63;
     for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
          i < nslots; i += (uint64_t)gridDim.x * blockDim.x) {
         const StrSlot& sl = slots[i];
-        if (sl.rep == 0) continue;
-        int seg = (int)(sl.rep >> 32) - 1;
-        int64_t id = (int64_t)(sl.rep & 0xFFFFFFFFULL);
-        uint32_t len;
-        const char* p = dict_entry(segs[key_seg_off + seg], segex[key_seg_off + seg],
-                                   id - 1, &len);
-        unsigned long long off = atomicAdd(pool_cursor, (unsigned long long)len);
+        const bool occ = sl.rep != 0;
+        uint32_t len = 0;
+        const char* p = nullptr;
+        if (occ) {
+            int seg = (int)(sl.rep >> 32) - 1;
+            int64_t id = (int64_t)(sl.rep & 0xFFFFFFFFULL);
+            p = dict_entry(segs[key_seg_off + seg], segex[key_seg_off + seg],
+                           id - 1, &len);
+        }
+        uint64_t mask = __ballot(occ);
+        if (mask == 0) continue;
+        /* inclusive shfl scan of len across the wave (inactive lanes add 0) */
+        unsigned long long run = len;
+        #pragma unroll
+        for (int d = 1; d < 64; d <<= 1) {
+            unsigned long long v = __shfl_up(run, d, 64);
+            if (lane >= d) run += v;
+        }
+        unsigned long long excl = run - len;
+        unsigned long long total = __shfl(run, 63, 64);
+        int leader = __ffsll((unsigned long long)mask) - 1;
+        unsigned long long poolbase = 0, cntbase = 0;
+        if (lane == leader) {
+            poolbase = atomicAdd(pool_cursor, total);
+            cntbase = atomicAdd(counter, (unsigned long long)__popcll(mask));
+        }
+        poolbase = __shfl(poolbase, leader, 64);
+        cntbase = __shfl(cntbase, leader, 64);
+        if (!occ) continue;
+        unsigned long long off = poolbase + excl;
         if (off + len > pool_cap) { th->overflow = 1; continue; }
         for (uint32_t k = 0; k < len; k++) pool[off + k] = p[k];
-        unsigned long long idx = atomicAdd(counter, 1ULL);
+        unsigned long long idx = cntbase
+            + (unsigned long long)__popcll(mask & ((1ULL << lane) - 1));
         OutStrGroup& g = out[idx];
         g.pool_off = off;
         g.len = len;
