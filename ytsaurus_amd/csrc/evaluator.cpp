@@ -1431,9 +1431,10 @@ static int run_string_group(const YtPlan* plan, const YtChunk* chunk,
         HIP_CHECK(pool_alloc(&d_slots, sizeof(StrSlot) * nslots));
         HIP_CHECK(hipMemsetAsync(d_slots, 0, sizeof(StrSlot) * nslots, R.stream));
 
-        hipEvent_t e0, e1;
+        hipEvent_t e0, e1, e2;
         HIP_CHECK(hipEventCreate(&e0));
         HIP_CHECK(hipEventCreate(&e1));
+        HIP_CHECK(hipEventCreate(&e2));
         HIP_CHECK(hipEventRecord(e0, R.stream));
         HIP_CHECK(ytql_launch_strgrp_accum(&sp, R.d_segs, R.d_segex, d_accbase,
                                            d_acc, R.d_th, R.stream));
@@ -1467,24 +1468,30 @@ static int run_string_group(const YtPlan* plan, const YtChunk* chunk,
                                                  d_pool, d_ctr + 1, pool_cap,
                                                  R.d_th, R.stream));
         }
-        std::vector<OutStrGroup> groups(ngroups ? ngroups : 0);
+        OutStrGroup* hgroups = nullptr;
+        HIP_CHECK(pool_alloc_host(&hgroups,
+                                  sizeof(OutStrGroup) * (ngroups ? ngroups : 1)));
         unsigned long long hctr[2] = {0, 0};
         if (ngroups > 0) {
-            HIP_CHECK(hipMemcpyAsync(groups.data(), d_out,
+            HIP_CHECK(hipMemcpyAsync(hgroups, d_out,
                                      sizeof(OutStrGroup) * ngroups,
                                      hipMemcpyDeviceToHost, R.stream));
         }
         HIP_CHECK(hipMemcpyAsync(hctr, d_ctr, 2 * sizeof(unsigned long long),
                                  hipMemcpyDeviceToHost, R.stream));
+        HIP_CHECK(hipEventRecord(e2, R.stream));
         HIP_CHECK(hipStreamSynchronize(R.stream));
-        float ms = 0;
+        float ms = 0, ms_other = 0;
         HIP_CHECK(hipEventElapsedTime(&ms, e0, e1));
+        HIP_CHECK(hipEventElapsedTime(&ms_other, e1, e2));
         hipEventDestroy(e0);
         hipEventDestroy(e1);
+        hipEventDestroy(e2);
         HIP_CHECK(hipMemcpy(&th, R.d_th, sizeof(th), hipMemcpyDeviceToHost));
         if (th.overflow == 1) {
             g_pool.put(d_accbase); g_pool.put(d_acc); g_pool.put(d_hashes);
-            g_pool.put(d_slots); g_pool.put(d_out); g_pool.put(d_pool); g_pool.put(d_ctr);
+            g_pool.put(d_slots); g_pool.put(d_out); g_pool.put(d_pool);
+            g_pool.put(d_ctr); g_pool.put(hgroups);
             set_err(errbuf, errlen, "string merge table/pool overflow");
             return YT_ERR_CAPACITY;
         }
@@ -1495,7 +1502,7 @@ static int run_string_group(const YtPlan* plan, const YtChunk* chunk,
          * rows with host threads. */
         int ncols = 1 + plan->agg_count;
         int out_limited = 0;
-        int64_t ng = (int64_t)groups.size();
+        int64_t ng = ngroups;
         int has_null_row = th.side_used[1] ? 1 : 0;
         if (options->output_row_limit == 0 &&
             ng + has_null_row <= output->capacity_rows &&
@@ -1511,7 +1518,7 @@ static int run_string_group(const YtPlan* plan, const YtChunk* chunk,
                 agg_is_sum1[a] = plan->aggs[a]->func == YT_AGG_SUM1;
             auto emit_range = [&](int64_t b, int64_t e) {
                 for (int64_t gI = b; gI < e; gI++) {
-                    const OutStrGroup& g = groups[gI];
+                    const OutStrGroup& g = hgroups[gI];
                     YtValue* dst = output->values + gI * ncols;
                     dst[0].id = 0;
                     dst[0].flags = 0;
@@ -1589,20 +1596,20 @@ static int run_string_group(const YtPlan* plan, const YtChunk* chunk,
         if (hctr[1]) {
             HIP_CHECK(hipMemcpy(pool.data(), d_pool, hctr[1], hipMemcpyDeviceToHost));
         }
-        for (int64_t gI = 0; gI < (int64_t)groups.size() + 2; gI++) {
+        for (int64_t gI = 0; gI < ngroups + 2; gI++) {
             const char* kstr = nullptr;
             uint32_t klen = 0;
             int knull = 0;
             uint64_t cnt, sum_bits, nonnull;
-            if (gI < (int64_t)groups.size()) {
-                const OutStrGroup& g = groups[gI];
+            if (gI < ngroups) {
+                const OutStrGroup& g = hgroups[gI];
                 kstr = pool.data() + g.pool_off;
                 klen = g.len;
                 cnt = g.cnt;
                 sum_bits = g.sum_bits;
                 nonnull = g.nonnull;
             } else {
-                int side = (int)(gI - (int64_t)groups.size());
+                int side = (int)(gI - ngroups);
                 if (side == 0) continue;                /* no sentinel for strings */
                 if (!th.side_used[1]) continue;
                 knull = 1;
@@ -1666,10 +1673,12 @@ emitted:
             stats->incomplete_output = out_limited || (th.overflow == 2);
             stats->kernel_scan_ms += ms;
             stats->kernel_scan_launches += 1;
+            stats->kernel_other_ms += ms_other;   /* compact + result D2H */
             stats->execute_time_ms = now_ms() - tw0;
         }
         g_pool.put(d_accbase); g_pool.put(d_acc); g_pool.put(d_hashes);
-        g_pool.put(d_slots); g_pool.put(d_out); g_pool.put(d_pool); g_pool.put(d_ctr);
+        g_pool.put(d_slots); g_pool.put(d_out); g_pool.put(d_pool);
+        g_pool.put(d_ctr); g_pool.put(hgroups);
     }
     return rc;
 fail:
