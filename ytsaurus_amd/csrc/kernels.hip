@@ -1477,7 +1477,14 @@ __global__ void k_strgrp_merge(const DevSeg* segs, const SegEx* segex,
 }
 
 /* S4: compact occupied slots; copy each group's key string into the device
- * pool */
+ * pool. Two-pass block compaction: counter/pool_cursor are single hot
+ * addresses and single-address atomic RMW serializes at ~160 M/s (probe
+ * tools/probe_strcompact.hip: per-group atomics = 39 ms, wave-aggregated
+ * still 78 ms at 2.4 % slot occupancy — 78 % of waves contain an occupied
+ * slot — while the raw 6.4 GB slot scan is 1.1 ms at 5.9 TB/s). So: each
+ * block scans its contiguous slot range twice — pass A counts groups and
+ * key bytes and claims space with ONE atomic pair per block, pass B emits
+ * with an in-block (wave shfl scan + LDS cross-wave) prefix. */
 __global__ void k_strgrp_compact(const DevSeg* segs, const SegEx* segex,
                                  int key_seg_off,
                                  const StrSlot* slots, uint64_t nslots,
@@ -1485,54 +1492,101 @@ __global__ void k_strgrp_compact(const DevSeg* segs, const SegEx* segex,
                                  char* pool, unsigned long long* pool_cursor,
                                  uint64_t pool_cap, TableHdr* th)
 {
-    /* counter/pool_cursor are single hot addresses: aggregate per wavefront
-     * (ballot + shfl scan), one atomic pair per 64 slots instead of per
-     * group — the naive form measured 78 ms at 6.3 M groups. */
     const int lane = threadIdx.x & 63;
-    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
-         i < nslots; i += (uint64_t)gridDim.x * blockDim.x) {
-        const StrSlot& sl = slots[i];
-        const bool occ = sl.rep != 0;
+    const int wid = (int)(threadIdx.x >> 6);
+    const int nw = (int)(blockDim.x >> 6);
+    __shared__ unsigned long long s_wcnt[16], s_wlen[16];
+    __shared__ unsigned long long s_cbase, s_pbase, s_crun, s_lrun;
+
+    uint64_t per = (nslots + gridDim.x - 1) / gridDim.x;
+    uint64_t b0 = (uint64_t)blockIdx.x * per;
+    if (b0 > nslots) b0 = nslots;
+    uint64_t b1 = b0 + per;
+    if (b1 > nslots) b1 = nslots;
+
+    /* pass A: count occupied slots + total key bytes in [b0, b1) */
+    unsigned long long c = 0, l = 0;
+    for (uint64_t i = b0 + threadIdx.x; i < b1; i += blockDim.x) {
+        unsigned long long rep = slots[i].rep;
+        if (rep) {
+            int seg = (int)(rep >> 32) - 1;
+            int64_t id = (int64_t)(rep & 0xFFFFFFFFULL);
+            uint32_t len;
+            (void)dict_entry(segs[key_seg_off + seg], segex[key_seg_off + seg],
+                             id - 1, &len);
+            c++;
+            l += len;
+        }
+    }
+    #pragma unroll
+    for (int d = 32; d; d >>= 1) {
+        c += __shfl_down(c, d, 64);
+        l += __shfl_down(l, d, 64);
+    }
+    if (lane == 0) { s_wcnt[wid] = c; s_wlen[wid] = l; }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        unsigned long long tc = 0, tl = 0;
+        for (int w = 0; w < nw; w++) { tc += s_wcnt[w]; tl += s_wlen[w]; }
+        s_cbase = tc ? atomicAdd(counter, tc) : 0;
+        s_pbase = tl ? atomicAdd(pool_cursor, tl) : 0;
+        s_crun = 0;
+        s_lrun = 0;
+    }
+    __syncthreads();
+    const unsigned long long cbase = s_cbase, pbase = s_pbase;
+
+    /* pass B: ordered emit with an in-block prefix per tile */
+    for (uint64_t base = b0; base < b1; base += blockDim.x) {
+        uint64_t i = base + threadIdx.x;
+        bool occ = false;
         uint32_t len = 0;
         const char* p = nullptr;
-        if (occ) {
-            int seg = (int)(sl.rep >> 32) - 1;
-            int64_t id = (int64_t)(sl.rep & 0xFFFFFFFFULL);
-            p = dict_entry(segs[key_seg_off + seg], segex[key_seg_off + seg],
-                           id - 1, &len);
+        const StrSlot* sl = nullptr;
+        if (i < b1) {
+            sl = &slots[i];
+            unsigned long long rep = sl->rep;
+            if (rep) {
+                occ = true;
+                int seg = (int)(rep >> 32) - 1;
+                int64_t id = (int64_t)(rep & 0xFFFFFFFFULL);
+                p = dict_entry(segs[key_seg_off + seg],
+                               segex[key_seg_off + seg], id - 1, &len);
+            }
         }
-        uint64_t mask = __ballot(occ);
-        if (mask == 0) continue;
-        /* inclusive shfl scan of len across the wave (inactive lanes add 0) */
-        unsigned long long run = len;
+        unsigned long long mask = __ballot(occ);
+        unsigned long long run = len;     /* inclusive wave scan of len */
         #pragma unroll
         for (int d = 1; d < 64; d <<= 1) {
             unsigned long long v = __shfl_up(run, d, 64);
             if (lane >= d) run += v;
         }
-        unsigned long long excl = run - len;
-        unsigned long long total = __shfl(run, 63, 64);
-        int leader = __ffsll((unsigned long long)mask) - 1;
-        unsigned long long poolbase = 0, cntbase = 0;
-        if (lane == leader) {
-            poolbase = atomicAdd(pool_cursor, total);
-            cntbase = atomicAdd(counter, (unsigned long long)__popcll(mask));
+        if (lane == 63) { s_wcnt[wid] = __popcll(mask); s_wlen[wid] = run; }
+        __syncthreads();
+        if (occ) {
+            unsigned long long wc = 0, wl = 0;
+            for (int w = 0; w < wid; w++) { wc += s_wcnt[w]; wl += s_wlen[w]; }
+            unsigned long long off = pbase + s_lrun + wl + (run - len);
+            unsigned long long idx = cbase + s_crun + wc
+                + (unsigned long long)__popcll(mask & ((1ULL << lane) - 1));
+            if (off + len > pool_cap) {
+                th->overflow = 1;
+            } else {
+                for (uint32_t k = 0; k < len; k++) pool[off + k] = p[k];
+                OutStrGroup& g = out[idx];
+                g.pool_off = off;
+                g.len = len;
+                g.pad_ = 0;
+                g.sum_bits = sl->sum_bits;
+                g.cnt = sl->cnt;
+                g.nonnull = sl->nonnull;
+            }
         }
-        poolbase = __shfl(poolbase, leader, 64);
-        cntbase = __shfl(cntbase, leader, 64);
-        if (!occ) continue;
-        unsigned long long off = poolbase + excl;
-        if (off + len > pool_cap) { th->overflow = 1; continue; }
-        for (uint32_t k = 0; k < len; k++) pool[off + k] = p[k];
-        unsigned long long idx = cntbase
-            + (unsigned long long)__popcll(mask & ((1ULL << lane) - 1));
-        OutStrGroup& g = out[idx];
-        g.pool_off = off;
-        g.len = len;
-        g.pad_ = 0;
-        g.sum_bits = sl.sum_bits;
-        g.cnt = sl.cnt;
-        g.nonnull = sl.nonnull;
+        __syncthreads();
+        if (threadIdx.x == 0) {
+            for (int w = 0; w < nw; w++) { s_crun += s_wcnt[w]; s_lrun += s_wlen[w]; }
+        }
+        __syncthreads();
     }
 }
 
