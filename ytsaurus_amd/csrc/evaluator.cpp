@@ -1426,8 +1426,22 @@ static int run_string_group(const YtPlan* plan, const YtChunk* chunk,
         HIP_CHECK(pool_alloc(&d_acc, sizeof(uint64_t) * 2 * (total_dict ? total_dict : 1)));
         HIP_CHECK(hipMemsetAsync(d_acc, 0, sizeof(uint64_t) * 2 * (total_dict ? total_dict : 1), R.stream));
         HIP_CHECK(pool_alloc(&d_hashes, sizeof(uint64_t) * (total_dict ? total_dict : 1)));
-        uint64_t nslots = next_pow2((uint64_t)(total_dict ? total_dict : 1) * 2);
+        /* distinct keys ≤ total_dict, usually far less (each key recurs in
+         * ~every segment). Size the merge table by an estimate and grow 4×
+         * on overflow (merge terminates with overflow=1 on a full table):
+         * a small table keeps the touched slots cache-resident for the
+         * merge's scattered atomics and shrinks compact's scan. */
+        int64_t max_dict = 0;
+        for (int i = 0; i < knseg; i++)
+            if ((int64_t)ksegex[i].dict_size > max_dict)
+                max_dict = ksegex[i].dict_size;
+        uint64_t nslots_cap = next_pow2((uint64_t)(total_dict ? total_dict : 1) * 2);
+        if (nslots_cap < 2048) nslots_cap = 2048;
+        uint64_t est = (uint64_t)max_dict;
+        if (est < (uint64_t)(total_dict / 8) + 1) est = (uint64_t)(total_dict / 8) + 1;
+        uint64_t nslots = next_pow2(est * 2);
         if (nslots < 2048) nslots = 2048;
+        if (nslots > nslots_cap) nslots = nslots_cap;
         HIP_CHECK(pool_alloc(&d_slots, sizeof(StrSlot) * nslots));
         HIP_CHECK(hipMemsetAsync(d_slots, 0, sizeof(StrSlot) * nslots, R.stream));
 
@@ -1440,14 +1454,25 @@ static int run_string_group(const YtPlan* plan, const YtChunk* chunk,
                                            d_acc, R.d_th, R.stream));
         HIP_CHECK(ytql_launch_strgrp_hash(R.d_segs, R.d_segex, koff, knseg,
                                           d_accbase, d_hashes, total_dict, R.stream));
-        HIP_CHECK(ytql_launch_strgrp_merge(R.d_segs, R.d_segex, koff, knseg,
-                                           d_accbase, d_acc, d_hashes,
-                                           d_slots, nslots, val_is_double,
-                                           R.d_th, total_dict, R.stream));
+        for (;;) {
+            HIP_CHECK(ytql_launch_strgrp_merge(R.d_segs, R.d_segex, koff, knseg,
+                                               d_accbase, d_acc, d_hashes,
+                                               d_slots, nslots, val_is_double,
+                                               R.d_th, total_dict, R.stream));
+            HIP_CHECK(hipMemcpy(&th, R.d_th, sizeof(th), hipMemcpyDeviceToHost));
+            if (th.overflow != 1 || nslots >= nslots_cap) break;
+            g_pool.put(d_slots);
+            nslots *= 4;
+            if (nslots > nslots_cap) nslots = nslots_cap;
+            HIP_CHECK(pool_alloc(&d_slots, sizeof(StrSlot) * nslots));
+            HIP_CHECK(hipMemsetAsync(d_slots, 0, sizeof(StrSlot) * nslots, R.stream));
+            th.ngroups = 0;
+            th.overflow = 0;
+            HIP_CHECK(hipMemcpy(R.d_th, &th, sizeof(th), hipMemcpyHostToDevice));
+        }
         HIP_CHECK(hipEventRecord(e1, R.stream));
 
         /* compact + materialize keys into a device pool */
-        HIP_CHECK(hipMemcpy(&th, R.d_th, sizeof(th), hipMemcpyDeviceToHost));
         int64_t ngroups = (int64_t)th.ngroups;
         uint64_t pool_cap = 0;
         for (int i = 0; i < knseg; i++) {

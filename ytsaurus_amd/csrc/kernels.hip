@@ -1430,7 +1430,12 @@ __global__ void k_strgrp_merge(const DevSeg* segs, const SegEx* segex,
 
         uint64_t sidx = mix64(h) & mask;
         StrSlot* slot = nullptr;
-        for (uint64_t it = 0; it <= mask; it++) {
+        /* probe chains this long only happen when the table is (nearly)
+         * full — give up early so a too-small estimated table overflows in
+         * bounded time and the host retries at 4x (never a wrong result:
+         * overflow always surfaces as retry or YT_ERR_CAPACITY) */
+        uint64_t max_probe = mask < 8192 ? mask : 8192;
+        for (uint64_t it = 0; it <= max_probe; it++) {
             StrSlot* cand = &slots[sidx];
             unsigned long long cur = __hip_atomic_load(&cand->rep, __ATOMIC_RELAXED,
                                                        __HIP_MEMORY_SCOPE_AGENT);
@@ -1463,10 +1468,11 @@ __global__ void k_strgrp_merge(const DevSeg* segs, const SegEx* segex,
             sidx = (sidx + 1) & mask;
         }
         if (!slot) { th->overflow = 1; continue; }
-        atomicAdd((unsigned long long*)&slot->cnt, cn & 0xFFFFFFFFULL);
+        /* cnt|nonnull<<32 packed exactly like the accumulator: one aligned
+         * 64-bit add updates both */
+        atomicAdd((unsigned long long*)&slot->cnt, cn);
         unsigned long long nn = cn >> 32;
         if (nn) {
-            atomicAdd((unsigned long long*)&slot->nonnull, nn);
             if (val_is_double)
                 atomicAdd((double*)&slot->sum_bits,
                           __longlong_as_double(acc[2 * g + 1]));
