@@ -1468,11 +1468,10 @@ __global__ void k_strgrp_merge(const DevSeg* segs, const SegEx* segex,
             sidx = (sidx + 1) & mask;
         }
         if (!slot) { th->overflow = 1; continue; }
-        /* cnt|nonnull<<32 packed exactly like the accumulator: one aligned
-         * 64-bit add updates both */
-        atomicAdd((unsigned long long*)&slot->cnt, cn);
+        atomicAdd((unsigned long long*)&slot->cnt, cn & 0xFFFFFFFFULL);
         unsigned long long nn = cn >> 32;
         if (nn) {
+            atomicAdd((unsigned long long*)&slot->nonnull, nn);
             if (val_is_double)
                 atomicAdd((double*)&slot->sum_bits,
                           __longlong_as_double(acc[2 * g + 1]));
