@@ -144,6 +144,22 @@ typedef struct YtPlan {
     const YtExpr* const* projects;
     int32_t is_merge;             /* front-query mode: input = [keys..., states...] rows
                                      (cg_fragment_compiler.cpp:4016-4124) */
+    /* ORDER BY ... LIMIT — TTopCollector semantics (engine_api/top_collector
+     * -inl.h AddRow: keep the (offset+limit) least rows by the lexicographic
+     * comparer, emit sorted ascending from index offset; registry.cpp
+     * OrderOpHelper:1948-1997). order_cols index the OUTPUT row of this plan
+     * stage (projections in scan mode, [keys..., aggs...] in group mode).
+     * Per-key comparison mirrors the codegen universal comparer
+     * (cg_fragment_compiler.cpp:400-530): null < any value, int64 signed,
+     * uint64/boolean unsigned, double by value (NaN comparison = error),
+     * string memcmp with length tiebreak; a descending key inverts the
+     * outcome (nulls go last). LIMIT is required with ORDER BY, as in the
+     * reference QL. */
+    int32_t order_count;
+    const int32_t* order_cols;    /* output-column indices */
+    const int32_t* order_desc;    /* 0 = ascending, 1 = descending */
+    int64_t order_limit;
+    int64_t order_offset;
 } YtPlan;
 
 /* ---- execution context / statistics ----

@@ -784,6 +784,101 @@ static int emit_group_row(const YtPlan* plan, GroupTable* t, int64_t g,
     return YT_OK;
 }
 
+/* ---- ORDER BY ... LIMIT (TTopCollector restatement) ----
+ * Comparer mirrors the codegen universal comparer
+ * (cg_fragment_compiler.cpp:400-530): null < any, int64 signed,
+ * uint64/boolean unsigned, double by value with NaN comparison an error,
+ * string memcmp + length tiebreak; descending inverts the per-key outcome.
+ * Collector mirrors top_collector-inl.h AddRow: keep the (offset+limit)
+ * least rows (strictly-less eviction of the current max), emit sorted
+ * ascending from index offset (registry.cpp OrderOpHelper:1948-1997). */
+static const YtPlan* g_ord_plan;
+static int g_ord_ncols;
+static int g_ord_err;
+
+static int cmp_order_vals(const YtValue* a, const YtValue* b)
+{
+    const YtPlan* p = g_ord_plan;
+    for (int i = 0; i < p->order_count; i++) {
+        int c = p->order_cols[i];
+        const YtValue* x = &a[c];
+        const YtValue* y = &b[c];
+        int xn = x->type == YT_VT_NULL, yn = y->type == YT_VT_NULL;
+        int r = 0;
+        if (xn || yn) {
+            r = (xn == yn) ? 0 : (xn ? -1 : 1);
+        } else if (x->type == YT_VT_DOUBLE) {
+            double xv = x->data.dbl, yv = y->data.dbl;
+            if (xv != xv || yv != yv) { g_ord_err = 1; return 0; }
+            r = xv < yv ? -1 : (xv > yv ? 1 : 0);
+        } else if (x->type == YT_VT_STRING) {
+            uint32_t lx = x->length, ly = y->length, m = lx < ly ? lx : ly;
+            int mc = m ? memcmp(x->data.str, y->data.str, m) : 0;
+            r = mc ? (mc < 0 ? -1 : 1) : (lx < ly ? -1 : (lx > ly ? 1 : 0));
+        } else if (x->type == YT_VT_INT64) {
+            r = x->data.i64 < y->data.i64 ? -1 : (x->data.i64 > y->data.i64 ? 1 : 0);
+        } else {  /* UINT64 / BOOLEAN */
+            r = x->data.u64 < y->data.u64 ? -1 : (x->data.u64 > y->data.u64 ? 1 : 0);
+        }
+        if (p->order_desc && p->order_desc[i]) r = -r;
+        if (r) return r;
+    }
+    return 0;
+}
+
+static int cmp_order_rowptr(const void* pa, const void* pb)
+{
+    return cmp_order_vals(*(const YtValue* const*)pa, *(const YtValue* const*)pb);
+}
+
+static int ord_validate(const YtPlan* plan, int ncols, char* errbuf, size_t errlen)
+{
+    if (plan->order_limit <= 0) {
+        set_err(errbuf, errlen, "ORDER BY requires LIMIT");
+        return YT_ERR_INVALID_PLAN;
+    }
+    for (int i = 0; i < plan->order_count; i++) {
+        if (!plan->order_cols || plan->order_cols[i] < 0 ||
+            plan->order_cols[i] >= ncols) {
+            set_err(errbuf, errlen, "ORDER BY column out of range");
+            return YT_ERR_INVALID_PLAN;
+        }
+    }
+    return YT_OK;
+}
+
+/* sort the materialized rowset rows and keep [offset, offset+limit) */
+static int apply_order_rowset(const YtPlan* plan, YtRowset* out,
+                              char* errbuf, size_t errlen)
+{
+    int ncols = out->column_count;
+    int rc = ord_validate(plan, ncols, errbuf, errlen);
+    if (rc != YT_OK) return rc;
+    int64_t n = out->row_count;
+    const YtValue** idx = malloc(sizeof(YtValue*) * (n ? n : 1));
+    YtValue* tmp = malloc(sizeof(YtValue) * (n ? n : 1) * ncols);
+    if (!idx || !tmp) { free(idx); free(tmp); return YT_ERR_CAPACITY; }
+    for (int64_t i = 0; i < n; i++) idx[i] = out->values + i * ncols;
+    g_ord_plan = plan;
+    g_ord_ncols = ncols;
+    g_ord_err = 0;
+    qsort(idx, (size_t)n, sizeof(YtValue*), cmp_order_rowptr);
+    if (g_ord_err) {
+        free(idx); free(tmp);
+        set_err(errbuf, errlen, "NaN in ORDER BY comparison");
+        return YT_ERR_LIMIT;
+    }
+    int64_t b = plan->order_offset < n ? plan->order_offset : n;
+    int64_t e = b + plan->order_limit;
+    if (e > n) e = n;
+    for (int64_t i = b; i < e; i++)
+        memcpy(tmp + (i - b) * ncols, idx[i], sizeof(YtValue) * ncols);
+    memcpy(out->values, tmp, sizeof(YtValue) * (e - b) * ncols);
+    out->row_count = e - b;
+    free(idx); free(tmp);
+    return YT_OK;
+}
+
 ORACLE_EXPORT
 int yto_execute(const YtPlan* plan, const YtChunk* chunk,
                 YtRowset* output, YtStatistics* stats,
@@ -865,6 +960,10 @@ int yto_execute(const YtPlan* plan, const YtChunk* chunk,
             output->row_count++;
         }
         output->column_count = np;
+        if (plan->order_count > 0) {
+            rc = apply_order_rowset(plan, output, errbuf, errlen);
+            if (rc != YT_OK) goto done;
+        }
         if (stats) {
             stats->rows_read = n;
             stats->rows_written = output->row_count;
@@ -931,6 +1030,11 @@ int yto_execute(const YtPlan* plan, const YtChunk* chunk,
             }
             for (int64_t g = 0; g < final_t->ngroups && rc == YT_OK; g++) {
                 rc = emit_group_row(plan, final_t, g, output);
+            }
+            output->column_count = plan->project_count
+                ? plan->project_count : plan->key_count + plan->agg_count;
+            if (rc == YT_OK && plan->order_count > 0) {
+                rc = apply_order_rowset(plan, output, errbuf, errlen);
             }
             if (stats) {
                 stats->rows_read = n;

@@ -98,7 +98,12 @@ class YtPlan(C.Structure):
                 ("aggs", C.POINTER(C.POINTER(YtAgg))),
                 ("project_count", C.c_int32),
                 ("projects", C.POINTER(C.POINTER(YtExpr))),
-                ("is_merge", C.c_int32)]
+                ("is_merge", C.c_int32),
+                ("order_count", C.c_int32),
+                ("order_cols", C.POINTER(C.c_int32)),
+                ("order_desc", C.POINTER(C.c_int32)),
+                ("order_limit", C.c_int64),
+                ("order_offset", C.c_int64)]
 
 
 class YtExecOptions(C.Structure):

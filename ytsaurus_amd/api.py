@@ -102,14 +102,20 @@ def agg_sum1():
 
 
 class Plan:
-    """Mirror of the restated TQuery{WhereClause,GroupClause} shape."""
+    """Mirror of the restated TQuery{WhereClause,GroupClause,OrderClause}
+    shape. order_by: list of (output_column_index, desc) pairs; limit is
+    required with order_by (reference QL rejects ORDER BY without LIMIT)."""
 
-    def __init__(self, filter=None, keys=(), aggs=(), projects=(), is_merge=False):
+    def __init__(self, filter=None, keys=(), aggs=(), projects=(), is_merge=False,
+                 order_by=(), limit=0, offset=0):
         self.filter = filter
         self.keys = list(keys)
         self.aggs = list(aggs)
         self.projects = list(projects)
         self.is_merge = is_merge
+        self.order_by = [(c, 1 if d else 0) for (c, d) in order_by]
+        self.limit = limit
+        self.offset = offset
         self._build()
 
     def _build(self):
@@ -148,6 +154,15 @@ class Plan:
             p.projects = arr
             self._keep.append(arr)
         p.is_merge = 1 if self.is_merge else 0
+        p.order_count = len(self.order_by)
+        if self.order_by:
+            oc = (C.c_int32 * len(self.order_by))(*[c for c, _ in self.order_by])
+            od = (C.c_int32 * len(self.order_by))(*[d for _, d in self.order_by])
+            p.order_cols = oc
+            p.order_desc = od
+            self._keep += [oc, od]
+        p.order_limit = self.limit
+        p.order_offset = self.offset
         self.c = p
 
 
