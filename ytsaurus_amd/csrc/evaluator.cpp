@@ -37,6 +37,21 @@ hipError_t ytql_launch_bucket_agg(const void*, const unsigned long long*, int64_
                                   OutGroup*, unsigned long long*, int64_t,
                                   TableHdr*, int, int, int, int,
                                   uint64_t, uint64_t, hipStream_t);
+hipError_t ytql_launch_topk_hist(const DevPlan*, const DevSeg*, const SegEx*,
+                                 const int32_t*, const int32_t*, int64_t,
+                                 const TopkPass*, unsigned long long*,
+                                 unsigned long long*, unsigned*, hipStream_t);
+hipError_t ytql_launch_topk_gather(const DevPlan*, const DevSeg*, const SegEx*,
+                                   const int32_t*, const int32_t*, int64_t,
+                                   const TopkGather*,
+                                   int64_t*, unsigned long long*,
+                                   int64_t*, unsigned long long*,
+                                   int64_t*, unsigned long long*,
+                                   unsigned*, hipStream_t);
+hipError_t ytql_launch_topk_materialize(const DevPlan*, const DevSeg*, const SegEx*,
+                                        const int32_t*, const int32_t*,
+                                        const int64_t*, int64_t, DevOutVal*,
+                                        unsigned*, hipStream_t);
 hipError_t ytql_launch_scan_project(const DevPlan*, const DevSeg*, const SegEx*,
                                     const int32_t*, const int32_t*, int64_t,
                                     DevOutVal*, uint8_t*, unsigned*, hipStream_t);
@@ -1213,6 +1228,403 @@ static int emit_rows(const YtPlan* plan, const YtChunk* chunk,
 }
 
 
+/* ---- ORDER BY ... LIMIT ----
+ * Comparer mirrors the reference codegen universal comparer
+ * (cg_fragment_compiler.cpp:400-530): null < any, int64 signed,
+ * uint64/boolean unsigned, double by value (NaN comparison = error), string
+ * memcmp + length tiebreak; a descending key inverts the outcome. The
+ * collector contract is TTopCollector's (top_collector-inl.h AddRow +
+ * OrderOpHelper registry.cpp:1948-1997): keep the (offset+limit) least
+ * rows, emit sorted ascending from offset. */
+static int ord_cmp_vals(const YtPlan* plan, const YtValue* a, const YtValue* b,
+                        int* nan_err)
+{
+    for (int i = 0; i < plan->order_count; i++) {
+        int c = plan->order_cols[i];
+        const YtValue* x = &a[c];
+        const YtValue* y = &b[c];
+        int xn = x->type == YT_VT_NULL, yn = y->type == YT_VT_NULL;
+        int r = 0;
+        if (xn || yn) {
+            r = (xn == yn) ? 0 : (xn ? -1 : 1);
+        } else if (x->type == YT_VT_DOUBLE) {
+            double xv = x->data.dbl, yv = y->data.dbl;
+            if (xv != xv || yv != yv) { *nan_err = 1; return 0; }
+            r = xv < yv ? -1 : (xv > yv ? 1 : 0);
+        } else if (x->type == YT_VT_STRING) {
+            uint32_t lx = x->length, ly = y->length, m = lx < ly ? lx : ly;
+            int mc = m ? memcmp(x->data.str, y->data.str, m) : 0;
+            r = mc ? (mc < 0 ? -1 : 1) : (lx < ly ? -1 : (lx > ly ? 1 : 0));
+        } else if (x->type == YT_VT_INT64) {
+            r = x->data.i64 < y->data.i64 ? -1 : (x->data.i64 > y->data.i64 ? 1 : 0);
+        } else {
+            r = x->data.u64 < y->data.u64 ? -1 : (x->data.u64 > y->data.u64 ? 1 : 0);
+        }
+        if (plan->order_desc && plan->order_desc[i]) r = -r;
+        if (r) return r;
+    }
+    return 0;
+}
+
+static int ord_validate_plan(const YtPlan* plan, int ncols, char* errbuf,
+                             size_t errlen)
+{
+    if (plan->order_limit <= 0) {
+        set_err(errbuf, errlen, "ORDER BY requires LIMIT");
+        return YT_ERR_INVALID_PLAN;
+    }
+    for (int i = 0; i < plan->order_count; i++) {
+        if (!plan->order_cols || plan->order_cols[i] < 0 ||
+            plan->order_cols[i] >= ncols) {
+            set_err(errbuf, errlen, "ORDER BY column out of range");
+            return YT_ERR_INVALID_PLAN;
+        }
+    }
+    return YT_OK;
+}
+
+/* sort the already-materialized rowset and keep [offset, offset+limit) —
+ * used for grouped output, where the group count is already bounded */
+static int apply_order_host(const YtPlan* plan, YtRowset* out,
+                            char* errbuf, size_t errlen)
+{
+    int ncols = out->column_count;
+    int rc = ord_validate_plan(plan, ncols, errbuf, errlen);
+    if (rc != YT_OK) return rc;
+    int64_t n = out->row_count;
+    std::vector<const YtValue*> idx((size_t)n);
+    for (int64_t i = 0; i < n; i++) idx[i] = out->values + i * ncols;
+    int nan_err = 0;
+    std::sort(idx.begin(), idx.end(),
+              [&](const YtValue* a, const YtValue* b) {
+                  return ord_cmp_vals(plan, a, b, &nan_err) < 0;
+              });
+    if (nan_err) {
+        set_err(errbuf, errlen, "NaN in ORDER BY comparison");
+        return YT_ERR_LIMIT;
+    }
+    int64_t b = plan->order_offset < n ? plan->order_offset : n;
+    int64_t e = b + plan->order_limit;
+    if (e > n) e = n;
+    std::vector<YtValue> tmp((size_t)(e - b) * ncols);
+    for (int64_t i = b; i < e; i++)
+        memcpy(tmp.data() + (i - b) * ncols, idx[i], sizeof(YtValue) * ncols);
+    memcpy(out->values, tmp.data(), sizeof(YtValue) * (e - b) * ncols);
+    out->row_count = e - b;
+    return YT_OK;
+}
+
+/* scan + ORDER BY ... LIMIT without materializing the scan: histogram
+ * k-selection on an order-isomorphic u64 mapping of the first order key
+ * (digits of 11 bits, host-driven refinement), then exact candidate gather
+ * and a full-comparer host sort of the <=(K + tie-cap) survivors.
+ * Selection runs the same expression program as the projection, so any
+ * filter/projection the generic path supports works here. */
+static int run_scan_topk(const YtPlan* plan, const YtChunk* chunk,
+                         const YtExecOptions* options, const DevPlan* dp,
+                         YtRowset* output, YtStatistics* stats, double tw0,
+                         char* errbuf, size_t errlen)
+{
+    int rc = YT_OK;
+    const int np = plan->project_count;
+    rc = ord_validate_plan(plan, np, errbuf, errlen);
+    if (rc != YT_OK) return rc;
+    const int64_t K = plan->order_offset + plan->order_limit;
+    if (K > ((int64_t)1 << 24)) {
+        set_err(errbuf, errlen, "ORDER BY LIMIT capped at 16M rows this round");
+        return YT_ERR_UNSUPPORTED;
+    }
+
+    DeviceRun R2;
+    R2.stream = (hipStream_t)(uintptr_t)options->stream;
+    unsigned mw = 0;
+    int in_clamped = 0;
+    rc = setup_chunk(chunk, &R2, &mw, options->input_row_limit, &in_clamped,
+                     errbuf, errlen);
+    if (rc) return rc;
+    int64_t n = chunk->row_count;
+    if (options->input_row_limit > 0 && options->input_row_limit < n)
+        n = options->input_row_limit;
+    output->row_count = 0;
+    output->column_count = np;
+    if (n == 0 || R2.nsegs == 0) return YT_OK;
+    if (in_clamped && stats) stats->incomplete_input = 1;
+
+    const int ord0 = plan->order_cols[0];
+    const int desc0 = plan->order_desc ? plan->order_desc[0] : 0;
+    const int64_t CAPB = (int64_t)1 << 17;
+
+    {
+        unsigned long long* d_bins = nullptr;   /* 2048 bins + null ctr + 3 ctrs */
+        HIP_CHECK(pool_alloc(&d_bins, sizeof(unsigned long long) * (2048 + 4)));
+        unsigned long long* d_nullc = d_bins + 2048;
+        unsigned long long* d_ctrs = d_bins + 2049;   /* strict, tie, null */
+        HIP_CHECK(hipMemsetAsync(R2.d_err, 0, sizeof(unsigned), R2.stream));
+
+        hipEvent_t e0, e1;
+        HIP_CHECK(hipEventCreate(&e0));
+        HIP_CHECK(hipEventCreate(&e1));
+        HIP_CHECK(hipEventRecord(e0, R2.stream));
+
+        TopkPass tp;
+        tp.order_proj = ord0;
+        tp.desc = desc0;
+        tp.level0 = 1;
+        tp.prefix_shift = 64;
+        tp.prefix = 0;
+        tp.shift = 53;
+        tp.pad_ = 0;
+
+        uint64_t prefix = 0;
+        int ps = 64, shift = 53;
+        int64_t S = 0, T = 0, nonnull = 0, k_nonnull = 0;
+        unsigned long long null_cnt = 0;
+        bool all_nonnull = false, only_nulls = false;
+        std::vector<unsigned long long> h_bins(2048);
+        int launches = 0;
+
+        for (;;) {
+            HIP_CHECK(hipMemsetAsync(d_bins, 0,
+                                     sizeof(unsigned long long) * 2049, R2.stream));
+            HIP_CHECK(ytql_launch_topk_hist(dp, R2.d_segs, R2.d_segex, R2.d_off,
+                                            R2.d_cnt, n, &tp, d_bins, d_nullc,
+                                            R2.d_err, R2.stream));
+            launches++;
+            HIP_CHECK(hipMemcpy(h_bins.data(), d_bins,
+                                sizeof(unsigned long long) * 2048,
+                                hipMemcpyDeviceToHost));
+            unsigned kerr = 0;
+            HIP_CHECK(hipMemcpy(&kerr, R2.d_err, sizeof(unsigned),
+                                hipMemcpyDeviceToHost));
+            if (kerr) {
+                g_pool.put(d_bins);
+                if (kerr == 100) {
+                    set_err(errbuf, errlen, "NaN in ORDER BY comparison");
+                    return YT_ERR_LIMIT;
+                }
+                set_err(errbuf, errlen, "unsupported ORDER BY key type");
+                return (int)kerr;
+            }
+            if (tp.level0) {
+                HIP_CHECK(hipMemcpy(&null_cnt, d_nullc,
+                                    sizeof(unsigned long long),
+                                    hipMemcpyDeviceToHost));
+                nonnull = 0;
+                for (int i = 0; i < 2048; i++) nonnull += (int64_t)h_bins[i];
+                const bool nulls_first = !desc0;
+                int64_t null_take = nulls_first
+                    ? ((int64_t)null_cnt < K ? (int64_t)null_cnt : K)
+                    : (K - nonnull > 0 ? K - nonnull : 0);
+                k_nonnull = K - (nulls_first ? null_take : 0);
+                if (!nulls_first && k_nonnull > nonnull) k_nonnull = nonnull;
+                if (k_nonnull <= 0) { only_nulls = true; break; }
+                if (k_nonnull >= nonnull) { all_nonnull = true; S = nonnull; break; }
+            }
+            int64_t cum = 0;
+            int t = -1;
+            for (int i = 0; i < 2048; i++) {
+                if (S + cum + (int64_t)h_bins[i] >= k_nonnull) { t = i; break; }
+                cum += (int64_t)h_bins[i];
+            }
+            if (t < 0) {   /* cannot happen: counts shrank between passes */
+                g_pool.put(d_bins);
+                set_err(errbuf, errlen, "ORDER BY selection internal error");
+                return YT_ERR_HIP;
+            }
+            S += cum;
+            T = (int64_t)h_bins[t];
+            int dw = ps - shift;
+            prefix = (ps == 64) ? (uint64_t)t : ((prefix << dw) | (uint64_t)t);
+            ps = shift;
+            if (T <= CAPB || shift == 0) break;
+            shift = shift >= 11 ? shift - 11 : 0;
+            tp.prefix_shift = ps;
+            tp.prefix = prefix;
+            tp.shift = shift;
+            tp.level0 = 0;
+        }
+
+        if (!all_nonnull && !only_nulls && T > CAPB && plan->order_count > 1) {
+            g_pool.put(d_bins);
+            set_err(errbuf, errlen,
+                    "order-key tie set too large for multi-key ORDER BY this round");
+            return YT_ERR_UNSUPPORTED;
+        }
+        const int64_t CAPN = K + CAPB;
+        {
+            const bool nulls_first = !desc0;
+            int64_t null_take = nulls_first
+                ? ((int64_t)null_cnt < K ? (int64_t)null_cnt : K)
+                : (K - nonnull > 0 ? K - nonnull : 0);
+            if ((int64_t)null_cnt > CAPN && plan->order_count > 1 &&
+                null_take > 0 && null_take < (int64_t)null_cnt) {
+                g_pool.put(d_bins);
+                set_err(errbuf, errlen,
+                        "null order-key set too large for multi-key ORDER BY this round");
+                return YT_ERR_UNSUPPORTED;
+            }
+        }
+
+        /* gather candidates: strict (exactly S), ties (cap), nulls (cap) */
+        TopkGather tg;
+        tg.order_proj = ord0;
+        tg.desc = desc0;
+        tg.all_nonnull = all_nonnull ? 1 : 0;
+        tg.prefix_shift = only_nulls ? 64 : ps;
+        tg.prefix = prefix;
+        tg.cap_tie = CAPB;
+        tg.cap_null = CAPN;
+        int64_t* d_rows_strict = nullptr;
+        int64_t* d_rows_tie = nullptr;
+        int64_t* d_rows_null = nullptr;
+        HIP_CHECK(pool_alloc(&d_rows_strict, sizeof(int64_t) * (S ? S : 1)));
+        HIP_CHECK(pool_alloc(&d_rows_tie, sizeof(int64_t) * CAPB));
+        HIP_CHECK(pool_alloc(&d_rows_null, sizeof(int64_t) * CAPN));
+        HIP_CHECK(hipMemsetAsync(d_ctrs, 0, 3 * sizeof(unsigned long long),
+                                 R2.stream));
+        HIP_CHECK(ytql_launch_topk_gather(dp, R2.d_segs, R2.d_segex, R2.d_off,
+                                          R2.d_cnt, n, &tg,
+                                          d_rows_strict, d_ctrs,
+                                          d_rows_tie, d_ctrs + 1,
+                                          d_rows_null, d_ctrs + 2,
+                                          R2.d_err, R2.stream));
+        launches++;
+        unsigned long long hctrs[3];
+        HIP_CHECK(hipMemcpy(hctrs, d_ctrs, 3 * sizeof(unsigned long long),
+                            hipMemcpyDeviceToHost));
+        int64_t nS = (int64_t)hctrs[0];
+        int64_t nB = (int64_t)hctrs[1] < CAPB ? (int64_t)hctrs[1] : CAPB;
+        int64_t nN = (int64_t)hctrs[2] < CAPN ? (int64_t)hctrs[2] : CAPN;
+        int64_t M = nS + nB + nN;
+
+        /* pack the three lists and materialize the projected rows */
+        int64_t* d_rows_all = nullptr;
+        DevOutVal* d_vals = nullptr;
+        DevOutVal* h_vals = nullptr;
+        HIP_CHECK(pool_alloc(&d_rows_all, sizeof(int64_t) * (M ? M : 1)));
+        if (nS) HIP_CHECK(hipMemcpyAsync(d_rows_all, d_rows_strict,
+                                         sizeof(int64_t) * nS,
+                                         hipMemcpyDeviceToDevice, R2.stream));
+        if (nB) HIP_CHECK(hipMemcpyAsync(d_rows_all + nS, d_rows_tie,
+                                         sizeof(int64_t) * nB,
+                                         hipMemcpyDeviceToDevice, R2.stream));
+        if (nN) HIP_CHECK(hipMemcpyAsync(d_rows_all + nS + nB, d_rows_null,
+                                         sizeof(int64_t) * nN,
+                                         hipMemcpyDeviceToDevice, R2.stream));
+        HIP_CHECK(pool_alloc(&d_vals, sizeof(DevOutVal) * (M ? M : 1) * np));
+        HIP_CHECK(pool_alloc_host(&h_vals, sizeof(DevOutVal) * (M ? M : 1) * np));
+        if (M) {
+            HIP_CHECK(ytql_launch_topk_materialize(dp, R2.d_segs, R2.d_segex,
+                                                   R2.d_off, R2.d_cnt,
+                                                   d_rows_all, M, d_vals,
+                                                   R2.d_err, R2.stream));
+            launches++;
+            HIP_CHECK(hipMemcpyAsync(h_vals, d_vals, sizeof(DevOutVal) * M * np,
+                                     hipMemcpyDeviceToHost, R2.stream));
+        }
+        HIP_CHECK(hipEventRecord(e1, R2.stream));
+        HIP_CHECK(hipStreamSynchronize(R2.stream));
+        float ms = 0;
+        HIP_CHECK(hipEventElapsedTime(&ms, e0, e1));
+        hipEventDestroy(e0);
+        hipEventDestroy(e1);
+        unsigned kerr = 0;
+        HIP_CHECK(hipMemcpy(&kerr, R2.d_err, sizeof(unsigned),
+                            hipMemcpyDeviceToHost));
+        auto put_all = [&]() {
+            g_pool.put(d_bins); g_pool.put(d_rows_strict); g_pool.put(d_rows_tie);
+            g_pool.put(d_rows_null); g_pool.put(d_rows_all); g_pool.put(d_vals);
+            g_pool.put(h_vals);
+        };
+        if (kerr) {
+            put_all();
+            if (kerr == 100) {
+                set_err(errbuf, errlen, "NaN in ORDER BY comparison");
+                return YT_ERR_LIMIT;
+            }
+            set_err(errbuf, errlen,
+                    kerr == YT_ERR_DIV_ZERO ? "Division by zero" : "expression error");
+            return (int)kerr;
+        }
+
+        /* host: full-comparer sort of the candidates, emit the slice */
+        std::vector<int64_t> idx((size_t)M);
+        for (int64_t i = 0; i < M; i++) idx[i] = i;
+        int nan_err = 0;
+        auto cmp_cand = [&](int64_t ia, int64_t ib) {
+            const DevOutVal* a = h_vals + ia * np;
+            const DevOutVal* b = h_vals + ib * np;
+            for (int i = 0; i < plan->order_count; i++) {
+                int c = plan->order_cols[i];
+                const DevOutVal& x = a[c];
+                const DevOutVal& y = b[c];
+                int xn = x.type == YT_VT_NULL, yn = y.type == YT_VT_NULL;
+                int r = 0;
+                if (xn || yn) {
+                    r = (xn == yn) ? 0 : (xn ? -1 : 1);
+                } else if (x.type == YT_VT_DOUBLE) {
+                    double xv, yv;
+                    memcpy(&xv, &x.bits, 8);
+                    memcpy(&yv, &y.bits, 8);
+                    if (xv != xv || yv != yv) { nan_err = 1; return false; }
+                    r = xv < yv ? -1 : (xv > yv ? 1 : 0);
+                } else if (x.type == YT_VT_INT64) {
+                    int64_t xv = (int64_t)x.bits, yv = (int64_t)y.bits;
+                    r = xv < yv ? -1 : (xv > yv ? 1 : 0);
+                } else {
+                    r = x.bits < y.bits ? -1 : (x.bits > y.bits ? 1 : 0);
+                }
+                if (plan->order_desc && plan->order_desc[i]) r = -r;
+                if (r) return r < 0;
+            }
+            return false;
+        };
+        std::sort(idx.begin(), idx.end(), cmp_cand);
+        if (nan_err) {
+            put_all();
+            set_err(errbuf, errlen, "NaN in ORDER BY comparison");
+            return YT_ERR_LIMIT;
+        }
+        int64_t b = plan->order_offset < M ? plan->order_offset : M;
+        int64_t e = b + plan->order_limit;
+        if (e > M) e = M;
+        int out_limited = 0;
+        for (int64_t i = b; i < e; i++) {
+            if (options->output_row_limit > 0 &&
+                output->row_count >= options->output_row_limit) {
+                out_limited = 1;
+                break;
+            }
+            if (output->row_count >= output->capacity_rows) {
+                put_all();
+                return YT_ERR_CAPACITY;
+            }
+            const DevOutVal* src = h_vals + idx[i] * np;
+            YtValue* dst = output->values + output->row_count * np;
+            for (int pj = 0; pj < np; pj++) {
+                dst[pj].id = (uint16_t)pj;
+                dst[pj].type = (uint8_t)src[pj].type;
+                dst[pj].flags = 0;
+                dst[pj].length = 0;
+                dst[pj].data.bits = src[pj].bits;
+            }
+            output->row_count++;
+        }
+        if (stats) {
+            stats->rows_read = n;
+            stats->rows_written = output->row_count;
+            stats->incomplete_output = out_limited;
+            stats->kernel_scan_ms += ms;
+            stats->kernel_scan_launches += launches;
+            stats->execute_time_ms = now_ms() - tw0;
+        }
+        put_all();
+    }
+    return YT_OK;
+fail:
+    return rc;
+}
+
 /* scan + filter + project: order-preserving (MakeCodegenProjectOp +
  * WriteOpHelper, registry.cpp:1999-2047). The device writes per-row values
  * and a pass mask; the host compacts in row order. */
@@ -1725,7 +2137,8 @@ extern "C" int yt_gpu_query_execute(
         set_err(errbuf, errlen, "empty plan");
         return YT_ERR_INVALID_PLAN;
     }
-    if (plan->agg_count == 0 && chunk->row_count > (int64_t)1 << 24) {
+    if (plan->agg_count == 0 && plan->order_count == 0 &&
+        chunk->row_count > (int64_t)1 << 24) {
         set_err(errbuf, errlen,
                 "GPU scan-project materializes per-row output; capped at 16M rows this round");
         return YT_ERR_UNSUPPORTED;
@@ -1736,8 +2149,11 @@ extern "C" int yt_gpu_query_execute(
         if (plan->key_count == 1 && plan->agg_count > 0 &&
             expr_is_col(plan->keys[0], &kc) && kc < chunk->column_count &&
             chunk->columns[kc].value_type == YT_VT_STRING) {
-            return run_string_group(plan, chunk, options, kc, output, stats,
-                                    tw0, errbuf, errlen);
+            rc = run_string_group(plan, chunk, options, kc, output, stats,
+                                  tw0, errbuf, errlen);
+            if (rc == YT_OK && plan->order_count > 0)
+                rc = apply_order_host(plan, output, errbuf, errlen);
+            return rc;
         }
     }
 
@@ -1746,6 +2162,11 @@ extern "C" int yt_gpu_query_execute(
     if (rc) return rc;
 
     if (plan->agg_count == 0) {
+        if (plan->order_count > 0) {
+            /* scan + ORDER BY ... LIMIT: k-selection, no full materialization */
+            return run_scan_topk(plan, chunk, options, &dp, output, stats, tw0,
+                                 errbuf, errlen);
+        }
         return run_scan_project(plan, chunk, options, &dp, output, stats, tw0,
                                 errbuf, errlen);
     }
@@ -1818,6 +2239,12 @@ extern "C" int yt_gpu_query_execute(
         g_pool.put(hgroups);
         if (rc) return rc;
         if (out_limited && stats) stats->incomplete_output = 1;
+    }
+    if (plan->order_count > 0) {
+        /* grouped output is already bounded: order on the host
+         * (combined group+order mode, registry.cpp:1677-1699) */
+        rc = apply_order_host(plan, output, errbuf, errlen);
+        if (rc) return rc;
     }
     if (stats) {
         stats->rows_read = (options->input_row_limit > 0 &&
@@ -2027,6 +2454,11 @@ extern "C" int yt_gpu_merge_states(
                        output, errbuf, errlen);
         if (rc) return rc;
         if (out_limited && stats) stats->incomplete_output = 1;
+        if (plan->order_count > 0) {
+            /* ORDER BY applies at the front (coordinator) query */
+            rc = apply_order_host(plan, output, errbuf, errlen);
+            if (rc) return rc;
+        }
         if (stats) {
             stats->rows_written = output->row_count;
             stats->grouped_row_count = ngroups + th.side_used[0] + th.side_used[1];

@@ -731,6 +731,163 @@ k_scan_project(DevPlan p, const DevSeg* segs, const SegEx* segex,
 }
 
 /* ------------------------------------------------------------------ */
+/* ORDER BY ... LIMIT: histogram k-selection over an order-isomorphic  */
+/* u64 key mapping (TTopCollector semantics; the host drives the digit */
+/* refinement and finishes with the exact comparer on ≤cap candidates) */
+
+/* map a value to u64 preserving the reference comparer order
+ * (cg_fragment_compiler.cpp:400-530): int64 signed, uint64/boolean
+ * unsigned, double by value (NaN -> error), null handled by caller */
+__device__ __forceinline__ bool topk_map(const DVal& v, int desc,
+                                         uint64_t* mk, unsigned* error_out)
+{
+    uint64_t m;
+    switch (v.type) {
+    case YT_VT_INT64:
+        m = v.bits ^ 0x8000000000000000ULL;
+        break;
+    case YT_VT_UINT64:
+    case YT_VT_BOOLEAN:
+        m = v.bits;
+        break;
+    case YT_VT_DOUBLE: {
+        double d = __longlong_as_double((long long)v.bits);
+        if (isnan(d)) { atomicMax(error_out, 100u); return false; }
+        m = (v.bits >> 63) ? ~v.bits : (v.bits | 0x8000000000000000ULL);
+        break;
+    }
+    default:
+        atomicMax(error_out, (unsigned)YT_ERR_UNSUPPORTED);
+        return false;
+    }
+    *mk = desc ? ~m : m;
+    return true;
+}
+
+__global__ void __launch_bounds__(256)
+k_topk_hist(DevPlan p, const DevSeg* segs, const SegEx* segex,
+            const int32_t* col_seg_off, const int32_t* col_seg_cnt,
+            int64_t row_count, TopkPass tp,
+            unsigned long long* bins,           /* 2048 */
+            unsigned long long* null_cnt,
+            unsigned* error_out)
+{
+    __shared__ unsigned lh[2048];
+    for (int i = threadIdx.x; i < 2048; i += blockDim.x) lh[i] = 0;
+    __syncthreads();
+
+    ColCtx c;
+    c.segs = segs;
+    c.segex = segex;
+    c.col_seg_off = col_seg_off;
+    c.col_seg_cnt = col_seg_cnt;
+    c.error = 0;
+    unsigned long long nulls = 0;
+
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         r < row_count; r += stride) {
+        c.row = r;
+        if (p.filter_len) {
+            DVal f = eval_prog(p, c, p.filter_off, p.filter_len);
+            if (f.null_ || f.bits == 0) continue;
+        }
+        DVal v = eval_prog(p, c, p.proj_off[tp.order_proj],
+                           p.proj_len[tp.order_proj]);
+        if (c.error) break;
+        if (v.null_) { nulls++; continue; }
+        uint64_t m;
+        if (!topk_map(v, tp.desc, &m, error_out)) continue;
+        if (tp.prefix_shift < 64 && (m >> tp.prefix_shift) != tp.prefix)
+            continue;
+        atomicAdd(&lh[(m >> tp.shift) & 2047], 1u);
+    }
+    if (c.error) atomicMax(error_out, c.error);
+    __syncthreads();
+    for (int i = threadIdx.x; i < 2048; i += blockDim.x) {
+        if (lh[i]) atomicAdd(&bins[i], (unsigned long long)lh[i]);
+    }
+    if (tp.level0 && nulls) atomicAdd(null_cnt, nulls);
+}
+
+__global__ void __launch_bounds__(256)
+k_topk_gather(DevPlan p, const DevSeg* segs, const SegEx* segex,
+              const int32_t* col_seg_off, const int32_t* col_seg_cnt,
+              int64_t row_count, TopkGather tg,
+              int64_t* rows_strict, unsigned long long* ctr_strict,
+              int64_t* rows_tie, unsigned long long* ctr_tie,
+              int64_t* rows_null, unsigned long long* ctr_null,
+              unsigned* error_out)
+{
+    ColCtx c;
+    c.segs = segs;
+    c.segex = segex;
+    c.col_seg_off = col_seg_off;
+    c.col_seg_cnt = col_seg_cnt;
+    c.error = 0;
+
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         r < row_count; r += stride) {
+        c.row = r;
+        if (p.filter_len) {
+            DVal f = eval_prog(p, c, p.filter_off, p.filter_len);
+            if (f.null_ || f.bits == 0) continue;
+        }
+        DVal v = eval_prog(p, c, p.proj_off[tg.order_proj],
+                           p.proj_len[tg.order_proj]);
+        if (c.error) break;
+        if (v.null_) {
+            unsigned long long j = atomicAdd(ctr_null, 1ULL);
+            if ((int64_t)j < tg.cap_null) rows_null[j] = r;
+            continue;
+        }
+        uint64_t m;
+        if (!topk_map(v, tg.desc, &m, error_out)) continue;
+        if (tg.all_nonnull ||
+            (tg.prefix_shift < 64 && (m >> tg.prefix_shift) < tg.prefix)) {
+            unsigned long long j = atomicAdd(ctr_strict, 1ULL);
+            rows_strict[j] = r;
+        } else if (tg.prefix_shift < 64 && (m >> tg.prefix_shift) == tg.prefix) {
+            unsigned long long j = atomicAdd(ctr_tie, 1ULL);
+            if ((int64_t)j < tg.cap_tie) rows_tie[j] = r;
+        }
+    }
+    if (c.error) atomicMax(error_out, c.error);
+}
+
+/* decode the full projected row for each selected chunk row */
+__global__ void __launch_bounds__(256)
+k_topk_materialize(DevPlan p, const DevSeg* segs, const SegEx* segex,
+                   const int32_t* col_seg_off, const int32_t* col_seg_cnt,
+                   const int64_t* rows, int64_t m, DevOutVal* out,
+                   unsigned* error_out)
+{
+    ColCtx c;
+    c.segs = segs;
+    c.segex = segex;
+    c.col_seg_off = col_seg_off;
+    c.col_seg_cnt = col_seg_cnt;
+    c.error = 0;
+
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         i < m; i += stride) {
+        c.row = rows[i];
+        for (int pj = 0; pj < p.proj_count; pj++) {
+            DVal v = eval_prog(p, c, p.proj_off[pj], p.proj_len[pj]);
+            DevOutVal o;
+            o.bits = v.bits;
+            o.type = v.null_ ? YT_VT_NULL : v.type;
+            o.pad_ = 0;
+            out[i * p.proj_count + pj] = o;
+        }
+        if (c.error) break;
+    }
+    if (c.error) atomicMax(error_out, c.error);
+}
+
+/* ------------------------------------------------------------------ */
 /* fast fused kernel — DirectDense segments, direct-column shapes      */
 /*                                                                     */
 /* One workgroup (256 threads) per tile of `tile_rows` rows inside one */
@@ -1746,6 +1903,61 @@ hipError_t ytql_launch_bucket_agg(const void* recs, const unsigned long long* cu
                        nrecs, ncursors, nbucket_stride,
                        out, out_counter, out_cap, th, sum_slot, agg_count,
                        packed_mode, bits_k, gmin_k, gmin_v);
+    return hipGetLastError();
+}
+
+hipError_t ytql_launch_topk_hist(const DevPlan* p, const DevSeg* segs,
+                                 const SegEx* segex,
+                                 const int32_t* col_seg_off,
+                                 const int32_t* col_seg_cnt,
+                                 int64_t row_count, const TopkPass* tp,
+                                 unsigned long long* bins,
+                                 unsigned long long* null_cnt,
+                                 unsigned* error_out, hipStream_t st)
+{
+    int block = 256;
+    int64_t want = (row_count + block - 1) / block;
+    int grid = (int)(want > 4096 ? 4096 : (want > 0 ? want : 1));
+    hipLaunchKernelGGL(k_topk_hist, dim3(grid), dim3(block), 0, st,
+                       *p, segs, segex, col_seg_off, col_seg_cnt, row_count,
+                       *tp, bins, null_cnt, error_out);
+    return hipGetLastError();
+}
+
+hipError_t ytql_launch_topk_gather(const DevPlan* p, const DevSeg* segs,
+                                   const SegEx* segex,
+                                   const int32_t* col_seg_off,
+                                   const int32_t* col_seg_cnt,
+                                   int64_t row_count, const TopkGather* tg,
+                                   int64_t* rows_strict, unsigned long long* ctr_strict,
+                                   int64_t* rows_tie, unsigned long long* ctr_tie,
+                                   int64_t* rows_null, unsigned long long* ctr_null,
+                                   unsigned* error_out, hipStream_t st)
+{
+    int block = 256;
+    int64_t want = (row_count + block - 1) / block;
+    int grid = (int)(want > 4096 ? 4096 : (want > 0 ? want : 1));
+    hipLaunchKernelGGL(k_topk_gather, dim3(grid), dim3(block), 0, st,
+                       *p, segs, segex, col_seg_off, col_seg_cnt, row_count,
+                       *tg, rows_strict, ctr_strict, rows_tie, ctr_tie,
+                       rows_null, ctr_null, error_out);
+    return hipGetLastError();
+}
+
+hipError_t ytql_launch_topk_materialize(const DevPlan* p, const DevSeg* segs,
+                                        const SegEx* segex,
+                                        const int32_t* col_seg_off,
+                                        const int32_t* col_seg_cnt,
+                                        const int64_t* rows, int64_t m,
+                                        DevOutVal* out, unsigned* error_out,
+                                        hipStream_t st)
+{
+    int block = 256;
+    int64_t want = (m + block - 1) / block;
+    int grid = (int)(want > 4096 ? 4096 : (want > 0 ? want : 1));
+    hipLaunchKernelGGL(k_topk_materialize, dim3(grid), dim3(block), 0, st,
+                       *p, segs, segex, col_seg_off, col_seg_cnt, rows, m,
+                       out, error_out);
     return hipGetLastError();
 }
 
