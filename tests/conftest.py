@@ -33,3 +33,10 @@ def _built():
     from ytsaurus_amd.build import build_all
     build_all()
     yield
+
+
+@pytest.fixture(scope="session")
+def cuda():
+    torch = pytest.importorskip("torch")
+    assert torch.cuda.is_available()
+    return torch

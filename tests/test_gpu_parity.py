@@ -16,12 +16,6 @@ pytestmark = pytest.mark.gpu
 torch = pytest.importorskip("torch")
 
 
-@pytest.fixture(scope="module")
-def cuda():
-    assert torch.cuda.is_available()
-    return torch
-
-
 def run_both(plan, chunk, cuda, hint=1 << 16):
     dev = chunk.c_device(cuda)
     got, stats = y.gpu_execute(plan, dev, max_groups_hint=hint)
