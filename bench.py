@@ -201,7 +201,9 @@ def main():
                     help="rows per GPU (weak scaling)")
     ap.add_argument("--keys", type=int, default=KEY_SPACE)
     ap.add_argument("--workload", default="groupby",
-                    choices=["groupby", "scanfilter", "strgroup"])
+                    choices=["groupby", "scanfilter", "strgroup", "topk"])
+    ap.add_argument("--limit", type=int, default=1000,
+                    help="topk workload: ORDER BY ... LIMIT n")
     ap.add_argument("--no-cpu-baseline", action="store_true")
     ap.add_argument("--traffic-bytes", type=float, default=0.0,
                     help="measured per-launch HBM bytes from a rocprofv3 --pmc run")
@@ -231,12 +233,19 @@ def main():
         if args.workload in ("groupby", "strgroup"):
             return y.Plan(keys=[y.col(0)],
                           aggs=[y.agg_sum(y.col(1)), y.agg_sum1()])
+        if args.workload == "topk":
+            # SELECT k, v ORDER BY k LIMIT n (TopCollector / OrderOpHelper
+            # shape): histogram k-selection, no scan materialization
+            return y.Plan(projects=[y.col(0), y.col(1)],
+                          order_by=[(0, False)], limit=args.limit)
         lo, hi = int(0.25 * 2**VAL_BITS), int(0.75 * 2**VAL_BITS)
         return y.Plan(filter=(y.col(0) >= lo).and_(y.col(0) <= hi),
                       aggs=[y.agg_sum(y.col(1)), y.agg_sum(y.col(2)),
                             y.agg_sum(y.col(3)), y.agg_sum1()])
 
     ncols = 4 if args.workload == "scanfilter" else 2
+    if args.workload == "topk":
+        assert world == 1, "topk: single-GPU this round (no top-K exchange yet)
     cores = os.cpu_count() or 8
 
     if args.workload == "strgroup":
@@ -272,8 +281,9 @@ def main():
     plan = make_plan()
     hint = key_space if args.workload == "groupby" else 0
     pool_b = (key_space * 10 + (1 << 20)) if args.workload == "strgroup" else 0
-    out_rs = y.make_rowset(key_space + 4096, 1 + len(plan.aggs),
-                           pool_bytes=pool_b)
+    out_cap = (args.limit + 64) if args.workload == "topk" else key_space + 4096
+    out_ncols = 2 if args.workload == "topk" else 1 + len(plan.aggs)
+    out_rs = y.make_rowset(out_cap, out_ncols, pool_bytes=pool_b)
 
     # multi-GPU state buffers
     if dist is not None:
@@ -368,10 +378,16 @@ def main():
     if rank == 0 and world == 1 and not args.no_cpu_baseline:
         bl_pool = 0
         bl_cap = KEY_SPACE + 1024
+        bl_cores = cores
         if args.workload == "strgroup":
             bl_cap = key_space + 4096
             bl_pool = key_space * 10 + (1 << 20)
-        cpu_baseline = cpu_baseline_leg(make_plan, enc_cols, n, cores,
+        if args.workload == "topk":
+            # the oracle scan+order path materializes all rows before the
+            # sort, and runs single-threaded
+            bl_cap = 8 * SLICE + 1024
+            bl_cores = 1
+        cpu_baseline = cpu_baseline_leg(make_plan, enc_cols, n, bl_cores,
                                         bl_cap, bl_pool)
 
     if rank == 0:
@@ -394,6 +410,7 @@ def main():
                     "scanfilter": "scan_filter_sum_4col@%drows" % n,
                     "strgroup": "strgroup_dict_%ddistinct@%drows"
                                 % (key_space, n),
+                    "topk": "orderby_limit%d@%drows" % (args.limit, n),
                 }[args.workload],
                 "rows_per_gpu": n,
                 "distinct_keys": key_space,
