@@ -245,7 +245,7 @@ def main():
 
     ncols = 4 if args.workload == "scanfilter" else 2
     if args.workload == "topk":
-        assert world == 1, "topk: single-GPU this round (no top-K exchange yet)
+        assert world == 1, "topk: single-GPU this round (no top-K exchange yet)"
     cores = os.cpu_count() or 8
 
     if args.workload == "strgroup":
