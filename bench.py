@@ -352,9 +352,9 @@ def main():
 
     st = getattr(step, "last", None)
     if st is not None:
-        log("last step: exec %.1fms scanK %.2fms otherK %.2fms rows_written %d"
+        log("last step: exec %.1fms scanK %.2fms otherK %.2fms setup %.2fms rows_written %d"
             % (st.execute_time_ms, st.kernel_scan_ms / max(st.kernel_scan_launches, 1),
-               st.kernel_other_ms, st.rows_written))
+               st.kernel_other_ms, st.decode_time_ms, st.rows_written))
 
     ms_per_step = elapsed / args.steps * 1000
     total_rows = n * world

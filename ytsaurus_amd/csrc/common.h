@@ -64,6 +64,10 @@ constexpr int kMaxProj = 8;
 struct DevPlan {
     int32_t ncols;
     uint8_t col_types[kMaxCols];       /* YT_VT_* */
+    /* log2(rows per segment) when the column's interior segments are all
+     * exactly 1<<shift rows (the writer's 128Ki cap makes this the norm):
+     * row→segment is then a shift instead of a binary search. 0 = ragged. */
+    int32_t col_uniform_shift[kMaxCols];
     int32_t filter_off, filter_len;    /* -1 len 0 = none */
     int32_t key_off, key_len;          /* key_count==1 only; len 0 = global agg */
     int32_t agg_count;

@@ -274,6 +274,23 @@ static int build_devplan(const YtPlan* plan, const YtChunk* chunk, DevPlan* p,
     p->ncols = chunk->column_count;
     if (p->ncols > kMaxCols) { set_err(errbuf, errlen, "too many columns"); return YT_ERR_UNSUPPORTED; }
     for (int c = 0; c < p->ncols; c++) p->col_types[c] = (uint8_t)chunk->columns[c].value_type;
+    for (int c = 0; c < p->ncols; c++) {
+        const YtColumn& col = chunk->columns[c];
+        int shift = 0;
+        if (col.segment_count > 0) {
+            int32_t r0 = col.segments[0].row_count;
+            if (r0 > 0 && (r0 & (r0 - 1)) == 0) {
+                int ok = 1;
+                for (int i = 0; i + 1 < col.segment_count; i++)
+                    if (col.segments[i].row_count != r0) { ok = 0; break; }
+                if (ok && col.segments[col.segment_count - 1].row_count <= r0) {
+                    while ((1 << (shift + 1)) <= r0) shift++;
+                    if ((1 << shift) != r0) shift = 0;
+                }
+            }
+        }
+        p->col_uniform_shift[c] = shift;
+    }
 
     if (plan->key_count > 1) { set_err(errbuf, errlen, "GPU path: at most 1 group key this round"); return YT_ERR_UNSUPPORTED; }
     if (plan->agg_count == 0) {
@@ -1342,6 +1359,7 @@ static int run_scan_topk(const YtPlan* plan, const YtChunk* chunk,
     rc = setup_chunk(chunk, &R2, &mw, options->input_row_limit, &in_clamped,
                      errbuf, errlen);
     if (rc) return rc;
+    if (stats) stats->decode_time_ms = now_ms() - tw0;   /* setup phase */
     int64_t n = chunk->row_count;
     if (options->input_row_limit > 0 && options->input_row_limit < n)
         n = options->input_row_limit;
@@ -1524,6 +1542,7 @@ static int run_scan_topk(const YtPlan* plan, const YtChunk* chunk,
         }
         HIP_CHECK(hipEventRecord(e1, R2.stream));
         HIP_CHECK(hipStreamSynchronize(R2.stream));
+        if (stats) stats->kernel_other_ms = now_ms() - tw0;  /* pre-emit wall */
         float ms = 0;
         HIP_CHECK(hipEventElapsedTime(&ms, e0, e1));
         hipEventDestroy(e0);

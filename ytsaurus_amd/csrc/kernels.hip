@@ -412,12 +412,18 @@ __device__ DVal col_value(const DevPlan& p, ColCtx& c, int col)
 {
     int off = c.col_seg_off[col];
     int cnt = c.col_seg_cnt[col];
-    /* binary search segment by start_row */
-    int lo = 0, hi = cnt;
-    while (lo + 1 < hi) {
-        int mid = (lo + hi) / 2;
-        if (c.segs[off + mid].start_row <= c.row) lo = mid;
-        else hi = mid;
+    int lo;
+    if (p.col_uniform_shift[col]) {
+        lo = (int)(c.row >> p.col_uniform_shift[col]);
+    } else {
+        /* binary search segment by start_row */
+        int hi = cnt;
+        lo = 0;
+        while (lo + 1 < hi) {
+            int mid = (lo + hi) / 2;
+            if (c.segs[off + mid].start_row <= c.row) lo = mid;
+            else hi = mid;
+        }
     }
     const DevSeg& s = c.segs[off + lo];
     return seg_value_at(s, c.segex[off + lo], c.row - s.start_row, p.col_types[col]);
