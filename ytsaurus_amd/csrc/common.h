@@ -220,23 +220,24 @@ struct StrSlot {
     uint64_t nonnull;
 };
 
-/* ORDER BY ... LIMIT k-selection (see kernels.hip k_topk_*) */
+/* ORDER BY ... LIMIT k-selection (see kernels.hip k_topk_*).
+ * Range-adaptive digits: candidates are mapped keys in [lo, hi]; digit =
+ * (m - lo) >> shift (< 2048 by host choice of shift). Pass 0 also reduces
+ * the data min/max so the next pass bins the ACTUAL key range instead of
+ * stepping fixed 11-bit prefixes through constant high bits. */
 struct TopkPass {
     int32_t order_proj;       /* projection index whose value is the order key */
     int32_t desc;             /* invert mapped key */
-    int32_t level0;           /* count nulls/passes on this pass */
-    int32_t prefix_shift;     /* 64 = no prefix constraint */
-    uint64_t prefix;          /* candidate iff mkey>>prefix_shift == prefix */
-    int32_t shift;            /* digit = (mkey>>shift) & 2047 */
-    int32_t pad_;
+    int32_t level0;           /* count nulls + reduce min/max on this pass */
+    int32_t shift;
+    uint64_t lo, hi;          /* inclusive candidate interval */
 };
 struct TopkGather {
     int32_t order_proj;
     int32_t desc;
     int32_t all_nonnull;      /* 1 = every non-null row is a strict hit */
-    int32_t prefix_shift;
-    uint64_t prefix;          /* strict: mkey>>prefix_shift < prefix;
-                                 tie: mkey>>prefix_shift == prefix */
+    int32_t pad_;
+    uint64_t lo, hi;          /* strict: m < lo; tie: lo <= m <= hi */
     int64_t cap_tie;
     int64_t cap_null;
 };

@@ -1373,10 +1373,10 @@ static int run_scan_topk(const YtPlan* plan, const YtChunk* chunk,
     const int64_t CAPB = (int64_t)1 << 17;
 
     {
-        unsigned long long* d_bins = nullptr;   /* 2048 bins + null ctr + 3 ctrs */
-        HIP_CHECK(pool_alloc(&d_bins, sizeof(unsigned long long) * (2048 + 4)));
-        unsigned long long* d_nullc = d_bins + 2048;
-        unsigned long long* d_ctrs = d_bins + 2049;   /* strict, tie, null */
+        unsigned long long* d_bins = nullptr;   /* 2048 bins + misc(3) + ctrs(3) */
+        HIP_CHECK(pool_alloc(&d_bins, sizeof(unsigned long long) * (2048 + 8)));
+        unsigned long long* d_misc = d_bins + 2048;   /* null_cnt, mmin, mmax */
+        unsigned long long* d_ctrs = d_bins + 2051;   /* strict, tie, null */
         HIP_CHECK(hipMemsetAsync(R2.d_err, 0, sizeof(unsigned), R2.stream));
 
         hipEvent_t e0, e1;
@@ -1388,28 +1388,34 @@ static int run_scan_topk(const YtPlan* plan, const YtChunk* chunk,
         tp.order_proj = ord0;
         tp.desc = desc0;
         tp.level0 = 1;
-        tp.prefix_shift = 64;
-        tp.prefix = 0;
         tp.shift = 53;
-        tp.pad_ = 0;
+        tp.lo = 0;
+        tp.hi = ~0ULL;
 
-        uint64_t prefix = 0;
-        int ps = 64, shift = 53;
+        uint64_t lo = 0, hi = ~0ULL;
+        int shift = 53;
         int64_t S = 0, T = 0, nonnull = 0, k_nonnull = 0;
         unsigned long long null_cnt = 0;
         bool all_nonnull = false, only_nulls = false;
-        std::vector<unsigned long long> h_bins(2048);
+        std::vector<unsigned long long> h_bins(2051);
         int launches = 0;
 
         for (;;) {
             HIP_CHECK(hipMemsetAsync(d_bins, 0,
                                      sizeof(unsigned long long) * 2049, R2.stream));
+            if (tp.level0) {
+                /* mmin = ~0, mmax = 0 */
+                HIP_CHECK(hipMemsetAsync(d_misc + 1, 0xFF,
+                                         sizeof(unsigned long long), R2.stream));
+                HIP_CHECK(hipMemsetAsync(d_misc + 2, 0,
+                                         sizeof(unsigned long long), R2.stream));
+            }
             HIP_CHECK(ytql_launch_topk_hist(dp, R2.d_segs, R2.d_segex, R2.d_off,
-                                            R2.d_cnt, n, &tp, d_bins, d_nullc,
+                                            R2.d_cnt, n, &tp, d_bins, d_misc,
                                             R2.d_err, R2.stream));
             launches++;
             HIP_CHECK(hipMemcpy(h_bins.data(), d_bins,
-                                sizeof(unsigned long long) * 2048,
+                                sizeof(unsigned long long) * 2051,
                                 hipMemcpyDeviceToHost));
             unsigned kerr = 0;
             HIP_CHECK(hipMemcpy(&kerr, R2.d_err, sizeof(unsigned),
@@ -1423,10 +1429,9 @@ static int run_scan_topk(const YtPlan* plan, const YtChunk* chunk,
                 set_err(errbuf, errlen, "unsupported ORDER BY key type");
                 return (int)kerr;
             }
+            uint64_t mmin = h_bins[2049], mmax = h_bins[2050];
             if (tp.level0) {
-                HIP_CHECK(hipMemcpy(&null_cnt, d_nullc,
-                                    sizeof(unsigned long long),
-                                    hipMemcpyDeviceToHost));
+                null_cnt = h_bins[2048];
                 nonnull = 0;
                 for (int i = 0; i < 2048; i++) nonnull += (int64_t)h_bins[i];
                 const bool nulls_first = !desc0;
@@ -1451,13 +1456,24 @@ static int run_scan_topk(const YtPlan* plan, const YtChunk* chunk,
             }
             S += cum;
             T = (int64_t)h_bins[t];
-            int dw = ps - shift;
-            prefix = (ps == 64) ? (uint64_t)t : ((prefix << dw) | (uint64_t)t);
-            ps = shift;
-            if (T <= CAPB || shift == 0) break;
-            shift = shift >= 11 ? shift - 11 : 0;
-            tp.prefix_shift = ps;
-            tp.prefix = prefix;
+            /* shrink [lo, hi] to the chosen bin (∩ data bounds on pass 0) */
+            uint64_t nlo = lo + ((uint64_t)t << shift);
+            uint64_t span = (uint64_t)(t + 1) << shift;
+            uint64_t nhi = (span == 0 || nlo + (span - ((uint64_t)t << shift)) - 1 < nlo)
+                ? hi : nlo + (((uint64_t)1 << shift) - 1);
+            if (nhi > hi) nhi = hi;
+            if (tp.level0) {
+                if (mmin > nlo) nlo = mmin;
+                if (mmax < nhi) nhi = mmax;
+            }
+            lo = nlo;
+            hi = nhi;
+            if (T <= CAPB || shift == 0 || lo >= hi) break;
+            uint64_t range = hi - lo;     /* >= 1 */
+            shift = 0;
+            while ((range >> shift) >= 2048) shift++;
+            tp.lo = lo;
+            tp.hi = hi;
             tp.shift = shift;
             tp.level0 = 0;
         }
@@ -1488,8 +1504,10 @@ static int run_scan_topk(const YtPlan* plan, const YtChunk* chunk,
         tg.order_proj = ord0;
         tg.desc = desc0;
         tg.all_nonnull = all_nonnull ? 1 : 0;
-        tg.prefix_shift = only_nulls ? 64 : ps;
-        tg.prefix = prefix;
+        tg.pad_ = 0;
+        tg.lo = only_nulls ? 0 : lo;       /* only_nulls: nothing strict */
+        tg.hi = only_nulls ? 0 : hi;
+        if (only_nulls) { tg.lo = 1; tg.hi = 0; }  /* empty tie interval */
         tg.cap_tie = CAPB;
         tg.cap_null = CAPN;
         int64_t* d_rows_strict = nullptr;

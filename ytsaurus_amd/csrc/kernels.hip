@@ -775,7 +775,7 @@ k_topk_hist(DevPlan p, const DevSeg* segs, const SegEx* segex,
             const int32_t* col_seg_off, const int32_t* col_seg_cnt,
             int64_t row_count, TopkPass tp,
             unsigned long long* bins,           /* 2048 */
-            unsigned long long* null_cnt,
+            unsigned long long* misc,           /* null_cnt, mmin, mmax */
             unsigned* error_out)
 {
     __shared__ unsigned lh[2048];
@@ -789,6 +789,7 @@ k_topk_hist(DevPlan p, const DevSeg* segs, const SegEx* segex,
     c.col_seg_cnt = col_seg_cnt;
     c.error = 0;
     unsigned long long nulls = 0;
+    uint64_t mmin = ~0ULL, mmax = 0;
 
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -804,16 +805,23 @@ k_topk_hist(DevPlan p, const DevSeg* segs, const SegEx* segex,
         if (v.null_) { nulls++; continue; }
         uint64_t m;
         if (!topk_map(v, tp.desc, &m, error_out)) continue;
-        if (tp.prefix_shift < 64 && (m >> tp.prefix_shift) != tp.prefix)
-            continue;
-        atomicAdd(&lh[(m >> tp.shift) & 2047], 1u);
+        if (tp.level0) {
+            if (m < mmin) mmin = m;
+            if (m > mmax) mmax = m;
+        }
+        if (m < tp.lo || m > tp.hi) continue;
+        atomicAdd(&lh[(m - tp.lo) >> tp.shift], 1u);
     }
     if (c.error) atomicMax(error_out, c.error);
     __syncthreads();
     for (int i = threadIdx.x; i < 2048; i += blockDim.x) {
         if (lh[i]) atomicAdd(&bins[i], (unsigned long long)lh[i]);
     }
-    if (tp.level0 && nulls) atomicAdd(null_cnt, nulls);
+    if (tp.level0) {
+        if (nulls) atomicAdd(&misc[0], nulls);
+        if (mmin != ~0ULL) atomicMin(&misc[1], mmin);
+        if (mmax || mmin != ~0ULL) atomicMax(&misc[2], mmax);
+    }
 }
 
 __global__ void __launch_bounds__(256)
@@ -850,11 +858,10 @@ k_topk_gather(DevPlan p, const DevSeg* segs, const SegEx* segex,
         }
         uint64_t m;
         if (!topk_map(v, tg.desc, &m, error_out)) continue;
-        if (tg.all_nonnull ||
-            (tg.prefix_shift < 64 && (m >> tg.prefix_shift) < tg.prefix)) {
+        if (tg.all_nonnull || m < tg.lo) {
             unsigned long long j = atomicAdd(ctr_strict, 1ULL);
             rows_strict[j] = r;
-        } else if (tg.prefix_shift < 64 && (m >> tg.prefix_shift) == tg.prefix) {
+        } else if (m <= tg.hi) {
             unsigned long long j = atomicAdd(ctr_tie, 1ULL);
             if ((int64_t)j < tg.cap_tie) rows_tie[j] = r;
         }
