@@ -268,3 +268,65 @@ def test_topk_gpu_nan_errors(cuda):
     plan = y.Plan(projects=[y.col(0)], order_by=[(0, False)], limit=10)
     with pytest.raises(RuntimeError, match="NaN"):
         y.gpu_execute(plan, chunk.c_device(cuda))
+
+
+# ---------------- goldens transcribed from the reference's own tests ----------------
+# TQueryEvaluateTest.GroupByOrderBy, ql_query_ut.cpp:2485-2622. Our seam
+# emits group rows as [keys..., aggs...] (the reference's projection wrapper
+# reorders them afterwards), so expected rows are stated in that layout.
+
+def _golden_chunk():
+    a = np.arange(1, 10, dtype=np.int64)             # a=1..9
+    b = np.array([0, 1, 2] * 3, dtype=np.int64)      # b cycles 0,1,2
+    d = 10 - a                                       # d=9..1
+    return a, b, d, y.Chunk([y.encode_int64(a), y.encode_int64(b),
+                             y.encode_int64(d)], 9)
+
+
+def test_golden_group_order_desc():
+    # "sum(a) as t, b ... group by b order by b desc limit 3"
+    # → t=18;b=2 / t=15;b=1 / t=12;b=0   (ql_query_ut.cpp:2514-2525)
+    _, _, _, chunk = _golden_chunk()
+    plan = y.Plan(keys=[y.col(1)], aggs=[y.agg_sum(y.col(0))],
+                  order_by=[(0, True)], limit=3)
+    rows, _ = y.oracle_execute(plan, chunk)
+    assert rows == [(2, 18), (1, 15), (0, 12)]
+
+
+def test_golden_group_order_desc_then_sum():
+    # "... order by b desc, sum(a) limit 3" → same rows
+    # (ql_query_ut.cpp:2540-2549)
+    _, _, _, chunk = _golden_chunk()
+    plan = y.Plan(keys=[y.col(1)], aggs=[y.agg_sum(y.col(0))],
+                  order_by=[(0, True), (1, False)], limit=3)
+    rows, _ = y.oracle_execute(plan, chunk)
+    assert rows == [(2, 18), (1, 15), (0, 12)]
+
+
+def test_golden_group_order_offset():
+    # "d, a, b ... group by d, a, b order by a, b offset 2 limit 3"
+    # → d=7;a=3;b=2 / d=6;a=4;b=0 / d=5;a=5;b=1 (ql_query_ut.cpp:2601-2611;
+    # multi-key GROUP BY → oracle only, the GPU path is 1-key this round)
+    _, _, _, chunk = _golden_chunk()
+    plan = y.Plan(keys=[y.col(2), y.col(0), y.col(1)],
+                  order_by=[(1, False), (2, False)], limit=3, offset=2)
+    rows, _ = y.oracle_execute(plan, chunk)
+    assert rows == [(7, 3, 2), (6, 4, 0), (5, 5, 1)]
+
+
+@pytest.mark.gpu
+def test_golden_group_order_desc_gpu(cuda):
+    _, _, _, chunk = _golden_chunk()
+    plan = y.Plan(keys=[y.col(1)], aggs=[y.agg_sum(y.col(0))],
+                  order_by=[(0, True)], limit=3)
+    rows, _ = y.gpu_execute(plan, chunk.c_device(cuda), max_groups_hint=64)
+    assert rows == [(2, 18), (1, 15), (0, 12)]
+
+
+@pytest.mark.gpu
+def test_golden_group_order_desc_then_sum_gpu(cuda):
+    _, _, _, chunk = _golden_chunk()
+    plan = y.Plan(keys=[y.col(1)], aggs=[y.agg_sum(y.col(0))],
+                  order_by=[(0, True), (1, False)], limit=3)
+    rows, _ = y.gpu_execute(plan, chunk.c_device(cuda), max_groups_hint=64)
+    assert rows == [(2, 18), (1, 15), (0, 12)]
