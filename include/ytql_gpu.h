@@ -160,6 +160,15 @@ typedef struct YtPlan {
     const int32_t* order_desc;    /* 0 = ascending, 1 = descending */
     int64_t order_limit;
     int64_t order_offset;
+    /* WITH TOTALS (TotalsMode::BeforeHaving; registry.cpp
+     * TGroupByClosure::InsertTotals/FlushTotals:1556-1650, test semantics
+     * ql_query_ut.cpp:3432-3510): one extra output row with null group keys
+     * whose aggregates cover ALL grouped rows (including the null-key
+     * group), computed before any ORDER BY/limit slicing. The totals row is
+     * appended last and flagged in YtRowset.totals_row — it mirrors the
+     * reference's separate EStreamTag::Totals stream. */
+    int32_t with_totals;
+    int32_t reserved_;
 } YtPlan;
 
 /* ---- execution context / statistics ----
@@ -200,6 +209,9 @@ typedef struct YtRowset {
     char* string_pool;            /* optional pool for string payloads */
     int64_t string_pool_capacity;
     int64_t string_pool_used;     /* out */
+    int32_t totals_row;           /* out: 1 = the LAST row is the WITH TOTALS
+                                     stream row (EStreamTag::Totals) */
+    int32_t pad_;
 } YtRowset;
 
 /* =========================== entry points =========================== */

@@ -927,6 +927,17 @@ int yto_execute(const YtPlan* plan, const YtChunk* chunk,
 
     output->row_count = 0;
     output->string_pool_used = 0;
+    output->totals_row = 0;
+    if (plan->with_totals && plan->key_count == 0) {
+        set_err(errbuf, errlen, "WITH TOTALS requires GROUP BY");
+        rc = YT_ERR_INVALID_PLAN;
+        goto done;
+    }
+    if (plan->with_totals && plan->project_count) {
+        set_err(errbuf, errlen, "WITH TOTALS with projections: not this round");
+        rc = YT_ERR_UNSUPPORTED;
+        goto done;
+    }
 
     if (plan->agg_count == 0 && plan->key_count == 0) {
         /* plain scan+filter+project */
@@ -1035,6 +1046,53 @@ int yto_execute(const YtPlan* plan, const YtChunk* chunk,
                 ? plan->project_count : plan->key_count + plan->agg_count;
             if (rc == YT_OK && plan->order_count > 0) {
                 rc = apply_order_rowset(plan, output, errbuf, errlen);
+            }
+            if (rc == YT_OK && plan->with_totals) {
+                /* totals stream: null keys + aggregates over ALL groups
+                 * (registry.cpp FlushTotals; ql_query_ut.cpp:3432-3476) */
+                int kc = plan->key_count, ac = plan->agg_count;
+                if (output->row_count >= output->capacity_rows) {
+                    rc = YT_ERR_CAPACITY;
+                } else {
+                    Val tst[16];
+                    uint64_t tcnt = 0;
+                    for (int a = 0; a < ac; a++) tst[a].type = YT_VT_NULL;
+                    for (int64_t g = 0; g < final_t->ngroups; g++) {
+                        tcnt += final_t->rowcounts[g];
+                        for (int a = 0; a < ac; a++) {
+                            Val st = final_t->states[g * ac + a];
+                            if (plan->aggs[a]->func == YT_AGG_SUM)
+                                sum_update_val(&tst[a], st);
+                            else if (plan->aggs[a]->func != YT_AGG_SUM1)
+                                minmax_update_val(&tst[a], st,
+                                    plan->aggs[a]->func == YT_AGG_MAX);
+                        }
+                    }
+                    YtValue* dst = output->values
+                        + output->row_count * (kc + ac);
+                    for (int k = 0; k < kc; k++) {
+                        dst[k].id = (uint16_t)k;
+                        dst[k].type = YT_VT_NULL;
+                        dst[k].flags = 0;
+                        dst[k].length = 0;
+                        dst[k].data.bits = 0;
+                    }
+                    for (int a = 0; a < ac; a++) {
+                        YtValue* v = &dst[kc + a];
+                        v->id = (uint16_t)(kc + a);
+                        v->flags = 0;
+                        v->length = 0;
+                        if (plan->aggs[a]->func == YT_AGG_SUM1) {
+                            v->type = YT_VT_INT64;
+                            v->data.bits = tcnt;
+                        } else {
+                            v->type = tst[a].type;
+                            v->data.bits = tst[a].bits;
+                        }
+                    }
+                    output->row_count++;
+                    output->totals_row = 1;
+                }
             }
             if (stats) {
                 stats->rows_read = n;

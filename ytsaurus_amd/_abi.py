@@ -103,7 +103,9 @@ class YtPlan(C.Structure):
                 ("order_cols", C.POINTER(C.c_int32)),
                 ("order_desc", C.POINTER(C.c_int32)),
                 ("order_limit", C.c_int64),
-                ("order_offset", C.c_int64)]
+                ("order_offset", C.c_int64),
+                ("with_totals", C.c_int32),
+                ("reserved_", C.c_int32)]
 
 
 class YtExecOptions(C.Structure):
@@ -125,7 +127,8 @@ class YtRowset(C.Structure):
     _fields_ = [("values", C.POINTER(YtValue)), ("capacity_rows", C.c_int64),
                 ("row_count", C.c_int64), ("column_count", C.c_int32),
                 ("string_pool", C.c_char_p), ("string_pool_capacity", C.c_int64),
-                ("string_pool_used", C.c_int64)]
+                ("string_pool_used", C.c_int64),
+                ("totals_row", C.c_int32), ("pad_", C.c_int32)]
 
 
 class YtStateRow(C.Structure):

@@ -107,7 +107,7 @@ class Plan:
     required with order_by (reference QL rejects ORDER BY without LIMIT)."""
 
     def __init__(self, filter=None, keys=(), aggs=(), projects=(), is_merge=False,
-                 order_by=(), limit=0, offset=0):
+                 order_by=(), limit=0, offset=0, with_totals=False):
         self.filter = filter
         self.keys = list(keys)
         self.aggs = list(aggs)
@@ -116,6 +116,7 @@ class Plan:
         self.order_by = [(c, 1 if d else 0) for (c, d) in order_by]
         self.limit = limit
         self.offset = offset
+        self.with_totals = with_totals
         self._build()
 
     def _build(self):
@@ -163,6 +164,7 @@ class Plan:
             self._keep += [oc, od]
         p.order_limit = self.limit
         p.order_offset = self.offset
+        p.with_totals = 1 if self.with_totals else 0
         self.c = p
 
 

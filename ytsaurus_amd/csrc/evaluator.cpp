@@ -1331,6 +1331,85 @@ static int apply_order_host(const YtPlan* plan, YtRowset* out,
     return YT_OK;
 }
 
+/* WITH TOTALS + ORDER BY finishing pass over a fully-materialized group
+ * rowset: totals (null keys + aggregates over ALL grouped rows — the
+ * reference's EStreamTag::Totals, registry.cpp FlushTotals) are computed
+ * BEFORE the order/limit slice and appended after it, flagged in
+ * YtRowset.totals_row. */
+static int finish_output(const YtPlan* plan, YtRowset* out,
+                         char* errbuf, size_t errlen)
+{
+    std::vector<YtValue> tot;
+    uint64_t tcnt_unused = 0;
+    (void)tcnt_unused;
+    if (plan->with_totals) {
+        int kc = plan->key_count, ac = plan->agg_count;
+        int ncols = out->column_count;
+        tot.resize(ac);
+        for (int a = 0; a < ac; a++) {
+            tot[a].id = (uint16_t)(kc + a);
+            tot[a].type = YT_VT_NULL;
+            tot[a].flags = 0;
+            tot[a].length = 0;
+            tot[a].data.bits = 0;
+        }
+        for (int64_t r = 0; r < out->row_count; r++) {
+            const YtValue* row = out->values + r * ncols;
+            for (int a = 0; a < ac; a++) {
+                const YtValue& v = row[kc + a];
+                if (v.type == YT_VT_NULL) continue;
+                YtValue& t = tot[a];
+                int f = plan->aggs[a]->func;
+                if (t.type == YT_VT_NULL) {
+                    t.type = v.type;
+                    t.data.bits = v.data.bits;
+                    continue;
+                }
+                if (f == YT_AGG_SUM || f == YT_AGG_SUM1) {
+                    if (v.type == YT_VT_DOUBLE) t.data.dbl += v.data.dbl;
+                    else t.data.bits += v.data.bits;   /* mod 2^64, udf/sum.c */
+                } else {
+                    bool take;
+                    if (v.type == YT_VT_DOUBLE)
+                        take = (f == YT_AGG_MAX) ? (v.data.dbl > t.data.dbl)
+                                                 : (v.data.dbl < t.data.dbl);
+                    else if (v.type == YT_VT_INT64)
+                        take = (f == YT_AGG_MAX) ? (v.data.i64 > t.data.i64)
+                                                 : (v.data.i64 < t.data.i64);
+                    else
+                        take = (f == YT_AGG_MAX) ? (v.data.u64 > t.data.u64)
+                                                 : (v.data.u64 < t.data.u64);
+                    if (take) t.data.bits = v.data.bits;
+                }
+            }
+        }
+    }
+    if (plan->order_count > 0) {
+        int rc = apply_order_host(plan, out, errbuf, errlen);
+        if (rc) return rc;
+    }
+    if (plan->with_totals) {
+        int kc = plan->key_count, ac = plan->agg_count;
+        int ncols = out->column_count;
+        if (out->row_count >= out->capacity_rows) {
+            set_err(errbuf, errlen, "rowset too small for the totals row");
+            return YT_ERR_CAPACITY;
+        }
+        YtValue* dst = out->values + out->row_count * ncols;
+        for (int k = 0; k < kc; k++) {
+            dst[k].id = (uint16_t)k;
+            dst[k].type = YT_VT_NULL;
+            dst[k].flags = 0;
+            dst[k].length = 0;
+            dst[k].data.bits = 0;
+        }
+        for (int a = 0; a < ac; a++) dst[kc + a] = tot[a];
+        out->row_count++;
+        out->totals_row = 1;
+    }
+    return YT_OK;
+}
+
 /* scan + ORDER BY ... LIMIT without materializing the scan: histogram
  * k-selection on an order-isomorphic u64 mapping of the first order key
  * (digits of 11 bits, host-driven refinement), then exact candidate gather
@@ -2174,6 +2253,22 @@ extern "C" int yt_gpu_query_execute(
         set_err(errbuf, errlen, "empty plan");
         return YT_ERR_INVALID_PLAN;
     }
+    output->totals_row = 0;
+    if (plan->with_totals) {
+        if (plan->key_count == 0) {
+            set_err(errbuf, errlen, "WITH TOTALS requires GROUP BY");
+            return YT_ERR_INVALID_PLAN;
+        }
+        if (plan->project_count) {
+            set_err(errbuf, errlen, "WITH TOTALS with projections: not this round");
+            return YT_ERR_UNSUPPORTED;
+        }
+        if (options->output_row_limit > 0) {
+            set_err(errbuf, errlen,
+                    "WITH TOTALS with OutputRowLimit: not this round");
+            return YT_ERR_UNSUPPORTED;
+        }
+    }
     if (plan->agg_count == 0 && plan->order_count == 0 &&
         chunk->row_count > (int64_t)1 << 24) {
         set_err(errbuf, errlen,
@@ -2188,8 +2283,8 @@ extern "C" int yt_gpu_query_execute(
             chunk->columns[kc].value_type == YT_VT_STRING) {
             rc = run_string_group(plan, chunk, options, kc, output, stats,
                                   tw0, errbuf, errlen);
-            if (rc == YT_OK && plan->order_count > 0)
-                rc = apply_order_host(plan, output, errbuf, errlen);
+            if (rc == YT_OK && (plan->order_count > 0 || plan->with_totals))
+                rc = finish_output(plan, output, errbuf, errlen);
             return rc;
         }
     }
@@ -2277,11 +2372,12 @@ extern "C" int yt_gpu_query_execute(
         if (rc) return rc;
         if (out_limited && stats) stats->incomplete_output = 1;
     }
-    if (plan->order_count > 0) {
-        /* grouped output is already bounded: order on the host
+    if (plan->order_count > 0 || plan->with_totals) {
+        /* grouped output is already bounded: order/totals on the host
          * (combined group+order mode, registry.cpp:1677-1699) */
-        rc = apply_order_host(plan, output, errbuf, errlen);
+        rc = finish_output(plan, output, errbuf, errlen);
         if (rc) return rc;
+        if (stats) stats->rows_written = output->row_count;
     }
     if (stats) {
         stats->rows_read = (options->input_row_limit > 0 &&
@@ -2433,6 +2529,7 @@ extern "C" int yt_gpu_merge_states(
     int rc = yt_gpu_available(errbuf, errlen);
     if (rc != YT_OK) return rc;
     if (plan->key_count != 1) { set_err(errbuf, errlen, "merge: need exactly 1 key"); return YT_ERR_UNSUPPORTED; }
+    output->totals_row = 0;
     int sum_slot = -1;
     for (int a = 0; a < plan->agg_count; a++) {
         if (plan->aggs[a]->func == YT_AGG_SUM) sum_slot = a;
@@ -2491,9 +2588,9 @@ extern "C" int yt_gpu_merge_states(
                        output, errbuf, errlen);
         if (rc) return rc;
         if (out_limited && stats) stats->incomplete_output = 1;
-        if (plan->order_count > 0) {
-            /* ORDER BY applies at the front (coordinator) query */
-            rc = apply_order_host(plan, output, errbuf, errlen);
+        if (plan->order_count > 0 || plan->with_totals) {
+            /* ORDER BY / WITH TOTALS apply at the front (coordinator) query */
+            rc = finish_output(plan, output, errbuf, errlen);
             if (rc) return rc;
         }
         if (stats) {
