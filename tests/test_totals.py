@@ -109,7 +109,7 @@ def test_totals_gpu_string_keys(cuda):
     keys = [keyset[int(i)] for i in rng.integers(0, 200, n)]
     vals = rng.random(n)
     chunk = y.Chunk([y.encode_string(keys, max_segment_values=8192),
-                     y.encode_double(vals)], n)
+                     y.encode_double(vals, max_segment_values=8192)], n)
     plan = y.Plan(keys=[y.col(0)], aggs=[y.agg_sum(y.col(1)), y.agg_sum1()],
                   with_totals=True)
     got, _ = y.gpu_execute(plan, chunk.c_device(cuda), max_groups_hint=1024)
