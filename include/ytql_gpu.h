@@ -168,7 +168,15 @@ typedef struct YtPlan {
      * appended last and flagged in YtRowset.totals_row — it mirrors the
      * reference's separate EStreamTag::Totals stream. */
     int32_t with_totals;
-    int32_t reserved_;
+    /* 1 = BeforeHaving (default), 2 = AfterHaving — syntactic position of
+     * WITH TOTALS relative to HAVING (parser.ypp:469-481): BeforeHaving
+     * totals cover all groups, AfterHaving totals only the groups that
+     * survive the having filter (folding_profiler.cpp:1810-1815). */
+    int32_t totals_mode;
+    /* HAVING clause (TQuery::HavingClause, base/query.h:499): a predicate
+     * over the group output row [keys..., aggs...], applied after
+     * aggregation and before ORDER BY/limit. */
+    const YtExpr* having;
 } YtPlan;
 
 /* ---- execution context / statistics ----

@@ -879,6 +879,136 @@ static int apply_order_rowset(const YtPlan* plan, YtRowset* out,
     return YT_OK;
 }
 
+/* fold the totals row from materialized [keys..., aggs...] rows */
+static void ord_fold_totals(const YtPlan* plan, const YtRowset* out,
+                            YtValue* tot)
+{
+    int kc = plan->key_count, ac = plan->agg_count;
+    int ncols = out->column_count;
+    for (int a = 0; a < ac; a++) {
+        tot[a].id = (uint16_t)(kc + a);
+        tot[a].type = YT_VT_NULL;
+        tot[a].flags = 0;
+        tot[a].length = 0;
+        tot[a].data.bits = 0;
+    }
+    for (int64_t r = 0; r < out->row_count; r++) {
+        const YtValue* row = out->values + r * ncols;
+        for (int a = 0; a < ac; a++) {
+            const YtValue* v = &row[kc + a];
+            if (v->type == YT_VT_NULL) continue;
+            YtValue* t = &tot[a];
+            int fagg = plan->aggs[a]->func;
+            if (t->type == YT_VT_NULL) { t->type = v->type; t->data.bits = v->data.bits; continue; }
+            if (fagg == YT_AGG_SUM || fagg == YT_AGG_SUM1) {
+                if (v->type == YT_VT_DOUBLE) t->data.dbl += v->data.dbl;
+                else t->data.bits += v->data.bits;   /* mod 2^64, udf/sum.c */
+            } else {
+                int take;
+                if (v->type == YT_VT_DOUBLE)
+                    take = fagg == YT_AGG_MAX ? v->data.dbl > t->data.dbl
+                                              : v->data.dbl < t->data.dbl;
+                else if (v->type == YT_VT_INT64)
+                    take = fagg == YT_AGG_MAX ? v->data.i64 > t->data.i64
+                                              : v->data.i64 < t->data.i64;
+                else
+                    take = fagg == YT_AGG_MAX ? v->data.u64 > t->data.u64
+                                              : v->data.u64 < t->data.u64;
+                if (take) t->data.bits = v->data.bits;
+            }
+        }
+    }
+}
+
+static void ord_expr_cols(const YtExpr* e, uint64_t* mask)
+{
+    if (!e) return;
+    if (e->op == YT_EX_COLUMN && e->col >= 0 && e->col < 64)
+        *mask |= 1ULL << e->col;
+    ord_expr_cols(e->a, mask);
+    ord_expr_cols(e->b, mask);
+}
+
+/* totals(Before) -> HAVING -> totals(After) -> ORDER/limit -> totals row
+ * (folding_profiler.cpp:1810-1815 Process(); parser.ypp:469-481 modes) */
+static int finish_group_rowset(const YtPlan* plan, YtRowset* out,
+                               char* errbuf, size_t errlen)
+{
+    int kc = plan->key_count, ac = plan->agg_count;
+    int ncols = out->column_count;
+    YtValue tot[16];
+    int after = plan->totals_mode == 2;
+    if (plan->with_totals && !after) ord_fold_totals(plan, out, tot);
+    if (plan->having) {
+        uint64_t mask = 0;
+        ord_expr_cols(plan->having, &mask);
+        int64_t w = 0;
+        for (int64_t r = 0; r < out->row_count; r++) {
+            YtValue* row = out->values + r * ncols;
+            int64_t pv[32];
+            uint8_t pn[32], pt[32];
+            const int64_t* pvp[32];
+            const uint8_t* pnp[32];
+            for (int i = 0; i < ncols && i < 32; i++) {
+                if ((mask >> i) & 1 && row[i].type == YT_VT_STRING) {
+                    set_err(errbuf, errlen,
+                            "HAVING over string values: not this round");
+                    return YT_ERR_UNSUPPORTED;
+                }
+                pv[i] = (int64_t)row[i].data.bits;
+                pn[i] = row[i].type == YT_VT_NULL;
+                pt[i] = pn[i] ? YT_VT_INT64 : row[i].type;
+                pvp[i] = &pv[i];
+                pnp[i] = &pn[i];
+            }
+            EvalCtx ctx;
+            memset(&ctx, 0, sizeof(ctx));
+            ctx.col_vals = pvp;
+            ctx.col_nulls = pnp;
+            ctx.col_types = pt;
+            ctx.ncols = ncols;
+            ctx.row = 0;
+            ctx.error = 0;
+            Val h = eval_expr(plan->having, &ctx);
+            if (ctx.error) {
+                set_err(errbuf, errlen, "HAVING expression error");
+                return ctx.error;
+            }
+            if (h.type == YT_VT_NULL || h.bits == 0) continue;
+            if (w != r)
+                memmove(out->values + w * ncols, row, sizeof(YtValue) * ncols);
+            w++;
+        }
+        out->row_count = w;
+    }
+    if (plan->with_totals && after) ord_fold_totals(plan, out, tot);
+    if (plan->order_count > 0) {
+        int rc = apply_order_rowset(plan, out, errbuf, errlen);
+        if (rc != YT_OK) return rc;
+    }
+    if (plan->with_totals) {
+        if (out->row_count >= out->capacity_rows) return YT_ERR_CAPACITY;
+        YtValue* dst = out->values + out->row_count * ncols;
+        for (int k = 0; k < kc; k++) {
+            dst[k].id = (uint16_t)k;
+            dst[k].type = YT_VT_NULL;
+            dst[k].flags = 0;
+            dst[k].length = 0;
+            dst[k].data.bits = 0;
+        }
+        for (int a = 0; a < ac; a++) {
+            dst[kc + a] = tot[a];
+            if (plan->aggs[a]->func == YT_AGG_SUM1 && dst[kc + a].type == YT_VT_NULL) {
+                dst[kc + a].type = YT_VT_INT64;   /* sum(1) over zero rows */
+                dst[kc + a].data.bits = 0;
+            }
+        }
+        out->row_count++;
+        out->totals_row = 1;
+    }
+    return YT_OK;
+}
+
 ORACLE_EXPORT
 int yto_execute(const YtPlan* plan, const YtChunk* chunk,
                 YtRowset* output, YtStatistics* stats,
@@ -928,13 +1058,13 @@ int yto_execute(const YtPlan* plan, const YtChunk* chunk,
     output->row_count = 0;
     output->string_pool_used = 0;
     output->totals_row = 0;
-    if (plan->with_totals && plan->key_count == 0) {
-        set_err(errbuf, errlen, "WITH TOTALS requires GROUP BY");
+    if ((plan->with_totals || plan->having) && plan->key_count == 0) {
+        set_err(errbuf, errlen, "WITH TOTALS / HAVING requires GROUP BY");
         rc = YT_ERR_INVALID_PLAN;
         goto done;
     }
-    if (plan->with_totals && plan->project_count) {
-        set_err(errbuf, errlen, "WITH TOTALS with projections: not this round");
+    if ((plan->with_totals || plan->having) && plan->project_count) {
+        set_err(errbuf, errlen, "WITH TOTALS / HAVING with projections: not this round");
         rc = YT_ERR_UNSUPPORTED;
         goto done;
     }
@@ -1044,55 +1174,9 @@ int yto_execute(const YtPlan* plan, const YtChunk* chunk,
             }
             output->column_count = plan->project_count
                 ? plan->project_count : plan->key_count + plan->agg_count;
-            if (rc == YT_OK && plan->order_count > 0) {
-                rc = apply_order_rowset(plan, output, errbuf, errlen);
-            }
-            if (rc == YT_OK && plan->with_totals) {
-                /* totals stream: null keys + aggregates over ALL groups
-                 * (registry.cpp FlushTotals; ql_query_ut.cpp:3432-3476) */
-                int kc = plan->key_count, ac = plan->agg_count;
-                if (output->row_count >= output->capacity_rows) {
-                    rc = YT_ERR_CAPACITY;
-                } else {
-                    Val tst[16];
-                    uint64_t tcnt = 0;
-                    for (int a = 0; a < ac; a++) tst[a].type = YT_VT_NULL;
-                    for (int64_t g = 0; g < final_t->ngroups; g++) {
-                        tcnt += final_t->rowcounts[g];
-                        for (int a = 0; a < ac; a++) {
-                            Val st = final_t->states[g * ac + a];
-                            if (plan->aggs[a]->func == YT_AGG_SUM)
-                                sum_update_val(&tst[a], st);
-                            else if (plan->aggs[a]->func != YT_AGG_SUM1)
-                                minmax_update_val(&tst[a], st,
-                                    plan->aggs[a]->func == YT_AGG_MAX);
-                        }
-                    }
-                    YtValue* dst = output->values
-                        + output->row_count * (kc + ac);
-                    for (int k = 0; k < kc; k++) {
-                        dst[k].id = (uint16_t)k;
-                        dst[k].type = YT_VT_NULL;
-                        dst[k].flags = 0;
-                        dst[k].length = 0;
-                        dst[k].data.bits = 0;
-                    }
-                    for (int a = 0; a < ac; a++) {
-                        YtValue* v = &dst[kc + a];
-                        v->id = (uint16_t)(kc + a);
-                        v->flags = 0;
-                        v->length = 0;
-                        if (plan->aggs[a]->func == YT_AGG_SUM1) {
-                            v->type = YT_VT_INT64;
-                            v->data.bits = tcnt;
-                        } else {
-                            v->type = tst[a].type;
-                            v->data.bits = tst[a].bits;
-                        }
-                    }
-                    output->row_count++;
-                    output->totals_row = 1;
-                }
+            if (rc == YT_OK && (plan->order_count > 0 || plan->with_totals ||
+                                plan->having)) {
+                rc = finish_group_rowset(plan, output, errbuf, errlen);
             }
             if (stats) {
                 stats->rows_read = n;

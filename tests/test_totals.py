@@ -138,3 +138,83 @@ def test_totals_gpu_two_phase(cuda):
     want, _ = y.oracle_execute(mplan, chunk)
     assert y.sort_rows(got[:-1]) == y.sort_rows(want[:-1])
     assert got[-1] == want[-1] == (None, int(v.sum()), 100_000)
+
+
+# --------- HAVING + totals-mode goldens (ql_query_ut.cpp:3916-3986) ---------
+# "x, sum(b) as t FROM t where a > 1 group by a % 2 = 1 as x ..."
+# Our seam emits [key, aggs] directly; having columns index that output row.
+
+def _having_chunk():
+    a = np.arange(1, 10, dtype=np.int64)
+    b = a * 10
+    return y.Chunk([y.encode_int64(a), y.encode_int64(b)], 9)
+
+
+def _having_plan(having=None, with_totals=False, after=False):
+    return y.Plan(filter=y.col(0) > 1,
+                  keys=[(y.col(0) % 2) == 1],
+                  aggs=[y.agg_sum(y.col(1))],
+                  having=having, with_totals=with_totals,
+                  totals_after_having=after)
+
+
+def test_golden_with_totals():
+    rows, _ = y.oracle_execute(_having_plan(with_totals=True), _having_chunk())
+    assert rows == [(False, 200), (True, 240), (None, 440)]
+
+
+def test_golden_having_then_totals():
+    # "having t > 200 with totals" → AfterHaving: totals over survivors
+    plan = _having_plan(having=y.col(1) > 200, with_totals=True, after=True)
+    rows, _ = y.oracle_execute(plan, _having_chunk())
+    assert rows == [(True, 240), (None, 240)]
+
+
+def test_golden_totals_then_having():
+    # "with totals having t > 200" → BeforeHaving: totals over all groups
+    plan = _having_plan(having=y.col(1) > 200, with_totals=True)
+    rows, _ = y.oracle_execute(plan, _having_chunk())
+    assert rows == [(True, 240), (None, 440)]
+    plan = _having_plan(having=y.col(1) < 220, with_totals=True)
+    rows, _ = y.oracle_execute(plan, _having_chunk())
+    assert rows == [(False, 200), (None, 440)]
+
+
+def test_having_requires_group_by():
+    with pytest.raises(RuntimeError, match="GROUP BY"):
+        y.oracle_execute(y.Plan(projects=[y.col(0)], having=y.col(0) > 1),
+                         _having_chunk())
+
+
+@pytest.mark.gpu
+def test_golden_having_totals_gpu(cuda):
+    chunk = _having_chunk()
+    for kwargs, want in [
+        (dict(with_totals=True),
+         [(False, 200), (True, 240), (None, 440)]),
+        (dict(having=y.col(1) > 200, with_totals=True, after=True),
+         [(True, 240), (None, 240)]),
+        (dict(having=y.col(1) > 200, with_totals=True),
+         [(True, 240), (None, 440)]),
+        (dict(having=y.col(1) < 220, with_totals=True),
+         [(False, 200), (None, 440)]),
+    ]:
+        got, _ = y.gpu_execute(_having_plan(**kwargs), chunk.c_device(cuda),
+                               max_groups_hint=64)
+        assert got[-1] == want[-1]                       # totals row
+        assert y.sort_rows(got[:-1]) == y.sort_rows(want[:-1])
+
+
+@pytest.mark.gpu
+def test_having_gpu_large(cuda):
+    rng = np.random.default_rng(60)
+    n = 400_000
+    gk = rng.integers(0, 2000, n, dtype=np.int64)
+    v = rng.integers(0, 10**6, n, dtype=np.int64)
+    chunk = y.Chunk([y.encode_int64(gk), y.encode_int64(v)], n)
+    plan = y.Plan(keys=[y.col(0)], aggs=[y.agg_sum(y.col(1)), y.agg_sum1()],
+                  having=y.col(1) > 10**8, order_by=[(1, True)], limit=20,
+                  with_totals=True, totals_after_having=True)
+    got, _ = y.gpu_execute(plan, chunk.c_device(cuda), max_groups_hint=4096)
+    want, _ = y.oracle_execute(plan, chunk, nthreads=4)
+    assert got == want

@@ -105,7 +105,8 @@ class YtPlan(C.Structure):
                 ("order_limit", C.c_int64),
                 ("order_offset", C.c_int64),
                 ("with_totals", C.c_int32),
-                ("reserved_", C.c_int32)]
+                ("totals_mode", C.c_int32),
+                ("having", C.POINTER(YtExpr))]
 
 
 class YtExecOptions(C.Structure):

@@ -107,7 +107,8 @@ class Plan:
     required with order_by (reference QL rejects ORDER BY without LIMIT)."""
 
     def __init__(self, filter=None, keys=(), aggs=(), projects=(), is_merge=False,
-                 order_by=(), limit=0, offset=0, with_totals=False):
+                 order_by=(), limit=0, offset=0, with_totals=False,
+                 having=None, totals_after_having=False):
         self.filter = filter
         self.keys = list(keys)
         self.aggs = list(aggs)
@@ -117,6 +118,8 @@ class Plan:
         self.limit = limit
         self.offset = offset
         self.with_totals = with_totals
+        self.having = having
+        self.totals_after_having = totals_after_having
         self._build()
 
     def _build(self):
@@ -165,6 +168,10 @@ class Plan:
         p.order_limit = self.limit
         p.order_offset = self.offset
         p.with_totals = 1 if self.with_totals else 0
+        p.totals_mode = 2 if self.totals_after_having else 1
+        if self.having is not None:
+            p.having = C.pointer(self.having.c)
+            self._keep.append(self.having)
         self.c = p
 
 

@@ -1336,54 +1336,94 @@ static int apply_order_host(const YtPlan* plan, YtRowset* out,
  * reference's EStreamTag::Totals, registry.cpp FlushTotals) are computed
  * BEFORE the order/limit slice and appended after it, flagged in
  * YtRowset.totals_row. */
+static void fold_totals_rows(const YtPlan* plan, const YtRowset* out,
+                             std::vector<YtValue>& tot)
+{
+    int kc = plan->key_count, ac = plan->agg_count;
+    int ncols = out->column_count;
+    tot.assign(ac, YtValue());
+    for (int a = 0; a < ac; a++) {
+        tot[a].id = (uint16_t)(kc + a);
+        tot[a].type = YT_VT_NULL;
+    }
+    for (int64_t r = 0; r < out->row_count; r++) {
+        const YtValue* row = out->values + r * ncols;
+        for (int a = 0; a < ac; a++) {
+            const YtValue& v = row[kc + a];
+            if (v.type == YT_VT_NULL) continue;
+            YtValue& t = tot[a];
+            int f = plan->aggs[a]->func;
+            if (t.type == YT_VT_NULL) {
+                t.type = v.type;
+                t.data.bits = v.data.bits;
+                continue;
+            }
+            if (f == YT_AGG_SUM || f == YT_AGG_SUM1) {
+                if (v.type == YT_VT_DOUBLE) t.data.dbl += v.data.dbl;
+                else t.data.bits += v.data.bits;   /* mod 2^64, udf/sum.c */
+            } else {
+                bool take;
+                if (v.type == YT_VT_DOUBLE)
+                    take = (f == YT_AGG_MAX) ? (v.data.dbl > t.data.dbl)
+                                             : (v.data.dbl < t.data.dbl);
+                else if (v.type == YT_VT_INT64)
+                    take = (f == YT_AGG_MAX) ? (v.data.i64 > t.data.i64)
+                                             : (v.data.i64 < t.data.i64);
+                else
+                    take = (f == YT_AGG_MAX) ? (v.data.u64 > t.data.u64)
+                                             : (v.data.u64 < t.data.u64);
+                if (take) t.data.bits = v.data.bits;
+            }
+        }
+    }
+}
+
+static void hexpr_cols(const YtExpr* e, uint64_t* mask)
+{
+    if (!e) return;
+    if (e->op == YT_EX_COLUMN && e->col >= 0 && e->col < 64)
+        *mask |= 1ULL << e->col;
+    hexpr_cols(e->a, mask);
+    hexpr_cols(e->b, mask);
+}
+
+/* totals(BeforeHaving) → HAVING → totals(AfterHaving) → ORDER/limit →
+ * append totals row (folding_profiler.cpp:1810-1815 Process() order;
+ * parser.ypp:469-481 totals modes) */
 static int finish_output(const YtPlan* plan, YtRowset* out,
                          char* errbuf, size_t errlen)
 {
     std::vector<YtValue> tot;
-    uint64_t tcnt_unused = 0;
-    (void)tcnt_unused;
-    if (plan->with_totals) {
-        int kc = plan->key_count, ac = plan->agg_count;
+    const bool after = plan->totals_mode == 2;
+    if (plan->with_totals && !after) fold_totals_rows(plan, out, tot);
+    if (plan->having) {
         int ncols = out->column_count;
-        tot.resize(ac);
-        for (int a = 0; a < ac; a++) {
-            tot[a].id = (uint16_t)(kc + a);
-            tot[a].type = YT_VT_NULL;
-            tot[a].flags = 0;
-            tot[a].length = 0;
-            tot[a].data.bits = 0;
-        }
+        uint64_t mask = 0;
+        hexpr_cols(plan->having, &mask);
+        int64_t w = 0;
+        HVal hrow[32];
         for (int64_t r = 0; r < out->row_count; r++) {
-            const YtValue* row = out->values + r * ncols;
-            for (int a = 0; a < ac; a++) {
-                const YtValue& v = row[kc + a];
-                if (v.type == YT_VT_NULL) continue;
-                YtValue& t = tot[a];
-                int f = plan->aggs[a]->func;
-                if (t.type == YT_VT_NULL) {
-                    t.type = v.type;
-                    t.data.bits = v.data.bits;
-                    continue;
+            YtValue* row = out->values + r * ncols;
+            for (int i = 0; i < ncols && i < 32; i++) {
+                if (((mask >> i) & 1) && row[i].type == YT_VT_STRING) {
+                    set_err(errbuf, errlen,
+                            "HAVING over string values: not this round");
+                    return YT_ERR_UNSUPPORTED;
                 }
-                if (f == YT_AGG_SUM || f == YT_AGG_SUM1) {
-                    if (v.type == YT_VT_DOUBLE) t.data.dbl += v.data.dbl;
-                    else t.data.bits += v.data.bits;   /* mod 2^64, udf/sum.c */
-                } else {
-                    bool take;
-                    if (v.type == YT_VT_DOUBLE)
-                        take = (f == YT_AGG_MAX) ? (v.data.dbl > t.data.dbl)
-                                                 : (v.data.dbl < t.data.dbl);
-                    else if (v.type == YT_VT_INT64)
-                        take = (f == YT_AGG_MAX) ? (v.data.i64 > t.data.i64)
-                                                 : (v.data.i64 < t.data.i64);
-                    else
-                        take = (f == YT_AGG_MAX) ? (v.data.u64 > t.data.u64)
-                                                 : (v.data.u64 < t.data.u64);
-                    if (take) t.data.bits = v.data.bits;
-                }
+                hrow[i].type = row[i].type;
+                hrow[i].bits = row[i].data.bits;
             }
+            HVal h;
+            int rc = heval(plan->having, hrow, ncols, &h);
+            if (rc) { set_err(errbuf, errlen, "HAVING expression error"); return rc; }
+            if (h.type == YT_VT_NULL || h.bits == 0) continue;
+            if (w != r)
+                memmove(out->values + w * ncols, row, sizeof(YtValue) * ncols);
+            w++;
         }
+        out->row_count = w;
     }
+    if (plan->with_totals && after) fold_totals_rows(plan, out, tot);
     if (plan->order_count > 0) {
         int rc = apply_order_host(plan, out, errbuf, errlen);
         if (rc) return rc;
@@ -1403,7 +1443,14 @@ static int finish_output(const YtPlan* plan, YtRowset* out,
             dst[k].length = 0;
             dst[k].data.bits = 0;
         }
-        for (int a = 0; a < ac; a++) dst[kc + a] = tot[a];
+        for (int a = 0; a < ac; a++) {
+            dst[kc + a] = tot[a];
+            if (plan->aggs[a]->func == YT_AGG_SUM1 &&
+                dst[kc + a].type == YT_VT_NULL) {
+                dst[kc + a].type = YT_VT_INT64;   /* sum(1) over zero rows */
+                dst[kc + a].data.bits = 0;
+            }
+        }
         out->row_count++;
         out->totals_row = 1;
     }
@@ -2254,18 +2301,19 @@ extern "C" int yt_gpu_query_execute(
         return YT_ERR_INVALID_PLAN;
     }
     output->totals_row = 0;
-    if (plan->with_totals) {
+    if (plan->with_totals || plan->having) {
         if (plan->key_count == 0) {
-            set_err(errbuf, errlen, "WITH TOTALS requires GROUP BY");
+            set_err(errbuf, errlen, "WITH TOTALS / HAVING requires GROUP BY");
             return YT_ERR_INVALID_PLAN;
         }
         if (plan->project_count) {
-            set_err(errbuf, errlen, "WITH TOTALS with projections: not this round");
+            set_err(errbuf, errlen,
+                    "WITH TOTALS / HAVING with projections: not this round");
             return YT_ERR_UNSUPPORTED;
         }
         if (options->output_row_limit > 0) {
             set_err(errbuf, errlen,
-                    "WITH TOTALS with OutputRowLimit: not this round");
+                    "WITH TOTALS / HAVING with OutputRowLimit: not this round");
             return YT_ERR_UNSUPPORTED;
         }
     }
@@ -2283,7 +2331,8 @@ extern "C" int yt_gpu_query_execute(
             chunk->columns[kc].value_type == YT_VT_STRING) {
             rc = run_string_group(plan, chunk, options, kc, output, stats,
                                   tw0, errbuf, errlen);
-            if (rc == YT_OK && (plan->order_count > 0 || plan->with_totals))
+            if (rc == YT_OK && (plan->order_count > 0 || plan->with_totals ||
+                                plan->having))
                 rc = finish_output(plan, output, errbuf, errlen);
             return rc;
         }
@@ -2372,8 +2421,8 @@ extern "C" int yt_gpu_query_execute(
         if (rc) return rc;
         if (out_limited && stats) stats->incomplete_output = 1;
     }
-    if (plan->order_count > 0 || plan->with_totals) {
-        /* grouped output is already bounded: order/totals on the host
+    if (plan->order_count > 0 || plan->with_totals || plan->having) {
+        /* grouped output is already bounded: order/totals/having on the host
          * (combined group+order mode, registry.cpp:1677-1699) */
         rc = finish_output(plan, output, errbuf, errlen);
         if (rc) return rc;
@@ -2588,8 +2637,8 @@ extern "C" int yt_gpu_merge_states(
                        output, errbuf, errlen);
         if (rc) return rc;
         if (out_limited && stats) stats->incomplete_output = 1;
-        if (plan->order_count > 0 || plan->with_totals) {
-            /* ORDER BY / WITH TOTALS apply at the front (coordinator) query */
+        if (plan->order_count > 0 || plan->with_totals || plan->having) {
+            /* ORDER BY / WITH TOTALS / HAVING apply at the front query */
             rc = finish_output(plan, output, errbuf, errlen);
             if (rc) return rc;
         }
