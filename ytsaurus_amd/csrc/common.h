@@ -61,9 +61,22 @@ constexpr int kMaxAggs = 4;
  * streams with (offset,len) per role. */
 constexpr int kMaxProj = 8;
 
+constexpr int kMaxPackKeys = 4;
+
 struct DevPlan {
     int32_t ncols;
     uint8_t col_types[kMaxCols];       /* YT_VT_* */
+    /* composite packed group key (key_count > 1): each component column's
+     * zigzag-space value is biased by kp_base and offset by 1 (0 = null),
+     * then packed at kp_shift. Total width <= 62 bits so the table's empty
+     * (0x8000...) sentinel can never collide; key_bits == 0 (all-null key)
+     * rides the existing side-accumulator path. */
+    int32_t kp_count;
+    int32_t kp_col[kMaxPackKeys];
+    int32_t kp_signed[kMaxPackKeys];
+    int32_t kp_shift[kMaxPackKeys];
+    int32_t kp_bits[kMaxPackKeys];
+    uint64_t kp_base[kMaxPackKeys];
     /* log2(rows per segment) when the column's interior segments are all
      * exactly 1<<shift rows (the writer's 128Ki cap makes this the norm):
      * row→segment is then a shift instead of a binary search. 0 = ragged. */

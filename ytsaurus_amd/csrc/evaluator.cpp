@@ -267,6 +267,8 @@ static uint8_t expr_static_type(const YtExpr* e, const uint8_t* col_types)
     }
 }
 
+static bool expr_is_col(const YtExpr* e, int* col);
+
 static int build_devplan(const YtPlan* plan, const YtChunk* chunk, DevPlan* p,
                          char* errbuf, size_t errlen)
 {
@@ -292,7 +294,31 @@ static int build_devplan(const YtPlan* plan, const YtChunk* chunk, DevPlan* p,
         p->col_uniform_shift[c] = shift;
     }
 
-    if (plan->key_count > 1) { set_err(errbuf, errlen, "GPU path: at most 1 group key this round"); return YT_ERR_UNSUPPORTED; }
+    if (plan->key_count > 1) {
+        /* composite packed key: plain int64/uint64/boolean key columns,
+         * widths resolved after segment parse (see pack_group_key) */
+        if (plan->key_count > kMaxPackKeys) {
+            set_err(errbuf, errlen, "GPU path: at most 4 group keys this round");
+            return YT_ERR_UNSUPPORTED;
+        }
+        p->kp_count = plan->key_count;
+        for (int i = 0; i < plan->key_count; i++) {
+            int c;
+            if (!expr_is_col(plan->keys[i], &c) || c >= p->ncols) {
+                set_err(errbuf, errlen,
+                        "multi-key GROUP BY: plain key columns this round");
+                return YT_ERR_UNSUPPORTED;
+            }
+            uint8_t vt = p->col_types[c];
+            if (vt != YT_VT_INT64 && vt != YT_VT_UINT64 && vt != YT_VT_BOOLEAN) {
+                set_err(errbuf, errlen,
+                        "multi-key GROUP BY: int64/uint64/boolean keys this round");
+                return YT_ERR_UNSUPPORTED;
+            }
+            p->kp_col[i] = c;
+            p->kp_signed[i] = vt == YT_VT_INT64;
+        }
+    }
     if (plan->agg_count == 0) {
         if (plan->project_count < 1 || plan->project_count > kMaxProj) {
             set_err(errbuf, errlen, "scan mode needs 1..8 projections");
@@ -1087,6 +1113,37 @@ fail:
     return rc;
 }
 
+static inline int64_t h_zz_dec(uint64_t z)
+{
+    return (int64_t)(z >> 1) ^ -(int64_t)(z & 1);
+}
+
+/* resolve composite-key packing widths from the parsed per-column
+ * zigzag-space ranges (k_parse_segments min/max atomics) */
+static int pack_group_key(DevPlan* dp, const DeviceRun* R,
+                          char* errbuf, size_t errlen)
+{
+    int shift = 0;
+    for (int i = 0; i < dp->kp_count; i++) {
+        int c = dp->kp_col[i];
+        uint64_t lo = R->col_zzmin[c], hi = R->col_zzmax[c];
+        uint64_t span = hi >= lo ? hi - lo : 0;
+        int bits = 1;
+        while (bits < 64 && ((span + 1) >> bits) != 0) bits++;
+        bits += 1;                       /* +1: code 0 reserved for null */
+        dp->kp_base[i] = lo;
+        dp->kp_bits[i] = bits;
+        dp->kp_shift[i] = shift;
+        shift += bits;
+    }
+    if (shift > 62) {
+        set_err(errbuf, errlen,
+                "multi-key GROUP BY: composite key wider than 62 bits this round");
+        return YT_ERR_UNSUPPORTED;
+    }
+    return YT_OK;
+}
+
 /* finalize a group (OutGroup record or side accumulator) into HVal row
  * [keys..., aggs...] */
 static void finalize_row(const YtPlan* plan, uint8_t key_type,
@@ -1133,6 +1190,7 @@ static void finalize_row(const YtPlan* plan, uint8_t key_type,
 }
 
 static int emit_rows(const YtPlan* plan, const YtChunk* chunk,
+                     const DevPlan* dp,
                      const OutGroup* groups, int64_t ngroups,
                      const TableHdr& th, int has_any_row_global,
                      const uint64_t* gaccum, int used_fast_global,
@@ -1143,6 +1201,7 @@ static int emit_rows(const YtPlan* plan, const YtChunk* chunk,
     for (int c = 0; c < chunk->column_count && c < kMaxCols; c++)
         col_types[c] = (uint8_t)chunk->columns[c].value_type;
 
+    const int kp = dp ? dp->kp_count : 0;
     uint8_t key_type = YT_VT_INT64;
     int has_key = plan->key_count == 1;
     if (has_key) key_type = expr_static_type(plan->keys[0], col_types);
@@ -1153,7 +1212,7 @@ static int emit_rows(const YtPlan* plan, const YtChunk* chunk,
         if (sum_type[a] == YT_VT_NULL) sum_type[a] = YT_VT_INT64;
     }
 
-    int base_cols = (has_key ? 1 : 0) + plan->agg_count;
+    int base_cols = (kp ? kp : (has_key ? 1 : 0)) + plan->agg_count;
     int out_cols = plan->project_count ? plan->project_count : base_cols;
 
     auto emit = [&](uint64_t key_bits, int key_null, uint64_t cnt,
@@ -1164,8 +1223,30 @@ static int emit_rows(const YtPlan* plan, const YtChunk* chunk,
             return -1;
         }
         if (output->row_count >= output->capacity_rows) return YT_ERR_CAPACITY;
-        HVal row[2 + kMaxAggs];
+        HVal row[kMaxPackKeys + 1 + kMaxAggs];
         int nrow = 0;
+        if (kp) {
+            /* unpack the composite key (DevPlan kp_*; code 0 = null) */
+            for (int i = 0; i < kp; i++) {
+                uint64_t mask = dp->kp_bits[i] >= 64
+                    ? ~0ULL : ((1ULL << dp->kp_bits[i]) - 1);
+                uint64_t code = (key_bits >> dp->kp_shift[i]) & mask;
+                if (code == 0) {
+                    row[nrow].type = YT_VT_NULL;
+                    row[nrow].bits = 0;
+                } else {
+                    uint64_t z = code - 1 + dp->kp_base[i];
+                    row[nrow].type = col_types[dp->kp_col[i]];
+                    row[nrow].bits = dp->kp_signed[i]
+                        ? (uint64_t)h_zz_dec(z) : z;
+                }
+                nrow++;
+            }
+            int an_ = 0;
+            finalize_row(plan, key_type, sum_type, 0, 0, cnt, ab, an,
+                         row + nrow, &an_, /*has_key=*/0);
+            nrow += an_;
+        } else
         finalize_row(plan, key_type, sum_type, key_bits, key_null, cnt, ab, an,
                      row, &nrow, has_key);
         YtValue* dst = output->values + output->row_count * out_cols;
@@ -1197,7 +1278,7 @@ static int emit_rows(const YtPlan* plan, const YtChunk* chunk,
     output->column_count = out_cols;
     int rc = YT_OK;
 
-    if (!has_key) {
+    if (!has_key && !kp) {
         /* global aggregate: one row iff any row passed the filter */
         uint64_t cnt;
         uint64_t ab[kMaxAggs] = {0}, an[kMaxAggs] = {0};
@@ -2367,6 +2448,10 @@ extern "C" int yt_gpu_query_execute(
     rc = setup_chunk(chunk, &R, &maxw, options->input_row_limit, &in_clamped,
                      errbuf, errlen);
     if (rc) return rc;
+    if (dp.kp_count && chunk->row_count > 0 && R.nsegs > 0) {
+        rc = pack_group_key(&dp, &R, errbuf, errlen);
+        if (rc) return rc;
+    }
 
     if (chunk->row_count == 0 || R.nsegs == 0) {
         /* zero groups -> single final flush emits nothing (registry.cpp:1481) */
@@ -2413,7 +2498,7 @@ extern "C" int yt_gpu_query_execute(
                                 sizeof(uint64_t) * (1 + 2 * kMaxAggs), hipMemcpyDeviceToHost));
         }
         int out_limited = 0;
-        rc = emit_rows(plan, chunk, hgroups, ngroups, th, 0, gaccum.data(),
+        rc = emit_rows(plan, chunk, &dp, hgroups, ngroups, th, 0, gaccum.data(),
                        fs.valid && fs.key_col < 0,
                        options->output_row_limit, &out_limited,
                        output, errbuf, errlen);
@@ -2632,7 +2717,7 @@ extern "C" int yt_gpu_merge_states(
         fake.column_count = kMaxCols;
         fake.columns = cols.data();
         int out_limited = 0;
-        rc = emit_rows(plan, &fake, groups.data(), (int64_t)groups.size(), th, 0,
+        rc = emit_rows(plan, &fake, nullptr, groups.data(), (int64_t)groups.size(), th, 0,
                        nullptr, 0, options->output_row_limit, &out_limited,
                        output, errbuf, errlen);
         if (rc) return rc;

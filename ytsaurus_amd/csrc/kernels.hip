@@ -680,7 +680,25 @@ k_scan_generic(DevPlan p, const DevSeg* segs, const SegEx* segex,
             if (f.null_ || f.bits == 0) continue;
         }
         DVal key;
-        if (p.key_len) {
+        if (p.kp_count) {
+            /* composite packed key (see DevPlan kp_*) */
+            uint64_t kb = 0;
+            for (int i = 0; i < p.kp_count; i++) {
+                DVal v = col_value(p, c, p.kp_col[i]);
+                uint64_t enc = 0;
+                if (!v.null_) {
+                    uint64_t z = p.kp_signed[i]
+                        ? (((uint64_t)v.bits << 1)
+                           ^ (uint64_t)(((int64_t)v.bits) >> 63))
+                        : v.bits;
+                    enc = 1 + (z - p.kp_base[i]);
+                }
+                kb |= enc << p.kp_shift[i];
+            }
+            key.bits = kb;
+            key.type = YT_VT_UINT64;
+            key.null_ = 0;
+        } else if (p.key_len) {
             key = eval_prog(p, c, p.key_off, p.key_len);
         } else {
             key.bits = 1; key.type = YT_VT_INT64; key.null_ = 0;  /* single group, side-stepped below */
