@@ -122,3 +122,17 @@ def test_multikey_bool_key(cuda):
     got, _ = y.gpu_execute(plan, chunk.c_device(cuda), max_groups_hint=256)
     want, _ = y.oracle_execute(plan, chunk)
     assert y.sort_rows(got) == y.sort_rows(want)
+
+
+@pytest.mark.gpu
+def test_single_key_distinct(cuda):
+    # GROUP BY with no aggregates = DISTINCT (newly routed to the group path)
+    rng = np.random.default_rng(77)
+    n = 100_000
+    b = rng.integers(0, 777, n, dtype=np.int64)
+    chunk = y.Chunk([y.encode_int64(b)], n)
+    plan = y.Plan(keys=[y.col(0)])
+    got, _ = y.gpu_execute(plan, chunk.c_device(cuda), max_groups_hint=2048)
+    want, _ = y.oracle_execute(plan, chunk)
+    assert y.sort_rows(got) == y.sort_rows(want)
+    assert len(got) == 777

@@ -319,12 +319,12 @@ static int build_devplan(const YtPlan* plan, const YtChunk* chunk, DevPlan* p,
             p->kp_signed[i] = vt == YT_VT_INT64;
         }
     }
-    if (plan->agg_count == 0) {
+    if (plan->agg_count == 0 && plan->key_count == 0) {
         if (plan->project_count < 1 || plan->project_count > kMaxProj) {
             set_err(errbuf, errlen, "scan mode needs 1..8 projections");
             return YT_ERR_UNSUPPORTED;
         }
-    } else
+    } else if (plan->agg_count > 0)
     if (plan->agg_count < 1 || plan->agg_count > kMaxAggs) { set_err(errbuf, errlen, "need 1..4 aggregates"); return YT_ERR_UNSUPPORTED; }
     for (int a = 0; a < plan->agg_count; a++) {
         int f = plan->aggs[a]->func;
@@ -1128,9 +1128,9 @@ static int pack_group_key(DevPlan* dp, const DeviceRun* R,
         int c = dp->kp_col[i];
         uint64_t lo = R->col_zzmin[c], hi = R->col_zzmax[c];
         uint64_t span = hi >= lo ? hi - lo : 0;
+        /* codes are 0 (null) .. span+1: smallest bits with span+1 < 2^bits */
         int bits = 1;
         while (bits < 64 && ((span + 1) >> bits) != 0) bits++;
-        bits += 1;                       /* +1: code 0 reserved for null */
         dp->kp_base[i] = lo;
         dp->kp_bits[i] = bits;
         dp->kp_shift[i] = shift;
@@ -2377,7 +2377,8 @@ extern "C" int yt_gpu_query_execute(
     if (!options) options = &defopt;
     if (stats) memset(stats, 0, sizeof(*stats));
 
-    if (plan->agg_count == 0 && plan->project_count == 0) {
+    if (plan->agg_count == 0 && plan->project_count == 0 &&
+        plan->key_count == 0) {
         set_err(errbuf, errlen, "empty plan");
         return YT_ERR_INVALID_PLAN;
     }
@@ -2423,7 +2424,7 @@ extern "C" int yt_gpu_query_execute(
     rc = build_devplan(plan, chunk, &dp, errbuf, errlen);
     if (rc) return rc;
 
-    if (plan->agg_count == 0) {
+    if (plan->agg_count == 0 && plan->key_count == 0) {
         if (plan->order_count > 0) {
             /* scan + ORDER BY ... LIMIT: k-selection, no full materialization */
             return run_scan_topk(plan, chunk, options, &dp, output, stats, tw0,
