@@ -177,7 +177,29 @@ typedef struct YtPlan {
      * over the group output row [keys..., aggs...], applied after
      * aggregation and before ORDER BY/limit. */
     const YtExpr* having;
+    /* Equi-join against a foreign rowset (TJoinClause, base/query.h:364-417;
+     * runtime semantics cg_routines/registry.cpp MultiJoinOpHelper:599-960:
+     * joined row = [primary columns..., foreign columns...], INNER drops
+     * unmatched primaries, LEFT null-extends, null keys join null keys —
+     * the codegen eq-comparer treats null == null). This round: one join,
+     * UNIQUE foreign keys (the dimension-lookup case; duplicate keys fail
+     * loudly — the reference's cross-product expansion is round-3 work),
+     * int64/uint64/boolean key columns. The foreign value columns appear to
+     * the rest of the plan as columns [P .. P+foreign_value_count) where P =
+     * the primary chunk's column count — filter/keys/aggs/order may
+     * reference them; the join applies BEFORE the WHERE clause, as in the
+     * reference pipeline. */
+    const struct YtJoin* join;
 } YtPlan;
+
+typedef struct YtJoin {
+    const YtChunk* foreign;          /* encoded foreign rowset */
+    int32_t primary_key_col;         /* equality key column in the primary chunk */
+    int32_t foreign_key_col;         /* equality key column in `foreign` */
+    int32_t foreign_value_count;
+    const int32_t* foreign_value_cols;  /* foreign columns appended to the row */
+    int32_t is_left;                 /* 0 = INNER, 1 = LEFT */
+} YtJoin;
 
 /* ---- execution context / statistics ----
  * Mirrors TExecutionContext limits and TQueryStatistics counters. */

@@ -651,6 +651,8 @@ typedef struct {
     uint8_t* types;
     char** strblob;
     int64_t** strend;
+    int ncols_eff;         /* primary + joined foreign columns */
+    const uint8_t* jdrop;  /* INNER-join misses (row excluded pre-filter) */
     GroupTable table;
     int error;
     int64_t rows_read;
@@ -666,7 +668,7 @@ static void* scan_worker(void* arg)
     ctx.col_types = t->types;
     ctx.col_str = (const char* const*)t->strblob;
     ctx.col_str_end = (const int64_t* const*)t->strend;
-    ctx.ncols = t->chunk->column_count;
+    ctx.ncols = t->ncols_eff;
     ctx.error = 0;
 
     Val keybuf[16];
@@ -675,6 +677,7 @@ static void* scan_worker(void* arg)
     for (int64_t r = t->row_begin; r < t->row_end; r++) {
         ctx.row = r;
         t->rows_read++;
+        if (t->jdrop && t->jdrop[r]) continue;
         if (plan->filter) {
             Val f = eval_expr(plan->filter, &ctx);
             if (ctx.error) { t->error = ctx.error; return NULL; }
@@ -1016,14 +1019,17 @@ int yto_execute(const YtPlan* plan, const YtChunk* chunk,
 {
     if (nthreads < 1) nthreads = 1;
     int ncols = chunk->column_count;
+    int jF = plan->join ? plan->join->foreign_value_count : 0;
+    int ncols_eff = ncols + jF;
+    uint8_t* jdrop = NULL;
     int64_t n = chunk->row_count;
     int rc = YT_OK;
 
-    int64_t** vals = calloc(ncols, sizeof(int64_t*));
-    uint8_t** nulls = calloc(ncols, sizeof(uint8_t*));
-    uint8_t* types = calloc(ncols, 1);
-    char** strblob = calloc(ncols, sizeof(char*));
-    int64_t** strend = calloc(ncols, sizeof(int64_t*));
+    int64_t** vals = calloc(ncols_eff, sizeof(int64_t*));
+    uint8_t** nulls = calloc(ncols_eff, sizeof(uint8_t*));
+    uint8_t* types = calloc(ncols_eff, 1);
+    char** strblob = calloc(ncols_eff, sizeof(char*));
+    int64_t** strend = calloc(ncols_eff, sizeof(int64_t*));
     for (int c = 0; c < ncols; c++) {
         nulls[c] = malloc(n ? n : 1);
         types[c] = (uint8_t)chunk->columns[c].value_type;
@@ -1055,6 +1061,94 @@ int yto_execute(const YtPlan* plan, const YtChunk* chunk,
         if (rc != YT_OK) { set_err(errbuf, errlen, "oracle: bad segment"); goto done; }
     }
 
+    /* equi-join materialization (registry.cpp MultiJoinOpHelper:599-960):
+     * unique foreign keys; joined foreign values become columns
+     * [ncols, ncols_eff); INNER misses drop the primary row pre-filter */
+    if (plan->join) {
+        const YtJoin* J = plan->join;
+        const YtChunk* fc = J->foreign;
+        int64_t fn = fc->row_count;
+        int64_t* fkey = malloc(sizeof(int64_t) * (fn ? fn : 1));
+        uint8_t* fknull = malloc(fn ? fn : 1);
+        rc = yto_decode_column(&fc->columns[J->foreign_key_col], fn, fkey, fknull);
+        if (rc != YT_OK) { free(fkey); free(fknull);
+            set_err(errbuf, errlen, "join: bad foreign key segment"); goto done; }
+        int64_t** fvals = calloc(jF, sizeof(int64_t*));
+        uint8_t** fnulls = calloc(jF, sizeof(uint8_t*));
+        for (int j = 0; j < jF && rc == YT_OK; j++) {
+            int cjf = J->foreign_value_cols[j];
+            if (fc->columns[cjf].value_type == YT_VT_STRING) {
+                set_err(errbuf, errlen, "join: string foreign values not this round");
+                rc = YT_ERR_UNSUPPORTED;
+                break;
+            }
+            fvals[j] = malloc(sizeof(int64_t) * (fn ? fn : 1));
+            fnulls[j] = malloc(fn ? fn : 1);
+            rc = yto_decode_column(&fc->columns[cjf], fn, fvals[j], fnulls[j]);
+        }
+        /* unique-key hash map: open addressing {key,null} -> row */
+        uint64_t cap = 2048;
+        while (cap < (uint64_t)fn * 2) cap <<= 1;
+        int64_t* hrow = NULL;
+        uint64_t* hkey = NULL;
+        uint8_t* hused = NULL;
+        int64_t null_row = -1;
+        if (rc == YT_OK) {
+            hrow = malloc(sizeof(int64_t) * cap);
+            hkey = malloc(sizeof(uint64_t) * cap);
+            hused = calloc(cap, 1);
+            for (int64_t r2 = 0; r2 < fn && rc == YT_OK; r2++) {
+                if (fknull[r2]) {
+                    if (null_row >= 0) { set_err(errbuf, errlen, "join: duplicate foreign key"); rc = YT_ERR_UNSUPPORTED; break; }
+                    null_row = r2;
+                    continue;
+                }
+                uint64_t h = splitmix64((uint64_t)fkey[r2]) & (cap - 1);
+                for (;;) {
+                    if (!hused[h]) { hused[h] = 1; hkey[h] = (uint64_t)fkey[r2]; hrow[h] = r2; break; }
+                    if (hkey[h] == (uint64_t)fkey[r2]) { set_err(errbuf, errlen, "join: duplicate foreign key"); rc = YT_ERR_UNSUPPORTED; break; }
+                    h = (h + 1) & (cap - 1);
+                }
+            }
+        }
+        if (rc == YT_OK) {
+            for (int j = 0; j < jF; j++) {
+                vals[ncols + j] = calloc(n ? n : 1, sizeof(int64_t));
+                nulls[ncols + j] = malloc(n ? n : 1);
+                memset(nulls[ncols + j], 1, n ? n : 1);
+                types[ncols + j] = (uint8_t)fc->columns[J->foreign_value_cols[j]].value_type;
+            }
+            jdrop = calloc(n ? n : 1, 1);
+            int pk = J->primary_key_col;
+            for (int64_t r2 = 0; r2 < n; r2++) {
+                int64_t frow = -1;
+                if (nulls[pk][r2]) {
+                    frow = null_row;      /* null joins null (eq-comparer) */
+                } else {
+                    uint64_t h = splitmix64((uint64_t)vals[pk][r2]) & (cap - 1);
+                    while (hused[h]) {
+                        if (hkey[h] == (uint64_t)vals[pk][r2]) { frow = hrow[h]; break; }
+                        h = (h + 1) & (cap - 1);
+                    }
+                }
+                if (frow < 0) {
+                    if (!J->is_left) jdrop[r2] = 1;
+                    continue;             /* LEFT: joined cols stay null */
+                }
+                for (int j = 0; j < jF; j++) {
+                    if (!fnulls[j][frow]) {
+                        vals[ncols + j][r2] = fvals[j][frow];
+                        nulls[ncols + j][r2] = 0;
+                    }
+                }
+            }
+        }
+        free(fkey); free(fknull);
+        for (int j = 0; j < jF; j++) { if (fvals) free(fvals[j]); if (fnulls) free(fnulls[j]); }
+        free(fvals); free(fnulls); free(hrow); free(hkey); free(hused);
+        if (rc != YT_OK) goto done;
+    }
+
     output->row_count = 0;
     output->string_pool_used = 0;
     output->totals_row = 0;
@@ -1077,11 +1171,12 @@ int yto_execute(const YtPlan* plan, const YtChunk* chunk,
         ctx.col_types = types;
         ctx.col_str = (const char* const*)strblob;
         ctx.col_str_end = (const int64_t* const*)strend;
-        ctx.ncols = ncols;
+        ctx.ncols = ncols_eff;
         ctx.error = 0;
         int np = plan->project_count;
         for (int64_t r = 0; r < n; r++) {
             ctx.row = r;
+            if (jdrop && jdrop[r]) continue;
             if (plan->filter) {
                 Val f = eval_expr(plan->filter, &ctx);
                 if (ctx.error) { rc = ctx.error; set_err(errbuf, errlen, "expr error"); goto done; }
@@ -1134,6 +1229,8 @@ int yto_execute(const YtPlan* plan, const YtChunk* chunk,
             tasks[actual].types = types;
             tasks[actual].strblob = strblob;
             tasks[actual].strend = strend;
+            tasks[actual].ncols_eff = ncols_eff;
+            tasks[actual].jdrop = jdrop;
             gt_init(&tasks[actual].table, plan->key_count, plan->agg_count, 1024);
             actual++;
         }
@@ -1191,13 +1288,14 @@ int yto_execute(const YtPlan* plan, const YtChunk* chunk,
     }
 
 done:
-    for (int c = 0; c < ncols; c++) {
+    for (int c = 0; c < ncols_eff; c++) {
         if (vals) free(vals[c]);
         if (nulls) free(nulls[c]);
         if (strblob) free(strblob[c]);
         if (strend) free(strend[c]);
     }
     free(vals); free(nulls); free(types); free(strblob); free(strend);
+    free(jdrop);
     return rc;
 }
 

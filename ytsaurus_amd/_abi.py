@@ -106,7 +106,17 @@ class YtPlan(C.Structure):
                 ("order_offset", C.c_int64),
                 ("with_totals", C.c_int32),
                 ("totals_mode", C.c_int32),
-                ("having", C.POINTER(YtExpr))]
+                ("having", C.POINTER(YtExpr)),
+                ("join", C.c_void_p)]
+
+
+class YtJoin(C.Structure):
+    _fields_ = [("foreign", C.POINTER(YtChunk)),
+                ("primary_key_col", C.c_int32),
+                ("foreign_key_col", C.c_int32),
+                ("foreign_value_count", C.c_int32),
+                ("foreign_value_cols", C.POINTER(C.c_int32)),
+                ("is_left", C.c_int32)]
 
 
 class YtExecOptions(C.Structure):

@@ -18,7 +18,7 @@ from ._abi import (
 )
 
 __all__ = [
-    "col", "lit", "null", "litf", "Plan", "agg_sum", "agg_sum1",
+    "col", "lit", "null", "litf", "Plan", "Join", "agg_sum", "agg_sum1",
     "encode_int64", "encode_double", "encode_bool", "encode_string", "encode_string_raw",
     "oracle_decode_strings",
     "Chunk", "oracle_execute",
@@ -101,6 +101,31 @@ def agg_sum1():
     return (AGG_SUM1, None)
 
 
+class Join:
+    """Single equi-join spec (TJoinClause slice — see include/ytql_gpu.h
+    YtJoin): foreign columns appear as columns [P .. P+len(value_cols)) to
+    the rest of the plan."""
+
+    def __init__(self, foreign_chunk, primary_key_col, foreign_key_col,
+                 value_cols, is_left=False):
+        self.chunk = foreign_chunk
+        self.primary_key_col = primary_key_col
+        self.foreign_key_col = foreign_key_col
+        self.value_cols = list(value_cols)
+        self.is_left = is_left
+
+    def c_struct(self, cchunk):
+        arr = (C.c_int32 * len(self.value_cols))(*self.value_cols)
+        j = _abi.YtJoin(foreign=C.pointer(cchunk),
+                        primary_key_col=self.primary_key_col,
+                        foreign_key_col=self.foreign_key_col,
+                        foreign_value_count=len(self.value_cols),
+                        foreign_value_cols=arr,
+                        is_left=1 if self.is_left else 0)
+        j._keep = (arr, cchunk)
+        return j
+
+
 class Plan:
     """Mirror of the restated TQuery{WhereClause,GroupClause,OrderClause}
     shape. order_by: list of (output_column_index, desc) pairs; limit is
@@ -108,7 +133,7 @@ class Plan:
 
     def __init__(self, filter=None, keys=(), aggs=(), projects=(), is_merge=False,
                  order_by=(), limit=0, offset=0, with_totals=False,
-                 having=None, totals_after_having=False):
+                 having=None, totals_after_having=False, join=None):
         self.filter = filter
         self.keys = list(keys)
         self.aggs = list(aggs)
@@ -120,6 +145,7 @@ class Plan:
         self.with_totals = with_totals
         self.having = having
         self.totals_after_having = totals_after_having
+        self.join = join
         self._build()
 
     def _build(self):
@@ -424,8 +450,20 @@ def _mk_rowset(capacity, ncols_max=8, pool_bytes=0):
 
 # ---------------- execution ----------------
 
+def _attach_join(plan, cchunk_factory):
+    """Point plan.c.join at a YtJoin built over the given chunk flavor;
+    returns the keep-alive object (None when the plan has no join)."""
+    if getattr(plan, "join", None) is None:
+        plan.c.join = None
+        return None
+    j = plan.join.c_struct(cchunk_factory(plan.join.chunk))
+    plan.c.join = C.cast(C.pointer(j), C.c_void_p)
+    return j
+
+
 def oracle_execute(plan, chunk, nthreads=1, expect_error=False):
     """TEST/BASELINE ONLY — runs the CPU oracle restatement."""
+    _j = _attach_join(plan, lambda c: c.c_host())
     ch = chunk.c_host()
     rs = _mk_rowset(max(chunk.row_count + 16, 1 << 16), pool_bytes=32 << 20)
     st = YtStatistics()
@@ -474,10 +512,20 @@ def gpu_available():
 
 
 def gpu_execute(plan, device_chunk, max_groups_hint=0, group_row_limit=0,
-                stream=0, out_capacity=None, rowset=None, raw_rowset=False):
+                stream=0, out_capacity=None, rowset=None, raw_rowset=False,
+                join_foreign=None):
     """device_chunk: YtChunk with device pointers (Chunk.c_device).
     Pass a preallocated `rowset` (from make_rowset) to avoid per-call
-    allocation; raw_rowset=True skips the Python row conversion."""
+    allocation; raw_rowset=True skips the Python row conversion.
+    join_foreign: device YtChunk for plan.join's foreign rowset (required
+    when the plan has a join)."""
+    _j = None
+    if getattr(plan, "join", None) is not None:
+        assert join_foreign is not None, "plan has a join: pass join_foreign="
+        _j = plan.join.c_struct(join_foreign)
+        plan.c.join = C.cast(C.pointer(_j), C.c_void_p)
+    else:
+        plan.c.join = None
     opts = YtExecOptions(max_groups_hint=max_groups_hint,
                          group_row_limit=group_row_limit, stream=stream)
     if rowset is not None:
