@@ -92,6 +92,31 @@ struct DevPlan {
     int32_t prog_len;
 };
 
+/* Equi-join device context (one join item, UNIQUE foreign keys — see
+ * include/ytql_gpu.h YtJoin). The hash table is open-addressing with the
+ * row index as the claim word (hrow -1 = empty); duplicate keys are
+ * detected by a post-build verify kernel, so inserts never read other
+ * slots' keys (no publish race). Probes run in later kernels only. */
+struct JoinDev {
+    int32_t active;
+    int32_t is_left;
+    int32_t pkey_col;             /* primary key column */
+    int32_t primary_ncols;        /* P: plan columns >= P are foreign */
+    int32_t fval_col[kMaxCols];   /* foreign column per appended slot */
+    int32_t f_shift[kMaxCols];    /* uniform row->segment shift per slot */
+    int32_t fkey_shift;
+    int32_t fkey_col;
+    const DevSeg* fsegs;
+    const SegEx* fsegex;
+    const int32_t* f_off;         /* per FOREIGN column: first segment */
+    const int32_t* f_cnt;
+    const uint64_t* hkey;
+    const int64_t* hrow;          /* -1 = empty */
+    uint64_t hmask;
+    int64_t null_row;             /* foreign row with null key, or -1 */
+    int64_t frows;
+};
+
 /* device output value for the scan+project path (16 B) */
 struct DevOutVal {
     uint64_t bits;

@@ -39,10 +39,12 @@ hipError_t ytql_launch_bucket_agg(const void*, const unsigned long long*, int64_
                                   uint64_t, uint64_t, hipStream_t);
 hipError_t ytql_launch_topk_hist(const DevPlan*, const DevSeg*, const SegEx*,
                                  const int32_t*, const int32_t*, int64_t,
+                                 const JoinDev*,
                                  const TopkPass*, unsigned long long*,
                                  unsigned long long*, unsigned*, hipStream_t);
 hipError_t ytql_launch_topk_gather(const DevPlan*, const DevSeg*, const SegEx*,
                                    const int32_t*, const int32_t*, int64_t,
+                                   const JoinDev*,
                                    const TopkGather*,
                                    int64_t*, unsigned long long*,
                                    int64_t*, unsigned long long*,
@@ -50,13 +52,21 @@ hipError_t ytql_launch_topk_gather(const DevPlan*, const DevSeg*, const SegEx*,
                                    unsigned*, hipStream_t);
 hipError_t ytql_launch_topk_materialize(const DevPlan*, const DevSeg*, const SegEx*,
                                         const int32_t*, const int32_t*,
+                                        const JoinDev*,
                                         const int64_t*, int64_t, DevOutVal*,
                                         unsigned*, hipStream_t);
+hipError_t ytql_launch_join_build(const JoinDev*, int64_t, uint64_t*,
+                                  long long*, unsigned long long*, unsigned*,
+                                  hipStream_t);
+hipError_t ytql_launch_join_verify(const JoinDev*, int64_t, unsigned*,
+                                   hipStream_t);
 hipError_t ytql_launch_scan_project(const DevPlan*, const DevSeg*, const SegEx*,
                                     const int32_t*, const int32_t*, int64_t,
+                                    const JoinDev*,
                                     DevOutVal*, uint8_t*, unsigned*, hipStream_t);
 hipError_t ytql_launch_scan_generic(const DevPlan*, const DevSeg*, const SegEx*,
                                     const int32_t*, const int32_t*, int64_t,
+                                    const JoinDev*,
                                     TableHdr*, unsigned long long*, unsigned*, hipStream_t);
 hipError_t ytql_launch_scan_fast(const FastParams*, const DevSeg*, const SegEx*,
                                  const FastCol*, TableHdr*, unsigned long long*,
@@ -274,9 +284,17 @@ static int build_devplan(const YtPlan* plan, const YtChunk* chunk, DevPlan* p,
 {
     memset(p, 0, sizeof(*p));
     p->ncols = chunk->column_count;
+    if (plan->join) p->ncols += plan->join->foreign_value_count;
     if (p->ncols > kMaxCols) { set_err(errbuf, errlen, "too many columns"); return YT_ERR_UNSUPPORTED; }
-    for (int c = 0; c < p->ncols; c++) p->col_types[c] = (uint8_t)chunk->columns[c].value_type;
-    for (int c = 0; c < p->ncols; c++) {
+    for (int c = 0; c < chunk->column_count; c++) p->col_types[c] = (uint8_t)chunk->columns[c].value_type;
+    if (plan->join) {
+        const YtJoin* J = plan->join;
+        for (int j = 0; j < J->foreign_value_count; j++) {
+            p->col_types[chunk->column_count + j] =
+                (uint8_t)J->foreign->columns[J->foreign_value_cols[j]].value_type;
+        }
+    }
+    for (int c = 0; c < chunk->column_count; c++) {
         const YtColumn& col = chunk->columns[c];
         int shift = 0;
         if (col.segment_count > 0) {
@@ -307,6 +325,11 @@ static int build_devplan(const YtPlan* plan, const YtChunk* chunk, DevPlan* p,
             if (!expr_is_col(plan->keys[i], &c) || c >= p->ncols) {
                 set_err(errbuf, errlen,
                         "multi-key GROUP BY: plain key columns this round");
+                return YT_ERR_UNSUPPORTED;
+            }
+            if (c >= chunk->column_count) {
+                set_err(errbuf, errlen,
+                        "multi-key GROUP BY over joined columns: not this round");
                 return YT_ERR_UNSUPPORTED;
             }
             uint8_t vt = p->col_types[c];
@@ -419,6 +442,7 @@ static void analyze_fast(const YtPlan* plan, const YtChunk* chunk, FastShape* fs
     memset(fs, 0, sizeof(*fs));
     fs->filter_col = -1;
     fs->key_col = -1;
+    if (plan->join) return;           /* joined plans run the generic path */
 
     const uint8_t* ct = nullptr;
     static uint8_t types[kMaxCols];
@@ -989,7 +1013,8 @@ fail:
  * Fills stats->kernel_scan_* from device events. */
 static int run_scan(const YtPlan* plan, const YtChunk* chunk,
                     const YtExecOptions* options, DeviceRun* R,
-                    const DevPlan* dp, const FastShape* fs, unsigned maxw,
+                    const DevPlan* dp, const JoinDev* jd,
+                    const FastShape* fs, unsigned maxw,
                     YtStatistics* stats, char* errbuf, size_t errlen)
 {
     int rc = YT_OK;
@@ -1082,7 +1107,7 @@ static int run_scan(const YtPlan* plan, const YtChunk* chunk,
         if (options->input_row_limit > 0 && options->input_row_limit < gen_rows)
             gen_rows = options->input_row_limit;
         HIP_CHECK(ytql_launch_scan_generic(dp, R->d_segs, R->d_segex, R->d_off, R->d_cnt,
-                                           gen_rows, R->d_th, R->d_slots,
+                                           gen_rows, jd, R->d_th, R->d_slots,
                                            R->d_err, R->stream));
         HIP_CHECK(hipEventRecord(ev1, R->stream));
     }
@@ -1417,6 +1442,132 @@ static int apply_order_host(const YtPlan* plan, YtRowset* out,
  * reference's EStreamTag::Totals, registry.cpp FlushTotals) are computed
  * BEFORE the order/limit slice and appended after it, flagged in
  * YtRowset.totals_row. */
+static int column_uniform_shift(const YtColumn& col)
+{
+    if (col.segment_count == 0) return 0;
+    int32_t r0 = col.segments[0].row_count;
+    if (r0 <= 0 || (r0 & (r0 - 1)) != 0) return 0;
+    for (int i = 0; i + 1 < col.segment_count; i++)
+        if (col.segments[i].row_count != r0) return 0;
+    if (col.segments[col.segment_count - 1].row_count > r0) return 0;
+    int shift = 0;
+    while ((1 << (shift + 1)) <= r0) shift++;
+    return (1 << shift) == r0 ? shift : 0;
+}
+
+/* equi-join runtime: parsed foreign chunk + unique-key hash table in HBM */
+struct JoinRun {
+    DeviceRun Rf;
+    JoinDev jd;
+    uint64_t* d_hkey = nullptr;
+    long long* d_hrow = nullptr;
+    unsigned long long* d_misc = nullptr;
+
+    JoinRun() { memset(&jd, 0, sizeof(jd)); }
+    ~JoinRun()
+    {
+        g_pool.put(d_hkey);
+        g_pool.put(d_hrow);
+        g_pool.put(d_misc);
+    }
+};
+
+static int setup_join(const YtPlan* plan, const YtChunk* chunk, JoinRun* JR,
+                      hipStream_t stream, char* errbuf, size_t errlen)
+{
+    int rc = YT_OK;
+    const YtJoin* J = plan->join;
+    const YtChunk* fc = J->foreign;
+    if (J->primary_key_col < 0 || J->primary_key_col >= chunk->column_count ||
+        J->foreign_key_col < 0 || J->foreign_key_col >= fc->column_count) {
+        set_err(errbuf, errlen, "join: key column out of range");
+        return YT_ERR_INVALID_PLAN;
+    }
+    auto intish = [](int vt) {
+        return vt == YT_VT_INT64 || vt == YT_VT_UINT64 || vt == YT_VT_BOOLEAN;
+    };
+    if (!intish(chunk->columns[J->primary_key_col].value_type) ||
+        !intish(fc->columns[J->foreign_key_col].value_type)) {
+        set_err(errbuf, errlen,
+                "join: int64/uint64/boolean key columns this round");
+        return YT_ERR_UNSUPPORTED;
+    }
+    for (int j = 0; j < J->foreign_value_count; j++) {
+        int cjf = J->foreign_value_cols[j];
+        if (cjf < 0 || cjf >= fc->column_count) {
+            set_err(errbuf, errlen, "join: foreign value column out of range");
+            return YT_ERR_INVALID_PLAN;
+        }
+        if (fc->columns[cjf].value_type == YT_VT_STRING) {
+            set_err(errbuf, errlen, "join: string foreign values not this round");
+            return YT_ERR_UNSUPPORTED;
+        }
+    }
+
+    JR->Rf.stream = stream;
+    unsigned mw = 0;
+    int cl = 0;
+    rc = setup_chunk(fc, &JR->Rf, &mw, 0, &cl, errbuf, errlen);
+    if (rc) return rc;
+
+    int64_t fn = fc->row_count;
+    uint64_t cap = next_pow2((uint64_t)(fn ? fn : 1) * 2);
+    if (cap < 2048) cap = 2048;
+    JoinDev& jd = JR->jd;
+    unsigned long long nr1 = 0;
+    unsigned kerr = 0;
+    HIP_CHECK(pool_alloc(&JR->d_hkey, sizeof(uint64_t) * cap));
+    HIP_CHECK(pool_alloc(&JR->d_hrow, sizeof(long long) * cap));
+    HIP_CHECK(pool_alloc(&JR->d_misc, sizeof(unsigned long long) * 2));
+    HIP_CHECK(hipMemsetAsync(JR->d_hrow, 0xFF, sizeof(long long) * cap, stream));
+    HIP_CHECK(hipMemsetAsync(JR->d_misc, 0, sizeof(unsigned long long) * 2, stream));
+
+    jd.active = 1;
+    jd.is_left = J->is_left ? 1 : 0;
+    jd.pkey_col = J->primary_key_col;
+    jd.primary_ncols = chunk->column_count;
+    jd.fkey_col = J->foreign_key_col;
+    jd.fkey_shift = column_uniform_shift(fc->columns[J->foreign_key_col]);
+    for (int j = 0; j < J->foreign_value_count; j++) {
+        jd.fval_col[j] = J->foreign_value_cols[j];
+        jd.f_shift[j] = column_uniform_shift(fc->columns[J->foreign_value_cols[j]]);
+    }
+    jd.fsegs = JR->Rf.d_segs;
+    jd.fsegex = JR->Rf.d_segex;
+    jd.f_off = JR->Rf.d_off;
+    jd.f_cnt = JR->Rf.d_cnt;
+    jd.hkey = JR->d_hkey;
+    jd.hrow = (const int64_t*)JR->d_hrow;
+    jd.hmask = cap - 1;
+    jd.null_row = -1;
+    jd.frows = fn;
+
+    if (fn > 0 && JR->Rf.nsegs > 0) {
+        HIP_CHECK(hipMemsetAsync(JR->Rf.d_err, 0, sizeof(unsigned), stream));
+        HIP_CHECK(ytql_launch_join_build(&jd, fn, JR->d_hkey, JR->d_hrow,
+                                         JR->d_misc, JR->Rf.d_err, stream));
+        HIP_CHECK(hipMemcpy(&nr1, JR->d_misc, sizeof(unsigned long long),
+                            hipMemcpyDeviceToHost));
+        jd.null_row = nr1 ? (int64_t)(nr1 - 1) : -1;
+        HIP_CHECK(ytql_launch_join_verify(&jd, fn, JR->Rf.d_err, stream));
+        HIP_CHECK(hipMemcpy(&kerr, JR->Rf.d_err, sizeof(unsigned),
+                            hipMemcpyDeviceToHost));
+        if (kerr == 200) {
+            set_err(errbuf, errlen,
+                    "join: duplicate foreign key (unique-key joins this round)");
+            return YT_ERR_UNSUPPORTED;
+        }
+        if (kerr) {
+            set_err(errbuf, errlen, "join: foreign decode error");
+            return (int)kerr;
+        }
+        HIP_CHECK(hipMemsetAsync(JR->Rf.d_err, 0, sizeof(unsigned), stream));
+    }
+    return YT_OK;
+fail:
+    return rc;
+}
+
 static void fold_totals_rows(const YtPlan* plan, const YtRowset* out,
                              std::vector<YtValue>& tot)
 {
@@ -1546,6 +1697,7 @@ static int finish_output(const YtPlan* plan, YtRowset* out,
  * filter/projection the generic path supports works here. */
 static int run_scan_topk(const YtPlan* plan, const YtChunk* chunk,
                          const YtExecOptions* options, const DevPlan* dp,
+                         const JoinDev* jd,
                          YtRowset* output, YtStatistics* stats, double tw0,
                          char* errbuf, size_t errlen)
 {
@@ -1618,7 +1770,7 @@ static int run_scan_topk(const YtPlan* plan, const YtChunk* chunk,
                                          sizeof(unsigned long long), R2.stream));
             }
             HIP_CHECK(ytql_launch_topk_hist(dp, R2.d_segs, R2.d_segex, R2.d_off,
-                                            R2.d_cnt, n, &tp, d_bins, d_misc,
+                                            R2.d_cnt, n, jd, &tp, d_bins, d_misc,
                                             R2.d_err, R2.stream));
             launches++;
             HIP_CHECK(hipMemcpy(h_bins.data(), d_bins,
@@ -1726,7 +1878,7 @@ static int run_scan_topk(const YtPlan* plan, const YtChunk* chunk,
         HIP_CHECK(hipMemsetAsync(d_ctrs, 0, 3 * sizeof(unsigned long long),
                                  R2.stream));
         HIP_CHECK(ytql_launch_topk_gather(dp, R2.d_segs, R2.d_segex, R2.d_off,
-                                          R2.d_cnt, n, &tg,
+                                          R2.d_cnt, n, jd, &tg,
                                           d_rows_strict, d_ctrs,
                                           d_rows_tie, d_ctrs + 1,
                                           d_rows_null, d_ctrs + 2,
@@ -1758,7 +1910,7 @@ static int run_scan_topk(const YtPlan* plan, const YtChunk* chunk,
         HIP_CHECK(pool_alloc_host(&h_vals, sizeof(DevOutVal) * (M ? M : 1) * np));
         if (M) {
             HIP_CHECK(ytql_launch_topk_materialize(dp, R2.d_segs, R2.d_segex,
-                                                   R2.d_off, R2.d_cnt,
+                                                   R2.d_off, R2.d_cnt, jd,
                                                    d_rows_all, M, d_vals,
                                                    R2.d_err, R2.stream));
             launches++;
@@ -1874,6 +2026,7 @@ fail:
  * and a pass mask; the host compacts in row order. */
 static int run_scan_project(const YtPlan* plan, const YtChunk* chunk,
                             const YtExecOptions* options, const DevPlan* dp,
+                            const JoinDev* jd,
                             YtRowset* output, YtStatistics* stats, double tw0,
                             char* errbuf, size_t errlen)
 {
@@ -1905,7 +2058,7 @@ static int run_scan_project(const YtPlan* plan, const YtChunk* chunk,
         HIP_CHECK(hipEventCreate(&e1));
         HIP_CHECK(hipEventRecord(e0, R2.stream));
         HIP_CHECK(ytql_launch_scan_project(dp, R2.d_segs, R2.d_segex, R2.d_off,
-                                           R2.d_cnt, n, d_out, d_pass, R2.d_err,
+                                           R2.d_cnt, n, jd, d_out, d_pass, R2.d_err,
                                            R2.stream));
         HIP_CHECK(hipEventRecord(e1, R2.stream));
         HIP_CHECK(pool_alloc_host(&h_out, sizeof(DevOutVal) * n * plan->project_count));
@@ -2411,6 +2564,11 @@ extern "C" int yt_gpu_query_execute(
         if (plan->key_count == 1 && plan->agg_count > 0 &&
             expr_is_col(plan->keys[0], &kc) && kc < chunk->column_count &&
             chunk->columns[kc].value_type == YT_VT_STRING) {
+            if (plan->join) {
+                set_err(errbuf, errlen,
+                        "join with string group keys: not this round");
+                return YT_ERR_UNSUPPORTED;
+            }
             rc = run_string_group(plan, chunk, options, kc, output, stats,
                                   tw0, errbuf, errlen);
             if (rc == YT_OK && (plan->order_count > 0 || plan->with_totals ||
@@ -2424,14 +2582,24 @@ extern "C" int yt_gpu_query_execute(
     rc = build_devplan(plan, chunk, &dp, errbuf, errlen);
     if (rc) return rc;
 
+    static const JoinDev kNoJoin = {};
+    JoinRun JR;
+    const JoinDev* jd = &kNoJoin;
+    if (plan->join) {
+        rc = setup_join(plan, chunk, &JR,
+                        (hipStream_t)(uintptr_t)options->stream, errbuf, errlen);
+        if (rc) return rc;
+        jd = &JR.jd;
+    }
+
     if (plan->agg_count == 0 && plan->key_count == 0) {
         if (plan->order_count > 0) {
             /* scan + ORDER BY ... LIMIT: k-selection, no full materialization */
-            return run_scan_topk(plan, chunk, options, &dp, output, stats, tw0,
-                                 errbuf, errlen);
+            return run_scan_topk(plan, chunk, options, &dp, jd, output, stats,
+                                 tw0, errbuf, errlen);
         }
-        return run_scan_project(plan, chunk, options, &dp, output, stats, tw0,
-                                errbuf, errlen);
+        return run_scan_project(plan, chunk, options, &dp, jd, output, stats,
+                                tw0, errbuf, errlen);
     }
 
     FastShape fs;
@@ -2467,7 +2635,7 @@ extern "C" int yt_gpu_query_execute(
                      options->group_row_limit, errbuf, errlen);
     if (rc) return rc;
 
-    rc = run_scan(plan, chunk, options, &R, &dp, &fs, maxw, stats, errbuf, errlen);
+    rc = run_scan(plan, chunk, options, &R, &dp, jd, &fs, maxw, stats, errbuf, errlen);
     if (rc) return rc;
 
     /* compact + readback (pinned staging from the pool) */
@@ -2543,6 +2711,7 @@ extern "C" int yt_gpu_query_partial(
     int rc = yt_gpu_available(errbuf, errlen);
     if (rc != YT_OK) return rc;
     if (plan->key_count != 1) { set_err(errbuf, errlen, "partial: need exactly 1 key"); return YT_ERR_UNSUPPORTED; }
+    if (plan->join) { set_err(errbuf, errlen, "partial: join at the bottom query not this round"); return YT_ERR_UNSUPPORTED; }
     int sum_slot = -1;
     for (int a = 0; a < plan->agg_count; a++) {
         if (plan->aggs[a]->func == YT_AGG_SUM) {
@@ -2580,7 +2749,8 @@ extern "C" int yt_gpu_query_partial(
     rc = setup_table(&R, plan->agg_count, options->max_groups_hint,
                      options->group_row_limit, errbuf, errlen);
     if (rc) return rc;
-    rc = run_scan(plan, chunk, options, &R, &dp, &fs, maxw, stats, errbuf, errlen);
+    static const JoinDev kNoJoin2 = {};
+    rc = run_scan(plan, chunk, options, &R, &dp, &kNoJoin2, &fs, maxw, stats, errbuf, errlen);
     if (rc) return rc;
 
     TableHdr th;

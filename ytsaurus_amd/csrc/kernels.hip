@@ -406,10 +406,31 @@ struct ColCtx {
     const int32_t* col_seg_cnt;
     int64_t row;                  /* chunk row */
     uint32_t error;               /* YT_ERR_* */
+    /* equi-join state: jrow_for caches which row the probe resolved */
+    const JoinDev* jt = nullptr;
+    int64_t jrow = -1;
+    int64_t jrow_for = -1;
 };
+
+__device__ int64_t join_resolve(const DevPlan& p, ColCtx& c);
+__device__ DVal jforeign_at(const JoinDev& jt, int fcol, int shift,
+                            int64_t frow, uint8_t vt);
 
 __device__ DVal col_value(const DevPlan& p, ColCtx& c, int col)
 {
+    if (c.jt && c.jt->active && col >= c.jt->primary_ncols) {
+        int slot = col - c.jt->primary_ncols;
+        int64_t frow = join_resolve(p, c);
+        if (frow < 0) {
+            DVal v;
+            v.type = p.col_types[col];
+            v.null_ = 1;
+            v.bits = 0;
+            return v;
+        }
+        return jforeign_at(*c.jt, c.jt->fval_col[slot], c.jt->f_shift[slot],
+                           frow, p.col_types[col]);
+    }
     int off = c.col_seg_off[col];
     int cnt = c.col_seg_cnt[col];
     int lo;
@@ -427,6 +448,63 @@ __device__ DVal col_value(const DevPlan& p, ColCtx& c, int col)
     }
     const DevSeg& s = c.segs[off + lo];
     return seg_value_at(s, c.segex[off + lo], c.row - s.start_row, p.col_types[col]);
+}
+
+/* fetch a FOREIGN chunk value (row, column index into the foreign chunk) */
+__device__ DVal jforeign_at(const JoinDev& jt, int fcol, int shift,
+                            int64_t frow, uint8_t vt)
+{
+    int off = jt.f_off[fcol];
+    int cnt = jt.f_cnt[fcol];
+    int lo;
+    if (shift) {
+        lo = (int)(frow >> shift);
+    } else {
+        int hi = cnt;
+        lo = 0;
+        while (lo + 1 < hi) {
+            int mid = (lo + hi) / 2;
+            if (jt.fsegs[off + mid].start_row <= frow) lo = mid;
+            else hi = mid;
+        }
+    }
+    const DevSeg& s = jt.fsegs[off + lo];
+    return seg_value_at(s, jt.fsegex[off + lo], frow - s.start_row, vt);
+}
+
+__device__ DVal col_value(const DevPlan& p, ColCtx& c, int col);
+
+/* resolve the join match for the current row (cached per row); returns the
+ * foreign row index or -1. null primary keys join the null foreign key —
+ * the reference eq-comparer treats null == null
+ * (cg_fragment_compiler.cpp:425-447). */
+__device__ int64_t join_resolve(const DevPlan& p, ColCtx& c)
+{
+    const JoinDev& jt = *c.jt;
+    if (c.jrow_for == c.row) return c.jrow;
+    c.jrow_for = c.row;
+    c.jrow = -1;
+    DVal k = col_value(p, c, jt.pkey_col);
+    if (k.null_) {
+        c.jrow = jt.null_row;
+        return c.jrow;
+    }
+    uint64_t h = mix64(k.bits) & jt.hmask;
+    for (;;) {
+        int64_t r = jt.hrow[h];
+        if (r < 0) break;
+        if (jt.hkey[h] == k.bits) { c.jrow = r; break; }
+        h = (h + 1) & jt.hmask;
+    }
+    return c.jrow;
+}
+
+/* INNER-join row gate: call once per row before evaluating expressions */
+__device__ __forceinline__ bool join_row_ok(const DevPlan& p, ColCtx& c)
+{
+    if (!c.jt || !c.jt->active) return true;
+    if (c.jt->is_left) return true;
+    return join_resolve(p, c) >= 0;
 }
 
 __device__ DVal eval_prog(const DevPlan& p, ColCtx& c, int off, int len)
@@ -653,12 +731,68 @@ __device__ void table_update_generic(TableHdr* th, unsigned long long* slots,
 }
 
 /* ------------------------------------------------------------------ */
+/* equi-join foreign-table build + unique-key verify                    */
+
+__global__ void k_join_build(JoinDev jt, int64_t frows,
+                             uint64_t* hkey, long long* hrow,
+                             unsigned long long* null_row_plus1,
+                             unsigned* error_out)
+{
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         r < frows; r += stride) {
+        DVal k = jforeign_at(jt, jt.fkey_col, jt.fkey_shift, r, YT_VT_INT64);
+        if (k.null_) {
+            /* at most one null-key row (unique keys) */
+            unsigned long long prev = atomicCAS(null_row_plus1, 0ULL,
+                                                (unsigned long long)(r + 1));
+            if (prev != 0ULL) atomicMax(error_out, 200u);
+            continue;
+        }
+        uint64_t h = mix64(k.bits) & jt.hmask;
+        for (;;) {
+            unsigned long long prev = atomicCAS(
+                (unsigned long long*)&hrow[h],
+                (unsigned long long)(long long)-1,
+                (unsigned long long)r);
+            if (prev == (unsigned long long)(long long)-1) {
+                hkey[h] = k.bits;
+                break;
+            }
+            h = (h + 1) & jt.hmask;
+        }
+    }
+}
+
+/* all inserts are visible now: the FIRST slot in a key's probe chain must
+ * be this row's own slot, else two foreign rows share a key */
+__global__ void k_join_verify(JoinDev jt, int64_t frows, unsigned* error_out)
+{
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         r < frows; r += stride) {
+        DVal k = jforeign_at(jt, jt.fkey_col, jt.fkey_shift, r, YT_VT_INT64);
+        if (k.null_) continue;
+        uint64_t h = mix64(k.bits) & jt.hmask;
+        for (;;) {
+            int64_t rr = jt.hrow[h];
+            if (rr < 0) break;                        /* unreachable */
+            if (jt.hkey[h] == k.bits) {
+                if (rr != r) atomicMax(error_out, 200u);
+                break;
+            }
+            h = (h + 1) & jt.hmask;
+        }
+    }
+}
+
+/* ------------------------------------------------------------------ */
 /* generic fused scan (any segment types / expressions)                */
 
 __global__ void __launch_bounds__(256)
 k_scan_generic(DevPlan p, const DevSeg* segs, const SegEx* segex,
                const int32_t* col_seg_off, const int32_t* col_seg_cnt,
-               int64_t row_count,
+               int64_t row_count, JoinDev jd,
                TableHdr* th, unsigned long long* slots,
                unsigned* error_out)
 {
@@ -668,6 +802,7 @@ k_scan_generic(DevPlan p, const DevSeg* segs, const SegEx* segex,
     c.col_seg_off = col_seg_off;
     c.col_seg_cnt = col_seg_cnt;
     c.error = 0;
+    c.jt = &jd;
 
     DVal aggv[kMaxAggs];
 
@@ -675,6 +810,7 @@ k_scan_generic(DevPlan p, const DevSeg* segs, const SegEx* segex,
     for (int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
          r < row_count; r += stride) {
         c.row = r;
+        if (!join_row_ok(p, c)) continue;
         if (p.filter_len) {
             DVal f = eval_prog(p, c, p.filter_off, p.filter_len);
             if (f.null_ || f.bits == 0) continue;
@@ -721,7 +857,7 @@ k_scan_generic(DevPlan p, const DevSeg* segs, const SegEx* segex,
 __global__ void __launch_bounds__(256)
 k_scan_project(DevPlan p, const DevSeg* segs, const SegEx* segex,
                const int32_t* col_seg_off, const int32_t* col_seg_cnt,
-               int64_t row_count, DevOutVal* out, uint8_t* pass,
+               int64_t row_count, JoinDev jd, DevOutVal* out, uint8_t* pass,
                unsigned* error_out)
 {
     ColCtx c;
@@ -730,12 +866,14 @@ k_scan_project(DevPlan p, const DevSeg* segs, const SegEx* segex,
     c.col_seg_off = col_seg_off;
     c.col_seg_cnt = col_seg_cnt;
     c.error = 0;
+    c.jt = &jd;
 
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
          r < row_count; r += stride) {
         c.row = r;
         pass[r] = 0;
+        if (!join_row_ok(p, c)) continue;
         if (p.filter_len) {
             DVal f = eval_prog(p, c, p.filter_off, p.filter_len);
             if (f.null_ || f.bits == 0) continue;
@@ -791,7 +929,7 @@ __device__ __forceinline__ bool topk_map(const DVal& v, int desc,
 __global__ void __launch_bounds__(256)
 k_topk_hist(DevPlan p, const DevSeg* segs, const SegEx* segex,
             const int32_t* col_seg_off, const int32_t* col_seg_cnt,
-            int64_t row_count, TopkPass tp,
+            int64_t row_count, JoinDev jd, TopkPass tp,
             unsigned long long* bins,           /* 2048 */
             unsigned long long* misc,           /* null_cnt, mmin, mmax */
             unsigned* error_out)
@@ -806,6 +944,7 @@ k_topk_hist(DevPlan p, const DevSeg* segs, const SegEx* segex,
     c.col_seg_off = col_seg_off;
     c.col_seg_cnt = col_seg_cnt;
     c.error = 0;
+    c.jt = &jd;
     unsigned long long nulls = 0;
     uint64_t mmin = ~0ULL, mmax = 0;
 
@@ -813,6 +952,7 @@ k_topk_hist(DevPlan p, const DevSeg* segs, const SegEx* segex,
     for (int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
          r < row_count; r += stride) {
         c.row = r;
+        if (!join_row_ok(p, c)) continue;
         if (p.filter_len) {
             DVal f = eval_prog(p, c, p.filter_off, p.filter_len);
             if (f.null_ || f.bits == 0) continue;
@@ -845,7 +985,7 @@ k_topk_hist(DevPlan p, const DevSeg* segs, const SegEx* segex,
 __global__ void __launch_bounds__(256)
 k_topk_gather(DevPlan p, const DevSeg* segs, const SegEx* segex,
               const int32_t* col_seg_off, const int32_t* col_seg_cnt,
-              int64_t row_count, TopkGather tg,
+              int64_t row_count, JoinDev jd, TopkGather tg,
               int64_t* rows_strict, unsigned long long* ctr_strict,
               int64_t* rows_tie, unsigned long long* ctr_tie,
               int64_t* rows_null, unsigned long long* ctr_null,
@@ -857,11 +997,13 @@ k_topk_gather(DevPlan p, const DevSeg* segs, const SegEx* segex,
     c.col_seg_off = col_seg_off;
     c.col_seg_cnt = col_seg_cnt;
     c.error = 0;
+    c.jt = &jd;
 
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
          r < row_count; r += stride) {
         c.row = r;
+        if (!join_row_ok(p, c)) continue;
         if (p.filter_len) {
             DVal f = eval_prog(p, c, p.filter_off, p.filter_len);
             if (f.null_ || f.bits == 0) continue;
@@ -891,7 +1033,7 @@ k_topk_gather(DevPlan p, const DevSeg* segs, const SegEx* segex,
 __global__ void __launch_bounds__(256)
 k_topk_materialize(DevPlan p, const DevSeg* segs, const SegEx* segex,
                    const int32_t* col_seg_off, const int32_t* col_seg_cnt,
-                   const int64_t* rows, int64_t m, DevOutVal* out,
+                   JoinDev jd, const int64_t* rows, int64_t m, DevOutVal* out,
                    unsigned* error_out)
 {
     ColCtx c;
@@ -900,6 +1042,7 @@ k_topk_materialize(DevPlan p, const DevSeg* segs, const SegEx* segex,
     c.col_seg_off = col_seg_off;
     c.col_seg_cnt = col_seg_cnt;
     c.error = 0;
+    c.jt = &jd;
 
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -1941,7 +2084,8 @@ hipError_t ytql_launch_topk_hist(const DevPlan* p, const DevSeg* segs,
                                  const SegEx* segex,
                                  const int32_t* col_seg_off,
                                  const int32_t* col_seg_cnt,
-                                 int64_t row_count, const TopkPass* tp,
+                                 int64_t row_count, const JoinDev* jd,
+                                 const TopkPass* tp,
                                  unsigned long long* bins,
                                  unsigned long long* null_cnt,
                                  unsigned* error_out, hipStream_t st)
@@ -1951,7 +2095,31 @@ hipError_t ytql_launch_topk_hist(const DevPlan* p, const DevSeg* segs,
     int grid = (int)(want > 4096 ? 4096 : (want > 0 ? want : 1));
     hipLaunchKernelGGL(k_topk_hist, dim3(grid), dim3(block), 0, st,
                        *p, segs, segex, col_seg_off, col_seg_cnt, row_count,
-                       *tp, bins, null_cnt, error_out);
+                       *jd, *tp, bins, null_cnt, error_out);
+    return hipGetLastError();
+}
+
+hipError_t ytql_launch_join_build(const JoinDev* jd, int64_t frows,
+                                  uint64_t* hkey, long long* hrow,
+                                  unsigned long long* null_row_plus1,
+                                  unsigned* error_out, hipStream_t st)
+{
+    int block = 256;
+    int64_t want = (frows + block - 1) / block;
+    int grid = (int)(want > 4096 ? 4096 : (want > 0 ? want : 1));
+    hipLaunchKernelGGL(k_join_build, dim3(grid), dim3(block), 0, st,
+                       *jd, frows, hkey, hrow, null_row_plus1, error_out);
+    return hipGetLastError();
+}
+
+hipError_t ytql_launch_join_verify(const JoinDev* jd, int64_t frows,
+                                   unsigned* error_out, hipStream_t st)
+{
+    int block = 256;
+    int64_t want = (frows + block - 1) / block;
+    int grid = (int)(want > 4096 ? 4096 : (want > 0 ? want : 1));
+    hipLaunchKernelGGL(k_join_verify, dim3(grid), dim3(block), 0, st,
+                       *jd, frows, error_out);
     return hipGetLastError();
 }
 
@@ -1959,7 +2127,8 @@ hipError_t ytql_launch_topk_gather(const DevPlan* p, const DevSeg* segs,
                                    const SegEx* segex,
                                    const int32_t* col_seg_off,
                                    const int32_t* col_seg_cnt,
-                                   int64_t row_count, const TopkGather* tg,
+                                   int64_t row_count, const JoinDev* jd,
+                                   const TopkGather* tg,
                                    int64_t* rows_strict, unsigned long long* ctr_strict,
                                    int64_t* rows_tie, unsigned long long* ctr_tie,
                                    int64_t* rows_null, unsigned long long* ctr_null,
@@ -1970,7 +2139,7 @@ hipError_t ytql_launch_topk_gather(const DevPlan* p, const DevSeg* segs,
     int grid = (int)(want > 4096 ? 4096 : (want > 0 ? want : 1));
     hipLaunchKernelGGL(k_topk_gather, dim3(grid), dim3(block), 0, st,
                        *p, segs, segex, col_seg_off, col_seg_cnt, row_count,
-                       *tg, rows_strict, ctr_strict, rows_tie, ctr_tie,
+                       *jd, *tg, rows_strict, ctr_strict, rows_tie, ctr_tie,
                        rows_null, ctr_null, error_out);
     return hipGetLastError();
 }
@@ -1979,6 +2148,7 @@ hipError_t ytql_launch_topk_materialize(const DevPlan* p, const DevSeg* segs,
                                         const SegEx* segex,
                                         const int32_t* col_seg_off,
                                         const int32_t* col_seg_cnt,
+                                        const JoinDev* jd,
                                         const int64_t* rows, int64_t m,
                                         DevOutVal* out, unsigned* error_out,
                                         hipStream_t st)
@@ -1987,7 +2157,7 @@ hipError_t ytql_launch_topk_materialize(const DevPlan* p, const DevSeg* segs,
     int64_t want = (m + block - 1) / block;
     int grid = (int)(want > 4096 ? 4096 : (want > 0 ? want : 1));
     hipLaunchKernelGGL(k_topk_materialize, dim3(grid), dim3(block), 0, st,
-                       *p, segs, segex, col_seg_off, col_seg_cnt, rows, m,
+                       *p, segs, segex, col_seg_off, col_seg_cnt, *jd, rows, m,
                        out, error_out);
     return hipGetLastError();
 }
@@ -1996,7 +2166,8 @@ hipError_t ytql_launch_scan_project(const DevPlan* p, const DevSeg* segs,
                                     const SegEx* segex,
                                     const int32_t* col_seg_off,
                                     const int32_t* col_seg_cnt,
-                                    int64_t row_count, DevOutVal* out,
+                                    int64_t row_count, const JoinDev* jd,
+                                    DevOutVal* out,
                                     uint8_t* pass, unsigned* error_out,
                                     hipStream_t st)
 {
@@ -2005,7 +2176,7 @@ hipError_t ytql_launch_scan_project(const DevPlan* p, const DevSeg* segs,
     int grid = (int)(want > 4096 ? 4096 : (want > 0 ? want : 1));
     hipLaunchKernelGGL(k_scan_project, dim3(grid), dim3(block), 0, st,
                        *p, segs, segex, col_seg_off, col_seg_cnt, row_count,
-                       out, pass, error_out);
+                       *jd, out, pass, error_out);
     return hipGetLastError();
 }
 
@@ -2013,7 +2184,7 @@ hipError_t ytql_launch_scan_generic(const DevPlan* p, const DevSeg* segs,
                                     const SegEx* segex,
                                     const int32_t* col_seg_off,
                                     const int32_t* col_seg_cnt,
-                                    int64_t row_count,
+                                    int64_t row_count, const JoinDev* jd,
                                     TableHdr* th, unsigned long long* slots,
                                     unsigned* error_out, hipStream_t st)
 {
@@ -2022,7 +2193,7 @@ hipError_t ytql_launch_scan_generic(const DevPlan* p, const DevSeg* segs,
     int grid = (int)(want > 4096 ? 4096 : (want > 0 ? want : 1));
     hipLaunchKernelGGL(k_scan_generic, dim3(grid), dim3(block), 0, st,
                        *p, segs, segex, col_seg_off, col_seg_cnt, row_count,
-                       th, slots, error_out);
+                       *jd, th, slots, error_out);
     return hipGetLastError();
 }
 
