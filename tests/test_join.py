@@ -103,7 +103,9 @@ def test_join_scan_project_and_topk(cuda):
     pk, v, fkey, fval, _, chunk, fchunk, j = _mk(rng, n=200_000)
     # scan+project with joined column
     plan = y.Plan(projects=[y.col(0), y.col(2)], join=j)
-    got, want = _both(plan, chunk, fchunk, cuda)
+    got, _ = y.gpu_execute(plan, chunk.c_device(cuda), out_capacity=250_000,
+                           join_foreign=fchunk.c_device(cuda))
+    want, _ = y.oracle_execute(plan, chunk)
     assert got == want
     # ORDER BY the joined column (k-selection path)
     plan = y.Plan(projects=[y.col(0), y.col(2), y.col(1)],

@@ -1223,8 +1223,15 @@ static int emit_rows(const YtPlan* plan, const YtChunk* chunk,
                      YtRowset* output, char* errbuf, size_t errlen)
 {
     uint8_t col_types[kMaxCols];
-    for (int c = 0; c < chunk->column_count && c < kMaxCols; c++)
-        col_types[c] = (uint8_t)chunk->columns[c].value_type;
+    memset(col_types, YT_VT_INT64, sizeof(col_types));
+    if (dp) {
+        /* dp carries the EXTENDED column space (primary + joined foreign) */
+        for (int c = 0; c < dp->ncols && c < kMaxCols; c++)
+            col_types[c] = dp->col_types[c];
+    } else {
+        for (int c = 0; c < chunk->column_count && c < kMaxCols; c++)
+            col_types[c] = (uint8_t)chunk->columns[c].value_type;
+    }
 
     const int kp = dp ? dp->kp_count : 0;
     uint8_t key_type = YT_VT_INT64;
