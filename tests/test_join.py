@@ -107,11 +107,21 @@ def test_join_scan_project_and_topk(cuda):
                            join_foreign=fchunk.c_device(cuda))
     want, _ = y.oracle_execute(plan, chunk)
     assert got == want
-    # ORDER BY the joined column (k-selection path)
+    # ORDER BY the joined column (k-selection path); ~333 primary rows share
+    # each foreign value, so boundary ties are arbitrary: compare the ordered
+    # key sequence and that every returned row exists in the joined data
     plan = y.Plan(projects=[y.col(0), y.col(2), y.col(1)],
                   order_by=[(1, True)], limit=50, join=j)
     got, want = _both(plan, chunk, fchunk, cuda)
-    assert got == want
+    assert [r[1] for r in got] == [r[1] for r in want]
+    fmap = dict(zip(fkey.tolist(), fval.tolist()))
+    import collections
+    src = collections.Counter(
+        (int(pk[i]), int(fmap[int(pk[i])]), int(v[i]))
+        for i in range(len(pk)) if int(pk[i]) in fmap)
+    for r in got:
+        assert src[tuple(r)] > 0
+        src[tuple(r)] -= 1
 
 
 @pytest.mark.gpu
