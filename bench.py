@@ -169,14 +169,16 @@ def cpu_baseline_leg(plan_f, enc_cols, n, cores, out_cap, pool_bytes=0):
         sample_rows = rows
     ch = YtChunk(row_count=sample_rows, column_count=len(enc_cols), columns=cols)
 
-    from ytsaurus_amd.api import _mk_rowset
+    from ytsaurus_amd.api import _mk_rowset, _attach_join
     from ytsaurus_amd import _abi
     rs = _mk_rowset(out_cap, 4, pool_bytes=pool_bytes)
     st = _abi.YtStatistics()
     err = ctypes.create_string_buffer(256)
+    plan_obj = plan_f()
+    _j = _attach_join(plan_obj, lambda c: c.c_host())
     t0 = time.monotonic()
     rc = _abi.oracle_lib().yto_execute(
-        ctypes.byref(plan_f().c), ctypes.byref(ch), ctypes.byref(rs),
+        ctypes.byref(plan_obj.c), ctypes.byref(ch), ctypes.byref(rs),
         ctypes.byref(st), cores, err, 256)
     dt = time.monotonic() - t0
     if rc != 0:
@@ -201,7 +203,8 @@ def main():
                     help="rows per GPU (weak scaling)")
     ap.add_argument("--keys", type=int, default=KEY_SPACE)
     ap.add_argument("--workload", default="groupby",
-                    choices=["groupby", "scanfilter", "strgroup", "topk"])
+                    choices=["groupby", "scanfilter", "strgroup", "topk",
+                             "join"])
     ap.add_argument("--limit", type=int, default=1000,
                     help="topk workload: ORDER BY ... LIMIT n")
     ap.add_argument("--no-cpu-baseline", action="store_true")
@@ -238,14 +241,20 @@ def main():
             # shape): histogram k-selection, no scan materialization
             return y.Plan(projects=[y.col(0), y.col(1)],
                           order_by=[(0, False)], limit=args.limit)
+        if args.workload == "join":
+            # fact JOIN dim (unique 1M-key dimension) + GROUP BY fact key:
+            # generic-path probe per row (MultiJoinOpHelper slice)
+            return y.Plan(keys=[y.col(0)],
+                          aggs=[y.agg_sum(y.col(2)), y.agg_sum1()],
+                          join=main.join_spec)
         lo, hi = int(0.25 * 2**VAL_BITS), int(0.75 * 2**VAL_BITS)
         return y.Plan(filter=(y.col(0) >= lo).and_(y.col(0) <= hi),
                       aggs=[y.agg_sum(y.col(1)), y.agg_sum(y.col(2)),
                             y.agg_sum(y.col(3)), y.agg_sum1()])
 
     ncols = 4 if args.workload == "scanfilter" else 2
-    if args.workload == "topk":
-        assert world == 1, "topk: single-GPU this round (no top-K exchange yet)"
+    if args.workload in ("topk", "join"):
+        assert world == 1, "%s: single-GPU this round" % args.workload
     cores = os.cpu_count() or 8
 
     if args.workload == "strgroup":
@@ -257,6 +266,16 @@ def main():
         key_space = ((n + SLICE - 1) // SLICE) * STR_WINDOW
 
     t0 = time.monotonic()
+    main.join_spec = None
+    if args.workload == "join":
+        # dimension: 1M unique keys (permuted), one value column
+        rng = np.random.default_rng([SEED, 999])
+        fn = key_space
+        fkey = rng.permutation(np.arange(fn, dtype=np.int64))
+        fval = rng.integers(0, 2**VAL_BITS, fn, dtype=np.int64)
+        fchunk = y.Chunk([y.encode_int64(fkey), y.encode_int64(fval)], fn)
+        main.join_chunk = fchunk
+        main.join_spec = y.Join(fchunk, 0, 0, [1])
     if args.workload == "strgroup":
         enc_cols = [gen_encode_strkey_column(n, SEED + rank, cores),
                     gen_encode_double_column(n, SEED + rank, cores)]
@@ -278,8 +297,11 @@ def main():
     log("rank %d: uploaded %.2f GB encoded in %.1fs (%.1f GB/s PCIe-inclusive)"
         % (rank, enc_bytes / 1e9, upload_s, enc_bytes / 1e9 / max(upload_s, 1e-9)))
 
+    main.join_dev = None
+    if args.workload == "join":
+        main.join_dev = main.join_chunk.c_device(torch)
     plan = make_plan()
-    hint = key_space if args.workload == "groupby" else 0
+    hint = key_space if args.workload in ("groupby", "join") else 0
     pool_b = (key_space * 10 + (1 << 20)) if args.workload == "strgroup" else 0
     out_cap = (args.limit + 64) if args.workload == "topk" else key_space + 4096
     out_ncols = 2 if args.workload == "topk" else 1 + len(plan.aggs)
@@ -298,7 +320,8 @@ def main():
         nonlocal scan_ms_total, scan_launches
         if dist is None:
             _, st = y.gpu_execute(plan, dev_chunk, max_groups_hint=hint,
-                                  rowset=out_rs, raw_rowset=True)
+                                  rowset=out_rs, raw_rowset=True,
+                                  join_foreign=main.join_dev)
             scan_ms_total += st.kernel_scan_ms
             scan_launches += st.kernel_scan_launches
             step.last = st
@@ -411,6 +434,7 @@ def main():
                     "strgroup": "strgroup_dict_%ddistinct@%drows"
                                 % (key_space, n),
                     "topk": "orderby_limit%d@%drows" % (args.limit, n),
+                    "join": "join_dim%d_groupby@%drows" % (key_space, n),
                 }[args.workload],
                 "rows_per_gpu": n,
                 "distinct_keys": key_space,
