@@ -282,7 +282,7 @@ def main():
     else:
         enc_cols = []
         for ci in range(ncols):
-            if args.workload == "groupby" and ci == 0:
+            if args.workload in ("groupby", "join") and ci == 0:
                 lo, hi = 0, key_space
             else:
                 lo, hi = 0, 2**VAL_BITS
