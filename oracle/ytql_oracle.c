@@ -1323,6 +1323,10 @@ int yto_partial(const YtPlan* plan, const YtChunk* chunk,
                 int64_t* part_counts,
                 int nthreads, char* errbuf, size_t errlen)
 {
+    if (plan->join) {
+        set_err(errbuf, errlen, "partial: join at the bottom query not this round");
+        return YT_ERR_UNSUPPORTED;
+    }
     if (plan->key_count != 1) { set_err(errbuf, errlen, "partial: need 1 key"); return YT_ERR_UNSUPPORTED; }
     int sum_idx = -1;
     for (int a = 0; a < plan->agg_count; a++) {
