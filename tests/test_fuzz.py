@@ -1,0 +1,177 @@
+"""Deterministic plan fuzzer: random-but-seeded plans over random chunks,
+GPU vs oracle. Covers the cross-product the directed tests cannot: random
+expression trees (arith/compare/logic with literals and nulls), 0-3 group
+keys, 1-4 aggregates (sum/sum1/min/max), optional ORDER BY/LIMIT/OFFSET,
+HAVING, WITH TOTALS (both modes), equi-JOIN (inner/left), null densities,
+and all segment encodings (segment size + value locality vary so the
+writer's min-size rule picks different formats). Rows compare sorted
+(hash-order output; OrderedResultMatcher-style), ordered plans compare the
+order-key sequence exactly and the rest as a multiset.
+"""
+import numpy as np
+import pytest
+
+import ytsaurus_amd as y
+from ytsaurus_amd._abi import AGG_MIN, AGG_MAX
+
+N_CASES = 24
+
+
+def _rand_expr(rng, ncols, depth=0):
+    r = rng.random()
+    if depth >= 3 or r < 0.35:
+        c = rng.random()
+        if c < 0.7:
+            return y.col(int(rng.integers(0, ncols)))
+        if c < 0.9:
+            return y.lit(int(rng.integers(-1000, 1000)))
+        return y.null()
+    a = _rand_expr(rng, ncols, depth + 1)
+    b = _rand_expr(rng, ncols, depth + 1)
+    op = rng.integers(0, 8)
+    if op == 0:
+        return a + b
+    if op == 1:
+        return a - b
+    if op == 2:
+        return a * y.lit(int(rng.integers(-3, 4)))
+    if op == 3:
+        return a < b
+    if op == 4:
+        return a >= b
+    if op == 5:
+        return (a == b)
+    if op == 6:
+        return (a < y.lit(0)).and_(b >= y.lit(-100))
+    return (a > y.lit(50)).or_(b == y.lit(0))
+
+
+def _rand_chunk(rng, n, ncols):
+    cols = []
+    arrays = []
+    seg = int(rng.choice([0, 1 << 10, 1 << 13, 100_000]))
+    for c in range(ncols):
+        style = rng.integers(0, 4)
+        if style == 0:       # narrow repeated -> dictionary-ish
+            v = rng.integers(-20, 20, n, dtype=np.int64)
+        elif style == 1:     # runs -> RLE-ish
+            v = np.repeat(rng.integers(-10**6, 10**6,
+                                       max(n // 50, 1), dtype=np.int64), 50)[:n]
+            if len(v) < n:
+                v = np.pad(v, (0, n - len(v)), constant_values=7)
+        elif style == 2:     # wide direct
+            v = rng.integers(-2**45, 2**45, n, dtype=np.int64)
+        else:
+            v = rng.integers(0, 1000, n, dtype=np.int64)
+        nulls = None
+        if rng.random() < 0.5:
+            nulls = (rng.random(n) < rng.choice([0.01, 0.2])).astype(np.uint8)
+        cols.append(y.encode_int64(v, nulls, max_segment_values=seg))
+        arrays.append((v, nulls))
+    return y.Chunk(cols, n), arrays
+
+
+def _rand_plan(rng, ncols, with_join_cols=0):
+    nc = ncols + with_join_cols
+    filt = _rand_expr(rng, nc) if rng.random() < 0.5 else None
+    kc = int(rng.integers(0, 4))
+    keys = []
+    if kc:
+        keycols = rng.choice(nc, size=min(kc, nc), replace=False)
+        keys = [y.col(int(c)) for c in keycols]
+    aggs = []
+    if kc or rng.random() < 0.7:
+        na = int(rng.integers(1, 4))
+        for _ in range(na):
+            f = rng.integers(0, 4)
+            c = y.col(int(rng.integers(0, nc)))
+            if f == 0:
+                aggs.append(y.agg_sum(c))
+            elif f == 1:
+                aggs.append(y.agg_sum1())
+            elif f == 2:
+                aggs.append((AGG_MIN, c))
+            else:
+                aggs.append((AGG_MAX, c))
+    projects = []
+    if not keys and not aggs:
+        projects = [_rand_expr(rng, nc) for _ in range(int(rng.integers(1, 4)))]
+    out_cols = len(projects) if projects else len(keys) + len(aggs)
+    order_by, limit, offset = (), 0, 0
+    if out_cols and rng.random() < 0.5:
+        order_by = [(int(rng.integers(0, out_cols)), bool(rng.integers(0, 2)))]
+        limit = int(rng.integers(1, 200))
+        offset = int(rng.integers(0, 5))
+    having, totals, after = None, False, False
+    if keys and aggs and not projects:
+        if rng.random() < 0.4:
+            having = _rand_expr(rng, out_cols)
+        if rng.random() < 0.4:
+            totals = True
+            after = bool(rng.integers(0, 2))
+    return dict(filter=filt, keys=keys, aggs=aggs, projects=projects,
+                order_by=order_by, limit=limit, offset=offset,
+                having=having, with_totals=totals, totals_after_having=after)
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("seed", range(N_CASES))
+def test_fuzz_parity(cuda, seed):
+    rng = np.random.default_rng([20260915, seed])
+    n = int(rng.choice([97, 5_000, 120_000]))
+    ncols = int(rng.integers(2, 5))
+    chunk, _ = _rand_chunk(rng, n, ncols)
+
+    join = None
+    jdev = None
+    jcols = 0
+    if rng.random() < 0.3:
+        fn = int(rng.integers(1, 500))
+        fkey = rng.permutation(np.arange(fn, dtype=np.int64))
+        fval = rng.integers(-10**6, 10**6, fn, dtype=np.int64)
+        fchunk = y.Chunk([y.encode_int64(fkey), y.encode_int64(fval)], fn)
+        join = y.Join(fchunk, int(rng.integers(0, ncols)), 0, [1],
+                      is_left=bool(rng.integers(0, 2)))
+        jdev = fchunk.c_device(cuda)
+        jcols = 1
+
+    kw = _rand_plan(rng, ncols, with_join_cols=jcols)
+    kw["join"] = join
+    plan = y.Plan(**kw)
+
+    def run_gpu():
+        return y.gpu_execute(plan, chunk.c_device(cuda),
+                             max_groups_hint=1 << 15,
+                             out_capacity=max(n + 1024, 1 << 16),
+                             join_foreign=jdev)
+
+    try:
+        want, _ = y.oracle_execute(plan, chunk, nthreads=2)
+        oracle_err = None
+    except RuntimeError as e:
+        want, oracle_err = None, str(e)
+
+    if oracle_err is not None:
+        # oracle rejected the plan (div-zero, NaN order key, ...): the GPU
+        # must error too (any message; same rc family not required)
+        with pytest.raises(RuntimeError):
+            run_gpu()
+        return
+
+    try:
+        got, _ = run_gpu()
+    except RuntimeError as e:
+        msg = str(e)
+        # the GPU path may refuse shapes the oracle covers — but only with
+        # a loud, known UNSUPPORTED reason, never a wrong answer
+        assert ("this round" in msg or "62 bits" in msg
+                or "at most" in msg), msg
+        return
+
+    if kw["order_by"]:
+        oc = kw["order_by"][0][0]
+        assert [r[oc] for r in got] == [r[oc] for r in want]
+        # boundary ties can differ row-wise; lengths must match
+        assert len(got) == len(want)
+    else:
+        assert y.sort_rows(got) == y.sort_rows(want)
