@@ -55,6 +55,16 @@ hipError_t ytql_launch_topk_materialize(const DevPlan*, const DevSeg*, const Seg
                                         const JoinDev*,
                                         const int64_t*, int64_t, DevOutVal*,
                                         unsigned*, hipStream_t);
+hipError_t ytql_launch_topk_hist_fast(const DevSeg*, const SegEx*, int, int,
+                                      int64_t, int, int, const TopkPass*,
+                                      unsigned long long*, unsigned long long*,
+                                      hipStream_t);
+hipError_t ytql_launch_topk_gather_fast(const DevSeg*, const SegEx*, int, int,
+                                        int64_t, int, int, const TopkGather*,
+                                        int64_t*, unsigned long long*,
+                                        int64_t*, unsigned long long*,
+                                        int64_t*, unsigned long long*,
+                                        hipStream_t);
 hipError_t ytql_launch_join_build(const JoinDev*, int64_t, uint64_t*,
                                   long long*, unsigned long long*, unsigned*,
                                   hipStream_t);
@@ -1738,6 +1748,34 @@ static int run_scan_topk(const YtPlan* plan, const YtChunk* chunk,
     const int desc0 = plan->order_desc ? plan->order_desc[0] : 0;
     const int64_t CAPB = (int64_t)1 << 17;
 
+    /* fast selection passes: plain DirectDense int column key, no filter,
+     * no join, uniform segmentation (k_topk_*_fast — 4 rows in flight per
+     * thread, no interpreter) */
+    int fast_col = -1, fast_signed = 0, fast_nulls = 0, fast_shift = 0;
+    {
+        int kcol;
+        if (!plan->filter && !plan->join &&
+            expr_is_col(plan->projects[ord0], &kcol) &&
+            kcol < chunk->column_count &&
+            (chunk->columns[kcol].value_type == YT_VT_INT64 ||
+             chunk->columns[kcol].value_type == YT_VT_UINT64) &&
+            dp->col_uniform_shift[kcol] != 0) {
+            bool all_dd = true;
+            for (int i = 0; i < chunk->columns[kcol].segment_count; i++) {
+                if (chunk->columns[kcol].segments[i].type != YT_SEG_DIRECT_DENSE) {
+                    all_dd = false;
+                    break;
+                }
+            }
+            if (all_dd) {
+                fast_col = kcol;
+                fast_signed = chunk->columns[kcol].value_type == YT_VT_INT64;
+                fast_nulls = R2.col_null_flags[kcol] != 0;
+                fast_shift = dp->col_uniform_shift[kcol];
+            }
+        }
+    }
+
     {
         unsigned long long* d_bins = nullptr;   /* 2048 bins + misc(3) + ctrs(3) */
         HIP_CHECK(pool_alloc(&d_bins, sizeof(unsigned long long) * (2048 + 8)));
@@ -1776,9 +1814,16 @@ static int run_scan_topk(const YtPlan* plan, const YtChunk* chunk,
                 HIP_CHECK(hipMemsetAsync(d_misc + 2, 0,
                                          sizeof(unsigned long long), R2.stream));
             }
-            HIP_CHECK(ytql_launch_topk_hist(dp, R2.d_segs, R2.d_segex, R2.d_off,
-                                            R2.d_cnt, n, jd, &tp, d_bins, d_misc,
-                                            R2.d_err, R2.stream));
+            if (fast_col >= 0) {
+                HIP_CHECK(ytql_launch_topk_hist_fast(
+                    R2.d_segs, R2.d_segex, R2.h_off[fast_col], fast_shift, n,
+                    fast_nulls, fast_signed, &tp, d_bins, d_misc, R2.stream));
+            } else {
+                HIP_CHECK(ytql_launch_topk_hist(dp, R2.d_segs, R2.d_segex,
+                                                R2.d_off, R2.d_cnt, n, jd, &tp,
+                                                d_bins, d_misc,
+                                                R2.d_err, R2.stream));
+            }
             launches++;
             HIP_CHECK(hipMemcpy(h_bins.data(), d_bins,
                                 sizeof(unsigned long long) * 2051,
@@ -1884,12 +1929,20 @@ static int run_scan_topk(const YtPlan* plan, const YtChunk* chunk,
         HIP_CHECK(pool_alloc(&d_rows_null, sizeof(int64_t) * CAPN));
         HIP_CHECK(hipMemsetAsync(d_ctrs, 0, 3 * sizeof(unsigned long long),
                                  R2.stream));
-        HIP_CHECK(ytql_launch_topk_gather(dp, R2.d_segs, R2.d_segex, R2.d_off,
-                                          R2.d_cnt, n, jd, &tg,
-                                          d_rows_strict, d_ctrs,
-                                          d_rows_tie, d_ctrs + 1,
-                                          d_rows_null, d_ctrs + 2,
-                                          R2.d_err, R2.stream));
+        if (fast_col >= 0) {
+            HIP_CHECK(ytql_launch_topk_gather_fast(
+                R2.d_segs, R2.d_segex, R2.h_off[fast_col], fast_shift, n,
+                fast_nulls, fast_signed, &tg,
+                d_rows_strict, d_ctrs, d_rows_tie, d_ctrs + 1,
+                d_rows_null, d_ctrs + 2, R2.stream));
+        } else {
+            HIP_CHECK(ytql_launch_topk_gather(dp, R2.d_segs, R2.d_segex,
+                                              R2.d_off, R2.d_cnt, n, jd, &tg,
+                                              d_rows_strict, d_ctrs,
+                                              d_rows_tie, d_ctrs + 1,
+                                              d_rows_null, d_ctrs + 2,
+                                              R2.d_err, R2.stream));
+        }
         launches++;
         unsigned long long hctrs[3];
         HIP_CHECK(hipMemcpy(hctrs, d_ctrs, 3 * sizeof(unsigned long long),
