@@ -164,3 +164,36 @@ def test_boolean_roundtrip():
     assert rc == 0
     np.testing.assert_array_equal(gn, nulls)
     np.testing.assert_array_equal(got[nulls == 0], vals[nulls == 0])
+
+
+def test_encoder_fuzz_roundtrip():
+    """Seeded sweep over widths 1..64, null densities, run/dict shapes and
+    segment splits — every encode must decode back exactly (the writer's
+    min-size rule picks whatever format it likes; the decode contract is
+    format-independent)."""
+    rng = np.random.default_rng(99)
+    for case in range(30):
+        n = int(rng.choice([1, 63, 64, 1000, 5000]))
+        bits = int(rng.integers(1, 63))
+        style = rng.integers(0, 4)
+        if style == 0:
+            vals = rng.integers(-(1 << bits), 1 << bits, n, dtype=np.int64)
+        elif style == 1:                     # few distinct (dictionary)
+            pool = rng.integers(-(1 << bits), 1 << bits, 5, dtype=np.int64)
+            vals = pool[rng.integers(0, 5, n)]
+        elif style == 2:                     # runs (RLE)
+            reps = max(n // 20, 1)
+            vals = np.repeat(
+                rng.integers(-(1 << bits), 1 << bits, reps, dtype=np.int64),
+                20)[:n]
+            if len(vals) < n:
+                vals = np.pad(vals, (0, n - len(vals)), constant_values=3)
+        else:                                # constant
+            vals = np.full(n, int(rng.integers(-(1 << bits), 1 << bits)),
+                           dtype=np.int64)
+        nulls = None
+        if rng.random() < 0.6:
+            nulls = (rng.random(n) < rng.choice([0.0, 0.05, 0.5, 1.0])
+                     ).astype(np.uint8)
+        seg = int(rng.choice([0, 100, 1024]))
+        check_roundtrip(vals, nulls, max_seg=seg)

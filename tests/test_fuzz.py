@@ -14,7 +14,7 @@ import pytest
 import ytsaurus_amd as y
 from ytsaurus_amd._abi import AGG_MIN, AGG_MAX
 
-N_CASES = 24
+N_CASES = 40
 
 
 def _rand_expr(rng, ncols, depth=0):
