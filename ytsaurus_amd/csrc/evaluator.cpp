@@ -1008,14 +1008,14 @@ static int run_partitioned(const YtPlan* plan, const YtChunk* chunk,
             *done = true;
         }
     }
-    hipEventDestroy(ev0);
-    hipEventDestroy(ev1);
-    hipEventDestroy(ev2);
+    (void)hipEventDestroy(ev0);
+    (void)hipEventDestroy(ev1);
+    (void)hipEventDestroy(ev2);
     return YT_OK;
 fail:
-    if (ev0) hipEventDestroy(ev0);
-    if (ev1) hipEventDestroy(ev1);
-    if (ev2) hipEventDestroy(ev2);
+    if (ev0) (void)hipEventDestroy(ev0);
+    if (ev1) (void)hipEventDestroy(ev1);
+    if (ev2) (void)hipEventDestroy(ev2);
     return rc;
 }
 
@@ -1139,12 +1139,12 @@ static int run_scan(const YtPlan* plan, const YtChunk* chunk,
             goto fail;
         }
     }
-    hipEventDestroy(ev0);
-    hipEventDestroy(ev1);
+    (void)hipEventDestroy(ev0);
+    (void)hipEventDestroy(ev1);
     return YT_OK;
 fail:
-    if (ev0) hipEventDestroy(ev0);
-    if (ev1) hipEventDestroy(ev1);
+    if (ev0) (void)hipEventDestroy(ev0);
+    if (ev1) (void)hipEventDestroy(ev1);
     return rc;
 }
 
@@ -1982,8 +1982,8 @@ static int run_scan_topk(const YtPlan* plan, const YtChunk* chunk,
         if (stats) stats->kernel_other_ms = now_ms() - tw0;  /* pre-emit wall */
         float ms = 0;
         HIP_CHECK(hipEventElapsedTime(&ms, e0, e1));
-        hipEventDestroy(e0);
-        hipEventDestroy(e1);
+        (void)hipEventDestroy(e0);
+        (void)hipEventDestroy(e1);
         unsigned kerr = 0;
         HIP_CHECK(hipMemcpy(&kerr, R2.d_err, sizeof(unsigned),
                             hipMemcpyDeviceToHost));
@@ -2130,8 +2130,8 @@ static int run_scan_project(const YtPlan* plan, const YtChunk* chunk,
         HIP_CHECK(hipStreamSynchronize(R2.stream));
         float ms = 0;
         HIP_CHECK(hipEventElapsedTime(&ms, e0, e1));
-        hipEventDestroy(e0);
-        hipEventDestroy(e1);
+        (void)hipEventDestroy(e0);
+        (void)hipEventDestroy(e1);
         unsigned kerr = 0;
         HIP_CHECK(hipMemcpy(&kerr, R2.d_err, sizeof(unsigned), hipMemcpyDeviceToHost));
         if (kerr) {
@@ -2372,9 +2372,9 @@ static int run_string_group(const YtPlan* plan, const YtChunk* chunk,
         float ms = 0, ms_other = 0;
         HIP_CHECK(hipEventElapsedTime(&ms, e0, e1));
         HIP_CHECK(hipEventElapsedTime(&ms_other, e1, e2));
-        hipEventDestroy(e0);
-        hipEventDestroy(e1);
-        hipEventDestroy(e2);
+        (void)hipEventDestroy(e0);
+        (void)hipEventDestroy(e1);
+        (void)hipEventDestroy(e2);
         HIP_CHECK(hipMemcpy(&th, R.d_th, sizeof(th), hipMemcpyDeviceToHost));
         if (th.overflow == 1) {
             g_pool.put(d_accbase); g_pool.put(d_acc); g_pool.put(d_hashes);
