@@ -1423,9 +1423,17 @@ int yto_merge(const YtPlan* plan, const YtStateRow* states, int64_t nstates,
         }
     }
     output->row_count = 0;
+    output->totals_row = 0;
     int rc = YT_OK;
     for (int64_t g = 0; g < t.ngroups && rc == YT_OK; g++) {
         rc = emit_group_row(plan, &t, g, output);
+    }
+    /* ORDER BY / WITH TOTALS / HAVING apply at the front (coordinator)
+     * query, exactly like the GPU merge path */
+    if (rc == YT_OK && (plan->order_count > 0 || plan->with_totals ||
+                        plan->having)) {
+        output->column_count = 1 + plan->agg_count;
+        rc = finish_group_rowset(plan, output, errbuf, errlen);
     }
     gt_free(&t);
     return rc;
