@@ -244,6 +244,69 @@ typedef struct YtRowset {
     int32_t pad_;
 } YtRowset;
 
+/* ---- versioned scan-format slice ----
+ * The MVCC "scan format" (ytlib/columnar_chunk_format) for flat tables:
+ * a timestamp segment (table_chunk_format/timestamp_writer.cpp DumpSegment:
+ * blob = five header-carrying bit-packed vectors
+ * [ts dictionary − base][write ts ids][delete ts ids]
+ * [cumulative write counts, diff-from-expected][cumulative delete counts,
+ * diff-from-expected]; meta = TTimestampMeta base/expected fields,
+ * prepared_meta.h:40-60) and versioned value segments (DirectDense int64
+ * this round: integer_column_writer.cpp:119-246 + column_writer_detail.cpp
+ * DumpVersionedData: blob = [cumulative values-per-row diffs][per-value
+ * timestamp index][values (zigzag − base)][null bitmap]; values correspond
+ * 1:1 with the row's write timestamps in this slice). Timestamps within a
+ * row are strictly descending (rowset_builder.cpp:1052-1060 asserts).
+ * Read-at-timestamp semantics (rowset_builder.cpp:1042-1166
+ * TRowAllocatorBase::DoAllocateRow, produceAll = false): latest write ≤ T
+ * that is newer than the latest delete ≤ T; no such write → the row is not
+ * visible at T. */
+typedef struct YtTimestampSeg {
+    int64_t row_count;
+    uint64_t base_timestamp;
+    uint32_t expected_writes_per_row;
+    uint32_t expected_deletes_per_row;
+    const void* data;
+    int64_t data_size;
+} YtTimestampSeg;
+
+typedef struct YtVersionedValueSeg {
+    int64_t row_count;
+    uint64_t base_value;              /* zigzag-space min over non-null values */
+    uint32_t expected_values_per_row;
+    uint32_t pad_;
+    const void* data;
+    int64_t data_size;
+} YtVersionedValueSeg;
+
+typedef struct YtVersionedColumn {
+    int32_t ts_seg_count;
+    int32_t val_seg_count;            /* == ts_seg_count, same row split */
+    const YtTimestampSeg* ts_segs;
+    const YtVersionedValueSeg* val_segs;
+} YtVersionedColumn;
+
+/* Encode one versioned int64 column + its timestamp column from flattened
+ * per-row write/delete lists (write_ts DESC within each row; values[i]
+ * belongs to the i-th write of its row). Synthetic-data generator, mirrors
+ * the reference writers byte-for-byte. Caller frees with
+ * yt_versioned_free. */
+int yt_encode_versioned_int64(
+    const uint32_t* writes_per_row, const uint64_t* write_ts,
+    const int64_t* values, const uint8_t* value_nulls,
+    const uint32_t* deletes_per_row, const uint64_t* delete_ts,
+    int64_t row_count, int64_t max_rows_per_segment,
+    YtVersionedColumn* out, char* errbuf, size_t errlen);
+void yt_versioned_free(YtVersionedColumn* col);
+
+/* Read the column as of timestamp T on the GPU: out_visible[r] = 1 iff the
+ * row has a visible write at T; out_bits/out_null = the column value of the
+ * visible version. All out pointers are DEVICE memory sized row_count. */
+int yt_gpu_versioned_read(
+    const YtVersionedColumn* col, uint64_t timestamp,
+    uint64_t* out_bits, uint8_t* out_null, uint8_t* out_visible,
+    uint64_t stream, char* errbuf, size_t errlen);
+
 /* =========================== entry points =========================== */
 
 /* Library/device probe. Returns YT_OK when a gfx950 HIP device is usable. */

@@ -1585,3 +1585,103 @@ int yto_decode_string_column(const YtColumn* col, int64_t row_count,
     }
     return (row == row_count) ? YT_OK : YT_ERR_INVALID_CHUNK;
 }
+
+/* ------------------------------------------------------------------ */
+/* versioned scan-format slice: restated reader.
+ * Layouts: include/ytql_gpu.h citations (timestamp_writer.cpp DumpSegment,
+ * column_writer_detail.cpp DumpVersionedData, integer_column_writer.cpp
+ * DumpDirectValues). Read-at-timestamp rule restates
+ * rowset_builder.cpp:1042-1166 (TRowAllocatorBase::DoAllocateRow,
+ * produceAll = false): per row, timestamps DESC; lower = first ts <= T;
+ * deleteTs = latest delete <= T (0 = none); visible writes are those <= T
+ * and > deleteTs; the column value is the first value (values sorted by
+ * ascending timestamp index) whose index lies in that range. */
+
+static uint64_t vcum_at(const BitReader* diffs, uint32_t expected, int64_t i)
+{
+    if (i < 0) return 0;
+    return (uint64_t)expected * (uint64_t)(i + 1)
+         + (uint64_t)(int64_t)zigzag_decode32((uint32_t)bitreader_get(diffs, i));
+}
+
+ORACLE_EXPORT
+int yto_versioned_read(const YtVersionedColumn* col, uint64_t timestamp,
+                       uint64_t* out_bits, uint8_t* out_null,
+                       uint8_t* out_visible, char* errbuf, size_t errlen)
+{
+    int64_t base_row = 0;
+    for (int si = 0; si < col->ts_seg_count; si++) {
+        const YtTimestampSeg* T = &col->ts_segs[si];
+        const YtVersionedValueSeg* V = &col->val_segs[si];
+        if (V->row_count != T->row_count) {
+            set_err(errbuf, errlen, "versioned: segment row mismatch");
+            return YT_ERR_INVALID_CHUNK;
+        }
+        const char* tp = (const char*)T->data;
+        BitReader dict = bitreader_init(tp);
+        tp += bitreader_byte_size(&dict);
+        BitReader wids = bitreader_init(tp);
+        tp += bitreader_byte_size(&wids);
+        BitReader dids = bitreader_init(tp);
+        tp += bitreader_byte_size(&dids);
+        BitReader wdiffs = bitreader_init(tp);
+        tp += bitreader_byte_size(&wdiffs);
+        BitReader ddiffs = bitreader_init(tp);
+
+        const char* vp = (const char*)V->data;
+        BitReader voffs = bitreader_init(vp);
+        vp += bitreader_byte_size(&voffs);
+        BitReader tsids = bitreader_init(vp);
+        vp += bitreader_byte_size(&tsids);
+        BitReader vvals = bitreader_init(vp);
+        vp += bitreader_byte_size(&vvals);
+        const uint8_t* vnull = (const uint8_t*)vp;
+
+        for (int64_t r = 0; r < T->row_count; r++) {
+            int64_t g = base_row + r;
+            out_visible[g] = 0;
+            out_null[g] = 1;
+            out_bits[g] = 0;
+            uint64_t wb = vcum_at(&wdiffs, T->expected_writes_per_row, r - 1);
+            uint64_t we = vcum_at(&wdiffs, T->expected_writes_per_row, r);
+            uint64_t db = vcum_at(&ddiffs, T->expected_deletes_per_row, r - 1);
+            uint64_t de = vcum_at(&ddiffs, T->expected_deletes_per_row, r);
+
+            /* latest delete <= T (lists are DESC: scan past those > T) */
+            uint64_t delete_ts = 0;
+            for (uint64_t i = db; i < de; i++) {
+                uint64_t ts = T->base_timestamp + bitreader_get(&dict, bitreader_get(&dids, i));
+                if (ts <= timestamp) { delete_ts = ts; break; }
+            }
+            /* lower/upper write indexes within the row (0-based in-row) */
+            int64_t wcount = (int64_t)(we - wb);
+            int64_t lower = wcount, upper = wcount;
+            for (int64_t i = 0; i < wcount; i++) {
+                uint64_t ts = T->base_timestamp + bitreader_get(&dict, bitreader_get(&wids, wb + i));
+                if (lower == wcount && ts <= timestamp) lower = i;
+                if (ts <= delete_ts) { upper = i; break; }
+            }
+            if (upper < lower) upper = lower;
+            if (lower >= upper) continue;     /* no visible write: row absent */
+            out_visible[g] = 1;
+
+            /* first value with in-row timestamp index in [lower, upper) */
+            uint64_t vb = vcum_at(&voffs, V->expected_values_per_row, r - 1);
+            uint64_t ve = vcum_at(&voffs, V->expected_values_per_row, r);
+            for (uint64_t j = vb; j < ve; j++) {
+                uint64_t ti = bitreader_get(&tsids, j);
+                if ((int64_t)ti < lower) continue;
+                if ((int64_t)ti >= upper) break;
+                int nul = (vnull[j / 8] >> (j % 8)) & 1;
+                if (!nul) {
+                    out_null[g] = 0;
+                    out_bits[g] = (uint64_t)zigzag_decode64(
+                        V->base_value + bitreader_get(&vvals, j));
+                }
+                break;
+            }
+        }
+        base_row += T->row_count;
+    }
+    return YT_OK;
+}

@@ -142,6 +142,25 @@ class YtRowset(C.Structure):
                 ("totals_row", C.c_int32), ("pad_", C.c_int32)]
 
 
+class YtTimestampSeg(C.Structure):
+    _fields_ = [("row_count", C.c_int64), ("base_timestamp", C.c_uint64),
+                ("expected_writes_per_row", C.c_uint32),
+                ("expected_deletes_per_row", C.c_uint32),
+                ("data", C.c_void_p), ("data_size", C.c_int64)]
+
+
+class YtVersionedValueSeg(C.Structure):
+    _fields_ = [("row_count", C.c_int64), ("base_value", C.c_uint64),
+                ("expected_values_per_row", C.c_uint32), ("pad_", C.c_uint32),
+                ("data", C.c_void_p), ("data_size", C.c_int64)]
+
+
+class YtVersionedColumn(C.Structure):
+    _fields_ = [("ts_seg_count", C.c_int32), ("val_seg_count", C.c_int32),
+                ("ts_segs", C.POINTER(YtTimestampSeg)),
+                ("val_segs", C.POINTER(YtVersionedValueSeg))]
+
+
 class YtStateRow(C.Structure):
     _fields_ = [("key_bits", C.c_uint64), ("meta", C.c_uint64),
                 ("sum_bits", C.c_uint64), ("row_count", C.c_uint64)]

@@ -20,7 +20,7 @@ from ._abi import (
 __all__ = [
     "col", "lit", "null", "litf", "Plan", "Join", "agg_sum", "agg_sum1",
     "encode_int64", "encode_double", "encode_bool", "encode_string", "encode_string_raw",
-    "oracle_decode_strings",
+    "oracle_decode_strings", "encode_versioned_int64", "oracle_versioned_read", "gpu_versioned_read", "VersionedColumn",
     "Chunk", "oracle_execute",
     "oracle_partial", "oracle_merge",
     "gpu_available", "gpu_execute", "gpu_partial", "gpu_merge",
@@ -289,6 +289,88 @@ def encode_string(strings, max_segment_values=0):
         at += len(v)
     blob = b"".join(blobs)
     return encode_string_raw(blob, begins, lens, nulls, max_segment_values)
+
+
+class VersionedColumn:
+    """Owns a YtVersionedColumn (versioned scan-format slice, SURVEY §8f
+    row 3): one int64 value column + its timestamp column."""
+
+    def __init__(self, ccol, row_count):
+        self._c = ccol
+        self.row_count = row_count
+
+    def __del__(self):
+        try:
+            _abi.gpu_lib().yt_versioned_free(C.byref(self._c))
+        except Exception:
+            pass
+
+
+def encode_versioned_int64(writes_per_row, write_ts, values, value_nulls,
+                           deletes_per_row, delete_ts,
+                           max_rows_per_segment=0):
+    """Synthetic versioned-chunk generator (byte-faithful to the reference
+    timestamp/versioned-int writers). write_ts DESC per row; values 1:1
+    with writes."""
+    wpr = np.ascontiguousarray(writes_per_row, dtype=np.uint32)
+    wts = np.ascontiguousarray(write_ts, dtype=np.uint64)
+    vals = np.ascontiguousarray(values, dtype=np.int64)
+    vn = np.ascontiguousarray(value_nulls, dtype=np.uint8) \
+        if value_nulls is not None else None
+    dpr = np.ascontiguousarray(deletes_per_row, dtype=np.uint32)
+    dts = np.ascontiguousarray(delete_ts, dtype=np.uint64)
+    cc = _abi.YtVersionedColumn()
+    err = C.create_string_buffer(256)
+    rc = _abi.gpu_lib().yt_encode_versioned_int64(
+        wpr.ctypes.data_as(C.POINTER(C.c_uint32)),
+        wts.ctypes.data_as(C.POINTER(C.c_uint64)),
+        vals.ctypes.data_as(C.POINTER(C.c_int64)),
+        vn.ctypes.data_as(C.POINTER(C.c_uint8)) if vn is not None else None,
+        dpr.ctypes.data_as(C.POINTER(C.c_uint32)),
+        dts.ctypes.data_as(C.POINTER(C.c_uint64)),
+        C.c_int64(len(wpr)), C.c_int64(max_rows_per_segment),
+        C.byref(cc), err, 256)
+    _check(rc, err)
+    return VersionedColumn(cc, len(wpr))
+
+
+def oracle_versioned_read(vcol, timestamp):
+    """TEST ONLY: read the column as of `timestamp` via the oracle; returns
+    (values list with None for nulls, visible bool list)."""
+    n = vcol.row_count
+    bits = np.zeros(max(n, 1), dtype=np.uint64)
+    nulls = np.zeros(max(n, 1), dtype=np.uint8)
+    vis = np.zeros(max(n, 1), dtype=np.uint8)
+    err = C.create_string_buffer(256)
+    rc = _abi.oracle_lib().yto_versioned_read(
+        C.byref(vcol._c), C.c_uint64(timestamp),
+        bits.ctypes.data_as(C.POINTER(C.c_uint64)),
+        nulls.ctypes.data_as(C.POINTER(C.c_uint8)),
+        vis.ctypes.data_as(C.POINTER(C.c_uint8)), err, 256)
+    _check(rc, err)
+    out = [None if nulls[i] else int(bits[i].astype(np.int64))
+           for i in range(n)]
+    return out, [bool(v) for v in vis[:n]]
+
+
+def gpu_versioned_read(vcol, timestamp, torch_mod):
+    """Read the versioned column as of `timestamp` on the GPU (the §8f
+    row-3 slice). Returns (values list with None, visible bool list)."""
+    n = vcol.row_count
+    bits = torch_mod.zeros(max(n, 1), dtype=torch_mod.int64, device="cuda")
+    nulls = torch_mod.zeros(max(n, 1), dtype=torch_mod.uint8, device="cuda")
+    vis = torch_mod.zeros(max(n, 1), dtype=torch_mod.uint8, device="cuda")
+    err = C.create_string_buffer(256)
+    rc = _abi.gpu_lib().yt_gpu_versioned_read(
+        C.byref(vcol._c), C.c_uint64(timestamp),
+        C.c_void_p(bits.data_ptr()), C.c_void_p(nulls.data_ptr()),
+        C.c_void_p(vis.data_ptr()), C.c_uint64(0), err, 256)
+    _check(rc, err)
+    hb = bits.cpu().numpy()
+    hn = nulls.cpu().numpy()
+    hv = vis.cpu().numpy()
+    out = [None if hn[i] else int(hb[i]) for i in range(n)]
+    return out, [bool(v) for v in hv[:n]]
 
 
 def oracle_decode_strings(enc, n):

@@ -117,6 +117,19 @@ struct JoinDev {
     int64_t frows;
 };
 
+/* versioned scan-format read: per-segment device
+ * descriptor; ts_data/val_data point at device copies of the encoded
+ * segment blobs (layouts in include/ytql_gpu.h) */
+struct VSegDev {
+    int64_t row_start;
+    int64_t row_count;
+    uint64_t base_timestamp;
+    uint32_t exp_w, exp_d, exp_v;
+    uint64_t base_value;
+    const char* ts_data;
+    const char* val_data;
+};
+
 /* device output value for the scan+project path (16 B) */
 struct DevOutVal {
     uint64_t bits;

@@ -623,3 +623,156 @@ extern "C" int yt_encode_string_column(
     flush();
     return finish_column(segs, out, errbuf, errlen);
 }
+
+/* ------------------------------------------------------------------ */
+/* versioned scan-format slice — see include/ytql_gpu.h
+ * for the layout citations (timestamp_writer.cpp DumpSegment,
+ * column_writer_detail.cpp DumpVersionedData,
+ * integer_column_writer.cpp:119-246 DumpDirectValues). */
+
+extern "C" int yt_encode_versioned_int64(
+    const uint32_t* writes_per_row, const uint64_t* write_ts,
+    const int64_t* values, const uint8_t* value_nulls,
+    const uint32_t* deletes_per_row, const uint64_t* delete_ts,
+    int64_t row_count, int64_t max_rows_per_segment,
+    YtVersionedColumn* out, char* errbuf, size_t errlen)
+{
+    if (max_rows_per_segment <= 0) max_rows_per_segment = 128 * 1024;
+    memset(out, 0, sizeof(*out));
+    int nseg = (int)((row_count + max_rows_per_segment - 1) / max_rows_per_segment);
+    if (nseg == 0) return YT_OK;
+
+    auto* tsegs = (YtTimestampSeg*)calloc(nseg, sizeof(YtTimestampSeg));
+    auto* vsegs = (YtVersionedValueSeg*)calloc(nseg, sizeof(YtVersionedValueSeg));
+
+    int64_t wat = 0, dat = 0;   /* global flattened cursors */
+    for (int si = 0; si < nseg; si++) {
+        int64_t r0 = (int64_t)si * max_rows_per_segment;
+        int64_t r1 = r0 + max_rows_per_segment;
+        if (r1 > row_count) r1 = row_count;
+        int64_t rows = r1 - r0;
+
+        /* --- timestamp segment: register per row (writes then deletes,
+         * first-appearance dictionary — timestamp_writer.cpp:45-58,185-196) */
+        std::vector<uint64_t> dict, wids, dids, wcnt, dcnt;
+        std::unordered_map<uint64_t, uint32_t> uniq;
+        uint64_t ts_min = ~0ULL, ts_max = 0;
+        auto reg = [&](uint64_t ts) -> uint32_t {
+            if (ts < ts_min) ts_min = ts;
+            if (ts > ts_max) ts_max = ts;
+            auto it = uniq.emplace(ts, (uint32_t)dict.size());
+            if (it.second) dict.push_back(ts);
+            return it.first->second;
+        };
+        std::vector<uint64_t> vals_zz, tsids, voff;
+        std::vector<uint8_t> vnull;
+        uint64_t vmin = ~0ULL, vmax = 0, max_tsid = 0;
+        for (int64_t r = r0; r < r1; r++) {
+            uint32_t wc = writes_per_row[r];
+            for (uint32_t i = 0; i < wc; i++) {
+                uint64_t ts = write_ts[wat + i];
+                if (i && write_ts[wat + i - 1] <= ts) {
+                    set_err(errbuf, errlen,
+                            "versioned: write timestamps must be strictly descending per row");
+                    free(tsegs); free(vsegs);
+                    return YT_ERR_INVALID_CHUNK;
+                }
+                wids.push_back(reg(ts));
+                /* value slice: 1:1 with writes, tsIndex = position in the
+                 * row's write list */
+                uint8_t nul = value_nulls ? value_nulls[wat + i] : 0;
+                uint64_t zz = 0;
+                if (!nul) {
+                    zz = zigzag_encode64(values[wat + i]);
+                    if (zz < vmin) vmin = zz;
+                    if (zz > vmax) vmax = zz;
+                }
+                vals_zz.push_back(zz);
+                vnull.push_back(nul);
+                tsids.push_back(i);
+                if (i > max_tsid) max_tsid = i;
+            }
+            wat += wc;
+            wcnt.push_back((wcnt.empty() ? 0 : wcnt.back()) + wc);
+            voff.push_back((voff.empty() ? 0 : voff.back()) + wc);
+            uint32_t dc = deletes_per_row[r];
+            for (uint32_t i = 0; i < dc; i++) {
+                uint64_t ts = delete_ts[dat + i];
+                if (i && delete_ts[dat + i - 1] <= ts) {
+                    set_err(errbuf, errlen,
+                            "versioned: delete timestamps must be strictly descending per row");
+                    free(tsegs); free(vsegs);
+                    return YT_ERR_INVALID_CHUNK;
+                }
+                dids.push_back(reg(ts));
+            }
+            dat += dc;
+            dcnt.push_back((dcnt.empty() ? 0 : dcnt.back()) + dc);
+        }
+        uint64_t ts_span = ts_max - ts_min;   /* wraps when no timestamps */
+        std::vector<uint64_t> dict_rel(dict.size());
+        for (size_t i = 0; i < dict.size(); i++) dict_rel[i] = dict[i] - ts_min;
+
+        uint64_t wdiff_max = 0, ddiff_max = 0;
+        uint32_t exp_w = prepare_diff_from_expected(&wcnt, &wdiff_max);
+        uint32_t exp_d = prepare_diff_from_expected(&dcnt, &ddiff_max);
+
+        Blob tb;
+        tb.pack(dict_rel, ts_span);
+        tb.pack(wids, dict.size());
+        tb.pack(dids, dict.size());
+        tb.pack(wcnt, wdiff_max);
+        tb.pack(dcnt, ddiff_max);
+
+        YtTimestampSeg& T = tsegs[si];
+        T.row_count = rows;
+        T.base_timestamp = ts_min;
+        T.expected_writes_per_row = exp_w;
+        T.expected_deletes_per_row = exp_d;
+        T.data_size = tb.size();
+        void* tp = malloc(tb.size() ? tb.size() : 1);
+        memcpy(tp, tb.bytes.data(), tb.size());
+        T.data = tp;
+
+        /* --- versioned int64 value segment (DirectDense) --- */
+        uint64_t vspan = vmax - vmin;         /* wraps when all null/empty */
+        for (size_t i = 0; i < vals_zz.size(); i++)
+            if (!vnull[i]) vals_zz[i] -= vmin;
+
+        uint64_t vdiff_max = 0;
+        uint32_t exp_v = prepare_diff_from_expected(&voff, &vdiff_max);
+
+        Blob vb;
+        vb.pack(voff, vdiff_max);
+        vb.pack(tsids, max_tsid);
+        vb.pack(vals_zz, vspan);
+        vb.bitmap(vnull);
+
+        YtVersionedValueSeg& V = vsegs[si];
+        V.row_count = rows;
+        V.base_value = vmin;
+        V.expected_values_per_row = exp_v;
+        V.data_size = vb.size();
+        void* vp = malloc(vb.size() ? vb.size() : 1);
+        memcpy(vp, vb.bytes.data(), vb.size());
+        V.data = vp;
+    }
+
+    out->ts_seg_count = nseg;
+    out->val_seg_count = nseg;
+    out->ts_segs = tsegs;
+    out->val_segs = vsegs;
+    return YT_OK;
+}
+
+extern "C" void yt_versioned_free(YtVersionedColumn* col)
+{
+    if (!col) return;
+    for (int i = 0; i < col->ts_seg_count; i++)
+        free((void*)col->ts_segs[i].data);
+    for (int i = 0; i < col->val_seg_count; i++)
+        free((void*)col->val_segs[i].data);
+    free((void*)col->ts_segs);
+    free((void*)col->val_segs);
+    memset(col, 0, sizeof(*col));
+}

@@ -65,6 +65,9 @@ hipError_t ytql_launch_topk_gather_fast(const DevSeg*, const SegEx*, int, int,
                                         int64_t*, unsigned long long*,
                                         int64_t*, unsigned long long*,
                                         hipStream_t);
+hipError_t ytql_launch_versioned_read(const VSegDev*, int, int64_t, uint64_t,
+                                      uint64_t*, uint8_t*, uint8_t*,
+                                      hipStream_t);
 hipError_t ytql_launch_join_build(const JoinDev*, int64_t, uint64_t*,
                                   long long*, unsigned long long*, unsigned*,
                                   hipStream_t);
@@ -2575,6 +2578,67 @@ fail:
 
 /* ------------------------------------------------------------------ */
 /* public entries                                                      */
+
+extern "C" int yt_gpu_versioned_read(
+    const YtVersionedColumn* col, uint64_t timestamp,
+    uint64_t* out_bits, uint8_t* out_null, uint8_t* out_visible,
+    uint64_t stream, char* errbuf, size_t errlen)
+{
+    int rc = yt_gpu_available(errbuf, errlen);
+    if (rc != YT_OK) return rc;
+    hipStream_t st = (hipStream_t)(uintptr_t)stream;
+    int nseg = col->ts_seg_count;
+    if (nseg == 0) return YT_OK;
+    if (col->val_seg_count != nseg) {
+        set_err(errbuf, errlen, "versioned: segment count mismatch");
+        return YT_ERR_INVALID_CHUNK;
+    }
+    std::vector<VSegDev> h_segs((size_t)nseg);
+    std::vector<void*> blobs;
+    int64_t total_rows = 0;
+    {
+        for (int i = 0; i < nseg; i++) {
+            const YtTimestampSeg& T = col->ts_segs[i];
+            const YtVersionedValueSeg& V = col->val_segs[i];
+            char* d_t = nullptr;
+            char* d_v = nullptr;
+            /* +8 pad: bp_gl may read one word past the last vector */
+            HIP_CHECK(pool_alloc(&d_t, (size_t)T.data_size + 8));
+            HIP_CHECK(pool_alloc(&d_v, (size_t)V.data_size + 8));
+            blobs.push_back(d_t);
+            blobs.push_back(d_v);
+            HIP_CHECK(hipMemcpyAsync(d_t, T.data, (size_t)T.data_size,
+                                     hipMemcpyHostToDevice, st));
+            HIP_CHECK(hipMemcpyAsync(d_v, V.data, (size_t)V.data_size,
+                                     hipMemcpyHostToDevice, st));
+            VSegDev& S = h_segs[i];
+            S.row_start = total_rows;
+            S.row_count = T.row_count;
+            S.base_timestamp = T.base_timestamp;
+            S.exp_w = T.expected_writes_per_row;
+            S.exp_d = T.expected_deletes_per_row;
+            S.exp_v = V.expected_values_per_row;
+            S.base_value = V.base_value;
+            S.ts_data = d_t;
+            S.val_data = d_v;
+            total_rows += T.row_count;
+        }
+        VSegDev* d_segs = nullptr;
+        HIP_CHECK(pool_alloc(&d_segs, sizeof(VSegDev) * nseg));
+        blobs.push_back(d_segs);
+        HIP_CHECK(hipMemcpyAsync(d_segs, h_segs.data(), sizeof(VSegDev) * nseg,
+                                 hipMemcpyHostToDevice, st));
+        HIP_CHECK(ytql_launch_versioned_read(d_segs, nseg, total_rows,
+                                             timestamp, out_bits, out_null,
+                                             out_visible, st));
+        HIP_CHECK(hipStreamSynchronize(st));
+    }
+    for (void* p : blobs) g_pool.put(p);
+    return YT_OK;
+fail:
+    for (void* p : blobs) g_pool.put(p);
+    return rc;
+}
 
 extern "C" int yt_gpu_query_execute(
     const YtPlan* plan, const YtChunk* chunk, const YtExecOptions* options,

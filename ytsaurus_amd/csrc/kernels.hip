@@ -731,6 +731,113 @@ __device__ void table_update_generic(TableHdr* th, unsigned long long* slots,
 }
 
 /* ------------------------------------------------------------------ */
+/* versioned scan-format read-at-timestamp           */
+/* One thread per row. Layout + semantics: include/ytql_gpu.h.          */
+
+struct VVecRef {
+    const uint64_t* data;
+    uint32_t width;
+    uint64_t size;
+};
+
+__device__ __forceinline__ VVecRef vvec_parse(const char** cursor)
+{
+    const uint64_t* p = (const uint64_t*)*cursor;
+    VVecRef v;
+    v.size = p[0] & ((1ULL << 56) - 1);
+    v.width = (uint32_t)(p[0] >> 56);
+    v.data = p + 1;
+    *cursor += (1 + (((uint64_t)v.width * v.size + 63) >> 6)) * 8;
+    return v;
+}
+
+__device__ __forceinline__ uint64_t vvec_get(const VVecRef& v, uint64_t i)
+{
+    if (v.width == 0) return 0;
+    uint64_t mask = v.width >= 64 ? ~0ULL : ((1ULL << v.width) - 1);
+    return bp_gl(v.data, mask, v.width, (int64_t)i);
+}
+
+__device__ __forceinline__ uint64_t vcum_dev(const VVecRef& diffs,
+                                             uint32_t expected, int64_t i)
+{
+    if (i < 0) return 0;
+    uint32_t zz = (uint32_t)vvec_get(diffs, i);
+    int32_t diff = (int32_t)((zz >> 1) ^ (~(zz & 1) + 1));
+    return (uint64_t)expected * (uint64_t)(i + 1) + (uint64_t)(int64_t)diff;
+}
+
+__global__ void __launch_bounds__(256)
+k_versioned_read(const VSegDev* segs, int nseg, int64_t total_rows,
+                 uint64_t timestamp,
+                 uint64_t* out_bits, uint8_t* out_null, uint8_t* out_vis)
+{
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t g = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         g < total_rows; g += stride) {
+        int lo = 0, hi = nseg;
+        while (lo + 1 < hi) {
+            int mid = (lo + hi) / 2;
+            if (segs[mid].row_start <= g) lo = mid;
+            else hi = mid;
+        }
+        const VSegDev& S = segs[lo];
+        int64_t r = g - S.row_start;
+
+        const char* tp = S.ts_data;
+        VVecRef dict = vvec_parse(&tp);
+        VVecRef wids = vvec_parse(&tp);
+        VVecRef dids = vvec_parse(&tp);
+        VVecRef wdiffs = vvec_parse(&tp);
+        VVecRef ddiffs = vvec_parse(&tp);
+        const char* vp = S.val_data;
+        VVecRef voffs = vvec_parse(&vp);
+        VVecRef tsids = vvec_parse(&vp);
+        VVecRef vvals = vvec_parse(&vp);
+        const uint8_t* vnull = (const uint8_t*)vp;
+
+        out_vis[g] = 0;
+        out_null[g] = 1;
+        out_bits[g] = 0;
+
+        uint64_t wb = vcum_dev(wdiffs, S.exp_w, r - 1);
+        uint64_t we = vcum_dev(wdiffs, S.exp_w, r);
+        uint64_t db = vcum_dev(ddiffs, S.exp_d, r - 1);
+        uint64_t de = vcum_dev(ddiffs, S.exp_d, r);
+
+        uint64_t delete_ts = 0;
+        for (uint64_t i = db; i < de; i++) {
+            uint64_t ts = S.base_timestamp + vvec_get(dict, vvec_get(dids, i));
+            if (ts <= timestamp) { delete_ts = ts; break; }
+        }
+        int64_t wcount = (int64_t)(we - wb);
+        int64_t lower = wcount, upper = wcount;
+        for (int64_t i = 0; i < wcount; i++) {
+            uint64_t ts = S.base_timestamp + vvec_get(dict, vvec_get(wids, wb + i));
+            if (lower == wcount && ts <= timestamp) lower = i;
+            if (ts <= delete_ts) { upper = i; break; }
+        }
+        if (upper < lower) upper = lower;
+        if (lower >= upper) continue;
+        out_vis[g] = 1;
+
+        uint64_t vb = vcum_dev(voffs, S.exp_v, r - 1);
+        uint64_t ve = vcum_dev(voffs, S.exp_v, r);
+        for (uint64_t j = vb; j < ve; j++) {
+            int64_t ti = (int64_t)vvec_get(tsids, j);
+            if (ti < lower) continue;
+            if (ti >= upper) break;
+            int nul = (vnull[j / 8] >> (j % 8)) & 1;
+            if (!nul) {
+                out_null[g] = 0;
+                out_bits[g] = (uint64_t)zz_dec(S.base_value + vvec_get(vvals, j));
+            }
+            break;
+        }
+    }
+}
+
+/* ------------------------------------------------------------------ */
 /* equi-join foreign-table build + unique-key verify                    */
 
 __global__ void k_join_build(JoinDev jt, int64_t frows,
@@ -2236,6 +2343,20 @@ hipError_t ytql_launch_topk_gather_fast(const DevSeg* segs, const SegEx* segex,
                        segs, segex, seg_off, shift, n, has_nulls, is_signed,
                        *tg, rows_strict, ctr_strict, rows_tie, ctr_tie,
                        rows_null, ctr_null);
+    return hipGetLastError();
+}
+
+hipError_t ytql_launch_versioned_read(const VSegDev* segs, int nseg,
+                                      int64_t total_rows, uint64_t timestamp,
+                                      uint64_t* out_bits, uint8_t* out_null,
+                                      uint8_t* out_vis, hipStream_t st)
+{
+    int block = 256;
+    int64_t want = (total_rows + block - 1) / block;
+    int grid = (int)(want > 4096 ? 4096 : (want > 0 ? want : 1));
+    hipLaunchKernelGGL(k_versioned_read, dim3(grid), dim3(block), 0, st,
+                       segs, nseg, total_rows, timestamp,
+                       out_bits, out_null, out_vis);
     return hipGetLastError();
 }
 
