@@ -740,6 +740,27 @@ extern "C" int yt_encode_versioned_int64(
             if (!vnull[i]) vals_zz[i] -= vmin;
 
         uint64_t vdiff_max = 0;
+        /* the reference writer picks dense iff denseSize <= sparseSize
+         * (column_writer_detail.cpp:199-210); emit only the dense layout
+         * this round and refuse inputs where the writer would pick sparse
+         * so the generated bytes always match the reference's choice */
+        {
+            int64_t dense_probe = 0;
+            {
+                std::vector<uint64_t> probe = voff;
+                uint64_t md = 0;
+                (void)prepare_diff_from_expected(&probe, &md);
+                dense_probe = cs_bytes(md, (int64_t)voff.size());
+            }
+            int64_t sparse_sz = cs_bytes((uint64_t)voff.size(),
+                                         (int64_t)vals_zz.size());
+            if (dense_probe > sparse_sz) {
+                set_err(errbuf, errlen,
+                        "versioned: sparse value-index layout not this round");
+                free(tsegs); free(vsegs);
+                return YT_ERR_UNSUPPORTED;
+            }
+        }
         uint32_t exp_v = prepare_diff_from_expected(&voff, &vdiff_max);
 
         Blob vb;
