@@ -330,3 +330,18 @@ def test_golden_group_order_desc_then_sum_gpu(cuda):
                   order_by=[(0, True), (1, False)], limit=3)
     rows, _ = y.gpu_execute(plan, chunk.c_device(cuda), max_groups_hint=64)
     assert rows == [(2, 18), (1, 15), (0, 12)]
+
+
+@pytest.mark.gpu
+def test_topk_gpu_only_nulls_with_min_keys(cuda):
+    # regression: limit <= null count AND a mapped-zero key (INT64_MIN asc)
+    # must not overflow the strict-candidate buffer
+    n = 50_000
+    rng = np.random.default_rng(41)
+    a = np.full(n, -2**63, dtype=np.int64)
+    an = np.ones(n, dtype=np.uint8)
+    an[: n // 2] = 0                      # half INT64_MIN, half null
+    chunk = y.Chunk([y.encode_int64(a, an)], n)
+    plan = y.Plan(projects=[y.col(0)], order_by=[(0, False)], limit=100)
+    got, _ = y.gpu_execute(plan, chunk.c_device(cuda))
+    assert got == [(None,)] * 100

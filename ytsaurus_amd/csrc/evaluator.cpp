@@ -1919,9 +1919,11 @@ static int run_scan_topk(const YtPlan* plan, const YtChunk* chunk,
         tg.desc = desc0;
         tg.all_nonnull = all_nonnull ? 1 : 0;
         tg.pad_ = 0;
-        tg.lo = only_nulls ? 0 : lo;       /* only_nulls: nothing strict */
+        /* only_nulls: nothing strict (strict is m < lo, so lo = 0 matches
+         * no row — a mapped key CAN be 0, e.g. INT64_MIN ascending, and the
+         * strict buffer is unbounded) and at most the capped tie bucket */
+        tg.lo = only_nulls ? 0 : lo;
         tg.hi = only_nulls ? 0 : hi;
-        if (only_nulls) { tg.lo = 1; tg.hi = 0; }  /* empty tie interval */
         tg.cap_tie = CAPB;
         tg.cap_null = CAPN;
         int64_t* d_rows_strict = nullptr;
