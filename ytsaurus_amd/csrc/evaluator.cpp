@@ -2698,8 +2698,11 @@ extern "C" int yt_gpu_query_execute(
             rc = run_string_group(plan, chunk, options, kc, output, stats,
                                   tw0, errbuf, errlen);
             if (rc == YT_OK && (plan->order_count > 0 || plan->with_totals ||
-                                plan->having))
+                                plan->having)) {
                 rc = finish_output(plan, output, errbuf, errlen);
+                if (rc == YT_OK && stats)
+                    stats->rows_written = output->row_count;
+            }
             return rc;
         }
     }
