@@ -307,6 +307,20 @@ int yt_gpu_versioned_read(
     uint64_t* out_bits, uint8_t* out_null, uint8_t* out_visible,
     uint64_t stream, char* errbuf, size_t errlen);
 
+/* Read the column at `timestamp` and compact the VISIBLE rows into a
+ * device-resident single-column unversioned chunk — reference-layout
+ * DirectDense width-64 segments (128 Ki row cap, min_value = 0, values in
+ * zigzag space) that yt_gpu_query_execute scans directly. This is the
+ * versioned→engine bridge: deleted / not-yet-written rows disappear, row
+ * order is preserved. out_chunk's column/segment arrays are heap-allocated
+ * host structs pointing at device blobs; release BOTH with
+ * yt_gpu_scan_chunk_free. */
+int yt_gpu_versioned_scan_chunk(
+    const YtVersionedColumn* col, uint64_t timestamp,
+    YtChunk* out_chunk, void** out_handle,
+    uint64_t stream, char* errbuf, size_t errlen);
+void yt_gpu_scan_chunk_free(YtChunk* chunk, void* handle);
+
 /* =========================== entry points =========================== */
 
 /* Library/device probe. Returns YT_OK when a gfx950 HIP device is usable. */

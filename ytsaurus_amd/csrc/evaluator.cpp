@@ -68,6 +68,12 @@ hipError_t ytql_launch_topk_gather_fast(const DevSeg*, const SegEx*, int, int,
 hipError_t ytql_launch_versioned_read(const VSegDev*, int, int64_t, uint64_t,
                                       uint64_t*, uint8_t*, uint8_t*,
                                       hipStream_t);
+hipError_t ytql_launch_vis_count(const uint8_t*, int64_t,
+                                 unsigned long long*, int, hipStream_t);
+hipError_t ytql_launch_vis_scatter(const uint8_t*, const uint8_t*,
+                                   const uint64_t*, int64_t,
+                                   const unsigned long long*, uint64_t,
+                                   const int64_t*, char*, int, hipStream_t);
 hipError_t ytql_launch_join_build(const JoinDev*, int64_t, uint64_t*,
                                   long long*, unsigned long long*, unsigned*,
                                   hipStream_t);
@@ -2640,6 +2646,136 @@ extern "C" int yt_gpu_versioned_read(
 fail:
     for (void* p : blobs) g_pool.put(p);
     return rc;
+}
+
+namespace {
+struct ScanChunkHandle {
+    char* blob = nullptr;              /* device */
+    YtSegment* segs = nullptr;         /* host heap */
+    YtColumn* col = nullptr;           /* host heap */
+};
+} /* namespace */
+
+extern "C" int yt_gpu_versioned_scan_chunk(
+    const YtVersionedColumn* col, uint64_t timestamp,
+    YtChunk* out_chunk, void** out_handle,
+    uint64_t stream, char* errbuf, size_t errlen)
+{
+    int rc = yt_gpu_available(errbuf, errlen);
+    if (rc != YT_OK) return rc;
+    hipStream_t st = (hipStream_t)(uintptr_t)stream;
+    memset(out_chunk, 0, sizeof(*out_chunk));
+    *out_handle = nullptr;
+
+    int64_t n = 0;
+    for (int i = 0; i < col->ts_seg_count; i++)
+        n += col->ts_segs[i].row_count;
+
+    auto* H = new ScanChunkHandle();
+    uint64_t* d_bits = nullptr;
+    uint8_t* d_null = nullptr;
+    uint8_t* d_vis = nullptr;
+    unsigned long long* d_blk = nullptr;
+    int64_t* d_off = nullptr;
+    const int64_t kSegCap = 128 * 1024;
+    int grid = (int)((n + 255) / 256);
+    if (grid > 2048) grid = 2048;
+    if (grid < 1) grid = 1;
+    int64_t total = 0;
+    int nseg = 0;
+    std::vector<unsigned long long> blk((size_t)grid);
+    std::vector<int64_t> seg_off;
+    std::vector<int64_t> seg_rows;
+    int64_t blob_bytes = 0;
+    auto cleanup = [&]() {
+        g_pool.put(d_bits); g_pool.put(d_null); g_pool.put(d_vis);
+        g_pool.put(d_blk); g_pool.put(d_off);
+    };
+    if (n == 0) { *out_handle = H; return YT_OK; }
+
+    HIP_CHECK(pool_alloc(&d_bits, sizeof(uint64_t) * n));
+    HIP_CHECK(pool_alloc(&d_null, (size_t)n));
+    HIP_CHECK(pool_alloc(&d_vis, (size_t)n));
+    HIP_CHECK(pool_alloc(&d_blk, sizeof(unsigned long long) * grid));
+    rc = yt_gpu_versioned_read(col, timestamp, d_bits, d_null, d_vis,
+                               stream, errbuf, errlen);
+    if (rc != YT_OK) { cleanup(); delete H; return rc; }
+
+    HIP_CHECK(ytql_launch_vis_count(d_vis, n, d_blk, grid, st));
+    HIP_CHECK(hipMemcpy(blk.data(), d_blk,
+                        sizeof(unsigned long long) * grid,
+                        hipMemcpyDeviceToHost));
+    {
+        unsigned long long run = 0;
+        for (int i = 0; i < grid; i++) {
+            unsigned long long c = blk[i];
+            blk[i] = run;
+            run += c;
+        }
+        total = (int64_t)run;
+    }
+    if (total == 0) { cleanup(); *out_handle = H; return YT_OK; }
+    HIP_CHECK(hipMemcpyAsync(d_blk, blk.data(),
+                             sizeof(unsigned long long) * grid,
+                             hipMemcpyHostToDevice, st));
+
+    nseg = (int)((total + kSegCap - 1) / kSegCap);
+    for (int i = 0; i < nseg; i++) {
+        int64_t rows = i + 1 < nseg ? kSegCap : total - (int64_t)i * kSegCap;
+        int64_t bm = (((rows + 7) / 8) + 7) & ~(int64_t)7;
+        seg_off.push_back(blob_bytes);
+        seg_rows.push_back(rows);
+        blob_bytes += 8 + rows * 8 + bm;
+    }
+    HIP_CHECK(pool_alloc(&H->blob, (size_t)blob_bytes + 8));
+    HIP_CHECK(hipMemsetAsync(H->blob, 0, (size_t)blob_bytes, st));
+    for (int i = 0; i < nseg; i++) {
+        uint64_t hdr = (uint64_t)seg_rows[i] | (64ULL << 56);
+        HIP_CHECK(hipMemcpyAsync(H->blob + seg_off[i], &hdr, 8,
+                                 hipMemcpyHostToDevice, st));
+    }
+    HIP_CHECK(pool_alloc(&d_off, sizeof(int64_t) * nseg));
+    HIP_CHECK(hipMemcpyAsync(d_off, seg_off.data(), sizeof(int64_t) * nseg,
+                             hipMemcpyHostToDevice, st));
+    HIP_CHECK(ytql_launch_vis_scatter(d_vis, d_null, d_bits, n, d_blk,
+                                      (uint64_t)kSegCap, d_off, H->blob,
+                                      grid, st));
+    HIP_CHECK(hipStreamSynchronize(st));
+
+    H->segs = (YtSegment*)calloc(nseg, sizeof(YtSegment));
+    H->col = (YtColumn*)calloc(1, sizeof(YtColumn));
+    for (int i = 0; i < nseg; i++) {
+        H->segs[i].type = YT_SEG_DIRECT_DENSE;
+        H->segs[i].row_count = (int32_t)seg_rows[i];
+        H->segs[i].min_value = 0;
+        H->segs[i].data = H->blob + seg_off[i];
+        H->segs[i].data_size = (i + 1 < nseg ? seg_off[i + 1] : blob_bytes)
+                             - seg_off[i];
+    }
+    H->col->value_type = YT_VT_INT64;
+    H->col->segment_count = nseg;
+    H->col->segments = H->segs;
+    out_chunk->row_count = total;
+    out_chunk->column_count = 1;
+    out_chunk->columns = H->col;
+    *out_handle = H;
+    cleanup();
+    return YT_OK;
+fail:
+    cleanup();
+    delete H;
+    return rc;
+}
+
+extern "C" void yt_gpu_scan_chunk_free(YtChunk* chunk, void* handle)
+{
+    auto* H = (ScanChunkHandle*)handle;
+    if (!H) return;
+    g_pool.put(H->blob);
+    free(H->segs);
+    free(H->col);
+    delete H;
+    if (chunk) memset(chunk, 0, sizeof(*chunk));
 }
 
 extern "C" int yt_gpu_query_execute(

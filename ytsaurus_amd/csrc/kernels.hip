@@ -837,6 +837,93 @@ k_versioned_read(const VSegDev* segs, int nseg, int64_t total_rows,
     }
 }
 
+/* versioned → unversioned chunk bridge: compact the visible rows of a
+ * read-at-timestamp into reference-layout DirectDense width-64 segments
+ * (min_value = 0, values in zigzag space) that the query engine scans
+ * directly. Stable two-pass block compaction (pass A per-block counts,
+ * host scans the ≤4096 block sums, pass B scatters). */
+__global__ void __launch_bounds__(256)
+k_vis_count(const uint8_t* vis, int64_t n, unsigned long long* block_counts)
+{
+    __shared__ unsigned long long s_cnt;
+    if (threadIdx.x == 0) s_cnt = 0;
+    __syncthreads();
+    int64_t per = (n + gridDim.x - 1) / gridDim.x;
+    int64_t b0 = (int64_t)blockIdx.x * per;
+    int64_t b1 = b0 + per > n ? n : b0 + per;
+    unsigned long long c = 0;
+    for (int64_t i = b0 + threadIdx.x; i < b1; i += blockDim.x)
+        c += vis[i];
+    #pragma unroll
+    for (int d = 32; d; d >>= 1) c += __shfl_down(c, d, 64);
+    if ((threadIdx.x & 63) == 0 && c) atomicAdd(&s_cnt, c);
+    __syncthreads();
+    if (threadIdx.x == 0) block_counts[blockIdx.x] = s_cnt;
+}
+
+__global__ void __launch_bounds__(256)
+k_vis_scatter(const uint8_t* vis, const uint8_t* nulls, const uint64_t* bits,
+              int64_t n, const unsigned long long* block_bases,
+              uint64_t seg_rows_cap,
+              const int64_t* seg_blob_off,   /* byte offset of each segment */
+              char* out_blob)
+{
+    __shared__ unsigned long long s_wcnt[16];
+    __shared__ unsigned long long s_run;
+    const int lane = threadIdx.x & 63;
+    const int wid = (int)(threadIdx.x >> 6);
+    const int nw = (int)(blockDim.x >> 6);
+
+    int64_t per = (n + gridDim.x - 1) / gridDim.x;
+    int64_t b0 = (int64_t)blockIdx.x * per;
+    int64_t b1 = b0 + per > n ? n : b0 + per;
+    if (threadIdx.x == 0) s_run = block_bases[blockIdx.x];
+    __syncthreads();
+
+    for (int64_t base = b0; base < b1; base += blockDim.x) {
+        int64_t i = base + threadIdx.x;
+        int v = (i < b1) ? vis[i] : 0;
+        unsigned long long mask = __ballot(v != 0);
+        unsigned long long run = v;
+        #pragma unroll
+        for (int d = 1; d < 64; d <<= 1) {
+            unsigned long long x = __shfl_up(run, d, 64);
+            if (lane >= d) run += x;
+        }
+        if (lane == 63) s_wcnt[wid] = run;
+        __syncthreads();
+        if (v) {
+            unsigned long long woff = 0;
+            for (int w = 0; w < wid; w++) woff += s_wcnt[w];
+            uint64_t idx = s_run + woff + (run - v);
+            uint64_t seg = idx / seg_rows_cap;
+            uint64_t j = idx - seg * seg_rows_cap;
+            char* sb = out_blob + seg_blob_off[seg];
+            /* [header][values w64][null bitmap] — values in zigzag space */
+            int64_t v64 = (int64_t)bits[i];
+            uint64_t zz = ((uint64_t)v64 << 1) ^ (uint64_t)(v64 >> 63);
+            ((uint64_t*)(sb + 8))[j] = nulls[i] ? 0 : zz;
+            if (nulls[i]) {
+                uint64_t rows_here = 0;   /* bitmap offset needs seg rows */
+                (void)rows_here;
+                /* bitmap starts after the per-segment value words; the host
+                 * passes per-segment offsets so compute from the NEXT
+                 * segment boundary: bitmap_off = 8 + seg_rows*8 where
+                 * seg_rows = min(cap, total - seg*cap); the host encodes
+                 * seg_rows in the header it wrote BEFORE this kernel runs */
+                uint64_t hdr = *(const uint64_t*)sb;
+                uint64_t rows = hdr & ((1ULL << 56) - 1);
+                uint32_t* bm = (uint32_t*)(sb + 8 + rows * 8);
+                atomicOr(&bm[j >> 5], 1u << (j & 31));
+            }
+        }
+        __syncthreads();
+        if (threadIdx.x == 0)
+            for (int w = 0; w < nw; w++) s_run += s_wcnt[w];
+        __syncthreads();
+    }
+}
+
 /* ------------------------------------------------------------------ */
 /* equi-join foreign-table build + unique-key verify                    */
 
@@ -2357,6 +2444,28 @@ hipError_t ytql_launch_versioned_read(const VSegDev* segs, int nseg,
     hipLaunchKernelGGL(k_versioned_read, dim3(grid), dim3(block), 0, st,
                        segs, nseg, total_rows, timestamp,
                        out_bits, out_null, out_vis);
+    return hipGetLastError();
+}
+
+hipError_t ytql_launch_vis_count(const uint8_t* vis, int64_t n,
+                                 unsigned long long* block_counts, int grid,
+                                 hipStream_t st)
+{
+    hipLaunchKernelGGL(k_vis_count, dim3(grid), dim3(256), 0, st,
+                       vis, n, block_counts);
+    return hipGetLastError();
+}
+
+hipError_t ytql_launch_vis_scatter(const uint8_t* vis, const uint8_t* nulls,
+                                   const uint64_t* bits, int64_t n,
+                                   const unsigned long long* block_bases,
+                                   uint64_t seg_rows_cap,
+                                   const int64_t* seg_blob_off,
+                                   char* out_blob, int grid, hipStream_t st)
+{
+    hipLaunchKernelGGL(k_vis_scatter, dim3(grid), dim3(256), 0, st,
+                       vis, nulls, bits, n, block_bases, seg_rows_cap,
+                       seg_blob_off, out_blob);
     return hipGetLastError();
 }
 
