@@ -20,7 +20,7 @@ from ._abi import (
 __all__ = [
     "col", "lit", "null", "litf", "Plan", "Join", "agg_sum", "agg_sum1",
     "encode_int64", "encode_double", "encode_bool", "encode_string", "encode_string_raw",
-    "oracle_decode_strings", "encode_versioned_int64", "oracle_versioned_read", "gpu_versioned_read", "VersionedColumn",
+    "oracle_decode_strings", "encode_versioned_int64", "oracle_versioned_read", "gpu_versioned_read", "gpu_versioned_scan_chunk", "ScanChunk", "VersionedColumn",
     "Chunk", "oracle_execute",
     "oracle_partial", "oracle_merge",
     "gpu_available", "gpu_execute", "gpu_partial", "gpu_merge",
@@ -371,6 +371,36 @@ def gpu_versioned_read(vcol, timestamp, torch_mod):
     hv = vis.cpu().numpy()
     out = [None if hn[i] else int(hb[i]) for i in range(n)]
     return out, [bool(v) for v in hv[:n]]
+
+
+class ScanChunk:
+    """Device-resident unversioned chunk produced by the versioned-to-engine
+    bridge (yt_gpu_versioned_scan_chunk); pass .chunk to gpu_execute."""
+
+    def __init__(self, chunk, handle):
+        self.chunk = chunk
+        self._handle = handle
+        self.row_count = chunk.row_count
+
+    def __del__(self):
+        try:
+            _abi.gpu_lib().yt_gpu_scan_chunk_free(C.byref(self.chunk),
+                                                  self._handle)
+        except Exception:
+            pass
+
+
+def gpu_versioned_scan_chunk(vcol, timestamp):
+    """Read `vcol` as of `timestamp` on the GPU and compact the visible rows
+    into an engine-scannable device chunk (SURVEY 8f row-3 bridge)."""
+    ch = _abi.YtChunk()
+    handle = C.c_void_p()
+    err = C.create_string_buffer(256)
+    rc = _abi.gpu_lib().yt_gpu_versioned_scan_chunk(
+        C.byref(vcol._c), C.c_uint64(timestamp), C.byref(ch),
+        C.byref(handle), C.c_uint64(0), err, 256)
+    _check(rc, err)
+    return ScanChunk(ch, handle)
 
 
 def oracle_decode_strings(enc, n):
