@@ -146,8 +146,8 @@ def test_versioned_scan_chunk_feeds_engine(cuda):
         if sc.row_count == 0:
             continue
         plan = y.Plan(keys=[y.col(0)], aggs=[y.agg_sum1()])
-        got, _ = y.gpu_execute(plan, sc.chunk, max_groups_hint=1 << 15,
-                               out_capacity=1 << 18)
+        got, _ = y.gpu_execute(plan, sc.chunk, max_groups_hint=1 << 19,
+                               out_capacity=1 << 19)
         want = collections.Counter(visible_vals)
         assert len(got) == len(want)
         for k, c in got:
