@@ -49,7 +49,7 @@ def _model(rows, T):
     return vals, vis
 
 
-@pytest.mark.parametrize("seed", range(6))
+@pytest.mark.parametrize("seed", range(12))
 def test_versioned_read_matches_model(seed):
     rng = np.random.default_rng([7, seed])
     n = int(rng.choice([1, 50, 3000]))
@@ -92,6 +92,17 @@ def test_versioned_delete_shadowing():
                          (30, 300, True), (99, 300, True)]:
         vals, viss = y.oracle_versioned_read(col, T)
         assert (vals[0], viss[0]) == (want, vis), T
+
+
+def test_versioned_all_empty_rows():
+    # rows with no writes and no deletes are never visible
+    n = 500
+    col = y.encode_versioned_int64(
+        np.zeros(n, dtype=np.uint32), np.array([], dtype=np.uint64),
+        np.array([], dtype=np.int64), None,
+        np.zeros(n, dtype=np.uint32), np.array([], dtype=np.uint64))
+    vals, vis = y.oracle_versioned_read(col, 10**18)
+    assert vals == [None] * n and vis == [False] * n
 
 
 def test_versioned_bad_order_errors():
