@@ -366,9 +366,9 @@ def test_string_group_by_gpu(cuda):
     vn = (rng.random(n) < 0.05).astype(np.uint8)
     chunk = y.Chunk([y.encode_string(keys, max_segment_values=20000),
                      y.encode_double(vals, vn, max_segment_values=20000)], n)
-    # ensure dictionary encoding was chosen (the GPU path requires it)
-    assert all(s.type in (1,) or s.type == 0 for s in chunk.columns[0].segments
-               if True) or True
+    # ensure dictionary encoding was chosen (the GPU path requires it):
+    # string segment types are 1 = DictionaryDense, 0 = DictionaryRle
+    assert all(s.type in (0, 1) for s in chunk.columns[0].segments)
     plan = y.Plan(keys=[y.col(0)], aggs=[y.agg_sum(y.col(1)), y.agg_sum1()])
     got, st = y.gpu_execute(plan, chunk.c_device(cuda), max_groups_hint=1024)
     want, _ = y.oracle_execute(plan, chunk)

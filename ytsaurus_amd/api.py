@@ -319,6 +319,13 @@ def encode_versioned_int64(writes_per_row, write_ts, values, value_nulls,
         if value_nulls is not None else None
     dpr = np.ascontiguousarray(deletes_per_row, dtype=np.uint32)
     dts = np.ascontiguousarray(delete_ts, dtype=np.uint64)
+    nw, nd = int(wpr.sum()), int(dpr.sum())
+    if len(wts) != nw or len(vals) != nw or (vn is not None and len(vn) != nw):
+        raise ValueError("versioned encode: write_ts/values/value_nulls length "
+                         f"must equal sum(writes_per_row)={nw}")
+    if len(dts) != nd:
+        raise ValueError("versioned encode: delete_ts length must equal "
+                         f"sum(deletes_per_row)={nd}")
     cc = _abi.YtVersionedColumn()
     err = C.create_string_buffer(256)
     rc = _abi.gpu_lib().yt_encode_versioned_int64(

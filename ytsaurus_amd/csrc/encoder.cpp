@@ -27,6 +27,7 @@
 #include <stdlib.h>
 #include <vector>
 #include <unordered_map>
+#include <unordered_set>
 #include <string_view>
 #include <string>
 
@@ -644,6 +645,16 @@ extern "C" int yt_encode_versioned_int64(
 
     auto* tsegs = (YtTimestampSeg*)calloc(nseg, sizeof(YtTimestampSeg));
     auto* vsegs = (YtVersionedValueSeg*)calloc(nseg, sizeof(YtVersionedValueSeg));
+    /* error exit: free every per-segment blob populated so far (calloc
+     * zeroed the arrays, so free(NULL) on untouched entries is fine) */
+    auto fail = [&](int rc) {
+        for (int k = 0; k < nseg; k++) {
+            free((void*)tsegs[k].data);
+            free((void*)vsegs[k].data);
+        }
+        free(tsegs); free(vsegs);
+        return rc;
+    };
 
     int64_t wat = 0, dat = 0;   /* global flattened cursors */
     for (int si = 0; si < nseg; si++) {
@@ -674,8 +685,7 @@ extern "C" int yt_encode_versioned_int64(
                 if (i && write_ts[wat + i - 1] <= ts) {
                     set_err(errbuf, errlen,
                             "versioned: write timestamps must be strictly descending per row");
-                    free(tsegs); free(vsegs);
-                    return YT_ERR_INVALID_CHUNK;
+                    return fail(YT_ERR_INVALID_CHUNK);
                 }
                 wids.push_back(reg(ts));
                 /* value slice: 1:1 with writes, tsIndex = position in the
@@ -701,8 +711,7 @@ extern "C" int yt_encode_versioned_int64(
                 if (i && delete_ts[dat + i - 1] <= ts) {
                     set_err(errbuf, errlen,
                             "versioned: delete timestamps must be strictly descending per row");
-                    free(tsegs); free(vsegs);
-                    return YT_ERR_INVALID_CHUNK;
+                    return fail(YT_ERR_INVALID_CHUNK);
                 }
                 dids.push_back(reg(ts));
             }
@@ -757,8 +766,29 @@ extern "C" int yt_encode_versioned_int64(
             if (dense_probe > sparse_sz) {
                 set_err(errbuf, errlen,
                         "versioned: sparse value-index layout not this round");
-                free(tsegs); free(vsegs);
-                return YT_ERR_UNSUPPORTED;
+                return fail(YT_ERR_UNSUPPORTED);
+            }
+        }
+        /* mirror the reference's direct-vs-dictionary choice
+         * (integer_column_writer.cpp GetDictionarySize/GetDirectSize,
+         * DumpSegment: dictionary wins iff dictionarySize < directSize):
+         * we emit only the DirectDense layout, so refuse inputs where the
+         * reference writer would have picked DictionaryDense — otherwise
+         * the generated bytes silently diverge from the reference's. */
+        {
+            std::unordered_set<uint64_t> distinct;
+            for (size_t i = 0; i < vals_zz.size(); i++)
+                if (!vnull[i]) distinct.insert(vals_zz[i]);
+            int64_t dict_sz = cs_bytes(vspan, (int64_t)distinct.size())
+                            + cs_bytes((uint64_t)distinct.size() + 1,
+                                       (int64_t)vals_zz.size());
+            int64_t direct_sz = cs_bytes(vspan, (int64_t)vals_zz.size())
+                              + (int64_t)((vals_zz.size() + 7) / 8);
+            if (dict_sz < direct_sz) {
+                set_err(errbuf, errlen,
+                        "versioned: reference writer would pick DictionaryDense "
+                        "(not built this round)");
+                return fail(YT_ERR_UNSUPPORTED);
             }
         }
         uint32_t exp_v = prepare_diff_from_expected(&voff, &vdiff_max);
