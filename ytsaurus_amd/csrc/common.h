@@ -238,6 +238,19 @@ struct PartParams {
     int32_t stage_bm_mask;        /* bit u: stage used-col u's null bitmap */
     int32_t has_filter_nulls;
     int32_t stage_val;            /* stage the value column's packed words too */
+    /* direct-span mode (small key zigzag span): buckets are key-RANGE
+     * slices (bucket = krel >> dshift) and phase B indexes an LDS array
+     * directly — no hash probe, no CAS, no stored keys. */
+    int32_t direct_mode;
+    int32_t dshift;               /* per-bucket key-slot range = 1 << dshift */
+    /* 64B-aligned record claims: each (tile,bucket) reserves a multiple of 8
+     * records and fills the tail with pad records, so every HBM line of the
+     * partition stream is written whole by one workgroup within one tile
+     * pass (round-1 PMC: unaligned 8B scatter wrote 42 GB for 8 GB of
+     * records). Pads: packed records use bit 63; 16B/null-stream records
+     * use key == kEmptyKey (real INT64_MIN keys never enter the streams —
+     * they are side-slotted). */
+    int32_t aligned;
 };
 
 /* string-keyed GROUP BY (BASELINE config 5 family): dictionary-encoded

@@ -36,7 +36,12 @@ hipError_t ytql_launch_bucket_agg(const void*, const unsigned long long*, int64_
                                   const uint64_t*, const unsigned long long*, int64_t,
                                   OutGroup*, unsigned long long*, int64_t,
                                   TableHdr*, int, int, int, int,
-                                  uint64_t, uint64_t, hipStream_t);
+                                  uint64_t, uint64_t, int, hipStream_t);
+hipError_t ytql_launch_bucket_agg_direct(const void*, const unsigned long long*, int64_t,
+                                         const uint64_t*, const unsigned long long*, int64_t,
+                                         OutGroup*, unsigned long long*, int64_t,
+                                         TableHdr*, int, int, int, int,
+                                         uint64_t, uint64_t, int, hipStream_t);
 hipError_t ytql_launch_topk_hist(const DevPlan*, const DevSeg*, const SegEx*,
                                  const int32_t*, const int32_t*, int64_t,
                                  const JoinDev*,
@@ -848,6 +853,7 @@ static int run_partitioned(const YtPlan* plan, const YtChunk* chunk,
                            const YtExecOptions* options, DeviceRun* R,
                            const FastShape* fs, unsigned maxw,
                            YtStatistics* stats, bool* done,
+                           int force_hash, bool* tried_direct,
                            char* errbuf, size_t errlen)
 {
     int rc = YT_OK;
@@ -901,6 +907,19 @@ static int run_partitioned(const YtPlan* plan, const YtChunk* chunk,
             pp.gmin_k = R->col_zzmin[fs->key_col];
             pp.gmin_v = gmin_v;
         }
+        /* aligned 64B claims need a pad encoding: bit 63 for packed records
+         * (spare iff bk+bv <= 63), kEmptyKey for 16B records (always) */
+        pp.aligned = (!pp.packed_mode || pp.bits_k + bv <= 63) ? 1 : 0;
+        /* direct-span mode: when the key zigzag span is small, partition by
+         * key RANGE and index phase B's per-bucket array directly (the
+         * headline config — 1M distinct keys — spans 21 bits). Needs the
+         * pad encoding (phase B skips by sign bit), so requires aligned. */
+        pp.gmin_k = R->col_zzmin[fs->key_col];
+        if (bk <= 22 && pp.aligned && !force_hash && !getenv("YTQL_NO_DIRECT")) {
+            pp.direct_mode = 1;
+            pp.dshift = bk > 10 ? bk - 10 : 0;
+            if (tried_direct) *tried_direct = true;
+        }
     }
 
     int64_t rows = chunk->row_count;
@@ -911,16 +930,22 @@ static int run_partitioned(const YtPlan* plan, const YtChunk* chunk,
     /* LDS: 4*kNB u32 histogram structures + staged columns (value column's
      * packed words are NOT staged — read from global in the write pass) */
     unsigned w = maxw ? maxw : 1;
-    int tile_rows = 8192;
+    /* bigger tiles make aligned claims cheaper (fewer pad tails per
+     * bucket); bounded by LDS (key staging) at 2 workgroups/CU (<=80KB) */
+    int tile_rows = 16384;
     {
         const char* ev = getenv("YTQL_TILE");   /* perf experiments only */
         if (ev && atoi(ev) >= 256) tile_rows = atoi(ev);
+        if (tile_rows > 16384) tile_rows = 16384;   /* per-thread row arrays bound */
     }
-    if (tile_rows > 8192) tile_rows = 8192;   /* per-thread row arrays bound */
     /* LDS: per-tile bucket histograms + the staged key column */
-    size_t lds = 4 * kNB * 4 + ((size_t)tile_rows * w / 64 + 2) * 8 + 256;
-    if (lds > 64 * 1024) { tile_rows = 4096; lds = 4 * kNB * 4 + ((size_t)tile_rows * w / 64 + 2) * 8 + 256; }
+    auto lds_for = [&](int tr) {
+        return (size_t)4 * kNB * 4 + ((size_t)tr * w / 64 + 2) * 8 + 256;
+    };
+    size_t lds = lds_for(tile_rows);
+    while (lds > 80 * 1024 && tile_rows > 4096) { tile_rows >>= 1; lds = lds_for(tile_rows); }
     while (tile_rows > 256 && tile_rows > seg0_rows) tile_rows >>= 1;
+    lds = lds_for(tile_rows);
     pp.tile_rows = tile_rows;
     pp.tiles_per_seg = (seg0_rows + tile_rows - 1) / tile_rows;
     {
@@ -928,8 +953,14 @@ static int run_partitioned(const YtPlan* plan, const YtChunk* chunk,
         int32_t last_rows = R->h_segs[R->h_off[fs->key_col] + nseg - 1].row_count;
         pp.ntiles = (nseg - 1) * pp.tiles_per_seg + (last_rows + tile_rows - 1) / tile_rows;
     }
-    /* per (bucket, XCD) sub-streams: 8x more cursors, 1/8 the rows each */
+    /* per (bucket, XCD) sub-streams: 8x more cursors, 1/8 the rows each.
+     * Aligned mode adds up to 7 pad records per (tile,bucket,sub) claim —
+     * budget the expected pad tail (~3.5 per claiming tile) explicitly. */
     pp.bucket_stride = rows / (kNB * 8) + (rows / (kNB * 8)) / 2 + 4096;
+    if (pp.aligned) {
+        int64_t tiles_per_sub = ((int64_t)pp.tiles_per_seg * R->h_cnt[fs->key_col] + 7) / 8;
+        pp.bucket_stride += 4 * tiles_per_sub + 4096;
+    }
     pp.nbucket_stride = pp.has_val_nulls ? pp.bucket_stride : 0;
 
     std::vector<FastCol> fc(nused);
@@ -977,12 +1008,21 @@ static int run_partitioned(const YtPlan* plan, const YtChunk* chunk,
                                              R->d_ncursors, R->d_nrecs,
                                              lds, grid, R->stream));
         HIP_CHECK(hipEventRecord(ev1, R->stream));
-        HIP_CHECK(ytql_launch_bucket_agg(R->d_recs, R->d_cursors, pp.bucket_stride,
-                                         R->d_nrecs, R->d_ncursors, pp.nbucket_stride,
-                                         R->d_groups, R->d_counter, cap_groups,
-                                         R->d_th, pp.sum_slot, pp.agg_count,
-                                         pp.packed_mode, pp.bits_k,
-                                         pp.gmin_k, pp.gmin_v, R->stream));
+        if (pp.direct_mode) {
+            HIP_CHECK(ytql_launch_bucket_agg_direct(R->d_recs, R->d_cursors, pp.bucket_stride,
+                                             R->d_nrecs, R->d_ncursors, pp.nbucket_stride,
+                                             R->d_groups, R->d_counter, cap_groups,
+                                             R->d_th, pp.sum_slot, pp.agg_count,
+                                             pp.packed_mode, pp.bits_k,
+                                             pp.gmin_k, pp.gmin_v, pp.dshift, R->stream));
+        } else {
+            HIP_CHECK(ytql_launch_bucket_agg(R->d_recs, R->d_cursors, pp.bucket_stride,
+                                             R->d_nrecs, R->d_ncursors, pp.nbucket_stride,
+                                             R->d_groups, R->d_counter, cap_groups,
+                                             R->d_th, pp.sum_slot, pp.agg_count,
+                                             pp.packed_mode, pp.bits_k,
+                                             pp.gmin_k, pp.gmin_v, pp.aligned, R->stream));
+        }
         HIP_CHECK(hipEventRecord(ev2, R->stream));
         HIP_CHECK(hipStreamSynchronize(R->stream));
         float msA = 0, msB = 0;
@@ -1041,9 +1081,17 @@ static int run_scan(const YtPlan* plan, const YtChunk* chunk,
 
     if (fs->valid && fs->key_col >= 0 && fs->nsum <= 1) {
         bool done = false;
+        bool tried_direct = false;
         rc = run_partitioned(plan, chunk, options, R, fs, maxw, stats, &done,
-                             errbuf, errlen);
+                             0, &tried_direct, errbuf, errlen);
         if (rc != YT_OK) return rc;
+        /* direct-span mode overflows on range-clustered key skew; the
+         * hash-partitioned layout spreads those — retry before giving up */
+        if (!done && tried_direct) {
+            rc = run_partitioned(plan, chunk, options, R, fs, maxw, stats, &done,
+                                 1, nullptr, errbuf, errlen);
+            if (rc != YT_OK) return rc;
+        }
         if (done) return YT_OK;
         /* else: capacity guard tripped — fall through to the direct path */
     }

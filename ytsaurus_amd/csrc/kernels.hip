@@ -1562,7 +1562,8 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
     unsigned* gbase = hist + kNB;
     unsigned* nhist = gbase + kNB;
     unsigned* ngbase = nhist + kNB;
-    uint64_t* klds = (uint64_t*)(ngbase + kNB);   /* staged key words */
+    unsigned* ovf = ngbase + kNB;                 /* [0]: this tile hit a guard */
+    uint64_t* klds = (uint64_t*)(ovf + 2);        /* staged key words */
 
     const bool has_filter = pp.filter_idx >= 0;
     const bool has_val = pp.val_idx >= 0;
@@ -1626,6 +1627,7 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
         }
 
         for (int i = tid; i < kNB; i += 256) { hist[i] = 0; nhist[i] = 0; }
+        if (tid == 0) ovf[0] = 0;
         __syncthreads();
 
         const int R = (pp.tile_rows + 255) / 256;
@@ -1641,34 +1643,40 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
             }
             int key_null = kbm && bm_get(kbm, j);
             if (key_null) continue;              /* side rows counted in pass 2 */
-            uint64_t key = (uint64_t)zz_dec(kmin + (bp_get_win(klds, kwd, j, kW0) & kmask));
-            if (key == kEmptyKey) continue;
-            unsigned b = (unsigned)(mix64(key) >> 40) & (kNB - 1);
+            uint64_t kzzfull = kmin + (bp_get_win(klds, kwd, j, kW0) & kmask);
+            if (kzzfull == ~0ULL) continue;      /* zz(INT64_MIN) = kEmptyKey */
+            unsigned b = pp.direct_mode
+                ? (unsigned)((kzzfull - pp.gmin_k) >> pp.dshift)
+                : (unsigned)(mix64((uint64_t)zz_dec(kzzfull)) >> 40) & (kNB - 1);
             if (has_val && vbm && bm_get(vbm, j)) atomicAdd(&nhist[b], 1u);
             else atomicAdd(&hist[b], 1u);
         }
         __syncthreads();
         /* reserve global space per bucket-sub (1 atomic per bucket/tile),
-         * then reuse hist/nhist as the pass-2 claim counters */
+         * then reuse hist/nhist as the pass-2 claim counters. Aligned mode
+         * reserves whole 8-record groups so no HBM line is split between
+         * two reservations (pad fill below). */
         for (int i = tid; i < kNB; i += 256) {
             unsigned c = hist[i];
             if (c) {
-                unsigned long long base = atomicAdd(&cursors[i * 8 + sub], (unsigned long long)c);
-                if ((int64_t)(base + c) > pp.bucket_stride) { th->overflow = 1; base = 0; }
+                unsigned res = pp.aligned ? ((c + 7u) & ~7u) : c;
+                unsigned long long base = atomicAdd(&cursors[i * 8 + sub], (unsigned long long)res);
+                if ((int64_t)(base + res) > pp.bucket_stride) { th->overflow = 1; ovf[0] = 1; base = 0; }
                 gbase[i] = (unsigned)base;
                 hist[i] = 0;
             }
             unsigned nc = nhist[i];
             if (nc) {
-                unsigned long long base = atomicAdd(&ncursors[i * 8 + sub], (unsigned long long)nc);
-                if ((int64_t)(base + nc) > pp.nbucket_stride) { th->overflow = 1; base = 0; }
+                unsigned res = pp.aligned ? ((nc + 7u) & ~7u) : nc;
+                unsigned long long base = atomicAdd(&ncursors[i * 8 + sub], (unsigned long long)res);
+                if ((int64_t)(base + res) > pp.nbucket_stride) { th->overflow = 1; ovf[0] = 1; base = 0; }
                 ngbase[i] = (unsigned)base;
                 nhist[i] = 0;
             }
         }
         __syncthreads();
         uint64_t* recs8 = (uint64_t*)recs;
-        if (th->overflow != 1) {
+        if (ovf[0] != 1) {   /* LDS flag: uniform across the block */
             /* pass 2: claim offset, decode, write */
             #pragma unroll 4
             for (int i = 0; i < R; i++) {
@@ -1700,7 +1708,9 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
                     continue;
                 }
 
-                unsigned b = (unsigned)(mix64(key) >> 40) & (kNB - 1);
+                unsigned b = pp.direct_mode
+                    ? (unsigned)((kzzfull - pp.gmin_k) >> pp.dshift)
+                    : (unsigned)(mix64(key) >> 40) & (kNB - 1);
                 int64_t sb = (int64_t)b * 8 + sub;
                 if (has_val && val_null) {
                     unsigned off = atomicAdd(&nhist[b], 1u);
@@ -1723,6 +1733,35 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
                         make_ulonglong2(key, val);
                 }
             }
+            /* pad the tail of every claim group up to the 8-record boundary
+             * (same lines the real records ended on — completes them) */
+            if (pp.aligned) {
+                __syncthreads();
+                for (int b = tid; b < kNB; b += 256) {
+                    unsigned c = hist[b];
+                    if (c & 7u) {
+                        int64_t sb = (int64_t)b * 8 + sub;
+                        unsigned pad = 8u - (c & 7u);
+                        if (pp.packed_mode) {
+                            for (unsigned p = 0; p < pad; p++)
+                                recs8[sb * pp.bucket_stride + gbase[b] + c + p] =
+                                    0x8000000000000000ULL;
+                        } else {
+                            for (unsigned p = 0; p < pad; p++)
+                                recs[sb * pp.bucket_stride + gbase[b] + c + p] =
+                                    make_ulonglong2(kEmptyKey, 0);
+                        }
+                    }
+                    unsigned nc = nhist[b];
+                    if (nc & 7u) {
+                        int64_t sb = (int64_t)b * 8 + sub;
+                        unsigned pad = 8u - (nc & 7u);
+                        for (unsigned p = 0; p < pad; p++)
+                            nrecs[sb * pp.nbucket_stride + ngbase[b] + nc + p] =
+                                kEmptyKey;
+                    }
+                }
+            }
         }
         __syncthreads();
     }
@@ -1739,7 +1778,8 @@ k_bucket_agg(const ulonglong2* recs, const unsigned long long* cursors,
              int64_t nbucket_stride,
              OutGroup* out, unsigned long long* out_counter, int64_t out_cap,
              TableHdr* th, int sum_slot, int agg_count,
-             int packed_mode, int bits_k, uint64_t gmin_k, uint64_t gmin_v)
+             int packed_mode, int bits_k, uint64_t gmin_k, uint64_t gmin_v,
+             int aligned)
 {
     __shared__ unsigned long long tab[kHSlots * 3];
     const int tid = threadIdx.x;
@@ -1760,12 +1800,19 @@ k_bucket_agg(const ulonglong2* recs, const unsigned long long* cursors,
     const uint64_t* rows8 = (const uint64_t*)recs + ((int64_t)bucket * 8 + sub) * bucket_stride;
     /* 4 records per thread per pass: independent probes overlap LDS latency */
     int64_t i = tid;
+    /* pads (aligned claims): packed = bit 63, 16B = key == kEmptyKey —
+     * both map kv.x to kEmptyKey, which PROBE skips (real INT64_MIN keys
+     * are side-slotted in phase A and never enter the streams) */
     #define LOADKV(kv, idx)                                                  \
         ulonglong2 kv;                                                       \
         if (packed_mode) {                                                   \
             uint64_t r_ = rows8[idx];                                        \
-            kv.x = (uint64_t)zz_dec(gmin_k + (r_ & kmask));                  \
-            kv.y = (uint64_t)zz_dec(gmin_v + (r_ >> bits_k));                \
+            if (aligned && (int64_t)r_ < 0) {                                \
+                kv.x = kEmptyKey; kv.y = 0;                                  \
+            } else {                                                         \
+                kv.x = (uint64_t)zz_dec(gmin_k + (r_ & kmask));              \
+                kv.y = (uint64_t)zz_dec(gmin_v + (r_ >> bits_k));            \
+            }                                                                \
         } else {                                                             \
             kv = rows[idx];                                                  \
         }
@@ -1783,7 +1830,7 @@ k_bucket_agg(const ulonglong2* recs, const unsigned long long* cursors,
         uint64_t s2 = mix64(kv2.x) & (kHSlots - 1);
         uint64_t s3 = mix64(kv3.x) & (kHSlots - 1);
         #define PROBE(kv, sv_)                                               \
-        {                                                                    \
+        if (kv.x != (unsigned long long)kEmptyKey) {                         \
             uint64_t sp = sv_;                                               \
             int found = 0;                                                   \
             for (int it = 0; it < kHSlots; it++) {                           \
@@ -1832,6 +1879,7 @@ k_bucket_agg(const ulonglong2* recs, const unsigned long long* cursors,
         const uint64_t* nrows = nrecs + ((int64_t)bucket * 8 + sub) * nbucket_stride;
         for (int64_t i = tid; i < nn && !full; i += 256) {
             uint64_t key = nrows[i];
+            if (key == kEmptyKey) continue;      /* aligned-claim pad */
             uint64_t s = mix64(key) & (kHSlots - 1);
             int found = 0;
             for (int it = 0; it < kHSlots; it++) {
@@ -1873,6 +1921,120 @@ k_bucket_agg(const ulonglong2* recs, const unsigned long long* cursors,
         if (sum_slot >= 0) {
             g.agg_bits[sum_slot] = tab[i * 3 + 2];
             g.agg_nonnull[sum_slot] = cntnn >> 32;
+        }
+    }
+}
+
+
+/* Phase B, direct-span mode: buckets are key-RANGE slices, so the LDS
+ * "table" is a dense array indexed by (krel - bucket<<dshift) — no hash,
+ * no probe loop, no CAS, no stored keys. Slot = {cnt|nonnull<<32, sum}.
+ * Semantics identical to k_bucket_agg (registry.cpp:1783-1834 insert +
+ * udf/sum.c null-propagating sum); only the table organisation differs. */
+__device__ __forceinline__ uint64_t zz_enc64(int64_t v)
+{
+    return ((uint64_t)v << 1) ^ (uint64_t)(v >> 63);
+}
+
+__global__ void __launch_bounds__(256)
+k_bucket_agg_direct(const ulonglong2* recs, const unsigned long long* cursors,
+                    int64_t bucket_stride,
+                    const uint64_t* nrecs, const unsigned long long* ncursors,
+                    int64_t nbucket_stride,
+                    OutGroup* out, unsigned long long* out_counter, int64_t out_cap,
+                    TableHdr* th, int sum_slot, int agg_count,
+                    int packed_mode, int bits_k, uint64_t gmin_k, uint64_t gmin_v,
+                    int dshift)
+{
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    const int SL = 1 << dshift;
+    unsigned long long* cntnn = (unsigned long long*)smem;   /* lo cnt, hi nonnull */
+    unsigned long long* sums = cntnn + SL;
+    const int tid = threadIdx.x;
+    const int bucket = blockIdx.x;
+
+    for (int i = tid; i < SL; i += 256) { cntnn[i] = 0; sums[i] = 0; }
+    __syncthreads();
+
+    const uint64_t kmask = (bits_k >= 64) ? ~0ULL : ((1ULL << bits_k) - 1);
+    const uint64_t base_rel = (uint64_t)bucket << dshift;
+    const bool has_sum = sum_slot >= 0;
+
+    for (int sub = 0; sub < 8; sub++) {
+        int64_t n = (int64_t)cursors[bucket * 8 + sub];
+        const ulonglong2* rows = recs + ((int64_t)bucket * 8 + sub) * bucket_stride;
+        const uint64_t* rows8 = (const uint64_t*)recs + ((int64_t)bucket * 8 + sub) * bucket_stride;
+        int64_t i = tid;
+        /* 8 records in flight per thread */
+        #define DLOAD(r, idx)                                                \
+            uint64_t r##k; uint64_t r##v;                                    \
+            if (packed_mode) {                                               \
+                uint64_t r_ = rows8[idx];                                    \
+                r##k = ((int64_t)r_ < 0) ? ~0ULL : (r_ & kmask) - base_rel;  \
+                r##v = (uint64_t)zz_dec(gmin_v + (r_ >> bits_k));            \
+            } else {                                                         \
+                ulonglong2 kv = rows[idx];                                   \
+                r##k = (kv.x == kEmptyKey) ? ~0ULL                           \
+                     : (zz_enc64((int64_t)kv.x) - gmin_k) - base_rel;        \
+                r##v = kv.y;                                                 \
+            }
+        #define DACC(r)                                                      \
+            if (r##k != ~0ULL) {                                             \
+                atomicAdd(&cntnn[r##k], 1ULL | (1ULL << 32));                \
+                if (has_sum) atomicAdd(&sums[r##k], r##v);                   \
+            }
+        for (; i + 1792 < n; i += 2048) {
+            DLOAD(a0, i)
+            DLOAD(a1, i + 256)
+            DLOAD(a2, i + 512)
+            DLOAD(a3, i + 768)
+            DLOAD(a4, i + 1024)
+            DLOAD(a5, i + 1280)
+            DLOAD(a6, i + 1536)
+            DLOAD(a7, i + 1792)
+            DACC(a0) DACC(a1) DACC(a2) DACC(a3)
+            DACC(a4) DACC(a5) DACC(a6) DACC(a7)
+        }
+        for (; i < n; i += 256) {
+            DLOAD(b0, i)
+            DACC(b0)
+        }
+        #undef DLOAD
+        #undef DACC
+    }
+    if (nrecs) {
+        for (int sub = 0; sub < 8; sub++) {
+            int64_t nn = (int64_t)ncursors[bucket * 8 + sub];
+            const uint64_t* nrows = nrecs + ((int64_t)bucket * 8 + sub) * nbucket_stride;
+            for (int64_t i = tid; i < nn; i += 256) {
+                uint64_t key = nrows[i];
+                if (key == kEmptyKey) continue;         /* pad */
+                uint64_t idx = (zz_enc64((int64_t)key) - gmin_k) - base_rel;
+                atomicAdd(&cntnn[idx], 1ULL);           /* cnt only: null value */
+            }
+        }
+    }
+    __syncthreads();
+
+    /* flush: slot index IS the key (zz-relative) */
+    for (int i = tid; i < SL; i += 256) {
+        uint64_t cn = cntnn[i];
+        if (!cn) continue;
+        unsigned long long idx = atomicAdd(out_counter, 1ULL);
+        unsigned long long t = atomicAdd(&th->ngroups, 1ULL);
+        if (th->group_limit > 0 && (int64_t)t >= th->group_limit) th->overflow = 2;
+        if ((int64_t)idx >= out_cap) { th->overflow = 1; continue; }
+        OutGroup& g = out[idx];
+        g.key_bits = (uint64_t)zz_dec(gmin_k + base_rel + (uint64_t)i);
+        g.key_meta = 0;
+        g.cnt = cn & 0xFFFFFFFFULL;
+        for (int a = 0; a < agg_count; a++) {
+            g.agg_bits[a] = 0;
+            g.agg_nonnull[a] = 0;
+        }
+        if (sum_slot >= 0) {
+            g.agg_bits[sum_slot] = sums[i];
+            g.agg_nonnull[sum_slot] = cn >> 32;
         }
     }
 }
@@ -2369,13 +2531,33 @@ hipError_t ytql_launch_bucket_agg(const void* recs, const unsigned long long* cu
                                   TableHdr* th, int sum_slot, int agg_count,
                                   int packed_mode, int bits_k,
                                   uint64_t gmin_k, uint64_t gmin_v,
-                                  hipStream_t st)
+                                  int aligned, hipStream_t st)
 {
     hipLaunchKernelGGL(k_bucket_agg, dim3(kNB), dim3(256), 0, st,
                        (const ulonglong2*)recs, cursors, bucket_stride,
                        nrecs, ncursors, nbucket_stride,
                        out, out_counter, out_cap, th, sum_slot, agg_count,
-                       packed_mode, bits_k, gmin_k, gmin_v);
+                       packed_mode, bits_k, gmin_k, gmin_v, aligned);
+    return hipGetLastError();
+}
+
+hipError_t ytql_launch_bucket_agg_direct(const void* recs, const unsigned long long* cursors,
+                                         int64_t bucket_stride,
+                                         const uint64_t* nrecs, const unsigned long long* ncursors,
+                                         int64_t nbucket_stride,
+                                         OutGroup* out, unsigned long long* out_counter,
+                                         int64_t out_cap,
+                                         TableHdr* th, int sum_slot, int agg_count,
+                                         int packed_mode, int bits_k,
+                                         uint64_t gmin_k, uint64_t gmin_v,
+                                         int dshift, hipStream_t st)
+{
+    size_t lds = (size_t)2 * sizeof(unsigned long long) << dshift;
+    hipLaunchKernelGGL(k_bucket_agg_direct, dim3(kNB), dim3(256), lds, st,
+                       (const ulonglong2*)recs, cursors, bucket_stride,
+                       nrecs, ncursors, nbucket_stride,
+                       out, out_counter, out_cap, th, sum_slot, agg_count,
+                       packed_mode, bits_k, gmin_k, gmin_v, dshift);
     return hipGetLastError();
 }
 
