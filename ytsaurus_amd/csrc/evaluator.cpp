@@ -28,6 +28,8 @@ using namespace ytql;
 extern "C" {
 hipError_t ytql_launch_parse_segments(const DevSeg*, int, SegEx*, unsigned*, unsigned*, unsigned long long*, unsigned long long*, hipStream_t);
 hipError_t ytql_launch_scan_nullflags(const DevSeg*, const SegEx*, int, unsigned*, hipStream_t);
+hipError_t ytql_launch_scan_zzrange(const DevSeg*, const SegEx*, int, int, int,
+                                    unsigned long long*, hipStream_t);
 hipError_t ytql_launch_scan_partition(const PartParams*, const DevSeg*, const SegEx*,
                                       const FastCol*, TableHdr*, unsigned long long*,
                                       void*, unsigned long long*, uint64_t*,
@@ -891,11 +893,55 @@ static int run_partitioned(const YtPlan* plan, const YtChunk* chunk,
             while (span) { b++; span >>= 1; }
             return b;
         };
+        auto exact_range = [&](int col) -> int {
+            /* refine the meta-only span (rounded up to min+2^w-1 per
+             * segment) by an exact device scan of the column */
+            uint64_t* d_ex = nullptr;
+            int rc2 = pool_alloc(&d_ex, 2 * sizeof(uint64_t));
+            if (rc2 != YT_OK) return rc2;
+            uint64_t init[2] = { ~0ULL, 0 };
+            hipError_t e;
+            if ((e = hipMemcpyAsync(d_ex, init, sizeof(init),
+                                    hipMemcpyHostToDevice, R->stream)) ||
+                (e = ytql_launch_scan_zzrange(R->d_segs, R->d_segex,
+                                              R->h_off[col], R->h_cnt[col],
+                                              R->col_null_flags[col] != 0,
+                                              (unsigned long long*)d_ex, R->stream)) ||
+                (e = hipMemcpyAsync(init, d_ex, sizeof(init),
+                                    hipMemcpyDeviceToHost, R->stream)) ||
+                (e = hipStreamSynchronize(R->stream))) {
+                g_pool.put(d_ex);
+                set_err(errbuf, errlen, hipGetErrorString(e));
+                return YT_ERR_HIP;
+            }
+            g_pool.put(d_ex);
+            if (init[0] <= init[1]) {   /* any non-null value seen */
+                R->col_zzmin[col] = init[0];
+                R->col_zzmax[col] = init[1];
+            }
+            return YT_OK;
+        };
         uint64_t span_k = R->col_zzmax[fs->key_col] - R->col_zzmin[fs->key_col];
         int bk = bits_of(span_k);
         int bv = 0;
         uint64_t gmin_v = 0;
         if (pp.val_idx >= 0) {
+            uint64_t span_v = R->col_zzmax[fs->sum_col[0]] - R->col_zzmin[fs->sum_col[0]];
+            bv = bits_of(span_v);
+            gmin_v = R->col_zzmin[fs->sum_col[0]];
+        }
+        /* the meta bound overstates each span by at most one bit; scan
+         * exactly when that one bit could enable direct-span mode or the
+         * aligned pad bit (cheap: one streaming read of the column) */
+        if ((bk >= 1 && bk <= 23) && !(bk <= 22 && bk + bv <= 63)) {
+            rc = exact_range(fs->key_col);
+            if (rc != YT_OK) return rc;
+            span_k = R->col_zzmax[fs->key_col] - R->col_zzmin[fs->key_col];
+            bk = bits_of(span_k);
+        }
+        if (pp.val_idx >= 0 && bk + bv >= 64 && bk + bv <= 65) {
+            rc = exact_range(fs->sum_col[0]);
+            if (rc != YT_OK) return rc;
             uint64_t span_v = R->col_zzmax[fs->sum_col[0]] - R->col_zzmin[fs->sum_col[0]];
             bv = bits_of(span_v);
             gmin_v = R->col_zzmin[fs->sum_col[0]];

@@ -301,6 +301,62 @@ __global__ void k_parse_segments(const DevSeg* segs, int nsegs, SegEx* out,
     }
 }
 
+/* exact zigzag-space min/max of one DirectDense int64 column (non-null
+ * values): k_parse_segments' meta-only bound rounds every segment up to
+ * min_value + 2^w - 1, which can overstate the span by one bit — enough to
+ * disable packed records / aligned claims / direct-span mode on data that
+ * actually fits (the headline config: 21+41 = 62 real bits, 22+42 = 64 by
+ * meta). One workgroup per segment; out[0] = min (init ~0), out[1] = max
+ * (init 0). */
+__global__ void __launch_bounds__(256)
+k_scan_zzrange(const DevSeg* segs, const SegEx* segex, int seg_off,
+               int has_nulls, unsigned long long* out)
+{
+    __shared__ unsigned long long red[8];   /* 4 waves x {min,max} */
+    const int si = seg_off + blockIdx.x;
+    const DevSeg& s = segs[si];
+    const SegEx& e = segex[si];
+    const uint64_t* words = s.blob + e.off_values_words;
+    const uint32_t w = e.w_values;
+    const uint64_t mask = (w >= 64) ? ~0ULL : ((1ULL << w) - 1);
+    const uint8_t* bm = has_nulls
+        ? (const uint8_t*)s.blob + e.off_bitmap_bytes : nullptr;
+    uint64_t mn = ~0ULL, mx = 0;
+    const int n = s.row_count;
+    int64_t j = threadIdx.x;
+    for (; j + 768 < n; j += 1024) {
+        uint64_t z0 = bp_gl(words, mask, w, j);
+        uint64_t z1 = bp_gl(words, mask, w, j + 256);
+        uint64_t z2 = bp_gl(words, mask, w, j + 512);
+        uint64_t z3 = bp_gl(words, mask, w, j + 768);
+        if (!bm || !bm_get(bm, j)) { mn = min(mn, z0); mx = max(mx, z0); }
+        if (!bm || !bm_get(bm, j + 256)) { mn = min(mn, z1); mx = max(mx, z1); }
+        if (!bm || !bm_get(bm, j + 512)) { mn = min(mn, z2); mx = max(mx, z2); }
+        if (!bm || !bm_get(bm, j + 768)) { mn = min(mn, z3); mx = max(mx, z3); }
+    }
+    for (; j < n; j += 256) {
+        if (bm && bm_get(bm, j)) continue;
+        uint64_t z = bp_gl(words, mask, w, j);
+        mn = min(mn, z);
+        mx = max(mx, z);
+    }
+    for (int sh = 32; sh >= 1; sh >>= 1) {
+        mn = min(mn, (uint64_t)__shfl_down((long long)mn, sh, 64));
+        mx = max(mx, (uint64_t)__shfl_down((long long)mx, sh, 64));
+    }
+    const int lane = threadIdx.x & 63, wave = threadIdx.x >> 6;
+    if (lane == 0) { red[wave * 2] = mn; red[wave * 2 + 1] = mx; }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        mn = min(min(red[0], red[2]), min(red[4], red[6]));
+        mx = max(max(red[1], red[3]), max(red[5], red[7]));
+        if (mn != ~0ULL) {
+            atomicMin(&out[0], s.min_value + mn);
+            atomicMax(&out[1], s.min_value + mx);
+        }
+    }
+}
+
 /* null-presence scan: one workgroup per DirectDense segment, threads stride
  * the null bitmap words, wave-OR reduce (writer zero-pads to 8 bytes,
  * bitmap.h) */
@@ -2506,6 +2562,15 @@ hipError_t ytql_launch_scan_nullflags(const DevSeg* segs, const SegEx* segex,
 {
     hipLaunchKernelGGL(k_scan_nullflags, dim3(nsegs), dim3(256), 0, st,
                        segs, segex, nsegs, col_null_flags);
+    return hipGetLastError();
+}
+
+hipError_t ytql_launch_scan_zzrange(const DevSeg* segs, const SegEx* segex,
+                                    int seg_off, int seg_cnt, int has_nulls,
+                                    unsigned long long* out, hipStream_t st)
+{
+    hipLaunchKernelGGL(k_scan_zzrange, dim3(seg_cnt), dim3(256), 0, st,
+                       segs, segex, seg_off, has_nulls, out);
     return hipGetLastError();
 }
 
