@@ -955,13 +955,14 @@ static int run_partitioned(const YtPlan* plan, const YtChunk* chunk,
         }
         /* aligned 64B claims need a pad encoding: bit 63 for packed records
          * (spare iff bk+bv <= 63), kEmptyKey for 16B records (always) */
-        pp.aligned = (!pp.packed_mode || pp.bits_k + bv <= 63) ? 1 : 0;
+        int can_pad = (!pp.packed_mode || pp.bits_k + bv <= 63) ? 1 : 0;
+        pp.aligned = can_pad && !getenv("YTQL_NO_ALIGN");
         /* direct-span mode: when the key zigzag span is small, partition by
          * key RANGE and index phase B's per-bucket array directly (the
          * headline config — 1M distinct keys — spans 21 bits). Needs the
          * pad encoding (phase B skips by sign bit), so requires aligned. */
         pp.gmin_k = R->col_zzmin[fs->key_col];
-        if (bk <= 22 && pp.aligned && !force_hash && !getenv("YTQL_NO_DIRECT")) {
+        if (bk <= 22 && can_pad && !force_hash && !getenv("YTQL_NO_DIRECT")) {
             pp.direct_mode = 1;
             pp.dshift = bk > 10 ? bk - 10 : 0;
             if (tried_direct) *tried_direct = true;
