@@ -243,6 +243,13 @@ struct PartParams {
      * directly — no hash probe, no CAS, no stored keys. */
     int32_t direct_mode;
     int32_t dshift;               /* per-bucket key-slot range = 1 << dshift */
+    /* LDS bucket-major reorder: pass 2 writes records into an LDS scratch
+     * (dense bucket-major within the tile), then a copy phase streams each
+     * bucket run to its aligned global claim — every HBM line of the record
+     * stream is written whole, by coalesced adjacent-lane stores (the 8B
+     * scatter wrote 45 GB for 8 GB of records: partial lines evicted from
+     * L2 between stores). Packed records, no val-null stream only. */
+    int32_t reorder;
     /* 64B-aligned record claims: each (tile,bucket) reserves a multiple of 8
      * records and fills the tail with pad records, so every HBM line of the
      * partition stream is written whole by one workgroup within one tile

@@ -954,9 +954,14 @@ static int run_partitioned(const YtPlan* plan, const YtChunk* chunk,
             pp.gmin_v = gmin_v;
         }
         /* aligned 64B claims need a pad encoding: bit 63 for packed records
-         * (spare iff bk+bv <= 63), kEmptyKey for 16B records (always) */
+         * (spare iff bk+bv <= 63), kEmptyKey for 16B records (always).
+         * Pads pay off only with the LDS reorder (coalesced full-line
+         * writes); without it the 8B scatter's partial-line eviction isn't
+         * fixed and pads just inflate the stream (measured r2c_ab). */
         int can_pad = (!pp.packed_mode || pp.bits_k + bv <= 63) ? 1 : 0;
-        pp.aligned = can_pad && !getenv("YTQL_NO_ALIGN");
+        pp.reorder = (pp.packed_mode && can_pad && !pp.has_val_nulls
+                      && !getenv("YTQL_NO_REORDER")) ? 1 : 0;
+        pp.aligned = pp.reorder;
         /* direct-span mode: when the key zigzag span is small, partition by
          * key RANGE and index phase B's per-bucket array directly (the
          * headline config — 1M distinct keys — spans 21 bits). Needs the
@@ -980,17 +985,20 @@ static int run_partitioned(const YtPlan* plan, const YtChunk* chunk,
     /* bigger tiles make aligned claims cheaper (fewer pad tails per
      * bucket); bounded by LDS (key staging) at 2 workgroups/CU (<=80KB) */
     int tile_rows = 16384;
+    bool tile_forced = false;
     {
         const char* ev = getenv("YTQL_TILE");   /* perf experiments only */
-        if (ev && atoi(ev) >= 256) tile_rows = atoi(ev);
+        if (ev && atoi(ev) >= 256) { tile_rows = atoi(ev); tile_forced = true; }
         if (tile_rows > 16384) tile_rows = 16384;   /* per-thread row arrays bound */
     }
     /* LDS: per-tile bucket histograms + the staged key column */
     auto lds_for = [&](int tr) {
-        return (size_t)4 * kNB * 4 + ((size_t)tr * w / 64 + 2) * 8 + 256;
+        return (pp.reorder ? (size_t)tr * 8 : 0)
+             + (size_t)4 * kNB * 4 + ((size_t)tr * w / 64 + 2) * 8 + 256;
     };
     size_t lds = lds_for(tile_rows);
-    while (lds > 80 * 1024 && tile_rows > 4096) { tile_rows >>= 1; lds = lds_for(tile_rows); }
+    while (!tile_forced && lds > 80 * 1024 && tile_rows > 4096) { tile_rows >>= 1; lds = lds_for(tile_rows); }
+    while (lds > 158 * 1024 && tile_rows > 1024) { tile_rows >>= 1; lds = lds_for(tile_rows); }
     while (tile_rows > 256 && tile_rows > seg0_rows) tile_rows >>= 1;
     lds = lds_for(tile_rows);
     pp.tile_rows = tile_rows;

@@ -1614,12 +1614,13 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
     extern __shared__ __attribute__((aligned(16))) char smem[];
     const int tid = threadIdx.x;
 
-    unsigned* hist = (unsigned*)smem;
+    uint64_t* rscratch = (uint64_t*)smem;         /* reorder mode: tile records */
+    unsigned* hist = (unsigned*)(smem + (pp.reorder ? (size_t)pp.tile_rows * 8 : 0));
     unsigned* gbase = hist + kNB;
-    unsigned* nhist = gbase + kNB;
-    unsigned* ngbase = nhist + kNB;
-    unsigned* ovf = ngbase + kNB;                 /* [0]: this tile hit a guard */
-    uint64_t* klds = (uint64_t*)(ovf + 2);        /* staged key words */
+    unsigned* nhist = gbase + kNB;                /* reorder mode: off[] prefix */
+    unsigned* ngbase = nhist + kNB;               /* reorder mode: claim counters */
+    unsigned* ovf = ngbase + kNB;                 /* [0] guard flag, [1] tile total, [2..5] scan */
+    uint64_t* klds = (uint64_t*)(ovf + 8);        /* staged key words */
 
     const bool has_filter = pp.filter_idx >= 0;
     const bool has_val = pp.val_idx >= 0;
@@ -1719,7 +1720,7 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
                 unsigned long long base = atomicAdd(&cursors[i * 8 + sub], (unsigned long long)res);
                 if ((int64_t)(base + res) > pp.bucket_stride) { th->overflow = 1; ovf[0] = 1; base = 0; }
                 gbase[i] = (unsigned)base;
-                hist[i] = 0;
+                if (!pp.reorder) hist[i] = 0;   /* reorder keeps counts for off[]/pads */
             }
             unsigned nc = nhist[i];
             if (nc) {
@@ -1731,6 +1732,31 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
             }
         }
         __syncthreads();
+        if (pp.reorder) {
+            /* exclusive prefix of hist -> nhist (off[]), zero claim counters;
+             * 256 threads x 4 consecutive buckets, wave scan + 4 wave sums */
+            const int lane_ = tid & 63, wave_ = tid >> 6;
+            const int b0 = tid * 4;
+            unsigned c0 = hist[b0], c1 = hist[b0 + 1], c2 = hist[b0 + 2], c3 = hist[b0 + 3];
+            unsigned local = c0 + c1 + c2 + c3;
+            unsigned pre = local;
+            for (int sh = 1; sh < 64; sh <<= 1) {
+                unsigned o = (unsigned)__shfl_up((int)pre, sh, 64);
+                if (lane_ >= sh) pre += o;
+            }
+            if (lane_ == 63) ovf[2 + wave_] = pre;
+            __syncthreads();
+            unsigned woff = 0;
+            for (int w2 = 0; w2 < wave_; w2++) woff += ovf[2 + w2];
+            unsigned excl = woff + pre - local;
+            nhist[b0] = excl;
+            nhist[b0 + 1] = excl + c0;
+            nhist[b0 + 2] = excl + c0 + c1;
+            nhist[b0 + 3] = excl + c0 + c1 + c2;
+            ngbase[b0] = 0; ngbase[b0 + 1] = 0; ngbase[b0 + 2] = 0; ngbase[b0 + 3] = 0;
+            if (tid == 255) ovf[1] = excl + local;
+            __syncthreads();
+        }
         uint64_t* recs8 = (uint64_t*)recs;
         if (ovf[0] != 1) {   /* LDS flag: uniform across the block */
             /* pass 2: claim offset, decode, write */
@@ -1771,6 +1797,14 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
                 if (has_val && val_null) {
                     unsigned off = atomicAdd(&nhist[b], 1u);
                     nrecs[sb * pp.nbucket_stride + ngbase[b] + off] = key;
+                } else if (pp.reorder) {
+                    uint64_t rec = kzzfull - pp.gmin_k;
+                    if (has_val) {
+                        uint64_t vzz = (vmin + bp_gl(vwords, vmask, vwd, j)) - pp.gmin_v;
+                        rec |= vzz << pp.bits_k;
+                    }
+                    unsigned off = atomicAdd(&ngbase[b], 1u);
+                    rscratch[nhist[b] + off] = rec;
                 } else if (pp.packed_mode) {
                     unsigned off = atomicAdd(&hist[b], 1u);
                     uint64_t rec = kzzfull - pp.gmin_k;
@@ -1787,6 +1821,24 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
                     }
                     recs[sb * pp.bucket_stride + gbase[b] + off] =
                         make_ulonglong2(key, val);
+                }
+            }
+            if (pp.reorder) {
+                /* stream the bucket-major tile to the aligned global claims:
+                 * adjacent lanes hit adjacent addresses within each bucket
+                 * run, and runs start 64B-aligned — whole lines only */
+                __syncthreads();
+                const uint64_t pmask = (pp.bits_k >= 64) ? ~0ULL
+                                     : ((1ULL << pp.bits_k) - 1);
+                const int tot = (int)ovf[1];
+                for (int t = tid; t < tot; t += 256) {
+                    uint64_t rec = rscratch[t];
+                    uint64_t krel = rec & pmask;
+                    unsigned b = pp.direct_mode
+                        ? (unsigned)(krel >> pp.dshift)
+                        : (unsigned)(mix64((uint64_t)zz_dec(pp.gmin_k + krel)) >> 40) & (kNB - 1);
+                    int64_t sb = (int64_t)b * 8 + sub;
+                    recs8[sb * pp.bucket_stride + gbase[b] + (t - nhist[b])] = rec;
                 }
             }
             /* pad the tail of every claim group up to the 8-record boundary
