@@ -1860,7 +1860,9 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
                                     make_ulonglong2(kEmptyKey, 0);
                         }
                     }
-                    unsigned nc = nhist[b];
+                    /* null-value stream pads (absent in reorder mode, where
+                     * nhist holds the off[] prefix and nrecs is null) */
+                    unsigned nc = (nrecs && !pp.reorder) ? nhist[b] : 0;
                     if (nc & 7u) {
                         int64_t sb = (int64_t)b * 8 + sub;
                         unsigned pad = 8u - (nc & 7u);
