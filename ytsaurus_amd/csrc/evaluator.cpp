@@ -984,7 +984,9 @@ static int run_partitioned(const YtPlan* plan, const YtChunk* chunk,
     unsigned w = maxw ? maxw : 1;
     /* bigger tiles make aligned claims cheaper (fewer pad tails per
      * bucket); bounded by LDS (key staging) at 2 workgroups/CU (<=80KB) */
-    int tile_rows = 16384;
+    /* reorder: 31x256 rows -> scratch 62KB + 16KB histograms = 2 WGs/CU
+     * (8192 would land 32 bytes over the 80KB half-LDS line) */
+    int tile_rows = pp.reorder ? 7936 : 16384;
     bool tile_forced = false;
     {
         const char* ev = getenv("YTQL_TILE");   /* perf experiments only */
@@ -993,8 +995,10 @@ static int run_partitioned(const YtPlan* plan, const YtChunk* chunk,
     }
     /* LDS: per-tile bucket histograms + the staged key column */
     auto lds_for = [&](int tr) {
-        return (pp.reorder ? (size_t)tr * 8 : 0)
-             + (size_t)4 * kNB * 4 + ((size_t)tr * w / 64 + 2) * 8 + 256;
+        /* reorder mode: record scratch instead of the staged key column */
+        return pp.reorder
+            ? (size_t)tr * 8 + (size_t)4 * kNB * 4 + 64
+            : (size_t)4 * kNB * 4 + ((size_t)tr * w / 64 + 2) * 8 + 256;
     };
     size_t lds = lds_for(tile_rows);
     while (!tile_forced && lds > 80 * 1024 && tile_rows > 4096) { tile_rows >>= 1; lds = lds_for(tile_rows); }

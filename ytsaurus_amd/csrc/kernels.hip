@@ -1659,8 +1659,12 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
         const uint64_t kmin = sgk.min_value;
         const uint8_t* kbm = pp.has_key_nulls
             ? (const uint8_t*)sgk.blob + ek.off_bitmap_bytes : nullptr;
+        const uint64_t* kwords = sgk.blob + ek.off_values_words;
         const int64_t kW0 = ((uint64_t)t0 * kwd) >> 6;
-        {
+        if (!pp.reorder) {
+            /* reorder mode reads keys straight from global both passes —
+             * the per-tile key window is L2-resident on the re-read, and
+             * the freed LDS buys the record scratch at 2 WGs/CU */
             int64_t kW1 = kwd ? ((((uint64_t)t1 * kwd) + 63) >> 6) : 0;
             int64_t nwords = (kwd == 0) ? 0 : (kW1 - kW0 + 1);
             int64_t vec_words = (kwd == 0) ? 0 : (((uint64_t)seg_rows * kwd + 63) >> 6);
@@ -1700,7 +1704,9 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
             }
             int key_null = kbm && bm_get(kbm, j);
             if (key_null) continue;              /* side rows counted in pass 2 */
-            uint64_t kzzfull = kmin + (bp_get_win(klds, kwd, j, kW0) & kmask);
+            uint64_t kraw = pp.reorder ? bp_gl(kwords, kmask, kwd, j)
+                                       : (bp_get_win(klds, kwd, j, kW0) & kmask);
+            uint64_t kzzfull = kmin + kraw;
             if (kzzfull == ~0ULL) continue;      /* zz(INT64_MIN) = kEmptyKey */
             unsigned b = pp.direct_mode
                 ? (unsigned)((kzzfull - pp.gmin_k) >> pp.dshift)
@@ -1773,7 +1779,9 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
                 uint64_t kzzfull = 0;
                 uint64_t key = 0;
                 if (!key_null) {
-                    kzzfull = kmin + (bp_get_win(klds, kwd, j, kW0) & kmask);
+                    kzzfull = kmin + (pp.reorder
+                        ? bp_gl(kwords, kmask, kwd, j)
+                        : (bp_get_win(klds, kwd, j, kW0) & kmask));
                     key = (uint64_t)zz_dec(kzzfull);
                 }
                 int val_null = has_val ? (vbm && bm_get(vbm, j)) : 1;
@@ -1824,26 +1832,29 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
                 }
             }
             if (pp.reorder) {
-                /* stream the bucket-major tile to the aligned global claims:
-                 * adjacent lanes hit adjacent addresses within each bucket
-                 * run, and runs start 64B-aligned — whole lines only */
+                /* stream the bucket-major tile to the aligned global claims,
+                 * pads fused in: one wave owns each bucket run, lanes write
+                 * adjacent records, every 64B line is covered by exactly one
+                 * wave store instruction — no partial lines, no separate pad
+                 * pass re-dirtying lines */
                 __syncthreads();
-                const uint64_t pmask = (pp.bits_k >= 64) ? ~0ULL
-                                     : ((1ULL << pp.bits_k) - 1);
-                const int tot = (int)ovf[1];
-                for (int t = tid; t < tot; t += 256) {
-                    uint64_t rec = rscratch[t];
-                    uint64_t krel = rec & pmask;
-                    unsigned b = pp.direct_mode
-                        ? (unsigned)(krel >> pp.dshift)
-                        : (unsigned)(mix64((uint64_t)zz_dec(pp.gmin_k + krel)) >> 40) & (kNB - 1);
-                    int64_t sb = (int64_t)b * 8 + sub;
-                    recs8[sb * pp.bucket_stride + gbase[b] + (t - nhist[b])] = rec;
+                const int lane_ = tid & 63, wave_ = tid >> 6;
+                for (int b = wave_; b < kNB; b += 4) {
+                    unsigned c = hist[b];
+                    if (!c) continue;
+                    unsigned cp = (c + 7u) & ~7u;
+                    const unsigned offb = nhist[b];
+                    uint64_t* dst = recs8 + ((int64_t)b * 8 + sub) * pp.bucket_stride
+                                  + gbase[b];
+                    for (unsigned r = lane_; r < cp; r += 64) {
+                        dst[r] = (r < c) ? rscratch[offb + r]
+                                         : 0x8000000000000000ULL;
+                    }
                 }
             }
             /* pad the tail of every claim group up to the 8-record boundary
              * (same lines the real records ended on — completes them) */
-            if (pp.aligned) {
+            if (pp.aligned && !pp.reorder) {
                 __syncthreads();
                 for (int b = tid; b < kNB; b += 256) {
                     unsigned c = hist[b];
