@@ -959,8 +959,11 @@ static int run_partitioned(const YtPlan* plan, const YtChunk* chunk,
          * writes); without it the 8B scatter's partial-line eviction isn't
          * fixed and pads just inflate the stream (measured r2c_ab). */
         int can_pad = (!pp.packed_mode || pp.bits_k + bv <= 63) ? 1 : 0;
+        /* measured slower than the plain scatter at every tile size
+         * (r2 A/Bs: 23.8-41 ms vs 18.7 ms) — L2 absorbs the 8B scatter
+         * better than any LDS round-trip; keep as an experiment knob */
         pp.reorder = (pp.packed_mode && can_pad && !pp.has_val_nulls
-                      && !getenv("YTQL_NO_REORDER")) ? 1 : 0;
+                      && getenv("YTQL_REORDER")) ? 1 : 0;
         pp.aligned = pp.reorder;
         /* direct-span mode: when the key zigzag span is small, partition by
          * key RANGE and index phase B's per-bucket array directly (the
@@ -3007,8 +3010,11 @@ extern "C" int yt_gpu_query_execute(
                      options->group_row_limit, errbuf, errlen);
     if (rc) return rc;
 
+    const bool timing = getenv("YTQL_TIMING") != nullptr;   /* phase breakdown to stderr */
+    double tp0 = now_ms();
     rc = run_scan(plan, chunk, options, &R, &dp, jd, &fs, maxw, stats, errbuf, errlen);
     if (rc) return rc;
+    double tp1 = now_ms();
 
     /* compact + readback (pinned staging from the pool) */
     TableHdr th;
@@ -3032,6 +3038,7 @@ extern "C" int yt_gpu_query_execute(
             HIP_CHECK(hipStreamSynchronize(R.stream));
         }
     }
+    double tp2 = now_ms();
     {
         std::vector<uint64_t> gaccum(1 + 2 * kMaxAggs, 0);
         if (fs.valid && fs.key_col < 0) {
@@ -3046,6 +3053,11 @@ extern "C" int yt_gpu_query_execute(
         g_pool.put(hgroups);
         if (rc) return rc;
         if (out_limited && stats) stats->incomplete_output = 1;
+    }
+    if (timing) {
+        double tp3 = now_ms();
+        fprintf(stderr, "[ytql timing] pre-scan %.2fms scan-wall %.2fms compact+D2H %.2fms emit %.2fms\n",
+                tp0 - tw0, tp1 - tp0, tp2 - tp1, tp3 - tp2);
     }
     if (plan->order_count > 0 || plan->with_totals || plan->having) {
         /* grouped output is already bounded: order/totals/having on the host
