@@ -2897,6 +2897,7 @@ extern "C" int yt_gpu_query_execute(
     if (!plan || !chunk || !output) { set_err(errbuf, errlen, "null argument"); return YT_ERR_INVALID_PLAN; }
 
     double tw0 = now_ms();
+    double tp0 = 0, tp1 = 0, tp2 = 0;   /* phase checkpoints (YTQL_TIMING) */
     YtExecOptions defopt;
     memset(&defopt, 0, sizeof(defopt));
     if (!options) options = &defopt;
@@ -3011,10 +3012,10 @@ extern "C" int yt_gpu_query_execute(
     if (rc) return rc;
 
     const bool timing = getenv("YTQL_TIMING") != nullptr;   /* phase breakdown to stderr */
-    double tp0 = now_ms();
+    tp0 = now_ms();
     rc = run_scan(plan, chunk, options, &R, &dp, jd, &fs, maxw, stats, errbuf, errlen);
     if (rc) return rc;
-    double tp1 = now_ms();
+    tp1 = now_ms();
 
     /* compact + readback (pinned staging from the pool) */
     TableHdr th;
@@ -3038,7 +3039,7 @@ extern "C" int yt_gpu_query_execute(
             HIP_CHECK(hipStreamSynchronize(R.stream));
         }
     }
-    double tp2 = now_ms();
+    tp2 = now_ms();
     {
         std::vector<uint64_t> gaccum(1 + 2 * kMaxAggs, 0);
         if (fs.valid && fs.key_col < 0) {
