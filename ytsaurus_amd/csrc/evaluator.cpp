@@ -18,6 +18,7 @@
 #include <mutex>
 #include <chrono>
 #include <thread>
+#include <atomic>
 
 #include "common.h"
 #include "../../include/ytql_gpu.h"
@@ -1064,6 +1065,10 @@ static int run_partitioned(const YtPlan* plan, const YtChunk* chunk,
         HIP_CHECK(hipEventCreate(&ev1));
         HIP_CHECK(hipEventCreate(&ev2));
         int grid = pp.ntiles < 2048 ? (pp.ntiles ? pp.ntiles : 1) : 2048;
+        {
+            const char* gv = getenv("YTQL_GRID");   /* perf experiments only */
+            if (gv && atoi(gv) >= 8 && atoi(gv) < grid) grid = atoi(gv) & ~7;
+        }
         HIP_CHECK(hipEventRecord(ev0, R->stream));
         HIP_CHECK(ytql_launch_scan_partition(&pp, R->d_segs, R->d_segex, R->d_fastcols,
                                              R->d_th, R->d_cursors, R->d_recs,
@@ -1376,14 +1381,8 @@ static int emit_rows(const YtPlan* plan, const YtChunk* chunk,
     int base_cols = (kp ? kp : (has_key ? 1 : 0)) + plan->agg_count;
     int out_cols = plan->project_count ? plan->project_count : base_cols;
 
-    auto emit = [&](uint64_t key_bits, int key_null, uint64_t cnt,
-                    const uint64_t* ab, const uint64_t* an) -> int {
-        if (output_row_limit > 0 && output->row_count >= output_row_limit) {
-            /* OutputRowLimit: soft stop — registry.cpp:297-305 WriteRow */
-            if (out_limited) *out_limited = 1;
-            return -1;
-        }
-        if (output->row_count >= output->capacity_rows) return YT_ERR_CAPACITY;
+    auto emit_at = [&](YtValue* dst, uint64_t key_bits, int key_null,
+                       uint64_t cnt, const uint64_t* ab, const uint64_t* an) -> int {
         HVal row[kMaxPackKeys + 1 + kMaxAggs];
         int nrow = 0;
         if (kp) {
@@ -1410,7 +1409,6 @@ static int emit_rows(const YtPlan* plan, const YtChunk* chunk,
         } else
         finalize_row(plan, key_type, sum_type, key_bits, key_null, cnt, ab, an,
                      row, &nrow, has_key);
-        YtValue* dst = output->values + output->row_count * out_cols;
         if (plan->project_count) {
             for (int p = 0; p < plan->project_count; p++) {
                 HVal v;
@@ -1431,8 +1429,20 @@ static int emit_rows(const YtPlan* plan, const YtChunk* chunk,
                 dst[i].data.bits = row[i].bits;
             }
         }
-        output->row_count++;
         return YT_OK;
+    };
+    auto emit = [&](uint64_t key_bits, int key_null, uint64_t cnt,
+                    const uint64_t* ab, const uint64_t* an) -> int {
+        if (output_row_limit > 0 && output->row_count >= output_row_limit) {
+            /* OutputRowLimit: soft stop — registry.cpp:297-305 WriteRow */
+            if (out_limited) *out_limited = 1;
+            return -1;
+        }
+        if (output->row_count >= output->capacity_rows) return YT_ERR_CAPACITY;
+        int rc_ = emit_at(output->values + output->row_count * out_cols,
+                          key_bits, key_null, cnt, ab, an);
+        if (rc_ == YT_OK) output->row_count++;
+        return rc_;
     };
 
     output->row_count = 0;
@@ -1464,11 +1474,46 @@ static int emit_rows(const YtPlan* plan, const YtChunk* chunk,
         return rc == -1 ? YT_OK : rc;
     }
 
-    for (int64_t i = 0; i < ngroups; i++) {
-        const OutGroup& g = groups[i];
-        rc = emit(g.key_bits, (int)(g.key_meta & 1), g.cnt, g.agg_bits, g.agg_nonnull);
-        if (rc == -1) return YT_OK;
-        if (rc) return rc;
+    {
+        const int64_t total_sides = (int64_t)(th.side_used[0] != 0)
+                                  + (int64_t)(th.side_used[1] != 0);
+        const bool par_ok = ngroups >= 65536
+            && (output_row_limit <= 0 ||
+                output_row_limit >= ngroups + total_sides)
+            && ngroups + total_sides <= output->capacity_rows;
+        if (par_ok) {
+            /* row i is group i — emit in parallel (group order preserved;
+             * the host loop was 5.5 ms of the 31 ms step at 1M groups) */
+            std::atomic<int> arc{YT_OK};
+            int nt = (int)std::min<int64_t>(std::thread::hardware_concurrency(),
+                                            (ngroups + 131071) / 131072);
+            if (nt > 32) nt = 32;
+            if (nt < 1) nt = 1;
+            std::vector<std::thread> ths;
+            for (int t = 0; t < nt; t++) {
+                ths.emplace_back([&, t]() {
+                    int64_t lo = ngroups * t / nt, hi = ngroups * (t + 1) / nt;
+                    for (int64_t i = lo; i < hi; i++) {
+                        const OutGroup& g = groups[i];
+                        int r2 = emit_at(output->values + i * out_cols,
+                                         g.key_bits, (int)(g.key_meta & 1),
+                                         g.cnt, g.agg_bits, g.agg_nonnull);
+                        if (r2 != YT_OK) { arc.store(r2); return; }
+                    }
+                });
+            }
+            for (auto& t : ths) t.join();
+            if (arc.load() != YT_OK) return arc.load();
+            output->row_count = ngroups;
+        } else {
+            for (int64_t i = 0; i < ngroups; i++) {
+                const OutGroup& g = groups[i];
+                rc = emit(g.key_bits, (int)(g.key_meta & 1), g.cnt,
+                          g.agg_bits, g.agg_nonnull);
+                if (rc == -1) return YT_OK;
+                if (rc) return rc;
+            }
+        }
     }
     /* side groups: the in-table sentinel key, then the null key */
     for (int side = 0; side < 2; side++) {

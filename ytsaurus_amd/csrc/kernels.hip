@@ -324,15 +324,13 @@ k_scan_zzrange(const DevSeg* segs, const SegEx* segex, int seg_off,
     uint64_t mn = ~0ULL, mx = 0;
     const int n = s.row_count;
     int64_t j = threadIdx.x;
-    for (; j + 768 < n; j += 1024) {
-        uint64_t z0 = bp_gl(words, mask, w, j);
-        uint64_t z1 = bp_gl(words, mask, w, j + 256);
-        uint64_t z2 = bp_gl(words, mask, w, j + 512);
-        uint64_t z3 = bp_gl(words, mask, w, j + 768);
-        if (!bm || !bm_get(bm, j)) { mn = min(mn, z0); mx = max(mx, z0); }
-        if (!bm || !bm_get(bm, j + 256)) { mn = min(mn, z1); mx = max(mx, z1); }
-        if (!bm || !bm_get(bm, j + 512)) { mn = min(mn, z2); mx = max(mx, z2); }
-        if (!bm || !bm_get(bm, j + 768)) { mn = min(mn, z3); mx = max(mx, z3); }
+    for (; j + 1792 < n; j += 2048) {
+        #pragma unroll
+        for (int u = 0; u < 8; u++) {
+            int64_t jj = j + u * 256;
+            uint64_t z = bp_gl(words, mask, w, jj);
+            if (!bm || !bm_get(bm, jj)) { mn = min(mn, z); mx = max(mx, z); }
+        }
     }
     for (; j < n; j += 256) {
         if (bm && bm_get(bm, j)) continue;
