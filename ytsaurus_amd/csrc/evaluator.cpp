@@ -1024,6 +1024,9 @@ static int run_partitioned(const YtPlan* plan, const YtChunk* chunk,
         int64_t tiles_per_sub = ((int64_t)pp.tiles_per_seg * R->h_cnt[fs->key_col] + 7) / 8;
         pp.bucket_stride += 4 * tiles_per_sub + 4096;
     }
+    /* every (bucket,sub) region must start 64B-aligned or aligned claims
+     * land mid-line and every run straddles lines */
+    pp.bucket_stride = (pp.bucket_stride + 7) & ~(int64_t)7;
     pp.nbucket_stride = pp.has_val_nulls ? pp.bucket_stride : 0;
 
     std::vector<FastCol> fc(nused);
