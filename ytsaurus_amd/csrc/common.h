@@ -250,6 +250,7 @@ struct PartParams {
      * scatter wrote 45 GB for 8 GB of records: partial lines evicted from
      * L2 between stores). Packed records, no val-null stream only. */
     int32_t reorder;
+    int32_t store_mode;           /* 0 plain, 1 sc1 write-through, 2 no store (timing floor probe) */
     /* 64B-aligned record claims: each (tile,bucket) reserves a multiple of 8
      * records and fills the tail with pad records, so every HBM line of the
      * partition stream is written whole by one workgroup within one tile

@@ -966,6 +966,10 @@ static int run_partitioned(const YtPlan* plan, const YtChunk* chunk,
         pp.reorder = (pp.packed_mode && can_pad && !pp.has_val_nulls
                       && getenv("YTQL_REORDER")) ? 1 : 0;
         pp.aligned = pp.reorder;
+        {
+            const char* sm = getenv("YTQL_STORE");   /* perf experiments only */
+            if (sm) pp.store_mode = atoi(sm);
+        }
         /* direct-span mode: when the key zigzag span is small, partition by
          * key RANGE and index phase B's per-bucket array directly (the
          * headline config — 1M distinct keys — spans 21 bits). Needs the

@@ -1818,7 +1818,18 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
                         uint64_t vzz = (vmin + bp_gl(vwords, vmask, vwd, j)) - pp.gmin_v;
                         rec |= vzz << pp.bits_k;
                     }
-                    recs8[sb * pp.bucket_stride + gbase[b] + off] = rec;
+                    uint64_t* dst = &recs8[sb * pp.bucket_stride + gbase[b] + off];
+                    if (pp.store_mode == 1) {
+                        /* write-through (sc1): drops the line from L2 — no
+                         * partial-line RMW fill on eviction */
+                        __hip_atomic_store((unsigned long long*)dst, rec,
+                                           __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+                    } else if (pp.store_mode == 2) {
+                        /* timing-floor probe: keep rec alive, skip the store */
+                        asm volatile("" :: "v"(rec));
+                    } else {
+                        *dst = rec;
+                    }
                 } else {
                     unsigned off = atomicAdd(&hist[b], 1u);
                     uint64_t val = 0;
