@@ -1691,7 +1691,7 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
 
         const int R = (pp.tile_rows + 255) / 256;
         /* pass 1: COUNT per bucket (no returns, no per-row state) */
-        #pragma unroll 4
+        #pragma unroll 8
         for (int i = 0; i < R; i++) {
             int64_t j = t0 + (int64_t)i * 256 + tid;
             if (j >= t1) continue;
@@ -1764,7 +1764,7 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
         uint64_t* recs8 = (uint64_t*)recs;
         if (ovf[0] != 1) {   /* LDS flag: uniform across the block */
             /* pass 2: claim offset, decode, write */
-            #pragma unroll 4
+            #pragma unroll 8
             for (int i = 0; i < R; i++) {
                 int64_t j = t0 + (int64_t)i * 256 + tid;
                 if (j >= t1) continue;
@@ -2095,42 +2095,55 @@ k_bucket_agg_direct(const ulonglong2* recs, const unsigned long long* cursors,
         const ulonglong2* rows = recs + ((int64_t)bucket * 8 + sub) * bucket_stride;
         const uint64_t* rows8 = (const uint64_t*)recs + ((int64_t)bucket * 8 + sub) * bucket_stride;
         int64_t i = tid;
-        /* 8 records in flight per thread */
-        #define DLOAD(r, idx)                                                \
-            uint64_t r##k; uint64_t r##v;                                    \
-            if (packed_mode) {                                               \
-                uint64_t r_ = rows8[idx];                                    \
-                r##k = ((int64_t)r_ < 0) ? ~0ULL : (r_ & kmask) - base_rel;  \
-                r##v = (uint64_t)zz_dec(gmin_v + (r_ >> bits_k));            \
-            } else {                                                         \
-                ulonglong2 kv = rows[idx];                                   \
-                r##k = (kv.x == kEmptyKey) ? ~0ULL                           \
-                     : (zz_enc64((int64_t)kv.x) - gmin_k) - base_rel;        \
-                r##v = kv.y;                                                 \
+        #define DACCR(r_)                                                    \
+            {                                                                \
+                uint64_t k_ = ((int64_t)(r_) < 0 && packed_mode)             \
+                    ? ~0ULL : ((r_) & kmask) - base_rel;                     \
+                if (k_ != ~0ULL) {                                           \
+                    atomicAdd(&cntnn[k_], 1ULL | (1ULL << 32));              \
+                    if (has_sum)                                             \
+                        atomicAdd(&sums[k_],                                 \
+                                  (uint64_t)zz_dec(gmin_v + ((r_) >> bits_k))); \
+                }                                                            \
             }
-        #define DACC(r)                                                      \
-            if (r##k != ~0ULL) {                                             \
-                atomicAdd(&cntnn[r##k], 1ULL | (1ULL << 32));                \
-                if (has_sum) atomicAdd(&sums[r##k], r##v);                   \
+        if (packed_mode) {
+            /* 16B paired loads: 8 in flight = 16 records per thread pass */
+            const ulonglong2* rows2 = (const ulonglong2*)rows8;
+            int64_t n2 = n >> 1;
+            for (; i + 1792 < n2; i += 2048) {
+                ulonglong2 p0 = rows2[i];
+                ulonglong2 p1 = rows2[i + 256];
+                ulonglong2 p2 = rows2[i + 512];
+                ulonglong2 p3 = rows2[i + 768];
+                ulonglong2 p4 = rows2[i + 1024];
+                ulonglong2 p5 = rows2[i + 1280];
+                ulonglong2 p6 = rows2[i + 1536];
+                ulonglong2 p7 = rows2[i + 1792];
+                DACCR(p0.x) DACCR(p0.y) DACCR(p1.x) DACCR(p1.y)
+                DACCR(p2.x) DACCR(p2.y) DACCR(p3.x) DACCR(p3.y)
+                DACCR(p4.x) DACCR(p4.y) DACCR(p5.x) DACCR(p5.y)
+                DACCR(p6.x) DACCR(p6.y) DACCR(p7.x) DACCR(p7.y)
             }
-        for (; i + 1792 < n; i += 2048) {
-            DLOAD(a0, i)
-            DLOAD(a1, i + 256)
-            DLOAD(a2, i + 512)
-            DLOAD(a3, i + 768)
-            DLOAD(a4, i + 1024)
-            DLOAD(a5, i + 1280)
-            DLOAD(a6, i + 1536)
-            DLOAD(a7, i + 1792)
-            DACC(a0) DACC(a1) DACC(a2) DACC(a3)
-            DACC(a4) DACC(a5) DACC(a6) DACC(a7)
+            for (; i < n2; i += 256) {
+                ulonglong2 pz = rows2[i];
+                DACCR(pz.x) DACCR(pz.y)
+            }
+            if ((n & 1) && tid == 0) {
+                uint64_t r_ = rows8[n - 1];
+                DACCR(r_)
+            }
+        } else {
+            for (; i < n; i += 256) {
+                ulonglong2 kv = rows[i];
+                uint64_t k_ = (kv.x == kEmptyKey) ? ~0ULL
+                     : (zz_enc64((int64_t)kv.x) - gmin_k) - base_rel;
+                if (k_ != ~0ULL) {
+                    atomicAdd(&cntnn[k_], 1ULL | (1ULL << 32));
+                    if (has_sum) atomicAdd(&sums[k_], kv.y);
+                }
+            }
         }
-        for (; i < n; i += 256) {
-            DLOAD(b0, i)
-            DACC(b0)
-        }
-        #undef DLOAD
-        #undef DACC
+        #undef DACCR
     }
     if (nrecs) {
         for (int sub = 0; sub < 8; sub++) {
