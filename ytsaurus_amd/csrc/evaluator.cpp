@@ -993,8 +993,10 @@ static int run_partitioned(const YtPlan* plan, const YtChunk* chunk,
     /* bigger tiles make aligned claims cheaper (fewer pad tails per
      * bucket); bounded by LDS (key staging) at 2 workgroups/CU (<=80KB) */
     /* reorder: 31x256 rows -> scratch 62KB + 16KB histograms = 2 WGs/CU
-     * (8192 would land 32 bytes over the 80KB half-LDS line) */
-    int tile_rows = pp.reorder ? 7936 : 16384;
+     * (8192 would land 32 bytes over the 80KB half-LDS line).
+     * Scatter mode: 4096 measured best (27KB LDS -> 5 WGs/CU; the kernel is
+     * 73% WAIT_ANY, occupancy is the lever — r2 tile sweeps). */
+    int tile_rows = pp.reorder ? 7936 : 4096;
     bool tile_forced = false;
     {
         const char* ev = getenv("YTQL_TILE");   /* perf experiments only */

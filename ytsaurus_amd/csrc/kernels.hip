@@ -1765,7 +1765,7 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
         if (ovf[0] != 1) {   /* LDS flag: uniform across the block */
             /* pass 2: claim offset, decode, write */
             #pragma unroll 8
-            for (int i = 0; i < R; i++) {
+            for (int i = 0; i < R && pp.store_mode != 3; i++) {
                 int64_t j = t0 + (int64_t)i * 256 + tid;
                 if (j >= t1) continue;
                 if (has_filter) {
