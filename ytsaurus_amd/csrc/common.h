@@ -251,6 +251,13 @@ struct PartParams {
      * L2 between stores). Packed records, no val-null stream only. */
     int32_t reorder;
     int32_t store_mode;           /* 0 plain, 1 sc1 write-through, 2 no store (timing floor probe) */
+    /* per-WORKGROUP record regions: region = bucket * gridDim + blockIdx,
+     * appended locally (LDS running base, init once per WG) — the per-tile
+     * global cursor reserve was ~1 atomic per 4 rows (250M global atomics
+     * per 1B-row query vs the ~30 G/s chip ceiling = ~8 ms of the pass).
+     * cursors[] becomes the per-(bucket,wg) final counts, published at WG
+     * end. Requires !has_val_nulls (the null stream keeps 8 XCD subs). */
+    int32_t wg_streams;
     /* 64B-aligned record claims: each (tile,bucket) reserves a multiple of 8
      * records and fills the tail with pad records, so every HBM line of the
      * partition stream is written whole by one workgroup within one tile

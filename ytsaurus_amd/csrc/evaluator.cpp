@@ -39,12 +39,12 @@ hipError_t ytql_launch_bucket_agg(const void*, const unsigned long long*, int64_
                                   const uint64_t*, const unsigned long long*, int64_t,
                                   OutGroup*, unsigned long long*, int64_t,
                                   TableHdr*, int, int, int, int,
-                                  uint64_t, uint64_t, int, hipStream_t);
+                                  uint64_t, uint64_t, int, int, hipStream_t);
 hipError_t ytql_launch_bucket_agg_direct(const void*, const unsigned long long*, int64_t,
                                          const uint64_t*, const unsigned long long*, int64_t,
                                          OutGroup*, unsigned long long*, int64_t,
                                          TableHdr*, int, int, int, int,
-                                         uint64_t, uint64_t, int, hipStream_t);
+                                         uint64_t, uint64_t, int, int, hipStream_t);
 hipError_t ytql_launch_topk_hist(const DevPlan*, const DevSeg*, const SegEx*,
                                  const int32_t*, const int32_t*, int64_t,
                                  const JoinDev*,
@@ -970,6 +970,8 @@ static int run_partitioned(const YtPlan* plan, const YtChunk* chunk,
             const char* sm = getenv("YTQL_STORE");   /* perf experiments only */
             if (sm) pp.store_mode = atoi(sm);
         }
+        pp.wg_streams = (!pp.reorder && !pp.has_val_nulls
+                         && !getenv("YTQL_NO_WGSTREAMS")) ? 1 : 0;
         /* direct-span mode: when the key zigzag span is small, partition by
          * key RANGE and index phase B's per-bucket array directly (the
          * headline config — 1M distinct keys — spans 21 bits). Needs the
@@ -1022,13 +1024,27 @@ static int run_partitioned(const YtPlan* plan, const YtChunk* chunk,
         int32_t last_rows = R->h_segs[R->h_off[fs->key_col] + nseg - 1].row_count;
         pp.ntiles = (nseg - 1) * pp.tiles_per_seg + (last_rows + tile_rows - 1) / tile_rows;
     }
-    /* per (bucket, XCD) sub-streams: 8x more cursors, 1/8 the rows each.
-     * Aligned mode adds up to 7 pad records per (tile,bucket,sub) claim —
-     * budget the expected pad tail (~3.5 per claiming tile) explicitly. */
-    pp.bucket_stride = rows / (kNB * 8) + (rows / (kNB * 8)) / 2 + 4096;
-    if (pp.aligned) {
-        int64_t tiles_per_sub = ((int64_t)pp.tiles_per_seg * R->h_cnt[fs->key_col] + 7) / 8;
-        pp.bucket_stride += 4 * tiles_per_sub + 4096;
+    int grid = pp.ntiles < 2048 ? (pp.ntiles ? pp.ntiles : 1) : 2048;
+    {
+        const char* gv = getenv("YTQL_GRID");   /* perf experiments only */
+        if (gv && atoi(gv) >= 8 && atoi(gv) < grid) grid = atoi(gv) & ~7;
+        if (grid < 1) grid = 1;
+    }
+    int nsub = pp.wg_streams ? grid : 8;
+    if (pp.wg_streams) {
+        /* per-(bucket, WG) regions: rows this WG scans / kNB + skew slack */
+        int64_t tiles_per_wg = (pp.ntiles + grid - 1) / grid;
+        int64_t rows_per_wg = tiles_per_wg * pp.tile_rows;
+        pp.bucket_stride = rows_per_wg / kNB + (rows_per_wg / kNB) / 2 + 64;
+    } else {
+        /* per (bucket, XCD) sub-streams: 8x more cursors, 1/8 the rows each.
+         * Aligned mode adds up to 7 pad records per (tile,bucket,sub) claim —
+         * budget the expected pad tail (~3.5 per claiming tile) explicitly. */
+        pp.bucket_stride = rows / (kNB * 8) + (rows / (kNB * 8)) / 2 + 4096;
+        if (pp.aligned) {
+            int64_t tiles_per_sub = ((int64_t)pp.tiles_per_seg * R->h_cnt[fs->key_col] + 7) / 8;
+            pp.bucket_stride += 4 * tiles_per_sub + 4096;
+        }
     }
     /* every (bucket,sub) region must start 64B-aligned or aligned claims
      * land mid-line and every run straddles lines */
@@ -1045,10 +1061,10 @@ static int run_partitioned(const YtPlan* plan, const YtChunk* chunk,
         HIP_CHECK(pool_alloc(&R->d_fastcols, sizeof(FastCol) * nused));
         HIP_CHECK(hipMemcpyAsync(R->d_fastcols, fc.data(), sizeof(FastCol) * nused,
                                  hipMemcpyHostToDevice, R->stream));
-        HIP_CHECK(pool_alloc(&R->d_cursors, sizeof(uint64_t) * kNB * 8));
-        HIP_CHECK(hipMemsetAsync(R->d_cursors, 0, sizeof(uint64_t) * kNB * 8, R->stream));
+        HIP_CHECK(pool_alloc(&R->d_cursors, sizeof(uint64_t) * kNB * nsub));
+        HIP_CHECK(hipMemsetAsync(R->d_cursors, 0, sizeof(uint64_t) * kNB * nsub, R->stream));
         HIP_CHECK(pool_alloc(&R->d_recs,
-                             (size_t)kNB * 8 * pp.bucket_stride * (pp.packed_mode ? 8 : 16)));
+                             (size_t)kNB * nsub * pp.bucket_stride * (pp.packed_mode ? 8 : 16)));
         if (pp.has_val_nulls) {
             HIP_CHECK(pool_alloc(&R->d_ncursors, sizeof(uint64_t) * kNB * 8));
             HIP_CHECK(hipMemsetAsync(R->d_ncursors, 0, sizeof(uint64_t) * kNB * 8, R->stream));
@@ -1073,11 +1089,6 @@ static int run_partitioned(const YtPlan* plan, const YtChunk* chunk,
         HIP_CHECK(hipEventCreate(&ev0));
         HIP_CHECK(hipEventCreate(&ev1));
         HIP_CHECK(hipEventCreate(&ev2));
-        int grid = pp.ntiles < 2048 ? (pp.ntiles ? pp.ntiles : 1) : 2048;
-        {
-            const char* gv = getenv("YTQL_GRID");   /* perf experiments only */
-            if (gv && atoi(gv) >= 8 && atoi(gv) < grid) grid = atoi(gv) & ~7;
-        }
         HIP_CHECK(hipEventRecord(ev0, R->stream));
         HIP_CHECK(ytql_launch_scan_partition(&pp, R->d_segs, R->d_segex, R->d_fastcols,
                                              R->d_th, R->d_cursors, R->d_recs,
@@ -1090,14 +1101,14 @@ static int run_partitioned(const YtPlan* plan, const YtChunk* chunk,
                                              R->d_groups, R->d_counter, cap_groups,
                                              R->d_th, pp.sum_slot, pp.agg_count,
                                              pp.packed_mode, pp.bits_k,
-                                             pp.gmin_k, pp.gmin_v, pp.dshift, R->stream));
+                                             pp.gmin_k, pp.gmin_v, pp.dshift, nsub, R->stream));
         } else {
             HIP_CHECK(ytql_launch_bucket_agg(R->d_recs, R->d_cursors, pp.bucket_stride,
                                              R->d_nrecs, R->d_ncursors, pp.nbucket_stride,
                                              R->d_groups, R->d_counter, cap_groups,
                                              R->d_th, pp.sum_slot, pp.agg_count,
                                              pp.packed_mode, pp.bits_k,
-                                             pp.gmin_k, pp.gmin_v, pp.aligned, R->stream));
+                                             pp.gmin_k, pp.gmin_v, pp.aligned, nsub, R->stream));
         }
         HIP_CHECK(hipEventRecord(ev2, R->stream));
         HIP_CHECK(hipStreamSynchronize(R->stream));

@@ -1626,6 +1626,12 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
      * record regions (cross-XCD partial-line sharing measured 3.7x write
      * amplification) */
     const int sub = blockIdx.x & 7;
+    if (pp.wg_streams) {
+        /* ngbase doubles as the per-bucket running base of this WG's own
+         * region — persistent across tiles */
+        for (int i = tid; i < kNB; i += 256) ngbase[i] = 0;
+        __syncthreads();
+    }
 
     for (int tile = blockIdx.x; tile < pp.ntiles; tile += gridDim.x) {
         const int seg_idx = tile / pp.tiles_per_seg;
@@ -1717,6 +1723,19 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
          * then reuse hist/nhist as the pass-2 claim counters. Aligned mode
          * reserves whole 8-record groups so no HBM line is split between
          * two reservations (pad fill below). */
+        if (pp.wg_streams) {
+            /* local append into this WG's own region: no global atomics */
+            for (int i = tid; i < kNB; i += 256) {
+                unsigned c = hist[i];
+                if (c) {
+                    unsigned base = ngbase[i];
+                    if ((int64_t)(base + c) > pp.bucket_stride) { th->overflow = 1; ovf[0] = 1; base = 0; }
+                    gbase[i] = base;
+                    ngbase[i] = base + c;
+                    hist[i] = 0;
+                }
+            }
+        } else {
         for (int i = tid; i < kNB; i += 256) {
             unsigned c = hist[i];
             if (c) {
@@ -1734,6 +1753,7 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
                 ngbase[i] = (unsigned)base;
                 nhist[i] = 0;
             }
+        }
         }
         __syncthreads();
         if (pp.reorder) {
@@ -1799,7 +1819,9 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
                 unsigned b = pp.direct_mode
                     ? (unsigned)((kzzfull - pp.gmin_k) >> pp.dshift)
                     : (unsigned)(mix64(key) >> 40) & (kNB - 1);
-                int64_t sb = (int64_t)b * 8 + sub;
+                int64_t sb = pp.wg_streams
+                    ? (int64_t)b * gridDim.x + blockIdx.x
+                    : (int64_t)b * 8 + sub;
                 if (has_val && val_null) {
                     unsigned off = atomicAdd(&nhist[b], 1u);
                     nrecs[sb * pp.nbucket_stride + ngbase[b] + off] = key;
@@ -1895,6 +1917,11 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
         }
         __syncthreads();
     }
+    if (pp.wg_streams) {
+        /* publish this WG's final per-bucket counts */
+        for (int i = tid; i < kNB; i += 256)
+            cursors[(int64_t)i * gridDim.x + blockIdx.x] = ngbase[i];
+    }
 }
 
 
@@ -1909,7 +1936,7 @@ k_bucket_agg(const ulonglong2* recs, const unsigned long long* cursors,
              OutGroup* out, unsigned long long* out_counter, int64_t out_cap,
              TableHdr* th, int sum_slot, int agg_count,
              int packed_mode, int bits_k, uint64_t gmin_k, uint64_t gmin_v,
-             int aligned)
+             int aligned, int nsub)
 {
     __shared__ unsigned long long tab[kHSlots * 3];
     const int tid = threadIdx.x;
@@ -1924,10 +1951,10 @@ k_bucket_agg(const ulonglong2* recs, const unsigned long long* cursors,
 
     const uint64_t kmask = (bits_k >= 64) ? ~0ULL : ((1ULL << bits_k) - 1);
     bool full = false;
-    for (int sub = 0; sub < 8 && !full; sub++) {
-    int64_t n = (int64_t)cursors[bucket * 8 + sub];
-    const ulonglong2* rows = recs + ((int64_t)bucket * 8 + sub) * bucket_stride;
-    const uint64_t* rows8 = (const uint64_t*)recs + ((int64_t)bucket * 8 + sub) * bucket_stride;
+    for (int sub = 0; sub < nsub && !full; sub++) {
+    int64_t n = (int64_t)cursors[(int64_t)bucket * nsub + sub];
+    const ulonglong2* rows = recs + ((int64_t)bucket * nsub + sub) * bucket_stride;
+    const uint64_t* rows8 = (const uint64_t*)recs + ((int64_t)bucket * nsub + sub) * bucket_stride;
     /* 4 records per thread per pass: independent probes overlap LDS latency */
     int64_t i = tid;
     /* pads (aligned claims): packed = bit 63, 16B = key == kEmptyKey —
@@ -2074,7 +2101,7 @@ k_bucket_agg_direct(const ulonglong2* recs, const unsigned long long* cursors,
                     OutGroup* out, unsigned long long* out_counter, int64_t out_cap,
                     TableHdr* th, int sum_slot, int agg_count,
                     int packed_mode, int bits_k, uint64_t gmin_k, uint64_t gmin_v,
-                    int dshift)
+                    int dshift, int nsub)
 {
     extern __shared__ __attribute__((aligned(16))) char smem[];
     const int SL = 1 << dshift;
@@ -2090,10 +2117,10 @@ k_bucket_agg_direct(const ulonglong2* recs, const unsigned long long* cursors,
     const uint64_t base_rel = (uint64_t)bucket << dshift;
     const bool has_sum = sum_slot >= 0;
 
-    for (int sub = 0; sub < 8; sub++) {
-        int64_t n = (int64_t)cursors[bucket * 8 + sub];
-        const ulonglong2* rows = recs + ((int64_t)bucket * 8 + sub) * bucket_stride;
-        const uint64_t* rows8 = (const uint64_t*)recs + ((int64_t)bucket * 8 + sub) * bucket_stride;
+    for (int sub = 0; sub < nsub; sub++) {
+        int64_t n = (int64_t)cursors[(int64_t)bucket * nsub + sub];
+        const ulonglong2* rows = recs + ((int64_t)bucket * nsub + sub) * bucket_stride;
+        const uint64_t* rows8 = (const uint64_t*)recs + ((int64_t)bucket * nsub + sub) * bucket_stride;
         int64_t i = tid;
         #define DACCR(r_)                                                    \
             {                                                                \
@@ -2683,13 +2710,13 @@ hipError_t ytql_launch_bucket_agg(const void* recs, const unsigned long long* cu
                                   TableHdr* th, int sum_slot, int agg_count,
                                   int packed_mode, int bits_k,
                                   uint64_t gmin_k, uint64_t gmin_v,
-                                  int aligned, hipStream_t st)
+                                  int aligned, int nsub, hipStream_t st)
 {
     hipLaunchKernelGGL(k_bucket_agg, dim3(kNB), dim3(256), 0, st,
                        (const ulonglong2*)recs, cursors, bucket_stride,
                        nrecs, ncursors, nbucket_stride,
                        out, out_counter, out_cap, th, sum_slot, agg_count,
-                       packed_mode, bits_k, gmin_k, gmin_v, aligned);
+                       packed_mode, bits_k, gmin_k, gmin_v, aligned, nsub);
     return hipGetLastError();
 }
 
@@ -2702,14 +2729,14 @@ hipError_t ytql_launch_bucket_agg_direct(const void* recs, const unsigned long l
                                          TableHdr* th, int sum_slot, int agg_count,
                                          int packed_mode, int bits_k,
                                          uint64_t gmin_k, uint64_t gmin_v,
-                                         int dshift, hipStream_t st)
+                                         int dshift, int nsub, hipStream_t st)
 {
     size_t lds = (size_t)2 * sizeof(unsigned long long) << dshift;
     hipLaunchKernelGGL(k_bucket_agg_direct, dim3(kNB), dim3(256), lds, st,
                        (const ulonglong2*)recs, cursors, bucket_stride,
                        nrecs, ncursors, nbucket_stride,
                        out, out_counter, out_cap, th, sum_slot, agg_count,
-                       packed_mode, bits_k, gmin_k, gmin_v, dshift);
+                       packed_mode, bits_k, gmin_k, gmin_v, dshift, nsub);
     return hipGetLastError();
 }
 
