@@ -1698,7 +1698,7 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
         const int R = (pp.tile_rows + 255) / 256;
         /* pass 1: COUNT per bucket (no returns, no per-row state) */
         #pragma unroll 8
-        for (int i = 0; i < R; i++) {
+        for (int i = 0; i < (pp.store_mode == 6 ? 0 : R); i++) {
             int64_t j = t0 + (int64_t)i * 256 + tid;
             if (j >= t1) continue;
             if (has_filter) {
@@ -1715,6 +1715,7 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
             unsigned b = pp.direct_mode
                 ? (unsigned)((kzzfull - pp.gmin_k) >> pp.dshift)
                 : (unsigned)(mix64((uint64_t)zz_dec(kzzfull)) >> 40) & (kNB - 1);
+            if (pp.store_mode == 7) { asm volatile("" :: "v"(b)); continue; }
             if (has_val && vbm && bm_get(vbm, j)) atomicAdd(&nhist[b], 1u);
             else atomicAdd(&hist[b], 1u);
         }
@@ -1785,7 +1786,7 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
         if (ovf[0] != 1) {   /* LDS flag: uniform across the block */
             /* pass 2: claim offset, decode, write */
             #pragma unroll 8
-            for (int i = 0; i < R && pp.store_mode != 3; i++) {
+            for (int i = 0; i < R && (pp.store_mode < 3); i++) {
                 int64_t j = t0 + (int64_t)i * 256 + tid;
                 if (j >= t1) continue;
                 if (has_filter) {
