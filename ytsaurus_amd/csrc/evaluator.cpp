@@ -970,8 +970,10 @@ static int run_partitioned(const YtPlan* plan, const YtChunk* chunk,
             const char* sm = getenv("YTQL_STORE");   /* perf experiments only */
             if (sm) pp.store_mode = atoi(sm);
         }
+        /* measured ~1ms slower than XCD sub-streams (phase B loses its
+         * paired-load ILP across 2048 short streams); keep as a knob */
         pp.wg_streams = (!pp.reorder && !pp.has_val_nulls
-                         && !getenv("YTQL_NO_WGSTREAMS")) ? 1 : 0;
+                         && getenv("YTQL_WGSTREAMS")) ? 1 : 0;
         /* direct-span mode: when the key zigzag span is small, partition by
          * key RANGE and index phase B's per-bucket array directly (the
          * headline config — 1M distinct keys — spans 21 bits). Needs the
