@@ -327,11 +327,14 @@ def main():
             step.last = st
             return st
         # bottom query: partial aggregate + hash partition on device
+        import time as _t
+        t0 = _t.monotonic()
         counts, st = y.gpu_partial(plan, dev_chunk, world,
                                    states_t.data_ptr(), cap,
                                    max_groups_hint=hint)
         scan_ms_total += st.kernel_scan_ms
         scan_launches += st.kernel_scan_launches
+        t1 = _t.monotonic()
         # exchange sizes then states (RCCL all-to-all over xGMI)
         sizes = torch.tensor(counts, dtype=torch.int64, device="cuda")
         rsizes = torch.zeros(world, dtype=torch.int64, device="cuda")
@@ -344,10 +347,14 @@ def main():
         dist.all_to_all_single(
             recv_t[:total_recv].view(-1, 4), states_t[:sum(counts)].view(-1, 4),
             output_split_sizes=recv_split, input_split_sizes=send_split)
+        t2 = _t.monotonic()
         # front query: merge + finalize
         _, mst = y.gpu_merge(plan, recv_t.data_ptr(), total_recv,
                              max_groups_hint=hint, rowset=out_rs,
                              raw_rowset=True)
+        if os.environ.get("YTQL_TIMING"):
+            log("2ph step: partial %.1fms a2a %.1fms merge %.1fms"
+                % ((t1 - t0) * 1e3, (t2 - t1) * 1e3, (_t.monotonic() - t2) * 1e3))
         return st
 
     # warmup

@@ -3180,6 +3180,7 @@ extern "C" int yt_gpu_query_partial(
     if (!options) options = &defopt;
     if (stats) memset(stats, 0, sizeof(*stats));
 
+    double tw0 = now_ms();
     DevPlan dp;
     rc = build_devplan(plan, chunk, &dp, errbuf, errlen);
     if (rc) return rc;
@@ -3201,9 +3202,13 @@ extern "C" int yt_gpu_query_partial(
     rc = setup_table(&R, plan->agg_count, options->max_groups_hint,
                      options->group_row_limit, errbuf, errlen);
     if (rc) return rc;
+    double tq0 = now_ms();
     static const JoinDev kNoJoin2 = {};
     rc = run_scan(plan, chunk, options, &R, &dp, &kNoJoin2, &fs, maxw, stats, errbuf, errlen);
     if (rc) return rc;
+    if (getenv("YTQL_TIMING"))
+        fprintf(stderr, "[ytql timing] partial: setup %.2fms scan %.2fms\n",
+                tq0 - tw0, now_ms() - tq0);
 
     TableHdr th;
     {
@@ -3267,6 +3272,8 @@ extern "C" int yt_gpu_query_partial(
         HIP_CHECK(hipStreamSynchronize(R.stream));
         g_pool.put(d_counts);
         for (int p = 0; p < partition_count; p++) part_counts[p] = (int64_t)counts[p];
+        if (getenv("YTQL_TIMING"))
+            fprintf(stderr, "[ytql timing] partial total %.2fms\n", now_ms() - tw0);
         if (stats) {
             stats->rows_read = chunk->row_count;
             stats->grouped_row_count = total;
@@ -3297,17 +3304,22 @@ extern "C" int yt_gpu_merge_states(
     if (!options) options = &defopt;
     if (stats) memset(stats, 0, sizeof(*stats));
 
+    double tw0 = now_ms();
     DeviceRun R;
     R.stream = (hipStream_t)(uintptr_t)options->stream;
     rc = setup_table(&R, plan->agg_count, options->max_groups_hint, 0, errbuf, errlen);
     if (rc) return rc;
     rc = ensure_slots(&R, plan->agg_count, errbuf, errlen);
     if (rc) return rc;
+    double tq1 = now_ms();
 
     HIP_CHECK(ytql_launch_merge_states((const YtStateRow*)states_device, state_row_count,
                                        plan->agg_count, sum_slot, R.d_th, R.d_slots,
                                        R.stream));
     HIP_CHECK(hipStreamSynchronize(R.stream));
+    if (getenv("YTQL_TIMING"))
+        fprintf(stderr, "[ytql timing] merge: setup %.2fms kernel %.2fms\n",
+                tq1 - tw0, now_ms() - tq1);
     {
         TableHdr th;
         HIP_CHECK(hipMemcpy(&th, R.d_th, sizeof(th), hipMemcpyDeviceToHost));
