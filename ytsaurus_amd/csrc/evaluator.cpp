@@ -3211,6 +3211,7 @@ extern "C" int yt_gpu_query_partial(
                 tq0 - tw0, now_ms() - tq0);
 
     TableHdr th;
+    double tq15 = now_ms();
     {
         HIP_CHECK(hipMemcpy(&th, R.d_th, sizeof(th), hipMemcpyDeviceToHost));
         if (th.overflow == 1) { set_err(errbuf, errlen, "group table overflow"); rc = YT_ERR_CAPACITY; goto fail; }
@@ -3251,6 +3252,7 @@ extern "C" int yt_gpu_query_partial(
         }
 
         /* partition counts → host prefix → scatter */
+        double tq2 = now_ms();
         unsigned long long* d_counts = nullptr;
         HIP_CHECK(pool_alloc(&d_counts, sizeof(unsigned long long) * partition_count));
         HIP_CHECK(hipMemsetAsync(d_counts, 0, sizeof(unsigned long long) * partition_count, R.stream));
@@ -3273,7 +3275,8 @@ extern "C" int yt_gpu_query_partial(
         g_pool.put(d_counts);
         for (int p = 0; p < partition_count; p++) part_counts[p] = (int64_t)counts[p];
         if (getenv("YTQL_TIMING"))
-            fprintf(stderr, "[ytql timing] partial total %.2fms\n", now_ms() - tw0);
+            fprintf(stderr, "[ytql timing] partial total %.2fms (compact+sides %.2fms count/scatter %.2fms)\n",
+                    now_ms() - tw0, tq2 - tq15, now_ms() - tq2);
         if (stats) {
             stats->rows_read = chunk->row_count;
             stats->grouped_row_count = total;
@@ -3325,15 +3328,15 @@ extern "C" int yt_gpu_merge_states(
         HIP_CHECK(hipMemcpy(&th, R.d_th, sizeof(th), hipMemcpyDeviceToHost));
         if (th.overflow == 1) { set_err(errbuf, errlen, "merge table overflow"); rc = YT_ERR_CAPACITY; goto fail; }
         int64_t ngroups = (int64_t)th.ngroups;
-        std::vector<OutGroup> groups;
+        OutGroup* hgroups = nullptr;
         if (ngroups > 0) {
             HIP_CHECK(pool_alloc(&R.d_groups, sizeof(OutGroup) * ngroups));
             HIP_CHECK(pool_alloc(&R.d_counter, sizeof(unsigned long long)));
             HIP_CHECK(hipMemsetAsync(R.d_counter, 0, sizeof(unsigned long long), R.stream));
             HIP_CHECK(ytql_launch_compact(nullptr, R.d_th, R.d_slots, plan->agg_count,
                                           R.d_groups, R.d_counter, R.nslots, R.stream));
-            groups.resize(ngroups);
-            HIP_CHECK(hipMemcpyAsync(groups.data(), R.d_groups, sizeof(OutGroup) * ngroups,
+            HIP_CHECK(pool_alloc_host(&hgroups, sizeof(OutGroup) * ngroups));
+            HIP_CHECK(hipMemcpyAsync(hgroups, R.d_groups, sizeof(OutGroup) * ngroups,
                                      hipMemcpyDeviceToHost, R.stream));
             HIP_CHECK(hipStreamSynchronize(R.stream));
         }
@@ -3352,9 +3355,10 @@ extern "C" int yt_gpu_merge_states(
         fake.column_count = kMaxCols;
         fake.columns = cols.data();
         int out_limited = 0;
-        rc = emit_rows(plan, &fake, nullptr, groups.data(), (int64_t)groups.size(), th, 0,
+        rc = emit_rows(plan, &fake, nullptr, hgroups, ngroups, th, 0,
                        nullptr, 0, options->output_row_limit, &out_limited,
                        output, errbuf, errlen);
+        g_pool.put(hgroups);
         if (rc) return rc;
         if (out_limited && stats) stats->incomplete_output = 1;
         if (plan->order_count > 0 || plan->with_totals || plan->having) {
