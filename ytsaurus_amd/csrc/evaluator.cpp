@@ -3163,6 +3163,7 @@ extern "C" int yt_gpu_query_partial(
     int rc = yt_gpu_available(errbuf, errlen);
     if (rc != YT_OK) return rc;
     if (plan->key_count != 1) { set_err(errbuf, errlen, "partial: need exactly 1 key"); return YT_ERR_UNSUPPORTED; }
+    if (partition_count < 1 || partition_count > 64) { set_err(errbuf, errlen, "partial: 1..64 partitions"); return YT_ERR_UNSUPPORTED; }
     if (plan->join) { set_err(errbuf, errlen, "partial: join at the bottom query not this round"); return YT_ERR_UNSUPPORTED; }
     int sum_slot = -1;
     for (int a = 0; a < plan->agg_count; a++) {

@@ -2587,26 +2587,57 @@ __global__ void k_compact(TableHdr* th, const unsigned long long* slots,
 /* partition counting + scatter of compacted groups into YtStateRow buckets.
  * Mirrors the reference's in-process shuffle hash partition
  * (shuffling_reader.cpp:40-42: destination = hash(key) % destinationCount). */
+/* per-block LDS reduction first: single-address global atomics serialize
+ * at ~160 M/s (tools/probe_strcompact), and nparts is 1..64 — a per-row
+ * global atomicAdd made this pair ~24 ms at 1M groups */
+constexpr int kMaxParts = 64;
 __global__ void k_part_count(const OutGroup* groups, int64_t n, int nparts,
                              int sum_slot, unsigned long long* counts)
 {
+    __shared__ unsigned lc[kMaxParts];
+    for (int i = threadIdx.x; i < nparts; i += blockDim.x) lc[i] = 0;
+    __syncthreads();
     for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
          i < n; i += (int64_t)gridDim.x * blockDim.x) {
         uint64_t h = partition_hash(groups[i].key_bits, (int)(groups[i].key_meta & 1));
-        atomicAdd(&counts[h % (uint64_t)nparts], 1ULL);
+        atomicAdd(&lc[h % (uint64_t)nparts], 1u);
     }
+    __syncthreads();
+    for (int i = threadIdx.x; i < nparts; i += blockDim.x)
+        if (lc[i]) atomicAdd(&counts[i], (unsigned long long)lc[i]);
 }
 
+/* block-chunked: count my chunk, claim one global range per partition,
+ * then place rows at block-local claims (order within a partition is
+ * arbitrary — the merge is order-free) */
 __global__ void k_part_scatter(const OutGroup* groups, int64_t n, int nparts,
                                int sum_slot, unsigned long long* cursors,
                                YtStateRow* out)
 {
-    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-         i < n; i += (int64_t)gridDim.x * blockDim.x) {
+    __shared__ unsigned lcnt[kMaxParts];
+    __shared__ unsigned lclaim[kMaxParts];
+    __shared__ unsigned long long lbase[kMaxParts];
+    const int64_t chunk = (n + gridDim.x - 1) / gridDim.x;
+    const int64_t lo = (int64_t)blockIdx.x * chunk;
+    int64_t hi = lo + chunk;
+    if (hi > n) hi = n;
+    if (lo >= hi) return;
+    for (int i = threadIdx.x; i < nparts; i += blockDim.x) { lcnt[i] = 0; lclaim[i] = 0; }
+    __syncthreads();
+    for (int64_t i = lo + threadIdx.x; i < hi; i += blockDim.x) {
+        uint64_t h = partition_hash(groups[i].key_bits, (int)(groups[i].key_meta & 1));
+        atomicAdd(&lcnt[h % (uint64_t)nparts], 1u);
+    }
+    __syncthreads();
+    for (int i = threadIdx.x; i < nparts; i += blockDim.x)
+        lbase[i] = lcnt[i] ? atomicAdd(&cursors[i], (unsigned long long)lcnt[i]) : 0;
+    __syncthreads();
+    for (int64_t i = lo + threadIdx.x; i < hi; i += blockDim.x) {
         const OutGroup& g = groups[i];
         int knull = (int)(g.key_meta & 1);
         uint64_t h = partition_hash(g.key_bits, knull);
-        unsigned long long pos = atomicAdd(&cursors[h % (uint64_t)nparts], 1ULL);
+        int p = (int)(h % (uint64_t)nparts);
+        unsigned long long pos = lbase[p] + atomicAdd(&lclaim[p], 1u);
         YtStateRow& sr = out[pos];
         sr.key_bits = g.key_bits;
         uint64_t nonnull = (sum_slot >= 0) ? (g.agg_nonnull[sum_slot] ? 1 : 0) : 0;
