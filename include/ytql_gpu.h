@@ -127,6 +127,16 @@ enum {
     YT_AGG_SUM1 = 1,      /* sum(1) == count; QL has no count aggregate (ql_query_ut.cpp:3200) */
     YT_AGG_MIN = 2,       /* udf/min.c */
     YT_AGG_MAX = 3,       /* udf/max.c */
+    YT_AGG_FIRST = 4,     /* udf/first.c + registry.cpp FirstIteration:3642-3663:
+                             the first NON-NULL value in scan order; scan order
+                             across GPU threads is nondeterministic (as it is
+                             across the reference's tablets), so with several
+                             distinct values per group the pick is arbitrary.
+                             String arguments: oracle only (GPU refuses). */
+    YT_AGG_AVG = 5,       /* builtin_function_profiler.cpp:1483-1607 avg:
+                             state {i64 count, arg-typed sum (int wraps,
+                             double fadd)}; finalize null when count==0 else
+                             double(sum)/count */
 };
 
 typedef struct YtAgg {

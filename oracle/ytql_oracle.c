@@ -479,6 +479,7 @@ static Val eval_expr(const YtExpr* e, EvalCtx* ctx)
 typedef struct {
     Val* keys;            /* ngroups * key_count */
     Val* states;          /* ngroups * agg_count (sum state: typed or null) */
+    uint64_t* acounts;    /* ngroups * agg_count non-null arg counts (avg) */
     uint64_t* rowcounts;  /* rows per group (sum(1) state) */
     int64_t ngroups;
     int64_t cap;
@@ -537,16 +538,17 @@ static int gt_init(GroupTable* t, int key_count, int agg_count, int64_t cap_hint
     while (t->nslots < t->cap * 2) t->nslots <<= 1;
     t->keys = malloc(sizeof(Val) * t->cap * (key_count ? key_count : 1));
     t->states = malloc(sizeof(Val) * t->cap * (agg_count ? agg_count : 1));
+    t->acounts = malloc(sizeof(uint64_t) * t->cap * (agg_count ? agg_count : 1));
     t->rowcounts = malloc(sizeof(uint64_t) * t->cap);
     t->slots = malloc(sizeof(int64_t) * t->nslots);
-    if (!t->keys || !t->states || !t->rowcounts || !t->slots) return -1;
+    if (!t->keys || !t->states || !t->acounts || !t->rowcounts || !t->slots) return -1;
     for (int64_t i = 0; i < t->nslots; i++) t->slots[i] = -1;
     return 0;
 }
 
 static void gt_free(GroupTable* t)
 {
-    free(t->keys); free(t->states); free(t->rowcounts); free(t->slots);
+    free(t->keys); free(t->states); free(t->acounts); free(t->rowcounts); free(t->slots);
 }
 
 static int gt_grow(GroupTable* t)
@@ -556,8 +558,9 @@ static int gt_grow(GroupTable* t)
     int ac = t->agg_count ? t->agg_count : 1;
     t->keys = realloc(t->keys, sizeof(Val) * newcap * kc);
     t->states = realloc(t->states, sizeof(Val) * newcap * ac);
+    t->acounts = realloc(t->acounts, sizeof(uint64_t) * newcap * ac);
     t->rowcounts = realloc(t->rowcounts, sizeof(uint64_t) * newcap);
-    if (!t->keys || !t->states || !t->rowcounts) return -1;
+    if (!t->keys || !t->states || !t->acounts || !t->rowcounts) return -1;
     t->cap = newcap;
     if (t->nslots < newcap * 2) {
         int64_t newslots = t->nslots;
@@ -598,6 +601,7 @@ static int64_t gt_upsert(GroupTable* t, const Val* keys)
     memcpy(&t->keys[g * t->key_count], keys, sizeof(Val) * t->key_count);
     for (int a = 0; a < t->agg_count; a++) {
         t->states[g * t->agg_count + a] = VNULL();   /* sum_init — udf/sum.c:3-10 */
+        t->acounts[g * t->agg_count + a] = 0;
     }
     t->rowcounts[g] = 0;
     return g;
@@ -697,7 +701,16 @@ static void* scan_worker(void* arg)
             if (ctx.error) { t->error = ctx.error; return NULL; }
             Val* st = &t->table.states[g * plan->agg_count + a];
             if (agg->func == YT_AGG_SUM) sum_update_val(st, nv);
-            else minmax_update_val(st, nv, agg->func == YT_AGG_MAX);
+            else if (agg->func == YT_AGG_AVG) {
+                /* avg state {count, arg-typed sum} — profiler avg codegen */
+                if (nv.type != YT_VT_NULL) {
+                    sum_update_val(st, nv);
+                    t->table.acounts[g * plan->agg_count + a]++;
+                }
+            } else if (agg->func == YT_AGG_FIRST) {
+                /* FirstIteration: keep the first non-null in scan order */
+                if (st->type == YT_VT_NULL) *st = nv;
+            } else minmax_update_val(st, nv, agg->func == YT_AGG_MAX);
         }
     }
     return NULL;
@@ -718,8 +731,24 @@ static int emit_group_row(const YtPlan* plan, GroupTable* t, int64_t g,
         if (plan->aggs[a]->func == YT_AGG_SUM1) {
             Val v; v.type = YT_VT_INT64; v.bits = t->rowcounts[g];
             rowvals[kc + a] = v;
+        } else if (plan->aggs[a]->func == YT_AGG_AVG) {
+            uint64_t c = t->acounts[g * ac + a];
+            Val st = t->states[g * ac + a];
+            Val v;
+            if (c == 0 || st.type == YT_VT_NULL) {
+                v = VNULL();
+            } else {
+                double sum;
+                if (st.type == YT_VT_DOUBLE) memcpy(&sum, &st.bits, 8);
+                else if (st.type == YT_VT_UINT64) sum = (double)st.bits;
+                else sum = (double)(int64_t)st.bits;
+                double r = sum / (double)c;
+                v.type = YT_VT_DOUBLE; v.str = 0; v.len = 0;
+                memcpy(&v.bits, &r, 8);
+            }
+            rowvals[kc + a] = v;
         } else {
-            rowvals[kc + a] = t->states[g * ac + a];  /* sum_finalize = copy */
+            rowvals[kc + a] = t->states[g * ac + a];  /* sum/first finalize = copy */
         }
     }
 
@@ -903,7 +932,9 @@ static void ord_fold_totals(const YtPlan* plan, const YtRowset* out,
             YtValue* t = &tot[a];
             int fagg = plan->aggs[a]->func;
             if (t->type == YT_VT_NULL) { t->type = v->type; t->data.bits = v->data.bits; continue; }
-            if (fagg == YT_AGG_SUM || fagg == YT_AGG_SUM1) {
+            if (fagg == YT_AGG_FIRST) {
+                /* FirstMerge: keep the first non-null (already in t) */
+            } else if (fagg == YT_AGG_SUM || fagg == YT_AGG_SUM1) {
                 if (v->type == YT_VT_DOUBLE) t->data.dbl += v->data.dbl;
                 else t->data.bits += v->data.bits;   /* mod 2^64, udf/sum.c */
             } else {
@@ -1018,6 +1049,18 @@ int yto_execute(const YtPlan* plan, const YtChunk* chunk,
                 int nthreads, char* errbuf, size_t errlen)
 {
     if (nthreads < 1) nthreads = 1;
+    for (int a = 0; a < plan->agg_count; a++) {
+        int f = plan->aggs[a]->func;
+        if (f < YT_AGG_SUM || f > YT_AGG_AVG) {
+            set_err(errbuf, errlen, "unknown aggregate function");
+            return YT_ERR_UNSUPPORTED;
+        }
+        if (f == YT_AGG_AVG && plan->with_totals) {
+            /* avg-of-avgs is wrong; the reference folds states — refuse */
+            set_err(errbuf, errlen, "avg: WITH TOTALS not this round");
+            return YT_ERR_UNSUPPORTED;
+        }
+    }
     int ncols = chunk->column_count;
     int jF = plan->join ? plan->join->foreign_value_count : 0;
     int ncols_eff = ncols + jF;
@@ -1259,7 +1302,14 @@ int yto_execute(const YtPlan* plan, const YtChunk* chunk,
                             Val st = pt->states[g * plan->agg_count + a];
                             Val* dst = &merged.states[mg * plan->agg_count + a];
                             if (plan->aggs[a]->func == YT_AGG_SUM) sum_update_val(dst, st);
-                            else if (plan->aggs[a]->func != YT_AGG_SUM1)
+                            else if (plan->aggs[a]->func == YT_AGG_AVG) {
+                                sum_update_val(dst, st);
+                                merged.acounts[mg * plan->agg_count + a] +=
+                                    pt->acounts[g * plan->agg_count + a];
+                            } else if (plan->aggs[a]->func == YT_AGG_FIRST) {
+                                /* FirstMerge: tasks iterate in row order */
+                                if (dst->type == YT_VT_NULL) *dst = st;
+                            } else if (plan->aggs[a]->func != YT_AGG_SUM1)
                                 minmax_update_val(dst, st, plan->aggs[a]->func == YT_AGG_MAX);
                         }
                     }

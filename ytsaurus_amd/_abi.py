@@ -49,6 +49,8 @@ AGG_SUM = 0
 AGG_SUM1 = 1
 AGG_MIN = 2
 AGG_MAX = 3
+AGG_FIRST = 4
+AGG_AVG = 5
 
 
 class YtValueData(C.Union):

@@ -14,11 +14,12 @@ from ._abi import (
     EX_COLUMN, EX_LIT_I64, EX_LIT_NULL, EX_LIT_DOUBLE,
     EX_ADD, EX_SUB, EX_MUL, EX_DIV, EX_MOD,
     EX_EQ, EX_NE, EX_LT, EX_LE, EX_GT, EX_GE, EX_AND, EX_OR, EX_NOT,
-    AGG_SUM, AGG_SUM1, YT_OK,
+    AGG_SUM, AGG_SUM1, AGG_MIN, AGG_MAX, AGG_FIRST, AGG_AVG, YT_OK,
 )
 
 __all__ = [
     "col", "lit", "null", "litf", "Plan", "Join", "agg_sum", "agg_sum1",
+    "agg_first", "agg_avg",
     "encode_int64", "encode_double", "encode_bool", "encode_string", "encode_string_raw",
     "oracle_decode_strings", "encode_versioned_int64", "oracle_versioned_read", "gpu_versioned_read", "gpu_versioned_scan_chunk", "ScanChunk", "VersionedColumn",
     "Chunk", "oracle_execute",
@@ -99,6 +100,18 @@ def agg_sum(e):
 
 def agg_sum1():
     return (AGG_SUM1, None)
+
+
+def agg_first(e):
+    """first(x): the first non-null value in scan order (udf/first.c,
+    registry.cpp FirstIteration) -- arbitrary across parallel scans."""
+    return (AGG_FIRST, e)
+
+
+def agg_avg(e):
+    """avg(x): double(sum)/count over non-null args
+    (builtin_function_profiler.cpp avg codegen)."""
+    return (AGG_AVG, e)
 
 
 class Join:

@@ -400,6 +400,31 @@ static int build_devplan(const YtPlan* plan, const YtChunk* chunk, DevPlan* p,
     p->agg_count = plan->agg_count;
     for (int a = 0; a < plan->agg_count; a++) {
         p->agg_func[a] = plan->aggs[a]->func;
+        if (plan->aggs[a]->func < YT_AGG_SUM ||
+            plan->aggs[a]->func > YT_AGG_AVG) {
+            set_err(errbuf, errlen, "unknown aggregate function");
+            return YT_ERR_UNSUPPORTED;
+        }
+        if (plan->aggs[a]->func == YT_AGG_FIRST ||
+            plan->aggs[a]->func == YT_AGG_AVG) {
+            /* no packed state format for these: strings can't live in one
+             * slot word (first), and avg-of-avgs is wrong (totals fold) */
+            uint8_t at = expr_static_type(plan->aggs[a]->arg, p->col_types);
+            if (plan->aggs[a]->func == YT_AGG_AVG &&
+                at != YT_VT_INT64 && at != YT_VT_UINT64 && at != YT_VT_DOUBLE) {
+                set_err(errbuf, errlen, "avg: int64/uint64/double argument only");
+                return YT_ERR_UNSUPPORTED;
+            }
+            if (plan->aggs[a]->func == YT_AGG_FIRST && at == YT_VT_STRING) {
+                set_err(errbuf, errlen,
+                        "first(): string argument is oracle-only this round");
+                return YT_ERR_UNSUPPORTED;
+            }
+            if (plan->aggs[a]->func == YT_AGG_AVG && plan->with_totals) {
+                set_err(errbuf, errlen, "avg: WITH TOTALS not this round");
+                return YT_ERR_UNSUPPORTED;
+            }
+        }
         if (plan->aggs[a]->func != YT_AGG_SUM1) {
             p->agg_off[a] = p->prog_len;
             rc = compile_expr(plan->aggs[a]->arg, p, &p->agg_len[a], errbuf, errlen);
@@ -1346,6 +1371,20 @@ static void finalize_row(const YtPlan* plan, uint8_t key_type,
         } else if (agg_nonnull[a] == 0) {
             row[n].type = YT_VT_NULL;
             row[n].bits = 0;
+        } else if (plan->aggs[a]->func == YT_AGG_AVG) {
+            /* finalize: double(sum)/count (profiler avg Finalize; count —
+             * the nonnull word — is nonzero here) */
+            double sum;
+            if (sum_type[a] == YT_VT_DOUBLE) {
+                memcpy(&sum, &agg_bits[a], 8);
+            } else if (sum_type[a] == YT_VT_UINT64) {
+                sum = (double)agg_bits[a];
+            } else {
+                sum = (double)(int64_t)agg_bits[a];
+            }
+            double r = sum / (double)agg_nonnull[a];
+            row[n].type = YT_VT_DOUBLE;
+            memcpy(&row[n].bits, &r, 8);
         } else {
             row[n].type = sum_type[a];
             uint64_t bits = agg_bits[a];
@@ -1796,6 +1835,8 @@ static void fold_totals_rows(const YtPlan* plan, const YtRowset* out,
             if (f == YT_AGG_SUM || f == YT_AGG_SUM1) {
                 if (v.type == YT_VT_DOUBLE) t.data.dbl += v.data.dbl;
                 else t.data.bits += v.data.bits;   /* mod 2^64, udf/sum.c */
+            } else if (f == YT_AGG_FIRST) {
+                /* keep the first non-null (already in t) */
             } else {
                 bool take;
                 if (v.type == YT_VT_DOUBLE)

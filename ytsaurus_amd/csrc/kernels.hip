@@ -713,7 +713,10 @@ __device__ __forceinline__ void agg_update_slot(unsigned long long* aggp,
     /* aggp -> { bits, nonnull } for agg a; semantics udf/sum.c, min.c, max.c */
     if (v.null_) return;
     int f = p.agg_func[a];
-    if (f == YT_AGG_SUM) {
+    if (f == YT_AGG_SUM || f == YT_AGG_AVG) {
+        /* avg state = {sum (arg-typed; int wraps, double fadd), count} —
+         * builtin_function_profiler.cpp avg codegen; count rides the
+         * nonnull word */
         if (v.type == YT_VT_DOUBLE) {
             atomicAdd((double*)aggp, __longlong_as_double(v.bits));
         } else {
@@ -723,6 +726,11 @@ __device__ __forceinline__ void agg_update_slot(unsigned long long* aggp,
         atomicMax(aggp, (unsigned long long)ord_map(v.bits, v.type));
     } else if (f == YT_AGG_MIN) {
         atomicMax(aggp, (unsigned long long)~ord_map(v.bits, v.type));
+    } else if (f == YT_AGG_FIRST) {
+        /* FirstIteration (registry.cpp:3642-3663): keep the first non-null;
+         * the CAS winner owns the bits word (read only after kernel end) */
+        if (atomicCAS(aggp + 1, 0ULL, 1ULL) == 0ULL) aggp[0] = v.bits;
+        return;
     }
     atomicAdd(aggp + 1, 1ULL);
 }
@@ -1697,8 +1705,9 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
 
         const int R = (pp.tile_rows + 255) / 256;
         /* pass 1: COUNT per bucket (no returns, no per-row state) */
+        if (pp.store_mode != 6)
         #pragma unroll 8
-        for (int i = 0; i < (pp.store_mode == 6 ? 0 : R); i++) {
+        for (int i = 0; i < R; i++) {
             int64_t j = t0 + (int64_t)i * 256 + tid;
             if (j >= t1) continue;
             if (has_filter) {
@@ -1785,8 +1794,9 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
         uint64_t* recs8 = (uint64_t*)recs;
         if (ovf[0] != 1) {   /* LDS flag: uniform across the block */
             /* pass 2: claim offset, decode, write */
+            if (pp.store_mode < 3)
             #pragma unroll 8
-            for (int i = 0; i < R && (pp.store_mode < 3); i++) {
+            for (int i = 0; i < R; i++) {
                 int64_t j = t0 + (int64_t)i * 256 + tid;
                 if (j >= t1) continue;
                 if (has_filter) {
