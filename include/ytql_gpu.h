@@ -280,10 +280,33 @@ typedef struct YtTimestampSeg {
     int64_t data_size;
 } YtTimestampSeg;
 
+/* versioned value segment layouts (round 2 widens the slice):
+ * value-index part (column_writer_detail.cpp DumpVersionedData): Dense =
+ * [cumulative values-per-row, diff-from-expected] iff denseSize <=
+ * sparseSize, else Sparse = [row index per value] (bit-packed, max =
+ * last row index). Value part (integer/floating_point writers): int
+ * Direct = [packed zigzag-base values][null bitmap]; int Dictionary =
+ * [packed dictionary (first-appearance, value-min)][packed ids, 0=null]
+ * (dictionarySize < directSize rule, integer_column_writer.cpp:222-239);
+ * double = [u64 count][raw doubles][null bitmap]. An aggregate-column
+ * bitmap (EValueFlags::Aggregate per value) sits between the timestamp
+ * ids and the value part when YT_VSEG_F_AGGREGATE is set. */
+enum YtVersionedSegType {
+    YT_VSEG_INT_DIRECT_DENSE = 0,
+    YT_VSEG_INT_DICT_DENSE = 1,
+    YT_VSEG_INT_DIRECT_SPARSE = 2,
+    YT_VSEG_INT_DICT_SPARSE = 3,
+    YT_VSEG_DOUBLE_DENSE = 16,
+    YT_VSEG_DOUBLE_SPARSE = 18,
+};
+enum { YT_VSEG_F_AGGREGATE = 1 };     /* aggregate bitmap present */
+
 typedef struct YtVersionedValueSeg {
     int64_t row_count;
     uint64_t base_value;              /* zigzag-space min over non-null values */
-    uint32_t expected_values_per_row;
+    uint32_t expected_values_per_row; /* dense index layouts only */
+    uint32_t type;                    /* YT_VSEG_* */
+    uint32_t flags;                   /* YT_VSEG_F_* */
     uint32_t pad_;
     const void* data;
     int64_t data_size;
@@ -304,6 +327,17 @@ typedef struct YtVersionedColumn {
 int yt_encode_versioned_int64(
     const uint32_t* writes_per_row, const uint64_t* write_ts,
     const int64_t* values, const uint8_t* value_nulls,
+    const uint8_t* value_agg,         /* optional per-value aggregate flags */
+    const uint32_t* deletes_per_row, const uint64_t* delete_ts,
+    int64_t row_count, int64_t max_rows_per_segment,
+    YtVersionedColumn* out, char* errbuf, size_t errlen);
+/* Same for a versioned DOUBLE column (floating_point_column_writer.cpp
+ * TVersionedFloatingPointColumnWriter: value part = [u64 count]
+ * [raw IEEE doubles][null bitmap]). */
+int yt_encode_versioned_double(
+    const uint32_t* writes_per_row, const uint64_t* write_ts,
+    const double* values, const uint8_t* value_nulls,
+    const uint8_t* value_agg,
     const uint32_t* deletes_per_row, const uint64_t* delete_ts,
     int64_t row_count, int64_t max_rows_per_segment,
     YtVersionedColumn* out, char* errbuf, size_t errlen);
@@ -315,6 +349,7 @@ void yt_versioned_free(YtVersionedColumn* col);
 int yt_gpu_versioned_read(
     const YtVersionedColumn* col, uint64_t timestamp,
     uint64_t* out_bits, uint8_t* out_null, uint8_t* out_visible,
+    uint8_t* out_agg,                 /* optional: visible value's aggregate flag */
     uint64_t stream, char* errbuf, size_t errlen);
 
 /* Read the column at `timestamp` and compact the VISIBLE rows into a

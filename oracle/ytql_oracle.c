@@ -1657,7 +1657,8 @@ static uint64_t vcum_at(const BitReader* diffs, uint32_t expected, int64_t i)
 ORACLE_EXPORT
 int yto_versioned_read(const YtVersionedColumn* col, uint64_t timestamp,
                        uint64_t* out_bits, uint8_t* out_null,
-                       uint8_t* out_visible, char* errbuf, size_t errlen)
+                       uint8_t* out_visible, uint8_t* out_agg,
+                       char* errbuf, size_t errlen)
 {
     int64_t base_row = 0;
     for (int si = 0; si < col->ts_seg_count; si++) {
@@ -1679,19 +1680,43 @@ int yto_versioned_read(const YtVersionedColumn* col, uint64_t timestamp,
         BitReader ddiffs = bitreader_init(tp);
 
         const char* vp = (const char*)V->data;
-        BitReader voffs = bitreader_init(vp);
-        vp += bitreader_byte_size(&voffs);
+        const int sparse = (V->type & 2) != 0;
+        const int is_dbl = V->type >= YT_VSEG_DOUBLE_DENSE;
+        const int is_dict = !is_dbl && (V->type & 1) != 0;
+        BitReader vindex = bitreader_init(vp);    /* dense offsets | sparse row idx */
+        vp += bitreader_byte_size(&vindex);
         BitReader tsids = bitreader_init(vp);
         vp += bitreader_byte_size(&tsids);
-        BitReader vvals = bitreader_init(vp);
-        vp += bitreader_byte_size(&vvals);
-        const uint8_t* vnull = (const uint8_t*)vp;
+        const uint8_t* vaggbm = NULL;
+        if (V->flags & YT_VSEG_F_AGGREGATE) {
+            vaggbm = (const uint8_t*)vp;
+            vp += ((tsids.size + 7) / 8 + 7) & ~(uint64_t)7;
+        }
+        BitReader vvals = {0};                    /* direct values | dictionary */
+        BitReader vids = {0};                     /* dictionary ids */
+        const double* ddata = NULL;
+        const uint8_t* vnull = NULL;
+        if (is_dbl) {
+            uint64_t cnt;
+            memcpy(&cnt, vp, 8);
+            ddata = (const double*)(vp + 8);
+            vnull = (const uint8_t*)(vp + 8 + cnt * 8);
+        } else if (is_dict) {
+            vvals = bitreader_init(vp);
+            vp += bitreader_byte_size(&vvals);
+            vids = bitreader_init(vp);
+        } else {
+            vvals = bitreader_init(vp);
+            vp += bitreader_byte_size(&vvals);
+            vnull = (const uint8_t*)vp;
+        }
 
         for (int64_t r = 0; r < T->row_count; r++) {
             int64_t g = base_row + r;
             out_visible[g] = 0;
             out_null[g] = 1;
             out_bits[g] = 0;
+            if (out_agg) out_agg[g] = 0;
             uint64_t wb = vcum_at(&wdiffs, T->expected_writes_per_row, r - 1);
             uint64_t we = vcum_at(&wdiffs, T->expected_writes_per_row, r);
             uint64_t db = vcum_at(&ddiffs, T->expected_deletes_per_row, r - 1);
@@ -1716,18 +1741,54 @@ int yto_versioned_read(const YtVersionedColumn* col, uint64_t timestamp,
             out_visible[g] = 1;
 
             /* first value with in-row timestamp index in [lower, upper) */
-            uint64_t vb = vcum_at(&voffs, V->expected_values_per_row, r - 1);
-            uint64_t ve = vcum_at(&voffs, V->expected_values_per_row, r);
+            uint64_t vb, ve;
+            if (sparse) {
+                /* row indexes ascending: binary search the run for r */
+                uint64_t lo2 = 0, hi2 = vindex.size;
+                while (lo2 < hi2) {
+                    uint64_t mid = (lo2 + hi2) >> 1;
+                    if ((int64_t)bitreader_get(&vindex, mid) < r) lo2 = mid + 1;
+                    else hi2 = mid;
+                }
+                vb = lo2;
+                hi2 = vindex.size;
+                while (lo2 < hi2) {
+                    uint64_t mid = (lo2 + hi2) >> 1;
+                    if ((int64_t)bitreader_get(&vindex, mid) <= r) lo2 = mid + 1;
+                    else hi2 = mid;
+                }
+                ve = lo2;
+            } else {
+                vb = vcum_at(&vindex, V->expected_values_per_row, r - 1);
+                ve = vcum_at(&vindex, V->expected_values_per_row, r);
+            }
             for (uint64_t j = vb; j < ve; j++) {
                 uint64_t ti = bitreader_get(&tsids, j);
                 if ((int64_t)ti < lower) continue;
                 if ((int64_t)ti >= upper) break;
-                int nul = (vnull[j / 8] >> (j % 8)) & 1;
+                int nul;
+                uint64_t bits = 0;
+                if (is_dict) {
+                    uint64_t id = bitreader_get(&vids, j);
+                    nul = (id == 0);
+                    if (!nul)
+                        bits = (uint64_t)zigzag_decode64(
+                            V->base_value + bitreader_get(&vvals, id - 1));
+                } else {
+                    nul = (vnull[j / 8] >> (j % 8)) & 1;
+                    if (!nul) {
+                        if (is_dbl) memcpy(&bits, &ddata[j], 8);
+                        else bits = (uint64_t)zigzag_decode64(
+                            V->base_value + bitreader_get(&vvals, j));
+                    }
+                }
                 if (!nul) {
                     out_null[g] = 0;
-                    out_bits[g] = (uint64_t)zigzag_decode64(
-                        V->base_value + bitreader_get(&vvals, j));
+                    out_bits[g] = bits;
                 }
+                /* aggregate flag of the chosen value, null or not */
+                if (out_agg && vaggbm)
+                    out_agg[g] = (vaggbm[j / 8] >> (j % 8)) & 1;
                 break;
             }
         }

@@ -1,0 +1,180 @@
+"""Round-2 versioned scan-format coverage (VERDICT r1 missing item 1):
+the sparse value-index layout, DictionaryDense/Sparse versioned int64
+values, versioned DOUBLE columns, and the aggregate bitmap
+(column_writer_detail.cpp DumpVersionedData:197-258,
+integer_column_writer.cpp:205-245, floating_point_column_writer.cpp
+versioned DumpSegment). Parity is pinned against an independent
+pure-Python model of the rowset_builder visibility rule.
+"""
+import numpy as np
+import pytest
+
+import ytsaurus_amd as y
+from ytsaurus_amd import _abi
+
+
+def model(rows, T):
+    """rows: (write_ts_desc, delete_ts_desc, values, nulls, aggs) per row ->
+    (value|None, visible, aggflag) per row."""
+    vals, vis, agg = [], [], []
+    for w, d, v, nl, ag in rows:
+        delete_ts = next((ts for ts in d if ts <= T), 0)
+        visible = [(i, ts) for i, ts in enumerate(w) if delete_ts < ts <= T]
+        if not visible:
+            vals.append(None)
+            vis.append(False)
+            agg.append(False)
+            continue
+        i, _ = visible[0]          # DESC order: first qualifying is newest
+        vals.append(None if nl[i] else v[i])
+        vis.append(True)
+        agg.append(bool(ag[i]))
+    return vals, vis, agg
+
+
+def flatten(rows):
+    wpr = np.array([len(r[0]) for r in rows], dtype=np.uint32)
+    dpr = np.array([len(r[1]) for r in rows], dtype=np.uint32)
+    wts = np.array([t for r in rows for t in r[0]], dtype=np.uint64)
+    dts = np.array([t for r in rows for t in r[1]], dtype=np.uint64)
+    vals = [x for r in rows for x in r[2]]
+    nuls = np.array([x for r in rows for x in r[3]], dtype=np.uint8)
+    aggs = np.array([x for r in rows for x in r[4]], dtype=np.uint8)
+    return wpr, wts, vals, nuls, dpr, dts, aggs
+
+
+def gen_rows(rng, n, max_writes=3, int_vals=True, val_pool=None,
+             sparse_shape=False):
+    rows = []
+    for r in range(n):
+        if sparse_shape:
+            # most rows empty: the reference writer picks the sparse index
+            wc = int(rng.integers(0, 8)) if rng.random() < 0.03 else 0
+        else:
+            wc = int(rng.integers(0, max_writes + 1))
+        dc = int(rng.integers(0, 2))
+        w = (np.sort(rng.choice(1000, size=wc, replace=False))[::-1] + 1).tolist()
+        d = (np.sort(rng.choice(1000, size=dc, replace=False))[::-1] + 1).tolist()
+        if val_pool is not None:
+            v = [val_pool[int(i)] for i in rng.integers(0, len(val_pool), wc)]
+        elif int_vals:
+            v = rng.integers(-10**9, 10**9, wc).tolist()
+        else:
+            v = (rng.random(wc) * 100 - 50).tolist()
+        nl = (rng.random(wc) < 0.15).astype(np.uint8).tolist()
+        ag = (rng.random(wc) < 0.5).astype(np.uint8).tolist()
+        rows.append((w, d, v, nl, ag))
+    return rows
+
+
+TS = [0, 1, 250, 500, 999, 1000, 10**15]
+
+
+def check_oracle(col, rows, with_agg):
+    for T in TS:
+        want_v, want_vis, want_agg = model(rows, T)
+        if with_agg:
+            got_v, got_vis, got_agg = y.oracle_versioned_read(col, T, with_agg=True)
+            assert got_agg == want_agg, f"T={T}"
+        else:
+            got_v, got_vis = y.oracle_versioned_read(col, T)
+        assert got_vis == want_vis, f"T={T}"
+        for a, b in zip(got_v, want_v):
+            if isinstance(b, float):
+                assert a == b or (a is None) == (b is None), f"T={T}"
+            else:
+                assert a == b, f"T={T}"
+
+
+def test_versioned_sparse_index():
+    rng = np.random.default_rng(101)
+    rows = gen_rows(rng, 20_000, sparse_shape=True)
+    wpr, wts, vals, nuls, dpr, dts, aggs = flatten(rows)
+    col = y.encode_versioned_int64(wpr, wts, np.array(vals, dtype=np.int64),
+                                   nuls, dpr, dts)
+    types = {col._c.val_segs[i].type for i in range(col._c.val_seg_count)}
+    assert any(t & 2 for t in types), f"expected a sparse segment, got {types}"
+    check_oracle(col, rows, with_agg=False)
+
+
+def test_versioned_dictionary_values():
+    rng = np.random.default_rng(102)
+    rows = gen_rows(rng, 8_000, max_writes=4, val_pool=[5, 7, -3])
+    wpr, wts, vals, nuls, dpr, dts, aggs = flatten(rows)
+    col = y.encode_versioned_int64(wpr, wts, np.array(vals, dtype=np.int64),
+                                   nuls, dpr, dts)
+    types = {col._c.val_segs[i].type for i in range(col._c.val_seg_count)}
+    assert any(t in (1, 3) for t in types), f"expected dictionary, got {types}"
+    check_oracle(col, rows, with_agg=False)
+
+
+def test_versioned_double_column():
+    rng = np.random.default_rng(103)
+    rows = gen_rows(rng, 5_000, int_vals=False)
+    wpr, wts, vals, nuls, dpr, dts, aggs = flatten(rows)
+    col = y.encode_versioned_double(wpr, wts, np.array(vals, dtype=np.float64),
+                                    nuls, dpr, dts)
+    types = {col._c.val_segs[i].type for i in range(col._c.val_seg_count)}
+    assert all(t >= 16 for t in types)
+    check_oracle(col, rows, with_agg=False)
+
+
+def test_versioned_aggregate_bitmap():
+    rng = np.random.default_rng(104)
+    rows = gen_rows(rng, 6_000)
+    wpr, wts, vals, nuls, dpr, dts, aggs = flatten(rows)
+    col = y.encode_versioned_int64(wpr, wts, np.array(vals, dtype=np.int64),
+                                   nuls, dpr, dts, value_agg=aggs)
+    assert all(col._c.val_segs[i].flags & _abi.VSEG_F_AGGREGATE
+               for i in range(col._c.val_seg_count))
+    check_oracle(col, rows, with_agg=True)
+
+
+def test_versioned_dict_sparse_combo():
+    rng = np.random.default_rng(105)
+    rows = gen_rows(rng, 30_000, sparse_shape=True, val_pool=[1, 2])
+    wpr, wts, vals, nuls, dpr, dts, aggs = flatten(rows)
+    col = y.encode_versioned_int64(wpr, wts, np.array(vals, dtype=np.int64),
+                                   nuls, dpr, dts, value_agg=aggs,
+                                   max_rows_per_segment=4096)
+    check_oracle(col, rows, with_agg=True)
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("kind", ["sparse", "dict", "double", "agg", "combo"])
+def test_versioned_gpu_parity_r2(cuda, kind):
+    rng = np.random.default_rng([106, hash(kind) % 1000])
+    if kind == "sparse":
+        rows = gen_rows(rng, 20_000, sparse_shape=True)
+    elif kind == "dict":
+        rows = gen_rows(rng, 8_000, val_pool=[5, 7, -3])
+    elif kind == "double":
+        rows = gen_rows(rng, 5_000, int_vals=False)
+    else:
+        rows = gen_rows(rng, 10_000)
+    wpr, wts, vals, nuls, dpr, dts, aggs = flatten(rows)
+    va = aggs if kind in ("agg", "combo") else None
+    seg = 4096 if kind == "combo" else 0
+    if kind == "double":
+        col = y.encode_versioned_double(wpr, wts, np.array(vals, dtype=np.float64),
+                                        nuls, dpr, dts, value_agg=va,
+                                        max_rows_per_segment=seg)
+    else:
+        col = y.encode_versioned_int64(wpr, wts, np.array(vals, dtype=np.int64),
+                                       nuls, dpr, dts, value_agg=va,
+                                       max_rows_per_segment=seg)
+    for T in [250, 750, 1000]:
+        want = y.oracle_versioned_read(col, T, with_agg=va is not None)
+        got = y.gpu_versioned_read(col, T, cuda, with_agg=va is not None)
+        assert got == want, f"T={T}"
+
+
+@pytest.mark.gpu
+def test_versioned_bridge_refuses_double(cuda):
+    rng = np.random.default_rng(107)
+    rows = gen_rows(rng, 100, int_vals=False)
+    wpr, wts, vals, nuls, dpr, dts, aggs = flatten(rows)
+    col = y.encode_versioned_double(wpr, wts, np.array(vals, dtype=np.float64),
+                                    nuls, dpr, dts)
+    with pytest.raises(RuntimeError, match="int64"):
+        y.gpu_versioned_scan_chunk(col, 500, cuda)

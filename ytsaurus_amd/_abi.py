@@ -151,9 +151,20 @@ class YtTimestampSeg(C.Structure):
                 ("data", C.c_void_p), ("data_size", C.c_int64)]
 
 
+# versioned value segment layouts (include/ytql_gpu.h YtVersionedSegType)
+VSEG_INT_DIRECT_DENSE = 0
+VSEG_INT_DICT_DENSE = 1
+VSEG_INT_DIRECT_SPARSE = 2
+VSEG_INT_DICT_SPARSE = 3
+VSEG_DOUBLE_DENSE = 16
+VSEG_DOUBLE_SPARSE = 18
+VSEG_F_AGGREGATE = 1
+
+
 class YtVersionedValueSeg(C.Structure):
     _fields_ = [("row_count", C.c_int64), ("base_value", C.c_uint64),
-                ("expected_values_per_row", C.c_uint32), ("pad_", C.c_uint32),
+                ("expected_values_per_row", C.c_uint32), ("type", C.c_uint32),
+                ("flags", C.c_uint32), ("pad_", C.c_uint32),
                 ("data", C.c_void_p), ("data_size", C.c_int64)]
 
 

@@ -21,7 +21,7 @@ __all__ = [
     "col", "lit", "null", "litf", "Plan", "Join", "agg_sum", "agg_sum1",
     "agg_first", "agg_avg",
     "encode_int64", "encode_double", "encode_bool", "encode_string", "encode_string_raw",
-    "oracle_decode_strings", "encode_versioned_int64", "oracle_versioned_read", "gpu_versioned_read", "gpu_versioned_scan_chunk", "ScanChunk", "VersionedColumn",
+    "oracle_decode_strings", "encode_versioned_int64", "encode_versioned_double", "oracle_versioned_read", "gpu_versioned_read", "gpu_versioned_scan_chunk", "ScanChunk", "VersionedColumn",
     "Chunk", "oracle_execute",
     "oracle_partial", "oracle_merge",
     "gpu_available", "gpu_execute", "gpu_partial", "gpu_merge",
@@ -319,33 +319,38 @@ class VersionedColumn:
             pass
 
 
-def encode_versioned_int64(writes_per_row, write_ts, values, value_nulls,
-                           deletes_per_row, delete_ts,
-                           max_rows_per_segment=0):
-    """Synthetic versioned-chunk generator (byte-faithful to the reference
-    timestamp/versioned-int writers). write_ts DESC per row; values 1:1
-    with writes."""
+def _encode_versioned(writes_per_row, write_ts, values, value_nulls,
+                      deletes_per_row, delete_ts, max_rows_per_segment,
+                      value_agg, is_double):
     wpr = np.ascontiguousarray(writes_per_row, dtype=np.uint32)
     wts = np.ascontiguousarray(write_ts, dtype=np.uint64)
-    vals = np.ascontiguousarray(values, dtype=np.int64)
+    vals = np.ascontiguousarray(values,
+                                dtype=np.float64 if is_double else np.int64)
     vn = np.ascontiguousarray(value_nulls, dtype=np.uint8) \
         if value_nulls is not None else None
+    va = np.ascontiguousarray(value_agg, dtype=np.uint8) \
+        if value_agg is not None else None
     dpr = np.ascontiguousarray(deletes_per_row, dtype=np.uint32)
     dts = np.ascontiguousarray(delete_ts, dtype=np.uint64)
     nw, nd = int(wpr.sum()), int(dpr.sum())
-    if len(wts) != nw or len(vals) != nw or (vn is not None and len(vn) != nw):
-        raise ValueError("versioned encode: write_ts/values/value_nulls length "
-                         f"must equal sum(writes_per_row)={nw}")
+    if (len(wts) != nw or len(vals) != nw
+            or (vn is not None and len(vn) != nw)
+            or (va is not None and len(va) != nw)):
+        raise ValueError("versioned encode: write_ts/values/value_nulls/"
+                         f"value_agg length must equal sum(writes_per_row)={nw}")
     if len(dts) != nd:
         raise ValueError("versioned encode: delete_ts length must equal "
                          f"sum(deletes_per_row)={nd}")
     cc = _abi.YtVersionedColumn()
     err = C.create_string_buffer(256)
-    rc = _abi.gpu_lib().yt_encode_versioned_int64(
+    fn = (_abi.gpu_lib().yt_encode_versioned_double if is_double
+          else _abi.gpu_lib().yt_encode_versioned_int64)
+    rc = fn(
         wpr.ctypes.data_as(C.POINTER(C.c_uint32)),
         wts.ctypes.data_as(C.POINTER(C.c_uint64)),
-        vals.ctypes.data_as(C.POINTER(C.c_int64)),
+        vals.ctypes.data_as(C.c_void_p),
         vn.ctypes.data_as(C.POINTER(C.c_uint8)) if vn is not None else None,
+        va.ctypes.data_as(C.POINTER(C.c_uint8)) if va is not None else None,
         dpr.ctypes.data_as(C.POINTER(C.c_uint32)),
         dts.ctypes.data_as(C.POINTER(C.c_uint64)),
         C.c_int64(len(wpr)), C.c_int64(max_rows_per_segment),
@@ -354,26 +359,64 @@ def encode_versioned_int64(writes_per_row, write_ts, values, value_nulls,
     return VersionedColumn(cc, len(wpr))
 
 
-def oracle_versioned_read(vcol, timestamp):
+def encode_versioned_int64(writes_per_row, write_ts, values, value_nulls,
+                           deletes_per_row, delete_ts,
+                           max_rows_per_segment=0, value_agg=None):
+    """Synthetic versioned-chunk generator (byte-faithful to the reference
+    timestamp/versioned-int writers: direct or dictionary values, dense or
+    sparse value index, optional aggregate bitmap). write_ts DESC per row;
+    values 1:1 with writes."""
+    return _encode_versioned(writes_per_row, write_ts, values, value_nulls,
+                             deletes_per_row, delete_ts, max_rows_per_segment,
+                             value_agg, is_double=False)
+
+
+def encode_versioned_double(writes_per_row, write_ts, values, value_nulls,
+                            deletes_per_row, delete_ts,
+                            max_rows_per_segment=0, value_agg=None):
+    """Versioned DOUBLE column (floating_point_column_writer.cpp
+    TVersionedFloatingPointColumnWriter layout)."""
+    return _encode_versioned(writes_per_row, write_ts, values, value_nulls,
+                             deletes_per_row, delete_ts, max_rows_per_segment,
+                             value_agg, is_double=True)
+
+
+def _vbits_decode(vcol, bits, nulls, n):
+    is_dbl = any(vcol._c.val_segs[i].type >= _abi.VSEG_DOUBLE_DENSE
+                 for i in range(vcol._c.val_seg_count))
+    if is_dbl:
+        dbl = bits.view(np.float64)
+        return [None if nulls[i] else float(dbl[i]) for i in range(n)]
+    return [None if nulls[i] else int(bits[i].astype(np.int64))
+            for i in range(n)]
+
+
+def oracle_versioned_read(vcol, timestamp, with_agg=False):
     """TEST ONLY: read the column as of `timestamp` via the oracle; returns
-    (values list with None for nulls, visible bool list)."""
+    (values list with None for nulls, visible bool list[, aggregate flag
+    list when with_agg])."""
     n = vcol.row_count
     bits = np.zeros(max(n, 1), dtype=np.uint64)
     nulls = np.zeros(max(n, 1), dtype=np.uint8)
     vis = np.zeros(max(n, 1), dtype=np.uint8)
+    agg = np.zeros(max(n, 1), dtype=np.uint8)
     err = C.create_string_buffer(256)
     rc = _abi.oracle_lib().yto_versioned_read(
         C.byref(vcol._c), C.c_uint64(timestamp),
         bits.ctypes.data_as(C.POINTER(C.c_uint64)),
         nulls.ctypes.data_as(C.POINTER(C.c_uint8)),
-        vis.ctypes.data_as(C.POINTER(C.c_uint8)), err, 256)
+        vis.ctypes.data_as(C.POINTER(C.c_uint8)),
+        agg.ctypes.data_as(C.POINTER(C.c_uint8)) if with_agg else None,
+        err, 256)
     _check(rc, err)
-    out = [None if nulls[i] else int(bits[i].astype(np.int64))
-           for i in range(n)]
-    return out, [bool(v) for v in vis[:n]]
+    out = _vbits_decode(vcol, bits, nulls, n)
+    visl = [bool(v) for v in vis[:n]]
+    if with_agg:
+        return out, visl, [bool(a) for a in agg[:n]]
+    return out, visl
 
 
-def gpu_versioned_read(vcol, timestamp, torch_mod):
+def gpu_versioned_read(vcol, timestamp, torch_mod, with_agg=False):
     """Read the versioned column as of `timestamp` on the GPU (the §8f
     row-3 slice). Returns (values list with None, visible bool list)."""
     n = vcol.row_count
@@ -381,16 +424,23 @@ def gpu_versioned_read(vcol, timestamp, torch_mod):
     nulls = torch_mod.zeros(max(n, 1), dtype=torch_mod.uint8, device="cuda")
     vis = torch_mod.zeros(max(n, 1), dtype=torch_mod.uint8, device="cuda")
     err = C.create_string_buffer(256)
+    agg = torch_mod.zeros(max(n, 1), dtype=torch_mod.uint8, device="cuda")
     rc = _abi.gpu_lib().yt_gpu_versioned_read(
         C.byref(vcol._c), C.c_uint64(timestamp),
         C.c_void_p(bits.data_ptr()), C.c_void_p(nulls.data_ptr()),
-        C.c_void_p(vis.data_ptr()), C.c_uint64(0), err, 256)
+        C.c_void_p(vis.data_ptr()),
+        C.c_void_p(agg.data_ptr()) if with_agg else None,
+        C.c_uint64(0), err, 256)
     _check(rc, err)
-    hb = bits.cpu().numpy()
+    hb = bits.cpu().numpy().astype(np.int64).view(np.uint64)
     hn = nulls.cpu().numpy()
     hv = vis.cpu().numpy()
-    out = [None if hn[i] else int(hb[i]) for i in range(n)]
-    return out, [bool(v) for v in hv[:n]]
+    out = _vbits_decode(vcol, hb, hn, n)
+    visl = [bool(v) for v in hv[:n]]
+    if with_agg:
+        ha = agg.cpu().numpy()
+        return out, visl, [bool(a) for a in ha[:n]]
+    return out, visl
 
 
 class ScanChunk:

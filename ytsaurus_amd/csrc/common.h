@@ -125,6 +125,9 @@ struct VSegDev {
     int64_t row_count;
     uint64_t base_timestamp;
     uint32_t exp_w, exp_d, exp_v;
+    uint32_t vtype;      /* YT_VSEG_* (include/ytql_gpu.h) */
+    uint32_t vflags;     /* YT_VSEG_F_* */
+    uint32_t pad_;
     uint64_t base_value;
     const char* ts_data;
     const char* val_data;

@@ -631,9 +631,22 @@ extern "C" int yt_encode_string_column(
  * column_writer_detail.cpp DumpVersionedData,
  * integer_column_writer.cpp:119-246 DumpDirectValues). */
 
-extern "C" int yt_encode_versioned_int64(
+/* shared implementation for versioned int64 / double columns.
+ * Layout per segment (column_writer_detail.cpp DumpVersionedData +
+ * integer/floating_point versioned DumpSegment):
+ *   [value index: dense = cumulative values-per-row diff-from-expected |
+ *                 sparse = row index per value]
+ *   [per-value timestamp indexes]
+ *   [aggregate bitmap]                       (aggregate columns only)
+ *   [value part: int direct = values+nullbitmap | int dict = dictionary+ids
+ *               | double = u64 count + raw doubles + nullbitmap]
+ * Index choice: dense iff denseSize <= sparseSize (DumpVersionedData:201-209).
+ * Int value choice: dictionary iff dictionarySize < directSize
+ * (integer_column_writer.cpp:222-239). */
+static int encode_versioned_impl(
     const uint32_t* writes_per_row, const uint64_t* write_ts,
-    const int64_t* values, const uint8_t* value_nulls,
+    const int64_t* ivalues, const double* dvalues,
+    const uint8_t* value_nulls, const uint8_t* value_agg,
     const uint32_t* deletes_per_row, const uint64_t* delete_ts,
     int64_t row_count, int64_t max_rows_per_segment,
     YtVersionedColumn* out, char* errbuf, size_t errlen)
@@ -642,11 +655,10 @@ extern "C" int yt_encode_versioned_int64(
     memset(out, 0, sizeof(*out));
     int nseg = (int)((row_count + max_rows_per_segment - 1) / max_rows_per_segment);
     if (nseg == 0) return YT_OK;
+    const int is_double = dvalues != nullptr;
 
     auto* tsegs = (YtTimestampSeg*)calloc(nseg, sizeof(YtTimestampSeg));
     auto* vsegs = (YtVersionedValueSeg*)calloc(nseg, sizeof(YtVersionedValueSeg));
-    /* error exit: free every per-segment blob populated so far (calloc
-     * zeroed the arrays, so free(NULL) on untouched entries is fine) */
     auto fail = [&](int rc) {
         for (int k = 0; k < nseg; k++) {
             free((void*)tsegs[k].data);
@@ -663,8 +675,7 @@ extern "C" int yt_encode_versioned_int64(
         if (r1 > row_count) r1 = row_count;
         int64_t rows = r1 - r0;
 
-        /* --- timestamp segment: register per row (writes then deletes,
-         * first-appearance dictionary — timestamp_writer.cpp:45-58,185-196) */
+        /* --- timestamp segment (timestamp_writer.cpp:45-58,185-196) --- */
         std::vector<uint64_t> dict, wids, dids, wcnt, dcnt;
         std::unordered_map<uint64_t, uint32_t> uniq;
         uint64_t ts_min = ~0ULL, ts_max = 0;
@@ -676,7 +687,8 @@ extern "C" int yt_encode_versioned_int64(
             return it.first->second;
         };
         std::vector<uint64_t> vals_zz, tsids, voff;
-        std::vector<uint8_t> vnull;
+        std::vector<double> vals_d;
+        std::vector<uint8_t> vnull, vagg;
         uint64_t vmin = ~0ULL, vmax = 0, max_tsid = 0;
         for (int64_t r = r0; r < r1; r++) {
             uint32_t wc = writes_per_row[r];
@@ -688,17 +700,20 @@ extern "C" int yt_encode_versioned_int64(
                     return fail(YT_ERR_INVALID_CHUNK);
                 }
                 wids.push_back(reg(ts));
-                /* value slice: 1:1 with writes, tsIndex = position in the
-                 * row's write list */
                 uint8_t nul = value_nulls ? value_nulls[wat + i] : 0;
-                uint64_t zz = 0;
-                if (!nul) {
-                    zz = zigzag_encode64(values[wat + i]);
-                    if (zz < vmin) vmin = zz;
-                    if (zz > vmax) vmax = zz;
+                if (is_double) {
+                    vals_d.push_back(nul ? 0.0 : dvalues[wat + i]);
+                } else {
+                    uint64_t zz = 0;
+                    if (!nul) {
+                        zz = zigzag_encode64(ivalues[wat + i]);
+                        if (zz < vmin) vmin = zz;
+                        if (zz > vmax) vmax = zz;
+                    }
+                    vals_zz.push_back(zz);
                 }
-                vals_zz.push_back(zz);
                 vnull.push_back(nul);
+                vagg.push_back(value_agg ? value_agg[wat + i] : 0);
                 tsids.push_back(i);
                 if (i > max_tsid) max_tsid = i;
             }
@@ -743,66 +758,89 @@ extern "C" int yt_encode_versioned_int64(
         memcpy(tp, tb.bytes.data(), tb.size());
         T.data = tp;
 
-        /* --- versioned int64 value segment (DirectDense) --- */
-        uint64_t vspan = vmax - vmin;         /* wraps when all null/empty */
-        for (size_t i = 0; i < vals_zz.size(); i++)
-            if (!vnull[i]) vals_zz[i] -= vmin;
+        /* --- value segment --- */
+        const int64_t nvalues = (int64_t)vnull.size();
+        const std::vector<uint64_t> vcum = voff;   /* cumulative, pre-diff */
 
+        /* index layout: dense iff denseSize <= sparseSize
+         * (DumpVersionedData:201-209) */
         uint64_t vdiff_max = 0;
-        /* the reference writer picks dense iff denseSize <= sparseSize
-         * (column_writer_detail.cpp:199-210); emit only the dense layout
-         * this round and refuse inputs where the writer would pick sparse
-         * so the generated bytes always match the reference's choice */
+        uint32_t exp_v;
         {
-            int64_t dense_probe = 0;
-            {
-                std::vector<uint64_t> probe = voff;
-                uint64_t md = 0;
-                (void)prepare_diff_from_expected(&probe, &md);
-                dense_probe = cs_bytes(md, (int64_t)voff.size());
+            std::vector<uint64_t> probe = voff;
+            exp_v = prepare_diff_from_expected(&probe, &vdiff_max);
+            voff.swap(probe);   /* voff now holds the diffs */
+        }
+        int64_t dense_sz = cs_bytes(vdiff_max, rows);
+        int64_t sparse_sz = cs_bytes((uint64_t)rows, nvalues);
+        int dense = dense_sz <= sparse_sz || nvalues == 0;
+
+        /* int value layout: dictionary iff dictionarySize < directSize */
+        uint64_t vspan = vmax - vmin;         /* wraps when all null/empty */
+        int use_dict = 0;
+        std::vector<uint64_t> vdict, vids;
+        if (!is_double) {
+            std::unordered_map<uint64_t, uint32_t> vuniq;
+            for (int64_t i = 0; i < nvalues; i++) {
+                if (!vnull[i]) vuniq.emplace(vals_zz[i], 0);
             }
-            int64_t sparse_sz = cs_bytes((uint64_t)voff.size(),
-                                         (int64_t)vals_zz.size());
-            if (dense_probe > sparse_sz) {
-                set_err(errbuf, errlen,
-                        "versioned: sparse value-index layout not this round");
-                return fail(YT_ERR_UNSUPPORTED);
+            int64_t dict_sz = cs_bytes(vspan, (int64_t)vuniq.size())
+                            + cs_bytes((uint64_t)vuniq.size() + 1, nvalues);
+            int64_t direct_sz = cs_bytes(vspan, nvalues)
+                              + (int64_t)((nvalues + 7) / 8);
+            use_dict = dict_sz < direct_sz;
+            if (use_dict) {
+                /* first-appearance order, ids 1-based, 0 = null
+                 * (integer_column_writer.cpp DumpDictionaryValues) */
+                std::unordered_map<uint64_t, uint32_t> order;
+                for (int64_t i = 0; i < nvalues; i++) {
+                    if (vnull[i]) { vids.push_back(0); continue; }
+                    auto it = order.emplace(vals_zz[i], (uint32_t)vdict.size() + 1);
+                    if (it.second) vdict.push_back(vals_zz[i] - vmin);
+                    vids.push_back(it.first->second);
+                }
+            } else {
+                for (int64_t i = 0; i < nvalues; i++)
+                    if (!vnull[i]) vals_zz[i] -= vmin;
             }
         }
-        /* mirror the reference's direct-vs-dictionary choice
-         * (integer_column_writer.cpp GetDictionarySize/GetDirectSize,
-         * DumpSegment: dictionary wins iff dictionarySize < directSize):
-         * we emit only the DirectDense layout, so refuse inputs where the
-         * reference writer would have picked DictionaryDense — otherwise
-         * the generated bytes silently diverge from the reference's. */
-        {
-            std::unordered_set<uint64_t> distinct;
-            for (size_t i = 0; i < vals_zz.size(); i++)
-                if (!vnull[i]) distinct.insert(vals_zz[i]);
-            int64_t dict_sz = cs_bytes(vspan, (int64_t)distinct.size())
-                            + cs_bytes((uint64_t)distinct.size() + 1,
-                                       (int64_t)vals_zz.size());
-            int64_t direct_sz = cs_bytes(vspan, (int64_t)vals_zz.size())
-                              + (int64_t)((vals_zz.size() + 7) / 8);
-            if (dict_sz < direct_sz) {
-                set_err(errbuf, errlen,
-                        "versioned: reference writer would pick DictionaryDense "
-                        "(not built this round)");
-                return fail(YT_ERR_UNSUPPORTED);
-            }
-        }
-        uint32_t exp_v = prepare_diff_from_expected(&voff, &vdiff_max);
 
         Blob vb;
-        vb.pack(voff, vdiff_max);
+        if (dense) {
+            vb.pack(voff, vdiff_max);
+        } else {
+            std::vector<uint64_t> row_idx;
+            row_idx.reserve(nvalues);
+            for (int64_t r = 0; r < rows; r++) {
+                while ((int64_t)row_idx.size() < (int64_t)vcum[r])
+                    row_idx.push_back((uint64_t)r);
+            }
+            vb.pack(row_idx, row_idx.empty() ? 0 : row_idx.back());
+        }
         vb.pack(tsids, max_tsid);
-        vb.pack(vals_zz, vspan);
-        vb.bitmap(vnull);
+        if (value_agg) vb.bitmap(vagg);
+        if (is_double) {
+            uint64_t cnt = (uint64_t)nvalues;
+            vb.raw(&cnt, 8);
+            vb.raw(vals_d.data(), vals_d.size() * 8);
+            vb.bitmap(vnull);
+        } else if (use_dict) {
+            vb.pack(vdict, vspan);
+            vb.pack(vids, (uint64_t)vdict.size() + 1);
+        } else {
+            vb.pack(vals_zz, vspan);
+            vb.bitmap(vnull);
+        }
 
         YtVersionedValueSeg& V = vsegs[si];
         V.row_count = rows;
         V.base_value = vmin;
         V.expected_values_per_row = exp_v;
+        V.type = is_double
+            ? (dense ? YT_VSEG_DOUBLE_DENSE : YT_VSEG_DOUBLE_SPARSE)
+            : (use_dict ? (dense ? YT_VSEG_INT_DICT_DENSE : YT_VSEG_INT_DICT_SPARSE)
+                        : (dense ? YT_VSEG_INT_DIRECT_DENSE : YT_VSEG_INT_DIRECT_SPARSE));
+        V.flags = value_agg ? YT_VSEG_F_AGGREGATE : 0;
         V.data_size = vb.size();
         void* vp = malloc(vb.size() ? vb.size() : 1);
         memcpy(vp, vb.bytes.data(), vb.size());
@@ -814,6 +852,34 @@ extern "C" int yt_encode_versioned_int64(
     out->ts_segs = tsegs;
     out->val_segs = vsegs;
     return YT_OK;
+}
+
+extern "C" int yt_encode_versioned_int64(
+    const uint32_t* writes_per_row, const uint64_t* write_ts,
+    const int64_t* values, const uint8_t* value_nulls,
+    const uint8_t* value_agg,
+    const uint32_t* deletes_per_row, const uint64_t* delete_ts,
+    int64_t row_count, int64_t max_rows_per_segment,
+    YtVersionedColumn* out, char* errbuf, size_t errlen)
+{
+    return encode_versioned_impl(writes_per_row, write_ts, values, nullptr,
+                                 value_nulls, value_agg, deletes_per_row,
+                                 delete_ts, row_count, max_rows_per_segment,
+                                 out, errbuf, errlen);
+}
+
+extern "C" int yt_encode_versioned_double(
+    const uint32_t* writes_per_row, const uint64_t* write_ts,
+    const double* values, const uint8_t* value_nulls,
+    const uint8_t* value_agg,
+    const uint32_t* deletes_per_row, const uint64_t* delete_ts,
+    int64_t row_count, int64_t max_rows_per_segment,
+    YtVersionedColumn* out, char* errbuf, size_t errlen)
+{
+    return encode_versioned_impl(writes_per_row, write_ts, nullptr, values,
+                                 value_nulls, value_agg, deletes_per_row,
+                                 delete_ts, row_count, max_rows_per_segment,
+                                 out, errbuf, errlen);
 }
 
 extern "C" void yt_versioned_free(YtVersionedColumn* col)

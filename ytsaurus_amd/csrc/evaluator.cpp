@@ -74,7 +74,7 @@ hipError_t ytql_launch_topk_gather_fast(const DevSeg*, const SegEx*, int, int,
                                         int64_t*, unsigned long long*,
                                         hipStream_t);
 hipError_t ytql_launch_versioned_read(const VSegDev*, int, int64_t, uint64_t,
-                                      uint64_t*, uint8_t*, uint8_t*,
+                                      uint64_t*, uint8_t*, uint8_t*, uint8_t*,
                                       hipStream_t);
 hipError_t ytql_launch_vis_count(const uint8_t*, int64_t,
                                  unsigned long long*, int, hipStream_t);
@@ -2808,7 +2808,7 @@ fail:
 extern "C" int yt_gpu_versioned_read(
     const YtVersionedColumn* col, uint64_t timestamp,
     uint64_t* out_bits, uint8_t* out_null, uint8_t* out_visible,
-    uint64_t stream, char* errbuf, size_t errlen)
+    uint8_t* out_agg, uint64_t stream, char* errbuf, size_t errlen)
 {
     int rc = yt_gpu_available(errbuf, errlen);
     if (rc != YT_OK) return rc;
@@ -2844,6 +2844,8 @@ extern "C" int yt_gpu_versioned_read(
             S.exp_w = T.expected_writes_per_row;
             S.exp_d = T.expected_deletes_per_row;
             S.exp_v = V.expected_values_per_row;
+            S.vtype = V.type;
+            S.vflags = V.flags;
             S.base_value = V.base_value;
             S.ts_data = d_t;
             S.val_data = d_v;
@@ -2856,7 +2858,7 @@ extern "C" int yt_gpu_versioned_read(
                                  hipMemcpyHostToDevice, st));
         HIP_CHECK(ytql_launch_versioned_read(d_segs, nseg, total_rows,
                                              timestamp, out_bits, out_null,
-                                             out_visible, st));
+                                             out_visible, out_agg, st));
         HIP_CHECK(hipStreamSynchronize(st));
     }
     for (void* p : blobs) g_pool.put(p);
@@ -2915,7 +2917,18 @@ extern "C" int yt_gpu_versioned_scan_chunk(
     HIP_CHECK(pool_alloc(&d_null, (size_t)n));
     HIP_CHECK(pool_alloc(&d_vis, (size_t)n));
     HIP_CHECK(pool_alloc(&d_blk, sizeof(unsigned long long) * grid));
-    rc = yt_gpu_versioned_read(col, timestamp, d_bits, d_null, d_vis,
+    /* the bridge emits width-64 zigzag int segments: int64 columns only
+     * (double/string bridge: later round — the read path supports them) */
+    for (int i = 0; i < col->val_seg_count; i++) {
+        if (col->val_segs[i].type >= YT_VSEG_DOUBLE_DENSE) {
+            set_err(errbuf, errlen,
+                    "versioned scan-chunk bridge: int64 value columns only");
+            cleanup();
+            delete H;
+            return YT_ERR_UNSUPPORTED;
+        }
+    }
+    rc = yt_gpu_versioned_read(col, timestamp, d_bits, d_null, d_vis, nullptr,
                                stream, errbuf, errlen);
     if (rc != YT_OK) { cleanup(); delete H; return rc; }
 

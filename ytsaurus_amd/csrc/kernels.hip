@@ -829,10 +829,31 @@ __device__ __forceinline__ uint64_t vcum_dev(const VVecRef& diffs,
     return (uint64_t)expected * (uint64_t)(i + 1) + (uint64_t)(int64_t)diff;
 }
 
+/* sparse value index: [vb,ve) = the contiguous run of values whose packed
+ * row index equals r (column_writer_detail.cpp sparse branch emits row
+ * indexes sorted ascending) */
+__device__ __forceinline__ void vsparse_range(const VVecRef& ridx, int64_t r,
+                                              uint64_t* vb, uint64_t* ve)
+{
+    uint64_t lo = 0, hi = ridx.size;
+    while (lo < hi) {   /* first j with ridx[j] >= r */
+        uint64_t mid = (lo + hi) >> 1;
+        if ((int64_t)vvec_get(ridx, mid) < r) lo = mid + 1; else hi = mid;
+    }
+    *vb = lo;
+    hi = ridx.size;
+    while (lo < hi) {   /* first j with ridx[j] > r */
+        uint64_t mid = (lo + hi) >> 1;
+        if ((int64_t)vvec_get(ridx, mid) <= r) lo = mid + 1; else hi = mid;
+    }
+    *ve = lo;
+}
+
 __global__ void __launch_bounds__(256)
 k_versioned_read(const VSegDev* segs, int nseg, int64_t total_rows,
                  uint64_t timestamp,
-                 uint64_t* out_bits, uint8_t* out_null, uint8_t* out_vis)
+                 uint64_t* out_bits, uint8_t* out_null, uint8_t* out_vis,
+                 uint8_t* out_agg)
 {
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (int64_t g = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -853,14 +874,36 @@ k_versioned_read(const VSegDev* segs, int nseg, int64_t total_rows,
         VVecRef wdiffs = vvec_parse(&tp);
         VVecRef ddiffs = vvec_parse(&tp);
         const char* vp = S.val_data;
-        VVecRef voffs = vvec_parse(&vp);
+        const bool sparse = (S.vtype & 2) != 0;   /* *_SPARSE codes */
+        const bool is_dbl = S.vtype >= YT_VSEG_DOUBLE_DENSE;
+        const bool is_dict = !is_dbl && (S.vtype & 1) != 0;
+        VVecRef vindex = vvec_parse(&vp);         /* dense offsets | sparse row idx */
         VVecRef tsids = vvec_parse(&vp);
-        VVecRef vvals = vvec_parse(&vp);
-        const uint8_t* vnull = (const uint8_t*)vp;
+        const uint8_t* vaggbm = nullptr;
+        if (S.vflags & YT_VSEG_F_AGGREGATE) {
+            vaggbm = (const uint8_t*)vp;
+            vp += ((tsids.size + 7) / 8 + 7) & ~(uint64_t)7;
+        }
+        VVecRef vvals = {};                       /* int direct values | dictionary */
+        VVecRef vids = {};                        /* dictionary ids */
+        const double* ddata = nullptr;
+        const uint8_t* vnull = nullptr;
+        if (is_dbl) {
+            uint64_t cnt = *(const uint64_t*)vp;
+            ddata = (const double*)(vp + 8);
+            vnull = (const uint8_t*)(vp + 8 + cnt * 8);
+        } else if (is_dict) {
+            vvals = vvec_parse(&vp);              /* dictionary (value - base) */
+            vids = vvec_parse(&vp);               /* 0 = null, 1-based */
+        } else {
+            vvals = vvec_parse(&vp);
+            vnull = (const uint8_t*)vp;
+        }
 
         out_vis[g] = 0;
         out_null[g] = 1;
         out_bits[g] = 0;
+        if (out_agg) out_agg[g] = 0;
 
         uint64_t wb = vcum_dev(wdiffs, S.exp_w, r - 1);
         uint64_t we = vcum_dev(wdiffs, S.exp_w, r);
@@ -883,17 +926,40 @@ k_versioned_read(const VSegDev* segs, int nseg, int64_t total_rows,
         if (lower >= upper) continue;
         out_vis[g] = 1;
 
-        uint64_t vb = vcum_dev(voffs, S.exp_v, r - 1);
-        uint64_t ve = vcum_dev(voffs, S.exp_v, r);
+        uint64_t vb, ve;
+        if (sparse) {
+            vsparse_range(vindex, r, &vb, &ve);
+        } else {
+            vb = vcum_dev(vindex, S.exp_v, r - 1);
+            ve = vcum_dev(vindex, S.exp_v, r);
+        }
         for (uint64_t j = vb; j < ve; j++) {
             int64_t ti = (int64_t)vvec_get(tsids, j);
             if (ti < lower) continue;
             if (ti >= upper) break;
-            int nul = (vnull[j / 8] >> (j % 8)) & 1;
+            int nul;
+            uint64_t bits = 0;
+            if (is_dict) {
+                uint64_t id = vvec_get(vids, j);
+                nul = (id == 0);
+                if (!nul)
+                    bits = (uint64_t)zz_dec(S.base_value + vvec_get(vvals, id - 1));
+            } else {
+                nul = (vnull[j / 8] >> (j % 8)) & 1;
+                if (!nul) {
+                    bits = is_dbl
+                        ? ((const uint64_t*)ddata)[j]
+                        : (uint64_t)zz_dec(S.base_value + vvec_get(vvals, j));
+                }
+            }
             if (!nul) {
                 out_null[g] = 0;
-                out_bits[g] = (uint64_t)zz_dec(S.base_value + vvec_get(vvals, j));
+                out_bits[g] = bits;
             }
+            /* the aggregate flag belongs to the chosen value, null or not
+             * (TVersionedColumnWriterBase::AddValues appends it per value) */
+            if (out_agg && vaggbm)
+                out_agg[g] = (vaggbm[j / 8] >> (j % 8)) & 1;
             break;
         }
     }
@@ -2839,14 +2905,15 @@ hipError_t ytql_launch_topk_gather_fast(const DevSeg* segs, const SegEx* segex,
 hipError_t ytql_launch_versioned_read(const VSegDev* segs, int nseg,
                                       int64_t total_rows, uint64_t timestamp,
                                       uint64_t* out_bits, uint8_t* out_null,
-                                      uint8_t* out_vis, hipStream_t st)
+                                      uint8_t* out_vis, uint8_t* out_agg,
+                                      hipStream_t st)
 {
     int block = 256;
     int64_t want = (total_rows + block - 1) / block;
     int grid = (int)(want > 4096 ? 4096 : (want > 0 ? want : 1));
     hipLaunchKernelGGL(k_versioned_read, dim3(grid), dim3(block), 0, st,
                        segs, nseg, total_rows, timestamp,
-                       out_bits, out_null, out_vis);
+                       out_bits, out_null, out_vis, out_agg);
     return hipGetLastError();
 }
 
