@@ -364,6 +364,19 @@ int yt_gpu_versioned_scan_chunk(
     const YtVersionedColumn* col, uint64_t timestamp,
     YtChunk* out_chunk, void** out_handle,
     uint64_t stream, char* errbuf, size_t errlen);
+
+/* Versioned TABLE bridge: nvcols versioned value columns (int64 or
+ * double; all sharing the table's per-row write/delete timestamp lists)
+ * plus optional unversioned int64 KEY columns (keys in the scan format
+ * are plain unversioned segments — rowset_builder.cpp key readers), read
+ * at `timestamp` and compacted into one device-resident unversioned
+ * chunk [keys..., values...] for yt_gpu_query_execute. Free with
+ * yt_gpu_scan_chunk_free. */
+int yt_gpu_versioned_scan_table(
+    const YtVersionedColumn* const* vcols, int nvcols,
+    const YtChunk* key_chunk, uint64_t timestamp,
+    YtChunk* out_chunk, void** out_handle,
+    uint64_t stream, char* errbuf, size_t errlen);
 void yt_gpu_scan_chunk_free(YtChunk* chunk, void* handle);
 
 /* =========================== entry points =========================== */

@@ -170,11 +170,80 @@ def test_versioned_gpu_parity_r2(cuda, kind):
 
 
 @pytest.mark.gpu
-def test_versioned_bridge_refuses_double(cuda):
+def test_versioned_table_bridge_end_to_end(cuda):
+    """MVCC table -> read at T -> GROUP BY key with sum(double), entirely on
+    device: key columns (unversioned DirectDense) + int64 + double versioned
+    value columns through yt_gpu_versioned_scan_table."""
     rng = np.random.default_rng(107)
-    rows = gen_rows(rng, 100, int_vals=False)
-    wpr, wts, vals, nuls, dpr, dts, aggs = flatten(rows)
+    n = 40_000
+    rows = gen_rows(rng, n)                      # int values
+    # same write/delete structure for the double column
+    drows = [(w, d, (rng.random(len(v)) * 10).tolist(), nl, ag)
+             for (w, d, v, nl, ag) in rows]
+    keys = rng.integers(0, 97, n).astype(np.int64)
+
+    wpr, wts, ivals, inuls, dpr, dts, _ = flatten(rows)
+    _, _, dvals, dnuls, _, _, _ = flatten(drows)
+    icol = y.encode_versioned_int64(wpr, wts, np.array(ivals, dtype=np.int64),
+                                    inuls, dpr, dts)
+    dcol = y.encode_versioned_double(wpr, wts, np.array(dvals, dtype=np.float64),
+                                     dnuls, dpr, dts)
+    key_chunk = y.Chunk([y.encode_int64(keys)], n)
+
+    T = 600
+    sc = y.gpu_versioned_scan_table([icol, dcol], T,
+                                    key_chunk=key_chunk.c_device(cuda))
+    plan = y.Plan(keys=[y.col(0)],
+                  aggs=[y.agg_sum(y.col(1)), y.agg_sum(y.col(2)), y.agg_sum1()])
+    got, _ = y.gpu_execute(plan, sc.chunk, max_groups_hint=256)
+
+    # expected: visible rows via the python model, grouped in python
+    iv, vis, _ = model(rows, T)
+    dv, _, _ = model(drows, T)
+    exp = {}
+    for r in range(n):
+        if not vis[r]:
+            continue
+        k = int(keys[r])
+        e = exp.setdefault(k, [0, 0.0, 0, False, False])
+        e[2] += 1
+        if iv[r] is not None:
+            e[0] += iv[r]
+            e[3] = True
+        if dv[r] is not None:
+            e[1] += dv[r]
+            e[4] = True
+    want = [(k, e[0] if e[3] else None, e[1] if e[4] else None, e[2])
+            for k, e in exp.items()]
+    gm = {r[0]: r for r in got}
+    assert len(got) == len(want)
+    for k, si, sd, cnt in want:
+        gk, gi, gd, gc = gm[k]
+        assert gc == cnt and gi == si
+        if sd is None:
+            assert gd is None
+        else:
+            assert gd == pytest.approx(sd, rel=1e-9)
+
+
+@pytest.mark.gpu
+def test_versioned_double_chunk_bridge(cuda):
+    """single double column through the (now type-general) chunk bridge"""
+    rng = np.random.default_rng(108)
+    rows = gen_rows(rng, 3_000, int_vals=False)
+    wpr, wts, vals, nuls, dpr, dts, _ = flatten(rows)
     col = y.encode_versioned_double(wpr, wts, np.array(vals, dtype=np.float64),
                                     nuls, dpr, dts)
-    with pytest.raises(RuntimeError, match="int64"):
-        y.gpu_versioned_scan_chunk(col, 500, cuda)
+    T = 500
+    sc = y.gpu_versioned_scan_chunk(col, T)
+    plan = y.Plan(aggs=[y.agg_sum(y.col(0)), y.agg_sum1()])
+    got, _ = y.gpu_execute(plan, sc.chunk, max_groups_hint=16)
+    dv, vis, _ = model(rows, T)
+    nz = [v for v, s in zip(dv, vis) if s and v is not None]
+    cnt = sum(1 for s in vis if s)
+    (gs, gc), = [tuple(r) for r in got]
+    assert gc == cnt
+    if nz:
+        assert gs == pytest.approx(sum(nz), rel=1e-9)
+    else:
+        assert gs is None

@@ -21,7 +21,7 @@ __all__ = [
     "col", "lit", "null", "litf", "Plan", "Join", "agg_sum", "agg_sum1",
     "agg_first", "agg_avg",
     "encode_int64", "encode_double", "encode_bool", "encode_string", "encode_string_raw",
-    "oracle_decode_strings", "encode_versioned_int64", "encode_versioned_double", "oracle_versioned_read", "gpu_versioned_read", "gpu_versioned_scan_chunk", "ScanChunk", "VersionedColumn",
+    "oracle_decode_strings", "encode_versioned_int64", "encode_versioned_double", "oracle_versioned_read", "gpu_versioned_read", "gpu_versioned_scan_chunk", "gpu_versioned_scan_table", "ScanChunk", "VersionedColumn",
     "Chunk", "oracle_execute",
     "oracle_partial", "oracle_merge",
     "gpu_available", "gpu_execute", "gpu_partial", "gpu_merge",
@@ -471,6 +471,28 @@ def gpu_versioned_scan_chunk(vcol, timestamp):
         C.byref(handle), C.c_uint64(0), err, 256)
     _check(rc, err)
     return ScanChunk(ch, handle)
+
+
+def gpu_versioned_scan_table(vcols, timestamp, key_chunk=None):
+    """Versioned TABLE bridge: versioned value columns (+ optional
+    unversioned int64 key chunk, device-encoded) read at `timestamp`,
+    compacted into one engine-scannable device chunk [keys..., values...]
+    (yt_gpu_versioned_scan_table)."""
+    arr = (C.POINTER(_abi.YtVersionedColumn) * len(vcols))()
+    for i, vc in enumerate(vcols):
+        arr[i] = C.pointer(vc._c)
+    ch = _abi.YtChunk()
+    handle = C.c_void_p()
+    err = C.create_string_buffer(256)
+    rc = _abi.gpu_lib().yt_gpu_versioned_scan_table(
+        arr, C.c_int32(len(vcols)),
+        C.byref(key_chunk) if key_chunk is not None else None,
+        C.c_uint64(timestamp), C.byref(ch), C.byref(handle),
+        C.c_uint64(0), err, 256)
+    _check(rc, err)
+    sc = ScanChunk(ch, handle)
+    sc._keep = (arr, key_chunk)
+    return sc
 
 
 def oracle_decode_strings(enc, n):

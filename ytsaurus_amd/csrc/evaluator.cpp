@@ -78,10 +78,13 @@ hipError_t ytql_launch_versioned_read(const VSegDev*, int, int64_t, uint64_t,
                                       hipStream_t);
 hipError_t ytql_launch_vis_count(const uint8_t*, int64_t,
                                  unsigned long long*, int, hipStream_t);
+hipError_t ytql_launch_unvcol_to_arrays(const DevSeg*, const SegEx*, int, int,
+                                        int64_t, uint64_t*, uint8_t*,
+                                        unsigned*, hipStream_t);
 hipError_t ytql_launch_vis_scatter(const uint8_t*, const uint8_t*,
                                    const uint64_t*, int64_t,
                                    const unsigned long long*, uint64_t,
-                                   const int64_t*, char*, int, hipStream_t);
+                                   const int64_t*, char*, int, int, hipStream_t);
 hipError_t ytql_launch_join_build(const JoinDev*, int64_t, uint64_t*,
                                   long long*, unsigned long long*, unsigned*,
                                   hipStream_t);
@@ -2876,8 +2879,33 @@ struct ScanChunkHandle {
 };
 } /* namespace */
 
+extern "C" int yt_gpu_versioned_scan_table(
+    const YtVersionedColumn* const* vcols, int nvcols,
+    const YtChunk* key_chunk, uint64_t timestamp,
+    YtChunk* out_chunk, void** out_handle,
+    uint64_t stream, char* errbuf, size_t errlen);
+
 extern "C" int yt_gpu_versioned_scan_chunk(
     const YtVersionedColumn* col, uint64_t timestamp,
+    YtChunk* out_chunk, void** out_handle,
+    uint64_t stream, char* errbuf, size_t errlen)
+{
+    /* single-column case of the table bridge below */
+    const YtVersionedColumn* arr[1] = { col };
+    return yt_gpu_versioned_scan_table(arr, 1, nullptr, timestamp,
+                                       out_chunk, out_handle, stream,
+                                       errbuf, errlen);
+}
+
+/* Versioned TABLE bridge: N versioned value
+ * columns sharing one timestamp layout + optional unversioned KEY columns
+ * (keys in the scan format are plain unversioned segments —
+ * rowset_builder.cpp key readers), read at T and compacted into one
+ * device-resident unversioned chunk [keys..., values...] that
+ * yt_gpu_query_execute scans directly. Int64 and double value columns. */
+extern "C" int yt_gpu_versioned_scan_table(
+    const YtVersionedColumn* const* vcols, int nvcols,
+    const YtChunk* key_chunk, uint64_t timestamp,
     YtChunk* out_chunk, void** out_handle,
     uint64_t stream, char* errbuf, size_t errlen)
 {
@@ -2886,108 +2914,163 @@ extern "C" int yt_gpu_versioned_scan_chunk(
     hipStream_t st = (hipStream_t)(uintptr_t)stream;
     memset(out_chunk, 0, sizeof(*out_chunk));
     *out_handle = nullptr;
+    if (nvcols < 1 || !vcols) { set_err(errbuf, errlen, "scan_table: need >=1 versioned column"); return YT_ERR_INVALID_PLAN; }
 
     int64_t n = 0;
-    for (int i = 0; i < col->ts_seg_count; i++)
-        n += col->ts_segs[i].row_count;
+    for (int i = 0; i < vcols[0]->ts_seg_count; i++)
+        n += vcols[0]->ts_segs[i].row_count;
+    for (int j = 1; j < nvcols; j++) {
+        int64_t nj = 0;
+        for (int i = 0; i < vcols[j]->ts_seg_count; i++)
+            nj += vcols[j]->ts_segs[i].row_count;
+        if (nj != n) { set_err(errbuf, errlen, "scan_table: column row counts differ"); return YT_ERR_INVALID_CHUNK; }
+    }
+    const int nkey = key_chunk ? key_chunk->column_count : 0;
+    if (key_chunk && key_chunk->row_count != n) {
+        set_err(errbuf, errlen, "scan_table: key chunk row count differs");
+        return YT_ERR_INVALID_CHUNK;
+    }
+    for (int c = 0; c < nkey; c++) {
+        if (key_chunk->columns[c].value_type != YT_VT_INT64) {
+            set_err(errbuf, errlen, "scan_table: int64 key columns this round");
+            return YT_ERR_UNSUPPORTED;
+        }
+    }
+    const int ncols = nkey + nvcols;
 
     auto* H = new ScanChunkHandle();
-    uint64_t* d_bits = nullptr;
-    uint8_t* d_null = nullptr;
+    std::vector<uint64_t*> d_bits(ncols, nullptr);
+    std::vector<uint8_t*> d_null(ncols, nullptr);
     uint8_t* d_vis = nullptr;
+    uint8_t* d_vis2 = nullptr;
     unsigned long long* d_blk = nullptr;
     int64_t* d_off = nullptr;
     const int64_t kSegCap = 128 * 1024;
     int grid = (int)((n + 255) / 256);
     if (grid > 2048) grid = 2048;
     if (grid < 1) grid = 1;
-    int64_t total = 0;
-    int nseg = 0;
-    std::vector<unsigned long long> blk((size_t)grid);
-    std::vector<int64_t> seg_off;
-    std::vector<int64_t> seg_rows;
-    int64_t blob_bytes = 0;
     auto cleanup = [&]() {
-        g_pool.put(d_bits); g_pool.put(d_null); g_pool.put(d_vis);
+        for (auto* p : d_bits) g_pool.put(p);
+        for (auto* p : d_null) g_pool.put(p);
+        g_pool.put(d_vis); g_pool.put(d_vis2);
         g_pool.put(d_blk); g_pool.put(d_off);
     };
+    std::vector<int> col_is_dbl(ncols, 0);
+    std::vector<unsigned long long> blk((size_t)grid);
+    std::vector<int64_t> seg_rows, seg_bytes;
+    int64_t total = 0, col_bytes = 0, blob_bytes = 0;
+    int nseg = 0;
     if (n == 0) { *out_handle = H; return YT_OK; }
 
-    HIP_CHECK(pool_alloc(&d_bits, sizeof(uint64_t) * n));
-    HIP_CHECK(pool_alloc(&d_null, (size_t)n));
+    for (int c = 0; c < ncols; c++) {
+        HIP_CHECK(pool_alloc(&d_bits[c], sizeof(uint64_t) * n));
+        HIP_CHECK(pool_alloc(&d_null[c], (size_t)n));
+    }
     HIP_CHECK(pool_alloc(&d_vis, (size_t)n));
+    HIP_CHECK(pool_alloc(&d_vis2, (size_t)n));
     HIP_CHECK(pool_alloc(&d_blk, sizeof(unsigned long long) * grid));
-    /* the bridge emits width-64 zigzag int segments: int64 columns only
-     * (double/string bridge: later round — the read path supports them) */
-    for (int i = 0; i < col->val_seg_count; i++) {
-        if (col->val_segs[i].type >= YT_VSEG_DOUBLE_DENSE) {
-            set_err(errbuf, errlen,
-                    "versioned scan-chunk bridge: int64 value columns only");
-            cleanup();
-            delete H;
+
+    /* key columns: decode unversioned DirectDense int64 */
+    if (nkey) {
+        DeviceRun R2;
+        R2.stream = st;
+        unsigned maxw = 0;
+        int clamped = 0;
+        rc = setup_chunk(key_chunk, &R2, &maxw, 0, &clamped, errbuf, errlen);
+        if (rc != YT_OK) { cleanup(); delete H; return rc; }
+        HIP_CHECK(hipMemsetAsync(R2.d_err, 0, sizeof(unsigned), st));
+        for (int c = 0; c < nkey; c++) {
+            HIP_CHECK(ytql_launch_unvcol_to_arrays(R2.d_segs, R2.d_segex,
+                                                   R2.h_off[c], R2.h_cnt[c], n,
+                                                   d_bits[c], d_null[c],
+                                                   R2.d_err, st));
+        }
+        unsigned kerr = 0;
+        HIP_CHECK(hipMemcpy(&kerr, R2.d_err, sizeof(unsigned), hipMemcpyDeviceToHost));
+        if (kerr) {
+            set_err(errbuf, errlen, "scan_table: DirectDense int64 key segments only this round");
+            cleanup(); delete H;
             return YT_ERR_UNSUPPORTED;
         }
     }
-    rc = yt_gpu_versioned_read(col, timestamp, d_bits, d_null, d_vis, nullptr,
-                               stream, errbuf, errlen);
-    if (rc != YT_OK) { cleanup(); delete H; return rc; }
+    /* value columns: read at T (col 0 supplies the visibility mask — the
+     * write/delete timestamp lists are row-level, shared by every column) */
+    for (int j = 0; j < nvcols; j++) {
+        const YtVersionedColumn* vc = vcols[j];
+        int dbl = 0;
+        for (int i = 0; i < vc->val_seg_count; i++)
+            if (vc->val_segs[i].type >= YT_VSEG_DOUBLE_DENSE) dbl = 1;
+        col_is_dbl[nkey + j] = dbl;
+        rc = yt_gpu_versioned_read(vc, timestamp, d_bits[nkey + j],
+                                   d_null[nkey + j], j == 0 ? d_vis : d_vis2,
+                                   nullptr, stream, errbuf, errlen);
+        if (rc != YT_OK) { cleanup(); delete H; return rc; }
+    }
 
     HIP_CHECK(ytql_launch_vis_count(d_vis, n, d_blk, grid, st));
-    HIP_CHECK(hipMemcpy(blk.data(), d_blk,
-                        sizeof(unsigned long long) * grid,
+    HIP_CHECK(hipMemcpy(blk.data(), d_blk, sizeof(unsigned long long) * grid,
                         hipMemcpyDeviceToHost));
     {
         unsigned long long run = 0;
-        for (int i = 0; i < grid; i++) {
-            unsigned long long c = blk[i];
-            blk[i] = run;
-            run += c;
-        }
+        for (int i = 0; i < grid; i++) { unsigned long long c = blk[i]; blk[i] = run; run += c; }
         total = (int64_t)run;
     }
-    if (total == 0) { cleanup(); *out_handle = H; return YT_OK; }
-    HIP_CHECK(hipMemcpyAsync(d_blk, blk.data(),
-                             sizeof(unsigned long long) * grid,
+    nseg = (int)((total + kSegCap - 1) / kSegCap);
+    if (total == 0) {
+        cleanup();
+        *out_handle = H;
+        return YT_OK;
+    }
+    HIP_CHECK(hipMemcpyAsync(d_blk, blk.data(), sizeof(unsigned long long) * grid,
                              hipMemcpyHostToDevice, st));
 
-    nseg = (int)((total + kSegCap - 1) / kSegCap);
+    /* one blob: per column, per segment [header][values w64][null bitmap] */
+    seg_rows.resize(nseg);
+    seg_bytes.resize(nseg);
     for (int i = 0; i < nseg; i++) {
-        int64_t rows = i + 1 < nseg ? kSegCap : total - (int64_t)i * kSegCap;
-        int64_t bm = (((rows + 7) / 8) + 7) & ~(int64_t)7;
-        seg_off.push_back(blob_bytes);
-        seg_rows.push_back(rows);
-        blob_bytes += 8 + rows * 8 + bm;
+        seg_rows[i] = i + 1 < nseg ? kSegCap : total - (int64_t)i * kSegCap;
+        int64_t bm = (((seg_rows[i] + 7) / 8) + 7) & ~(int64_t)7;
+        seg_bytes[i] = 8 + seg_rows[i] * 8 + bm;
+        col_bytes += seg_bytes[i];
     }
+    blob_bytes = col_bytes * ncols;
     HIP_CHECK(pool_alloc(&H->blob, (size_t)blob_bytes + 8));
     HIP_CHECK(hipMemsetAsync(H->blob, 0, (size_t)blob_bytes, st));
-    for (int i = 0; i < nseg; i++) {
-        uint64_t hdr = (uint64_t)seg_rows[i] | (64ULL << 56);
-        HIP_CHECK(hipMemcpyAsync(H->blob + seg_off[i], &hdr, 8,
-                                 hipMemcpyHostToDevice, st));
-    }
     HIP_CHECK(pool_alloc(&d_off, sizeof(int64_t) * nseg));
-    HIP_CHECK(hipMemcpyAsync(d_off, seg_off.data(), sizeof(int64_t) * nseg,
-                             hipMemcpyHostToDevice, st));
-    HIP_CHECK(ytql_launch_vis_scatter(d_vis, d_null, d_bits, n, d_blk,
-                                      (uint64_t)kSegCap, d_off, H->blob,
-                                      grid, st));
-    HIP_CHECK(hipStreamSynchronize(st));
 
-    H->segs = (YtSegment*)calloc(nseg, sizeof(YtSegment));
-    H->col = (YtColumn*)calloc(1, sizeof(YtColumn));
-    for (int i = 0; i < nseg; i++) {
-        H->segs[i].type = YT_SEG_DIRECT_DENSE;
-        H->segs[i].row_count = (int32_t)seg_rows[i];
-        H->segs[i].min_value = 0;
-        H->segs[i].data = H->blob + seg_off[i];
-        H->segs[i].data_size = (i + 1 < nseg ? seg_off[i + 1] : blob_bytes)
-                             - seg_off[i];
+    H->segs = (YtSegment*)calloc((size_t)nseg * ncols, sizeof(YtSegment));
+    H->col = (YtColumn*)calloc(ncols, sizeof(YtColumn));
+    for (int c = 0; c < ncols; c++) {
+        std::vector<int64_t> off(nseg);
+        int64_t at = (int64_t)c * col_bytes;
+        for (int i = 0; i < nseg; i++) {
+            off[i] = at;
+            /* int: bitpack header rows|64<<56; double: [u64 count] */
+            uint64_t hdr = col_is_dbl[c] ? (uint64_t)seg_rows[i]
+                                         : ((uint64_t)seg_rows[i] | (64ULL << 56));
+            HIP_CHECK(hipMemcpyAsync(H->blob + at, &hdr, 8,
+                                     hipMemcpyHostToDevice, st));
+            YtSegment& sg = H->segs[c * nseg + i];
+            sg.type = col_is_dbl[c] ? YT_SEG_DOUBLE : YT_SEG_DIRECT_DENSE;
+            sg.row_count = (int32_t)seg_rows[i];
+            sg.min_value = 0;
+            sg.data = H->blob + at;
+            sg.data_size = seg_bytes[i];
+            at += seg_bytes[i];
+        }
+        HIP_CHECK(hipMemcpyAsync(d_off, off.data(), sizeof(int64_t) * nseg,
+                                 hipMemcpyHostToDevice, st));
+        HIP_CHECK(ytql_launch_vis_scatter(d_vis, d_null[c], d_bits[c], n, d_blk,
+                                          (uint64_t)kSegCap, d_off,
+                                          H->blob, col_is_dbl[c], grid, st));
+        HIP_CHECK(hipStreamSynchronize(st));
+        YtColumn& C2 = H->col[c];
+        C2.value_type = col_is_dbl[c] ? YT_VT_DOUBLE : YT_VT_INT64;
+        C2.segment_count = nseg;
+        C2.segments = &H->segs[c * nseg];
     }
-    H->col->value_type = YT_VT_INT64;
-    H->col->segment_count = nseg;
-    H->col->segments = H->segs;
     out_chunk->row_count = total;
-    out_chunk->column_count = 1;
+    out_chunk->column_count = ncols;
     out_chunk->columns = H->col;
     *out_handle = H;
     cleanup();

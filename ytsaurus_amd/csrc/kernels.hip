@@ -965,6 +965,41 @@ k_versioned_read(const VSegDev* segs, int nseg, int64_t total_rows,
     }
 }
 
+/* decode one unversioned DirectDense int64 column into per-row
+ * (decoded bits, null) arrays — the key-column side of the versioned
+ * table bridge (keys in the scan format are plain unversioned segments,
+ * rowset_builder.cpp key readers) */
+__global__ void __launch_bounds__(256)
+k_unvcol_to_arrays(const DevSeg* segs, const SegEx* segex,
+                   int seg_off, int seg_cnt, int64_t n,
+                   uint64_t* out_bits, uint8_t* out_null,
+                   unsigned* error_out)
+{
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t g = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         g < n; g += stride) {
+        int lo = 0, hi = seg_cnt;
+        while (lo + 1 < hi) {
+            int mid = (lo + hi) / 2;
+            if (segs[seg_off + mid].start_row <= g) lo = mid;
+            else hi = mid;
+        }
+        const DevSeg& sg = segs[seg_off + lo];
+        const SegEx& e = segex[seg_off + lo];
+        if (sg.type != YT_SEG_DIRECT_DENSE) { *error_out = 1; return; }
+        int64_t r = g - sg.start_row;
+        const uint8_t* bm = (const uint8_t*)sg.blob + e.off_bitmap_bytes;
+        int nul = bm_get(bm, r);
+        out_null[g] = (uint8_t)nul;
+        uint64_t mask = (e.w_values >= 64) ? ~0ULL : ((1ULL << e.w_values) - 1);
+        uint64_t raw = e.w_values
+            ? bp_gl(sg.blob + e.off_values_words, mask, e.w_values, r) : 0;
+        out_bits[g] = nul ? 0
+            : (sg.is_signed ? (uint64_t)zz_dec(sg.min_value + raw)
+                            : sg.min_value + raw);
+    }
+}
+
 /* versioned → unversioned chunk bridge: compact the visible rows of a
  * read-at-timestamp into reference-layout DirectDense width-64 segments
  * (min_value = 0, values in zigzag space) that the query engine scans
@@ -994,7 +1029,7 @@ k_vis_scatter(const uint8_t* vis, const uint8_t* nulls, const uint64_t* bits,
               int64_t n, const unsigned long long* block_bases,
               uint64_t seg_rows_cap,
               const int64_t* seg_blob_off,   /* byte offset of each segment */
-              char* out_blob)
+              char* out_blob, int is_double)
 {
     __shared__ unsigned long long s_wcnt[16];
     __shared__ unsigned long long s_run;
@@ -1027,10 +1062,17 @@ k_vis_scatter(const uint8_t* vis, const uint8_t* nulls, const uint64_t* bits,
             uint64_t seg = idx / seg_rows_cap;
             uint64_t j = idx - seg * seg_rows_cap;
             char* sb = out_blob + seg_blob_off[seg];
-            /* [header][values w64][null bitmap] — values in zigzag space */
-            int64_t v64 = (int64_t)bits[i];
-            uint64_t zz = ((uint64_t)v64 << 1) ^ (uint64_t)(v64 >> 63);
-            ((uint64_t*)(sb + 8))[j] = nulls[i] ? 0 : zz;
+            /* int64: [bitpack header][values w64 zigzag][null bitmap];
+             * double: [u64 count][raw doubles][null bitmap] — identical
+             * offsets, only the header/types differ (host writes them) */
+            uint64_t stored;
+            if (is_double) {
+                stored = bits[i];
+            } else {
+                int64_t v64 = (int64_t)bits[i];
+                stored = ((uint64_t)v64 << 1) ^ (uint64_t)(v64 >> 63);
+            }
+            ((uint64_t*)(sb + 8))[j] = nulls[i] ? 0 : stored;
             if (nulls[i]) {
                 uint64_t rows_here = 0;   /* bitmap offset needs seg rows */
                 (void)rows_here;
@@ -1040,7 +1082,7 @@ k_vis_scatter(const uint8_t* vis, const uint8_t* nulls, const uint64_t* bits,
                  * seg_rows = min(cap, total - seg*cap); the host encodes
                  * seg_rows in the header it wrote BEFORE this kernel runs */
                 uint64_t hdr = *(const uint64_t*)sb;
-                uint64_t rows = hdr & ((1ULL << 56) - 1);
+                uint64_t rows = hdr & ((1ULL << 56) - 1);   /* both layouts */
                 uint32_t* bm = (uint32_t*)(sb + 8 + rows * 8);
                 atomicOr(&bm[j >> 5], 1u << (j & 31));
             }
@@ -2917,6 +2959,20 @@ hipError_t ytql_launch_versioned_read(const VSegDev* segs, int nseg,
     return hipGetLastError();
 }
 
+hipError_t ytql_launch_unvcol_to_arrays(const DevSeg* segs, const SegEx* segex,
+                                        int seg_off, int seg_cnt, int64_t n,
+                                        uint64_t* out_bits, uint8_t* out_null,
+                                        unsigned* error_out, hipStream_t st)
+{
+    int block = 256;
+    int64_t want = (n + block - 1) / block;
+    int grid = (int)(want > 4096 ? 4096 : (want > 0 ? want : 1));
+    hipLaunchKernelGGL(k_unvcol_to_arrays, dim3(grid), dim3(block), 0, st,
+                       segs, segex, seg_off, seg_cnt, n, out_bits, out_null,
+                       error_out);
+    return hipGetLastError();
+}
+
 hipError_t ytql_launch_vis_count(const uint8_t* vis, int64_t n,
                                  unsigned long long* block_counts, int grid,
                                  hipStream_t st)
@@ -2931,11 +2987,12 @@ hipError_t ytql_launch_vis_scatter(const uint8_t* vis, const uint8_t* nulls,
                                    const unsigned long long* block_bases,
                                    uint64_t seg_rows_cap,
                                    const int64_t* seg_blob_off,
-                                   char* out_blob, int grid, hipStream_t st)
+                                   char* out_blob, int is_double,
+                                   int grid, hipStream_t st)
 {
     hipLaunchKernelGGL(k_vis_scatter, dim3(grid), dim3(256), 0, st,
                        vis, nulls, bits, n, block_bases, seg_rows_cap,
-                       seg_blob_off, out_blob);
+                       seg_blob_off, out_blob, is_double);
     return hipGetLastError();
 }
 
