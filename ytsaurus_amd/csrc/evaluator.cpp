@@ -381,7 +381,7 @@ static int build_devplan(const YtPlan* plan, const YtChunk* chunk, DevPlan* p,
     if (plan->agg_count < 1 || plan->agg_count > kMaxAggs) { set_err(errbuf, errlen, "need 1..4 aggregates"); return YT_ERR_UNSUPPORTED; }
     for (int a = 0; a < plan->agg_count; a++) {
         int f = plan->aggs[a]->func;
-        if (f != YT_AGG_SUM && f != YT_AGG_SUM1 && f != YT_AGG_MIN && f != YT_AGG_MAX) {
+        if (f < YT_AGG_SUM || f > YT_AGG_AVG) {
             set_err(errbuf, errlen, "unsupported aggregate");
             return YT_ERR_UNSUPPORTED;
         }
