@@ -2988,7 +2988,7 @@ extern "C" int yt_gpu_versioned_scan_table(
         unsigned kerr = 0;
         HIP_CHECK(hipMemcpy(&kerr, R2.d_err, sizeof(unsigned), hipMemcpyDeviceToHost));
         if (kerr) {
-            set_err(errbuf, errlen, "scan_table: DirectDense int64 key segments only this round");
+            set_err(errbuf, errlen, "scan_table: string key columns not bridged this round");
             cleanup(); delete H;
             return YT_ERR_UNSUPPORTED;
         }

@@ -986,17 +986,11 @@ k_unvcol_to_arrays(const DevSeg* segs, const SegEx* segex,
         }
         const DevSeg& sg = segs[seg_off + lo];
         const SegEx& e = segex[seg_off + lo];
-        if (sg.type != YT_SEG_DIRECT_DENSE) { *error_out = 1; return; }
+        if (sg.is_signed == 2) { *error_out = 1; return; }   /* string keys: not yet */
         int64_t r = g - sg.start_row;
-        const uint8_t* bm = (const uint8_t*)sg.blob + e.off_bitmap_bytes;
-        int nul = bm_get(bm, r);
-        out_null[g] = (uint8_t)nul;
-        uint64_t mask = (e.w_values >= 64) ? ~0ULL : ((1ULL << e.w_values) - 1);
-        uint64_t raw = e.w_values
-            ? bp_gl(sg.blob + e.off_values_words, mask, e.w_values, r) : 0;
-        out_bits[g] = nul ? 0
-            : (sg.is_signed ? (uint64_t)zz_dec(sg.min_value + raw)
-                            : sg.min_value + raw);
+        DVal v = seg_value_at(sg, e, r, YT_VT_INT64);
+        out_null[g] = (uint8_t)v.null_;
+        out_bits[g] = v.null_ ? 0 : v.bits;
     }
 }
 
