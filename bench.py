@@ -66,11 +66,13 @@ def gen_encode_column(n, seed, lo, hi, nthreads, base_seed_tag):
 STR_WINDOW = 65536             # distinct keys per generation slice (strgroup)
 
 
-def gen_encode_strkey_column(n, seed, nthreads):
+def gen_encode_strkey_column(n, seed, nthreads, window=STR_WINDOW):
     """String key column for the config-5 family: each 2Mi-row slice draws
-    from its own disjoint window of STR_WINDOW keys, so every 128Ki-row
-    segment sees ~56K distinct values and the reference's min-size rule
-    picks DictionaryDense — global distinct ≈ (n/SLICE)·STR_WINDOW."""
+    from its own disjoint window of `window` keys, so the reference's
+    min-size rule picks DictionaryDense (window <= ~210K keeps per-128Ki-
+    segment dictionaries smaller than direct) — global distinct ≈
+    (n/SLICE)·window. window ≈ 210K at 1B rows gives the NAMED config-5
+    scale: 100M global distinct keys."""
     import ytsaurus_amd as y
 
     slices = [(s, min(s + SLICE, n)) for s in range(0, n, SLICE)]
@@ -79,7 +81,7 @@ def gen_encode_strkey_column(n, seed, nthreads):
         i, (b, e) = item
         m = e - b
         rng = np.random.default_rng([seed, 777, i])
-        ids = i * STR_WINDOW + rng.integers(0, STR_WINDOW, m)
+        ids = i * window + rng.integers(0, window, m)
         # fixed-width 10-byte keys "k%09d", built with numpy (no per-row Python)
         dig = (ids[:, None] // 10 ** np.arange(8, -1, -1)) % 10
         arr = np.empty((m, 10), dtype=np.uint8)
@@ -262,8 +264,10 @@ def main():
                            "states not in the exchange format yet)"
         # BASELINE configs[4] family: dictionary-encoded string key GROUP BY
         # + double sum, via the global-atomic per-dictionary-id accumulate
-        # path; distinct keys ≈ slices × STR_WINDOW
-        key_space = ((n + SLICE - 1) // SLICE) * STR_WINDOW
+        # path; window sized so distinct ≈ --keys
+        nslices = (n + SLICE - 1) // SLICE
+        str_window = max(1024, min(210_000, (args.keys + nslices - 1) // nslices))
+        key_space = nslices * str_window
 
     t0 = time.monotonic()
     main.join_spec = None
@@ -277,7 +281,7 @@ def main():
         main.join_chunk = fchunk
         main.join_spec = y.Join(fchunk, 0, 0, [1])
     if args.workload == "strgroup":
-        enc_cols = [gen_encode_strkey_column(n, SEED + rank, cores),
+        enc_cols = [gen_encode_strkey_column(n, SEED + rank, cores, str_window),
                     gen_encode_double_column(n, SEED + rank, cores)]
     else:
         enc_cols = []
