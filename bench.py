@@ -77,11 +77,14 @@ def gen_encode_strkey_column(n, seed, nthreads, window=STR_WINDOW):
 
     slices = [(s, min(s + SLICE, n)) for s in range(0, n, SLICE)]
 
+    wstep = SLICE // 2   # window advances every 1Mi rows (8 segments)
+
     def work(item):
         i, (b, e) = item
         m = e - b
         rng = np.random.default_rng([seed, 777, i])
-        ids = i * window + rng.integers(0, window, m)
+        widx = (b + np.arange(m)) // wstep
+        ids = widx * window + rng.integers(0, window, m)
         # fixed-width 10-byte keys "k%09d", built with numpy (no per-row Python)
         dig = (ids[:, None] // 10 ** np.arange(8, -1, -1)) % 10
         arr = np.empty((m, 10), dtype=np.uint8)
@@ -264,10 +267,13 @@ def main():
                            "states not in the exchange format yet)"
         # BASELINE configs[4] family: dictionary-encoded string key GROUP BY
         # + double sum, via the global-atomic per-dictionary-id accumulate
-        # path; window sized so distinct ≈ --keys
-        nslices = (n + SLICE - 1) // SLICE
-        str_window = max(1024, min(210_000, (args.keys + nslices - 1) // nslices))
-        key_space = nslices * str_window
+        # path; per-1Mi-row key windows sized so global
+        # distinct ≈ --keys while each 128Ki-row segment still sees a
+        # dictionary-favoured density (the reference's min-size rule flips
+        # to direct strings above ~94K distinct per segment)
+        nwindows = (n + SLICE // 2 - 1) // (SLICE // 2)
+        str_window = max(1024, min(131_072, (args.keys + nwindows - 1) // nwindows))
+        key_space = nwindows * str_window
 
     t0 = time.monotonic()
     main.join_spec = None
