@@ -307,6 +307,8 @@ typedef struct {
     const int64_t* const* col_str_end;
     int ncols;
     int64_t row;
+    const int64_t* rowmap;   /* join expansion: expanded idx -> primary row */
+    int jf_base;             /* first joined foreign column (= ncols_primary) */
     int error;        /* YT_ERR_DIV_ZERO etc. */
 } EvalCtx;
 
@@ -317,16 +319,21 @@ static Val eval_expr(const YtExpr* e, EvalCtx* ctx)
     Val v = VNULL();
     switch (e->op) {
     case YT_EX_COLUMN: {
-        if (ctx->col_nulls[e->col][ctx->row]) return VNULL();
+        /* duplicate-key joins expand rows: primary columns read through
+         * rowmap (expanded index -> primary row), joined foreign columns
+         * (col >= jf_base) are materialized per EXPANDED row */
+        int64_t r = (ctx->rowmap && e->col < ctx->jf_base)
+            ? ctx->rowmap[ctx->row] : ctx->row;
+        if (ctx->col_nulls[e->col][r]) return VNULL();
         v.type = ctx->col_types[e->col];
         if (v.type == YT_VT_STRING) {
             const int64_t* ends = ctx->col_str_end[e->col];
-            v.str = ctx->col_str[e->col] + ends[ctx->row];
-            v.len = (uint32_t)(ends[ctx->row + 1] - ends[ctx->row]);
+            v.str = ctx->col_str[e->col] + ends[r];
+            v.len = (uint32_t)(ends[r + 1] - ends[r]);
             v.bits = 0;
             return v;
         }
-        v.bits = (uint64_t)ctx->col_vals[e->col][ctx->row];
+        v.bits = (uint64_t)ctx->col_vals[e->col][r];
         return v;
     }
     case YT_EX_LIT_I64:
@@ -657,6 +664,8 @@ typedef struct {
     int64_t** strend;
     int ncols_eff;         /* primary + joined foreign columns */
     const uint8_t* jdrop;  /* INNER-join misses (row excluded pre-filter) */
+    const int64_t* rowmap; /* duplicate-key join expansion (or NULL) */
+    int jf_base;
     GroupTable table;
     int error;
     int64_t rows_read;
@@ -673,6 +682,8 @@ static void* scan_worker(void* arg)
     ctx.col_str = (const char* const*)t->strblob;
     ctx.col_str_end = (const int64_t* const*)t->strend;
     ctx.ncols = t->ncols_eff;
+    ctx.rowmap = t->rowmap;
+    ctx.jf_base = t->jf_base;
     ctx.error = 0;
 
     Val keybuf[16];
@@ -787,6 +798,8 @@ static int emit_group_row(const YtPlan* plan, GroupTable* t, int64_t g,
         ctx.col_types = pt;
         ctx.ncols = kc + ac;
         ctx.row = 0;
+        ctx.rowmap = NULL;
+        ctx.jf_base = 0;
         ctx.error = 0;
         for (int p = 0; p < plan->project_count; p++) {
             Val v = eval_expr(plan->projects[p], &ctx);
@@ -1066,6 +1079,8 @@ int yto_execute(const YtPlan* plan, const YtChunk* chunk,
     int ncols_eff = ncols + jF;
     uint8_t* jdrop = NULL;
     int64_t n = chunk->row_count;
+    int64_t n_scan = chunk->row_count;   /* expanded row count under dup joins */
+    int64_t* join_rowmap = NULL;
     int rc = YT_OK;
 
     int64_t** vals = calloc(ncols_eff, sizeof(int64_t*));
@@ -1129,32 +1144,52 @@ int yto_execute(const YtPlan* plan, const YtChunk* chunk,
             fnulls[j] = malloc(fn ? fn : 1);
             rc = yto_decode_column(&fc->columns[cjf], fn, fvals[j], fnulls[j]);
         }
-        /* unique-key hash map: open addressing {key,null} -> row */
+        /* hash map with per-key CHAIN lists (duplicate foreign keys —
+         * registry.cpp MultiJoinOpHelper cross-product expansion):
+         * hrow[h] = chain head, hnext[r] = next same-key foreign row */
         uint64_t cap = 2048;
         while (cap < (uint64_t)fn * 2) cap <<= 1;
         int64_t* hrow = NULL;
         uint64_t* hkey = NULL;
         uint8_t* hused = NULL;
+        int64_t* hnext = NULL;
         int64_t null_row = -1;
+        int has_dups = 0;
         if (rc == YT_OK) {
             hrow = malloc(sizeof(int64_t) * cap);
             hkey = malloc(sizeof(uint64_t) * cap);
             hused = calloc(cap, 1);
+            hnext = malloc(sizeof(int64_t) * (fn ? fn : 1));
+            for (int64_t r2 = 0; r2 < fn; r2++) hnext[r2] = -1;
             for (int64_t r2 = 0; r2 < fn && rc == YT_OK; r2++) {
                 if (fknull[r2]) {
-                    if (null_row >= 0) { set_err(errbuf, errlen, "join: duplicate foreign key"); rc = YT_ERR_UNSUPPORTED; break; }
+                    if (null_row >= 0) has_dups = 1;
+                    hnext[r2] = null_row;
                     null_row = r2;
                     continue;
                 }
                 uint64_t h = splitmix64((uint64_t)fkey[r2]) & (cap - 1);
                 for (;;) {
                     if (!hused[h]) { hused[h] = 1; hkey[h] = (uint64_t)fkey[r2]; hrow[h] = r2; break; }
-                    if (hkey[h] == (uint64_t)fkey[r2]) { set_err(errbuf, errlen, "join: duplicate foreign key"); rc = YT_ERR_UNSUPPORTED; break; }
+                    if (hkey[h] == (uint64_t)fkey[r2]) {
+                        has_dups = 1;
+                        hnext[r2] = hrow[h];
+                        hrow[h] = r2;
+                        break;
+                    }
                     h = (h + 1) & (cap - 1);
                 }
             }
         }
-        if (rc == YT_OK) {
+        if (rc == YT_OK && has_dups &&
+            (plan->order_count > 0 || plan->agg_count == 0)) {
+            set_err(errbuf, errlen,
+                    "join: duplicate foreign keys with ORDER BY / plain scan "
+                    "not this round (GROUP BY plans supported)");
+            rc = YT_ERR_UNSUPPORTED;
+        }
+        if (rc == YT_OK && !has_dups) {
+            /* unique keys: per-primary-row foreign columns (fast path) */
             for (int j = 0; j < jF; j++) {
                 vals[ncols + j] = calloc(n ? n : 1, sizeof(int64_t));
                 nulls[ncols + j] = malloc(n ? n : 1);
@@ -1185,10 +1220,63 @@ int yto_execute(const YtPlan* plan, const YtChunk* chunk,
                     }
                 }
             }
+        } else if (rc == YT_OK) {
+            /* duplicate keys: expand (primary, match) pairs; primary
+             * columns read through rowmap, foreign columns materialized
+             * per EXPANDED row */
+            int64_t exp_cap = n ? n : 1;
+            int64_t n_exp = 0;
+            int64_t* eprim = malloc(sizeof(int64_t) * exp_cap);
+            int64_t* ematch = malloc(sizeof(int64_t) * exp_cap);
+            int pk = J->primary_key_col;
+            for (int64_t r2 = 0; r2 < n && rc == YT_OK; r2++) {
+                int64_t frow = -1;
+                if (nulls[pk][r2]) {
+                    frow = null_row;
+                } else {
+                    uint64_t h = splitmix64((uint64_t)vals[pk][r2]) & (cap - 1);
+                    while (hused[h]) {
+                        if (hkey[h] == (uint64_t)vals[pk][r2]) { frow = hrow[h]; break; }
+                        h = (h + 1) & (cap - 1);
+                    }
+                }
+                if (frow < 0 && !J->is_left) continue;
+                do {
+                    if (n_exp == exp_cap) {
+                        exp_cap *= 2;
+                        eprim = realloc(eprim, sizeof(int64_t) * exp_cap);
+                        ematch = realloc(ematch, sizeof(int64_t) * exp_cap);
+                        if (!eprim || !ematch) { rc = YT_ERR_CAPACITY; break; }
+                    }
+                    eprim[n_exp] = r2;
+                    ematch[n_exp] = frow;
+                    n_exp++;
+                    frow = frow >= 0 ? hnext[frow] : -1;
+                } while (frow >= 0);
+            }
+            if (rc == YT_OK) {
+                for (int j = 0; j < jF; j++) {
+                    vals[ncols + j] = calloc(n_exp ? n_exp : 1, sizeof(int64_t));
+                    nulls[ncols + j] = malloc(n_exp ? n_exp : 1);
+                    memset(nulls[ncols + j], 1, n_exp ? n_exp : 1);
+                    types[ncols + j] = (uint8_t)fc->columns[J->foreign_value_cols[j]].value_type;
+                    for (int64_t e2 = 0; e2 < n_exp; e2++) {
+                        int64_t m2 = ematch[e2];
+                        if (m2 >= 0 && !fnulls[j][m2]) {
+                            vals[ncols + j][e2] = fvals[j][m2];
+                            nulls[ncols + j][e2] = 0;
+                        }
+                    }
+                }
+                join_rowmap = eprim;
+                eprim = NULL;
+                n_scan = n_exp;
+            }
+            free(eprim); free(ematch);
         }
         free(fkey); free(fknull);
         for (int j = 0; j < jF; j++) { if (fvals) free(fvals[j]); if (fnulls) free(fnulls[j]); }
-        free(fvals); free(fnulls); free(hrow); free(hkey); free(hused);
+        free(fvals); free(fnulls); free(hrow); free(hkey); free(hused); free(hnext);
         if (rc != YT_OK) goto done;
     }
 
@@ -1215,6 +1303,8 @@ int yto_execute(const YtPlan* plan, const YtChunk* chunk,
         ctx.col_str = (const char* const*)strblob;
         ctx.col_str_end = (const int64_t* const*)strend;
         ctx.ncols = ncols_eff;
+        ctx.rowmap = NULL;
+        ctx.jf_base = 0;
         ctx.error = 0;
         int np = plan->project_count;
         for (int64_t r = 0; r < n; r++) {
@@ -1257,11 +1347,11 @@ int yto_execute(const YtPlan* plan, const YtChunk* chunk,
     {
         ScanTask* tasks = calloc(nthreads, sizeof(ScanTask));
         pthread_t* tids = malloc(sizeof(pthread_t) * nthreads);
-        int64_t per = (n + nthreads - 1) / nthreads;
+        int64_t per = (n_scan + nthreads - 1) / nthreads;
         int actual = 0;
         for (int i = 0; i < nthreads; i++) {
             int64_t b = (int64_t)i * per;
-            int64_t e = b + per > n ? n : b + per;
+            int64_t e = b + per > n_scan ? n_scan : b + per;
             if (b >= e) break;
             tasks[actual].plan = plan;
             tasks[actual].chunk = chunk;
@@ -1274,6 +1364,8 @@ int yto_execute(const YtPlan* plan, const YtChunk* chunk,
             tasks[actual].strend = strend;
             tasks[actual].ncols_eff = ncols_eff;
             tasks[actual].jdrop = jdrop;
+            tasks[actual].rowmap = join_rowmap;
+            tasks[actual].jf_base = ncols;
             gt_init(&tasks[actual].table, plan->key_count, plan->agg_count, 1024);
             actual++;
         }
@@ -1346,6 +1438,7 @@ done:
     }
     free(vals); free(nulls); free(types); free(strblob); free(strend);
     free(jdrop);
+    free(join_rowmap);
     return rc;
 }
 

@@ -111,10 +111,19 @@ struct JoinDev {
     const int32_t* f_off;         /* per FOREIGN column: first segment */
     const int32_t* f_cnt;
     const uint64_t* hkey;
-    const int64_t* hrow;          /* -1 = empty */
+    const int64_t* hrow;          /* -1 = empty (slot claim; dup keys may
+                                     occupy several slots — probes stop at
+                                     the FIRST matching slot, whose chead
+                                     holds the match list) */
     uint64_t hmask;
-    int64_t null_row;             /* foreign row with null key, or -1 */
+    int64_t null_row;             /* head of the null-key chain, or -1 */
     int64_t frows;
+    /* duplicate foreign keys (registry.cpp MultiJoinOpHelper cross-product
+     * expansion): per-slot match-list head + per-foreign-row next links */
+    const int64_t* chead;         /* [nslots] chain head row, -1 */
+    const int64_t* fnext;         /* [frows] next same-key row, -1 */
+    int32_t has_dups;             /* any key (incl. null) with >1 row */
+    int32_t pad2_;
 };
 
 /* versioned scan-format read: per-segment device

@@ -88,7 +88,7 @@ hipError_t ytql_launch_vis_scatter(const uint8_t*, const uint8_t*,
 hipError_t ytql_launch_join_build(const JoinDev*, int64_t, uint64_t*,
                                   long long*, unsigned long long*, unsigned*,
                                   hipStream_t);
-hipError_t ytql_launch_join_verify(const JoinDev*, int64_t, unsigned*,
+hipError_t ytql_launch_join_chain(const JoinDev*, int64_t, unsigned*,
                                    hipStream_t);
 hipError_t ytql_launch_scan_project(const DevPlan*, const DevSeg*, const SegEx*,
                                     const int32_t*, const int32_t*, int64_t,
@@ -1708,12 +1708,17 @@ struct JoinRun {
     long long* d_hrow = nullptr;
     unsigned long long* d_misc = nullptr;
 
+    long long* d_chead = nullptr;
+    long long* d_fnext = nullptr;
+
     JoinRun() { memset(&jd, 0, sizeof(jd)); }
     ~JoinRun()
     {
         g_pool.put(d_hkey);
         g_pool.put(d_hrow);
         g_pool.put(d_misc);
+        g_pool.put(d_chead);
+        g_pool.put(d_fnext);
     }
 };
 
@@ -1764,7 +1769,11 @@ static int setup_join(const YtPlan* plan, const YtChunk* chunk, JoinRun* JR,
     HIP_CHECK(pool_alloc(&JR->d_hkey, sizeof(uint64_t) * cap));
     HIP_CHECK(pool_alloc(&JR->d_hrow, sizeof(long long) * cap));
     HIP_CHECK(pool_alloc(&JR->d_misc, sizeof(unsigned long long) * 2));
+    HIP_CHECK(pool_alloc(&JR->d_chead, sizeof(long long) * cap));
+    HIP_CHECK(pool_alloc(&JR->d_fnext, sizeof(long long) * (fn ? fn : 1)));
     HIP_CHECK(hipMemsetAsync(JR->d_hrow, 0xFF, sizeof(long long) * cap, stream));
+    HIP_CHECK(hipMemsetAsync(JR->d_chead, 0xFF, sizeof(long long) * cap, stream));
+    HIP_CHECK(hipMemsetAsync(JR->d_fnext, 0xFF, sizeof(long long) * (fn ? fn : 1), stream));
     HIP_CHECK(hipMemsetAsync(JR->d_misc, 0, sizeof(unsigned long long) * 2, stream));
 
     jd.active = 1;
@@ -1786,22 +1795,26 @@ static int setup_join(const YtPlan* plan, const YtChunk* chunk, JoinRun* JR,
     jd.hmask = cap - 1;
     jd.null_row = -1;
     jd.frows = fn;
+    jd.chead = (const int64_t*)JR->d_chead;
+    jd.fnext = (const int64_t*)JR->d_fnext;
+    jd.has_dups = 0;
 
     if (fn > 0 && JR->Rf.nsegs > 0) {
+        /* d_err[0] = decode error, d_err[1] = has_dups flag (two words) */
         HIP_CHECK(hipMemsetAsync(JR->Rf.d_err, 0, sizeof(unsigned), stream));
+        HIP_CHECK(hipMemsetAsync(JR->d_misc + 1, 0, sizeof(unsigned long long), stream));
         HIP_CHECK(ytql_launch_join_build(&jd, fn, JR->d_hkey, JR->d_hrow,
-                                         JR->d_misc, JR->Rf.d_err, stream));
+                                         JR->d_misc, (unsigned*)(JR->d_misc + 1), stream));
         HIP_CHECK(hipMemcpy(&nr1, JR->d_misc, sizeof(unsigned long long),
                             hipMemcpyDeviceToHost));
         jd.null_row = nr1 ? (int64_t)(nr1 - 1) : -1;
-        HIP_CHECK(ytql_launch_join_verify(&jd, fn, JR->Rf.d_err, stream));
+        HIP_CHECK(ytql_launch_join_chain(&jd, fn, (unsigned*)(JR->d_misc + 1), stream));
+        unsigned long long dupw = 0;
+        HIP_CHECK(hipMemcpy(&dupw, JR->d_misc + 1, sizeof(unsigned long long),
+                            hipMemcpyDeviceToHost));
         HIP_CHECK(hipMemcpy(&kerr, JR->Rf.d_err, sizeof(unsigned),
                             hipMemcpyDeviceToHost));
-        if (kerr == 200) {
-            set_err(errbuf, errlen,
-                    "join: duplicate foreign key (unique-key joins this round)");
-            return YT_ERR_UNSUPPORTED;
-        }
+        jd.has_dups = ((unsigned*)&dupw)[1] != 0;
         if (kerr) {
             set_err(errbuf, errlen, "join: foreign decode error");
             return (int)kerr;
@@ -3170,6 +3183,15 @@ extern "C" int yt_gpu_query_execute(
                         (hipStream_t)(uintptr_t)options->stream, errbuf, errlen);
         if (rc) return rc;
         jd = &JR.jd;
+        if (JR.jd.has_dups && (plan->order_count > 0 || plan->agg_count == 0)) {
+            /* the ORDER BY / scan-project machinery tracks candidates by
+             * primary row id; a one-to-many join breaks that identity —
+             * grouped plans take the cross-product path instead */
+            set_err(errbuf, errlen,
+                    "join: duplicate foreign keys with ORDER BY / plain scan "
+                    "not this round (GROUP BY plans supported)");
+            return YT_ERR_UNSUPPORTED;
+        }
     }
 
     if (plan->agg_count == 0 && plan->key_count == 0) {

@@ -547,7 +547,7 @@ __device__ int64_t join_resolve(const DevPlan& p, ColCtx& c)
     for (;;) {
         int64_t r = jt.hrow[h];
         if (r < 0) break;
-        if (jt.hkey[h] == k.bits) { c.jrow = r; break; }
+        if (jt.hkey[h] == k.bits) { c.jrow = jt.chead[h]; break; }
         h = (h + 1) & jt.hmask;
     }
     return c.jrow;
@@ -1101,10 +1101,11 @@ __global__ void k_join_build(JoinDev jt, int64_t frows,
          r < frows; r += stride) {
         DVal k = jforeign_at(jt, jt.fkey_col, jt.fkey_shift, r, YT_VT_INT64);
         if (k.null_) {
-            /* at most one null-key row (unique keys) */
-            unsigned long long prev = atomicCAS(null_row_plus1, 0ULL,
-                                                (unsigned long long)(r + 1));
-            if (prev != 0ULL) atomicMax(error_out, 200u);
+            /* null-key rows chain off null_row (null joins null) */
+            unsigned long long prev = atomicExch(null_row_plus1,
+                                                 (unsigned long long)(r + 1));
+            ((int64_t*)jt.fnext)[r] = (int64_t)prev - 1;
+            if (prev != 0ULL) atomicOr(error_out + 1, 1u);   /* has_dups */
             continue;
         }
         uint64_t h = mix64(k.bits) & jt.hmask;
@@ -1122,9 +1123,11 @@ __global__ void k_join_build(JoinDev jt, int64_t frows,
     }
 }
 
-/* all inserts are visible now: the FIRST slot in a key's probe chain must
- * be this row's own slot, else two foreign rows share a key */
-__global__ void k_join_verify(JoinDev jt, int64_t frows, unsigned* error_out)
+/* all inserts are visible now: push every row onto the match list of its
+ * key's CANONICAL slot (first matching slot in probe order — duplicate
+ * keys may have claimed later slots too; those stay as tombstones that
+ * only lengthen probes). error_out[1] = has_dups. */
+__global__ void k_join_chain(JoinDev jt, int64_t frows, unsigned* error_out)
 {
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -1136,7 +1139,11 @@ __global__ void k_join_verify(JoinDev jt, int64_t frows, unsigned* error_out)
             int64_t rr = jt.hrow[h];
             if (rr < 0) break;                        /* unreachable */
             if (jt.hkey[h] == k.bits) {
-                if (rr != r) atomicMax(error_out, 200u);
+                int64_t prev = (int64_t)atomicExch(
+                    (unsigned long long*)&((int64_t*)jt.chead)[h],
+                    (unsigned long long)r);
+                ((int64_t*)jt.fnext)[r] = prev;
+                if (prev >= 0) atomicOr(error_out + 1, 1u);   /* has_dups */
                 break;
             }
             h = (h + 1) & jt.hmask;
@@ -1168,11 +1175,23 @@ k_scan_generic(DevPlan p, const DevSeg* segs, const SegEx* segex,
     for (int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
          r < row_count; r += stride) {
         c.row = r;
-        if (!join_row_ok(p, c)) continue;
+        /* duplicate foreign keys: one pass per match in the key's chain
+         * (registry.cpp MultiJoinOpHelper cross-product); unjoined plans
+         * and unique keys run the single pass they always did */
+        const bool joined = jd.active != 0;
+        int64_t m = -1;
+        if (joined) {
+            m = join_resolve(p, c);
+            if (m < 0 && !jd.is_left) continue;
+        }
+        for (;;) {
+        if (joined) { c.jrow = m; c.jrow_for = r; }
+        bool pass = true;
         if (p.filter_len) {
             DVal f = eval_prog(p, c, p.filter_off, p.filter_len);
-            if (f.null_ || f.bits == 0) continue;
+            if (f.null_ || f.bits == 0) pass = false;
         }
+        if (pass) {
         DVal key;
         if (p.kp_count) {
             /* composite packed key (see DevPlan kp_*) */
@@ -1204,6 +1223,11 @@ k_scan_generic(DevPlan p, const DevSeg* segs, const SegEx* segex,
         }
         if (c.error) { atomicMax(error_out, c.error); return; }
         table_update_generic(th, slots, p, key, aggv);
+        }   /* pass */
+        if (!joined || m < 0) break;
+        m = jd.fnext[m];
+        if (m < 0) break;
+        }   /* match loop */
     }
     if (c.error) atomicMax(error_out, c.error);
 }
@@ -3003,13 +3027,13 @@ hipError_t ytql_launch_join_build(const JoinDev* jd, int64_t frows,
     return hipGetLastError();
 }
 
-hipError_t ytql_launch_join_verify(const JoinDev* jd, int64_t frows,
+hipError_t ytql_launch_join_chain(const JoinDev* jd, int64_t frows,
                                    unsigned* error_out, hipStream_t st)
 {
     int block = 256;
     int64_t want = (frows + block - 1) / block;
     int grid = (int)(want > 4096 ? 4096 : (want > 0 ? want : 1));
-    hipLaunchKernelGGL(k_join_verify, dim3(grid), dim3(block), 0, st,
+    hipLaunchKernelGGL(k_join_chain, dim3(grid), dim3(block), 0, st,
                        *jd, frows, error_out);
     return hipGetLastError();
 }
