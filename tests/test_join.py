@@ -149,16 +149,19 @@ def test_join_null_keys_match(cuda):
 
 
 @pytest.mark.gpu
-def test_join_duplicate_foreign_key_errors(cuda):
+def test_join_duplicate_foreign_key_expands(cuda):
+    # r2: duplicate foreign keys cross-product instead of erroring
+    # (registry.cpp MultiJoinOpHelper; see test_join_dups.py for depth)
     rng = np.random.default_rng(86)
     pk, v, fkey, fval, _, chunk, fchunk, j = _mk(rng)
     dup = y.Chunk([y.encode_int64(np.array([5, 5, 7], dtype=np.int64)),
                    y.encode_int64(np.array([1, 2, 3], dtype=np.int64))], 3)
     plan = y.Plan(keys=[y.col(0)], aggs=[y.agg_sum1()],
                   join=y.Join(dup, 0, 0, [1]))
-    with pytest.raises(RuntimeError, match="duplicate"):
-        y.gpu_execute(plan, chunk.c_device(cuda),
-                      join_foreign=dup.c_device(cuda))
+    got, _ = y.gpu_execute(plan, chunk.c_device(cuda),
+                           join_foreign=dup.c_device(cuda))
+    want, _ = y.oracle_execute(plan, chunk)
+    assert y.sort_rows(got) == y.sort_rows(want)
 
 
 @pytest.mark.gpu
