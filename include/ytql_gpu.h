@@ -436,6 +436,9 @@ int yt_gpu_merge_states(
     const YtPlan* plan,
     const void* states_device,
     int64_t state_row_count,
+    const uint8_t* col_types,        /* optional YT_VT_* per bottom-query column
+                                        (double sums finalize as doubles); NULL
+                                        = all int64 */
     const YtExecOptions* options,
     YtRowset* output,
     YtStatistics* stats,

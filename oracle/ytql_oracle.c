@@ -1524,7 +1524,9 @@ int yto_partial(const YtPlan* plan, const YtChunk* chunk,
                 nonnull = (row[1 + a].type != YT_VT_NULL);
             }
         }
-        sr->meta = (uint64_t)knull | (nonnull << 8);
+        uint64_t sum_dbl = 0;
+        if (sum_idx >= 0 && row[1 + sum_idx].type == YT_VT_DOUBLE) sum_dbl = 2;
+        sr->meta = (uint64_t)knull | sum_dbl | (nonnull << 8);
         sr->sum_bits = sum_bits;
         sr->row_count = rowcount;
     }
@@ -1559,8 +1561,9 @@ int yto_merge(const YtPlan* plan, const YtStateRow* states, int64_t nstates,
             if (plan->aggs[a]->func != YT_AGG_SUM) continue;
             if (nonnull) {
                 Val nv;
-                nv.type = YT_VT_INT64;
+                nv.type = (states[i].meta & 2) ? YT_VT_DOUBLE : YT_VT_INT64;
                 nv.bits = states[i].sum_bits;
+                nv.str = 0; nv.len = 0;
                 sum_update_val(&t.states[g * plan->agg_count + a], nv);
             }
         }

@@ -221,7 +221,8 @@ def gpu_lib():
               C.c_int32, C.c_void_p, C.c_int64, C.POINTER(C.c_int64),
               C.POINTER(YtStatistics), C.c_char_p, C.c_size_t])
         _sig(lib, "yt_gpu_merge_states", C.c_int,
-             [C.POINTER(YtPlan), C.c_void_p, C.c_int64, C.POINTER(YtExecOptions),
+             [C.POINTER(YtPlan), C.c_void_p, C.c_int64,
+              C.POINTER(C.c_uint8), C.POINTER(YtExecOptions),
               C.POINTER(YtRowset), C.POINTER(YtStatistics), C.c_char_p, C.c_size_t])
         _sig(lib, "yt_encode_int64_column", C.c_int,
              [C.POINTER(C.c_int64), C.POINTER(C.c_uint8), C.c_int64, C.c_int32,

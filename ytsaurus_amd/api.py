@@ -766,7 +766,8 @@ def gpu_partial(plan, device_chunk, nparts, states_dev_ptr, capacity_rows,
 
 
 def gpu_merge(plan, states_dev_ptr, n_states, max_groups_hint=0, stream=0,
-              out_capacity=None, rowset=None, raw_rowset=False):
+              out_capacity=None, rowset=None, raw_rowset=False,
+              col_types=None):
     opts = YtExecOptions(max_groups_hint=max_groups_hint, stream=stream)
     if rowset is not None:
         rs = rowset
@@ -775,9 +776,13 @@ def gpu_merge(plan, states_dev_ptr, n_states, max_groups_hint=0, stream=0,
         rs = _mk_rowset(cap)
     st = YtStatistics()
     err = C.create_string_buffer(512)
+    ct = None
+    if col_types is not None:
+        arr = list(col_types)[:8] + [VT_INT64] * max(0, 8 - len(col_types))
+        ct = (C.c_uint8 * 8)(*arr)
     rc = _abi.gpu_lib().yt_gpu_merge_states(
-        C.byref(plan.c), C.c_void_p(states_dev_ptr), n_states, C.byref(opts),
-        C.byref(rs), C.byref(st), err, 512)
+        C.byref(plan.c), C.c_void_p(states_dev_ptr), n_states, ct,
+        C.byref(opts), C.byref(rs), C.byref(st), err, 512)
     _check(rc, err)
     if raw_rowset:
         return rs, st

@@ -105,7 +105,7 @@ hipError_t ytql_launch_compact(TableHdr*, TableHdr*, const unsigned long long*, 
                                OutGroup*, unsigned long long*, uint64_t, hipStream_t);
 hipError_t ytql_launch_part_count(const OutGroup*, int64_t, int, int,
                                   unsigned long long*, hipStream_t);
-hipError_t ytql_launch_part_scatter(const OutGroup*, int64_t, int, int,
+hipError_t ytql_launch_part_scatter(const OutGroup*, int64_t, int, int, int,
                                     unsigned long long*, YtStateRow*, hipStream_t);
 hipError_t ytql_launch_merge_states(const YtStateRow*, int64_t, int, int,
                                     TableHdr*, unsigned long long*, hipStream_t);
@@ -3429,7 +3429,17 @@ extern "C" int yt_gpu_query_partial(
         HIP_CHECK(hipMemcpyAsync(d_counts, cursors.data(),
                                  sizeof(unsigned long long) * partition_count,
                                  hipMemcpyHostToDevice, R.stream));
+        int sum_is_double = 0;
+        if (sum_slot >= 0) {
+            uint8_t ct2[kMaxCols];
+            memset(ct2, YT_VT_INT64, sizeof(ct2));
+            for (int c2 = 0; c2 < chunk->column_count && c2 < kMaxCols; c2++)
+                ct2[c2] = (uint8_t)chunk->columns[c2].value_type;
+            sum_is_double =
+                expr_static_type(plan->aggs[sum_slot]->arg, ct2) == YT_VT_DOUBLE;
+        }
         HIP_CHECK(ytql_launch_part_scatter(R.d_groups, total, partition_count, sum_slot,
+                                           sum_is_double,
                                            d_counts, (YtStateRow*)states_device, R.stream));
         HIP_CHECK(hipStreamSynchronize(R.stream));
         g_pool.put(d_counts);
@@ -3450,6 +3460,7 @@ fail:
 
 extern "C" int yt_gpu_merge_states(
     const YtPlan* plan, const void* states_device, int64_t state_row_count,
+    const uint8_t* col_types,
     const YtExecOptions* options, YtRowset* output, YtStatistics* stats,
     char* errbuf, size_t errlen)
 {
@@ -3510,6 +3521,10 @@ extern "C" int yt_gpu_merge_states(
         col.value_type = YT_VT_INT64;
         col.segment_count = 0;
         std::vector<YtColumn> cols(kMaxCols, col);
+        if (col_types) {
+            for (int c2 = 0; c2 < kMaxCols; c2++)
+                cols[c2].value_type = col_types[c2];
+        }
         YtChunk fake;
         fake.row_count = 0;
         fake.column_count = kMaxCols;

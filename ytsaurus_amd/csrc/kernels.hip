@@ -2747,7 +2747,8 @@ __global__ void k_part_count(const OutGroup* groups, int64_t n, int nparts,
  * then place rows at block-local claims (order within a partition is
  * arbitrary — the merge is order-free) */
 __global__ void k_part_scatter(const OutGroup* groups, int64_t n, int nparts,
-                               int sum_slot, unsigned long long* cursors,
+                               int sum_slot, int sum_is_double,
+                               unsigned long long* cursors,
                                YtStateRow* out)
 {
     __shared__ unsigned lcnt[kMaxParts];
@@ -2777,7 +2778,8 @@ __global__ void k_part_scatter(const OutGroup* groups, int64_t n, int nparts,
         YtStateRow& sr = out[pos];
         sr.key_bits = g.key_bits;
         uint64_t nonnull = (sum_slot >= 0) ? (g.agg_nonnull[sum_slot] ? 1 : 0) : 0;
-        sr.meta = (uint64_t)knull | (nonnull << 8);
+        sr.meta = (uint64_t)knull | (nonnull << 8)
+                | (sum_is_double ? 2ULL : 0ULL);
         sr.sum_bits = (sum_slot >= 0) ? g.agg_bits[sum_slot] : 0;
         sr.row_count = g.cnt;
     }
@@ -2809,7 +2811,13 @@ __global__ void k_merge_states(const YtStateRow* states, int64_t n,
         }
         atomicAdd(cntp, (unsigned long long)sr.row_count);
         if (sum_slot >= 0 && (sr.meta >> 8)) {
-            atomicAdd(aggp + 2 * sum_slot, (unsigned long long)sr.sum_bits);
+            if (sr.meta & 2) {
+                /* double sum state: FP merge (udf/sum.c:27-35) */
+                atomicAdd((double*)(aggp + 2 * sum_slot),
+                          __longlong_as_double((long long)sr.sum_bits));
+            } else {
+                atomicAdd(aggp + 2 * sum_slot, (unsigned long long)sr.sum_bits);
+            }
             atomicAdd(aggp + 2 * sum_slot + 1, 1ULL);
         }
     }
@@ -3149,14 +3157,15 @@ hipError_t ytql_launch_part_count(const OutGroup* groups, int64_t n, int nparts,
 }
 
 hipError_t ytql_launch_part_scatter(const OutGroup* groups, int64_t n, int nparts,
-                                    int sum_slot, unsigned long long* cursors,
+                                    int sum_slot, int sum_is_double,
+                                    unsigned long long* cursors,
                                     YtStateRow* out, hipStream_t st)
 {
     int block = 256;
     int64_t want = (n + block - 1) / block;
     int grid = (int)(want > 2048 ? 2048 : (want ? want : 1));
     hipLaunchKernelGGL(k_part_scatter, dim3(grid), dim3(block), 0, st,
-                       groups, n, nparts, sum_slot, cursors, out);
+                       groups, n, nparts, sum_slot, sum_is_double, cursors, out);
     return hipGetLastError();
 }
 
