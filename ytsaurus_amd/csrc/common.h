@@ -304,12 +304,25 @@ struct StrGroupParams {
  * The representative's hash is NOT stored: readers recompute it from the
  * precomputed per-entry hash array via (seg, id), which avoids any
  * publication ordering between the claim and a hash field. */
+/* 64-byte slot, one cache line: rep claims the slot (CAS); the claimer
+ * then publishes the key identity in-slot with RELAXED agent stores —
+ * ident = hash48|len16, pfx = the key's first 16 bytes. Readers verify
+ * with relaxed loads; a stale (zero / partially visible) identity only
+ * demotes the probe to the exact owner-array compare, never a wrong
+ * accept: ident mismatch is exact (FNV48+len of equal strings always
+ * match), ident match with len <= 16 is decided by pfx (the FULL bytes),
+ * len > 16 falls back to the byte compare against the owner's dictionary
+ * entry. */
 struct StrSlot {
-    unsigned long long rep;
+    unsigned long long rep;       /* (seg+1)<<32 | id; 0 = empty */
+    uint64_t ident;               /* hash48<<16 | len; 0 = unpublished */
+    uint64_t pfx[2];              /* first 16 key bytes, zero-padded */
     uint64_t sum_bits;
     uint64_t cnt;
     uint64_t nonnull;
+    uint64_t pad_;
 };
+static_assert(sizeof(StrSlot) == 64, "one cache line per slot");
 
 /* ORDER BY ... LIMIT k-selection (see kernels.hip k_topk_*).
  * Range-adaptive digits: candidates are mapped keys in [lo, hi]; digit =

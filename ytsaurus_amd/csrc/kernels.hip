@@ -2526,6 +2526,17 @@ __global__ void k_strgrp_merge(const DevSeg* segs, const SegEx* segex,
         uint64_t h = hashes[g];
         unsigned long long rep = ((unsigned long long)(lo + 1) << 32)
                                | (unsigned long long)my_id;
+        /* in-slot identity (see common.h StrSlot): hash48|len + 16-byte
+         * prefix; a true duplicate resolves from the slot's own cache line
+         * instead of three scattered owner reads (the merge was 59% of the
+         * named config-5 step, fully latency-bound: WAIT_ANY 0.66 + 0.34) */
+        const uint64_t my_ident = ((h >> 16) << 16) | (uint64_t)(my_len & 0xFFFF);
+        uint64_t my_pfx[2] = {0, 0};
+        {
+            uint32_t npfx = my_len < 16 ? my_len : 16;
+            for (uint32_t k = 0; k < npfx; k++)
+                ((char*)my_pfx)[k] = my_p[k];
+        }
 
         uint64_t sidx = mix64(h) & mask;
         StrSlot* slot = nullptr;
@@ -2541,6 +2552,15 @@ __global__ void k_strgrp_merge(const DevSeg* segs, const SegEx* segex,
             if (cur == 0ULL) {
                 cur = atomicCAS(&cand->rep, 0ULL, rep);
                 if (cur == 0ULL) {
+                    /* claimed: publish the identity (relaxed agent stores;
+                     * readers that catch it unpublished just fall back to
+                     * the exact owner-array compare) */
+                    __hip_atomic_store(&cand->pfx[0], my_pfx[0],
+                                       __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+                    __hip_atomic_store(&cand->pfx[1], my_pfx[1],
+                                       __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+                    __hip_atomic_store(&cand->ident, my_ident,
+                                       __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
                     unsigned long long t = atomicAdd(&th->ngroups, 1ULL);
                     if (th->group_limit > 0 && (int64_t)t >= th->group_limit)
                         th->overflow = 2;
@@ -2550,6 +2570,21 @@ __global__ void k_strgrp_merge(const DevSeg* segs, const SegEx* segex,
             }
             if (cur == rep) { slot = cand; break; }
             {
+                uint64_t oident = __hip_atomic_load(&cand->ident, __ATOMIC_RELAXED,
+                                                    __HIP_MEMORY_SCOPE_AGENT);
+                if (oident != 0) {
+                    if (oident != my_ident) { sidx = (sidx + 1) & mask; continue; }
+                    if (my_len <= 16) {
+                        uint64_t p0 = __hip_atomic_load(&cand->pfx[0], __ATOMIC_RELAXED,
+                                                        __HIP_MEMORY_SCOPE_AGENT);
+                        uint64_t p1 = __hip_atomic_load(&cand->pfx[1], __ATOMIC_RELAXED,
+                                                        __HIP_MEMORY_SCOPE_AGENT);
+                        if (p0 == my_pfx[0] && p1 == my_pfx[1]) { slot = cand; break; }
+                        /* prefix mismatch on a hash48|len match: either a
+                         * genuine collision or a not-yet-visible prefix —
+                         * decide exactly from the owner's dictionary entry */
+                    }
+                }
                 int oseg = (int)(cur >> 32) - 1;
                 int64_t oid = (int64_t)(cur & 0xFFFFFFFFULL);
                 uint64_t oh = hashes[acc_base[oseg] + oid - 1];
