@@ -318,8 +318,8 @@ struct StrSlot {
     uint64_t ident;               /* hash48<<16 | len; 0 = unpublished */
     uint64_t pfx[2];              /* first 16 key bytes, zero-padded */
     uint64_t sum_bits;
-    uint64_t cnt;
-    uint64_t nonnull;
+    uint64_t cnt;                 /* low 32 = row count, high 32 = nonnull */
+    uint64_t pad0_;
     uint64_t pad_;
 };
 static_assert(sizeof(StrSlot) == 64, "one cache line per slot");

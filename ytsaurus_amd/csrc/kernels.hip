@@ -2602,10 +2602,11 @@ __global__ void k_strgrp_merge(const DevSeg* segs, const SegEx* segex,
             sidx = (sidx + 1) & mask;
         }
         if (!slot) { th->overflow = 1; continue; }
-        atomicAdd((unsigned long long*)&slot->cnt, cn & 0xFFFFFFFFULL);
+        /* one packed atomic: cnt in the low half, nonnull in the high half
+         * (both < 2^32 — segment row caps; carries cannot cross) */
+        atomicAdd((unsigned long long*)&slot->cnt, cn);
         unsigned long long nn = cn >> 32;
         if (nn) {
-            atomicAdd((unsigned long long*)&slot->nonnull, nn);
             if (val_is_double)
                 atomicAdd((double*)&slot->sum_bits,
                           __longlong_as_double(acc[2 * g + 1]));
@@ -2717,8 +2718,8 @@ __global__ void k_strgrp_compact(const DevSeg* segs, const SegEx* segex,
                 g.len = len;
                 g.pad_ = 0;
                 g.sum_bits = sl->sum_bits;
-                g.cnt = sl->cnt;
-                g.nonnull = sl->nonnull;
+                g.cnt = sl->cnt & 0xFFFFFFFFULL;
+                g.nonnull = sl->cnt >> 32;
             }
         }
         __syncthreads();
