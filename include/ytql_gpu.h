@@ -429,6 +429,28 @@ int yt_gpu_query_partial(
     YtStatistics* stats,
     char* errbuf, size_t errlen);
 
+/* Multi-key sharding support: a multi-key GROUP BY packs its key columns'
+ * zigzag-space values into one <=62-bit composite (DevPlan kp_*, mirroring
+ * the coordinated shuffle's key-prefix hash, shuffling_reader.cpp:40-42).
+ * Across ranks the packing BASES must agree, so the caller
+ *   (1) asks each rank for its local per-key-column zigzag ranges,
+ *   (2) all-reduces min over zzmin and max over zzmax,
+ *   (3) passes the common ranges to partial AND merge.
+ * Single-key plans ignore the ranges. key_zzmin/key_zzmax have
+ * plan->key_count entries; each key must be a plain int64/uint64/boolean
+ * column (the single-GPU composite-key rule). */
+int yt_gpu_key_ranges(
+    const YtPlan* plan, const YtChunk* chunk,
+    uint64_t* key_zzmin, uint64_t* key_zzmax,   /* out, length key_count */
+    uint64_t stream, char* errbuf, size_t errlen);
+
+int yt_gpu_query_partial_mk(
+    const YtPlan* plan, const YtChunk* chunk, const YtExecOptions* options,
+    int32_t partition_count, void* states_device, int64_t capacity_rows,
+    int64_t* part_counts,
+    const uint64_t* key_zzmin, const uint64_t* key_zzmax,  /* common ranges */
+    YtStatistics* stats, char* errbuf, size_t errlen);
+
 /* front query: merge state rows (device buffer, e.g. post all-to-all) and
  * finalize into `output` (host). Implements Merge+finalize semantics of
  * cg_fragment_compiler.cpp:4116-4134 + udf/sum.c:47-65. */
@@ -442,6 +464,15 @@ int yt_gpu_merge_states(
     const YtExecOptions* options,
     YtRowset* output,
     YtStatistics* stats,
+    char* errbuf, size_t errlen);
+
+/* multi-key front merge: state key_bits are composites packed with the
+ * COMMON ranges (must equal the ones passed to every rank's partial). */
+int yt_gpu_merge_states_mk(
+    const YtPlan* plan, const void* states_device, int64_t state_row_count,
+    const uint8_t* col_types,
+    const uint64_t* key_zzmin, const uint64_t* key_zzmax,
+    const YtExecOptions* options, YtRowset* output, YtStatistics* stats,
     char* errbuf, size_t errlen);
 
 /* ---- chunk encoder (host-side product component; the synthetic-data

@@ -1714,6 +1714,224 @@ static int decode_string_segment(const YtSegment* seg,
     return 0;
 }
 
+/* ---- multi-key two-phase (common-range composite packing) ----
+ * Mirrors the GPU path exactly (evaluator pack_group_key + DevPlan kp_*):
+ * component i (plain int64/uint64/boolean column) is coded as
+ *   0                      when null
+ *   1 + (zzspace(v) - lo)  otherwise   (zigzag for int64, raw bits else)
+ * packed at a running shift; total width <= 62 bits. The caller passes the
+ * CROSS-RANK-reduced [lo, hi] per component so every rank and the merge
+ * pack identically. */
+typedef struct {
+    int count;
+    int col[4];
+    int is_signed[4];
+    int shift[4];
+    int bits[4];
+    uint64_t base[4];
+} MkPack;
+
+static int mk_resolve(const YtPlan* plan, const uint8_t* col_types,
+                      const uint64_t* kzmin, const uint64_t* kzmax,
+                      MkPack* mk, char* errbuf, size_t errlen)
+{
+    mk->count = plan->key_count;
+    int shift = 0;
+    for (int i = 0; i < plan->key_count; i++) {
+        const YtExpr* e = plan->keys[i];
+        if (!e || e->op != YT_EX_COLUMN) {
+            set_err(errbuf, errlen, "mk: plain key columns only");
+            return YT_ERR_UNSUPPORTED;
+        }
+        int c = e->col;
+        mk->col[i] = c;
+        uint8_t vt = col_types ? col_types[c] : YT_VT_INT64;
+        mk->is_signed[i] = vt == YT_VT_INT64;
+        uint64_t lo = kzmin[i], hi = kzmax[i];
+        uint64_t span = hi >= lo ? hi - lo : 0;
+        int bits = 1;
+        while (bits < 64 && ((span + 1) >> bits) != 0) bits++;
+        mk->base[i] = lo;
+        mk->bits[i] = bits;
+        mk->shift[i] = shift;
+        shift += bits;
+    }
+    if (shift > 62) {
+        set_err(errbuf, errlen, "mk: composite key wider than 62 bits");
+        return YT_ERR_UNSUPPORTED;
+    }
+    return YT_OK;
+}
+
+static uint64_t mk_pack_row(const MkPack* mk, const YtValue* row)
+{
+    uint64_t kb = 0;
+    for (int i = 0; i < mk->count; i++) {
+        uint64_t enc = 0;
+        if (row[i].type != YT_VT_NULL) {
+            uint64_t z = mk->is_signed[i]
+                ? zigzag_encode64((int64_t)row[i].data.bits)
+                : row[i].data.bits;
+            enc = 1 + (z - mk->base[i]);
+        }
+        kb |= enc << mk->shift[i];
+    }
+    return kb;
+}
+
+ORACLE_EXPORT
+int yto_partial_mk(const YtPlan* plan, const YtChunk* chunk,
+                   int32_t partition_count,
+                   const uint64_t* key_zzmin, const uint64_t* key_zzmax,
+                   YtStateRow* out, int64_t capacity_rows,
+                   int64_t* part_counts,
+                   int nthreads, char* errbuf, size_t errlen)
+{
+    if (plan->join) { set_err(errbuf, errlen, "partial: no join"); return YT_ERR_UNSUPPORTED; }
+    if (plan->key_count < 1 || plan->key_count > 4) { set_err(errbuf, errlen, "partial: 1..4 keys"); return YT_ERR_UNSUPPORTED; }
+    int sum_idx = -1;
+    for (int a = 0; a < plan->agg_count; a++) {
+        if (plan->aggs[a]->func == YT_AGG_SUM) sum_idx = a;
+        else if (plan->aggs[a]->func != YT_AGG_SUM1) { set_err(errbuf, errlen, "partial: sum/sum1 only"); return YT_ERR_UNSUPPORTED; }
+    }
+    uint8_t ct[64];
+    memset(ct, YT_VT_INT64, sizeof(ct));
+    for (int c = 0; c < chunk->column_count && c < 64; c++)
+        ct[c] = (uint8_t)chunk->columns[c].value_type;
+    MkPack mk;
+    int rc = mk_resolve(plan, ct, key_zzmin, key_zzmax, &mk, errbuf, errlen);
+    if (rc != YT_OK) return rc;
+
+    YtPlan local = *plan;
+    local.project_count = 0;
+    local.projects = NULL;
+    int kc = plan->key_count;
+    int ncols = kc + plan->agg_count;
+    int64_t cap = capacity_rows;
+    YtValue* tmp = malloc(sizeof(YtValue) * cap * ncols);
+    if (!tmp) { set_err(errbuf, errlen, "partial_mk: oom"); return YT_ERR_CAPACITY; }
+    YtRowset rs;
+    memset(&rs, 0, sizeof(rs));
+    rs.values = tmp;
+    rs.capacity_rows = cap;
+    YtStatistics st;
+    memset(&st, 0, sizeof(st));
+    rc = yto_execute(&local, chunk, &rs, &st, nthreads, errbuf, errlen);
+    if (rc != YT_OK) {
+        if (rc == YT_ERR_CAPACITY) set_err(errbuf, errlen, "partial_mk: local group rowset too small");
+        free(tmp);
+        return rc;
+    }
+
+    int64_t* counts = calloc(partition_count, sizeof(int64_t));
+    for (int64_t r = 0; r < rs.row_count; r++) {
+        uint64_t kb = mk_pack_row(&mk, rs.values + r * ncols);
+        counts[yto_partition_hash(kb, 0) % (uint64_t)partition_count]++;
+    }
+    int64_t total = 0;
+    int64_t* offs = calloc(partition_count, sizeof(int64_t));
+    for (int p = 0; p < partition_count; p++) { offs[p] = total; total += counts[p]; }
+    if (total > capacity_rows) { free(tmp); free(counts); free(offs); set_err(errbuf, errlen, "partial_mk: state buffer too small"); return YT_ERR_CAPACITY; }
+    for (int64_t r = 0; r < rs.row_count; r++) {
+        const YtValue* row = rs.values + r * ncols;
+        uint64_t kb = mk_pack_row(&mk, row);
+        int64_t p = (int64_t)(yto_partition_hash(kb, 0) % (uint64_t)partition_count);
+        YtStateRow* sr = &out[offs[p]++];
+        sr->key_bits = kb;
+        uint64_t rowcount = 0, sum_bits = 0, nonnull = 0, sum_dbl = 0;
+        for (int a = 0; a < plan->agg_count; a++) {
+            if (plan->aggs[a]->func == YT_AGG_SUM1) rowcount = row[kc + a].data.bits;
+            else if (a == sum_idx) {
+                sum_bits = row[kc + a].data.bits;
+                nonnull = (row[kc + a].type != YT_VT_NULL);
+                if (row[kc + a].type == YT_VT_DOUBLE) sum_dbl = 2;
+            }
+        }
+        sr->meta = sum_dbl | (nonnull << 8);
+        sr->sum_bits = sum_bits;
+        sr->row_count = rowcount;
+    }
+    for (int p = 0; p < partition_count; p++) part_counts[p] = counts[p];
+    free(tmp); free(counts); free(offs);
+    return YT_OK;
+}
+
+ORACLE_EXPORT
+int yto_merge_mk(const YtPlan* plan, const YtStateRow* states, int64_t nstates,
+                 const uint8_t* col_types,
+                 const uint64_t* key_zzmin, const uint64_t* key_zzmax,
+                 YtRowset* output, char* errbuf, size_t errlen)
+{
+    if (plan->key_count < 1 || plan->key_count > 4) { set_err(errbuf, errlen, "merge: 1..4 keys"); return YT_ERR_UNSUPPORTED; }
+    MkPack mk;
+    int rc = mk_resolve(plan, col_types, key_zzmin, key_zzmax, &mk, errbuf, errlen);
+    if (rc != YT_OK) return rc;
+    GroupTable t;
+    gt_init(&t, 1, plan->agg_count, 1024);
+    for (int64_t i = 0; i < nstates; i++) {
+        Val key;
+        key.type = YT_VT_UINT64;
+        key.bits = states[i].key_bits;
+        key.str = 0; key.len = 0;
+        int64_t g = gt_upsert(&t, &key);
+        if (g < 0) { gt_free(&t); return YT_ERR_CAPACITY; }
+        t.rowcounts[g] += states[i].row_count;
+        uint64_t nonnull = states[i].meta >> 8;
+        for (int a = 0; a < plan->agg_count; a++) {
+            if (plan->aggs[a]->func != YT_AGG_SUM) continue;
+            if (nonnull) {
+                Val nv;
+                nv.type = (states[i].meta & 2) ? YT_VT_DOUBLE : YT_VT_INT64;
+                nv.bits = states[i].sum_bits;
+                nv.str = 0; nv.len = 0;
+                sum_update_val(&t.states[g * plan->agg_count + a], nv);
+            }
+        }
+    }
+    output->row_count = 0;
+    output->totals_row = 0;
+    int ncols = plan->key_count + plan->agg_count;
+    output->column_count = ncols;
+    for (int64_t g = 0; g < t.ngroups; g++) {
+        if (output->row_count >= output->capacity_rows) { gt_free(&t); return YT_ERR_CAPACITY; }
+        YtValue* dst = output->values + output->row_count * ncols;
+        uint64_t kb = t.keys[g].bits;
+        for (int i = 0; i < mk.count; i++) {
+            uint64_t mask = mk.bits[i] >= 64 ? ~0ULL : ((1ULL << mk.bits[i]) - 1);
+            uint64_t code = (kb >> mk.shift[i]) & mask;
+            dst[i].id = (uint16_t)i;
+            dst[i].flags = 0;
+            dst[i].length = 0;
+            if (code == 0) {
+                dst[i].type = YT_VT_NULL;
+                dst[i].data.bits = 0;
+            } else {
+                uint64_t z = code - 1 + mk.base[i];
+                dst[i].type = col_types ? col_types[mk.col[i]] : YT_VT_INT64;
+                dst[i].data.bits = mk.is_signed[i]
+                    ? (uint64_t)zigzag_decode64(z) : z;
+            }
+        }
+        for (int a = 0; a < plan->agg_count; a++) {
+            YtValue* v = &dst[mk.count + a];
+            v->id = (uint16_t)(mk.count + a);
+            v->flags = 0;
+            v->length = 0;
+            if (plan->aggs[a]->func == YT_AGG_SUM1) {
+                v->type = YT_VT_INT64;
+                v->data.bits = t.rowcounts[g];
+            } else {
+                Val st2 = t.states[g * plan->agg_count + a];
+                v->type = st2.type;
+                v->data.bits = st2.bits;
+            }
+        }
+        output->row_count++;
+    }
+    gt_free(&t);
+    return YT_OK;
+}
+
 ORACLE_EXPORT
 int yto_decode_string_column(const YtColumn* col, int64_t row_count,
                              char* out_blob, int64_t blob_cap,

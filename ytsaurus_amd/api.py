@@ -23,7 +23,8 @@ __all__ = [
     "encode_int64", "encode_double", "encode_bool", "encode_string", "encode_string_raw",
     "oracle_decode_strings", "encode_versioned_int64", "encode_versioned_double", "oracle_versioned_read", "gpu_versioned_read", "gpu_versioned_scan_chunk", "gpu_versioned_scan_table", "ScanChunk", "VersionedColumn",
     "Chunk", "oracle_execute",
-    "oracle_partial", "oracle_merge",
+    "oracle_partial", "oracle_merge", "oracle_partial_mk", "oracle_merge_mk",
+    "gpu_key_ranges", "gpu_partial_mk", "gpu_merge_mk",
     "gpu_available", "gpu_execute", "gpu_partial", "gpu_merge",
     "rows_from_rowset", "sort_rows", "make_rowset",
 ]
@@ -690,6 +691,96 @@ def oracle_partial(plan, chunk, nparts, nthreads=1):
                                        states, cap, counts, nthreads, err, 256)
     _check(rc, err)
     return states, [counts[i] for i in range(nparts)]
+
+
+def oracle_partial_mk(plan, chunk, nparts, key_ranges, nthreads=1):
+    """multi-key bottom query: key_ranges = (zzmin list, zzmax list) reduced
+    across ranks (yt_gpu_key_ranges / oracle equivalents)."""
+    ch = chunk.c_host()
+    cap = chunk.row_count + 16
+    states = (YtStateRow * cap)()
+    counts = (C.c_int64 * nparts)()
+    kzmin = (C.c_uint64 * len(plan.keys))(*[int(v) for v in key_ranges[0]])
+    kzmax = (C.c_uint64 * len(plan.keys))(*[int(v) for v in key_ranges[1]])
+    err = C.create_string_buffer(256)
+    rc = _abi.oracle_lib().yto_partial_mk(C.byref(plan.c), C.byref(ch), nparts,
+                                          kzmin, kzmax,
+                                          states, cap, counts, nthreads, err, 256)
+    _check(rc, err)
+    return states, [counts[i] for i in range(nparts)]
+
+
+def oracle_merge_mk(plan, states_list, key_ranges, col_types=None):
+    total = sum(n for _, n in states_list)
+    arr = (YtStateRow * max(total, 1))()
+    at = 0
+    for st, n in states_list:
+        for i in range(n):
+            arr[at] = st[i]
+            at += 1
+    kzmin = (C.c_uint64 * len(plan.keys))(*[int(v) for v in key_ranges[0]])
+    kzmax = (C.c_uint64 * len(plan.keys))(*[int(v) for v in key_ranges[1]])
+    ct = None
+    if col_types is not None:
+        lst = list(col_types)[:8] + [VT_INT64] * max(0, 8 - len(col_types))
+        ct = (C.c_uint8 * 8)(*lst)
+    rs = _mk_rowset(max(total + 1024, 1 << 14))
+    err = C.create_string_buffer(256)
+    rc = _abi.oracle_lib().yto_merge_mk(C.byref(plan.c), arr, C.c_int64(total),
+                                        ct, kzmin, kzmax,
+                                        C.byref(rs), err, 256)
+    _check(rc, err)
+    return rows_from_rowset(rs)
+
+
+def gpu_key_ranges(plan, device_chunk):
+    k = len(plan.keys)
+    kzmin = (C.c_uint64 * k)()
+    kzmax = (C.c_uint64 * k)()
+    err = C.create_string_buffer(256)
+    rc = _abi.gpu_lib().yt_gpu_key_ranges(C.byref(plan.c), C.byref(device_chunk),
+                                          kzmin, kzmax, C.c_uint64(0), err, 256)
+    _check(rc, err)
+    return [int(v) for v in kzmin], [int(v) for v in kzmax]
+
+
+def gpu_partial_mk(plan, device_chunk, nparts, states_dev_ptr, capacity_rows,
+                   key_ranges, max_groups_hint=0, stream=0):
+    opts = YtExecOptions(max_groups_hint=max_groups_hint, stream=stream)
+    counts = (C.c_int64 * nparts)()
+    st = YtStatistics()
+    k = len(plan.keys)
+    kzmin = (C.c_uint64 * k)(*[int(v) for v in key_ranges[0]])
+    kzmax = (C.c_uint64 * k)(*[int(v) for v in key_ranges[1]])
+    err = C.create_string_buffer(512)
+    rc = _abi.gpu_lib().yt_gpu_query_partial_mk(
+        C.byref(plan.c), C.byref(device_chunk), C.byref(opts),
+        C.c_int32(nparts), C.c_void_p(states_dev_ptr),
+        C.c_int64(capacity_rows), counts, kzmin, kzmax,
+        C.byref(st), err, 512)
+    _check(rc, err)
+    return [counts[i] for i in range(nparts)], st
+
+
+def gpu_merge_mk(plan, states_dev_ptr, n_states, key_ranges, col_types=None,
+                 max_groups_hint=0, stream=0, out_capacity=None):
+    opts = YtExecOptions(max_groups_hint=max_groups_hint, stream=stream)
+    cap = out_capacity or max(n_states + 1024, 1 << 16)
+    rs = _mk_rowset(cap)
+    st = YtStatistics()
+    k = len(plan.keys)
+    kzmin = (C.c_uint64 * k)(*[int(v) for v in key_ranges[0]])
+    kzmax = (C.c_uint64 * k)(*[int(v) for v in key_ranges[1]])
+    ct = None
+    if col_types is not None:
+        lst = list(col_types)[:8] + [VT_INT64] * max(0, 8 - len(col_types))
+        ct = (C.c_uint8 * 8)(*lst)
+    err = C.create_string_buffer(512)
+    rc = _abi.gpu_lib().yt_gpu_merge_states_mk(
+        C.byref(plan.c), C.c_void_p(states_dev_ptr), C.c_int64(n_states),
+        ct, kzmin, kzmax, C.byref(opts), C.byref(rs), C.byref(st), err, 512)
+    _check(rc, err)
+    return rows_from_rowset(rs), st
 
 
 def oracle_merge(plan, states_list):

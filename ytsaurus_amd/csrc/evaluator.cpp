@@ -3314,14 +3314,25 @@ fail:
     return rc;
 }
 
-extern "C" int yt_gpu_query_partial(
+static int query_partial_impl(
     const YtPlan* plan, const YtChunk* chunk, const YtExecOptions* options,
     int32_t partition_count, void* states_device, int64_t capacity_rows,
-    int64_t* part_counts, YtStatistics* stats, char* errbuf, size_t errlen)
+    int64_t* part_counts,
+    const uint64_t* key_zzmin, const uint64_t* key_zzmax,
+    YtStatistics* stats, char* errbuf, size_t errlen)
 {
     int rc = yt_gpu_available(errbuf, errlen);
     if (rc != YT_OK) return rc;
-    if (plan->key_count != 1) { set_err(errbuf, errlen, "partial: need exactly 1 key"); return YT_ERR_UNSUPPORTED; }
+    if (plan->key_count < 1 || plan->key_count > kMaxPackKeys) {
+        set_err(errbuf, errlen, "partial: 1..4 group keys");
+        return YT_ERR_UNSUPPORTED;
+    }
+    if (plan->key_count > 1 && (!key_zzmin || !key_zzmax)) {
+        set_err(errbuf, errlen,
+                "partial: multi-key sharding needs the common key ranges "
+                "(yt_gpu_key_ranges + cross-rank min/max reduce)");
+        return YT_ERR_UNSUPPORTED;
+    }
     if (partition_count < 1 || partition_count > 64) { set_err(errbuf, errlen, "partial: 1..64 partitions"); return YT_ERR_UNSUPPORTED; }
     if (plan->join) { set_err(errbuf, errlen, "partial: join at the bottom query not this round"); return YT_ERR_UNSUPPORTED; }
     int sum_slot = -1;
@@ -3358,6 +3369,20 @@ extern "C" int yt_gpu_query_partial(
     if (chunk->row_count == 0 || R.nsegs == 0) {
         for (int p = 0; p < partition_count; p++) part_counts[p] = 0;
         return YT_OK;
+    }
+    if (dp.kp_count && key_zzmin && key_zzmax) {
+        /* pack the composite with the COMMON cross-rank ranges so every
+         * rank's key_bits agree (partition hash + merge identity) */
+        for (int i = 0; i < dp.kp_count; i++) {
+            int c = dp.kp_col[i];
+            R.col_zzmin[c] = key_zzmin[i];
+            R.col_zzmax[c] = key_zzmax[i];
+        }
+        rc = pack_group_key(&dp, &R, errbuf, errlen);
+        if (rc) return rc;
+    } else if (dp.kp_count) {
+        rc = pack_group_key(&dp, &R, errbuf, errlen);
+        if (rc) return rc;
     }
     rc = setup_table(&R, plan->agg_count, options->max_groups_hint,
                      options->group_row_limit, errbuf, errlen);
@@ -3458,15 +3483,80 @@ fail:
     return rc;
 }
 
-extern "C" int yt_gpu_merge_states(
+extern "C" int yt_gpu_query_partial(
+    const YtPlan* plan, const YtChunk* chunk, const YtExecOptions* options,
+    int32_t partition_count, void* states_device, int64_t capacity_rows,
+    int64_t* part_counts, YtStatistics* stats, char* errbuf, size_t errlen)
+{
+    if (plan && plan->key_count != 1) {
+        set_err(errbuf, errlen, "partial: need exactly 1 key (use _mk)");
+        return YT_ERR_UNSUPPORTED;
+    }
+    return query_partial_impl(plan, chunk, options, partition_count,
+                              states_device, capacity_rows, part_counts,
+                              nullptr, nullptr, stats, errbuf, errlen);
+}
+
+extern "C" int yt_gpu_query_partial_mk(
+    const YtPlan* plan, const YtChunk* chunk, const YtExecOptions* options,
+    int32_t partition_count, void* states_device, int64_t capacity_rows,
+    int64_t* part_counts,
+    const uint64_t* key_zzmin, const uint64_t* key_zzmax,
+    YtStatistics* stats, char* errbuf, size_t errlen)
+{
+    return query_partial_impl(plan, chunk, options, partition_count,
+                              states_device, capacity_rows, part_counts,
+                              key_zzmin, key_zzmax, stats, errbuf, errlen);
+}
+
+/* per-rank key-column zigzag ranges for the cross-rank reduce (meta-only
+ * bound: conservative is fine, the reduce of bounds is a bound) */
+extern "C" int yt_gpu_key_ranges(
+    const YtPlan* plan, const YtChunk* chunk,
+    uint64_t* key_zzmin, uint64_t* key_zzmax,
+    uint64_t stream, char* errbuf, size_t errlen)
+{
+    int rc = yt_gpu_available(errbuf, errlen);
+    if (rc != YT_OK) return rc;
+    if (plan->key_count < 1 || plan->key_count > kMaxPackKeys) {
+        set_err(errbuf, errlen, "key_ranges: 1..4 group keys");
+        return YT_ERR_UNSUPPORTED;
+    }
+    DeviceRun R;
+    R.stream = (hipStream_t)(uintptr_t)stream;
+    unsigned maxw = 0;
+    int cl = 0;
+    rc = setup_chunk(chunk, &R, &maxw, 0, &cl, errbuf, errlen);
+    if (rc) return rc;
+    for (int i = 0; i < plan->key_count; i++) {
+        int c;
+        if (!expr_is_col(plan->keys[i], &c) || c >= chunk->column_count) {
+            set_err(errbuf, errlen, "key_ranges: plain key columns only");
+            return YT_ERR_UNSUPPORTED;
+        }
+        key_zzmin[i] = R.col_zzmin[c];
+        key_zzmax[i] = R.col_zzmax[c];
+    }
+    return YT_OK;
+}
+
+static int merge_states_impl(
     const YtPlan* plan, const void* states_device, int64_t state_row_count,
     const uint8_t* col_types,
+    const uint64_t* key_zzmin, const uint64_t* key_zzmax,
     const YtExecOptions* options, YtRowset* output, YtStatistics* stats,
     char* errbuf, size_t errlen)
 {
     int rc = yt_gpu_available(errbuf, errlen);
     if (rc != YT_OK) return rc;
-    if (plan->key_count != 1) { set_err(errbuf, errlen, "merge: need exactly 1 key"); return YT_ERR_UNSUPPORTED; }
+    if (plan->key_count < 1 || plan->key_count > kMaxPackKeys) {
+        set_err(errbuf, errlen, "merge: 1..4 group keys");
+        return YT_ERR_UNSUPPORTED;
+    }
+    if (plan->key_count > 1 && (!key_zzmin || !key_zzmax)) {
+        set_err(errbuf, errlen, "merge: multi-key needs the common key ranges");
+        return YT_ERR_UNSUPPORTED;
+    }
     output->totals_row = 0;
     int sum_slot = -1;
     for (int a = 0; a < plan->agg_count; a++) {
@@ -3529,8 +3619,36 @@ extern "C" int yt_gpu_merge_states(
         fake.row_count = 0;
         fake.column_count = kMaxCols;
         fake.columns = cols.data();
+        /* multi-key: rebuild the composite packing from the COMMON ranges
+         * (identical math to every rank's partial — pack_group_key) so
+         * emit unpacks the key components */
+        DevPlan dpk;
+        DevPlan* dpk_p = nullptr;
+        if (plan->key_count > 1) {
+            memset(&dpk, 0, sizeof(dpk));
+            dpk.ncols = kMaxCols;
+            for (int c2 = 0; c2 < kMaxCols; c2++)
+                dpk.col_types[c2] = col_types ? col_types[c2] : YT_VT_INT64;
+            dpk.kp_count = plan->key_count;
+            DeviceRun R2;   /* host-side ranges only */
+            for (int i = 0; i < plan->key_count; i++) {
+                int c2;
+                if (!expr_is_col(plan->keys[i], &c2) || c2 >= kMaxCols) {
+                    set_err(errbuf, errlen, "merge: plain key columns only");
+                    g_pool.put(hgroups);
+                    return YT_ERR_UNSUPPORTED;
+                }
+                dpk.kp_col[i] = c2;
+                dpk.kp_signed[i] = dpk.col_types[c2] == YT_VT_INT64;
+                R2.col_zzmin[c2] = key_zzmin[i];
+                R2.col_zzmax[c2] = key_zzmax[i];
+            }
+            rc = pack_group_key(&dpk, &R2, errbuf, errlen);
+            if (rc) { g_pool.put(hgroups); return rc; }
+            dpk_p = &dpk;
+        }
         int out_limited = 0;
-        rc = emit_rows(plan, &fake, nullptr, hgroups, ngroups, th, 0,
+        rc = emit_rows(plan, &fake, dpk_p, hgroups, ngroups, th, 0,
                        nullptr, 0, options->output_row_limit, &out_limited,
                        output, errbuf, errlen);
         g_pool.put(hgroups);
@@ -3549,4 +3667,31 @@ extern "C" int yt_gpu_merge_states(
     return YT_OK;
 fail:
     return rc;
+}
+
+extern "C" int yt_gpu_merge_states(
+    const YtPlan* plan, const void* states_device, int64_t state_row_count,
+    const uint8_t* col_types,
+    const YtExecOptions* options, YtRowset* output, YtStatistics* stats,
+    char* errbuf, size_t errlen)
+{
+    if (plan && plan->key_count != 1) {
+        set_err(errbuf, errlen, "merge: need exactly 1 key (use _mk)");
+        return YT_ERR_UNSUPPORTED;
+    }
+    return merge_states_impl(plan, states_device, state_row_count, col_types,
+                             nullptr, nullptr, options, output, stats,
+                             errbuf, errlen);
+}
+
+extern "C" int yt_gpu_merge_states_mk(
+    const YtPlan* plan, const void* states_device, int64_t state_row_count,
+    const uint8_t* col_types,
+    const uint64_t* key_zzmin, const uint64_t* key_zzmax,
+    const YtExecOptions* options, YtRowset* output, YtStatistics* stats,
+    char* errbuf, size_t errlen)
+{
+    return merge_states_impl(plan, states_device, state_row_count, col_types,
+                             key_zzmin, key_zzmax, options, output, stats,
+                             errbuf, errlen);
 }
