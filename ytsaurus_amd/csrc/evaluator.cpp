@@ -1000,8 +1000,12 @@ static int run_partitioned(const YtPlan* plan, const YtChunk* chunk,
         }
         /* measured ~1ms slower than XCD sub-streams (phase B loses its
          * paired-load ILP across 2048 short streams); keep as a knob */
-        pp.wg_streams = (!pp.reorder && !pp.has_val_nulls
-                         && getenv("YTQL_WGSTREAMS")) ? 1 : 0;
+        {
+            const char* wgm = getenv("YTQL_WGSTREAMS");   /* 1 two-pass, 2 single-pass */
+            pp.wg_streams = (!pp.reorder && !pp.has_val_nulls && wgm)
+                ? atoi(wgm) : 0;
+            if (pp.wg_streams == 2 && !pp.packed_mode) pp.wg_streams = 1;
+        }
         /* direct-span mode: when the key zigzag span is small, partition by
          * key RANGE and index phase B's per-bucket array directly (the
          * headline config — 1M distinct keys — spans 21 bits). Needs the
@@ -1037,10 +1041,13 @@ static int run_partitioned(const YtPlan* plan, const YtChunk* chunk,
     }
     /* LDS: per-tile bucket histograms + the staged key column */
     auto lds_for = [&](int tr) {
-        /* reorder mode: record scratch instead of the staged key column */
-        return pp.reorder
-            ? (size_t)tr * 8 + (size_t)4 * kNB * 4 + 64
-            : (size_t)4 * kNB * 4 + ((size_t)tr * w / 64 + 2) * 8 + 256;
+        /* reorder mode: record scratch instead of the staged key column;
+         * wg single-pass: histograms only (keys read once, from global) */
+        if (pp.reorder)
+            return (size_t)tr * 8 + (size_t)4 * kNB * 4 + 64;
+        if (pp.wg_streams == 2)
+            return (size_t)4 * kNB * 4 + 256;
+        return (size_t)4 * kNB * 4 + ((size_t)tr * w / 64 + 2) * 8 + 256;
     };
     size_t lds = lds_for(tile_rows);
     while (!tile_forced && lds > 80 * 1024 && tile_rows > 4096) { tile_rows >>= 1; lds = lds_for(tile_rows); }

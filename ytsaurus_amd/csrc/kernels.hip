@@ -1799,7 +1799,7 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
             ? (const uint8_t*)sgk.blob + ek.off_bitmap_bytes : nullptr;
         const uint64_t* kwords = sgk.blob + ek.off_values_words;
         const int64_t kW0 = ((uint64_t)t0 * kwd) >> 6;
-        if (!pp.reorder) {
+        if (!pp.reorder && pp.wg_streams != 2) {
             /* reorder mode reads keys straight from global both passes —
              * the per-tile key window is L2-resident on the re-read, and
              * the freed LDS buys the record scratch at 2 WGs/CU */
@@ -1830,8 +1830,10 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
         __syncthreads();
 
         const int R = (pp.tile_rows + 255) / 256;
+        const bool single_pass = pp.wg_streams == 2;   /* claims straight off
+            the per-WG running bases: no count pass, no reserve */
         /* pass 1: COUNT per bucket (no returns, no per-row state) */
-        if (pp.store_mode != 6)
+        if (pp.store_mode != 6 && !single_pass)
         #pragma unroll 8
         for (int i = 0; i < R; i++) {
             int64_t j = t0 + (int64_t)i * 256 + tid;
@@ -1859,7 +1861,9 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
          * then reuse hist/nhist as the pass-2 claim counters. Aligned mode
          * reserves whole 8-record groups so no HBM line is split between
          * two reservations (pad fill below). */
-        if (pp.wg_streams) {
+        if (single_pass) {
+            /* no reserve: pass 2 claims ngbase[b]++ per record */
+        } else if (pp.wg_streams) {
             /* local append into this WG's own region: no global atomics */
             for (int i = tid; i < kNB; i += 256) {
                 unsigned c = hist[i];
@@ -1934,7 +1938,7 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
                 uint64_t kzzfull = 0;
                 uint64_t key = 0;
                 if (!key_null) {
-                    kzzfull = kmin + (pp.reorder
+                    kzzfull = kmin + ((pp.reorder || pp.wg_streams == 2)
                         ? bp_gl(kwords, kmask, kwd, j)
                         : (bp_get_win(klds, kwd, j, kW0) & kmask));
                     key = (uint64_t)zz_dec(kzzfull);
@@ -1971,13 +1975,25 @@ k_scan_partition(PartParams pp, const DevSeg* segs, const SegEx* segex,
                     unsigned off = atomicAdd(&ngbase[b], 1u);
                     rscratch[nhist[b] + off] = rec;
                 } else if (pp.packed_mode) {
-                    unsigned off = atomicAdd(&hist[b], 1u);
+                    unsigned off;
+                    unsigned base;
+                    if (single_pass) {
+                        off = atomicAdd(&ngbase[b], 1u);
+                        base = 0;
+                        if ((int64_t)off >= pp.bucket_stride) {
+                            th->overflow = 1;
+                            continue;
+                        }
+                    } else {
+                        off = atomicAdd(&hist[b], 1u);
+                        base = gbase[b];
+                    }
                     uint64_t rec = kzzfull - pp.gmin_k;
                     if (has_val) {
                         uint64_t vzz = (vmin + bp_gl(vwords, vmask, vwd, j)) - pp.gmin_v;
                         rec |= vzz << pp.bits_k;
                     }
-                    uint64_t* dst = &recs8[sb * pp.bucket_stride + gbase[b] + off];
+                    uint64_t* dst = &recs8[sb * pp.bucket_stride + base + off];
                     if (pp.store_mode == 1) {
                         /* write-through (sc1): drops the line from L2 — no
                          * partial-line RMW fill on eviction */
