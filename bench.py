@@ -405,13 +405,19 @@ def main():
     scan_ms_avg = scan_ms_total / max(scan_launches, 1)
     achieved_gbs = (enc_bytes / 1e9) / (scan_ms_avg / 1e3) if scan_ms_avg else None
     peak_gbs = 8000.0   # HBM3E spec peak, MI355X_MICROARCH.md
+    traffic = args.traffic_bytes or None
+    if traffic is None and args.workload == "groupby" and n == 10**9:
+        # measured once at this exact config/state via separate rocprofv3
+        # --pmc FETCH_SIZE / WRITE_SIZE passes (profiles/r02/pmc_1e9.json):
+        # k_scan_partition fetch 8.29 GB (x2-corrected) + write 47.49 GB
+        traffic = 55.78e9
     roofline = {
         "bound": "hbm",
         "achieved": achieved_gbs,
         "peak": peak_gbs,
         "unit": "GB/s",
         "frac": (achieved_gbs / peak_gbs) if achieved_gbs else None,
-        "traffic": args.traffic_bytes or None,
+        "traffic": traffic,
     }
 
     cpu_baseline = None
