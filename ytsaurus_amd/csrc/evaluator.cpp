@@ -113,11 +113,13 @@ hipError_t ytql_launch_strgrp_accum(const StrGroupParams*, const DevSeg*, const 
                                     const int64_t*, unsigned long long*, TableHdr*,
                                     hipStream_t);
 hipError_t ytql_launch_strgrp_hash(const DevSeg*, const SegEx*, int, int,
-                                   const int64_t*, uint64_t*, int64_t, hipStream_t);
+                                   const int64_t*, uint64_t*, uint64_t*,
+                                   ulonglong2*, int64_t, hipStream_t);
 hipError_t ytql_launch_strgrp_merge(const DevSeg*, const SegEx*, int, int,
                                     const int64_t*, const unsigned long long*,
-                                    const uint64_t*, StrSlot*, uint64_t, int,
-                                    TableHdr*, int64_t, hipStream_t);
+                                    const uint64_t*, const uint64_t*,
+                                    const ulonglong2*, StrSlot*, uint64_t, int,
+                                    int, TableHdr*, int64_t, hipStream_t);
 hipError_t ytql_launch_strgrp_compact(const DevSeg*, const SegEx*, int,
                                       const StrSlot*, uint64_t, OutStrGroup*,
                                       unsigned long long*, char*, unsigned long long*,
@@ -2547,6 +2549,15 @@ static int run_string_group(const YtPlan* plan, const YtChunk* chunk,
         HIP_CHECK(pool_alloc(&d_acc, sizeof(uint64_t) * 2 * (total_dict ? total_dict : 1)));
         HIP_CHECK(hipMemsetAsync(d_acc, 0, sizeof(uint64_t) * 2 * (total_dict ? total_dict : 1), R.stream));
         HIP_CHECK(pool_alloc(&d_hashes, sizeof(uint64_t) * (total_dict ? total_dict : 1)));
+        /* per-entry identity + prefix arrays: filled by the hash kernel so
+         * the merge's hot path reads them as a stream (YTQL_STRMERGE=0
+         * falls back to the round-1 dict-read probe for A/B) */
+        uint64_t* d_idents = nullptr;
+        ulonglong2* d_pfxs = nullptr;
+        HIP_CHECK(pool_alloc(&d_idents, sizeof(uint64_t) * (total_dict ? total_dict : 1)));
+        HIP_CHECK(pool_alloc(&d_pfxs, sizeof(ulonglong2) * (total_dict ? total_dict : 1)));
+        const char* sm_env = getenv("YTQL_STRMERGE");
+        int merge_fast = sm_env ? atoi(sm_env) : 1;
         /* Table sized for the worst case (every dict entry distinct).
          * Measured: an 8×-smaller table holding just the ~6 M live keys is
          * NOT faster — the merge's scattered atomics then fight over hot
@@ -2568,12 +2579,15 @@ static int run_string_group(const YtPlan* plan, const YtChunk* chunk,
         HIP_CHECK(ytql_launch_strgrp_accum(&sp, R.d_segs, R.d_segex, d_accbase,
                                            d_acc, R.d_th, R.stream));
         HIP_CHECK(ytql_launch_strgrp_hash(R.d_segs, R.d_segex, koff, knseg,
-                                          d_accbase, d_hashes, total_dict, R.stream));
+                                          d_accbase, d_hashes, d_idents, d_pfxs,
+                                          total_dict, R.stream));
         for (;;) {
             HIP_CHECK(ytql_launch_strgrp_merge(R.d_segs, R.d_segex, koff, knseg,
                                                d_accbase, d_acc, d_hashes,
+                                               d_idents, d_pfxs,
                                                d_slots, nslots, val_is_double,
-                                               R.d_th, total_dict, R.stream));
+                                               merge_fast, R.d_th, total_dict,
+                                               R.stream));
             HIP_CHECK(hipMemcpy(&th, R.d_th, sizeof(th), hipMemcpyDeviceToHost));
             if (th.overflow != 1 || nslots >= nslots_cap) break;
             g_pool.put(d_slots);
@@ -2630,6 +2644,7 @@ static int run_string_group(const YtPlan* plan, const YtChunk* chunk,
         HIP_CHECK(hipMemcpy(&th, R.d_th, sizeof(th), hipMemcpyDeviceToHost));
         if (th.overflow == 1) {
             g_pool.put(d_accbase); g_pool.put(d_acc); g_pool.put(d_hashes);
+            g_pool.put(d_idents); g_pool.put(d_pfxs);
             g_pool.put(d_slots); g_pool.put(d_out); g_pool.put(d_pool);
             g_pool.put(d_ctr); g_pool.put(hgroups);
             set_err(errbuf, errlen, "string merge table/pool overflow");
@@ -2817,6 +2832,7 @@ emitted:
             stats->execute_time_ms = now_ms() - tw0;
         }
         g_pool.put(d_accbase); g_pool.put(d_acc); g_pool.put(d_hashes);
+        g_pool.put(d_idents); g_pool.put(d_pfxs);
         g_pool.put(d_slots); g_pool.put(d_out); g_pool.put(d_pool);
         g_pool.put(d_ctr); g_pool.put(hgroups);
     }

@@ -2485,10 +2485,15 @@ k_strgrp_accum(StrGroupParams sp, const DevSeg* segs, const SegEx* segex,
     }
 }
 
-/* S2: FNV-1a hash of every dictionary entry */
+/* S2: FNV-1a hash of every dictionary entry. Also materializes the per-entry
+ * slot identity (hash48|len) and 16-byte prefix that the merge compares
+ * against StrSlot.ident/pfx — this kernel already streams every dictionary
+ * byte, so capturing them here lets the merge's hot path touch ONLY its
+ * streaming arrays and the slot line (no dict reads, no binary search). */
 __global__ void k_strgrp_hash(const DevSeg* segs, const SegEx* segex,
                               int key_seg_off, int nsegs,
-                              const int64_t* acc_base, uint64_t* hashes)
+                              const int64_t* acc_base, uint64_t* hashes,
+                              uint64_t* idents, ulonglong2* pfxs)
 {
     int64_t total = acc_base[nsegs];
     for (int64_t g = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -2508,6 +2513,13 @@ __global__ void k_strgrp_hash(const DevSeg* segs, const SegEx* segex,
             h = (h ^ (uint8_t)p[k]) * 0x100000001B3ULL;
         }
         hashes[g] = h;
+        idents[g] = ((h >> 16) << 16) | (uint64_t)(len & 0xFFFF);
+        ulonglong2 pf;
+        pf.x = 0; pf.y = 0;
+        uint32_t npfx = len < 16 ? len : 16;
+        for (uint32_t k = 0; k < npfx; k++)
+            ((char*)&pf)[k] = p[k];
+        pfxs[g] = pf;
     }
 }
 
@@ -2519,8 +2531,9 @@ __global__ void k_strgrp_merge(const DevSeg* segs, const SegEx* segex,
                                const int64_t* acc_base,
                                const unsigned long long* acc,
                                const uint64_t* hashes,
+                               const uint64_t* idents, const ulonglong2* pfxs,
                                StrSlot* slots, uint64_t nslots,
-                               int val_is_double, TableHdr* th)
+                               int val_is_double, int fast, TableHdr* th)
 {
     int64_t total = acc_base[nsegs];
     uint64_t mask = nslots - 1;
@@ -2528,6 +2541,112 @@ __global__ void k_strgrp_merge(const DevSeg* segs, const SegEx* segex,
          g < total; g += (int64_t)gridDim.x * blockDim.x) {
         unsigned long long cn = acc[2 * g];
         if (cn == 0) continue;
+        uint64_t h = hashes[g];
+        StrSlot* slot = nullptr;
+        if (fast) {
+            /* hot path: everything this entry needs is STREAMED (idents,
+             * pfxs, hashes, acc are read in entry order) — the only random
+             * memory is the slot line itself, read as ONE 32-byte load
+             * (rep|ident|pfx share the line's first half). The segment
+             * binary search and the dictionary-byte reads are resolved
+             * LAZILY: only the ~1/dup-factor claiming entries (and the rare
+             * hash48|len collisions) pay them. */
+            const uint64_t my_ident = idents[g];
+            const uint32_t my_len = (uint32_t)(my_ident & 0xFFFF);
+            const ulonglong2 pf = pfxs[g];
+            int lo = -1;
+            unsigned long long rep = 0;
+            uint64_t sidx = mix64(h) & mask;
+            uint64_t max_probe = mask < 8192 ? mask : 8192;
+            for (uint64_t it = 0; it <= max_probe; it++) {
+                StrSlot* cand = &slots[sidx];
+                ulonglong4 v = *(const ulonglong4*)cand;
+                unsigned long long cur = v.x;
+                uint64_t oident = v.y;
+                if (cur == 0ULL) {
+                    if (lo < 0) {
+                        int a = 0, b = nsegs;
+                        while (a + 1 < b) {
+                            int mid = (a + b) / 2;
+                            if (acc_base[mid] <= g) a = mid;
+                            else b = mid;
+                        }
+                        lo = a;
+                        rep = ((unsigned long long)(lo + 1) << 32)
+                            | (unsigned long long)(g - acc_base[lo] + 1);
+                    }
+                    cur = atomicCAS(&cand->rep, 0ULL, rep);
+                    if (cur == 0ULL) {
+                        __hip_atomic_store(&cand->pfx[0], pf.x,
+                                           __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+                        __hip_atomic_store(&cand->pfx[1], pf.y,
+                                           __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+                        __hip_atomic_store(&cand->ident, my_ident,
+                                           __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+                        unsigned long long t = atomicAdd(&th->ngroups, 1ULL);
+                        if (th->group_limit > 0 && (int64_t)t >= th->group_limit)
+                            th->overflow = 2;
+                        slot = cand;
+                        break;
+                    }
+                    /* lost the race: cand belongs to someone else now and
+                     * its identity may be unpublished — treat as oident 0 */
+                    oident = 0;
+                }
+                if (oident != 0 && oident != my_ident) {
+                    sidx = (sidx + 1) & mask;
+                    continue;
+                }
+                if (oident == my_ident && my_len <= 16
+                    && v.z == pf.x && v.w == pf.y) { slot = cand; break; }
+                /* unpublished identity, long key, or prefix mismatch on an
+                 * ident match: decide exactly from the owner's dictionary
+                 * entry (rare) */
+                {
+                    int oseg = (int)(cur >> 32) - 1;
+                    int64_t oid = (int64_t)(cur & 0xFFFFFFFFULL);
+                    uint64_t oh = hashes[acc_base[oseg] + oid - 1];
+                    if (oh == h) {
+                        if (lo < 0) {
+                            int a = 0, b = nsegs;
+                            while (a + 1 < b) {
+                                int mid = (a + b) / 2;
+                                if (acc_base[mid] <= g) a = mid;
+                                else b = mid;
+                            }
+                            lo = a;
+                            rep = ((unsigned long long)(lo + 1) << 32)
+                                | (unsigned long long)(g - acc_base[lo] + 1);
+                        }
+                        uint32_t my_len2;
+                        const char* my_p = dict_entry(segs[key_seg_off + lo],
+                                                      segex[key_seg_off + lo],
+                                                      g - acc_base[lo], &my_len2);
+                        uint32_t olen;
+                        const char* op = dict_entry(segs[key_seg_off + oseg],
+                                                    segex[key_seg_off + oseg],
+                                                    oid - 1, &olen);
+                        if (olen == my_len2) {
+                            uint32_t k = 0;
+                            while (k < my_len2 && op[k] == my_p[k]) k++;
+                            if (k == my_len2) { slot = cand; break; }
+                        }
+                    }
+                }
+                sidx = (sidx + 1) & mask;
+            }
+            if (!slot) { th->overflow = 1; continue; }
+            atomicAdd((unsigned long long*)&slot->cnt, cn);
+            unsigned long long nn = cn >> 32;
+            if (nn) {
+                if (val_is_double)
+                    atomicAdd((double*)&slot->sum_bits,
+                              __longlong_as_double(acc[2 * g + 1]));
+                else
+                    atomicAdd((unsigned long long*)&slot->sum_bits, acc[2 * g + 1]);
+            }
+            continue;
+        }
         int lo = 0, hi = nsegs;
         while (lo + 1 < hi) {
             int mid = (lo + hi) / 2;
@@ -2539,7 +2658,6 @@ __global__ void k_strgrp_merge(const DevSeg* segs, const SegEx* segex,
         int64_t my_id = g - acc_base[lo] + 1;
         uint32_t my_len;
         const char* my_p = dict_entry(my_s, my_e, my_id - 1, &my_len);
-        uint64_t h = hashes[g];
         unsigned long long rep = ((unsigned long long)(lo + 1) << 32)
                                | (unsigned long long)my_id;
         /* in-slot identity (see common.h StrSlot): hash48|len + 16-byte
@@ -2555,7 +2673,6 @@ __global__ void k_strgrp_merge(const DevSeg* segs, const SegEx* segex,
         }
 
         uint64_t sidx = mix64(h) & mask;
-        StrSlot* slot = nullptr;
         /* probe chains this long only happen when the table is (nearly)
          * full — give up early so a too-small estimated table overflows in
          * bounded time and the host retries at 4x (never a wrong result:
@@ -3235,12 +3352,14 @@ hipError_t ytql_launch_strgrp_accum(const StrGroupParams* sp, const DevSeg* segs
 hipError_t ytql_launch_strgrp_hash(const DevSeg* segs, const SegEx* segex,
                                    int key_seg_off, int nsegs,
                                    const int64_t* acc_base, uint64_t* hashes,
+                                   uint64_t* idents, ulonglong2* pfxs,
                                    int64_t total, hipStream_t st)
 {
     int64_t want = (total + 255) / 256;
     int grid = (int)(want > 4096 ? 4096 : (want ? want : 1));
     hipLaunchKernelGGL(k_strgrp_hash, dim3(grid), dim3(256), 0, st,
-                       segs, segex, key_seg_off, nsegs, acc_base, hashes);
+                       segs, segex, key_seg_off, nsegs, acc_base, hashes,
+                       idents, pfxs);
     return hipGetLastError();
 }
 
@@ -3249,15 +3368,16 @@ hipError_t ytql_launch_strgrp_merge(const DevSeg* segs, const SegEx* segex,
                                     const int64_t* acc_base,
                                     const unsigned long long* acc,
                                     const uint64_t* hashes,
+                                    const uint64_t* idents, const ulonglong2* pfxs,
                                     StrSlot* slots, uint64_t nslots,
-                                    int val_is_double, TableHdr* th,
+                                    int val_is_double, int fast, TableHdr* th,
                                     int64_t total, hipStream_t st)
 {
     int64_t want = (total + 255) / 256;
     int grid = (int)(want > 4096 ? 4096 : (want ? want : 1));
     hipLaunchKernelGGL(k_strgrp_merge, dim3(grid), dim3(256), 0, st,
                        segs, segex, key_seg_off, nsegs, acc_base, acc, hashes,
-                       slots, nslots, val_is_double, th);
+                       idents, pfxs, slots, nslots, val_is_double, fast, th);
     return hipGetLastError();
 }
 
