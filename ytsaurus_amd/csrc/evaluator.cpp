@@ -2571,16 +2571,20 @@ static int run_string_group(const YtPlan* plan, const YtChunk* chunk,
         HIP_CHECK(pool_alloc(&d_slots, sizeof(StrSlot) * nslots));
         HIP_CHECK(hipMemsetAsync(d_slots, 0, sizeof(StrSlot) * nslots, R.stream));
 
-        hipEvent_t e0, e1, e2;
+        hipEvent_t e0, e1, e2, ea, eh;
         HIP_CHECK(hipEventCreate(&e0));
         HIP_CHECK(hipEventCreate(&e1));
         HIP_CHECK(hipEventCreate(&e2));
+        HIP_CHECK(hipEventCreate(&ea));
+        HIP_CHECK(hipEventCreate(&eh));
         HIP_CHECK(hipEventRecord(e0, R.stream));
         HIP_CHECK(ytql_launch_strgrp_accum(&sp, R.d_segs, R.d_segex, d_accbase,
                                            d_acc, R.d_th, R.stream));
+        HIP_CHECK(hipEventRecord(ea, R.stream));
         HIP_CHECK(ytql_launch_strgrp_hash(R.d_segs, R.d_segex, koff, knseg,
                                           d_accbase, d_hashes, d_idents, d_pfxs,
                                           total_dict, R.stream));
+        HIP_CHECK(hipEventRecord(eh, R.stream));
         for (;;) {
             HIP_CHECK(ytql_launch_strgrp_merge(R.d_segs, R.d_segex, koff, knseg,
                                                d_accbase, d_acc, d_hashes,
@@ -2638,6 +2642,16 @@ static int run_string_group(const YtPlan* plan, const YtChunk* chunk,
         float ms = 0, ms_other = 0;
         HIP_CHECK(hipEventElapsedTime(&ms, e0, e1));
         HIP_CHECK(hipEventElapsedTime(&ms_other, e1, e2));
+        if (getenv("YTQL_TIMING")) {
+            float ms_acc = 0, ms_hash = 0;
+            HIP_CHECK(hipEventElapsedTime(&ms_acc, e0, ea));
+            HIP_CHECK(hipEventElapsedTime(&ms_hash, ea, eh));
+            fprintf(stderr, "[ytql timing] strgrp: accum %.2fms hash %.2fms "
+                    "merge %.2fms compact+D2H %.2fms\n",
+                    ms_acc, ms_hash, ms - ms_acc - ms_hash, ms_other);
+        }
+        (void)hipEventDestroy(ea);
+        (void)hipEventDestroy(eh);
         (void)hipEventDestroy(e0);
         (void)hipEventDestroy(e1);
         (void)hipEventDestroy(e2);

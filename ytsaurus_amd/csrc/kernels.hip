@@ -2537,6 +2537,14 @@ __global__ void k_strgrp_merge(const DevSeg* segs, const SegEx* segex,
 {
     int64_t total = acc_base[nsegs];
     uint64_t mask = nslots - 1;
+    /* fast path: claims are counted per block in LDS and folded into
+     * th->ngroups ONCE per block — a per-claim global atomicAdd on that
+     * single address serializes (~160 M/s measured, tools/probe_strcompact)
+     * and throttled the whole merge at 100 M distinct keys. The group-limit
+     * check moves to the per-block fold (same flag, coarser timing). */
+    __shared__ unsigned long long s_claims;
+    if (threadIdx.x == 0) s_claims = 0;
+    __syncthreads();
     for (int64_t g = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
          g < total; g += (int64_t)gridDim.x * blockDim.x) {
         unsigned long long cn = acc[2 * g];
@@ -2583,9 +2591,7 @@ __global__ void k_strgrp_merge(const DevSeg* segs, const SegEx* segex,
                                            __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
                         __hip_atomic_store(&cand->ident, my_ident,
                                            __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
-                        unsigned long long t = atomicAdd(&th->ngroups, 1ULL);
-                        if (th->group_limit > 0 && (int64_t)t >= th->group_limit)
-                            th->overflow = 2;
+                        atomicAdd(&s_claims, 1ULL);
                         slot = cand;
                         break;
                     }
@@ -2746,6 +2752,13 @@ __global__ void k_strgrp_merge(const DevSeg* segs, const SegEx* segex,
             else
                 atomicAdd((unsigned long long*)&slot->sum_bits, acc[2 * g + 1]);
         }
+    }
+    /* fold this block's claims (fast path; the slow path added per claim) */
+    __syncthreads();
+    if (threadIdx.x == 0 && s_claims) {
+        unsigned long long t = atomicAdd(&th->ngroups, s_claims);
+        if (th->group_limit > 0 && (int64_t)(t + s_claims) > th->group_limit)
+            th->overflow = 2;
     }
 }
 
