@@ -296,7 +296,8 @@ struct StrGroupParams {
     int32_t tiles_per_seg;
     int32_t ntiles;
     int32_t has_val_nulls;
-    int32_t pad_;
+    int32_t xcd_affine;           /* keep each segment's accumulators in one
+                                     XCD's L2 (blockIdx%8 = XCD round-robin) */
     int64_t row_count;
 };
 
@@ -346,14 +347,15 @@ struct TopkGather {
     int64_t cap_null;
 };
 
-/* compacted string group (device→host) */
+/* compacted string group (device→host). 24 B: this array is the bulk of
+ * the query's D2H at 100 M groups, so the key ref packs off|len (pool is
+ * bounded by the dict blob bytes << 2^48) and counts stay in the slot's
+ * cnt|nonnull 32-bit-halves encoding. */
 struct OutStrGroup {
-    uint64_t pool_off;
-    uint32_t len;
-    uint32_t pad_;
+    uint64_t off_len;             /* pool_off<<24 | len (len cap 16 MB =
+                                     the reference MaxStringValueLength) */
     uint64_t sum_bits;
-    uint64_t cnt;
-    uint64_t nonnull;
+    uint64_t cnt_nonnull;         /* low 32 = row count, high 32 = nonnull */
 };
 
 struct KernelTimes {

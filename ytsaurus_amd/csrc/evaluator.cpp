@@ -2533,6 +2533,10 @@ static int run_string_group(const YtPlan* plan, const YtChunk* chunk,
         sp.tile_rows = 8192;
         if (sp.tile_rows > rows0) sp.tile_rows = rows0 > 256 ? rows0 : 256;
         sp.tiles_per_seg = (rows0 + sp.tile_rows - 1) / sp.tile_rows;
+        {
+            const char* sa = getenv("YTQL_STRACCUM");
+            sp.xcd_affine = sa ? atoi(sa) : 1;
+        }
         int32_t last_rows = R.h_segs[koff + knseg - 1].row_count;
         sp.ntiles = (knseg - 1) * sp.tiles_per_seg
                   + (last_rows + sp.tile_rows - 1) / sp.tile_rows;
@@ -2692,8 +2696,8 @@ static int run_string_group(const YtPlan* plan, const YtChunk* chunk,
                     dst[0].id = 0;
                     dst[0].flags = 0;
                     dst[0].type = YT_VT_STRING;
-                    dst[0].length = g.len;
-                    dst[0].data.str = output->string_pool + g.pool_off;
+                    dst[0].length = (uint32_t)(g.off_len & 0xFFFFFF);
+                    dst[0].data.str = output->string_pool + (g.off_len >> 24);
                     for (int a = 0; a < plan->agg_count; a++) {
                         YtValue& v = dst[1 + a];
                         v.id = (uint16_t)(1 + a);
@@ -2701,8 +2705,8 @@ static int run_string_group(const YtPlan* plan, const YtChunk* chunk,
                         v.length = 0;
                         if (agg_is_sum1[a]) {
                             v.type = YT_VT_INT64;
-                            v.data.bits = g.cnt;
-                        } else if (g.nonnull == 0) {
+                            v.data.bits = g.cnt_nonnull & 0xFFFFFFFFULL;
+                        } else if ((g.cnt_nonnull >> 32) == 0) {
                             v.type = YT_VT_NULL;
                             v.data.bits = 0;
                         } else {
@@ -2772,11 +2776,11 @@ static int run_string_group(const YtPlan* plan, const YtChunk* chunk,
             uint64_t cnt, sum_bits, nonnull;
             if (gI < ngroups) {
                 const OutStrGroup& g = hgroups[gI];
-                kstr = pool.data() + g.pool_off;
-                klen = g.len;
-                cnt = g.cnt;
+                kstr = pool.data() + (g.off_len >> 24);
+                klen = (uint32_t)(g.off_len & 0xFFFFFF);
+                cnt = g.cnt_nonnull & 0xFFFFFFFFULL;
                 sum_bits = g.sum_bits;
-                nonnull = g.nonnull;
+                nonnull = g.cnt_nonnull >> 32;
             } else {
                 int side = (int)(gI - ngroups);
                 if (side == 0) continue;                /* no sentinel for strings */

@@ -2402,9 +2402,30 @@ k_strgrp_accum(StrGroupParams sp, const DevSeg* segs, const SegEx* segex,
                const int64_t* acc_base, unsigned long long* acc,
                TableHdr* th)
 {
-    for (int tile = blockIdx.x; tile < sp.ntiles; tile += gridDim.x) {
-        const int seg_idx = tile / sp.tiles_per_seg;
-        const int tile_in_seg = tile % sp.tiles_per_seg;
+    /* XCD-affine schedule: workgroups land on XCD blockIdx%8 (round-robin,
+     * MI355X_MICROARCH.md), so striding segments by 8 from blockIdx&7 keeps
+     * ALL tiles of a segment — and therefore all atomics into its ~1.3 MB
+     * accumulator region — inside one XCD's 4 MB L2 instead of ping-ponging
+     * the lines across eight of them. */
+    const bool affine = sp.xcd_affine && gridDim.x >= 64;
+    const int xcd = blockIdx.x & 7;
+    const int sub = blockIdx.x >> 3;
+    const int nsub = gridDim.x >> 3;
+    const int nwork = affine
+        ? ((sp.key_seg_cnt + 7) >> 3) * sp.tiles_per_seg   /* per-XCD upper bound */
+        : sp.ntiles;
+    for (int w = affine ? sub : blockIdx.x; w < nwork;
+         w += affine ? nsub : gridDim.x) {
+        int seg_idx, tile_in_seg;
+        if (affine) {
+            seg_idx = xcd + 8 * (w / sp.tiles_per_seg);
+            tile_in_seg = w % sp.tiles_per_seg;
+            if (seg_idx >= sp.key_seg_cnt) continue;
+            if (seg_idx * sp.tiles_per_seg + tile_in_seg >= sp.ntiles) continue;
+        } else {
+            seg_idx = w / sp.tiles_per_seg;
+            tile_in_seg = w % sp.tiles_per_seg;
+        }
         const DevSeg& sk = segs[sp.key_seg_off + seg_idx];
         const SegEx& ek = segex[sp.key_seg_off + seg_idx];
         const int64_t t0 = (int64_t)tile_in_seg * sp.tile_rows;
@@ -2860,12 +2881,9 @@ __global__ void k_strgrp_compact(const DevSeg* segs, const SegEx* segex,
             } else {
                 for (uint32_t k = 0; k < len; k++) pool[off + k] = p[k];
                 OutStrGroup& g = out[idx];
-                g.pool_off = off;
-                g.len = len;
-                g.pad_ = 0;
+                g.off_len = ((unsigned long long)off << 24) | len;
                 g.sum_bits = sl->sum_bits;
-                g.cnt = sl->cnt & 0xFFFFFFFFULL;
-                g.nonnull = sl->cnt >> 32;
+                g.cnt_nonnull = sl->cnt;
             }
         }
         __syncthreads();
