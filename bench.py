@@ -311,7 +311,7 @@ def main():
     if args.workload == "join":
         main.join_dev = main.join_chunk.c_device(torch)
     plan = make_plan()
-    hint = key_space if args.workload in ("groupby", "join") else 0
+    hint = key_space if args.workload in ("groupby", "join", "strgroup") else 0
     pool_b = (key_space * 10 + (1 << 20)) if args.workload == "strgroup" else 0
     out_cap = (args.limit + 64) if args.workload == "topk" else key_space + 4096
     out_ncols = 2 if args.workload == "topk" else 1 + len(plan.aggs)
