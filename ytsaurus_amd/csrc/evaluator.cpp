@@ -2579,7 +2579,7 @@ static int run_string_group(const YtPlan* plan, const YtChunk* chunk,
          * plan). The grow-on-overflow loop below keeps correctness for
          * adversarial cardinalities (one 4x retry reaches 2x entries). */
         const char* sx_env = getenv("YTQL_STRSLOTS_X");
-        uint64_t slots_x = sx_env ? (uint64_t)atoll(sx_env) : 2;
+        uint64_t slots_x = sx_env ? (uint64_t)atoll(sx_env) : 4;
         if (slots_x < 1) slots_x = 1;
         uint64_t nslots = options->max_groups_hint > 0
             ? next_pow2((uint64_t)options->max_groups_hint * slots_x)
