@@ -298,6 +298,22 @@ enum YtVersionedSegType {
     YT_VSEG_INT_DICT_SPARSE = 3,
     YT_VSEG_DOUBLE_DENSE = 16,
     YT_VSEG_DOUBLE_SPARSE = 18,
+    /* versioned STRING value segments (string_column_writer.cpp
+     * TVersionedStringColumnWriter::DumpSegment:325-360; the Any and
+     * Composite versioned writers share this byte layout :365-401).
+     * Value part — Direct (DumpDirectValues:205-229): [packed cumulative
+     * END offsets, diff-from-expected (null values step 0)][null bitmap]
+     * [value bytes]; Dictionary (DumpDictionaryValues:153-203):
+     * [packed ids, 0 = null, first-appearance 1-based][packed cumulative
+     * dictionary END offsets, diff-from-expected][dictionary bytes].
+     * Chosen by dictionaryByteSize < directByteSize (:338-352, size
+     * estimates :80-92). base_value carries
+     * TStringSegmentMeta.expected_length. The read entry points return
+     * bits = (byte offset within the value-segment blob) << 24 | length. */
+    YT_VSEG_STR_DIRECT_DENSE = 32,
+    YT_VSEG_STR_DICT_DENSE = 33,
+    YT_VSEG_STR_DIRECT_SPARSE = 34,
+    YT_VSEG_STR_DICT_SPARSE = 35,
 };
 enum { YT_VSEG_F_AGGREGATE = 1 };     /* aggregate bitmap present */
 
@@ -338,6 +354,17 @@ int yt_encode_versioned_double(
     const uint32_t* writes_per_row, const uint64_t* write_ts,
     const double* values, const uint8_t* value_nulls,
     const uint8_t* value_agg,
+    const uint32_t* deletes_per_row, const uint64_t* delete_ts,
+    int64_t row_count, int64_t max_rows_per_segment,
+    YtVersionedColumn* out, char* errbuf, size_t errlen);
+/* Versioned STRING column (also the byte layout of versioned Any /
+ * Composite columns, string_column_writer.cpp:365-401): value_bytes is the
+ * concatenation of every write's bytes in flattened write order;
+ * value_lens[i] is the i-th write's length (0 for null writes). */
+int yt_encode_versioned_string(
+    const uint32_t* writes_per_row, const uint64_t* write_ts,
+    const char* value_bytes, const uint32_t* value_lens,
+    const uint8_t* value_nulls, const uint8_t* value_agg,
     const uint32_t* deletes_per_row, const uint64_t* delete_ts,
     int64_t row_count, int64_t max_rows_per_segment,
     YtVersionedColumn* out, char* errbuf, size_t errlen);

@@ -1995,8 +1995,10 @@ int yto_versioned_read(const YtVersionedColumn* col, uint64_t timestamp,
 
         const char* vp = (const char*)V->data;
         const int sparse = (V->type & 2) != 0;
-        const int is_dbl = V->type >= YT_VSEG_DOUBLE_DENSE;
-        const int is_dict = !is_dbl && (V->type & 1) != 0;
+        const int is_str = V->type >= YT_VSEG_STR_DIRECT_DENSE;
+        const int is_dbl = !is_str && V->type >= YT_VSEG_DOUBLE_DENSE;
+        const int is_dict = !is_dbl && !is_str && (V->type & 1) != 0;
+        const int is_sdict = is_str && (V->type & 1) != 0;
         BitReader vindex = bitreader_init(vp);    /* dense offsets | sparse row idx */
         vp += bitreader_byte_size(&vindex);
         BitReader tsids = bitreader_init(vp);
@@ -2010,7 +2012,25 @@ int yto_versioned_read(const YtVersionedColumn* col, uint64_t timestamp,
         BitReader vids = {0};                     /* dictionary ids */
         const double* ddata = NULL;
         const uint8_t* vnull = NULL;
-        if (is_dbl) {
+        const char* sdata = NULL;                 /* string bytes / dict bytes */
+        if (is_str) {
+            /* string_column_writer.cpp DumpDirectValues/DumpDictionaryValues:
+             * direct = [END offsets diff-from-expected][null bitmap][bytes];
+             * dict = [ids 0=null][dict END offsets diff-from-expected][bytes] */
+            if (is_sdict) {
+                vids = bitreader_init(vp);
+                vp += bitreader_byte_size(&vids);
+                vvals = bitreader_init(vp);       /* dictionary offsets */
+                vp += bitreader_byte_size(&vvals);
+                sdata = vp;
+            } else {
+                vvals = bitreader_init(vp);       /* value offsets */
+                vp += bitreader_byte_size(&vvals);
+                vnull = (const uint8_t*)vp;
+                vp += ((vvals.size + 7) / 8 + 7) & ~(uint64_t)7;
+                sdata = vp;
+            }
+        } else if (is_dbl) {
             uint64_t cnt;
             memcpy(&cnt, vp, 8);
             ddata = (const double*)(vp + 8);
@@ -2082,7 +2102,31 @@ int yto_versioned_read(const YtVersionedColumn* col, uint64_t timestamp,
                 if ((int64_t)ti >= upper) break;
                 int nul;
                 uint64_t bits = 0;
-                if (is_dict) {
+                if (is_str) {
+                    /* bits = (offset within V->data) << 24 | length */
+                    if (is_sdict) {
+                        uint64_t id = bitreader_get(&vids, j);
+                        nul = (id == 0);
+                        if (!nul) {
+                            uint64_t e2 = vcum_at(&vvals, (uint32_t)V->base_value,
+                                                  (int64_t)id - 1);
+                            uint64_t b2 = vcum_at(&vvals, (uint32_t)V->base_value,
+                                                  (int64_t)id - 2);
+                            bits = ((uint64_t)(sdata - (const char*)V->data + b2) << 24)
+                                 | (e2 - b2);
+                        }
+                    } else {
+                        nul = (vnull[j / 8] >> (j % 8)) & 1;
+                        if (!nul) {
+                            uint64_t e2 = vcum_at(&vvals, (uint32_t)V->base_value,
+                                                  (int64_t)j);
+                            uint64_t b2 = vcum_at(&vvals, (uint32_t)V->base_value,
+                                                  (int64_t)j - 1);
+                            bits = ((uint64_t)(sdata - (const char*)V->data + b2) << 24)
+                                 | (e2 - b2);
+                        }
+                    }
+                } else if (is_dict) {
                     uint64_t id = bitreader_get(&vids, j);
                     nul = (id == 0);
                     if (!nul)

@@ -158,6 +158,10 @@ VSEG_INT_DIRECT_SPARSE = 2
 VSEG_INT_DICT_SPARSE = 3
 VSEG_DOUBLE_DENSE = 16
 VSEG_DOUBLE_SPARSE = 18
+VSEG_STR_DIRECT_DENSE = 32
+VSEG_STR_DICT_DENSE = 33
+VSEG_STR_DIRECT_SPARSE = 34
+VSEG_STR_DICT_SPARSE = 35
 VSEG_F_AGGREGATE = 1
 
 

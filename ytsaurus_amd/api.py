@@ -21,7 +21,7 @@ __all__ = [
     "col", "lit", "null", "litf", "Plan", "Join", "agg_sum", "agg_sum1",
     "agg_first", "agg_avg",
     "encode_int64", "encode_double", "encode_bool", "encode_string", "encode_string_raw",
-    "oracle_decode_strings", "encode_versioned_int64", "encode_versioned_double", "oracle_versioned_read", "gpu_versioned_read", "gpu_versioned_scan_chunk", "gpu_versioned_scan_table", "ScanChunk", "VersionedColumn",
+    "oracle_decode_strings", "encode_versioned_int64", "encode_versioned_double", "encode_versioned_string", "oracle_versioned_read", "gpu_versioned_read", "gpu_versioned_scan_chunk", "gpu_versioned_scan_table", "ScanChunk", "VersionedColumn",
     "Chunk", "oracle_execute",
     "oracle_partial", "oracle_merge", "oracle_partial_mk", "oracle_merge_mk",
     "gpu_key_ranges", "gpu_partial_mk", "gpu_merge_mk",
@@ -382,10 +382,73 @@ def encode_versioned_double(writes_per_row, write_ts, values, value_nulls,
                              value_agg, is_double=True)
 
 
+def encode_versioned_string(writes_per_row, write_ts, values, value_nulls,
+                            deletes_per_row, delete_ts,
+                            max_rows_per_segment=0, value_agg=None):
+    """Versioned STRING column (string_column_writer.cpp
+    TVersionedStringColumnWriter; versioned Any/Composite share the byte
+    layout). `values` is a list of bytes per write (b"" allowed); null
+    writes (value_nulls[i] == 1) must be b""."""
+    wpr = np.ascontiguousarray(writes_per_row, dtype=np.uint32)
+    wts = np.ascontiguousarray(write_ts, dtype=np.uint64)
+    vn = np.ascontiguousarray(value_nulls, dtype=np.uint8) \
+        if value_nulls is not None else None
+    va = np.ascontiguousarray(value_agg, dtype=np.uint8) \
+        if value_agg is not None else None
+    dpr = np.ascontiguousarray(deletes_per_row, dtype=np.uint32)
+    dts = np.ascontiguousarray(delete_ts, dtype=np.uint64)
+    nw, nd = int(wpr.sum()), int(dpr.sum())
+    if (len(wts) != nw or len(values) != nw
+            or (vn is not None and len(vn) != nw)
+            or (va is not None and len(va) != nw)):
+        raise ValueError("versioned string encode: write_ts/values/nulls/agg "
+                         f"lengths must equal sum(writes_per_row)={nw}")
+    if len(dts) != nd:
+        raise ValueError("versioned string encode: delete_ts length must "
+                         f"equal sum(deletes_per_row)={nd}")
+    lens = np.array([len(v) for v in values], dtype=np.uint32)
+    blob = b"".join(bytes(v) for v in values)
+    cc = _abi.YtVersionedColumn()
+    err = C.create_string_buffer(256)
+    rc = _abi.gpu_lib().yt_encode_versioned_string(
+        wpr.ctypes.data_as(C.POINTER(C.c_uint32)),
+        wts.ctypes.data_as(C.POINTER(C.c_uint64)),
+        C.c_char_p(blob),
+        lens.ctypes.data_as(C.POINTER(C.c_uint32)),
+        vn.ctypes.data_as(C.POINTER(C.c_uint8)) if vn is not None else None,
+        va.ctypes.data_as(C.POINTER(C.c_uint8)) if va is not None else None,
+        dpr.ctypes.data_as(C.POINTER(C.c_uint32)),
+        dts.ctypes.data_as(C.POINTER(C.c_uint64)),
+        C.c_int64(len(wpr)), C.c_int64(max_rows_per_segment),
+        C.byref(cc), err, 256)
+    _check(rc, err)
+    return VersionedColumn(cc, len(wpr))
+
+
 def _vbits_decode(vcol, bits, nulls, n):
-    is_dbl = any(vcol._c.val_segs[i].type >= _abi.VSEG_DOUBLE_DENSE
-                 for i in range(vcol._c.val_seg_count))
-    if is_dbl:
+    types = {vcol._c.val_segs[i].type for i in range(vcol._c.val_seg_count)}
+    if any(t >= _abi.VSEG_STR_DIRECT_DENSE for t in types):
+        # bits = (byte offset within the owning value-segment blob)<<24 | len
+        segs = vcol._c.val_segs
+        bounds = []
+        at = 0
+        for i in range(vcol._c.val_seg_count):
+            at += segs[i].row_count
+            bounds.append(at)
+        out = []
+        si = 0
+        for r in range(n):
+            while r >= bounds[si]:
+                si += 1
+            if nulls[r]:
+                out.append(None)
+                continue
+            b = int(bits[r])
+            off, ln = b >> 24, b & 0xFFFFFF
+            blob = C.string_at(segs[si].data, segs[si].data_size)
+            out.append(blob[off:off + ln])
+        return out
+    if any(t >= _abi.VSEG_DOUBLE_DENSE for t in types):
         dbl = bits.view(np.float64)
         return [None if nulls[i] else float(dbl[i]) for i in range(n)]
     return [None if nulls[i] else int(bits[i].astype(np.int64))

@@ -646,6 +646,7 @@ extern "C" int yt_encode_string_column(
 static int encode_versioned_impl(
     const uint32_t* writes_per_row, const uint64_t* write_ts,
     const int64_t* ivalues, const double* dvalues,
+    const char* sbytes, const uint32_t* slens,
     const uint8_t* value_nulls, const uint8_t* value_agg,
     const uint32_t* deletes_per_row, const uint64_t* delete_ts,
     int64_t row_count, int64_t max_rows_per_segment,
@@ -656,6 +657,7 @@ static int encode_versioned_impl(
     int nseg = (int)((row_count + max_rows_per_segment - 1) / max_rows_per_segment);
     if (nseg == 0) return YT_OK;
     const int is_double = dvalues != nullptr;
+    const int is_str = sbytes != nullptr;
 
     auto* tsegs = (YtTimestampSeg*)calloc(nseg, sizeof(YtTimestampSeg));
     auto* vsegs = (YtVersionedValueSeg*)calloc(nseg, sizeof(YtVersionedValueSeg));
@@ -669,6 +671,7 @@ static int encode_versioned_impl(
     };
 
     int64_t wat = 0, dat = 0;   /* global flattened cursors */
+    int64_t sbat = 0;           /* global byte cursor (string values) */
     for (int si = 0; si < nseg; si++) {
         int64_t r0 = (int64_t)si * max_rows_per_segment;
         int64_t r1 = r0 + max_rows_per_segment;
@@ -688,6 +691,8 @@ static int encode_versioned_impl(
         };
         std::vector<uint64_t> vals_zz, tsids, voff;
         std::vector<double> vals_d;
+        std::vector<const char*> sptr;          /* string values */
+        std::vector<uint32_t> slen2;
         std::vector<uint8_t> vnull, vagg;
         uint64_t vmin = ~0ULL, vmax = 0, max_tsid = 0;
         for (int64_t r = r0; r < r1; r++) {
@@ -701,7 +706,17 @@ static int encode_versioned_impl(
                 }
                 wids.push_back(reg(ts));
                 uint8_t nul = value_nulls ? value_nulls[wat + i] : 0;
-                if (is_double) {
+                if (is_str) {
+                    uint32_t sl = slens[wat + i];
+                    if (nul && sl != 0) {
+                        set_err(errbuf, errlen,
+                                "versioned string: null write with nonzero length");
+                        return fail(YT_ERR_INVALID_CHUNK);
+                    }
+                    sptr.push_back(sbytes + sbat);
+                    slen2.push_back(sl);
+                    sbat += sl;
+                } else if (is_double) {
                     vals_d.push_back(nul ? 0.0 : dvalues[wat + i]);
                 } else {
                     uint64_t zz = 0;
@@ -775,11 +790,70 @@ static int encode_versioned_impl(
         int64_t sparse_sz = cs_bytes((uint64_t)rows, nvalues);
         int dense = dense_sz <= sparse_sz || nvalues == 0;
 
+        /* string value layout: dictionary iff dictionaryByteSize <
+         * directByteSize (string_column_writer.cpp:80-92,338-352); direct =
+         * [cumulative END offsets, diff-from-expected][null bitmap][bytes],
+         * dictionary = [ids, 0=null first-appearance][cumulative dictionary
+         * END offsets, diff-from-expected][dictionary bytes] */
+        int use_sdict = 0;
+        uint32_t s_expected = 0;
+        std::vector<uint64_t> s_offs, s_ids;
+        std::vector<char> s_data;
+        if (is_str) {
+            uint64_t direct_bytes = 0, max_len = 0;
+            for (int64_t i = 0; i < nvalues; i++) {
+                direct_bytes += slen2[i];
+                if (slen2[i] > max_len) max_len = slen2[i];
+            }
+            std::unordered_map<std::string, uint32_t> suniq;
+            std::vector<std::pair<const char*, uint32_t>> sdict;
+            uint64_t dict_bytes = 0;
+            std::vector<uint64_t> ids1;
+            ids1.reserve(nvalues);
+            for (int64_t i = 0; i < nvalues; i++) {
+                if (vnull[i]) { ids1.push_back(0); continue; }
+                auto it = suniq.emplace(std::string(sptr[i], slen2[i]),
+                                        (uint32_t)sdict.size() + 1);
+                if (it.second) {
+                    sdict.push_back({sptr[i], slen2[i]});
+                    dict_bytes += slen2[i];
+                }
+                ids1.push_back(it.first->second);
+            }
+            int64_t direct_sz = (int64_t)direct_bytes
+                              + cs_bytes(max_len, nvalues);
+            int64_t dict_sz = (int64_t)dict_bytes
+                            + cs_bytes(max_len, (int64_t)sdict.size())
+                            + cs_bytes((uint64_t)sdict.size() + 1, nvalues);
+            use_sdict = dict_sz < direct_sz;
+            uint64_t sdiff_max = 0;
+            if (use_sdict) {
+                s_ids.swap(ids1);
+                uint64_t off = 0;
+                for (auto& dv : sdict) {
+                    s_data.insert(s_data.end(), dv.first, dv.first + dv.second);
+                    off += dv.second;
+                    s_offs.push_back(off);
+                }
+                s_expected = prepare_diff_from_expected(&s_offs, &sdiff_max);
+            } else {
+                uint64_t off = 0;
+                for (int64_t i = 0; i < nvalues; i++) {
+                    if (!vnull[i])
+                        s_data.insert(s_data.end(), sptr[i], sptr[i] + slen2[i]);
+                    off += slen2[i];            /* null steps 0 */
+                    s_offs.push_back(off);
+                }
+                s_expected = prepare_diff_from_expected(&s_offs, &sdiff_max);
+            }
+            vmax = sdiff_max; vmin = 0;         /* reuse vspan for the pack */
+        }
+
         /* int value layout: dictionary iff dictionarySize < directSize */
         uint64_t vspan = vmax - vmin;         /* wraps when all null/empty */
         int use_dict = 0;
         std::vector<uint64_t> vdict, vids;
-        if (!is_double) {
+        if (!is_double && !is_str) {
             std::unordered_map<uint64_t, uint32_t> vuniq;
             for (int64_t i = 0; i < nvalues; i++) {
                 if (!vnull[i]) vuniq.emplace(vals_zz[i], 0);
@@ -819,7 +893,17 @@ static int encode_versioned_impl(
         }
         vb.pack(tsids, max_tsid);
         if (value_agg) vb.bitmap(vagg);
-        if (is_double) {
+        if (is_str) {
+            if (use_sdict) {
+                vb.pack(s_ids, s_offs.size() + 1);
+                vb.pack(s_offs, vspan);         /* vspan = offset diff max */
+                vb.raw(s_data.data(), s_data.size());
+            } else {
+                vb.pack(s_offs, vspan);
+                vb.bitmap(vnull);
+                vb.raw(s_data.data(), s_data.size());
+            }
+        } else if (is_double) {
             uint64_t cnt = (uint64_t)nvalues;
             vb.raw(&cnt, 8);
             vb.raw(vals_d.data(), vals_d.size() * 8);
@@ -834,9 +918,11 @@ static int encode_versioned_impl(
 
         YtVersionedValueSeg& V = vsegs[si];
         V.row_count = rows;
-        V.base_value = vmin;
+        V.base_value = is_str ? (uint64_t)s_expected : vmin;
         V.expected_values_per_row = exp_v;
-        V.type = is_double
+        V.type = is_str
+            ? (YT_VSEG_STR_DIRECT_DENSE | (use_sdict ? 1 : 0) | (dense ? 0 : 2))
+            : is_double
             ? (dense ? YT_VSEG_DOUBLE_DENSE : YT_VSEG_DOUBLE_SPARSE)
             : (use_dict ? (dense ? YT_VSEG_INT_DICT_DENSE : YT_VSEG_INT_DICT_SPARSE)
                         : (dense ? YT_VSEG_INT_DIRECT_DENSE : YT_VSEG_INT_DIRECT_SPARSE));
@@ -863,6 +949,7 @@ extern "C" int yt_encode_versioned_int64(
     YtVersionedColumn* out, char* errbuf, size_t errlen)
 {
     return encode_versioned_impl(writes_per_row, write_ts, values, nullptr,
+                                 nullptr, nullptr,
                                  value_nulls, value_agg, deletes_per_row,
                                  delete_ts, row_count, max_rows_per_segment,
                                  out, errbuf, errlen);
@@ -877,6 +964,25 @@ extern "C" int yt_encode_versioned_double(
     YtVersionedColumn* out, char* errbuf, size_t errlen)
 {
     return encode_versioned_impl(writes_per_row, write_ts, nullptr, values,
+                                 nullptr, nullptr,
+                                 value_nulls, value_agg, deletes_per_row,
+                                 delete_ts, row_count, max_rows_per_segment,
+                                 out, errbuf, errlen);
+}
+
+/* versioned STRING column (string_column_writer.cpp
+ * TVersionedStringColumnWriter; the Any/Composite writers share the byte
+ * layout :365-401) */
+extern "C" int yt_encode_versioned_string(
+    const uint32_t* writes_per_row, const uint64_t* write_ts,
+    const char* value_bytes, const uint32_t* value_lens,
+    const uint8_t* value_nulls, const uint8_t* value_agg,
+    const uint32_t* deletes_per_row, const uint64_t* delete_ts,
+    int64_t row_count, int64_t max_rows_per_segment,
+    YtVersionedColumn* out, char* errbuf, size_t errlen)
+{
+    return encode_versioned_impl(writes_per_row, write_ts, nullptr, nullptr,
+                                 value_bytes, value_lens,
                                  value_nulls, value_agg, deletes_per_row,
                                  delete_ts, row_count, max_rows_per_segment,
                                  out, errbuf, errlen);

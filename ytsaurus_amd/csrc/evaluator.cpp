@@ -3029,6 +3029,17 @@ extern "C" int yt_gpu_versioned_scan_table(
             return YT_ERR_UNSUPPORTED;
         }
     }
+    for (int j = 0; j < nvcols; j++) {
+        for (int i = 0; i < vcols[j]->val_seg_count; i++) {
+            if (vcols[j]->val_segs[i].type >= YT_VSEG_STR_DIRECT_DENSE) {
+                set_err(errbuf, errlen,
+                        "scan_table: versioned string value columns are "
+                        "readable (yt_gpu_versioned_read) but not yet "
+                        "bridged into engine chunks");
+                return YT_ERR_UNSUPPORTED;
+            }
+        }
+    }
     const int ncols = nkey + nvcols;
 
     auto* H = new ScanChunkHandle();

@@ -875,8 +875,10 @@ k_versioned_read(const VSegDev* segs, int nseg, int64_t total_rows,
         VVecRef ddiffs = vvec_parse(&tp);
         const char* vp = S.val_data;
         const bool sparse = (S.vtype & 2) != 0;   /* *_SPARSE codes */
-        const bool is_dbl = S.vtype >= YT_VSEG_DOUBLE_DENSE;
-        const bool is_dict = !is_dbl && (S.vtype & 1) != 0;
+        const bool is_str = S.vtype >= YT_VSEG_STR_DIRECT_DENSE;
+        const bool is_dbl = !is_str && S.vtype >= YT_VSEG_DOUBLE_DENSE;
+        const bool is_dict = !is_dbl && !is_str && (S.vtype & 1) != 0;
+        const bool is_sdict = is_str && (S.vtype & 1) != 0;
         VVecRef vindex = vvec_parse(&vp);         /* dense offsets | sparse row idx */
         VVecRef tsids = vvec_parse(&vp);
         const uint8_t* vaggbm = nullptr;
@@ -888,7 +890,22 @@ k_versioned_read(const VSegDev* segs, int nseg, int64_t total_rows,
         VVecRef vids = {};                        /* dictionary ids */
         const double* ddata = nullptr;
         const uint8_t* vnull = nullptr;
-        if (is_dbl) {
+        const char* sdata = nullptr;              /* string / dict bytes */
+        if (is_str) {
+            /* string_column_writer.cpp layouts (see include/ytql_gpu.h):
+             * direct = [END offsets diff][null bitmap][bytes];
+             * dict   = [ids 0=null][dict END offsets diff][bytes] */
+            if (is_sdict) {
+                vids = vvec_parse(&vp);
+                vvals = vvec_parse(&vp);          /* dictionary offsets */
+                sdata = vp;
+            } else {
+                vvals = vvec_parse(&vp);          /* value END offsets */
+                vnull = (const uint8_t*)vp;
+                vp += ((vvals.size + 7) / 8 + 7) & ~(uint64_t)7;
+                sdata = vp;
+            }
+        } else if (is_dbl) {
             uint64_t cnt = *(const uint64_t*)vp;
             ddata = (const double*)(vp + 8);
             vnull = (const uint8_t*)(vp + 8 + cnt * 8);
@@ -939,7 +956,31 @@ k_versioned_read(const VSegDev* segs, int nseg, int64_t total_rows,
             if (ti >= upper) break;
             int nul;
             uint64_t bits = 0;
-            if (is_dict) {
+            if (is_str) {
+                /* bits = (byte offset within the value blob) << 24 | len */
+                if (is_sdict) {
+                    uint64_t id = vvec_get(vids, j);
+                    nul = (id == 0);
+                    if (!nul) {
+                        uint64_t e2 = vcum_dev(vvals, (uint32_t)S.base_value,
+                                               (int64_t)id - 1);
+                        uint64_t b2 = vcum_dev(vvals, (uint32_t)S.base_value,
+                                               (int64_t)id - 2);
+                        bits = ((uint64_t)(sdata - S.val_data + b2) << 24)
+                             | (e2 - b2);
+                    }
+                } else {
+                    nul = (vnull[j / 8] >> (j % 8)) & 1;
+                    if (!nul) {
+                        uint64_t e2 = vcum_dev(vvals, (uint32_t)S.base_value,
+                                               (int64_t)j);
+                        uint64_t b2 = vcum_dev(vvals, (uint32_t)S.base_value,
+                                               (int64_t)j - 1);
+                        bits = ((uint64_t)(sdata - S.val_data + b2) << 24)
+                             | (e2 - b2);
+                    }
+                }
+            } else if (is_dict) {
                 uint64_t id = vvec_get(vids, j);
                 nul = (id == 0);
                 if (!nul)
