@@ -1466,14 +1466,17 @@ int yto_partial(const YtPlan* plan, const YtChunk* chunk,
                 int64_t* part_counts,
                 int nthreads, char* errbuf, size_t errlen)
 {
-    if (plan->join) {
-        set_err(errbuf, errlen, "partial: join at the bottom query not this round");
-        return YT_ERR_UNSUPPORTED;
-    }
+    /* join at the bottom query: the local group-by below delegates to
+     * yto_execute, whose join path (incl. dup-key expansion) applies before
+     * grouping — matching coordinator.cpp:130-170 keeping the JoinClause in
+     * the bottom query. Multi-key + join stays refused (yto_partial_mk). */
     if (plan->key_count != 1) { set_err(errbuf, errlen, "partial: need 1 key"); return YT_ERR_UNSUPPORTED; }
     int sum_idx = -1;
     for (int a = 0; a < plan->agg_count; a++) {
-        if (plan->aggs[a]->func == YT_AGG_SUM) sum_idx = a;
+        if (plan->aggs[a]->func == YT_AGG_SUM) {
+            if (sum_idx >= 0) { set_err(errbuf, errlen, "partial: one sum agg max this round"); return YT_ERR_UNSUPPORTED; }
+            sum_idx = a;
+        }
         else if (plan->aggs[a]->func != YT_AGG_SUM1) { set_err(errbuf, errlen, "partial: sum/sum1 only"); return YT_ERR_UNSUPPORTED; }
     }
 
@@ -1787,7 +1790,7 @@ int yto_partial_mk(const YtPlan* plan, const YtChunk* chunk,
                    int64_t* part_counts,
                    int nthreads, char* errbuf, size_t errlen)
 {
-    if (plan->join) { set_err(errbuf, errlen, "partial: no join"); return YT_ERR_UNSUPPORTED; }
+    if (plan->join) { set_err(errbuf, errlen, "partial: join with multi-key GROUP BY not this round"); return YT_ERR_UNSUPPORTED; }
     if (plan->key_count < 1 || plan->key_count > 4) { set_err(errbuf, errlen, "partial: 1..4 keys"); return YT_ERR_UNSUPPORTED; }
     int sum_idx = -1;
     for (int a = 0; a < plan->agg_count; a++) {

@@ -745,6 +745,7 @@ def oracle_execute(plan, chunk, nthreads=1, expect_error=False):
 
 
 def oracle_partial(plan, chunk, nparts, nthreads=1):
+    _j = _attach_join(plan, lambda c: c.c_host())
     ch = chunk.c_host()
     cap = chunk.row_count + 16
     states = (YtStateRow * cap)()
@@ -759,6 +760,7 @@ def oracle_partial(plan, chunk, nparts, nthreads=1):
 def oracle_partial_mk(plan, chunk, nparts, key_ranges, nthreads=1):
     """multi-key bottom query: key_ranges = (zzmin list, zzmax list) reduced
     across ranks (yt_gpu_key_ranges / oracle equivalents)."""
+    _j = _attach_join(plan, lambda c: c.c_host())
     ch = chunk.c_host()
     cap = chunk.row_count + 16
     states = (YtStateRow * cap)()
@@ -907,7 +909,14 @@ def make_rowset(capacity, ncols, pool_bytes=0):
 
 
 def gpu_partial(plan, device_chunk, nparts, states_dev_ptr, capacity_rows,
-                max_groups_hint=0, stream=0):
+                max_groups_hint=0, stream=0, join_foreign=None):
+    _j = None
+    if getattr(plan, "join", None) is not None:
+        assert join_foreign is not None, "plan has a join: pass join_foreign="
+        _j = plan.join.c_struct(join_foreign)
+        plan.c.join = C.cast(C.pointer(_j), C.c_void_p)
+    else:
+        plan.c.join = None
     opts = YtExecOptions(max_groups_hint=max_groups_hint, stream=stream)
     counts = (C.c_int64 * nparts)()
     st = YtStatistics()

@@ -3425,7 +3425,17 @@ static int query_partial_impl(
         return YT_ERR_UNSUPPORTED;
     }
     if (partition_count < 1 || partition_count > 64) { set_err(errbuf, errlen, "partial: 1..64 partitions"); return YT_ERR_UNSUPPORTED; }
-    if (plan->join) { set_err(errbuf, errlen, "partial: join at the bottom query not this round"); return YT_ERR_UNSUPPORTED; }
+    /* join at the bottom query (the reference's coordinated split keeps
+     * JoinClause in the bottom, coordinator.cpp:130-170: the dimension
+     * table is available on every node) — each rank builds the foreign
+     * hash table from its own copy of the dimension chunk and the generic
+     * scan probes it while producing partial states. Multi-key packing
+     * would need zigzag ranges of foreign columns; deferred. */
+    if (plan->join && plan->key_count > 1) {
+        set_err(errbuf, errlen,
+                "partial: join with multi-key GROUP BY not this round");
+        return YT_ERR_UNSUPPORTED;
+    }
     int sum_slot = -1;
     for (int a = 0; a < plan->agg_count; a++) {
         if (plan->aggs[a]->func == YT_AGG_SUM) {
@@ -3480,7 +3490,16 @@ static int query_partial_impl(
     if (rc) return rc;
     double tq0 = now_ms();
     static const JoinDev kNoJoin2 = {};
-    rc = run_scan(plan, chunk, options, &R, &dp, &kNoJoin2, &fs, maxw, stats, errbuf, errlen);
+    JoinRun JR2;
+    const JoinDev* jd2 = &kNoJoin2;
+    if (plan->join) {
+        rc = setup_join(plan, chunk, &JR2,
+                        (hipStream_t)(uintptr_t)options->stream, errbuf, errlen);
+        if (rc) return rc;
+        jd2 = &JR2.jd;
+        /* dup foreign keys are fine here: a partial is always grouped */
+    }
+    rc = run_scan(plan, chunk, options, &R, &dp, jd2, &fs, maxw, stats, errbuf, errlen);
     if (rc) return rc;
     if (getenv("YTQL_TIMING"))
         fprintf(stderr, "[ytql timing] partial: setup %.2fms scan %.2fms\n",
@@ -3547,12 +3566,10 @@ static int query_partial_impl(
                                  hipMemcpyHostToDevice, R.stream));
         int sum_is_double = 0;
         if (sum_slot >= 0) {
-            uint8_t ct2[kMaxCols];
-            memset(ct2, YT_VT_INT64, sizeof(ct2));
-            for (int c2 = 0; c2 < chunk->column_count && c2 < kMaxCols; c2++)
-                ct2[c2] = (uint8_t)chunk->columns[c2].value_type;
+            /* dp.col_types already appends the joined foreign columns */
             sum_is_double =
-                expr_static_type(plan->aggs[sum_slot]->arg, ct2) == YT_VT_DOUBLE;
+                expr_static_type(plan->aggs[sum_slot]->arg, dp.col_types)
+                == YT_VT_DOUBLE;
         }
         HIP_CHECK(ytql_launch_part_scatter(R.d_groups, total, partition_count, sum_slot,
                                            sum_is_double,
