@@ -502,6 +502,36 @@ int yt_gpu_merge_states_mk(
     const YtExecOptions* options, YtRowset* output, YtStatistics* stats,
     char* errbuf, size_t errlen);
 
+/* STRING-keyed coordinated split (the key shuffle of
+ * shuffling_reader.cpp:40-42 applied to string group keys). The bottom
+ * query partitions its local string group-by results into YtStateRow
+ * records — key_bits = (byte offset within the state's own partition pool
+ * slice) << 24 | length — plus the per-partition key-byte pool slices.
+ * Partition = splitmix64(FNV-1a(key bytes)) % partition_count; the null
+ * key partitions as splitmix64(FNV-basis ^ 0xDEADBEEF12345678) with
+ * key_bits 0 and meta bit0 set. The caller exchanges BOTH buffers (counts
+ * and byte counts returned per partition); the receiver concatenates its
+ * received slices in a fixed segment order and hands the extents to the
+ * merge. */
+int yt_gpu_query_partial_str(
+    const YtPlan* plan, const YtChunk* chunk, const YtExecOptions* options,
+    int32_t partition_count, void* states_device, int64_t capacity_rows,
+    void* pool_device, int64_t pool_capacity,
+    int64_t* part_counts, int64_t* part_pool_bytes,
+    YtStatistics* stats, char* errbuf, size_t errlen);
+
+/* string-keyed front merge: states/pool are DEVICE buffers holding the
+ * received segments concatenated in order; seg_counts / seg_pool_bytes
+ * give each segment's extent (a state's slice-local key reference is
+ * rebased by its segment's pool offset). col_types = the original chunk's
+ * column types (resolves the sum result type). */
+int yt_gpu_merge_states_str(
+    const YtPlan* plan, const void* states_device, const int64_t* seg_counts,
+    int32_t nseg_in, const void* pool_device, const int64_t* seg_pool_bytes,
+    const uint8_t* col_types,
+    const YtExecOptions* options, YtRowset* output, YtStatistics* stats,
+    char* errbuf, size_t errlen);
+
 /* ---- chunk encoder (host-side product component; the synthetic-data
  * generator — reference writer semantics, integer_column_writer.cpp). ---- */
 

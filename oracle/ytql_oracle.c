@@ -1935,6 +1935,276 @@ int yto_merge_mk(const YtPlan* plan, const YtStateRow* states, int64_t nstates,
     return YT_OK;
 }
 
+/* ---- string-keyed two-phase (shuffling_reader.cpp key shuffle applied to
+ * string group keys). State rows reuse YtStateRow with key_bits =
+ * (offset within the state's own partition pool slice)<<24 | len; the
+ * null key has key_bits 0 and meta bit0. Partition =
+ * splitmix64(FNV-1a(key bytes)) % world; null key partitions as
+ * splitmix64(FNV-basis ^ 0xDEADBEEF12345678). Must match the GPU
+ * k_strst_* kernels bit-for-bit. */
+static uint64_t fnv1a_c(const char* p, uint32_t len)
+{
+    uint64_t h = 0xCBF29CE484222325ULL;
+    for (uint32_t k = 0; k < len; k++)
+        h = (h ^ (uint8_t)p[k]) * 0x100000001B3ULL;
+    return h;
+}
+
+ORACLE_EXPORT
+int yto_partial_str(const YtPlan* plan, const YtChunk* chunk,
+                    int32_t partition_count,
+                    YtStateRow* out, int64_t capacity_rows,
+                    char* pool_out, int64_t pool_capacity,
+                    int64_t* part_counts, int64_t* part_pool_bytes,
+                    int nthreads, char* errbuf, size_t errlen)
+{
+    if (plan->key_count != 1) { set_err(errbuf, errlen, "partial_str: need 1 key"); return YT_ERR_UNSUPPORTED; }
+    int sum_idx = -1;
+    for (int a = 0; a < plan->agg_count; a++) {
+        if (plan->aggs[a]->func == YT_AGG_SUM) {
+            if (sum_idx >= 0) { set_err(errbuf, errlen, "partial_str: one sum agg max"); return YT_ERR_UNSUPPORTED; }
+            sum_idx = a;
+        } else if (plan->aggs[a]->func != YT_AGG_SUM1) {
+            set_err(errbuf, errlen, "partial_str: sum/sum1 only");
+            return YT_ERR_UNSUPPORTED;
+        }
+    }
+    YtPlan local = *plan;
+    local.project_count = 0;
+    local.projects = NULL;
+
+    int64_t cap = chunk->row_count + 16;
+    int ncols = 1 + plan->agg_count;
+    YtValue* tmp = malloc(sizeof(YtValue) * cap * ncols);
+    char* tpool = malloc((size_t)(pool_capacity > 0 ? pool_capacity : 1));
+    if (!tmp || !tpool) { free(tmp); free(tpool); return YT_ERR_CAPACITY; }
+    YtRowset rs;
+    memset(&rs, 0, sizeof(rs));
+    rs.values = tmp;
+    rs.capacity_rows = cap;
+    rs.string_pool = tpool;
+    rs.string_pool_capacity = pool_capacity;
+    YtStatistics st;
+    memset(&st, 0, sizeof(st));
+    int rc = yto_execute(&local, chunk, &rs, &st, nthreads, errbuf, errlen);
+    if (rc != YT_OK) { free(tmp); free(tpool); return rc; }
+
+    const int p_null = (int)(splitmix64(
+        0xCBF29CE484222325ULL ^ 0xDEADBEEF12345678ULL) % (uint64_t)partition_count);
+    int64_t* counts = calloc(partition_count, sizeof(int64_t));
+    int64_t* bytes = calloc(partition_count, sizeof(int64_t));
+    for (int64_t r = 0; r < rs.row_count; r++) {
+        const YtValue* row = rs.values + r * ncols;
+        if (row[0].type == YT_VT_NULL) { counts[p_null]++; continue; }
+        uint64_t h = splitmix64(fnv1a_c(row[0].data.str, row[0].length));
+        int p = (int)(h % (uint64_t)partition_count);
+        counts[p]++;
+        bytes[p] += row[0].length;
+    }
+    int64_t rtot = 0, btot = 0;
+    int64_t* roff = calloc(partition_count, sizeof(int64_t));
+    int64_t* boff = calloc(partition_count, sizeof(int64_t));
+    int64_t* bbase = calloc(partition_count, sizeof(int64_t));
+    for (int p = 0; p < partition_count; p++) {
+        roff[p] = rtot;
+        bbase[p] = btot;
+        rtot += counts[p];
+        btot += bytes[p];
+    }
+    if (rtot > capacity_rows || btot > pool_capacity) {
+        free(tmp); free(tpool); free(counts); free(bytes);
+        free(roff); free(boff); free(bbase);
+        set_err(errbuf, errlen, "partial_str: state/pool buffer too small");
+        return YT_ERR_CAPACITY;
+    }
+    for (int64_t r = 0; r < rs.row_count; r++) {
+        const YtValue* row = rs.values + r * ncols;
+        int knull = (row[0].type == YT_VT_NULL);
+        int p;
+        uint64_t key_bits = 0;
+        if (knull) {
+            p = p_null;
+        } else {
+            uint64_t h = splitmix64(fnv1a_c(row[0].data.str, row[0].length));
+            p = (int)(h % (uint64_t)partition_count);
+            int64_t bo = boff[p];
+            memcpy(pool_out + bbase[p] + bo, row[0].data.str, row[0].length);
+            key_bits = ((uint64_t)bo << 24) | row[0].length;
+            boff[p] += row[0].length;
+        }
+        YtStateRow* sr = &out[roff[p]++];
+        uint64_t rowcount = 0, sum_bits = 0, nonnull = 0, sum_dbl = 0;
+        for (int a = 0; a < plan->agg_count; a++) {
+            if (plan->aggs[a]->func == YT_AGG_SUM1) rowcount = row[1 + a].data.bits;
+            else if (a == sum_idx) {
+                sum_bits = row[1 + a].data.bits;
+                nonnull = (row[1 + a].type != YT_VT_NULL);
+                if (row[1 + a].type == YT_VT_DOUBLE) sum_dbl = 2;
+            }
+        }
+        sr->key_bits = key_bits;
+        sr->meta = (uint64_t)knull | sum_dbl | (nonnull << 8);
+        sr->sum_bits = sum_bits;
+        sr->row_count = rowcount;
+    }
+    for (int p = 0; p < partition_count; p++) {
+        part_counts[p] = counts[p];
+        part_pool_bytes[p] = bytes[p];
+    }
+    free(tmp); free(tpool); free(counts); free(bytes);
+    free(roff); free(boff); free(bbase);
+    return YT_OK;
+}
+
+ORACLE_EXPORT
+int yto_merge_str(const YtPlan* plan, const YtStateRow* states,
+                  const int64_t* seg_counts, int nseg,
+                  const char* pool, const int64_t* seg_pool_bytes,
+                  YtRowset* output, char* errbuf, size_t errlen)
+{
+    if (plan->key_count != 1) { set_err(errbuf, errlen, "merge_str: need 1 key"); return YT_ERR_UNSUPPORTED; }
+    int sum_idx = -1;
+    for (int a = 0; a < plan->agg_count; a++) {
+        if (plan->aggs[a]->func == YT_AGG_SUM) sum_idx = a;
+    }
+    int64_t n = 0;
+    for (int s = 0; s < nseg; s++) n += seg_counts[s];
+    /* open-addressing table keyed by string bytes */
+    uint64_t nslots = 64;
+    while (nslots < (uint64_t)n * 2 + 8) nslots <<= 1;
+    typedef struct {
+        const char* p;         /* NULL = empty */
+        uint32_t len;
+        uint64_t cnt, sum_bits, nonnull, sum_dbl;
+    } MSlot;
+    MSlot* slots = calloc(nslots, sizeof(MSlot));
+    MSlot nullg;
+    memset(&nullg, 0, sizeof(nullg));
+    int has_null = 0;
+    int64_t at = 0, ngroups = 0;
+    for (int s = 0; s < nseg; s++) {
+        int64_t pbase = 0;
+        for (int s2 = 0; s2 < s; s2++) pbase += seg_pool_bytes[s2];
+        for (int64_t i = 0; i < seg_counts[s]; i++, at++) {
+            const YtStateRow* sr = &states[at];
+            uint64_t nonnull = sr->meta >> 8;
+            if (sr->meta & 1) {
+                has_null = 1;
+                nullg.cnt += sr->row_count;
+                if (nonnull) {
+                    if (sr->meta & 2) {
+                        double d;
+                        memcpy(&d, &nullg.sum_bits, 8);
+                        double d2;
+                        uint64_t b2 = sr->sum_bits;
+                        memcpy(&d2, &b2, 8);
+                        d += d2;
+                        memcpy(&nullg.sum_bits, &d, 8);
+                        nullg.sum_dbl = 1;
+                    } else {
+                        nullg.sum_bits += sr->sum_bits;
+                    }
+                    nullg.nonnull += nonnull;
+                }
+                continue;
+            }
+            const char* kp = pool + pbase + (int64_t)(sr->key_bits >> 24);
+            uint32_t klen = (uint32_t)(sr->key_bits & 0xFFFFFF);
+            uint64_t h = splitmix64(fnv1a_c(kp, klen));
+            uint64_t idx = h & (nslots - 1);
+            for (;;) {
+                MSlot* sl = &slots[idx];
+                if (!sl->p) {
+                    sl->p = kp;
+                    sl->len = klen;
+                    ngroups++;
+                    /* fall through to accumulate */
+                }
+                if (sl->len == klen && memcmp(sl->p, kp, klen) == 0) {
+                    sl->cnt += sr->row_count;
+                    if (nonnull) {
+                        if (sr->meta & 2) {
+                            double d;
+                            memcpy(&d, &sl->sum_bits, 8);
+                            double d2;
+                            uint64_t b2 = sr->sum_bits;
+                            memcpy(&d2, &b2, 8);
+                            d += d2;
+                            memcpy(&sl->sum_bits, &d, 8);
+                            sl->sum_dbl = 1;
+                        } else {
+                            sl->sum_bits += sr->sum_bits;
+                        }
+                        sl->nonnull += nonnull;
+                    }
+                    break;
+                }
+                idx = (idx + 1) & (nslots - 1);
+            }
+        }
+    }
+    output->row_count = 0;
+    output->string_pool_used = 0;
+    output->column_count = 1 + plan->agg_count;
+    if (ngroups + has_null > output->capacity_rows) {
+        free(slots);
+        set_err(errbuf, errlen, "merge_str: output rowset too small");
+        return YT_ERR_CAPACITY;
+    }
+    int ncols = 1 + plan->agg_count;
+    for (uint64_t i = 0; i <= nslots; i++) {
+        const MSlot* sl;
+        int is_null_row = (i == nslots);
+        if (is_null_row) {
+            if (!has_null) break;
+            sl = &nullg;
+        } else {
+            sl = &slots[i];
+            if (!sl->p) continue;
+        }
+        YtValue* dst = output->values + output->row_count * ncols;
+        dst[0].id = 0;
+        dst[0].flags = 0;
+        if (is_null_row) {
+            dst[0].type = YT_VT_NULL;
+            dst[0].length = 0;
+            dst[0].data.bits = 0;
+        } else {
+            if (output->string_pool_used + sl->len >
+                (uint64_t)output->string_pool_capacity) {
+                free(slots);
+                set_err(errbuf, errlen, "merge_str: string pool too small");
+                return YT_ERR_CAPACITY;
+            }
+            char* dstp = output->string_pool + output->string_pool_used;
+            memcpy(dstp, sl->p, sl->len);
+            dst[0].type = YT_VT_STRING;
+            dst[0].length = sl->len;
+            dst[0].data.str = dstp;
+            output->string_pool_used += sl->len;
+        }
+        for (int a = 0; a < plan->agg_count; a++) {
+            YtValue* v = &dst[1 + a];
+            v->id = (uint16_t)(1 + a);
+            v->flags = 0;
+            v->length = 0;
+            if (plan->aggs[a]->func == YT_AGG_SUM1) {
+                v->type = YT_VT_INT64;
+                v->data.bits = sl->cnt;
+            } else if (a == sum_idx && sl->nonnull) {
+                v->type = sl->sum_dbl ? YT_VT_DOUBLE : YT_VT_INT64;
+                v->data.bits = sl->sum_bits;
+            } else {
+                v->type = YT_VT_NULL;
+                v->data.bits = 0;
+            }
+        }
+        output->row_count++;
+    }
+    free(slots);
+    return YT_OK;
+}
+
 ORACLE_EXPORT
 int yto_decode_string_column(const YtColumn* col, int64_t row_count,
                              char* out_blob, int64_t blob_cap,

@@ -242,6 +242,16 @@ def gpu_lib():
               C.c_int32, C.c_void_p, C.c_int64, C.POINTER(C.c_int64),
               C.POINTER(C.c_uint64), C.POINTER(C.c_uint64),
               C.POINTER(YtStatistics), C.c_char_p, C.c_size_t])
+        _sig(lib, "yt_gpu_query_partial_str", C.c_int,
+             [C.POINTER(YtPlan), C.POINTER(YtChunk), C.POINTER(YtExecOptions),
+              C.c_int32, C.c_void_p, C.c_int64, C.c_void_p, C.c_int64,
+              C.POINTER(C.c_int64), C.POINTER(C.c_int64),
+              C.POINTER(YtStatistics), C.c_char_p, C.c_size_t])
+        _sig(lib, "yt_gpu_merge_states_str", C.c_int,
+             [C.POINTER(YtPlan), C.c_void_p, C.POINTER(C.c_int64), C.c_int32,
+              C.c_void_p, C.POINTER(C.c_int64), C.POINTER(C.c_uint8),
+              C.POINTER(YtExecOptions), C.POINTER(YtRowset),
+              C.POINTER(YtStatistics), C.c_char_p, C.c_size_t])
         _sig(lib, "yt_encode_int64_column", C.c_int,
              [C.POINTER(C.c_int64), C.POINTER(C.c_uint8), C.c_int64, C.c_int32,
               C.c_int32, C.c_int64, C.POINTER(YtEncodedColumn), C.c_char_p,
@@ -294,6 +304,17 @@ def oracle_lib():
              [C.POINTER(YtPlan), C.POINTER(YtStateRow), C.c_int64,
               C.POINTER(C.c_uint8),
               C.POINTER(C.c_uint64), C.POINTER(C.c_uint64),
+              C.POINTER(YtRowset), C.c_char_p, C.c_size_t])
+        _sig(lib, "yto_partial_str", C.c_int,
+             [C.POINTER(YtPlan), C.POINTER(YtChunk), C.c_int32,
+              C.POINTER(YtStateRow), C.c_int64,
+              C.c_char_p, C.c_int64,
+              C.POINTER(C.c_int64), C.POINTER(C.c_int64),
+              C.c_int, C.c_char_p, C.c_size_t])
+        _sig(lib, "yto_merge_str", C.c_int,
+             [C.POINTER(YtPlan), C.POINTER(YtStateRow),
+              C.POINTER(C.c_int64), C.c_int,
+              C.c_char_p, C.POINTER(C.c_int64),
               C.POINTER(YtRowset), C.c_char_p, C.c_size_t])
         _sig(lib, "yto_partition_hash", C.c_uint64, [C.c_uint64, C.c_int])
         _sig(lib, "yto_bitunpack", C.c_int64,

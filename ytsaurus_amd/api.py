@@ -25,6 +25,7 @@ __all__ = [
     "Chunk", "oracle_execute",
     "oracle_partial", "oracle_merge", "oracle_partial_mk", "oracle_merge_mk",
     "gpu_key_ranges", "gpu_partial_mk", "gpu_merge_mk",
+    "oracle_partial_str", "oracle_merge_str", "gpu_partial_str", "gpu_merge_str",
     "gpu_available", "gpu_execute", "gpu_partial", "gpu_merge",
     "rows_from_rowset", "sort_rows", "make_rowset",
 ]
@@ -926,6 +927,93 @@ def gpu_partial(plan, device_chunk, nparts, states_dev_ptr, capacity_rows,
         C.c_void_p(states_dev_ptr), capacity_rows, counts, C.byref(st), err, 512)
     _check(rc, err)
     return [counts[i] for i in range(nparts)], st
+
+
+def oracle_partial_str(plan, chunk, nparts, nthreads=1):
+    """string-keyed bottom query: returns (states ctypes array, counts list,
+    pool bytes, pool_bytes-per-partition list). Each partition's states are
+    contiguous; key_bits reference that partition's pool slice."""
+    ch = chunk.c_host()
+    cap = chunk.row_count + 16
+    pool_cap = 64 + 32 * chunk.row_count
+    states = (YtStateRow * cap)()
+    pool = C.create_string_buffer(pool_cap)
+    counts = (C.c_int64 * nparts)()
+    pbytes = (C.c_int64 * nparts)()
+    err = C.create_string_buffer(256)
+    rc = _abi.oracle_lib().yto_partial_str(
+        C.byref(plan.c), C.byref(ch), nparts, states, cap,
+        C.cast(pool, C.c_char_p), pool_cap, counts, pbytes, nthreads,
+        err, 256)
+    _check(rc, err)
+    return (states, [counts[i] for i in range(nparts)],
+            pool.raw, [pbytes[i] for i in range(nparts)])
+
+
+def oracle_merge_str(plan, segments, out_capacity=None, pool_capacity=None):
+    """segments: list of (states ctypes array/list, count, pool bytes obj,
+    pool_byte_count) received by this partition, in a fixed order."""
+    nseg = len(segments)
+    total = sum(s[1] for s in segments)
+    allst = (YtStateRow * max(total, 1))()
+    at = 0
+    pool = b"".join(bytes(s[2][:s[3]]) for s in segments)
+    for st_arr, cnt, _, _ in segments:
+        for i in range(cnt):
+            allst[at] = st_arr[i]
+            at += 1
+    seg_counts = (C.c_int64 * nseg)(*[s[1] for s in segments])
+    seg_bytes = (C.c_int64 * nseg)(*[s[3] for s in segments])
+    cap = out_capacity or (total + 16)
+    pc = pool_capacity or max(len(pool), 1024)
+    rs = _mk_rowset(cap, pool_bytes=pc)
+    err = C.create_string_buffer(256)
+    rc = _abi.oracle_lib().yto_merge_str(
+        C.byref(plan.c), allst, seg_counts, nseg,
+        C.c_char_p(pool), seg_bytes, C.byref(rs), err, 256)
+    _check(rc, err)
+    return rows_from_rowset(rs)
+
+
+def gpu_partial_str(plan, device_chunk, nparts, states_dev_ptr, capacity_rows,
+                    pool_dev_ptr, pool_capacity, max_groups_hint=0, stream=0):
+    opts = YtExecOptions(max_groups_hint=max_groups_hint, stream=stream)
+    counts = (C.c_int64 * nparts)()
+    pbytes = (C.c_int64 * nparts)()
+    st = YtStatistics()
+    err = C.create_string_buffer(512)
+    rc = _abi.gpu_lib().yt_gpu_query_partial_str(
+        C.byref(plan.c), C.byref(device_chunk), C.byref(opts), nparts,
+        C.c_void_p(states_dev_ptr), capacity_rows,
+        C.c_void_p(pool_dev_ptr), pool_capacity, counts, pbytes,
+        C.byref(st), err, 512)
+    _check(rc, err)
+    return ([counts[i] for i in range(nparts)],
+            [pbytes[i] for i in range(nparts)], st)
+
+
+def gpu_merge_str(plan, states_dev_ptr, seg_counts, pool_dev_ptr,
+                  seg_pool_bytes, col_types=None, max_groups_hint=0,
+                  stream=0, out_capacity=None, pool_capacity=None):
+    opts = YtExecOptions(max_groups_hint=max_groups_hint, stream=stream)
+    nseg = len(seg_counts)
+    sc = (C.c_int64 * nseg)(*seg_counts)
+    sb = (C.c_int64 * nseg)(*seg_pool_bytes)
+    ct = None
+    if col_types is not None:
+        ct = (C.c_uint8 * 8)(*([int(x) for x in col_types] + [0] * (8 - len(col_types))))
+    total = sum(seg_counts)
+    cap = out_capacity or (total + 16)
+    pc = pool_capacity or max(sum(seg_pool_bytes), 1024)
+    rs = _mk_rowset(cap, pool_bytes=pc)
+    st = YtStatistics()
+    err = C.create_string_buffer(512)
+    rc = _abi.gpu_lib().yt_gpu_merge_states_str(
+        C.byref(plan.c), C.c_void_p(states_dev_ptr), sc, nseg,
+        C.c_void_p(pool_dev_ptr), sb, ct, C.byref(opts), C.byref(rs),
+        C.byref(st), err, 512)
+    _check(rc, err)
+    return rows_from_rowset(rs), st
 
 
 def gpu_merge(plan, states_dev_ptr, n_states, max_groups_hint=0, stream=0,

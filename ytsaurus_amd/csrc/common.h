@@ -347,6 +347,16 @@ struct TopkGather {
     int64_t cap_null;
 };
 
+/* merged string-state group (device→host) for the string-keyed two-phase
+ * front query: counts are full u64 (multi-rank accumulation can exceed the
+ * 32-bit halves the single-pass path packs). */
+struct OutStrState {
+    uint64_t off_len;             /* out_pool offset<<24 | len */
+    uint64_t sum_bits;
+    uint64_t cnt;
+    uint64_t nonnull;
+};
+
 /* compacted string group (device→host). 24 B: this array is the bulk of
  * the query's D2H at 100 M groups, so the key ref packs off|len (pool is
  * bounded by the dict blob bytes << 2^48) and counts stay in the slot's

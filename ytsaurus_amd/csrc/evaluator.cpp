@@ -124,7 +124,49 @@ hipError_t ytql_launch_strgrp_compact(const DevSeg*, const SegEx*, int,
                                       const StrSlot*, uint64_t, OutStrGroup*,
                                       unsigned long long*, char*, unsigned long long*,
                                       uint64_t, TableHdr*, hipStream_t);
+hipError_t ytql_launch_strst_count(const OutStrGroup*, int64_t, const char*,
+                                   int, unsigned long long*,
+                                   unsigned long long*, hipStream_t);
+hipError_t ytql_launch_strst_scatter(const OutStrGroup*, int64_t, const char*,
+                                     int, int, unsigned long long*,
+                                     unsigned long long*, YtStateRow*, char*,
+                                     const unsigned long long*, hipStream_t);
+hipError_t ytql_launch_strst_hash(const YtStateRow*, int64_t, const int64_t*,
+                                  const unsigned long long*, int, const char*,
+                                  uint64_t*, uint64_t*, ulonglong2*,
+                                  uint64_t*, hipStream_t);
+hipError_t ytql_launch_strst_merge(const YtStateRow*, int64_t, const char*,
+                                   const uint64_t*, const uint64_t*,
+                                   const ulonglong2*, const uint64_t*, int,
+                                   StrSlot*, uint64_t, TableHdr*, hipStream_t);
+hipError_t ytql_launch_strst_compact(const StrSlot*, uint64_t,
+                                     const uint64_t*, const char*,
+                                     OutStrState*, unsigned long long*,
+                                     char*, unsigned long long*, uint64_t,
+                                     TableHdr*, hipStream_t);
 }
+
+/* must match kernels.hip mix64 and the oracle's splitmix64 */
+static inline uint64_t splitmix64_host(uint64_t x)
+{
+    x += 0x9E3779B97F4A7C15ULL;
+    x = (x ^ (x >> 30)) * 0xBF58476D1CE4E5B9ULL;
+    x = (x ^ (x >> 27)) * 0x94D049BB133111EBULL;
+    return x ^ (x >> 31);
+}
+
+/* partition-output request for run_string_group (string-keyed bottom
+ * query): when set, the compacted groups are hash-partitioned into
+ * state rows + pool slices instead of being emitted as a rowset. */
+struct StrPartialOut {
+    int32_t partition_count;
+    YtStateRow* states_device;
+    int64_t capacity_rows;
+    char* pool_device;
+    int64_t pool_capacity;
+    int64_t* part_counts;        /* out */
+    int64_t* part_pool_bytes;    /* out */
+};
 
 /* ------------------------------------------------------------------ */
 /* device/pinned buffer pool: query executions reuse large allocations
@@ -2441,7 +2483,8 @@ fail:
 static int run_string_group(const YtPlan* plan, const YtChunk* chunk,
                             const YtExecOptions* options, int key_col,
                             YtRowset* output, YtStatistics* stats, double tw0,
-                            char* errbuf, size_t errlen)
+                            char* errbuf, size_t errlen,
+                            StrPartialOut* po = nullptr)
 {
     int rc = YT_OK;
     int sum_slot = -1, sum_col = -1, val_is_double = 0;
@@ -2477,10 +2520,20 @@ static int run_string_group(const YtPlan* plan, const YtChunk* chunk,
     rc = setup_chunk(chunk, &R, &mw, options->input_row_limit, &in_clamped,
                      errbuf, errlen);
     if (rc) return rc;
-    output->row_count = 0;
-    output->string_pool_used = 0;   /* rowsets are reusable across queries */
-    output->column_count = 1 + plan->agg_count;
-    if (chunk->row_count == 0 || R.nsegs == 0) return YT_OK;
+    if (output) {
+        output->row_count = 0;
+        output->string_pool_used = 0;   /* rowsets reusable across queries */
+        output->column_count = 1 + plan->agg_count;
+    }
+    if (chunk->row_count == 0 || R.nsegs == 0) {
+        if (po) {
+            for (int p = 0; p < po->partition_count; p++) {
+                po->part_counts[p] = 0;
+                po->part_pool_bytes[p] = 0;
+            }
+        }
+        return YT_OK;
+    }
     rc = setup_table(&R, plan->agg_count, options->max_groups_hint,
                      options->group_row_limit, errbuf, errlen);
     if (rc) return rc;
@@ -2648,7 +2701,7 @@ static int run_string_group(const YtPlan* plan, const YtChunk* chunk,
         HIP_CHECK(pool_alloc_host(&hgroups,
                                   sizeof(OutStrGroup) * (ngroups ? ngroups : 1)));
         unsigned long long hctr[2] = {0, 0};
-        if (ngroups > 0) {
+        if (ngroups > 0 && !po) {
             HIP_CHECK(hipMemcpyAsync(hgroups, d_out,
                                      sizeof(OutStrGroup) * ngroups,
                                      hipMemcpyDeviceToHost, R.stream));
@@ -2681,6 +2734,101 @@ static int run_string_group(const YtPlan* plan, const YtChunk* chunk,
             g_pool.put(d_ctr); g_pool.put(hgroups);
             set_err(errbuf, errlen, "string merge table/pool overflow");
             return YT_ERR_CAPACITY;
+        }
+
+        if (po) {
+            /* bottom query: hash-partition the compacted groups + the null
+             * side group into state rows and per-partition pool slices
+             * (key_bits = slice-local offset<<24 | len) */
+            auto putall = [&]() {
+                g_pool.put(d_accbase); g_pool.put(d_acc); g_pool.put(d_hashes);
+                g_pool.put(d_idents); g_pool.put(d_pfxs);
+                g_pool.put(d_slots); g_pool.put(d_out); g_pool.put(d_pool);
+                g_pool.put(d_ctr); g_pool.put(hgroups);
+            };
+            const int np = po->partition_count;
+            const int p_null = (int)(splitmix64_host(
+                0xCBF29CE484222325ULL ^ 0xDEADBEEF12345678ULL) % (uint64_t)np);
+            unsigned long long* d_pc = nullptr;   /* [np] counts, [np] bytes,
+                                                     [np] cursors, [np] bcursors,
+                                                     [np] pool_base */
+            HIP_CHECK(pool_alloc(&d_pc, sizeof(unsigned long long) * np * 5));
+            HIP_CHECK(hipMemsetAsync(d_pc, 0,
+                                     sizeof(unsigned long long) * np * 5,
+                                     R.stream));
+            if (ngroups > 0) {
+                HIP_CHECK(ytql_launch_strst_count(d_out, ngroups, d_pool, np,
+                                                  d_pc, d_pc + np, R.stream));
+            }
+            std::vector<unsigned long long> pc(2 * np);
+            HIP_CHECK(hipMemcpyAsync(pc.data(), d_pc,
+                                     sizeof(unsigned long long) * 2 * np,
+                                     hipMemcpyDeviceToHost, R.stream));
+            HIP_CHECK(hipStreamSynchronize(R.stream));
+            if (th.side_used[1]) pc[p_null] += 1;
+            unsigned long long rrun = 0, brun = 0;
+            std::vector<unsigned long long> rcur(np), pbase(np);
+            for (int p = 0; p < np; p++) {
+                rcur[p] = rrun;
+                pbase[p] = brun;
+                rrun += pc[p];
+                brun += pc[np + p];
+            }
+            if ((int64_t)rrun > po->capacity_rows
+                || (int64_t)brun > po->pool_capacity) {
+                putall();
+                g_pool.put(d_pc);
+                set_err(errbuf, errlen, "partial_str: state/pool buffer too small");
+                return YT_ERR_CAPACITY;
+            }
+            HIP_CHECK(hipMemcpyAsync(d_pc + 2 * np, rcur.data(),
+                                     sizeof(unsigned long long) * np,
+                                     hipMemcpyHostToDevice, R.stream));
+            HIP_CHECK(hipMemsetAsync(d_pc + 3 * np, 0,
+                                     sizeof(unsigned long long) * np, R.stream));
+            HIP_CHECK(hipMemcpyAsync(d_pc + 4 * np, pbase.data(),
+                                     sizeof(unsigned long long) * np,
+                                     hipMemcpyHostToDevice, R.stream));
+            if (ngroups > 0) {
+                HIP_CHECK(ytql_launch_strst_scatter(d_out, ngroups, d_pool, np,
+                                                    val_is_double,
+                                                    d_pc + 2 * np, d_pc + 3 * np,
+                                                    po->states_device,
+                                                    po->pool_device,
+                                                    d_pc + 4 * np, R.stream));
+            }
+            HIP_CHECK(hipStreamSynchronize(R.stream));
+            if (th.side_used[1]) {
+                unsigned long long pos = 0;
+                HIP_CHECK(hipMemcpy(&pos, d_pc + 2 * np + p_null,
+                                    sizeof(pos), hipMemcpyDeviceToHost));
+                YtStateRow sr;
+                memset(&sr, 0, sizeof(sr));
+                sr.key_bits = 0;
+                uint64_t nn = sum_slot >= 0 ? th.side_agg[1][2 * sum_slot + 1] : 0;
+                sr.meta = 1ULL | (val_is_double ? 2ULL : 0ULL) | (nn << 8);
+                sr.sum_bits = sum_slot >= 0 ? th.side_agg[1][2 * sum_slot] : 0;
+                sr.row_count = th.side_cnt[1];
+                HIP_CHECK(hipMemcpy(po->states_device + pos, &sr, sizeof(sr),
+                                    hipMemcpyHostToDevice));
+            }
+            for (int p = 0; p < np; p++) {
+                po->part_counts[p] = (int64_t)pc[p];
+                po->part_pool_bytes[p] = (int64_t)pc[np + p];
+            }
+            if (stats) {
+                stats->rows_read = chunk->row_count;
+                stats->grouped_row_count = (int64_t)rrun;
+                stats->incomplete_input = in_clamped;
+                stats->incomplete_output = (th.overflow == 2);
+                stats->kernel_scan_ms += ms;
+                stats->kernel_scan_launches += 1;
+                stats->kernel_other_ms += ms_other;
+                stats->execute_time_ms = now_ms() - tw0;
+            }
+            putall();
+            g_pool.put(d_pc);
+            return YT_OK;
         }
 
         /* emit [key(string), aggs...] rows. Fast path (no output limit,
@@ -3617,6 +3765,48 @@ extern "C" int yt_gpu_query_partial_mk(
                               key_zzmin, key_zzmax, stats, errbuf, errlen);
 }
 
+/* STRING-keyed bottom query: the local string group-by (run_string_group)
+ * followed by a hash partition of the groups into YtStateRow records +
+ * per-partition key-byte pool slices. The caller all-to-alls BOTH buffers
+ * (counts and byte counts per partition are returned); each received
+ * state's key_bits is an offset within its own slice. Mirrors the
+ * reference's key shuffle (shuffling_reader.cpp:40-42) applied to string
+ * group keys. */
+extern "C" int yt_gpu_query_partial_str(
+    const YtPlan* plan, const YtChunk* chunk, const YtExecOptions* options,
+    int32_t partition_count, void* states_device, int64_t capacity_rows,
+    void* pool_device, int64_t pool_capacity,
+    int64_t* part_counts, int64_t* part_pool_bytes,
+    YtStatistics* stats, char* errbuf, size_t errlen)
+{
+    int rc = yt_gpu_available(errbuf, errlen);
+    if (rc != YT_OK) return rc;
+    if (!plan || !chunk) { set_err(errbuf, errlen, "null argument"); return YT_ERR_INVALID_PLAN; }
+    if (partition_count < 1 || partition_count > 64) { set_err(errbuf, errlen, "partial_str: 1..64 partitions"); return YT_ERR_UNSUPPORTED; }
+    int kc = -1;
+    if (!(plan->key_count == 1 && plan->agg_count > 0 &&
+          expr_is_col(plan->keys[0], &kc) && kc < chunk->column_count &&
+          chunk->columns[kc].value_type == YT_VT_STRING)) {
+        set_err(errbuf, errlen, "partial_str: needs one string key column + aggregates");
+        return YT_ERR_UNSUPPORTED;
+    }
+    if (plan->join) { set_err(errbuf, errlen, "partial_str: join with string keys not this round"); return YT_ERR_UNSUPPORTED; }
+    YtExecOptions defopt;
+    memset(&defopt, 0, sizeof(defopt));
+    if (!options) options = &defopt;
+    if (stats) memset(stats, 0, sizeof(*stats));
+    StrPartialOut po;
+    po.partition_count = partition_count;
+    po.states_device = (YtStateRow*)states_device;
+    po.capacity_rows = capacity_rows;
+    po.pool_device = (char*)pool_device;
+    po.pool_capacity = pool_capacity;
+    po.part_counts = part_counts;
+    po.part_pool_bytes = part_pool_bytes;
+    return run_string_group(plan, chunk, options, kc, nullptr, stats,
+                            now_ms(), errbuf, errlen, &po);
+}
+
 /* per-rank key-column zigzag ranges for the cross-rank reduce (meta-only
  * bound: conservative is fine, the reduce of bounds is a bound) */
 extern "C" int yt_gpu_key_ranges(
@@ -3802,4 +3992,235 @@ extern "C" int yt_gpu_merge_states_mk(
     return merge_states_impl(plan, states_device, state_row_count, col_types,
                              key_zzmin, key_zzmax, options, output, stats,
                              errbuf, errlen);
+}
+
+/* STRING-keyed front query: merge exchanged state rows (key_bits =
+ * slice-local pool reference) whose pool slices were concatenated in
+ * segment order on the receiver. seg_counts / seg_pool_bytes give each
+ * segment's row and byte extent; states and pool are DEVICE buffers. */
+extern "C" int yt_gpu_merge_states_str(
+    const YtPlan* plan, const void* states_device, const int64_t* seg_counts,
+    int32_t nseg_in, const void* pool_device, const int64_t* seg_pool_bytes,
+    const uint8_t* col_types,      /* original chunk column types (sum arg) */
+    const YtExecOptions* options, YtRowset* output, YtStatistics* stats,
+    char* errbuf, size_t errlen)
+{
+    int rc = yt_gpu_available(errbuf, errlen);
+    if (rc != YT_OK) return rc;
+    if (!plan || !output || nseg_in < 1) { set_err(errbuf, errlen, "merge_str: bad arguments"); return YT_ERR_INVALID_PLAN; }
+    if (plan->key_count != 1) { set_err(errbuf, errlen, "merge_str: one string key"); return YT_ERR_UNSUPPORTED; }
+    if (plan->with_totals || plan->having || plan->order_count) {
+        set_err(errbuf, errlen, "merge_str: totals/having/order not this round");
+        return YT_ERR_UNSUPPORTED;
+    }
+    int sum_slot = -1;
+    for (int a = 0; a < plan->agg_count; a++) {
+        if (plan->aggs[a]->func == YT_AGG_SUM) {
+            if (sum_slot >= 0) { set_err(errbuf, errlen, "merge_str: one sum agg max"); return YT_ERR_UNSUPPORTED; }
+            sum_slot = a;
+        } else if (plan->aggs[a]->func != YT_AGG_SUM1) {
+            set_err(errbuf, errlen, "merge_str: sum/sum(1) only");
+            return YT_ERR_UNSUPPORTED;
+        }
+    }
+    int sum_is_double = 0;
+    if (sum_slot >= 0) {
+        uint8_t ct2[kMaxCols];
+        memset(ct2, YT_VT_INT64, sizeof(ct2));
+        if (col_types) {
+            for (int c2 = 0; c2 < kMaxCols; c2++) ct2[c2] = col_types[c2];
+        }
+        sum_is_double =
+            expr_static_type(plan->aggs[sum_slot]->arg, ct2) == YT_VT_DOUBLE;
+    }
+    YtExecOptions defopt;
+    memset(&defopt, 0, sizeof(defopt));
+    if (!options) options = &defopt;
+    if (stats) memset(stats, 0, sizeof(*stats));
+    hipStream_t st = (hipStream_t)(uintptr_t)options->stream;
+    double tw0 = now_ms();
+
+    int64_t n = 0, pool_total = 0;
+    std::vector<int64_t> row_base(nseg_in);
+    std::vector<unsigned long long> pool_base(nseg_in);
+    for (int s = 0; s < nseg_in; s++) {
+        row_base[s] = n;
+        pool_base[s] = (unsigned long long)pool_total;
+        n += seg_counts[s];
+        pool_total += seg_pool_bytes[s];
+    }
+    output->row_count = 0;
+    output->string_pool_used = 0;
+    output->column_count = 1 + plan->agg_count;
+    if (n == 0) return YT_OK;
+
+    int64_t* d_rowb = nullptr;
+    unsigned long long* d_poolb = nullptr;
+    uint64_t* d_hashes = nullptr;
+    uint64_t* d_idents = nullptr;
+    ulonglong2* d_pfxs = nullptr;
+    uint64_t* d_absoff = nullptr;
+    StrSlot* d_slots = nullptr;
+    TableHdr* d_th = nullptr;
+    OutStrState* d_out = nullptr;
+    char* d_opool = nullptr;
+    unsigned long long* d_ctr = nullptr;
+    OutStrState* hgroups = nullptr;
+    char* hpool = nullptr;
+    auto putall = [&]() {
+        g_pool.put(d_rowb); g_pool.put(d_poolb); g_pool.put(d_hashes);
+        g_pool.put(d_idents); g_pool.put(d_pfxs); g_pool.put(d_absoff);
+        g_pool.put(d_slots); g_pool.put(d_th); g_pool.put(d_out);
+        g_pool.put(d_opool); g_pool.put(d_ctr); g_pool.put(hgroups);
+        g_pool.put(hpool);
+    };
+    {
+    HIP_CHECK(pool_alloc(&d_rowb, sizeof(int64_t) * nseg_in));
+    HIP_CHECK(pool_alloc(&d_poolb, sizeof(unsigned long long) * nseg_in));
+    HIP_CHECK(hipMemcpyAsync(d_rowb, row_base.data(),
+                             sizeof(int64_t) * nseg_in,
+                             hipMemcpyHostToDevice, st));
+    HIP_CHECK(hipMemcpyAsync(d_poolb, pool_base.data(),
+                             sizeof(unsigned long long) * nseg_in,
+                             hipMemcpyHostToDevice, st));
+    HIP_CHECK(pool_alloc(&d_hashes, sizeof(uint64_t) * n));
+    HIP_CHECK(pool_alloc(&d_idents, sizeof(uint64_t) * n));
+    HIP_CHECK(pool_alloc(&d_pfxs, sizeof(ulonglong2) * n));
+    HIP_CHECK(pool_alloc(&d_absoff, sizeof(uint64_t) * n));
+    uint64_t nslots_cap = next_pow2((uint64_t)n * 2);
+    if (nslots_cap < 2048) nslots_cap = 2048;
+    uint64_t nslots = options->max_groups_hint > 0
+        ? next_pow2((uint64_t)options->max_groups_hint * 2)
+        : nslots_cap;
+    if (nslots < 2048) nslots = 2048;
+    if (nslots > nslots_cap) nslots = nslots_cap;
+    HIP_CHECK(pool_alloc(&d_th, sizeof(TableHdr)));
+    HIP_CHECK(ytql_launch_strst_hash((const YtStateRow*)states_device, n,
+                                     d_rowb, d_poolb, nseg_in,
+                                     (const char*)pool_device,
+                                     d_hashes, d_idents, d_pfxs, d_absoff, st));
+    TableHdr th;
+    for (;;) {
+        HIP_CHECK(pool_alloc(&d_slots, sizeof(StrSlot) * nslots));
+        HIP_CHECK(hipMemsetAsync(d_slots, 0, sizeof(StrSlot) * nslots, st));
+        memset(&th, 0, sizeof(th));
+        th.group_limit = options->group_row_limit;
+        HIP_CHECK(hipMemcpyAsync(d_th, &th, sizeof(th),
+                                 hipMemcpyHostToDevice, st));
+        HIP_CHECK(ytql_launch_strst_merge((const YtStateRow*)states_device, n,
+                                          (const char*)pool_device, d_hashes,
+                                          d_idents, d_pfxs, d_absoff, sum_slot,
+                                          d_slots, nslots, d_th, st));
+        HIP_CHECK(hipMemcpy(&th, d_th, sizeof(th), hipMemcpyDeviceToHost));
+        if (th.overflow != 1 || nslots >= nslots_cap) break;
+        g_pool.put(d_slots);
+        d_slots = nullptr;
+        nslots *= 4;
+        if (nslots > nslots_cap) nslots = nslots_cap;
+    }
+    if (th.overflow == 1) {
+        putall();
+        set_err(errbuf, errlen, "merge_str: table overflow");
+        return YT_ERR_CAPACITY;
+    }
+    int64_t ngroups = (int64_t)th.ngroups;
+    int has_null = th.side_used[1] ? 1 : 0;
+    if (ngroups + has_null > output->capacity_rows) {
+        putall();
+        set_err(errbuf, errlen, "merge_str: output rowset too small");
+        return YT_ERR_CAPACITY;
+    }
+    uint64_t opool_cap = (uint64_t)(pool_total ? pool_total : 1);
+    HIP_CHECK(pool_alloc(&d_out, sizeof(OutStrState) * (ngroups ? ngroups : 1)));
+    HIP_CHECK(pool_alloc(&d_opool, opool_cap));
+    HIP_CHECK(pool_alloc(&d_ctr, 2 * sizeof(unsigned long long)));
+    HIP_CHECK(hipMemsetAsync(d_ctr, 0, 2 * sizeof(unsigned long long), st));
+    if (ngroups > 0) {
+        HIP_CHECK(ytql_launch_strst_compact(d_slots, nslots, d_absoff,
+                                            (const char*)pool_device, d_out,
+                                            d_ctr, d_opool, d_ctr + 1,
+                                            opool_cap, d_th, st));
+    }
+    unsigned long long hctr[2] = {0, 0};
+    HIP_CHECK(pool_alloc_host(&hgroups,
+                              sizeof(OutStrState) * (ngroups ? ngroups : 1)));
+    if (ngroups > 0) {
+        HIP_CHECK(hipMemcpyAsync(hgroups, d_out,
+                                 sizeof(OutStrState) * ngroups,
+                                 hipMemcpyDeviceToHost, st));
+    }
+    HIP_CHECK(hipMemcpyAsync(hctr, d_ctr, 2 * sizeof(unsigned long long),
+                             hipMemcpyDeviceToHost, st));
+    HIP_CHECK(hipStreamSynchronize(st));
+    if ((unsigned long long)hctr[1] > (unsigned long long)output->string_pool_capacity
+        || (output->string_pool == nullptr && hctr[1] != 0)) {
+        putall();
+        set_err(errbuf, errlen, "merge_str: string pool too small");
+        return YT_ERR_CAPACITY;
+    }
+    if (hctr[1]) {
+        HIP_CHECK(pool_alloc_host(&hpool, hctr[1]));
+        HIP_CHECK(hipMemcpyAsync(hpool, d_opool, hctr[1],
+                                 hipMemcpyDeviceToHost, st));
+        HIP_CHECK(hipStreamSynchronize(st));
+        memcpy(output->string_pool, hpool, hctr[1]);
+    }
+    output->string_pool_used = hctr[1];
+    int ncols = 1 + plan->agg_count;
+    for (int64_t g = 0; g < ngroups + has_null; g++) {
+        const int is_null_row = g >= ngroups;
+        uint64_t cnt, sum_bits, nonnull, off_len = 0;
+        if (is_null_row) {
+            cnt = th.side_cnt[1];
+            sum_bits = sum_slot >= 0 ? th.side_agg[1][2 * sum_slot] : 0;
+            nonnull = sum_slot >= 0 ? th.side_agg[1][2 * sum_slot + 1] : 0;
+        } else {
+            const OutStrState& gg = hgroups[g];
+            cnt = gg.cnt;
+            sum_bits = gg.sum_bits;
+            nonnull = gg.nonnull;
+            off_len = gg.off_len;
+        }
+        YtValue* dst = output->values + output->row_count * ncols;
+        dst[0].id = 0;
+        dst[0].flags = 0;
+        if (is_null_row) {
+            dst[0].type = YT_VT_NULL;
+            dst[0].length = 0;
+            dst[0].data.bits = 0;
+        } else {
+            dst[0].type = YT_VT_STRING;
+            dst[0].length = (uint32_t)(off_len & 0xFFFFFF);
+            dst[0].data.str = output->string_pool + (off_len >> 24);
+        }
+        for (int a = 0; a < plan->agg_count; a++) {
+            YtValue& v = dst[1 + a];
+            v.id = (uint16_t)(1 + a);
+            v.flags = 0;
+            v.length = 0;
+            if (plan->aggs[a]->func == YT_AGG_SUM1) {
+                v.type = YT_VT_INT64;
+                v.data.bits = cnt;
+            } else if (nonnull == 0) {
+                v.type = YT_VT_NULL;
+                v.data.bits = 0;
+            } else {
+                v.type = sum_is_double ? YT_VT_DOUBLE : YT_VT_INT64;
+                v.data.bits = sum_bits;
+            }
+        }
+        output->row_count++;
+    }
+    if (stats) {
+        stats->rows_written = output->row_count;
+        stats->grouped_row_count = ngroups + has_null;
+        stats->incomplete_output = (th.overflow == 2);
+        stats->execute_time_ms = now_ms() - tw0;
+    }
+    putall();
+    return YT_OK;
+    }
+fail:
+    putall();
+    return rc;
 }

@@ -3026,6 +3026,337 @@ __global__ void k_part_scatter(const OutGroup* groups, int64_t n, int nparts,
     }
 }
 
+/* ---- string-keyed two-phase exchange (§8e; shuffling_reader.cpp key
+ * shuffle applied to string group keys) ----
+ * The 32-byte YtStateRow is reused with key_bits = (byte offset within the
+ * state's own partition pool slice) << 24 | length; the pool slices travel
+ * beside the states in the all-to-all. Partition = splitmix64(FNV-1a(key
+ * bytes) [^ null marker]) % world — the byte-level mirror of
+ * partition_hash, matching the oracle bit-for-bit. */
+__device__ __forceinline__ uint64_t fnv1a_bytes(const char* p, uint32_t len)
+{
+    uint64_t h = 0xCBF29CE484222325ULL;
+    for (uint32_t k = 0; k < len; k++)
+        h = (h ^ (uint8_t)p[k]) * 0x100000001B3ULL;
+    return h;
+}
+
+__global__ void k_strst_count(const OutStrGroup* groups, int64_t n,
+                              const char* pool, int nparts,
+                              unsigned long long* counts,
+                              unsigned long long* byte_counts)
+{
+    __shared__ unsigned long long lc[kMaxParts], lb[kMaxParts];
+    for (int i = threadIdx.x; i < nparts; i += blockDim.x) { lc[i] = 0; lb[i] = 0; }
+    __syncthreads();
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         i < n; i += (int64_t)gridDim.x * blockDim.x) {
+        uint64_t ol = groups[i].off_len;
+        uint32_t len = (uint32_t)(ol & 0xFFFFFF);
+        uint64_t h = mix64(fnv1a_bytes(pool + (ol >> 24), len));
+        int p = (int)(h % (uint64_t)nparts);
+        atomicAdd(&lc[p], 1ULL);
+        atomicAdd(&lb[p], (unsigned long long)len);
+    }
+    __syncthreads();
+    for (int i = threadIdx.x; i < nparts; i += blockDim.x) {
+        if (lc[i]) atomicAdd(&counts[i], lc[i]);
+        if (lb[i]) atomicAdd(&byte_counts[i], lb[i]);
+    }
+}
+
+/* block-chunked scatter (rows AND bytes claimed per block, order within a
+ * partition arbitrary — the merge is order-free) */
+__global__ void k_strst_scatter(const OutStrGroup* groups, int64_t n,
+                                const char* pool, int nparts,
+                                int sum_is_double,
+                                unsigned long long* cursors,   /* global row idx */
+                                unsigned long long* bcursors,  /* slice-local bytes */
+                                YtStateRow* states, char* out_pool,
+                                const unsigned long long* pool_base)
+{
+    __shared__ unsigned lcnt[kMaxParts], lclaim[kMaxParts];
+    __shared__ unsigned long long lbytes[kMaxParts], lbclaim[kMaxParts];
+    __shared__ unsigned long long lbase[kMaxParts], lbbase[kMaxParts];
+    const int64_t chunk = (n + gridDim.x - 1) / gridDim.x;
+    const int64_t lo = (int64_t)blockIdx.x * chunk;
+    int64_t hi = lo + chunk;
+    if (hi > n) hi = n;
+    if (lo >= hi) return;
+    for (int i = threadIdx.x; i < nparts; i += blockDim.x) {
+        lcnt[i] = 0; lclaim[i] = 0; lbytes[i] = 0; lbclaim[i] = 0;
+    }
+    __syncthreads();
+    for (int64_t i = lo + threadIdx.x; i < hi; i += blockDim.x) {
+        uint64_t ol = groups[i].off_len;
+        uint32_t len = (uint32_t)(ol & 0xFFFFFF);
+        uint64_t h = mix64(fnv1a_bytes(pool + (ol >> 24), len));
+        int p = (int)(h % (uint64_t)nparts);
+        atomicAdd(&lcnt[p], 1u);
+        atomicAdd(&lbytes[p], (unsigned long long)len);
+    }
+    __syncthreads();
+    for (int i = threadIdx.x; i < nparts; i += blockDim.x) {
+        lbase[i] = lcnt[i] ? atomicAdd(&cursors[i], (unsigned long long)lcnt[i]) : 0;
+        lbbase[i] = lbytes[i] ? atomicAdd(&bcursors[i], lbytes[i]) : 0;
+    }
+    __syncthreads();
+    for (int64_t i = lo + threadIdx.x; i < hi; i += blockDim.x) {
+        const OutStrGroup& g = groups[i];
+        uint64_t ol = g.off_len;
+        uint32_t len = (uint32_t)(ol & 0xFFFFFF);
+        const char* src = pool + (ol >> 24);
+        uint64_t h = mix64(fnv1a_bytes(src, len));
+        int p = (int)(h % (uint64_t)nparts);
+        unsigned long long pos = lbase[p] + atomicAdd(&lclaim[p], 1u);
+        unsigned long long boff = lbbase[p] + atomicAdd(&lbclaim[p],
+                                                        (unsigned long long)len);
+        char* dst = out_pool + pool_base[p] + boff;
+        for (uint32_t k = 0; k < len; k++) dst[k] = src[k];
+        YtStateRow& sr = states[pos];
+        sr.key_bits = (boff << 24) | len;       /* slice-local reference */
+        uint64_t nonnull = (g.cnt_nonnull >> 32) ? 1 : 0;
+        sr.meta = (nonnull << 8) | (sum_is_double ? 2ULL : 0ULL);
+        sr.sum_bits = g.sum_bits;
+        sr.row_count = g.cnt_nonnull & 0xFFFFFFFFULL;
+    }
+}
+
+/* merge side: per-state hash/ident/prefix + absolute pool offsets (the
+ * received slices are concatenated; a state's slice is found by its row
+ * segment) */
+__global__ void k_strst_hash(const YtStateRow* states, int64_t n,
+                             const int64_t* seg_row_base,
+                             const unsigned long long* seg_pool_base,
+                             int nsegin, const char* pool,
+                             uint64_t* hashes, uint64_t* idents,
+                             ulonglong2* pfxs, uint64_t* absoff)
+{
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         i < n; i += (int64_t)gridDim.x * blockDim.x) {
+        int lo = 0, hi = nsegin;
+        while (lo + 1 < hi) {
+            int mid = (lo + hi) / 2;
+            if (seg_row_base[mid] <= i) lo = mid;
+            else hi = mid;
+        }
+        uint64_t ol = states[i].key_bits;
+        uint32_t len = (uint32_t)(ol & 0xFFFFFF);
+        uint64_t ab = seg_pool_base[lo] + (ol >> 24);
+        absoff[i] = (ab << 24) | len;
+        if (states[i].meta & 1) { idents[i] = 0; hashes[i] = 0; continue; }
+        uint64_t h = fnv1a_bytes(pool + ab, len);
+        hashes[i] = h;
+        idents[i] = ((h >> 16) << 16) | (uint64_t)(len & 0xFFFF);
+        ulonglong2 pf;
+        pf.x = 0; pf.y = 0;
+        uint32_t npfx = len < 16 ? len : 16;
+        const char* p = pool + ab;
+        for (uint32_t k = 0; k < npfx; k++) ((char*)&pf)[k] = p[k];
+        pfxs[i] = pf;
+    }
+}
+
+/* front-query merge over string states: StrSlot probe (rep = state index+1,
+ * identity from the streamed arrays, exact compare against the owner's pool
+ * bytes); counts are FULL u64 (slot->cnt = rows, slot->pad0_ = nonnull) */
+__global__ void k_strst_merge(const YtStateRow* states, int64_t n,
+                              const char* pool, const uint64_t* hashes,
+                              const uint64_t* idents, const ulonglong2* pfxs,
+                              const uint64_t* absoff, int sum_slot,
+                              StrSlot* slots, uint64_t nslots, TableHdr* th)
+{
+    uint64_t mask = nslots - 1;
+    __shared__ unsigned long long s_claims;
+    if (threadIdx.x == 0) s_claims = 0;
+    __syncthreads();
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         i < n; i += (int64_t)gridDim.x * blockDim.x) {
+        const YtStateRow sr = states[i];
+        if (sr.meta & 1) {
+            /* null-key state → side accumulator (k_merge_states pattern) */
+            th->side_used[1] = 1;
+            atomicAdd((unsigned long long*)&th->side_cnt[1],
+                      (unsigned long long)sr.row_count);
+            if (sum_slot >= 0 && (sr.meta >> 8)) {
+                if (sr.meta & 2)
+                    atomicAdd((double*)&th->side_agg[1][2 * sum_slot],
+                              __longlong_as_double((long long)sr.sum_bits));
+                else
+                    atomicAdd((unsigned long long*)&th->side_agg[1][2 * sum_slot],
+                              (unsigned long long)sr.sum_bits);
+                atomicAdd((unsigned long long*)&th->side_agg[1][2 * sum_slot + 1],
+                          (unsigned long long)(sr.meta >> 8));
+            }
+            continue;
+        }
+        const uint64_t my_ident = idents[i];
+        const uint32_t my_len = (uint32_t)(my_ident & 0xFFFF);
+        const ulonglong2 pf = pfxs[i];
+        const uint64_t h = hashes[i];
+        const char* my_p = pool + (absoff[i] >> 24);
+        unsigned long long rep = (unsigned long long)(i + 1);
+        StrSlot* slot = nullptr;
+        uint64_t sidx = mix64(h) & mask;
+        uint64_t max_probe = mask < 8192 ? mask : 8192;
+        for (uint64_t it = 0; it <= max_probe; it++) {
+            StrSlot* cand = &slots[sidx];
+            ulonglong4 v = *(const ulonglong4*)cand;
+            unsigned long long cur = v.x;
+            uint64_t oident = v.y;
+            if (cur == 0ULL) {
+                cur = atomicCAS(&cand->rep, 0ULL, rep);
+                if (cur == 0ULL) {
+                    __hip_atomic_store(&cand->pfx[0], pf.x,
+                                       __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+                    __hip_atomic_store(&cand->pfx[1], pf.y,
+                                       __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+                    __hip_atomic_store(&cand->ident, my_ident,
+                                       __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+                    atomicAdd(&s_claims, 1ULL);
+                    slot = cand;
+                    break;
+                }
+                oident = 0;
+            }
+            if (oident != 0 && oident != my_ident) {
+                sidx = (sidx + 1) & mask;
+                continue;
+            }
+            if (oident == my_ident && my_len <= 16
+                && v.z == pf.x && v.w == pf.y) { slot = cand; break; }
+            {
+                int64_t oi = (int64_t)cur - 1;
+                if (hashes[oi] == h) {
+                    uint64_t oab = absoff[oi];
+                    uint32_t olen = (uint32_t)(oab & 0xFFFFFF);
+                    uint64_t full_len = (uint32_t)(absoff[i] & 0xFFFFFF);
+                    if (olen == full_len) {
+                        const char* op = pool + (oab >> 24);
+                        uint32_t k = 0;
+                        while (k < olen && op[k] == my_p[k]) k++;
+                        if (k == olen) { slot = cand; break; }
+                    }
+                }
+            }
+            sidx = (sidx + 1) & mask;
+        }
+        if (!slot) { th->overflow = 1; continue; }
+        atomicAdd((unsigned long long*)&slot->cnt,
+                  (unsigned long long)sr.row_count);
+        if (sum_slot >= 0 && (sr.meta >> 8)) {
+            if (sr.meta & 2)
+                atomicAdd((double*)&slot->sum_bits,
+                          __longlong_as_double((long long)sr.sum_bits));
+            else
+                atomicAdd((unsigned long long*)&slot->sum_bits,
+                          (unsigned long long)sr.sum_bits);
+            atomicAdd((unsigned long long*)&slot->pad0_,
+                      (unsigned long long)(sr.meta >> 8));
+        }
+    }
+    __syncthreads();
+    if (threadIdx.x == 0 && s_claims) {
+        unsigned long long t = atomicAdd(&th->ngroups, s_claims);
+        if (th->group_limit > 0 && (int64_t)(t + s_claims) > th->group_limit)
+            th->overflow = 2;
+    }
+}
+
+/* compact the merged string-state table; strings come from the exchanged
+ * pool via the owner state's absolute offset (two-pass block compaction
+ * like k_strgrp_compact) */
+__global__ void k_strst_compact(const StrSlot* slots, uint64_t nslots,
+                                const uint64_t* absoff, const char* pool,
+                                OutStrState* out, unsigned long long* counter,
+                                char* out_pool, unsigned long long* pool_cursor,
+                                uint64_t pool_cap, TableHdr* th)
+{
+    const int lane = threadIdx.x & 63;
+    const int wid = (int)(threadIdx.x >> 6);
+    const int nw = (int)(blockDim.x >> 6);
+    __shared__ unsigned long long s_wcnt[16], s_wlen[16];
+    __shared__ unsigned long long s_cbase, s_pbase, s_crun, s_lrun;
+
+    uint64_t per = (nslots + gridDim.x - 1) / gridDim.x;
+    uint64_t b0 = (uint64_t)blockIdx.x * per;
+    if (b0 > nslots) b0 = nslots;
+    uint64_t b1 = b0 + per;
+    if (b1 > nslots) b1 = nslots;
+
+    unsigned long long c = 0, l = 0;
+    for (uint64_t i = b0 + threadIdx.x; i < b1; i += blockDim.x) {
+        unsigned long long rep = slots[i].rep;
+        if (rep) { c++; l += (uint32_t)(absoff[rep - 1] & 0xFFFFFF); }
+    }
+    #pragma unroll
+    for (int d = 32; d; d >>= 1) {
+        c += __shfl_down(c, d, 64);
+        l += __shfl_down(l, d, 64);
+    }
+    if (lane == 0) { s_wcnt[wid] = c; s_wlen[wid] = l; }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        unsigned long long tc = 0, tl = 0;
+        for (int w = 0; w < nw; w++) { tc += s_wcnt[w]; tl += s_wlen[w]; }
+        s_cbase = tc ? atomicAdd(counter, tc) : 0;
+        s_pbase = tl ? atomicAdd(pool_cursor, tl) : 0;
+        s_crun = 0;
+        s_lrun = 0;
+    }
+    __syncthreads();
+    const unsigned long long cbase = s_cbase, pbase = s_pbase;
+
+    for (uint64_t base = b0; base < b1; base += blockDim.x) {
+        uint64_t i = base + threadIdx.x;
+        bool occ = false;
+        uint32_t len = 0;
+        const char* p = nullptr;
+        const StrSlot* sl = nullptr;
+        if (i < b1) {
+            sl = &slots[i];
+            unsigned long long rep = sl->rep;
+            if (rep) {
+                occ = true;
+                uint64_t ab = absoff[rep - 1];
+                len = (uint32_t)(ab & 0xFFFFFF);
+                p = pool + (ab >> 24);
+            }
+        }
+        unsigned long long mask = __ballot(occ);
+        unsigned long long run = len;     /* inclusive wave scan of len */
+        #pragma unroll
+        for (int d = 1; d < 64; d <<= 1) {
+            unsigned long long v = __shfl_up(run, d, 64);
+            if (lane >= d) run += v;
+        }
+        if (lane == 63) { s_wcnt[wid] = __popcll(mask); s_wlen[wid] = run; }
+        __syncthreads();
+        if (occ) {
+            unsigned long long wc = 0, wl = 0;
+            for (int w = 0; w < wid; w++) { wc += s_wcnt[w]; wl += s_wlen[w]; }
+            unsigned long long off = pbase + s_lrun + wl + (run - len);
+            unsigned long long idx = cbase + s_crun + wc
+                + (unsigned long long)__popcll(mask & ((1ULL << lane) - 1));
+            if (off + len > pool_cap) {
+                th->overflow = 1;
+            } else {
+                for (uint32_t k = 0; k < len; k++) out_pool[off + k] = p[k];
+                OutStrState& g = out[idx];
+                g.off_len = (off << 24) | len;
+                g.sum_bits = sl->sum_bits;
+                g.cnt = sl->cnt;
+                g.nonnull = sl->pad0_;
+            }
+        }
+        __syncthreads();
+        if (threadIdx.x == 0) {
+            for (int w = 0; w < nw; w++) { s_crun += s_wcnt[w]; s_lrun += s_wlen[w]; }
+        }
+        __syncthreads();
+    }
+}
+
 /* merge state rows into a (fresh) table — front-query Merge semantics
  * (cg_fragment_compiler.cpp:4116-4134 + udf/sum.c sum_merge). */
 __global__ void k_merge_states(const YtStateRow* states, int64_t n,
@@ -3466,6 +3797,85 @@ hipError_t ytql_launch_strgrp_compact(const DevSeg* segs, const SegEx* segex,
     hipLaunchKernelGGL(k_strgrp_compact, dim3(grid), dim3(256), 0, st,
                        segs, segex, key_seg_off, slots, nslots, out, counter,
                        pool, pool_cursor, pool_cap, th);
+    return hipGetLastError();
+}
+
+hipError_t ytql_launch_strst_count(const OutStrGroup* groups, int64_t n,
+                                   const char* pool, int nparts,
+                                   unsigned long long* counts,
+                                   unsigned long long* byte_counts,
+                                   hipStream_t st)
+{
+    int64_t want = (n + 255) / 256;
+    int grid = (int)(want > 2048 ? 2048 : (want ? want : 1));
+    hipLaunchKernelGGL(k_strst_count, dim3(grid), dim3(256), 0, st,
+                       groups, n, pool, nparts, counts, byte_counts);
+    return hipGetLastError();
+}
+
+hipError_t ytql_launch_strst_scatter(const OutStrGroup* groups, int64_t n,
+                                     const char* pool, int nparts,
+                                     int sum_is_double,
+                                     unsigned long long* cursors,
+                                     unsigned long long* bcursors,
+                                     YtStateRow* states, char* out_pool,
+                                     const unsigned long long* pool_base,
+                                     hipStream_t st)
+{
+    int64_t want = (n + 255) / 256;
+    int grid = (int)(want > 2048 ? 2048 : (want ? want : 1));
+    hipLaunchKernelGGL(k_strst_scatter, dim3(grid), dim3(256), 0, st,
+                       groups, n, pool, nparts, sum_is_double, cursors,
+                       bcursors, states, out_pool, pool_base);
+    return hipGetLastError();
+}
+
+hipError_t ytql_launch_strst_hash(const YtStateRow* states, int64_t n,
+                                  const int64_t* seg_row_base,
+                                  const unsigned long long* seg_pool_base,
+                                  int nsegin, const char* pool,
+                                  uint64_t* hashes, uint64_t* idents,
+                                  ulonglong2* pfxs, uint64_t* absoff,
+                                  hipStream_t st)
+{
+    int64_t want = (n + 255) / 256;
+    int grid = (int)(want > 2048 ? 2048 : (want ? want : 1));
+    hipLaunchKernelGGL(k_strst_hash, dim3(grid), dim3(256), 0, st,
+                       states, n, seg_row_base, seg_pool_base, nsegin, pool,
+                       hashes, idents, pfxs, absoff);
+    return hipGetLastError();
+}
+
+hipError_t ytql_launch_strst_merge(const YtStateRow* states, int64_t n,
+                                   const char* pool, const uint64_t* hashes,
+                                   const uint64_t* idents,
+                                   const ulonglong2* pfxs,
+                                   const uint64_t* absoff, int sum_slot,
+                                   StrSlot* slots, uint64_t nslots,
+                                   TableHdr* th, hipStream_t st)
+{
+    int64_t want = (n + 255) / 256;
+    int grid = (int)(want > 4096 ? 4096 : (want ? want : 1));
+    hipLaunchKernelGGL(k_strst_merge, dim3(grid), dim3(256), 0, st,
+                       states, n, pool, hashes, idents, pfxs, absoff,
+                       sum_slot, slots, nslots, th);
+    return hipGetLastError();
+}
+
+hipError_t ytql_launch_strst_compact(const StrSlot* slots, uint64_t nslots,
+                                     const uint64_t* absoff, const char* pool,
+                                     OutStrState* out,
+                                     unsigned long long* counter,
+                                     char* out_pool,
+                                     unsigned long long* pool_cursor,
+                                     uint64_t pool_cap, TableHdr* th,
+                                     hipStream_t st)
+{
+    uint64_t want = (nslots + 255) / 256;
+    int grid = (int)(want > 2048 ? 2048 : (want ? want : 1));
+    hipLaunchKernelGGL(k_strst_compact, dim3(grid), dim3(256), 0, st,
+                       slots, nslots, absoff, pool, out, counter, out_pool,
+                       pool_cursor, pool_cap, th);
     return hipGetLastError();
 }
 
