@@ -322,6 +322,10 @@ def main():
         cap = 2 * key_space + 1024
         states_t = torch.zeros((cap, 4), dtype=torch.int64, device="cuda")
         recv_t = torch.zeros((2 * cap, 4), dtype=torch.int64, device="cuda")
+        if args.workload == "strgroup":
+            # string-key exchange: key-byte pool slices travel with the states
+            spool_t = torch.zeros(cap * 12, dtype=torch.uint8, device="cuda")
+            rpool_t = torch.zeros(2 * cap * 12, dtype=torch.uint8, device="cuda")
 
     scan_ms_total = 0.0
     scan_launches = 0
@@ -339,6 +343,8 @@ def main():
         # bottom query: partial aggregate + hash partition on device
         import time as _t
         t0 = _t.monotonic()
+        if args.workload == "strgroup":
+            return step_str()
         counts, st = y.gpu_partial(plan, dev_chunk, world,
                                    states_t.data_ptr(), cap,
                                    max_groups_hint=hint)
@@ -366,6 +372,55 @@ def main():
             log("2ph step: partial %.1fms a2a %.1fms merge %.1fms"
                 % ((t1 - t0) * 1e3, (t2 - t1) * 1e3, (_t.monotonic() - t2) * 1e3))
         return st
+
+    def step_str():
+        # string-keyed two-phase: exchange states AND key-byte pool slices
+        import time as _t
+        t0 = _t.monotonic()
+        counts, pbytes, st = y.gpu_partial_str(plan, dev_chunk, world,
+                                               states_t.data_ptr(), cap,
+                                               spool_t.data_ptr(),
+                                               spool_t.numel(),
+                                               max_groups_hint=hint)
+        scan_ms_total_add(st)
+        t1 = _t.monotonic()
+        sizes = torch.tensor(counts + pbytes, dtype=torch.int64, device="cuda")
+        rsizes = torch.zeros(2 * world, dtype=torch.int64, device="cuda")
+        dist.all_to_all_single(rsizes.view(world, 2),
+                               sizes.view(2, world).t().contiguous())
+        rs = rsizes.cpu().view(world, 2)
+        recv_rows = [int(x) for x in rs[:, 0]]
+        recv_bytes = [int(x) for x in rs[:, 1]]
+        dist.all_to_all_single(
+            recv_t[:sum(recv_rows)].view(-1, 4),
+            states_t[:sum(counts)].view(-1, 4),
+            output_split_sizes=recv_rows, input_split_sizes=counts)
+        dist.all_to_all_single(
+            rpool_t[:max(sum(recv_bytes), 1)],
+            spool_t[:max(sum(pbytes), 1)],
+            output_split_sizes=recv_bytes, input_split_sizes=pbytes)
+        t2 = _t.monotonic()
+        _, mst = y.gpu_merge_str(plan, recv_t.data_ptr(), recv_rows,
+                                 rpool_t.data_ptr(), recv_bytes,
+                                 col_types=col_types_of(),
+                                 max_groups_hint=hint,
+                                 out_capacity=key_space + 4096,
+                                 pool_capacity=key_space * 10 + (1 << 20))
+        if os.environ.get("YTQL_TIMING"):
+            log("2ph-str step: partial %.1fms a2a %.1fms merge %.1fms"
+                % ((t1 - t0) * 1e3, (t2 - t1) * 1e3,
+                   (_t.monotonic() - t2) * 1e3))
+        return st
+
+    def scan_ms_total_add(st):
+        nonlocal scan_ms_total, scan_launches
+        scan_ms_total += st.kernel_scan_ms
+        scan_launches += st.kernel_scan_launches
+        step.last = st
+
+    def col_types_of():
+        from ytsaurus_amd._abi import VT_STRING, VT_DOUBLE
+        return [VT_STRING, VT_DOUBLE]
 
     # warmup
     for _ in range(args.warmup):
