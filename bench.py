@@ -404,8 +404,7 @@ def main():
                                  rpool_t.data_ptr(), recv_bytes,
                                  col_types=col_types_of(),
                                  max_groups_hint=hint,
-                                 out_capacity=key_space + 4096,
-                                 pool_capacity=key_space * 10 + (1 << 20))
+                                 rowset=out_rs, raw_rowset=True)
         if os.environ.get("YTQL_TIMING"):
             log("2ph-str step: partial %.1fms a2a %.1fms merge %.1fms"
                 % ((t1 - t0) * 1e3, (t2 - t1) * 1e3,

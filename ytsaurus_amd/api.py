@@ -994,7 +994,8 @@ def gpu_partial_str(plan, device_chunk, nparts, states_dev_ptr, capacity_rows,
 
 def gpu_merge_str(plan, states_dev_ptr, seg_counts, pool_dev_ptr,
                   seg_pool_bytes, col_types=None, max_groups_hint=0,
-                  stream=0, out_capacity=None, pool_capacity=None):
+                  stream=0, out_capacity=None, pool_capacity=None,
+                  rowset=None, raw_rowset=False):
     opts = YtExecOptions(max_groups_hint=max_groups_hint, stream=stream)
     nseg = len(seg_counts)
     sc = (C.c_int64 * nseg)(*seg_counts)
@@ -1003,9 +1004,12 @@ def gpu_merge_str(plan, states_dev_ptr, seg_counts, pool_dev_ptr,
     if col_types is not None:
         ct = (C.c_uint8 * 8)(*([int(x) for x in col_types] + [0] * (8 - len(col_types))))
     total = sum(seg_counts)
-    cap = out_capacity or (total + 16)
-    pc = pool_capacity or max(sum(seg_pool_bytes), 1024)
-    rs = _mk_rowset(cap, pool_bytes=pc)
+    if rowset is not None:
+        rs = rowset
+    else:
+        cap = out_capacity or (total + 16)
+        pc = pool_capacity or max(sum(seg_pool_bytes), 1024)
+        rs = _mk_rowset(cap, pool_bytes=pc)
     st = YtStatistics()
     err = C.create_string_buffer(512)
     rc = _abi.gpu_lib().yt_gpu_merge_states_str(
@@ -1013,6 +1017,8 @@ def gpu_merge_str(plan, states_dev_ptr, seg_counts, pool_dev_ptr,
         C.c_void_p(pool_dev_ptr), sb, ct, C.byref(opts), C.byref(rs),
         C.byref(st), err, 512)
     _check(rc, err)
+    if raw_rowset:
+        return rs, st
     return rows_from_rowset(rs), st
 
 
