@@ -985,6 +985,23 @@ static int finish_group_rowset(const YtPlan* plan, YtRowset* out,
     int ncols = out->column_count;
     YtValue tot[16];
     int after = plan->totals_mode == 2;
+    /* all-null group keys are forbidden on the INTERMEDIATE stream (WITH
+     * TOTALS re-folds rows) and in the group-combined-with-order op —
+     * registry.cpp ValidateGroupKeyIsNotNull:1460-1476, call sites
+     * :1795,:1820; pinned by GroupByWithTotalsNulls ql_query_ut.cpp:3989 */
+    if (plan->with_totals || plan->order_count > 0) {
+        for (int64_t r = 0; r < out->row_count; r++) {
+            const YtValue* row = out->values + r * ncols;
+            int allnull = 1;
+            for (int k = 0; k < kc; k++)
+                if (row[k].type != YT_VT_NULL) { allnull = 0; break; }
+            if (allnull && kc > 0) {
+                set_err(errbuf, errlen, "Null values are forbidden in group key");
+                return YT_ERR_INVALID_PLAN;
+            }
+        }
+    }
+    const int64_t had_group_rows = out->row_count;
     if (plan->with_totals && !after) ord_fold_totals(plan, out, tot);
     if (plan->having) {
         uint64_t mask = 0;
@@ -1033,7 +1050,7 @@ static int finish_group_rowset(const YtPlan* plan, YtRowset* out,
         int rc = apply_order_rowset(plan, out, errbuf, errlen);
         if (rc != YT_OK) return rc;
     }
-    if (plan->with_totals) {
+    if (plan->with_totals && had_group_rows > 0) {
         if (out->row_count >= out->capacity_rows) return YT_ERR_CAPACITY;
         YtValue* dst = out->values + out->row_count * ncols;
         for (int k = 0; k < kc; k++) {

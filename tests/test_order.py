@@ -94,7 +94,7 @@ def test_oracle_group_order_by_sum():
 
 def test_oracle_order_string_group_keys():
     # group mode with string keys, ordered by key (memcmp + length tiebreak)
-    keys = ["b", "aa", "a", "ab", "b", "a", None, "aa"]
+    keys = ["b", "aa", "a", "ab", "b", "a", "c", "aa"]
     vals = list(range(8))
     n = len(keys)
     chunk = y.Chunk([y.encode_string(keys),
@@ -102,7 +102,20 @@ def test_oracle_order_string_group_keys():
     plan = y.Plan(keys=[y.col(0)], aggs=[y.agg_sum1()],
                   order_by=[(0, False)], limit=10)
     rows, _ = y.oracle_execute(plan, chunk)
-    assert [r[0] for r in rows] == [None, b"a", b"aa", b"ab", b"b"]
+    assert [r[0] for r in rows] == [b"a", b"aa", b"ab", b"b", b"c"]
+
+
+def test_oracle_order_group_null_key_forbidden():
+    # the group-combined-with-order op validates keys: an all-null group key
+    # is forbidden (registry.cpp ValidateGroupKeyIsNotNull:1460-1476,
+    # call site :1795)
+    keys = ["b", None, "a"]
+    chunk = y.Chunk([y.encode_string(keys),
+                     y.encode_int64(np.array([1, 2, 3], dtype=np.int64))], 3)
+    plan = y.Plan(keys=[y.col(0)], aggs=[y.agg_sum1()],
+                  order_by=[(0, False)], limit=10)
+    with pytest.raises(RuntimeError, match="forbidden in group key"):
+        y.oracle_execute(plan, chunk)
 
 
 def test_oracle_order_double_and_nan():

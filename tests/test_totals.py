@@ -42,8 +42,10 @@ def test_oracle_totals_requires_group_by():
 
 
 def test_oracle_totals_with_null_key_group():
-    # a null-key GROUP coexists with the totals row: both have null keys,
-    # the totals row is LAST (the stream-tag mirror)
+    # WITH TOTALS re-folds rows through the INTERMEDIATE stream, and the
+    # reference forbids all-null group keys there
+    # (registry.cpp ValidateGroupKeyIsNotNull:1460-1476; pinned by
+    # GroupByWithTotalsNulls ql_query_ut.cpp:3989-4016)
     rng = np.random.default_rng(53)
     n = 2000
     gk = rng.integers(0, 5, n, dtype=np.int64)
@@ -52,10 +54,8 @@ def test_oracle_totals_with_null_key_group():
     chunk = y.Chunk([y.encode_int64(gk, kn), y.encode_int64(v)], n)
     plan = y.Plan(keys=[y.col(0)], aggs=[y.agg_sum(y.col(1)), y.agg_sum1()],
                   with_totals=True)
-    rows, _ = y.oracle_execute(plan, chunk)
-    assert rows[-1] == (None, int(v.sum()), n)            # totals
-    nullgrp = [r for r in rows[:-1] if r[0] is None]
-    assert nullgrp == [(None, int(v[kn == 1].sum()), int(kn.sum()))]
+    with pytest.raises(RuntimeError, match="forbidden in group key"):
+        y.oracle_execute(plan, chunk)
 
 
 def test_oracle_totals_min_max_and_order():
@@ -90,15 +90,19 @@ def test_totals_gpu_order_and_nullkeys(cuda):
     rng = np.random.default_rng(56)
     n = 200_000
     gk = rng.integers(0, 500, n, dtype=np.int64)
-    kn = (rng.random(n) < 0.01).astype(np.uint8)
     v = rng.integers(0, 10**9, n, dtype=np.int64)
-    chunk = y.Chunk([y.encode_int64(gk, kn), y.encode_int64(v)], n)
+    chunk = y.Chunk([y.encode_int64(gk), y.encode_int64(v)], n)
     plan = y.Plan(keys=[y.col(0)], aggs=[y.agg_sum(y.col(1)), y.agg_sum1()],
                   order_by=[(1, True)], limit=10, with_totals=True)
     got, _ = y.gpu_execute(plan, chunk.c_device(cuda), max_groups_hint=4096)
     want, _ = y.oracle_execute(plan, chunk)
     assert got == want
     assert got[-1] == (None, int(v.sum()), n)
+    # a null group key under WITH TOTALS is forbidden on the GPU too
+    kn = (rng.random(n) < 0.01).astype(np.uint8)
+    chunk2 = y.Chunk([y.encode_int64(gk, kn), y.encode_int64(v)], n)
+    with pytest.raises(RuntimeError, match="forbidden in group key"):
+        y.gpu_execute(plan, chunk2.c_device(cuda), max_groups_hint=4096)
 
 
 @pytest.mark.gpu

@@ -1938,6 +1938,25 @@ static int finish_output(const YtPlan* plan, YtRowset* out,
 {
     std::vector<YtValue> tot;
     const bool after = plan->totals_mode == 2;
+    /* all-null group keys are forbidden on the INTERMEDIATE stream (WITH
+     * TOTALS re-folds rows) and in the group-combined-with-order op —
+     * registry.cpp ValidateGroupKeyIsNotNull:1460-1476, call sites
+     * :1795,:1820; pinned by GroupByWithTotalsNulls ql_query_ut.cpp:3989 */
+    if (plan->with_totals || plan->order_count > 0) {
+        const int kc0 = plan->key_count;
+        const int nc0 = out->column_count;
+        for (int64_t r = 0; r < out->row_count; r++) {
+            const YtValue* row = out->values + r * nc0;
+            bool allnull = kc0 > 0;
+            for (int k = 0; k < kc0; k++)
+                if (row[k].type != YT_VT_NULL) { allnull = false; break; }
+            if (allnull) {
+                set_err(errbuf, errlen, "Null values are forbidden in group key");
+                return YT_ERR_INVALID_PLAN;
+            }
+        }
+    }
+    const int64_t had_group_rows = out->row_count;
     if (plan->with_totals && !after) fold_totals_rows(plan, out, tot);
     if (plan->having) {
         int ncols = out->column_count;
@@ -1971,7 +1990,7 @@ static int finish_output(const YtPlan* plan, YtRowset* out,
         int rc = apply_order_host(plan, out, errbuf, errlen);
         if (rc) return rc;
     }
-    if (plan->with_totals) {
+    if (plan->with_totals && had_group_rows > 0) {
         int kc = plan->key_count, ac = plan->agg_count;
         int ncols = out->column_count;
         if (out->row_count >= out->capacity_rows) {
