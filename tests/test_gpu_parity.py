@@ -396,3 +396,19 @@ def test_string_group_by_gpu_int_sum(cuda):
     got, _ = y.gpu_execute(plan, chunk.c_device(cuda), max_groups_hint=1024)
     want, _ = y.oracle_execute(plan, chunk)
     assert y.sort_rows(got) == y.sort_rows(want)
+
+
+@pytest.mark.gpu
+def test_scan_project_streams_past_16M(cuda):
+    """the per-row materialization buffer is windowed (16Mi rows): larger
+    inputs stream through bounded memory (a selective filter keeps the
+    OUTPUT small while the INPUT spans two windows)"""
+    n = 17_000_000
+    a = np.arange(n, dtype=np.int64)
+    chunk = y.Chunk([y.encode_int64(a)], n)
+    plan = y.Plan(filter=(y.col(0) % 1_000_000) == 7,
+                  projects=[y.col(0), y.col(0) * 2])
+    got, st = y.gpu_execute(plan, chunk.c_device(cuda), out_capacity=64)
+    want = [(int(v), int(v) * 2) for v in range(7, n, 1_000_000)]
+    assert got == want
+    assert st.rows_read == n

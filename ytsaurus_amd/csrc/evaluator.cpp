@@ -92,7 +92,7 @@ hipError_t ytql_launch_join_chain(const JoinDev*, int64_t, unsigned*,
                                    hipStream_t);
 hipError_t ytql_launch_scan_project(const DevPlan*, const DevSeg*, const SegEx*,
                                     const int32_t*, const int32_t*, int64_t,
-                                    const JoinDev*,
+                                    int64_t, const JoinDev*,
                                     DevOutVal*, uint8_t*, unsigned*, hipStream_t);
 hipError_t ytql_launch_scan_generic(const DevPlan*, const DevSeg*, const SegEx*,
                                     const int32_t*, const int32_t*, int64_t,
@@ -2421,30 +2421,40 @@ static int run_scan_project(const YtPlan* plan, const YtChunk* chunk,
     output->column_count = plan->project_count;
     if (n == 0 || R2.nsegs == 0) return YT_OK;
     {
+        /* stream the scan through bounded windows: the per-row
+         * materialization buffer is at most kWin rows, so arbitrarily
+         * large inputs need only bounded device/host memory */
+        const int64_t kWin = (int64_t)1 << 24;
+        const int64_t win = n < kWin ? n : kWin;
         DevOutVal* d_out = nullptr;
         uint8_t* d_pass = nullptr;
         DevOutVal* h_out = nullptr;
         uint8_t* h_pass = nullptr;
-        HIP_CHECK(pool_alloc(&d_out, sizeof(DevOutVal) * n * plan->project_count));
-        HIP_CHECK(pool_alloc(&d_pass, (size_t)n));
+        HIP_CHECK(pool_alloc(&d_out, sizeof(DevOutVal) * win * plan->project_count));
+        HIP_CHECK(pool_alloc(&d_pass, (size_t)win));
+        HIP_CHECK(pool_alloc_host(&h_out, sizeof(DevOutVal) * win * plan->project_count));
+        HIP_CHECK(pool_alloc_host(&h_pass, (size_t)win));
         HIP_CHECK(hipMemsetAsync(R2.d_err, 0, sizeof(unsigned), R2.stream));
+        float ms = 0;
+        int np = plan->project_count;
+        for (int64_t w0 = 0; w0 < n; w0 += win) {
+        const int64_t wlen = (w0 + win <= n) ? win : (n - w0);
         hipEvent_t e0, e1;
         HIP_CHECK(hipEventCreate(&e0));
         HIP_CHECK(hipEventCreate(&e1));
         HIP_CHECK(hipEventRecord(e0, R2.stream));
         HIP_CHECK(ytql_launch_scan_project(dp, R2.d_segs, R2.d_segex, R2.d_off,
-                                           R2.d_cnt, n, jd, d_out, d_pass, R2.d_err,
-                                           R2.stream));
+                                           R2.d_cnt, w0, wlen, jd, d_out,
+                                           d_pass, R2.d_err, R2.stream));
         HIP_CHECK(hipEventRecord(e1, R2.stream));
-        HIP_CHECK(pool_alloc_host(&h_out, sizeof(DevOutVal) * n * plan->project_count));
-        HIP_CHECK(pool_alloc_host(&h_pass, (size_t)n));
-        HIP_CHECK(hipMemcpyAsync(h_out, d_out, sizeof(DevOutVal) * n * plan->project_count,
+        HIP_CHECK(hipMemcpyAsync(h_out, d_out, sizeof(DevOutVal) * wlen * plan->project_count,
                                  hipMemcpyDeviceToHost, R2.stream));
-        HIP_CHECK(hipMemcpyAsync(h_pass, d_pass, (size_t)n,
+        HIP_CHECK(hipMemcpyAsync(h_pass, d_pass, (size_t)wlen,
                                  hipMemcpyDeviceToHost, R2.stream));
         HIP_CHECK(hipStreamSynchronize(R2.stream));
-        float ms = 0;
-        HIP_CHECK(hipEventElapsedTime(&ms, e0, e1));
+        float wms = 0;
+        HIP_CHECK(hipEventElapsedTime(&wms, e0, e1));
+        ms += wms;
         (void)hipEventDestroy(e0);
         (void)hipEventDestroy(e1);
         unsigned kerr = 0;
@@ -2455,8 +2465,7 @@ static int run_scan_project(const YtPlan* plan, const YtChunk* chunk,
             set_err(errbuf, errlen, kerr == YT_ERR_DIV_ZERO ? "Division by zero" : "expression error");
             return (int)kerr;
         }
-        int np = plan->project_count;
-        for (int64_t r2 = 0; r2 < n; r2++) {
+        for (int64_t r2 = 0; r2 < wlen; r2++) {
             if (!h_pass[r2]) continue;
             if (options->output_row_limit > 0 &&
                 output->row_count >= options->output_row_limit) {
@@ -2478,6 +2487,10 @@ static int run_scan_project(const YtPlan* plan, const YtChunk* chunk,
                 dst[pj].data.bits = o.bits;
             }
             output->row_count++;
+        }
+        if (options->output_row_limit > 0 &&
+            output->row_count >= options->output_row_limit)
+            break;
         }
         if (stats) {
             stats->rows_read = n;
@@ -3399,12 +3412,6 @@ extern "C" int yt_gpu_query_execute(
                     "WITH TOTALS / HAVING with OutputRowLimit: not this round");
             return YT_ERR_UNSUPPORTED;
         }
-    }
-    if (plan->agg_count == 0 && plan->order_count == 0 &&
-        chunk->row_count > (int64_t)1 << 24) {
-        set_err(errbuf, errlen,
-                "GPU scan-project materializes per-row output; capped at 16M rows this round");
-        return YT_ERR_UNSUPPORTED;
     }
 
     {

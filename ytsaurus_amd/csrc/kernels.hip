@@ -1280,9 +1280,12 @@ k_scan_generic(DevPlan p, const DevSeg* segs, const SegEx* segex,
 __global__ void __launch_bounds__(256)
 k_scan_project(DevPlan p, const DevSeg* segs, const SegEx* segex,
                const int32_t* col_seg_off, const int32_t* col_seg_cnt,
-               int64_t row_count, JoinDev jd, DevOutVal* out, uint8_t* pass,
-               unsigned* error_out)
+               int64_t row_base, int64_t row_count, JoinDev jd,
+               DevOutVal* out, uint8_t* pass, unsigned* error_out)
 {
+    /* [row_base, row_base+row_count) window: out/pass are window-local so
+     * the host can STREAM arbitrarily large scans through a bounded
+     * materialization buffer */
     ColCtx c;
     c.segs = segs;
     c.segex = segex;
@@ -1294,7 +1297,7 @@ k_scan_project(DevPlan p, const DevSeg* segs, const SegEx* segex,
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
          r < row_count; r += stride) {
-        c.row = r;
+        c.row = row_base + r;
         pass[r] = 0;
         if (!join_row_ok(p, c)) continue;
         if (p.filter_len) {
@@ -3661,6 +3664,7 @@ hipError_t ytql_launch_scan_project(const DevPlan* p, const DevSeg* segs,
                                     const SegEx* segex,
                                     const int32_t* col_seg_off,
                                     const int32_t* col_seg_cnt,
+                                    int64_t row_base,
                                     int64_t row_count, const JoinDev* jd,
                                     DevOutVal* out,
                                     uint8_t* pass, unsigned* error_out,
@@ -3670,8 +3674,8 @@ hipError_t ytql_launch_scan_project(const DevPlan* p, const DevSeg* segs,
     int64_t want = (row_count + block - 1) / block;
     int grid = (int)(want > 4096 ? 4096 : (want > 0 ? want : 1));
     hipLaunchKernelGGL(k_scan_project, dim3(grid), dim3(block), 0, st,
-                       *p, segs, segex, col_seg_off, col_seg_cnt, row_count,
-                       *jd, out, pass, error_out);
+                       *p, segs, segex, col_seg_off, col_seg_cnt, row_base,
+                       row_count, *jd, out, pass, error_out);
     return hipGetLastError();
 }
 
