@@ -204,11 +204,21 @@ typedef struct YtPlan {
 
 typedef struct YtJoin {
     const YtChunk* foreign;          /* encoded foreign rowset */
-    int32_t primary_key_col;         /* equality key column in the primary chunk */
+    int32_t primary_key_col;         /* equality key column: an index into the
+                                        row AS EXTENDED SO FAR — primary
+                                        columns plus every earlier join item's
+                                        appended values — so a later item can
+                                        key on an earlier item's output
+                                        (snowflake chains, the reference's
+                                        TMultiJoinParameters item list,
+                                        registry.cpp MultiJoinOpHelper) */
     int32_t foreign_key_col;         /* equality key column in `foreign` */
     int32_t foreign_value_count;
     const int32_t* foreign_value_cols;  /* foreign columns appended to the row */
     int32_t is_left;                 /* 0 = INNER, 1 = LEFT */
+    const struct YtJoin* next;       /* next join item (2 items max this
+                                        round; duplicate foreign keys only on
+                                        the FIRST item) */
 } YtJoin;
 
 /* ---- execution context / statistics ----

@@ -1093,7 +1093,13 @@ int yto_execute(const YtPlan* plan, const YtChunk* chunk,
     }
     int ncols = chunk->column_count;
     int jF = plan->join ? plan->join->foreign_value_count : 0;
-    int ncols_eff = ncols + jF;
+    const YtJoin* J2chain = plan->join ? plan->join->next : NULL;
+    int jF2 = J2chain ? J2chain->foreign_value_count : 0;
+    if (J2chain && J2chain->next) {
+        set_err(errbuf, errlen, "join: more than two join items not this round");
+        return YT_ERR_UNSUPPORTED;
+    }
+    int ncols_eff = ncols + jF + jF2;
     uint8_t* jdrop = NULL;
     int64_t n = chunk->row_count;
     int64_t n_scan = chunk->row_count;   /* expanded row count under dup joins */
@@ -1294,6 +1300,119 @@ int yto_execute(const YtPlan* plan, const YtChunk* chunk,
         free(fkey); free(fknull);
         for (int j = 0; j < jF; j++) { if (fvals) free(fvals[j]); if (fnulls) free(fnulls[j]); }
         free(fvals); free(fnulls); free(hrow); free(hkey); free(hused); free(hnext);
+        if (rc != YT_OK) goto done;
+    }
+
+    /* second join item (snowflake chain): joins the row AS EXTENDED by item
+     * 0 — its key may be one of item 0's appended columns. UNIQUE foreign
+     * keys only (the cross-product machinery binds item 0). Columns
+     * [ncols+jF, ncols_eff) are indexed by SCAN row like item 0's. */
+    if (J2chain) {
+        const YtJoin* J2 = J2chain;
+        const YtChunk* fc = J2->foreign;
+        int base2 = ncols + jF;
+        int kp = J2->primary_key_col;
+        if (kp < 0 || kp >= base2) {
+            set_err(errbuf, errlen, "join: key column out of range");
+            rc = YT_ERR_INVALID_PLAN;
+            goto done;
+        }
+        if (types[kp] == YT_VT_STRING || types[kp] == YT_VT_DOUBLE) {
+            set_err(errbuf, errlen,
+                    "join: int64/uint64/boolean key columns this round");
+            rc = YT_ERR_UNSUPPORTED;
+            goto done;
+        }
+        int64_t fn = fc->row_count;
+        int64_t* fkey = malloc(sizeof(int64_t) * (fn ? fn : 1));
+        uint8_t* fknull = malloc(fn ? fn : 1);
+        rc = yto_decode_column(&fc->columns[J2->foreign_key_col], fn, fkey, fknull);
+        if (rc != YT_OK) { free(fkey); free(fknull);
+            set_err(errbuf, errlen, "join: bad foreign key segment"); goto done; }
+        int64_t** fvals = calloc(jF2, sizeof(int64_t*));
+        uint8_t** fnulls = calloc(jF2, sizeof(uint8_t*));
+        for (int j = 0; j < jF2 && rc == YT_OK; j++) {
+            int cjf = J2->foreign_value_cols[j];
+            if (fc->columns[cjf].value_type == YT_VT_STRING) {
+                set_err(errbuf, errlen, "join: string foreign values not this round");
+                rc = YT_ERR_UNSUPPORTED;
+                break;
+            }
+            fvals[j] = malloc(sizeof(int64_t) * (fn ? fn : 1));
+            fnulls[j] = malloc(fn ? fn : 1);
+            rc = yto_decode_column(&fc->columns[cjf], fn, fvals[j], fnulls[j]);
+        }
+        uint64_t cap = 2048;
+        while (cap < (uint64_t)fn * 2) cap <<= 1;
+        int64_t* hrow = NULL;
+        uint64_t* hkey = NULL;
+        uint8_t* hused = NULL;
+        int64_t null_row = -1;
+        if (rc == YT_OK) {
+            hrow = malloc(sizeof(int64_t) * cap);
+            hkey = malloc(sizeof(uint64_t) * cap);
+            hused = calloc(cap, 1);
+            for (int64_t r2 = 0; r2 < fn && rc == YT_OK; r2++) {
+                if (fknull[r2]) {
+                    if (null_row >= 0) {
+                        set_err(errbuf, errlen,
+                                "join: duplicate foreign keys on a non-first "
+                                "join item not this round");
+                        rc = YT_ERR_UNSUPPORTED;
+                    }
+                    null_row = r2;
+                    continue;
+                }
+                uint64_t h = splitmix64((uint64_t)fkey[r2]) & (cap - 1);
+                for (;;) {
+                    if (!hused[h]) { hused[h] = 1; hkey[h] = (uint64_t)fkey[r2]; hrow[h] = r2; break; }
+                    if (hkey[h] == (uint64_t)fkey[r2]) {
+                        set_err(errbuf, errlen,
+                                "join: duplicate foreign keys on a non-first "
+                                "join item not this round");
+                        rc = YT_ERR_UNSUPPORTED;
+                        break;
+                    }
+                    h = (h + 1) & (cap - 1);
+                }
+            }
+        }
+        if (rc == YT_OK) {
+            for (int j = 0; j < jF2; j++) {
+                vals[base2 + j] = calloc(n_scan ? n_scan : 1, sizeof(int64_t));
+                nulls[base2 + j] = malloc(n_scan ? n_scan : 1);
+                memset(nulls[base2 + j], 1, n_scan ? n_scan : 1);
+                types[base2 + j] = (uint8_t)fc->columns[J2->foreign_value_cols[j]].value_type;
+            }
+            if (!jdrop) jdrop = calloc(n_scan ? n_scan : 1, 1);
+            for (int64_t e2 = 0; e2 < n_scan; e2++) {
+                if (jdrop[e2]) continue;
+                int64_t kr = (join_rowmap && kp < ncols) ? join_rowmap[e2] : e2;
+                int64_t frow = -1;
+                if (nulls[kp][kr]) {
+                    frow = null_row;
+                } else {
+                    uint64_t h = splitmix64((uint64_t)vals[kp][kr]) & (cap - 1);
+                    while (hused[h]) {
+                        if (hkey[h] == (uint64_t)vals[kp][kr]) { frow = hrow[h]; break; }
+                        h = (h + 1) & (cap - 1);
+                    }
+                }
+                if (frow < 0) {
+                    if (!J2->is_left) jdrop[e2] = 1;
+                    continue;
+                }
+                for (int j = 0; j < jF2; j++) {
+                    if (!fnulls[j][frow]) {
+                        vals[base2 + j][e2] = fvals[j][frow];
+                        nulls[base2 + j][e2] = 0;
+                    }
+                }
+            }
+        }
+        free(fkey); free(fknull);
+        for (int j = 0; j < jF2; j++) { if (fvals) free(fvals[j]); if (fnulls) free(fnulls[j]); }
+        free(fvals); free(fnulls); free(hrow); free(hkey); free(hused);
         if (rc != YT_OK) goto done;
     }
 

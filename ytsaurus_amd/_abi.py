@@ -113,12 +113,16 @@ class YtPlan(C.Structure):
 
 
 class YtJoin(C.Structure):
-    _fields_ = [("foreign", C.POINTER(YtChunk)),
-                ("primary_key_col", C.c_int32),
-                ("foreign_key_col", C.c_int32),
-                ("foreign_value_count", C.c_int32),
-                ("foreign_value_cols", C.POINTER(C.c_int32)),
-                ("is_left", C.c_int32)]
+    pass
+
+
+YtJoin._fields_ = [("foreign", C.POINTER(YtChunk)),
+                   ("primary_key_col", C.c_int32),
+                   ("foreign_key_col", C.c_int32),
+                   ("foreign_value_count", C.c_int32),
+                   ("foreign_value_cols", C.POINTER(C.c_int32)),
+                   ("is_left", C.c_int32),
+                   ("next", C.POINTER(YtJoin))]
 
 
 class YtExecOptions(C.Structure):

@@ -719,15 +719,49 @@ def _mk_rowset(capacity, ncols_max=8, pool_bytes=0):
 
 # ---------------- execution ----------------
 
+def _join_items(plan):
+    j = getattr(plan, "join", None)
+    if j is None:
+        return []
+    return list(j) if isinstance(j, (list, tuple)) else [j]
+
+
 def _attach_join(plan, cchunk_factory):
-    """Point plan.c.join at a YtJoin built over the given chunk flavor;
-    returns the keep-alive object (None when the plan has no join)."""
-    if getattr(plan, "join", None) is None:
+    """Point plan.c.join at the YtJoin chain built over the given chunk
+    flavor; returns the keep-alive object (None when the plan has no
+    join). plan.join may be a Join or a list of Joins (2 items max —
+    snowflake chains, later items may key on earlier items\' columns)."""
+    items = _join_items(plan)
+    if not items:
         plan.c.join = None
         return None
-    j = plan.join.c_struct(cchunk_factory(plan.join.chunk))
-    plan.c.join = C.cast(C.pointer(j), C.c_void_p)
-    return j
+    structs = [it.c_struct(cchunk_factory(it.chunk)) for it in items]
+    for i in range(len(structs) - 1):
+        structs[i].next = C.pointer(structs[i + 1])
+    for s in structs[1:]:
+        structs[0]._keep = (structs[0]._keep, s)
+    plan.c.join = C.cast(C.pointer(structs[0]), C.c_void_p)
+    return structs
+
+
+def _attach_join_dev(plan, join_foreign):
+    """GPU flavor: join_foreign is the device YtChunk (or list of them,
+    matching the plan\'s join list)."""
+    items = _join_items(plan)
+    if not items:
+        plan.c.join = None
+        return None
+    assert join_foreign is not None, "plan has a join: pass join_foreign="
+    devs = join_foreign if isinstance(join_foreign, (list, tuple)) \
+        else [join_foreign]
+    assert len(devs) == len(items), "one device chunk per join item"
+    structs = [it.c_struct(devs[i]) for i, it in enumerate(items)]
+    for i in range(len(structs) - 1):
+        structs[i].next = C.pointer(structs[i + 1])
+    for s in structs[1:]:
+        structs[0]._keep = (structs[0]._keep, s)
+    plan.c.join = C.cast(C.pointer(structs[0]), C.c_void_p)
+    return structs
 
 
 def oracle_execute(plan, chunk, nthreads=1, expect_error=False):
@@ -880,13 +914,7 @@ def gpu_execute(plan, device_chunk, max_groups_hint=0, group_row_limit=0,
     allocation; raw_rowset=True skips the Python row conversion.
     join_foreign: device YtChunk for plan.join's foreign rowset (required
     when the plan has a join)."""
-    _j = None
-    if getattr(plan, "join", None) is not None:
-        assert join_foreign is not None, "plan has a join: pass join_foreign="
-        _j = plan.join.c_struct(join_foreign)
-        plan.c.join = C.cast(C.pointer(_j), C.c_void_p)
-    else:
-        plan.c.join = None
+    _j = _attach_join_dev(plan, join_foreign)
     opts = YtExecOptions(max_groups_hint=max_groups_hint,
                          group_row_limit=group_row_limit, stream=stream)
     if rowset is not None:
@@ -911,13 +939,7 @@ def make_rowset(capacity, ncols, pool_bytes=0):
 
 def gpu_partial(plan, device_chunk, nparts, states_dev_ptr, capacity_rows,
                 max_groups_hint=0, stream=0, join_foreign=None):
-    _j = None
-    if getattr(plan, "join", None) is not None:
-        assert join_foreign is not None, "plan has a join: pass join_foreign="
-        _j = plan.join.c_struct(join_foreign)
-        plan.c.join = C.cast(C.pointer(_j), C.c_void_p)
-    else:
-        plan.c.join = None
+    _j = _attach_join_dev(plan, join_foreign)
     opts = YtExecOptions(max_groups_hint=max_groups_hint, stream=stream)
     counts = (C.c_int64 * nparts)()
     st = YtStatistics()

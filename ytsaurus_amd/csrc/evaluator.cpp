@@ -47,12 +47,12 @@ hipError_t ytql_launch_bucket_agg_direct(const void*, const unsigned long long*,
                                          uint64_t, uint64_t, int, int, hipStream_t);
 hipError_t ytql_launch_topk_hist(const DevPlan*, const DevSeg*, const SegEx*,
                                  const int32_t*, const int32_t*, int64_t,
-                                 const JoinDev*,
+                                 const JoinDev*, const JoinDev*,
                                  const TopkPass*, unsigned long long*,
                                  unsigned long long*, unsigned*, hipStream_t);
 hipError_t ytql_launch_topk_gather(const DevPlan*, const DevSeg*, const SegEx*,
                                    const int32_t*, const int32_t*, int64_t,
-                                   const JoinDev*,
+                                   const JoinDev*, const JoinDev*,
                                    const TopkGather*,
                                    int64_t*, unsigned long long*,
                                    int64_t*, unsigned long long*,
@@ -60,7 +60,7 @@ hipError_t ytql_launch_topk_gather(const DevPlan*, const DevSeg*, const SegEx*,
                                    unsigned*, hipStream_t);
 hipError_t ytql_launch_topk_materialize(const DevPlan*, const DevSeg*, const SegEx*,
                                         const int32_t*, const int32_t*,
-                                        const JoinDev*,
+                                        const JoinDev*, const JoinDev*,
                                         const int64_t*, int64_t, DevOutVal*,
                                         unsigned*, hipStream_t);
 hipError_t ytql_launch_topk_hist_fast(const DevSeg*, const SegEx*, int, int,
@@ -92,11 +92,11 @@ hipError_t ytql_launch_join_chain(const JoinDev*, int64_t, unsigned*,
                                    hipStream_t);
 hipError_t ytql_launch_scan_project(const DevPlan*, const DevSeg*, const SegEx*,
                                     const int32_t*, const int32_t*, int64_t,
-                                    int64_t, const JoinDev*,
+                                    int64_t, const JoinDev*, const JoinDev*,
                                     DevOutVal*, uint8_t*, unsigned*, hipStream_t);
 hipError_t ytql_launch_scan_generic(const DevPlan*, const DevSeg*, const SegEx*,
                                     const int32_t*, const int32_t*, int64_t,
-                                    const JoinDev*,
+                                    const JoinDev*, const JoinDev*,
                                     TableHdr*, unsigned long long*, unsigned*, hipStream_t);
 hipError_t ytql_launch_scan_fast(const FastParams*, const DevSeg*, const SegEx*,
                                  const FastCol*, TableHdr*, unsigned long long*,
@@ -358,15 +358,16 @@ static int build_devplan(const YtPlan* plan, const YtChunk* chunk, DevPlan* p,
 {
     memset(p, 0, sizeof(*p));
     p->ncols = chunk->column_count;
-    if (plan->join) p->ncols += plan->join->foreign_value_count;
+    for (const YtJoin* J = plan->join; J; J = J->next)
+        p->ncols += J->foreign_value_count;
     if (p->ncols > kMaxCols) { set_err(errbuf, errlen, "too many columns"); return YT_ERR_UNSUPPORTED; }
     for (int c = 0; c < chunk->column_count; c++) p->col_types[c] = (uint8_t)chunk->columns[c].value_type;
-    if (plan->join) {
-        const YtJoin* J = plan->join;
-        for (int j = 0; j < J->foreign_value_count; j++) {
-            p->col_types[chunk->column_count + j] =
-                (uint8_t)J->foreign->columns[J->foreign_value_cols[j]].value_type;
-        }
+    {
+        int at = chunk->column_count;
+        for (const YtJoin* J = plan->join; J; J = J->next)
+            for (int j = 0; j < J->foreign_value_count; j++)
+                p->col_types[at++] =
+                    (uint8_t)J->foreign->columns[J->foreign_value_cols[j]].value_type;
     }
     for (int c = 0; c < chunk->column_count; c++) {
         const YtColumn& col = chunk->columns[c];
@@ -1241,6 +1242,7 @@ fail:
 static int run_scan(const YtPlan* plan, const YtChunk* chunk,
                     const YtExecOptions* options, DeviceRun* R,
                     const DevPlan* dp, const JoinDev* jd,
+                    const JoinDev* jd2,
                     const FastShape* fs, unsigned maxw,
                     YtStatistics* stats, char* errbuf, size_t errlen)
 {
@@ -1342,7 +1344,7 @@ static int run_scan(const YtPlan* plan, const YtChunk* chunk,
         if (options->input_row_limit > 0 && options->input_row_limit < gen_rows)
             gen_rows = options->input_row_limit;
         HIP_CHECK(ytql_launch_scan_generic(dp, R->d_segs, R->d_segex, R->d_off, R->d_cnt,
-                                           gen_rows, jd, R->d_th, R->d_slots,
+                                           gen_rows, jd, jd2, R->d_th, R->d_slots,
                                            R->d_err, R->stream));
         HIP_CHECK(hipEventRecord(ev1, R->stream));
     }
@@ -1773,13 +1775,16 @@ struct JoinRun {
     }
 };
 
-static int setup_join(const YtPlan* plan, const YtChunk* chunk, JoinRun* JR,
-                      hipStream_t stream, char* errbuf, size_t errlen)
+/* one join item: the primary side is the row AS EXTENDED SO FAR
+ * (row_types/row_ncols — primary columns plus earlier items' values, so a
+ * later item can key on an earlier item's output). */
+static int setup_join_item(const YtJoin* J, const uint8_t* row_types,
+                           int row_ncols, JoinRun* JR,
+                           hipStream_t stream, char* errbuf, size_t errlen)
 {
     int rc = YT_OK;
-    const YtJoin* J = plan->join;
     const YtChunk* fc = J->foreign;
-    if (J->primary_key_col < 0 || J->primary_key_col >= chunk->column_count ||
+    if (J->primary_key_col < 0 || J->primary_key_col >= row_ncols ||
         J->foreign_key_col < 0 || J->foreign_key_col >= fc->column_count) {
         set_err(errbuf, errlen, "join: key column out of range");
         return YT_ERR_INVALID_PLAN;
@@ -1787,7 +1792,7 @@ static int setup_join(const YtPlan* plan, const YtChunk* chunk, JoinRun* JR,
     auto intish = [](int vt) {
         return vt == YT_VT_INT64 || vt == YT_VT_UINT64 || vt == YT_VT_BOOLEAN;
     };
-    if (!intish(chunk->columns[J->primary_key_col].value_type) ||
+    if (!intish(row_types[J->primary_key_col]) ||
         !intish(fc->columns[J->foreign_key_col].value_type)) {
         set_err(errbuf, errlen,
                 "join: int64/uint64/boolean key columns this round");
@@ -1830,7 +1835,7 @@ static int setup_join(const YtPlan* plan, const YtChunk* chunk, JoinRun* JR,
     jd.active = 1;
     jd.is_left = J->is_left ? 1 : 0;
     jd.pkey_col = J->primary_key_col;
-    jd.primary_ncols = chunk->column_count;
+    jd.primary_ncols = row_ncols;
     jd.fkey_col = J->foreign_key_col;
     jd.fkey_shift = column_uniform_shift(fc->columns[J->foreign_key_col]);
     for (int j = 0; j < J->foreign_value_count; j++) {
@@ -1875,6 +1880,46 @@ static int setup_join(const YtPlan* plan, const YtChunk* chunk, JoinRun* JR,
     return YT_OK;
 fail:
     return rc;
+}
+
+
+/* drive the join chain (2 items max this round; duplicate foreign keys on
+ * the FIRST item only — the cross-product machinery binds item 0) */
+static int setup_join_chain(const YtPlan* plan, const YtChunk* chunk,
+                            JoinRun* JR, JoinDev const** jd0,
+                            JoinDev const** jd1,
+                            hipStream_t stream, char* errbuf, size_t errlen)
+{
+    static const JoinDev kNone = {};
+    *jd0 = &kNone;
+    *jd1 = &kNone;
+    uint8_t rt[kMaxCols];
+    int rn = chunk->column_count;
+    if (rn > kMaxCols) { set_err(errbuf, errlen, "too many columns"); return YT_ERR_UNSUPPORTED; }
+    for (int c = 0; c < rn; c++) rt[c] = (uint8_t)chunk->columns[c].value_type;
+    int nj = 0;
+    for (const YtJoin* J = plan->join; J; J = J->next) {
+        if (nj >= 2) {
+            set_err(errbuf, errlen, "join: more than two join items not this round");
+            return YT_ERR_UNSUPPORTED;
+        }
+        int rc = setup_join_item(J, rt, rn, &JR[nj], stream, errbuf, errlen);
+        if (rc != YT_OK) return rc;
+        if (nj == 1 && JR[1].jd.has_dups) {
+            set_err(errbuf, errlen,
+                    "join: duplicate foreign keys on a non-first join item "
+                    "not this round");
+            return YT_ERR_UNSUPPORTED;
+        }
+        for (int j = 0; j < J->foreign_value_count; j++) {
+            if (rn >= kMaxCols) { set_err(errbuf, errlen, "too many columns"); return YT_ERR_UNSUPPORTED; }
+            rt[rn++] = (uint8_t)J->foreign->columns[J->foreign_value_cols[j]].value_type;
+        }
+        nj++;
+    }
+    if (nj >= 1) *jd0 = &JR[0].jd;
+    if (nj >= 2) *jd1 = &JR[1].jd;
+    return YT_OK;
 }
 
 static void fold_totals_rows(const YtPlan* plan, const YtRowset* out,
@@ -2027,7 +2072,7 @@ static int finish_output(const YtPlan* plan, YtRowset* out,
  * filter/projection the generic path supports works here. */
 static int run_scan_topk(const YtPlan* plan, const YtChunk* chunk,
                          const YtExecOptions* options, const DevPlan* dp,
-                         const JoinDev* jd,
+                         const JoinDev* jd, const JoinDev* jd2,
                          YtRowset* output, YtStatistics* stats, double tw0,
                          char* errbuf, size_t errlen)
 {
@@ -2133,7 +2178,8 @@ static int run_scan_topk(const YtPlan* plan, const YtChunk* chunk,
                     fast_nulls, fast_signed, &tp, d_bins, d_misc, R2.stream));
             } else {
                 HIP_CHECK(ytql_launch_topk_hist(dp, R2.d_segs, R2.d_segex,
-                                                R2.d_off, R2.d_cnt, n, jd, &tp,
+                                                R2.d_off, R2.d_cnt, n, jd,
+                                                jd2, &tp,
                                                 d_bins, d_misc,
                                                 R2.d_err, R2.stream));
             }
@@ -2252,7 +2298,8 @@ static int run_scan_topk(const YtPlan* plan, const YtChunk* chunk,
                 d_rows_null, d_ctrs + 2, R2.stream));
         } else {
             HIP_CHECK(ytql_launch_topk_gather(dp, R2.d_segs, R2.d_segex,
-                                              R2.d_off, R2.d_cnt, n, jd, &tg,
+                                              R2.d_off, R2.d_cnt, n, jd,
+                                              jd2, &tg,
                                               d_rows_strict, d_ctrs,
                                               d_rows_tie, d_ctrs + 1,
                                               d_rows_null, d_ctrs + 2,
@@ -2285,7 +2332,7 @@ static int run_scan_topk(const YtPlan* plan, const YtChunk* chunk,
         HIP_CHECK(pool_alloc_host(&h_vals, sizeof(DevOutVal) * (M ? M : 1) * np));
         if (M) {
             HIP_CHECK(ytql_launch_topk_materialize(dp, R2.d_segs, R2.d_segex,
-                                                   R2.d_off, R2.d_cnt, jd,
+                                                   R2.d_off, R2.d_cnt, jd, jd2,
                                                    d_rows_all, M, d_vals,
                                                    R2.d_err, R2.stream));
             launches++;
@@ -2401,7 +2448,7 @@ fail:
  * and a pass mask; the host compacts in row order. */
 static int run_scan_project(const YtPlan* plan, const YtChunk* chunk,
                             const YtExecOptions* options, const DevPlan* dp,
-                            const JoinDev* jd,
+                            const JoinDev* jd, const JoinDev* jd2,
                             YtRowset* output, YtStatistics* stats, double tw0,
                             char* errbuf, size_t errlen)
 {
@@ -2444,7 +2491,7 @@ static int run_scan_project(const YtPlan* plan, const YtChunk* chunk,
         HIP_CHECK(hipEventCreate(&e1));
         HIP_CHECK(hipEventRecord(e0, R2.stream));
         HIP_CHECK(ytql_launch_scan_project(dp, R2.d_segs, R2.d_segex, R2.d_off,
-                                           R2.d_cnt, w0, wlen, jd, d_out,
+                                           R2.d_cnt, w0, wlen, jd, jd2, d_out,
                                            d_pass, R2.d_err, R2.stream));
         HIP_CHECK(hipEventRecord(e1, R2.stream));
         HIP_CHECK(hipMemcpyAsync(h_out, d_out, sizeof(DevOutVal) * wlen * plan->project_count,
@@ -3440,33 +3487,32 @@ extern "C" int yt_gpu_query_execute(
     rc = build_devplan(plan, chunk, &dp, errbuf, errlen);
     if (rc) return rc;
 
-    static const JoinDev kNoJoin = {};
-    JoinRun JR;
-    const JoinDev* jd = &kNoJoin;
-    if (plan->join) {
-        rc = setup_join(plan, chunk, &JR,
-                        (hipStream_t)(uintptr_t)options->stream, errbuf, errlen);
-        if (rc) return rc;
-        jd = &JR.jd;
-        if (JR.jd.has_dups && (plan->order_count > 0 || plan->agg_count == 0)) {
-            /* the ORDER BY / scan-project machinery tracks candidates by
-             * primary row id; a one-to-many join breaks that identity —
-             * grouped plans take the cross-product path instead */
-            set_err(errbuf, errlen,
-                    "join: duplicate foreign keys with ORDER BY / plain scan "
-                    "not this round (GROUP BY plans supported)");
-            return YT_ERR_UNSUPPORTED;
-        }
+    JoinRun JR[2];
+    const JoinDev* jd = nullptr;
+    const JoinDev* jd2 = nullptr;
+    rc = setup_join_chain(plan, chunk, JR, &jd, &jd2,
+                          (hipStream_t)(uintptr_t)options->stream,
+                          errbuf, errlen);
+    if (rc) return rc;
+    if (jd->active && jd->has_dups &&
+        (plan->order_count > 0 || plan->agg_count == 0)) {
+        /* the ORDER BY / scan-project machinery tracks candidates by
+         * primary row id; a one-to-many join breaks that identity —
+         * grouped plans take the cross-product path instead */
+        set_err(errbuf, errlen,
+                "join: duplicate foreign keys with ORDER BY / plain scan "
+                "not this round (GROUP BY plans supported)");
+        return YT_ERR_UNSUPPORTED;
     }
 
     if (plan->agg_count == 0 && plan->key_count == 0) {
         if (plan->order_count > 0) {
             /* scan + ORDER BY ... LIMIT: k-selection, no full materialization */
-            return run_scan_topk(plan, chunk, options, &dp, jd, output, stats,
-                                 tw0, errbuf, errlen);
+            return run_scan_topk(plan, chunk, options, &dp, jd, jd2, output,
+                                 stats, tw0, errbuf, errlen);
         }
-        return run_scan_project(plan, chunk, options, &dp, jd, output, stats,
-                                tw0, errbuf, errlen);
+        return run_scan_project(plan, chunk, options, &dp, jd, jd2, output,
+                                stats, tw0, errbuf, errlen);
     }
 
     FastShape fs;
@@ -3504,7 +3550,8 @@ extern "C" int yt_gpu_query_execute(
 
     const bool timing = getenv("YTQL_TIMING") != nullptr;   /* phase breakdown to stderr */
     tp0 = now_ms();
-    rc = run_scan(plan, chunk, options, &R, &dp, jd, &fs, maxw, stats, errbuf, errlen);
+    rc = run_scan(plan, chunk, options, &R, &dp, jd, jd2, &fs, maxw, stats,
+                  errbuf, errlen);
     if (rc) return rc;
     tp1 = now_ms();
 
@@ -3663,17 +3710,16 @@ static int query_partial_impl(
                      options->group_row_limit, errbuf, errlen);
     if (rc) return rc;
     double tq0 = now_ms();
-    static const JoinDev kNoJoin2 = {};
-    JoinRun JR2;
-    const JoinDev* jd2 = &kNoJoin2;
-    if (plan->join) {
-        rc = setup_join(plan, chunk, &JR2,
-                        (hipStream_t)(uintptr_t)options->stream, errbuf, errlen);
-        if (rc) return rc;
-        jd2 = &JR2.jd;
-        /* dup foreign keys are fine here: a partial is always grouped */
-    }
-    rc = run_scan(plan, chunk, options, &R, &dp, jd2, &fs, maxw, stats, errbuf, errlen);
+    JoinRun JRp[2];
+    const JoinDev* pjd = nullptr;
+    const JoinDev* pjd2 = nullptr;
+    rc = setup_join_chain(plan, chunk, JRp, &pjd, &pjd2,
+                          (hipStream_t)(uintptr_t)options->stream,
+                          errbuf, errlen);
+    if (rc) return rc;
+    /* dup foreign keys on item 0 are fine here: a partial is always grouped */
+    rc = run_scan(plan, chunk, options, &R, &dp, pjd, pjd2, &fs, maxw, stats,
+                  errbuf, errlen);
     if (rc) return rc;
     if (getenv("YTQL_TIMING"))
         fprintf(stderr, "[ytql timing] partial: setup %.2fms scan %.2fms\n",

@@ -460,18 +460,39 @@ struct ColCtx {
     const int32_t* col_seg_cnt;
     int64_t row;                  /* chunk row */
     uint32_t error;               /* YT_ERR_* */
-    /* equi-join state: jrow_for caches which row the probe resolved */
+    /* equi-join state: jrow_for caches which row the probe resolved.
+     * jt = join item 0 (may have duplicate keys — the cross-product
+     * machinery binds c.jrow per match); jt2 = item 1 (unique keys). */
     const JoinDev* jt = nullptr;
     int64_t jrow = -1;
     int64_t jrow_for = -1;
+    const JoinDev* jt2 = nullptr;
+    int64_t jrow2 = -1;
+    int64_t jrow2_for = -1;
 };
 
 __device__ int64_t join_resolve(const DevPlan& p, ColCtx& c);
+__device__ int64_t join_resolve2(const DevPlan& p, ColCtx& c);
 __device__ DVal jforeign_at(const JoinDev& jt, int fcol, int shift,
                             int64_t frow, uint8_t vt);
 
 __device__ DVal col_value(const DevPlan& p, ColCtx& c, int col)
 {
+    /* join item 1's columns sit ABOVE item 0's (primary_ncols of item 1 =
+     * primary + item-0 values), so check it first */
+    if (c.jt2 && c.jt2->active && col >= c.jt2->primary_ncols) {
+        int slot = col - c.jt2->primary_ncols;
+        int64_t frow = join_resolve2(p, c);
+        if (frow < 0) {
+            DVal v;
+            v.type = p.col_types[col];
+            v.null_ = 1;
+            v.bits = 0;
+            return v;
+        }
+        return jforeign_at(*c.jt2, c.jt2->fval_col[slot], c.jt2->f_shift[slot],
+                           frow, p.col_types[col]);
+    }
     if (c.jt && c.jt->active && col >= c.jt->primary_ncols) {
         int slot = col - c.jt->primary_ncols;
         int64_t frow = join_resolve(p, c);
@@ -553,12 +574,40 @@ __device__ int64_t join_resolve(const DevPlan& p, ColCtx& c)
     return c.jrow;
 }
 
+/* item-1 resolver: its key column may be one of item 0's appended values
+ * (col_value recursion handles the chain; key col < jt2->primary_ncols so
+ * it can never recurse back into item 1) */
+__device__ int64_t join_resolve2(const DevPlan& p, ColCtx& c)
+{
+    const JoinDev& jt = *c.jt2;
+    if (c.jrow2_for == c.row) return c.jrow2;
+    c.jrow2_for = c.row;
+    c.jrow2 = -1;
+    DVal k = col_value(p, c, jt.pkey_col);
+    if (k.null_) {
+        c.jrow2 = jt.null_row;
+        return c.jrow2;
+    }
+    uint64_t h = mix64(k.bits) & jt.hmask;
+    for (;;) {
+        int64_t r = jt.hrow[h];
+        if (r < 0) break;
+        if (jt.hkey[h] == k.bits) { c.jrow2 = jt.chead[h]; break; }
+        h = (h + 1) & jt.hmask;
+    }
+    return c.jrow2;
+}
+
 /* INNER-join row gate: call once per row before evaluating expressions */
 __device__ __forceinline__ bool join_row_ok(const DevPlan& p, ColCtx& c)
 {
-    if (!c.jt || !c.jt->active) return true;
-    if (c.jt->is_left) return true;
-    return join_resolve(p, c) >= 0;
+    if (c.jt && c.jt->active && !c.jt->is_left) {
+        if (join_resolve(p, c) < 0) return false;
+    }
+    if (c.jt2 && c.jt2->active && !c.jt2->is_left) {
+        if (join_resolve2(p, c) < 0) return false;
+    }
+    return true;
 }
 
 __device__ DVal eval_prog(const DevPlan& p, ColCtx& c, int off, int len)
@@ -1198,7 +1247,7 @@ __global__ void k_join_chain(JoinDev jt, int64_t frows, unsigned* error_out)
 __global__ void __launch_bounds__(256)
 k_scan_generic(DevPlan p, const DevSeg* segs, const SegEx* segex,
                const int32_t* col_seg_off, const int32_t* col_seg_cnt,
-               int64_t row_count, JoinDev jd,
+               int64_t row_count, JoinDev jd, JoinDev jd2,
                TableHdr* th, unsigned long long* slots,
                unsigned* error_out)
 {
@@ -1209,6 +1258,7 @@ k_scan_generic(DevPlan p, const DevSeg* segs, const SegEx* segex,
     c.col_seg_cnt = col_seg_cnt;
     c.error = 0;
     c.jt = &jd;
+    c.jt2 = jd2.active ? &jd2 : nullptr;
 
     DVal aggv[kMaxAggs];
 
@@ -1226,7 +1276,7 @@ k_scan_generic(DevPlan p, const DevSeg* segs, const SegEx* segex,
             if (m < 0 && !jd.is_left) continue;
         }
         for (;;) {
-        if (joined) { c.jrow = m; c.jrow_for = r; }
+        if (joined) { c.jrow = m; c.jrow_for = r; c.jrow2_for = -9; }
         bool pass = true;
         if (p.filter_len) {
             DVal f = eval_prog(p, c, p.filter_off, p.filter_len);
@@ -1280,7 +1330,8 @@ k_scan_generic(DevPlan p, const DevSeg* segs, const SegEx* segex,
 __global__ void __launch_bounds__(256)
 k_scan_project(DevPlan p, const DevSeg* segs, const SegEx* segex,
                const int32_t* col_seg_off, const int32_t* col_seg_cnt,
-               int64_t row_base, int64_t row_count, JoinDev jd,
+               int64_t row_base, int64_t row_count,
+               JoinDev jd, JoinDev jd2,
                DevOutVal* out, uint8_t* pass, unsigned* error_out)
 {
     /* [row_base, row_base+row_count) window: out/pass are window-local so
@@ -1293,6 +1344,7 @@ k_scan_project(DevPlan p, const DevSeg* segs, const SegEx* segex,
     c.col_seg_cnt = col_seg_cnt;
     c.error = 0;
     c.jt = &jd;
+    c.jt2 = jd2.active ? &jd2 : nullptr;
 
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -1355,7 +1407,7 @@ __device__ __forceinline__ bool topk_map(const DVal& v, int desc,
 __global__ void __launch_bounds__(256)
 k_topk_hist(DevPlan p, const DevSeg* segs, const SegEx* segex,
             const int32_t* col_seg_off, const int32_t* col_seg_cnt,
-            int64_t row_count, JoinDev jd, TopkPass tp,
+            int64_t row_count, JoinDev jd, JoinDev jd2, TopkPass tp,
             unsigned long long* bins,           /* 2048 */
             unsigned long long* misc,           /* null_cnt, mmin, mmax */
             unsigned* error_out)
@@ -1371,6 +1423,7 @@ k_topk_hist(DevPlan p, const DevSeg* segs, const SegEx* segex,
     c.col_seg_cnt = col_seg_cnt;
     c.error = 0;
     c.jt = &jd;
+    c.jt2 = jd2.active ? &jd2 : nullptr;
     unsigned long long nulls = 0;
     uint64_t mmin = ~0ULL, mmax = 0;
 
@@ -1411,7 +1464,7 @@ k_topk_hist(DevPlan p, const DevSeg* segs, const SegEx* segex,
 __global__ void __launch_bounds__(256)
 k_topk_gather(DevPlan p, const DevSeg* segs, const SegEx* segex,
               const int32_t* col_seg_off, const int32_t* col_seg_cnt,
-              int64_t row_count, JoinDev jd, TopkGather tg,
+              int64_t row_count, JoinDev jd, JoinDev jd2, TopkGather tg,
               int64_t* rows_strict, unsigned long long* ctr_strict,
               int64_t* rows_tie, unsigned long long* ctr_tie,
               int64_t* rows_null, unsigned long long* ctr_null,
@@ -1424,6 +1477,7 @@ k_topk_gather(DevPlan p, const DevSeg* segs, const SegEx* segex,
     c.col_seg_cnt = col_seg_cnt;
     c.error = 0;
     c.jt = &jd;
+    c.jt2 = jd2.active ? &jd2 : nullptr;
 
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -1564,7 +1618,7 @@ k_topk_gather_fast(const DevSeg* segs, const SegEx* segex, int seg_off,
 __global__ void __launch_bounds__(256)
 k_topk_materialize(DevPlan p, const DevSeg* segs, const SegEx* segex,
                    const int32_t* col_seg_off, const int32_t* col_seg_cnt,
-                   JoinDev jd, const int64_t* rows, int64_t m, DevOutVal* out,
+                   JoinDev jd, JoinDev jd2, const int64_t* rows, int64_t m, DevOutVal* out,
                    unsigned* error_out)
 {
     ColCtx c;
@@ -1574,6 +1628,7 @@ k_topk_materialize(DevPlan p, const DevSeg* segs, const SegEx* segex,
     c.col_seg_cnt = col_seg_cnt;
     c.error = 0;
     c.jt = &jd;
+    c.jt2 = jd2.active ? &jd2 : nullptr;
 
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -3496,7 +3551,7 @@ hipError_t ytql_launch_topk_hist(const DevPlan* p, const DevSeg* segs,
                                  const int32_t* col_seg_off,
                                  const int32_t* col_seg_cnt,
                                  int64_t row_count, const JoinDev* jd,
-                                 const TopkPass* tp,
+                                 const JoinDev* jd2, const TopkPass* tp,
                                  unsigned long long* bins,
                                  unsigned long long* null_cnt,
                                  unsigned* error_out, hipStream_t st)
@@ -3506,7 +3561,7 @@ hipError_t ytql_launch_topk_hist(const DevPlan* p, const DevSeg* segs,
     int grid = (int)(want > 4096 ? 4096 : (want > 0 ? want : 1));
     hipLaunchKernelGGL(k_topk_hist, dim3(grid), dim3(block), 0, st,
                        *p, segs, segex, col_seg_off, col_seg_cnt, row_count,
-                       *jd, *tp, bins, null_cnt, error_out);
+                       *jd, *jd2, *tp, bins, null_cnt, error_out);
     return hipGetLastError();
 }
 
@@ -3626,7 +3681,7 @@ hipError_t ytql_launch_topk_gather(const DevPlan* p, const DevSeg* segs,
                                    const int32_t* col_seg_off,
                                    const int32_t* col_seg_cnt,
                                    int64_t row_count, const JoinDev* jd,
-                                   const TopkGather* tg,
+                                   const JoinDev* jd2, const TopkGather* tg,
                                    int64_t* rows_strict, unsigned long long* ctr_strict,
                                    int64_t* rows_tie, unsigned long long* ctr_tie,
                                    int64_t* rows_null, unsigned long long* ctr_null,
@@ -3637,7 +3692,7 @@ hipError_t ytql_launch_topk_gather(const DevPlan* p, const DevSeg* segs,
     int grid = (int)(want > 4096 ? 4096 : (want > 0 ? want : 1));
     hipLaunchKernelGGL(k_topk_gather, dim3(grid), dim3(block), 0, st,
                        *p, segs, segex, col_seg_off, col_seg_cnt, row_count,
-                       *jd, *tg, rows_strict, ctr_strict, rows_tie, ctr_tie,
+                       *jd, *jd2, *tg, rows_strict, ctr_strict, rows_tie, ctr_tie,
                        rows_null, ctr_null, error_out);
     return hipGetLastError();
 }
@@ -3647,6 +3702,7 @@ hipError_t ytql_launch_topk_materialize(const DevPlan* p, const DevSeg* segs,
                                         const int32_t* col_seg_off,
                                         const int32_t* col_seg_cnt,
                                         const JoinDev* jd,
+                                        const JoinDev* jd2,
                                         const int64_t* rows, int64_t m,
                                         DevOutVal* out, unsigned* error_out,
                                         hipStream_t st)
@@ -3655,8 +3711,8 @@ hipError_t ytql_launch_topk_materialize(const DevPlan* p, const DevSeg* segs,
     int64_t want = (m + block - 1) / block;
     int grid = (int)(want > 4096 ? 4096 : (want > 0 ? want : 1));
     hipLaunchKernelGGL(k_topk_materialize, dim3(grid), dim3(block), 0, st,
-                       *p, segs, segex, col_seg_off, col_seg_cnt, *jd, rows, m,
-                       out, error_out);
+                       *p, segs, segex, col_seg_off, col_seg_cnt, *jd, *jd2,
+                       rows, m, out, error_out);
     return hipGetLastError();
 }
 
@@ -3666,7 +3722,7 @@ hipError_t ytql_launch_scan_project(const DevPlan* p, const DevSeg* segs,
                                     const int32_t* col_seg_cnt,
                                     int64_t row_base,
                                     int64_t row_count, const JoinDev* jd,
-                                    DevOutVal* out,
+                                    const JoinDev* jd2, DevOutVal* out,
                                     uint8_t* pass, unsigned* error_out,
                                     hipStream_t st)
 {
@@ -3675,7 +3731,7 @@ hipError_t ytql_launch_scan_project(const DevPlan* p, const DevSeg* segs,
     int grid = (int)(want > 4096 ? 4096 : (want > 0 ? want : 1));
     hipLaunchKernelGGL(k_scan_project, dim3(grid), dim3(block), 0, st,
                        *p, segs, segex, col_seg_off, col_seg_cnt, row_base,
-                       row_count, *jd, out, pass, error_out);
+                       row_count, *jd, *jd2, out, pass, error_out);
     return hipGetLastError();
 }
 
@@ -3684,6 +3740,7 @@ hipError_t ytql_launch_scan_generic(const DevPlan* p, const DevSeg* segs,
                                     const int32_t* col_seg_off,
                                     const int32_t* col_seg_cnt,
                                     int64_t row_count, const JoinDev* jd,
+                                    const JoinDev* jd2,
                                     TableHdr* th, unsigned long long* slots,
                                     unsigned* error_out, hipStream_t st)
 {
@@ -3692,7 +3749,7 @@ hipError_t ytql_launch_scan_generic(const DevPlan* p, const DevSeg* segs,
     int grid = (int)(want > 4096 ? 4096 : (want > 0 ? want : 1));
     hipLaunchKernelGGL(k_scan_generic, dim3(grid), dim3(block), 0, st,
                        *p, segs, segex, col_seg_off, col_seg_cnt, row_count,
-                       *jd, th, slots, error_out);
+                       *jd, *jd2, th, slots, error_out);
     return hipGetLastError();
 }
 
