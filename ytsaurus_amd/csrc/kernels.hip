@@ -1278,7 +1278,12 @@ k_scan_generic(DevPlan p, const DevSeg* segs, const SegEx* segex,
         for (;;) {
         if (joined) { c.jrow = m; c.jrow_for = r; c.jrow2_for = -9; }
         bool pass = true;
-        if (p.filter_len) {
+        /* INNER gate for join item 1 (its key may depend on item 0's match,
+         * so it re-resolves per match pass) */
+        if (c.jt2 && c.jt2->active && !c.jt2->is_left &&
+            join_resolve2(p, c) < 0)
+            pass = false;
+        if (pass && p.filter_len) {
             DVal f = eval_prog(p, c, p.filter_off, p.filter_len);
             if (f.null_ || f.bits == 0) pass = false;
         }
