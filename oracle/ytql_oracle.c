@@ -49,6 +49,11 @@
 /* ------------------------------------------------------------------ */
 /* small helpers                                                       */
 
+/* oracle-internal aggregate: count of non-null args (avg's count leg —
+ * used by yto_partial to materialize the exact {count,sum} avg state;
+ * never part of the public plan surface) */
+#define YTO_AGG_NNCNT 100
+
 static void set_err(char* errbuf, size_t errlen, const char* msg)
 {
     if (errbuf && errlen) {
@@ -721,6 +726,9 @@ static void* scan_worker(void* arg)
             } else if (agg->func == YT_AGG_FIRST) {
                 /* FirstIteration: keep the first non-null in scan order */
                 if (st->type == YT_VT_NULL) *st = nv;
+            } else if (agg->func == YTO_AGG_NNCNT) {
+                if (nv.type != YT_VT_NULL)
+                    t->table.acounts[g * plan->agg_count + a]++;
             } else minmax_update_val(st, nv, agg->func == YT_AGG_MAX);
         }
     }
@@ -741,6 +749,10 @@ static int emit_group_row(const YtPlan* plan, GroupTable* t, int64_t g,
     for (int a = 0; a < ac; a++) {
         if (plan->aggs[a]->func == YT_AGG_SUM1) {
             Val v; v.type = YT_VT_INT64; v.bits = t->rowcounts[g];
+            rowvals[kc + a] = v;
+        } else if (plan->aggs[a]->func == YTO_AGG_NNCNT) {
+            Val v; v.type = YT_VT_INT64; v.bits = t->acounts[g * ac + a];
+            v.str = 0; v.len = 0;
             rowvals[kc + a] = v;
         } else if (plan->aggs[a]->func == YT_AGG_AVG) {
             uint64_t c = t->acounts[g * ac + a];
@@ -1081,7 +1093,7 @@ int yto_execute(const YtPlan* plan, const YtChunk* chunk,
     if (nthreads < 1) nthreads = 1;
     for (int a = 0; a < plan->agg_count; a++) {
         int f = plan->aggs[a]->func;
-        if (f < YT_AGG_SUM || f > YT_AGG_AVG) {
+        if ((f < YT_AGG_SUM || f > YT_AGG_AVG) && f != YTO_AGG_NNCNT) {
             set_err(errbuf, errlen, "unknown aggregate function");
             return YT_ERR_UNSUPPORTED;
         }
@@ -1608,18 +1620,42 @@ int yto_partial(const YtPlan* plan, const YtChunk* chunk,
      * the bottom query. Multi-key + join stays refused (yto_partial_mk). */
     if (plan->key_count != 1) { set_err(errbuf, errlen, "partial: need 1 key"); return YT_ERR_UNSUPPORTED; }
     int sum_idx = -1;
+    int is_avg = 0;
     for (int a = 0; a < plan->agg_count; a++) {
-        if (plan->aggs[a]->func == YT_AGG_SUM) {
-            if (sum_idx >= 0) { set_err(errbuf, errlen, "partial: one sum agg max this round"); return YT_ERR_UNSUPPORTED; }
+        int f = plan->aggs[a]->func;
+        if (f == YT_AGG_SUM || f == YT_AGG_AVG) {
+            if (sum_idx >= 0) { set_err(errbuf, errlen, "partial: one sum/avg agg max this round"); return YT_ERR_UNSUPPORTED; }
             sum_idx = a;
+            is_avg = (f == YT_AGG_AVG);
         }
-        else if (plan->aggs[a]->func != YT_AGG_SUM1) { set_err(errbuf, errlen, "partial: sum/sum1 only"); return YT_ERR_UNSUPPORTED; }
+        else if (f != YT_AGG_SUM1) { set_err(errbuf, errlen, "partial: sum/avg/sum1 only"); return YT_ERR_UNSUPPORTED; }
     }
 
-    /* run the local group-by via yto_execute on a plan without projection */
+    /* run the local group-by via yto_execute on a plan without projection.
+     * avg(x) lowers to the {count,sum} state the reference's coordinated
+     * avg carries (GroupByWithAvgCoordinated ql_query_ut.cpp:2760): the
+     * local avg slot becomes sum(x), and a second pass with the filter
+     * extended by (x == x) — null comparison yields null, which the WHERE
+     * clause drops — counts the non-null args per group exactly. */
     YtPlan local = *plan;
     local.project_count = 0;
     local.projects = NULL;
+    YtAgg local_aggs[17];
+    const YtAgg* local_agg_ptrs[17];
+    if (is_avg) {
+        if (plan->agg_count > 16) { set_err(errbuf, errlen, "partial: too many aggregates"); return YT_ERR_UNSUPPORTED; }
+        for (int a = 0; a < plan->agg_count; a++) {
+            local_aggs[a] = *plan->aggs[a];
+            if (a == sum_idx) local_aggs[a].func = YT_AGG_SUM;
+            local_agg_ptrs[a] = &local_aggs[a];
+        }
+        /* the avg count leg: exact non-null arg count per group */
+        local_aggs[plan->agg_count].func = YTO_AGG_NNCNT;
+        local_aggs[plan->agg_count].arg = plan->aggs[sum_idx]->arg;
+        local_agg_ptrs[plan->agg_count] = &local_aggs[plan->agg_count];
+        local.agg_count = plan->agg_count + 1;
+        local.aggs = local_agg_ptrs;
+    }
 
     int64_t cap = capacity_rows;
     YtValue* tmp = malloc(sizeof(YtValue) * cap * (1 + plan->agg_count));
@@ -1637,7 +1673,8 @@ int yto_partial(const YtPlan* plan, const YtChunk* chunk,
      * nonnull >= 1; exact nonnull count is not needed for sum-merge parity —
      * only null-ness matters (sum.c:12-22). Encode nonnull = 1 for non-null
      * state, 0 for null state. */
-    int ncols = 1 + plan->agg_count;
+    int ncols = 1 + local.agg_count;   /* incl. the internal count leg */
+
     int64_t* counts = calloc(partition_count, sizeof(int64_t));
     for (int64_t r = 0; r < rs.row_count; r++) {
         const YtValue* row = rs.values + r * ncols;
@@ -1663,6 +1700,8 @@ int yto_partial(const YtPlan* plan, const YtChunk* chunk,
                 nonnull = (row[1 + a].type != YT_VT_NULL);
             }
         }
+        if (is_avg)
+            nonnull = row[1 + plan->agg_count].data.bits;   /* count leg */
         uint64_t sum_dbl = 0;
         if (sum_idx >= 0 && row[1 + sum_idx].type == YT_VT_DOUBLE) sum_dbl = 2;
         sr->meta = (uint64_t)knull | sum_dbl | (nonnull << 8);
@@ -1697,13 +1736,16 @@ int yto_merge(const YtPlan* plan, const YtStateRow* states, int64_t nstates,
         t.rowcounts[g] += states[i].row_count;
         uint64_t nonnull = states[i].meta >> 8;
         for (int a = 0; a < plan->agg_count; a++) {
-            if (plan->aggs[a]->func != YT_AGG_SUM) continue;
+            int f = plan->aggs[a]->func;
+            if (f != YT_AGG_SUM && f != YT_AGG_AVG) continue;
             if (nonnull) {
                 Val nv;
                 nv.type = (states[i].meta & 2) ? YT_VT_DOUBLE : YT_VT_INT64;
                 nv.bits = states[i].sum_bits;
                 nv.str = 0; nv.len = 0;
                 sum_update_val(&t.states[g * plan->agg_count + a], nv);
+                /* avg finalize divides by the accumulated non-null count */
+                t.acounts[g * plan->agg_count + a] += nonnull;
             }
         }
     }

@@ -3657,13 +3657,17 @@ static int query_partial_impl(
                 "partial: join with multi-key GROUP BY not this round");
         return YT_ERR_UNSUPPORTED;
     }
+    /* the 32-byte state carries ONE running {sum, nonnull-count} slot plus
+     * the row count: sum(x) or avg(x) (the reference's coordinated avg is
+     * exactly {count,sum}: GroupByWithAvgCoordinated ql_query_ut.cpp:2760) */
     int sum_slot = -1;
     for (int a = 0; a < plan->agg_count; a++) {
-        if (plan->aggs[a]->func == YT_AGG_SUM) {
-            if (sum_slot >= 0) { set_err(errbuf, errlen, "partial: one sum agg max this round"); return YT_ERR_UNSUPPORTED; }
+        int f = plan->aggs[a]->func;
+        if (f == YT_AGG_SUM || f == YT_AGG_AVG) {
+            if (sum_slot >= 0) { set_err(errbuf, errlen, "partial: one sum/avg agg max this round"); return YT_ERR_UNSUPPORTED; }
             sum_slot = a;
-        } else if (plan->aggs[a]->func != YT_AGG_SUM1) {
-            set_err(errbuf, errlen, "partial: sum/sum(1) only");
+        } else if (f != YT_AGG_SUM1) {
+            set_err(errbuf, errlen, "partial: sum/avg/sum(1) only");
             return YT_ERR_UNSUPPORTED;
         }
     }
@@ -3930,7 +3934,8 @@ static int merge_states_impl(
     output->totals_row = 0;
     int sum_slot = -1;
     for (int a = 0; a < plan->agg_count; a++) {
-        if (plan->aggs[a]->func == YT_AGG_SUM) sum_slot = a;
+        if (plan->aggs[a]->func == YT_AGG_SUM ||
+            plan->aggs[a]->func == YT_AGG_AVG) sum_slot = a;
     }
 
     YtExecOptions defopt;

@@ -3081,7 +3081,9 @@ __global__ void k_part_scatter(const OutGroup* groups, int64_t n, int nparts,
         unsigned long long pos = lbase[p] + atomicAdd(&lclaim[p], 1u);
         YtStateRow& sr = out[pos];
         sr.key_bits = g.key_bits;
-        uint64_t nonnull = (sum_slot >= 0) ? (g.agg_nonnull[sum_slot] ? 1 : 0) : 0;
+        /* meta bits 8+ carry the EXACT non-null count (needed for avg; any
+         * value >0 means a non-null sum) */
+        uint64_t nonnull = (sum_slot >= 0) ? g.agg_nonnull[sum_slot] : 0;
         sr.meta = (uint64_t)knull | (nonnull << 8)
                 | (sum_is_double ? 2ULL : 0ULL);
         sr.sum_bits = (sum_slot >= 0) ? g.agg_bits[sum_slot] : 0;
@@ -3453,7 +3455,9 @@ __global__ void k_merge_states(const YtStateRow* states, int64_t n,
             } else {
                 atomicAdd(aggp + 2 * sum_slot, (unsigned long long)sr.sum_bits);
             }
-            atomicAdd(aggp + 2 * sum_slot + 1, 1ULL);
+            /* states carry EXACT non-null counts (avg = sum/count) */
+            atomicAdd(aggp + 2 * sum_slot + 1,
+                      (unsigned long long)(sr.meta >> 8));
         }
     }
 }
