@@ -445,13 +445,21 @@ int yt_gpu_query_execute(
  * (key_count + 3*agg_count... ) — packed as YtStateRow records below.
  * part_offsets/part_counts (length partition_count, host) describe the layout.
  */
-/* Round-1 state-row record: the {one int64/boolean key, sum(expr), sum(1)}
- * plan family (BASELINE configs 3/4). Other shapes return YT_ERR_UNSUPPORTED
- * from the two-phase entries for now. 32 bytes, device-native layout. */
+/* State-row record: {group key, ONE value-carrying aggregate state, sum(1)}
+ * plans — the value slot is sum(x) / avg(x) / min(x) / max(x); the plan
+ * the merge receives names the function (the reference's merge codegen is
+ * likewise driven by the front query's aggregate list,
+ * cg_fragment_compiler.cpp:4016-4134). avg's state is exactly the
+ * reference's coordinated {count,sum} (GroupByWithAvgCoordinated
+ * ql_query_ut.cpp:2760). Multi-key plans pack composites into key_bits
+ * (the _mk entries); string keys reference an exchanged pool slice (the
+ * _str entries). 32 bytes, device-native layout. */
 typedef struct YtStateRow {
-    uint64_t key_bits;               /* raw key bits (i64), undefined when key null */
-    uint64_t meta;                   /* bit0 = key is null; bits 8..63 = nonnull count of sum arg */
-    uint64_t sum_bits;               /* wrapping int64 sum of the sum() argument */
+    uint64_t key_bits;               /* raw key bits / packed composite / pool ref */
+    uint64_t meta;                   /* bit0 = key null; bit1 = value is double;
+                                        bits 8..63 = EXACT non-null arg count */
+    uint64_t sum_bits;               /* running sum (wrapping i64 / double bits)
+                                        or running min/max */
     uint64_t row_count;              /* rows in group == sum(1) state */
 } YtStateRow;
 

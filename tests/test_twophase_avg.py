@@ -156,3 +156,105 @@ def test_gpu_avg_states_merged_by_oracle(cuda):
     for kk, av, cnt in union:
         _, wav, wc = wm[kk]
         assert cnt == wc and av == pytest.approx(wav, rel=1e-12)
+
+
+def mm_plan(is_max, col=1):
+    f = y.agg_max if is_max else y.agg_min
+    return y.Plan(keys=[y.col(0)], aggs=[f(y.col(col)), y.agg_sum1()])
+
+
+@pytest.mark.parametrize("is_max", [False, True])
+@pytest.mark.parametrize("double_arg", [False, True])
+def test_oracle_minmax_two_phase(is_max, double_arg):
+    """min/max through the exchange: states carry the RAW running value
+    (meta bit1 types it); the merge re-min/maxes (udf/min.c merge =
+    another min)."""
+    world = 3
+    shards = []
+    for r in range(world):
+        rng = np.random.default_rng(720 + r)
+        n = 4000
+        k = rng.integers(-40, 40, n, dtype=np.int64)
+        kn = (rng.random(n) < 0.03).astype(np.uint8)
+        vn = (rng.random(n) < 0.3).astype(np.uint8)
+        if double_arg:
+            v = rng.random(n) * 100 - 50
+            vcol = y.encode_double(v, vn)
+        else:
+            v = rng.integers(-10**9, 10**9, n, dtype=np.int64)
+            vcol = y.encode_int64(v, vn)
+        shards.append((k, kn, v, vn,
+                       y.Chunk([y.encode_int64(k, kn), vcol], n)))
+    parts = [[] for _ in range(world)]
+    for *_x, ch in shards:
+        states, counts = y.oracle_partial(mm_plan(is_max), ch, world)
+        at = 0
+        for p in range(world):
+            seg = (YtStateRow * max(counts[p], 1))()
+            for i in range(counts[p]):
+                seg[i] = states[at + i]
+            parts[p].append((seg, counts[p]))
+            at += counts[p]
+    union = []
+    for p in range(world):
+        union += y.oracle_merge(mm_plan(is_max), parts[p])
+    allk = np.concatenate([s[0] for s in shards])
+    allkn = np.concatenate([s[1] for s in shards])
+    allv = np.concatenate([s[2] for s in shards])
+    allvn = np.concatenate([s[3] for s in shards])
+    vcol = (y.encode_double(allv, allvn) if double_arg
+            else y.encode_int64(allv.astype(np.int64), allvn))
+    big = y.Chunk([y.encode_int64(allk, allkn), vcol], len(allk))
+    want, _ = y.oracle_execute(mm_plan(is_max), big)
+    assert y.sort_rows(union) == y.sort_rows(want)
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("is_max", [False, True])
+@pytest.mark.parametrize("double_arg", [False, True])
+def test_gpu_minmax_two_phase(cuda, is_max, double_arg):
+    rng = np.random.default_rng(721)
+    n = 120_000
+    k = rng.integers(0, 503, n, dtype=np.int64)
+    vn = (rng.random(n) < 0.25).astype(np.uint8)
+    if double_arg:
+        vcol = y.encode_double(rng.random(n) * 20 - 10, vn)
+    else:
+        vcol = y.encode_int64(
+            rng.integers(-10**12, 10**12, n, dtype=np.int64), vn)
+    ch = y.Chunk([y.encode_int64(k), vcol], n)
+    cap = 4 * 503 + 1024
+    states_t = cuda.zeros((cap, 4), dtype=cuda.int64, device="cuda")
+    counts, _ = y.gpu_partial(mm_plan(is_max), ch.c_device(cuda), 1,
+                              states_t.data_ptr(), cap, max_groups_hint=2048)
+    got, _ = y.gpu_merge(mm_plan(is_max), states_t.data_ptr(), sum(counts),
+                         max_groups_hint=2048,
+                         col_types=[VT_INT64,
+                                    VT_DOUBLE if double_arg else VT_INT64])
+    want, _ = y.oracle_execute(mm_plan(is_max), ch)
+    assert y.sort_rows(got) == y.sort_rows(want)
+
+
+@pytest.mark.gpu
+def test_gpu_minmax_states_cross_impl(cuda):
+    """GPU min states merged by the ORACLE: the raw-value state encoding
+    must agree across implementations."""
+    rng = np.random.default_rng(722)
+    n = 20_000
+    k = rng.integers(0, 37, n, dtype=np.int64)
+    v = rng.integers(-10**6, 10**6, n, dtype=np.int64)
+    vn = (rng.random(n) < 0.2).astype(np.uint8)
+    ch = y.Chunk([y.encode_int64(k), y.encode_int64(v, vn)], n)
+    cap = n + 16
+    states_t = cuda.zeros((cap, 4), dtype=cuda.int64, device="cuda")
+    counts, _ = y.gpu_partial(mm_plan(False), ch.c_device(cuda), 1,
+                              states_t.data_ptr(), cap, max_groups_hint=128)
+    host = states_t.cpu().numpy().view(np.uint64)
+    seg = (YtStateRow * max(counts[0], 1))()
+    for i in range(counts[0]):
+        row = host[i]
+        seg[i] = YtStateRow(key_bits=int(row[0]), meta=int(row[1]),
+                            sum_bits=int(row[2]), row_count=int(row[3]))
+    union = y.oracle_merge(mm_plan(False), [(seg, counts[0])])
+    want, _ = y.oracle_execute(mm_plan(False), ch)
+    assert y.sort_rows(union) == y.sort_rows(want)

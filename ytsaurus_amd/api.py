@@ -19,7 +19,7 @@ from ._abi import (
 
 __all__ = [
     "col", "lit", "null", "litf", "Plan", "Join", "agg_sum", "agg_sum1",
-    "agg_first", "agg_avg",
+    "agg_first", "agg_avg", "agg_min", "agg_max",
     "encode_int64", "encode_double", "encode_bool", "encode_string", "encode_string_raw",
     "oracle_decode_strings", "encode_versioned_int64", "encode_versioned_double", "encode_versioned_string", "oracle_versioned_read", "gpu_versioned_read", "gpu_versioned_scan_chunk", "gpu_versioned_scan_table", "ScanChunk", "VersionedColumn",
     "Chunk", "oracle_execute",
@@ -108,6 +108,18 @@ def agg_first(e):
     """first(x): the first non-null value in scan order (udf/first.c,
     registry.cpp FirstIteration) -- arbitrary across parallel scans."""
     return (AGG_FIRST, e)
+
+
+def agg_min(e):
+    """min(expr) — udf/min.c"""
+    from ytsaurus_amd._abi import AGG_MIN
+    return (AGG_MIN, e)
+
+
+def agg_max(e):
+    """max(expr) — udf/max.c"""
+    from ytsaurus_amd._abi import AGG_MAX
+    return (AGG_MAX, e)
 
 
 def agg_avg(e):

@@ -106,9 +106,11 @@ hipError_t ytql_launch_compact(TableHdr*, TableHdr*, const unsigned long long*, 
 hipError_t ytql_launch_part_count(const OutGroup*, int64_t, int, int,
                                   unsigned long long*, hipStream_t);
 hipError_t ytql_launch_part_scatter(const OutGroup*, int64_t, int, int, int,
-                                    unsigned long long*, YtStateRow*, hipStream_t);
+                                    int, unsigned long long*, YtStateRow*,
+                                    hipStream_t);
 hipError_t ytql_launch_merge_states(const YtStateRow*, int64_t, int, int,
-                                    TableHdr*, unsigned long long*, hipStream_t);
+                                    int, TableHdr*, unsigned long long*,
+                                    hipStream_t);
 hipError_t ytql_launch_strgrp_accum(const StrGroupParams*, const DevSeg*, const SegEx*,
                                     const int64_t*, unsigned long long*, TableHdr*,
                                     hipStream_t);
@@ -3661,13 +3663,16 @@ static int query_partial_impl(
      * the row count: sum(x) or avg(x) (the reference's coordinated avg is
      * exactly {count,sum}: GroupByWithAvgCoordinated ql_query_ut.cpp:2760) */
     int sum_slot = -1;
+    int state_func = YT_AGG_SUM;
     for (int a = 0; a < plan->agg_count; a++) {
         int f = plan->aggs[a]->func;
-        if (f == YT_AGG_SUM || f == YT_AGG_AVG) {
-            if (sum_slot >= 0) { set_err(errbuf, errlen, "partial: one sum/avg agg max this round"); return YT_ERR_UNSUPPORTED; }
+        if (f == YT_AGG_SUM || f == YT_AGG_AVG || f == YT_AGG_MIN ||
+            f == YT_AGG_MAX) {
+            if (sum_slot >= 0) { set_err(errbuf, errlen, "partial: one value-carrying agg max this round"); return YT_ERR_UNSUPPORTED; }
             sum_slot = a;
+            state_func = f;
         } else if (f != YT_AGG_SUM1) {
-            set_err(errbuf, errlen, "partial: sum/avg/sum(1) only");
+            set_err(errbuf, errlen, "partial: sum/avg/min/max/sum(1) only");
             return YT_ERR_UNSUPPORTED;
         }
     }
@@ -3791,12 +3796,18 @@ static int query_partial_impl(
         int sum_is_double = 0;
         if (sum_slot >= 0) {
             /* dp.col_types already appends the joined foreign columns */
-            sum_is_double =
-                expr_static_type(plan->aggs[sum_slot]->arg, dp.col_types)
-                == YT_VT_DOUBLE;
+            int at = expr_static_type(plan->aggs[sum_slot]->arg, dp.col_types);
+            sum_is_double = at == YT_VT_DOUBLE;
+            if ((state_func == YT_AGG_MIN || state_func == YT_AGG_MAX) &&
+                at != YT_VT_INT64 && at != YT_VT_DOUBLE) {
+                set_err(errbuf, errlen,
+                        "partial: min/max over int64/double args this round");
+                rc = YT_ERR_UNSUPPORTED;
+                goto fail;
+            }
         }
         HIP_CHECK(ytql_launch_part_scatter(R.d_groups, total, partition_count, sum_slot,
-                                           sum_is_double,
+                                           sum_is_double, state_func,
                                            d_counts, (YtStateRow*)states_device, R.stream));
         HIP_CHECK(hipStreamSynchronize(R.stream));
         g_pool.put(d_counts);
@@ -3933,9 +3944,11 @@ static int merge_states_impl(
     }
     output->totals_row = 0;
     int sum_slot = -1;
+    int state_func = YT_AGG_SUM;
     for (int a = 0; a < plan->agg_count; a++) {
-        if (plan->aggs[a]->func == YT_AGG_SUM ||
-            plan->aggs[a]->func == YT_AGG_AVG) sum_slot = a;
+        int f = plan->aggs[a]->func;
+        if (f == YT_AGG_SUM || f == YT_AGG_AVG || f == YT_AGG_MIN ||
+            f == YT_AGG_MAX) { sum_slot = a; state_func = f; }
     }
 
     YtExecOptions defopt;
@@ -3953,8 +3966,8 @@ static int merge_states_impl(
     double tq1 = now_ms();
 
     HIP_CHECK(ytql_launch_merge_states((const YtStateRow*)states_device, state_row_count,
-                                       plan->agg_count, sum_slot, R.d_th, R.d_slots,
-                                       R.stream));
+                                       plan->agg_count, sum_slot, state_func,
+                                       R.d_th, R.d_slots, R.stream));
     HIP_CHECK(hipStreamSynchronize(R.stream));
     if (getenv("YTQL_TIMING"))
         fprintf(stderr, "[ytql timing] merge: setup %.2fms kernel %.2fms\n",

@@ -755,6 +755,15 @@ __device__ __forceinline__ uint64_t ord_map(uint64_t bits, uint8_t type)
     return bits;   /* uint64 / boolean */
 }
 
+__device__ __forceinline__ uint64_t ord_unmap(uint64_t m, uint8_t type)
+{
+    if (type == YT_VT_DOUBLE)
+        return (m & 0x8000000000000000ULL) ? (m & ~0x8000000000000000ULL)
+                                           : ~m;
+    if (type == YT_VT_INT64) return m ^ 0x8000000000000000ULL;
+    return m;
+}
+
 __device__ __forceinline__ void agg_update_slot(unsigned long long* aggp,
                                                 const DevPlan& p, int a,
                                                 DVal v)
@@ -3051,7 +3060,7 @@ __global__ void k_part_count(const OutGroup* groups, int64_t n, int nparts,
  * then place rows at block-local claims (order within a partition is
  * arbitrary — the merge is order-free) */
 __global__ void k_part_scatter(const OutGroup* groups, int64_t n, int nparts,
-                               int sum_slot, int sum_is_double,
+                               int sum_slot, int sum_is_double, int state_func,
                                unsigned long long* cursors,
                                YtStateRow* out)
 {
@@ -3086,7 +3095,16 @@ __global__ void k_part_scatter(const OutGroup* groups, int64_t n, int nparts,
         uint64_t nonnull = (sum_slot >= 0) ? g.agg_nonnull[sum_slot] : 0;
         sr.meta = (uint64_t)knull | (nonnull << 8)
                 | (sum_is_double ? 2ULL : 0ULL);
-        sr.sum_bits = (sum_slot >= 0) ? g.agg_bits[sum_slot] : 0;
+        uint64_t sb = (sum_slot >= 0) ? g.agg_bits[sum_slot] : 0;
+        if (sum_slot >= 0 && nonnull &&
+            (state_func == YT_AGG_MIN || state_func == YT_AGG_MAX)) {
+            /* the table stores min/max in the order-preserving mapped space
+             * (ord_map above); the exchanged STATE carries the RAW value so
+             * oracle- and GPU-produced states are interchangeable */
+            uint8_t vt = sum_is_double ? YT_VT_DOUBLE : YT_VT_INT64;
+            sb = ord_unmap(state_func == YT_AGG_MIN ? ~sb : sb, vt);
+        }
+        sr.sum_bits = sb;
         sr.row_count = g.cnt;
     }
 }
@@ -3425,7 +3443,7 @@ __global__ void k_strst_compact(const StrSlot* slots, uint64_t nslots,
 /* merge state rows into a (fresh) table — front-query Merge semantics
  * (cg_fragment_compiler.cpp:4116-4134 + udf/sum.c sum_merge). */
 __global__ void k_merge_states(const YtStateRow* states, int64_t n,
-                               int agg_count, int sum_slot,
+                               int agg_count, int sum_slot, int state_func,
                                TableHdr* th, unsigned long long* slots)
 {
     int stride = 2 + 2 * agg_count;
@@ -3448,7 +3466,14 @@ __global__ void k_merge_states(const YtStateRow* states, int64_t n,
         }
         atomicAdd(cntp, (unsigned long long)sr.row_count);
         if (sum_slot >= 0 && (sr.meta >> 8)) {
-            if (sr.meta & 2) {
+            if (state_func == YT_AGG_MIN || state_func == YT_AGG_MAX) {
+                /* states carry RAW values; the table accumulates in the
+                 * mapped space (same convention as the single-pass table) */
+                uint8_t vt = (sr.meta & 2) ? YT_VT_DOUBLE : YT_VT_INT64;
+                uint64_t m = ord_map(sr.sum_bits, vt);
+                atomicMax(aggp + 2 * sum_slot,
+                          (unsigned long long)(state_func == YT_AGG_MIN ? ~m : m));
+            } else if (sr.meta & 2) {
                 /* double sum state: FP merge (udf/sum.c:27-35) */
                 atomicAdd((double*)(aggp + 2 * sum_slot),
                           __longlong_as_double((long long)sr.sum_bits));
@@ -3800,6 +3825,7 @@ hipError_t ytql_launch_part_count(const OutGroup* groups, int64_t n, int nparts,
 
 hipError_t ytql_launch_part_scatter(const OutGroup* groups, int64_t n, int nparts,
                                     int sum_slot, int sum_is_double,
+                                    int state_func,
                                     unsigned long long* cursors,
                                     YtStateRow* out, hipStream_t st)
 {
@@ -3807,7 +3833,8 @@ hipError_t ytql_launch_part_scatter(const OutGroup* groups, int64_t n, int npart
     int64_t want = (n + block - 1) / block;
     int grid = (int)(want > 2048 ? 2048 : (want ? want : 1));
     hipLaunchKernelGGL(k_part_scatter, dim3(grid), dim3(block), 0, st,
-                       groups, n, nparts, sum_slot, sum_is_double, cursors, out);
+                       groups, n, nparts, sum_slot, sum_is_double, state_func,
+                       cursors, out);
     return hipGetLastError();
 }
 
@@ -3950,7 +3977,7 @@ hipError_t ytql_launch_strst_compact(const StrSlot* slots, uint64_t nslots,
 }
 
 hipError_t ytql_launch_merge_states(const YtStateRow* states, int64_t n,
-                                    int agg_count, int sum_slot,
+                                    int agg_count, int sum_slot, int state_func,
                                     TableHdr* th, unsigned long long* slots,
                                     hipStream_t st)
 {
@@ -3958,7 +3985,7 @@ hipError_t ytql_launch_merge_states(const YtStateRow* states, int64_t n,
     int64_t want = (n + block - 1) / block;
     int grid = (int)(want > 2048 ? 2048 : (want ? want : 1));
     hipLaunchKernelGGL(k_merge_states, dim3(grid), dim3(block), 0, st,
-                       states, n, agg_count, sum_slot, th, slots);
+                       states, n, agg_count, sum_slot, state_func, th, slots);
     return hipGetLastError();
 }
 
