@@ -1624,12 +1624,12 @@ int yto_partial(const YtPlan* plan, const YtChunk* chunk,
     for (int a = 0; a < plan->agg_count; a++) {
         int f = plan->aggs[a]->func;
         if (f == YT_AGG_SUM || f == YT_AGG_AVG || f == YT_AGG_MIN ||
-            f == YT_AGG_MAX) {
+            f == YT_AGG_MAX || f == YT_AGG_FIRST) {
             if (sum_idx >= 0) { set_err(errbuf, errlen, "partial: one value-carrying agg max this round"); return YT_ERR_UNSUPPORTED; }
             sum_idx = a;
             is_avg = (f == YT_AGG_AVG);
         }
-        else if (f != YT_AGG_SUM1) { set_err(errbuf, errlen, "partial: sum/avg/min/max/sum1 only"); return YT_ERR_UNSUPPORTED; }
+        else if (f != YT_AGG_SUM1) { set_err(errbuf, errlen, "partial: sum/avg/min/max/first/sum1 only"); return YT_ERR_UNSUPPORTED; }
     }
 
     /* run the local group-by via yto_execute on a plan without projection.
@@ -1739,14 +1739,16 @@ int yto_merge(const YtPlan* plan, const YtStateRow* states, int64_t nstates,
         for (int a = 0; a < plan->agg_count; a++) {
             int f = plan->aggs[a]->func;
             if (f != YT_AGG_SUM && f != YT_AGG_AVG && f != YT_AGG_MIN &&
-                f != YT_AGG_MAX) continue;
+                f != YT_AGG_MAX && f != YT_AGG_FIRST) continue;
             if (nonnull) {
                 Val nv;
                 nv.type = (states[i].meta & 2) ? YT_VT_DOUBLE : YT_VT_INT64;
                 nv.bits = states[i].sum_bits;
                 nv.str = 0; nv.len = 0;
                 Val* st2 = &t.states[g * plan->agg_count + a];
-                if (f == YT_AGG_MIN || f == YT_AGG_MAX)
+                if (f == YT_AGG_FIRST) {
+                    if (st2->type == YT_VT_NULL) *st2 = nv;
+                } else if (f == YT_AGG_MIN || f == YT_AGG_MAX)
                     minmax_update_val(st2, nv, f == YT_AGG_MAX);
                 else
                     sum_update_val(st2, nv);

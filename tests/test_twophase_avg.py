@@ -258,3 +258,56 @@ def test_gpu_minmax_states_cross_impl(cuda):
     union = y.oracle_merge(mm_plan(False), [(seg, counts[0])])
     want, _ = y.oracle_execute(mm_plan(False), ch)
     assert y.sort_rows(union) == y.sort_rows(want)
+
+
+def f_plan():
+    return y.Plan(keys=[y.col(0)], aggs=[y.agg_first(y.col(1)), y.agg_sum1()])
+
+
+def test_oracle_first_two_phase():
+    """first() through the exchange: the first ARRIVING non-null state
+    claims the slot (FirstIteration across tablets is likewise
+    order-arbitrary; the test data has ONE distinct value per group so the
+    pick is deterministic)."""
+    world = 3
+    shards = []
+    for r in range(world):
+        rng = np.random.default_rng(730 + r)
+        n = 3000
+        k = rng.integers(0, 40, n, dtype=np.int64)
+        v = k * 7 + 1                      # value determined by the group
+        vn = (rng.random(n) < 0.3).astype(np.uint8)
+        shards.append(y.Chunk([enc(k), y.encode_int64(v, vn)], n))
+    parts = [[] for _ in range(world)]
+    for ch in shards:
+        states, counts = y.oracle_partial(f_plan(), ch, world)
+        at = 0
+        for p in range(world):
+            seg = (YtStateRow * max(counts[p], 1))()
+            for i in range(counts[p]):
+                seg[i] = states[at + i]
+            parts[p].append((seg, counts[p]))
+            at += counts[p]
+    union = []
+    for p in range(world):
+        union += y.oracle_merge(f_plan(), parts[p])
+    for k, fv, cnt in union:
+        assert fv is None or fv == k * 7 + 1
+
+
+@pytest.mark.gpu
+def test_gpu_first_two_phase(cuda):
+    rng = np.random.default_rng(731)
+    n = 80_000
+    k = rng.integers(0, 211, n, dtype=np.int64)
+    v = k * 3 - 5
+    vn = (rng.random(n) < 0.2).astype(np.uint8)
+    ch = y.Chunk([y.encode_int64(k), y.encode_int64(v, vn)], n)
+    cap = 4 * 211 + 1024
+    states_t = cuda.zeros((cap, 4), dtype=cuda.int64, device="cuda")
+    counts, _ = y.gpu_partial(f_plan(), ch.c_device(cuda), 1,
+                              states_t.data_ptr(), cap, max_groups_hint=1024)
+    got, _ = y.gpu_merge(f_plan(), states_t.data_ptr(), sum(counts),
+                         max_groups_hint=1024, col_types=[VT_INT64, VT_INT64])
+    want, _ = y.oracle_execute(f_plan(), ch)
+    assert y.sort_rows(got) == y.sort_rows(want)

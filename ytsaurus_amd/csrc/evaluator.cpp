@@ -3667,12 +3667,12 @@ static int query_partial_impl(
     for (int a = 0; a < plan->agg_count; a++) {
         int f = plan->aggs[a]->func;
         if (f == YT_AGG_SUM || f == YT_AGG_AVG || f == YT_AGG_MIN ||
-            f == YT_AGG_MAX) {
+            f == YT_AGG_MAX || f == YT_AGG_FIRST) {
             if (sum_slot >= 0) { set_err(errbuf, errlen, "partial: one value-carrying agg max this round"); return YT_ERR_UNSUPPORTED; }
             sum_slot = a;
             state_func = f;
         } else if (f != YT_AGG_SUM1) {
-            set_err(errbuf, errlen, "partial: sum/avg/min/max/sum(1) only");
+            set_err(errbuf, errlen, "partial: sum/avg/min/max/first/sum(1) only");
             return YT_ERR_UNSUPPORTED;
         }
     }
@@ -3948,7 +3948,7 @@ static int merge_states_impl(
     for (int a = 0; a < plan->agg_count; a++) {
         int f = plan->aggs[a]->func;
         if (f == YT_AGG_SUM || f == YT_AGG_AVG || f == YT_AGG_MIN ||
-            f == YT_AGG_MAX) { sum_slot = a; state_func = f; }
+            f == YT_AGG_MAX || f == YT_AGG_FIRST) { sum_slot = a; state_func = f; }
     }
 
     YtExecOptions defopt;

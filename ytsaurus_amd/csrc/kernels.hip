@@ -3466,6 +3466,15 @@ __global__ void k_merge_states(const YtStateRow* states, int64_t n,
         }
         atomicAdd(cntp, (unsigned long long)sr.row_count);
         if (sum_slot >= 0 && (sr.meta >> 8)) {
+            if (state_func == YT_AGG_FIRST) {
+                /* FirstIteration merge: the first ARRIVING non-null state
+                 * claims the slot via CAS on the nonnull word (same idiom
+                 * as the single-pass agg_update_slot FIRST case) */
+                if (atomicCAS(aggp + 2 * sum_slot + 1, 0ULL,
+                              (unsigned long long)(sr.meta >> 8)) == 0ULL)
+                    aggp[2 * sum_slot] = sr.sum_bits;
+                continue;
+            }
             if (state_func == YT_AGG_MIN || state_func == YT_AGG_MAX) {
                 /* states carry RAW values; the table accumulates in the
                  * mapped space (same convention as the single-pass table) */
