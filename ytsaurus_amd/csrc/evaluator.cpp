@@ -4253,42 +4253,89 @@ extern "C" int yt_gpu_merge_states_str(
         HIP_CHECK(hipMemcpyAsync(hpool, d_opool, hctr[1],
                                  hipMemcpyDeviceToHost, st));
         HIP_CHECK(hipStreamSynchronize(st));
-        memcpy(output->string_pool, hpool, hctr[1]);
+        size_t total_b = (size_t)hctr[1];
+        int ct = (int)std::min<size_t>(std::thread::hardware_concurrency(),
+                                       (total_b + (64 << 20) - 1) / (64 << 20));
+        if (ct > 1) {
+            std::vector<std::thread> cth;
+            size_t per = (total_b + ct - 1) / ct;
+            for (int t2 = 0; t2 < ct; t2++) {
+                size_t b = (size_t)t2 * per;
+                size_t e = std::min(total_b, b + per);
+                if (b >= e) break;
+                cth.emplace_back([&, b, e] {
+                    memcpy(output->string_pool + b, hpool + b, e - b);
+                });
+            }
+            for (auto& th2 : cth) th2.join();
+        } else {
+            memcpy(output->string_pool, hpool, total_b);
+        }
     }
     output->string_pool_used = hctr[1];
     int ncols = 1 + plan->agg_count;
-    for (int64_t g = 0; g < ngroups + has_null; g++) {
-        const int is_null_row = g >= ngroups;
-        uint64_t cnt, sum_bits, nonnull, off_len = 0;
-        if (is_null_row) {
-            cnt = th.side_cnt[1];
-            sum_bits = sum_slot >= 0 ? th.side_agg[1][2 * sum_slot] : 0;
-            nonnull = sum_slot >= 0 ? th.side_agg[1][2 * sum_slot + 1] : 0;
-        } else {
+    int agg_is_sum1[kMaxAggs];
+    for (int a = 0; a < plan->agg_count; a++)
+        agg_is_sum1[a] = plan->aggs[a]->func == YT_AGG_SUM1;
+    auto emit_range = [&](int64_t b, int64_t e) {
+        for (int64_t g = b; g < e; g++) {
             const OutStrState& gg = hgroups[g];
-            cnt = gg.cnt;
-            sum_bits = gg.sum_bits;
-            nonnull = gg.nonnull;
-            off_len = gg.off_len;
+            YtValue* dst = output->values + g * ncols;
+            dst[0].id = 0;
+            dst[0].flags = 0;
+            dst[0].type = YT_VT_STRING;
+            dst[0].length = (uint32_t)(gg.off_len & 0xFFFFFF);
+            dst[0].data.str = output->string_pool + (gg.off_len >> 24);
+            for (int a = 0; a < plan->agg_count; a++) {
+                YtValue& v = dst[1 + a];
+                v.id = (uint16_t)(1 + a);
+                v.flags = 0;
+                v.length = 0;
+                if (agg_is_sum1[a]) {
+                    v.type = YT_VT_INT64;
+                    v.data.bits = gg.cnt;
+                } else if (gg.nonnull == 0) {
+                    v.type = YT_VT_NULL;
+                    v.data.bits = 0;
+                } else {
+                    v.type = sum_is_double ? YT_VT_DOUBLE : YT_VT_INT64;
+                    v.data.bits = gg.sum_bits;
+                }
+            }
         }
+    };
+    if (ngroups >= 65536) {
+        int nt = (int)std::min<int64_t>(std::thread::hardware_concurrency(),
+                                        (ngroups + 65535) / 65536);
+        std::vector<std::thread> ths;
+        int64_t per = (ngroups + nt - 1) / nt;
+        for (int t2 = 0; t2 < nt; t2++) {
+            int64_t b = (int64_t)t2 * per;
+            int64_t e = std::min<int64_t>(ngroups, b + per);
+            if (b >= e) break;
+            ths.emplace_back(emit_range, b, e);
+        }
+        for (auto& th2 : ths) th2.join();
+    } else {
+        emit_range(0, ngroups);
+    }
+    output->row_count = ngroups;
+    if (has_null) {
         YtValue* dst = output->values + output->row_count * ncols;
+        uint64_t cnt = th.side_cnt[1];
+        uint64_t sum_bits = sum_slot >= 0 ? th.side_agg[1][2 * sum_slot] : 0;
+        uint64_t nonnull = sum_slot >= 0 ? th.side_agg[1][2 * sum_slot + 1] : 0;
         dst[0].id = 0;
         dst[0].flags = 0;
-        if (is_null_row) {
-            dst[0].type = YT_VT_NULL;
-            dst[0].length = 0;
-            dst[0].data.bits = 0;
-        } else {
-            dst[0].type = YT_VT_STRING;
-            dst[0].length = (uint32_t)(off_len & 0xFFFFFF);
-            dst[0].data.str = output->string_pool + (off_len >> 24);
-        }
+        dst[0].type = YT_VT_NULL;
+        dst[0].length = 0;
+        dst[0].data.bits = 0;
         for (int a = 0; a < plan->agg_count; a++) {
             YtValue& v = dst[1 + a];
             v.id = (uint16_t)(1 + a);
             v.flags = 0;
             v.length = 0;
-            if (plan->aggs[a]->func == YT_AGG_SUM1) {
+            if (agg_is_sum1[a]) {
                 v.type = YT_VT_INT64;
                 v.data.bits = cnt;
             } else if (nonnull == 0) {
