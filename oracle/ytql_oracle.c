@@ -1979,9 +1979,16 @@ int yto_partial_mk(const YtPlan* plan, const YtChunk* chunk,
     if (plan->join) { set_err(errbuf, errlen, "partial: join with multi-key GROUP BY not this round"); return YT_ERR_UNSUPPORTED; }
     if (plan->key_count < 1 || plan->key_count > 4) { set_err(errbuf, errlen, "partial: 1..4 keys"); return YT_ERR_UNSUPPORTED; }
     int sum_idx = -1;
+    int is_avg = 0;
     for (int a = 0; a < plan->agg_count; a++) {
-        if (plan->aggs[a]->func == YT_AGG_SUM) sum_idx = a;
-        else if (plan->aggs[a]->func != YT_AGG_SUM1) { set_err(errbuf, errlen, "partial: sum/sum1 only"); return YT_ERR_UNSUPPORTED; }
+        int f = plan->aggs[a]->func;
+        if (f == YT_AGG_SUM || f == YT_AGG_AVG || f == YT_AGG_MIN ||
+            f == YT_AGG_MAX || f == YT_AGG_FIRST) {
+            if (sum_idx >= 0) { set_err(errbuf, errlen, "partial: one value-carrying agg max this round"); return YT_ERR_UNSUPPORTED; }
+            sum_idx = a;
+            is_avg = (f == YT_AGG_AVG);
+        }
+        else if (f != YT_AGG_SUM1) { set_err(errbuf, errlen, "partial: sum/avg/min/max/first/sum1 only"); return YT_ERR_UNSUPPORTED; }
     }
     uint8_t ct[64];
     memset(ct, YT_VT_INT64, sizeof(ct));
@@ -1994,8 +2001,23 @@ int yto_partial_mk(const YtPlan* plan, const YtChunk* chunk,
     YtPlan local = *plan;
     local.project_count = 0;
     local.projects = NULL;
+    YtAgg local_aggs[17];
+    const YtAgg* local_agg_ptrs[17];
+    if (is_avg) {
+        if (plan->agg_count > 16) { set_err(errbuf, errlen, "partial: too many aggregates"); return YT_ERR_UNSUPPORTED; }
+        for (int a = 0; a < plan->agg_count; a++) {
+            local_aggs[a] = *plan->aggs[a];
+            if (a == sum_idx) local_aggs[a].func = YT_AGG_SUM;
+            local_agg_ptrs[a] = &local_aggs[a];
+        }
+        local_aggs[plan->agg_count].func = YTO_AGG_NNCNT;
+        local_aggs[plan->agg_count].arg = plan->aggs[sum_idx]->arg;
+        local_agg_ptrs[plan->agg_count] = &local_aggs[plan->agg_count];
+        local.agg_count = plan->agg_count + 1;
+        local.aggs = local_agg_ptrs;
+    }
     int kc = plan->key_count;
-    int ncols = kc + plan->agg_count;
+    int ncols = kc + local.agg_count;
     int64_t cap = capacity_rows;
     YtValue* tmp = malloc(sizeof(YtValue) * cap * ncols);
     if (!tmp) { set_err(errbuf, errlen, "partial_mk: oom"); return YT_ERR_CAPACITY; }
@@ -2036,6 +2058,8 @@ int yto_partial_mk(const YtPlan* plan, const YtChunk* chunk,
                 if (row[kc + a].type == YT_VT_DOUBLE) sum_dbl = 2;
             }
         }
+        if (is_avg)
+            nonnull = row[kc + plan->agg_count].data.bits;   /* count leg */
         sr->meta = sum_dbl | (nonnull << 8);
         sr->sum_bits = sum_bits;
         sr->row_count = rowcount;
@@ -2067,13 +2091,22 @@ int yto_merge_mk(const YtPlan* plan, const YtStateRow* states, int64_t nstates,
         t.rowcounts[g] += states[i].row_count;
         uint64_t nonnull = states[i].meta >> 8;
         for (int a = 0; a < plan->agg_count; a++) {
-            if (plan->aggs[a]->func != YT_AGG_SUM) continue;
+            int f = plan->aggs[a]->func;
+            if (f != YT_AGG_SUM && f != YT_AGG_AVG && f != YT_AGG_MIN &&
+                f != YT_AGG_MAX && f != YT_AGG_FIRST) continue;
             if (nonnull) {
                 Val nv;
                 nv.type = (states[i].meta & 2) ? YT_VT_DOUBLE : YT_VT_INT64;
                 nv.bits = states[i].sum_bits;
                 nv.str = 0; nv.len = 0;
-                sum_update_val(&t.states[g * plan->agg_count + a], nv);
+                Val* st2 = &t.states[g * plan->agg_count + a];
+                if (f == YT_AGG_FIRST) {
+                    if (st2->type == YT_VT_NULL) *st2 = nv;
+                } else if (f == YT_AGG_MIN || f == YT_AGG_MAX)
+                    minmax_update_val(st2, nv, f == YT_AGG_MAX);
+                else
+                    sum_update_val(st2, nv);
+                t.acounts[g * plan->agg_count + a] += nonnull;
             }
         }
     }
@@ -2109,6 +2142,20 @@ int yto_merge_mk(const YtPlan* plan, const YtStateRow* states, int64_t nstates,
             if (plan->aggs[a]->func == YT_AGG_SUM1) {
                 v->type = YT_VT_INT64;
                 v->data.bits = t.rowcounts[g];
+            } else if (plan->aggs[a]->func == YT_AGG_AVG) {
+                uint64_t c2 = t.acounts[g * plan->agg_count + a];
+                Val st2 = t.states[g * plan->agg_count + a];
+                if (c2 == 0 || st2.type == YT_VT_NULL) {
+                    v->type = YT_VT_NULL;
+                    v->data.bits = 0;
+                } else {
+                    double sum2;
+                    if (st2.type == YT_VT_DOUBLE) memcpy(&sum2, &st2.bits, 8);
+                    else sum2 = (double)(int64_t)st2.bits;
+                    double rr = sum2 / (double)c2;
+                    v->type = YT_VT_DOUBLE;
+                    memcpy(&v->data.bits, &rr, 8);
+                }
             } else {
                 Val st2 = t.states[g * plan->agg_count + a];
                 v->type = st2.type;

@@ -154,3 +154,50 @@ def test_gpu_multikey_two_phase_matches_oracle_states(cuda):
         sum(len(s[0]) for s in shards))
     want, _ = y.oracle_execute(plan(), big)
     assert y.sort_rows(union) == y.sort_rows(want)
+
+
+def test_oracle_multikey_avg_minmax_two_phase():
+    """the mk exchange carries the same aggregate family as single-key:
+    avg (exact counts) and min/max (raw-value states)"""
+    from ytsaurus_amd._abi import AGG_MIN
+    world = 2
+    shards = make_shards(world, n=3000)
+    for aggs in ([y.agg_avg(y.col(2)), y.agg_sum1()],
+                 [y.agg_min(y.col(2)), y.agg_sum1()]):
+        p = y.Plan(keys=[y.col(0), y.col(1)], aggs=list(aggs))
+        ranges = reduce_ranges([oracle_key_ranges(s) for s in shards])
+        parts = [[] for _ in range(world)]
+        for r in range(world):
+            states, counts = y.oracle_partial_mk(p, chunk_of(shards[r]),
+                                                 world, ranges)
+            at = 0
+            for q in range(world):
+                seg = (type(states[0]) * max(counts[q], 1))()
+                for i in range(counts[q]):
+                    seg[i] = states[at + i]
+                parts[q].append((seg, counts[q]))
+                at += counts[q]
+        union = []
+        for q in range(world):
+            union += y.oracle_merge_mk(p, parts[q], ranges,
+                                       col_types=[VT_INT64, VT_INT64,
+                                                  VT_INT64])
+        big = y.Chunk(
+            [y.encode_int64(np.concatenate([s[0] for s in shards]),
+                            np.concatenate([s[1] for s in shards])),
+             y.encode_int64(np.concatenate([s[2] for s in shards])),
+             y.encode_int64(np.concatenate([s[3] for s in shards]),
+                            np.concatenate([s[4] for s in shards]))],
+            sum(len(s[0]) for s in shards))
+        want, _ = y.oracle_execute(p, big)
+        wm = {(r[0], r[1]): r for r in want}
+        assert len(union) == len(want)
+        for row in union:
+            w = wm[(row[0], row[1])]
+            assert row[3] == w[3]
+            if w[2] is None:
+                assert row[2] is None
+            elif isinstance(w[2], float):
+                assert row[2] == pytest.approx(w[2], rel=1e-9)
+            else:
+                assert row[2] == w[2]
