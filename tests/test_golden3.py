@@ -103,3 +103,51 @@ def test_golden3_gpu(cuda):
     chunk = y.Chunk([enc([1] * 10), enc(list(range(10)))], 10)
     plan = y.Plan(keys=[y.col(0)], aggs=[y.agg_first(y.col(1))])
     assert run(plan, chunk, cuda) == [(1, 0)]
+
+
+# GroupByOrderBy3 (:2704-2758): filter + group + order by key + limit
+def test_golden_group_order3():
+    a = list(range(1, 10))
+    b = [0, 1, 2] * 3
+    chunk = y.Chunk([enc(a), enc(b)], 9)
+    # "sum(a) as t, b where b = 0 group by b order by b limit 3" -> t=12;b=0
+    plan = y.Plan(filter=y.col(1) == 0, keys=[y.col(1)],
+                  aggs=[y.agg_sum(y.col(0))], order_by=[(0, False)], limit=3)
+    rows, _ = y.oracle_execute(plan, chunk)
+    assert rows == [(0, 12)]            # our emit order: [key, agg]
+    # "... where b = 4 ..." -> empty
+    plan = y.Plan(filter=y.col(1) == 4, keys=[y.col(1)],
+                  aggs=[y.agg_sum(y.col(0))], order_by=[(0, False)], limit=3)
+    rows, _ = y.oracle_execute(plan, chunk)
+    assert rows == []
+
+
+@pytest.mark.gpu
+def test_golden_group_order3_gpu(cuda):
+    a = list(range(1, 10))
+    b = [0, 1, 2] * 3
+    chunk = y.Chunk([enc(a), enc(b)], 9)
+    plan = y.Plan(filter=y.col(1) == 0, keys=[y.col(1)],
+                  aggs=[y.agg_sum(y.col(0))], order_by=[(0, False)], limit=3)
+    rows, _ = y.gpu_execute(plan, chunk.c_device(cuda), max_groups_hint=16)
+    assert rows == [(0, 12)]
+    plan = y.Plan(filter=y.col(1) == 4, keys=[y.col(1)],
+                  aggs=[y.agg_sum(y.col(0))], order_by=[(0, False)], limit=3)
+    rows, _ = y.gpu_execute(plan, chunk.c_device(cuda), max_groups_hint=16)
+    assert rows == []
+
+
+# GroupByOrderBy2 first case (:2623-2668): string+int multi-key GROUP BY,
+# ordered by the keys — oracle (string components refuse GPU composite
+# packing loudly)
+def test_golden_group_order2_string_multikey():
+    a = list(range(1, 10))
+    bs = ["a", "a", "b", "a", "b", "a", "b", "b", "a"]
+    c = [1, 2, 3, 4, 1, 2, 3, 4, 1]
+    chunk = y.Chunk([enc(a), y.encode_string(bs), enc(c)], 9)
+    plan = y.Plan(keys=[y.col(1), y.col(2)], aggs=[y.agg_sum(y.col(0))],
+                  order_by=[(0, False), (1, False)], limit=6)
+    rows, _ = y.oracle_execute(plan, chunk)
+    # reference rows (t,b,c) reordered to our [b, c, t] emit
+    assert rows == [(b"a", 1, 10), (b"a", 2, 8), (b"a", 4, 4),
+                    (b"b", 1, 5), (b"b", 3, 10), (b"b", 4, 8)]
