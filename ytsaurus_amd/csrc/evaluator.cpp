@@ -2782,11 +2782,8 @@ static int run_string_group(const YtPlan* plan, const YtChunk* chunk,
         HIP_CHECK(pool_alloc_host(&hgroups,
                                   sizeof(OutStrGroup) * (ngroups ? ngroups : 1)));
         unsigned long long hctr[2] = {0, 0};
-        if (ngroups > 0 && !po) {
-            HIP_CHECK(hipMemcpyAsync(hgroups, d_out,
-                                     sizeof(OutStrGroup) * ngroups,
-                                     hipMemcpyDeviceToHost, R.stream));
-        }
+        /* the fast emit path below streams hgroups in SLICES so the host
+         * emit overlaps the remaining D2H; the limited path copies here */
         HIP_CHECK(hipMemcpyAsync(hctr, d_ctr, 2 * sizeof(unsigned long long),
                                  hipMemcpyDeviceToHost, R.stream));
         HIP_CHECK(hipEventRecord(e2, R.stream));
@@ -2924,34 +2921,60 @@ static int run_string_group(const YtPlan* plan, const YtChunk* chunk,
             ng + has_null_row <= output->capacity_rows &&
             (unsigned long long)hctr[1] <= (unsigned long long)output->string_pool_capacity &&
             (output->string_pool != nullptr || hctr[1] == 0)) {
+            /* pipeline: pool D2H first, then hgroups in slices — the
+             * parallel host copies/emits of slice s overlap slice s+1's
+             * D2H (the serial tail was D2H then copy then emit) */
+            char* stage = nullptr;
             if (hctr[1]) {
-                /* stage through pinned memory + parallel host copy: a sync
-                 * D2H into the caller's pageable pool runs at a fraction of
-                 * the pinned rate at ~1 GB of key bytes */
-                char* stage = nullptr;
                 HIP_CHECK(pool_alloc_host(&stage, hctr[1]));
                 HIP_CHECK(hipMemcpyAsync(stage, d_pool, hctr[1],
                                          hipMemcpyDeviceToHost, R.stream));
-                HIP_CHECK(hipStreamSynchronize(R.stream));
-                size_t total_b = (size_t)hctr[1];
-                int ct = (int)std::min<size_t>(std::thread::hardware_concurrency(),
-                                               (total_b + (64 << 20) - 1) / (64 << 20));
-                if (ct > 1) {
-                    std::vector<std::thread> cth;
-                    size_t per = (total_b + ct - 1) / ct;
-                    for (int t = 0; t < ct; t++) {
-                        size_t b = (size_t)t * per;
-                        size_t e = std::min(total_b, b + per);
-                        if (b >= e) break;
-                        cth.emplace_back([&, b, e] {
-                            memcpy(output->string_pool + b, stage + b, e - b);
-                        });
-                    }
-                    for (auto& th2 : cth) th2.join();
-                } else {
-                    memcpy(output->string_pool, stage, total_b);
+            }
+            const int kSlices = (ng >= (1 << 21)) ? 8 : 1;
+            std::vector<hipEvent_t> sl_ev(kSlices, nullptr);
+            int64_t per_sl = (ng + kSlices - 1) / kSlices;
+            for (int s2 = 0; s2 < kSlices; s2++) {
+                int64_t b = (int64_t)s2 * per_sl;
+                int64_t e = std::min<int64_t>(ng, b + per_sl);
+                if (b >= e) break;
+                HIP_CHECK(hipMemcpyAsync(hgroups + b, d_out + b,
+                                         sizeof(OutStrGroup) * (e - b),
+                                         hipMemcpyDeviceToHost, R.stream));
+                HIP_CHECK(hipEventCreate(&sl_ev[s2]));
+                HIP_CHECK(hipEventRecord(sl_ev[s2], R.stream));
+            }
+            std::thread pool_thr;
+            if (hctr[1]) {
+                /* pool D2H was queued first — wait for the FIRST slice event
+                 * (>= pool completion), then copy to the caller's pool off
+                 * the critical path */
+                hipEvent_t after_pool = sl_ev[0];
+                if (!after_pool) {
+                    HIP_CHECK(hipStreamSynchronize(R.stream));
                 }
-                g_pool.put(stage);
+                size_t total_b = (size_t)hctr[1];
+                char* dst_pool = output->string_pool;
+                pool_thr = std::thread([stage, dst_pool, total_b, after_pool] {
+                    if (after_pool) (void)hipEventSynchronize(after_pool);
+                    int ct = (int)std::min<size_t>(
+                        std::thread::hardware_concurrency(),
+                        (total_b + (64 << 20) - 1) / (64 << 20));
+                    if (ct > 1) {
+                        std::vector<std::thread> cth;
+                        size_t per = (total_b + ct - 1) / ct;
+                        for (int t = 0; t < ct; t++) {
+                            size_t b = (size_t)t * per;
+                            size_t e = std::min(total_b, b + per);
+                            if (b >= e) break;
+                            cth.emplace_back([=] {
+                                memcpy(dst_pool + b, stage + b, e - b);
+                            });
+                        }
+                        for (auto& th2 : cth) th2.join();
+                    } else if (total_b) {
+                        memcpy(dst_pool, stage, total_b);
+                    }
+                });
             }
             output->string_pool_used = hctr[1];
             int agg_is_sum1[kMaxAggs];
@@ -2984,22 +3007,34 @@ static int run_string_group(const YtPlan* plan, const YtChunk* chunk,
                     }
                 }
             };
-            int nt = (int)std::thread::hardware_concurrency();
-            if (nt < 1) nt = 1;
-            if (nt > 64) nt = 64;
-            if (ng < (1 << 16)) nt = 1;
-            if (nt == 1) {
-                emit_range(0, ng);
-            } else {
+            {
+                int nt = (int)std::thread::hardware_concurrency();
+                if (nt < 1) nt = 1;
+                if (nt > 64) nt = 64;
+                if (ng < (1 << 16)) nt = 1;
                 std::vector<std::thread> ths;
-                int64_t per = (ng + nt - 1) / nt;
-                for (int t = 0; t < nt; t++) {
-                    int64_t b = (int64_t)t * per, e = b + per;
-                    if (b >= ng) break;
-                    if (e > ng) e = ng;
-                    ths.emplace_back(emit_range, b, e);
+                for (int s2 = 0; s2 < kSlices; s2++) {
+                    int64_t sb = (int64_t)s2 * per_sl;
+                    int64_t se = std::min<int64_t>(ng, sb + per_sl);
+                    if (sb >= se) break;
+                    if (sl_ev[s2]) HIP_CHECK(hipEventSynchronize(sl_ev[s2]));
+                    if (nt == 1) {
+                        emit_range(sb, se);
+                        continue;
+                    }
+                    int kt = nt / kSlices > 0 ? nt / kSlices : 1;
+                    int64_t per = (se - sb + kt - 1) / kt;
+                    for (int t = 0; t < kt; t++) {
+                        int64_t b = sb + (int64_t)t * per, e = b + per;
+                        if (b >= se) break;
+                        if (e > se) e = se;
+                        ths.emplace_back(emit_range, b, e);
+                    }
                 }
                 for (auto& t : ths) t.join();
+                if (pool_thr.joinable()) pool_thr.join();
+                for (auto& ev : sl_ev) if (ev) (void)hipEventDestroy(ev);
+                g_pool.put(stage);
             }
             output->row_count = ng;
             if (has_null_row) {
@@ -3036,6 +3071,12 @@ static int run_string_group(const YtPlan* plan, const YtChunk* chunk,
         std::vector<char> pool(hctr[1] ? hctr[1] : 1);
         if (hctr[1]) {
             HIP_CHECK(hipMemcpy(pool.data(), d_pool, hctr[1], hipMemcpyDeviceToHost));
+        }
+        if (ngroups > 0) {
+            /* the fast path streams hgroups in slices; this limited path
+             * copies them in one go */
+            HIP_CHECK(hipMemcpy(hgroups, d_out, sizeof(OutStrGroup) * ngroups,
+                                hipMemcpyDeviceToHost));
         }
         for (int64_t gI = 0; gI < ngroups + 2; gI++) {
             const char* kstr = nullptr;
