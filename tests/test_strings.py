@@ -114,3 +114,49 @@ def test_string_filter_oracle():
                   keys=[y.col(0)], aggs=[y.agg_sum1()])
     rows, _ = y.oracle_execute(plan, chunk2)
     assert rows == [(b"banana", 2)]
+
+
+@pytest.mark.gpu
+def test_gpu_direct_string_keys(cuda):
+    """direct-dense string KEY segments group on the GPU via the identity
+    dictionary (entry j = row j; the per-entry accumulators just have one
+    reference each) — string_column_writer.cpp DumpDirectValues layout."""
+    rng = np.random.default_rng(91)
+    n = 60_000
+    # mostly-unique strings => the writer's min-size rule picks DirectDense
+    keys = ["u%07d" % int(i) for i in rng.integers(0, 10**7, n)]
+    kn = rng.random(n) < 0.02
+    keys = [None if kn[i] else keys[i] for i in range(n)]
+    v = rng.integers(0, 1000, n, dtype=np.int64)
+    chunk = y.Chunk([y.encode_string(keys), y.encode_int64(v)], n)
+    # assert the chosen layout really is direct (type 3 = DirectDense)
+    types = {chunk.columns[0]._cenc.segments[j].type
+             for j in range(chunk.columns[0]._cenc.segment_count)}
+    assert 3 in types, types
+    plan = y.Plan(keys=[y.col(0)], aggs=[y.agg_sum(y.col(1)), y.agg_sum1()])
+    got, _ = y.gpu_execute(plan, chunk.c_device(cuda),
+                           max_groups_hint=n + 16,
+                           out_capacity=n + 16)
+    want, _ = y.oracle_execute(plan, chunk)
+    assert y.sort_rows(got) == y.sort_rows(want)
+
+
+@pytest.mark.gpu
+def test_gpu_mixed_direct_dict_string_segments(cuda):
+    """per-segment layout choice can differ across one column: repeated keys
+    in the first 8Ki rows (dictionary) then unique keys (direct)"""
+    rep = ["k%03d" % (i % 50) for i in range(8192)]
+    uni = ["z%06d" % i for i in range(8192)]
+    keys = rep + uni
+    n = len(keys)
+    v = np.arange(n, dtype=np.int64)
+    chunk = y.Chunk([y.encode_string(keys, max_segment_values=8192),
+                     y.encode_int64(v, max_segment_values=8192)], n)
+    types = [chunk.columns[0]._cenc.segments[j].type
+             for j in range(chunk.columns[0]._cenc.segment_count)]
+    assert len(set(types)) == 2, types     # one dict + one direct segment
+    plan = y.Plan(keys=[y.col(0)], aggs=[y.agg_sum(y.col(1)), y.agg_sum1()])
+    got, _ = y.gpu_execute(plan, chunk.c_device(cuda),
+                           max_groups_hint=n + 16, out_capacity=n + 16)
+    want, _ = y.oracle_execute(plan, chunk)
+    assert y.sort_rows(got) == y.sort_rows(want)

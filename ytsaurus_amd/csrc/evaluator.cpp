@@ -2629,9 +2629,10 @@ static int run_string_group(const YtPlan* plan, const YtChunk* chunk,
         HIP_CHECK(hipMemcpy(ksegex.data(), R.d_segex + koff,
                             sizeof(SegEx) * knseg, hipMemcpyDeviceToHost));
         for (int i = 0; i < knseg; i++) {
-            if (!(ksegex[i].flags & 8)) {
+            if (!(ksegex[i].flags & (8 | 32))) {
                 set_err(errbuf, errlen,
-                        "string keys: only dictionary-encoded segments this round");
+                        "string keys: dictionary or direct-dense segments "
+                        "(RLE strings not this round)");
                 return YT_ERR_UNSUPPORTED;
             }
         }

@@ -218,8 +218,23 @@ __global__ void k_parse_segments(const DevSeg* segs, int nsegs, SegEx* out,
             e.flags |= 8;
             break;
         }
+        case YT_SEG_DIRECT_DENSE: {
+            /* [offsets, diff-from-expected][null bitmap][data]
+             * (string_column_writer.cpp DumpDirectValues). As a GROUP key
+             * this rides the dictionary machinery with an IDENTITY
+             * dictionary: entry j = row j's string, null rows skipped via
+             * the bitmap (their accumulators stay zero). */
+            e.w_values = w0;
+            e.dict_size = (uint32_t)n0;      /* = row_count */
+            e.off_values_words = 1;
+            e.off_bitmap_bytes = words0 * 8;
+            int64_t bm = (((int64_t)s.row_count + 7) / 8 + 7) & ~(int64_t)7;
+            e.off_doubles_bytes = words0 * 8 + bm;
+            e.flags |= 32;  /* direct-dense string segment */
+            break;
+        }
         default:
-            e.flags |= 16;  /* direct string segment: no GPU key path yet */
+            e.flags |= 16;  /* RLE direct string: no GPU key path yet */
             break;
         }
         out[i] = e;
@@ -2562,6 +2577,8 @@ k_strgrp_accum(StrGroupParams sp, const DevSeg* segs, const SegEx* segex,
         }
 
         const bool is_rle = (sk.type == YT_SEG_DICTIONARY_RLE);
+        const bool is_direct = (ek.flags & 32) != 0;
+        const uint8_t* kbm = (const uint8_t*)sk.blob + ek.off_bitmap_bytes;
         const uint64_t* ids = sk.blob + ek.off_ids_words;
         const uint64_t* starts = sk.blob + ek.off_starts_words;
         const int R = (sp.tile_rows + 255) / 256;
@@ -2569,7 +2586,10 @@ k_strgrp_accum(StrGroupParams sp, const DevSeg* segs, const SegEx* segex,
             int64_t j = t0 + (int64_t)i * 256 + threadIdx.x;
             if (j >= t1) continue;
             uint64_t id;
-            if (is_rle) {
+            if (is_direct) {
+                /* identity dictionary: row j IS entry j; nulls by bitmap */
+                id = bm_get(kbm, j) ? 0 : (uint64_t)(j + 1);
+            } else if (is_rle) {
                 uint32_t lo = 0, hi = ek.run_count;
                 while (lo + 1 < hi) {
                     uint32_t mid = (lo + hi) / 2;
