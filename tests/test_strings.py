@@ -160,3 +160,25 @@ def test_gpu_mixed_direct_dict_string_segments(cuda):
                            max_groups_hint=n + 16, out_capacity=n + 16)
     want, _ = y.oracle_execute(plan, chunk)
     assert y.sort_rows(got) == y.sort_rows(want)
+
+
+@pytest.mark.gpu
+def test_gpu_string_keys_dict_value_column(cuda):
+    """REGRESSION: the sum-argument column can itself be DICTIONARY (or
+    RLE) encoded — repeating int values make the writer pick it — and the
+    accumulate must use the generic per-row fetch for those layouts (the
+    fast path only decodes DirectDense/double)."""
+    rng = np.random.default_rng(92)
+    n = 30_000
+    keyset = ["k%05d" % i for i in range(300)]
+    keys = [keyset[int(i)] for i in rng.integers(0, 300, n)]
+    v = rng.integers(0, 50, n, dtype=np.int64)      # 50 distinct -> dict
+    vn = (rng.random(n) < 0.1).astype(np.uint8)
+    chunk = y.Chunk([y.encode_string(keys), y.encode_int64(v, vn)], n)
+    vtypes = {chunk.columns[1]._cenc.segments[j].type
+              for j in range(chunk.columns[1]._cenc.segment_count)}
+    assert vtypes & {0, 1}, vtypes          # dictionary layout chosen
+    plan = y.Plan(keys=[y.col(0)], aggs=[y.agg_sum(y.col(1)), y.agg_sum1()])
+    got, _ = y.gpu_execute(plan, chunk.c_device(cuda), max_groups_hint=1024)
+    want, _ = y.oracle_execute(plan, chunk)
+    assert y.sort_rows(got) == y.sort_rows(want)

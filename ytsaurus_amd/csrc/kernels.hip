@@ -2604,16 +2604,28 @@ k_strgrp_accum(StrGroupParams sp, const DevSeg* segs, const SegEx* segex,
             uint64_t vbits = 0;
             double vdbl = 0;
             if (sv) {
-                val_null = vbm && bm_get(vbm, j);
-                if (!val_null) {
-                    if (sv->type == YT_SEG_DOUBLE) {
-                        vbits = ((const uint64_t*)((const uint8_t*)sv->blob
-                                 + ev->off_doubles_bytes))[j];
-                        vdbl = __longlong_as_double(vbits);
-                    } else {
-                        uint64_t pv = bp_gl(vwords, vmask, vwd, j);
-                        vbits = (uint64_t)zz_dec(vmin + pv);
+                if (sv->type == YT_SEG_DOUBLE ||
+                    (sv->type == YT_SEG_DIRECT_DENSE && sv->is_signed != 2)) {
+                    val_null = vbm && bm_get(vbm, j);
+                    if (!val_null) {
+                        if (sv->type == YT_SEG_DOUBLE) {
+                            vbits = ((const uint64_t*)((const uint8_t*)sv->blob
+                                     + ev->off_doubles_bytes))[j];
+                            vdbl = __longlong_as_double(vbits);
+                        } else {
+                            uint64_t pv = bp_gl(vwords, vmask, vwd, j);
+                            vbits = (uint64_t)zz_dec(vmin + pv);
+                        }
                     }
+                } else {
+                    /* dictionary / RLE value segments: generic per-row
+                     * fetch (handles the layout's own null encoding) */
+                    DVal dv = seg_value_at(*sv, *ev, j,
+                                           sp.val_is_double ? YT_VT_DOUBLE
+                                                            : YT_VT_INT64);
+                    val_null = dv.null_;
+                    vbits = dv.bits;
+                    if (sp.val_is_double) vdbl = __longlong_as_double(dv.bits);
                 }
             }
             if (id == 0) {
