@@ -83,7 +83,7 @@ def _rand_plan(rng, ncols, with_join_cols=0):
     if kc or rng.random() < 0.7:
         na = int(rng.integers(1, 4))
         for _ in range(na):
-            f = rng.integers(0, 4)
+            f = rng.integers(0, 5)
             c = y.col(int(rng.integers(0, nc)))
             if f == 0:
                 aggs.append(y.agg_sum(c))
@@ -91,8 +91,11 @@ def _rand_plan(rng, ncols, with_join_cols=0):
                 aggs.append(y.agg_sum1())
             elif f == 2:
                 aggs.append((AGG_MIN, c))
-            else:
+            elif f == 3:
                 aggs.append((AGG_MAX, c))
+            else:
+                # avg over int args is exact (int sum / exact count)
+                aggs.append(y.agg_avg(c))
     projects = []
     if not keys and not aggs:
         projects = [_rand_expr(rng, nc) for _ in range(int(rng.integers(1, 4)))]
@@ -125,9 +128,15 @@ def test_fuzz_parity(cuda, seed):
     join = None
     jdev = None
     jcols = 0
+    join_dups = False
     if rng.random() < 0.3:
         fn = int(rng.integers(1, 500))
         fkey = rng.permutation(np.arange(fn, dtype=np.int64))
+        if rng.random() < 0.4 and fn >= 4:
+            # duplicate foreign keys: cross-product expansion (grouped only)
+            join_dups = True
+            fkey = np.concatenate([fkey, fkey[:3]])
+            fn = len(fkey)
         fval = rng.integers(-10**6, 10**6, fn, dtype=np.int64)
         fchunk = y.Chunk([y.encode_int64(fkey), y.encode_int64(fval)], fn)
         join = y.Join(fchunk, int(rng.integers(0, ncols)), 0, [1],
@@ -137,6 +146,15 @@ def test_fuzz_parity(cuda, seed):
 
     kw = _rand_plan(rng, ncols, with_join_cols=jcols)
     kw["join"] = join
+    if join_dups:
+        # dup keys + ORDER BY / plain scan are refused loudly; steer the
+        # fuzz onto the supported grouped shape
+        if not kw["keys"]:
+            kw["keys"] = [y.col(0)]
+            if not kw["aggs"]:
+                kw["aggs"] = [y.agg_sum1()]
+            kw["projects"] = []
+        kw["order_by"], kw["limit"], kw["offset"] = (), 0, 0
     plan = y.Plan(**kw)
 
     def run_gpu():

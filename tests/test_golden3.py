@@ -151,3 +151,28 @@ def test_golden_group_order2_string_multikey():
     # reference rows (t,b,c) reordered to our [b, c, t] emit
     assert rows == [(b"a", 1, 10), (b"a", 2, 8), (b"a", 4, 4),
                     (b"b", 1, 5), (b"b", 3, 10), (b"b", 4, 8)]
+
+
+# GroupByAlias (:3805-3838): "a % 3 as a, sum(a + b) as b group by a" — the
+# alias SHADOWS the source column inside the aggregate (sum adds the
+# aliased a%3, not the raw a: group 1 = (1+10)+(1+40)+(1+70) = 123). Our
+# seam is post-parse, so the shadowing is restated explicitly.
+def test_golden_group_by_alias():
+    a = list(range(1, 10))
+    b = [10 * i for i in a]
+    chunk = y.Chunk([enc(a), enc(b)], 9)
+    plan = y.Plan(keys=[y.col(0) % 3],
+                  aggs=[y.agg_sum((y.col(0) % 3) + y.col(1))])
+    rows = run(plan, chunk)
+    assert y.sort_rows(rows) == y.sort_rows([(1, 123), (2, 156), (0, 180)])
+
+
+@pytest.mark.gpu
+def test_golden_group_by_alias_gpu(cuda):
+    a = list(range(1, 10))
+    b = [10 * i for i in a]
+    chunk = y.Chunk([enc(a), enc(b)], 9)
+    plan = y.Plan(keys=[y.col(0) % 3],
+                  aggs=[y.agg_sum((y.col(0) % 3) + y.col(1))])
+    rows = run(plan, chunk, cuda)
+    assert y.sort_rows(rows) == y.sort_rows([(1, 123), (2, 156), (0, 180)])
