@@ -201,3 +201,49 @@ def test_oracle_multikey_avg_minmax_two_phase():
                 assert row[2] == pytest.approx(w[2], rel=1e-9)
             else:
                 assert row[2] == w[2]
+
+
+def test_coordinator_combine_totals_and_order():
+    """the coordinator tail (DoCoordinateAndExecute, executor.cpp:761):
+    key-disjoint front outputs concatenate, per-partition TOTALS rows
+    re-fold into one, and the global ORDER BY ... LIMIT merges the
+    per-partition-ordered unions — combined result == the single-pass
+    query (GroupByCoordinatedWithTotals* family semantics)."""
+    from ytsaurus_amd._abi import YtStateRow
+    world = 3
+    rng = np.random.default_rng(61)
+    n = 9000
+    k = rng.integers(-50, 50, n, dtype=np.int64)
+    v = rng.integers(0, 10**6, n, dtype=np.int64)
+    vn = (rng.random(n) < 0.1).astype(np.uint8)
+    shards = []
+    for r in range(world):
+        sl = slice(r * n // world, (r + 1) * n // world)
+        shards.append(y.Chunk([y.encode_int64(k[sl]),
+                               y.encode_int64(v[sl], vn[sl])],
+                              len(k[sl])))
+
+    def planf():
+        return y.Plan(keys=[y.col(0)],
+                      aggs=[y.agg_sum(y.col(1)), y.agg_sum1()],
+                      order_by=[(1, True)], limit=12, with_totals=True)
+
+    parts = [[] for _ in range(world)]
+    for s in shards:
+        states, counts = y.oracle_partial(
+            y.Plan(keys=[y.col(0)],
+                   aggs=[y.agg_sum(y.col(1)), y.agg_sum1()]), s, world)
+        at = 0
+        for p in range(world):
+            seg = (YtStateRow * max(counts[p], 1))()
+            for i in range(counts[p]):
+                seg[i] = states[at + i]
+            parts[p].append((seg, counts[p]))
+            at += counts[p]
+    results = [y.oracle_merge(planf(), parts[p]) for p in range(world)]
+    combined = y.coordinate_results(planf(), results)
+    big = y.Chunk([y.encode_int64(k), y.encode_int64(v, vn)], n)
+    want, _ = y.oracle_execute(planf(), big)
+    assert combined[-1] == want[-1]                  # totals row
+    assert [r[1] for r in combined[:-1]] == [r[1] for r in want[:-1]]
+    assert len(combined) == len(want)

@@ -27,7 +27,7 @@ __all__ = [
     "gpu_key_ranges", "gpu_partial_mk", "gpu_merge_mk",
     "oracle_partial_str", "oracle_merge_str", "gpu_partial_str", "gpu_merge_str",
     "gpu_available", "gpu_execute", "gpu_partial", "gpu_merge",
-    "rows_from_rowset", "sort_rows", "make_rowset",
+    "rows_from_rowset", "sort_rows", "make_rowset", "coordinate_results",
 ]
 
 
@@ -911,6 +911,82 @@ def oracle_merge(plan, states_list):
                                      err, 256)
     _check(rc, err)
     return rows_from_rowset(rs)
+
+
+def coordinate_results(plan, partition_results):
+    """COORDINATOR combine for the key-partitioned front queries (the
+    reference's DoCoordinateAndExecute tail, ytlib/query_client/
+    executor.cpp:761: each partition's front query produced its slice of
+    the grouped output; the coordinator concatenates the key-disjoint
+    rows, re-folds the per-partition TOTALS rows into one, and applies the
+    global ORDER BY ... LIMIT over the per-partition-ordered unions —
+    valid because every partition already kept its own top
+    (offset+limit)). partition_results: list of row lists as returned by
+    oracle_merge / gpu_merge (totals row LAST when plan.with_totals)."""
+    from ytsaurus_amd._abi import (AGG_SUM, AGG_SUM1, AGG_MIN, AGG_MAX,
+                                   AGG_FIRST)
+    kc = len(plan.keys)
+    aggs = [a if isinstance(a, tuple) else a for a in plan.aggs]
+    rows = []
+    totals_parts = []
+    for pr in partition_results:
+        pr = list(pr)
+        if plan.with_totals and pr:
+            totals_parts.append(pr[-1])
+            pr = pr[:-1]
+        rows += pr
+
+    if plan.order_by:
+        def cmp_key(r):
+            out = []
+            for col, desc in plan.order_by:
+                v = r[col]
+                null = v is None
+                k = (0 if null else 1, v if not null else 0)
+                if desc:
+                    out.append((-k[0], _neg(v)))
+                else:
+                    out.append(k)
+            return tuple(out)
+        def _neg(v):
+            if v is None:
+                return 0
+            if isinstance(v, bytes):
+                return tuple(-b for b in v) + (1,)   # inverted memcmp
+            return -v
+        rows.sort(key=cmp_key)
+        off = plan.offset or 0
+        rows = rows[off:off + plan.limit] if plan.limit else rows[off:]
+
+    if plan.with_totals and totals_parts:
+        tot = list(totals_parts[0])
+        for tp in totals_parts[1:]:
+            for ai, agg in enumerate(plan.aggs):
+                f = agg[0]
+                i = kc + ai
+                a, b = tot[i], tp[i]
+                if f == AGG_SUM1:
+                    tot[i] = (a or 0) + (b or 0)
+                elif f == AGG_SUM:
+                    if b is not None:
+                        tot[i] = b if a is None else (
+                            a + b if isinstance(a, float) else _wrap(a + b))
+                elif f == AGG_MIN:
+                    if b is not None:
+                        tot[i] = b if a is None else min(a, b)
+                elif f == AGG_MAX:
+                    if b is not None:
+                        tot[i] = b if a is None else max(a, b)
+                elif f == AGG_FIRST:
+                    if a is None:
+                        tot[i] = b
+        rows.append(tuple(tot))
+    return rows
+
+
+def _wrap(v):
+    v &= (1 << 64) - 1
+    return v - (1 << 64) if v >= (1 << 63) else v
 
 
 def gpu_available():
