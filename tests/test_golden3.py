@@ -176,3 +176,54 @@ def test_golden_group_by_alias_gpu(cuda):
                   aggs=[y.agg_sum((y.col(0) % 3) + y.col(1))])
     rows = run(plan, chunk, cuda)
     assert y.sort_rows(rows) == y.sort_rows([(1, 123), (2, 156), (0, 180)])
+
+
+# JoinEmpty (:5117-5164): odd b's join even b's -> nothing; group by a
+# FOREIGN expression (c % 2)
+def test_golden_join_empty(cuda=None):
+    left = y.Chunk([enc([1, 3, 5, 7, 9]), enc([10, 30, 50, 70, 90])], 5)
+    right = y.Chunk([enc([20, 40, 60, 80]), enc([2, 4, 6, 8])], 4)
+    j = y.Join(right, primary_key_col=1, foreign_key_col=0, value_cols=[1])
+    plan = y.Plan(keys=[y.col(2) % 2],
+                  aggs=[y.agg_sum(y.col(0)), y.agg_sum(y.col(1))], join=j)
+    rows, _ = y.oracle_execute(plan, left)
+    assert rows == []
+
+
+# JoinSimple2 (:5166-5203): unique keys, plain-scan join projection
+def test_golden_join_simple2():
+    left = y.Chunk([enc([1, 2])], 1 + 1)
+    right = y.Chunk([enc([2, 1])], 2)
+    j = y.Join(right, primary_key_col=0, foreign_key_col=0, value_cols=[])
+    plan = y.Plan(projects=[y.col(0)], join=j)
+    rows, _ = y.oracle_execute(plan, left)
+    assert sorted(r[0] for r in rows) == [1, 2]
+
+
+# JoinSimple3 (:5205-5242): DUPLICATE PRIMARY rows (a=1,1) x unique foreign
+def test_golden_join_simple3():
+    left = y.Chunk([enc([1, 1])], 2)
+    right = y.Chunk([enc([2, 1])], 2)
+    j = y.Join(right, primary_key_col=0, foreign_key_col=0, value_cols=[])
+    plan = y.Plan(projects=[y.col(0)], join=j)
+    rows, _ = y.oracle_execute(plan, left)
+    assert [r[0] for r in rows] == [1, 1]
+
+
+@pytest.mark.gpu
+def test_golden_join_simple_gpu(cuda):
+    left = y.Chunk([enc([1, 3, 5, 7, 9]), enc([10, 30, 50, 70, 90])], 5)
+    right = y.Chunk([enc([20, 40, 60, 80]), enc([2, 4, 6, 8])], 4)
+    j = y.Join(right, primary_key_col=1, foreign_key_col=0, value_cols=[1])
+    plan = y.Plan(keys=[y.col(2) % 2],
+                  aggs=[y.agg_sum(y.col(0)), y.agg_sum(y.col(1))], join=j)
+    rows, _ = y.gpu_execute(plan, left.c_device(cuda), max_groups_hint=16,
+                            join_foreign=right.c_device(cuda))
+    assert rows == []
+    left2 = y.Chunk([enc([1, 1])], 2)
+    right2 = y.Chunk([enc([2, 1])], 2)
+    j2 = y.Join(right2, primary_key_col=0, foreign_key_col=0, value_cols=[])
+    plan2 = y.Plan(projects=[y.col(0)], join=j2)
+    rows2, _ = y.gpu_execute(plan2, left2.c_device(cuda),
+                             join_foreign=right2.c_device(cuda))
+    assert [r[0] for r in rows2] == [1, 1]
