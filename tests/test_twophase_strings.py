@@ -245,3 +245,32 @@ def test_gpu_string_states_merged_by_oracle(cuda):
     want, _ = y.oracle_execute(plan(), concat_raws(
         [make_raw(50), make_raw(51)]))
     approx_rows(sorted(union, key=rkey), sorted(want, key=rkey))
+
+
+def test_coordinator_combine_string_partitions():
+    """coordinator tail over STRING-keyed front outputs: key-disjoint
+    concatenation + a global ORDER BY sum LIMIT applied by the combiner
+    (the string merge emits full group sets, so the global sort is exact)"""
+    world = 3
+    raws = [make_raw(70 + r, n=4000, nkeys=61, with_null=False)
+            for r in range(world)]
+    shards = [chunk_of_raw(r) for r in raws]
+    outs = [y.oracle_partial_str(plan(), s, world) for s in shards]
+    results = []
+    for p in range(world):
+        segs = []
+        for r in range(world):
+            states, counts, pool, pbytes = outs[r]
+            rbase = sum(counts[:p])
+            bbase = sum(pbytes[:p])
+            segs.append(([states[rbase + i] for i in range(counts[p])],
+                         counts[p], pool[bbase:bbase + pbytes[p]], pbytes[p]))
+        results.append(y.oracle_merge_str(plan(), segs))
+    oplan = y.Plan(keys=[y.col(0)], aggs=[y.agg_sum(y.col(1)), y.agg_sum1()],
+                   order_by=[(1, True)], limit=9)
+    combined = y.coordinate_results(oplan, results)
+    big = concat_raws(raws)
+    want, _ = y.oracle_execute(oplan, big)
+    assert [r[1] for r in combined] == pytest.approx([r[1] for r in want],
+                                                     rel=1e-9)
+    assert len(combined) == 9
