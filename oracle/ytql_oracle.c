@@ -1160,6 +1160,18 @@ int yto_execute(const YtPlan* plan, const YtChunk* chunk,
     if (plan->join) {
         const YtJoin* J = plan->join;
         const YtChunk* fc = J->foreign;
+        {
+            int fkt = fc->columns[J->foreign_key_col].value_type;
+            int pkt = (J->primary_key_col >= 0 && J->primary_key_col < ncols)
+                ? types[J->primary_key_col] : YT_VT_INT64;
+            if (fkt == YT_VT_STRING || fkt == YT_VT_DOUBLE ||
+                pkt == YT_VT_STRING || pkt == YT_VT_DOUBLE) {
+                set_err(errbuf, errlen,
+                        "join: int64/uint64/boolean key columns this round");
+                rc = YT_ERR_UNSUPPORTED;
+                goto done;
+            }
+        }
         int64_t fn = fc->row_count;
         int64_t* fkey = malloc(sizeof(int64_t) * (fn ? fn : 1));
         uint8_t* fknull = malloc(fn ? fn : 1);

@@ -227,3 +227,15 @@ def test_golden_join_simple_gpu(cuda):
     rows2, _ = y.gpu_execute(plan2, left2.c_device(cuda),
                              join_foreign=right2.c_device(cuda))
     assert [r[0] for r in rows2] == [1, 1]
+
+
+# JoinNonPrefixColumns (:5659-5703) joins on a STRING key — a documented
+# refusal this round (int64/uint64/boolean join keys only), asserted loud
+def test_golden_join_string_key_refused():
+    left = y.Chunk([y.encode_string(["a", "b", "c"]),
+                    y.encode_string([None, None, None])], 3)
+    right = y.Chunk([enc([1, 2, 3]), y.encode_string(["a", "b", "c"])], 3)
+    j = y.Join(right, primary_key_col=0, foreign_key_col=1, value_cols=[0])
+    plan = y.Plan(projects=[y.col(0), y.col(2)], join=j)
+    with pytest.raises(RuntimeError, match="key columns"):
+        y.oracle_execute(plan, left)
