@@ -126,7 +126,7 @@ def model(kw, arrays, n):
     return out
 
 
-@pytest.mark.parametrize("seed", range(30))
+@pytest.mark.parametrize("seed", range(40))
 def test_oracle_vs_python_model(seed):
     rng = np.random.default_rng([20260916, seed])
     n = int(rng.choice([97, 1500]))
@@ -232,7 +232,7 @@ def finish_model(kw, rows, kc):
     return rows
 
 
-@pytest.mark.parametrize("seed", range(25))
+@pytest.mark.parametrize("seed", range(35))
 def test_oracle_vs_python_model_finish(seed):
     """having / WITH TOTALS (both modes) / ORDER BY..LIMIT over grouped
     output, cross-checked against the python restatement of the finish

@@ -51,7 +51,7 @@ def gen(rng, n, kind):
     return rows
 
 
-@pytest.mark.parametrize("seed", range(18))
+@pytest.mark.parametrize("seed", range(24))
 def test_versioned_oracle_fuzz(seed):
     rng = np.random.default_rng([20260917, seed])
     kind = ["int", "double", "str"][seed % 3]
